@@ -1,0 +1,167 @@
+"""ctypes bindings for the arroyo-amd C ABI (include/arroyo_amd_types.h).
+
+Generic over the implementing library: the product HIP library
+(libarroyo_amd.so, prefix ``arroyo_amd_``) and the CPU oracle
+(oracle/liboracle.so, prefix ``oracle_``) export the same function set; tests
+drive both through :class:`WindowOp` and compare.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+NS = 10**9
+U64MAX = 2**64 - 1
+
+COUNT, SUM, MIN, MAX, AVG = 0, 1, 2, 3, 4
+
+
+class AmdWindowConfig(ctypes.Structure):
+    _fields_ = [
+        ("width_nanos", ctypes.c_uint64),
+        ("slide_nanos", ctypes.c_uint64),
+        ("is_tumbling", ctypes.c_int32),
+        ("n_keys", ctypes.c_int32),
+        ("n_aggs", ctypes.c_int32),
+        ("agg_ops", ctypes.c_int32 * 8),
+        ("agg_col", ctypes.c_int32 * 8),
+        ("n_value_cols", ctypes.c_int32),
+        ("log2_capacity", ctypes.c_uint32),
+        ("ring_panes", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+class AmdOutBatch(ctypes.Structure):
+    _fields_ = [
+        ("n_rows", ctypes.c_int64),
+        ("n_cols", ctypes.c_int32),
+        ("cols", ctypes.POINTER(ctypes.c_void_p)),
+        ("is_f64", ctypes.POINTER(ctypes.c_int32)),
+        ("on_device", ctypes.c_int32),
+    ]
+
+
+def make_config(width_ns, slide_ns, aggs, n_keys=1, n_value_cols=0,
+                is_tumbling=False, log2_capacity=20, ring_panes=64,
+                device=0, emit_to_host=True):
+    """aggs: list of (op, value_col_index); value_col_index -1 for COUNT(*)."""
+    cfg = AmdWindowConfig()
+    cfg.width_nanos = width_ns
+    cfg.slide_nanos = width_ns if is_tumbling else slide_ns
+    cfg.is_tumbling = 1 if is_tumbling else 0
+    cfg.n_keys = n_keys
+    cfg.n_aggs = len(aggs)
+    for i, (op, col) in enumerate(aggs):
+        cfg.agg_ops[i] = op
+        cfg.agg_col[i] = col
+    cfg.n_value_cols = n_value_cols
+    cfg.log2_capacity = log2_capacity
+    cfg.ring_panes = ring_panes
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+def bind(lib, prefix):
+    fn = {}
+    g = lambda n: getattr(lib, prefix + n)
+    fn["create"] = g("create")
+    fn["create"].restype = ctypes.c_void_p
+    fn["create"].argtypes = [ctypes.POINTER(AmdWindowConfig)]
+    fn["process_batch"] = g("process_batch")
+    fn["process_batch"].restype = ctypes.c_int
+    fn["process_batch"].argtypes = [ctypes.c_void_p,
+                                    ctypes.POINTER(ctypes.c_void_p),
+                                    ctypes.c_int32, ctypes.c_int64]
+    fn["handle_watermark"] = g("handle_watermark")
+    fn["handle_watermark"].restype = ctypes.c_int
+    fn["handle_watermark"].argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                       ctypes.POINTER(AmdOutBatch)]
+    fn["checkpoint_drain"] = g("checkpoint_drain")
+    fn["checkpoint_drain"].restype = ctypes.c_int
+    fn["checkpoint_drain"].argtypes = [ctypes.c_void_p,
+                                       ctypes.POINTER(AmdOutBatch)]
+    fn["free_out"] = g("free_out")
+    fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+    fn["destroy"] = g("destroy")
+    fn["destroy"].argtypes = [ctypes.c_void_p]
+    fn["last_error"] = g("last_error")
+    fn["last_error"].restype = ctypes.c_char_p
+    fn["last_error"].argtypes = [ctypes.c_void_p]
+    return fn
+
+
+def _out_to_numpy(out):
+    """Copy an AmdOutBatch (host memory) into numpy arrays."""
+    n = out.n_rows
+    cols = []
+    for i in range(out.n_cols):
+        dt = np.float64 if out.is_f64[i] else np.int64
+        if n == 0:
+            cols.append(np.empty(0, dtype=dt))
+            continue
+        buf = ctypes.cast(out.cols[i], ctypes.POINTER(ctypes.c_int64 * n))
+        arr = np.frombuffer(bytearray(bytes(buf.contents)), dtype=dt).copy()
+        cols.append(arr)
+    return cols
+
+
+class WindowOp:
+    """One window-aggregate operator instance behind the C ABI.
+
+    Mirrors the ArrowOperator calling convention
+    (crates/arroyo-operator/src/operator.rs:1144-1257): process_batch
+    ingests an owned batch; handle_watermark may emit result batches;
+    single-threaded per handle."""
+
+    def __init__(self, lib, prefix, cfg):
+        self._fn = bind(lib, prefix)
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{prefix}create failed (invalid config)")
+
+    def process_batch(self, cols):
+        """cols: list of np.int64 arrays [keys..., values..., _timestamp]."""
+        n_rows = len(cols[0]) if cols else 0
+        arr = (ctypes.c_void_p * len(cols))()
+        keep = []
+        for i, c in enumerate(cols):
+            c = np.ascontiguousarray(c, dtype=np.int64)
+            keep.append(c)
+            arr[i] = c.ctypes.data_as(ctypes.c_void_p).value
+        rc = self._fn["process_batch"](self._h, arr, len(cols), n_rows)
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        return keep  # keep references alive through the call
+
+    def handle_watermark(self, wm):
+        out = AmdOutBatch()
+        rc = self._fn["handle_watermark"](self._h, wm, ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def checkpoint_drain(self):
+        out = AmdOutBatch()
+        rc = self._fn["checkpoint_drain"](self._h, ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

@@ -1,0 +1,73 @@
+/* Shared POD types for the arroyo-amd C ABI.
+ *
+ * These mirror the configuration and batch contract of the reference's
+ * windowed-aggregate operators (ArroyoSystems/arroyo):
+ *   - AmdWindowConfig <-> api::SlidingWindowAggregateOperator /
+ *     api::TumblingWindowAggregateOperator protobufs decoded at
+ *     crates/arroyo-worker/src/arrow/sliding_aggregating_window.rs:449-528 and
+ *     tumbling_aggregating_window.rs:111-201 (width/slide micros, input
+ *     schema with key columns first, partial/final aggregation plans).  The
+ *     serialized DataFusion plans are replaced by an explicit aggregate spec
+ *     (op + input column), which is the information content of the plans for
+ *     the supported aggregate set.
+ *   - batch columns <-> Arrow RecordBatch per ArroyoSchema
+ *     (crates/arroyo-rpc/src/df.rs:24-30: schema + timestamp_index +
+ *     key_indices); key columns first, `_timestamp` (ns) last.
+ *
+ * Round-1 scope: fixed-width i64 key (0 or 1 key columns), i64 value
+ * columns, non-nullable.  Timestamps are u64 nanoseconds since the epoch;
+ * the end-of-stream watermark is UINT64_MAX (arroyo-worker watermark
+ * generator on_close, watermark_generator.rs:131-148).
+ */
+#ifndef ARROYO_AMD_TYPES_H
+#define ARROYO_AMD_TYPES_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum AmdAggOp {
+    AMD_AGG_COUNT = 0, /* partial: count        final: sum of counts  */
+    AMD_AGG_SUM   = 1, /* partial: sum (i64)    final: sum            */
+    AMD_AGG_MIN   = 2, /* partial: min          final: min            */
+    AMD_AGG_MAX   = 3, /* partial: max          final: max            */
+    AMD_AGG_AVG   = 4  /* partial: (count,sum f64)  final: sum/divide */
+};
+
+#define AMD_MAX_AGGS 8
+
+typedef struct {
+    uint64_t width_nanos;
+    uint64_t slide_nanos;     /* ignored when is_tumbling */
+    int32_t  is_tumbling;     /* 1: TumblingAggregatingWindowFunc semantics */
+    int32_t  n_keys;          /* 0 or 1 (i64 key columns, first) */
+    int32_t  n_aggs;
+    int32_t  agg_ops[AMD_MAX_AGGS];
+    int32_t  agg_col[AMD_MAX_AGGS];  /* value-column index; -1 for COUNT(*) */
+    int32_t  n_value_cols;    /* i64 value columns between keys and _timestamp */
+    uint32_t log2_capacity;   /* hash slots per pane (GPU path) */
+    uint32_t ring_panes;      /* live-pane ring size (GPU path, power of 2) */
+    int32_t  device;          /* HIP device ordinal (GPU path) */
+    int32_t  emit_to_host;    /* GPU path: 1 = outputs copied to host memory */
+} AmdWindowConfig;
+
+/* Output batch, allocated by the callee; free with *_free_out.
+ * Column order: [key (if n_keys)], agg outputs (one column per agg),
+ * window_start, window_end, _timestamp.  All columns are 8-byte elements;
+ * is_f64[i] marks double columns (AVG outputs).  on_device=1 means `cols`
+ * are HIP device pointers. */
+typedef struct {
+    int64_t   n_rows;
+    int32_t   n_cols;
+    void    **cols;
+    int32_t  *is_f64;
+    int32_t   on_device;
+} AmdOutBatch;
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* ARROYO_AMD_TYPES_H */

@@ -1,0 +1,520 @@
+/* CPU restatement of Arroyo's sliding/tumbling window-aggregate operator.
+ *
+ * TEST INFRASTRUCTURE + CPU BASELINE ONLY.  Only tests/, __graft_entry__'s
+ * smoke() and bench.py's cpu_baseline leg may load this library; the product
+ * path is the HIP library (arroyo_amd/csrc) and fails loudly if its extension
+ * is missing -- it never falls back to this code.
+ *
+ * Restated, statement for statement, from the reference
+ * (ArroyoSystems/arroyo, mounted read-only at /root/reference during the
+ * build; citations are file:line into that tree):
+ *   sliding: crates/arroyo-worker/src/arrow/sliding_aggregating_window.rs
+ *     bin_start :90-99, should_advance :102-113, advance :115-210,
+ *     process_batch :598-674 (late drop :631-633)
+ *   tumbling: crates/arroyo-worker/src/arrow/tumbling_aggregating_window.rs
+ *     process_batch :250-319, handle_watermark :321-392
+ *   state table: crates/arroyo-state/src/tables/expiring_time_key_map.rs
+ *     :826-929 (flush retention cutoff, exact-timestamp expire, get_min_time);
+ *     retention = width (timestamp_table_config, sliding :739-752)
+ *   output projection: crates/arroyo-planner/src/extension/aggregate.rs
+ *     :292-390 (window=(bin_start, bin_start+width), _timestamp=end-1ns)
+ *   partial/final aggregate split: crates/arroyo-planner/src/builder.rs
+ *     :135-199 over DataFusion 48.0.1 (ArroyoSystems fork 48.0.1/arroyo,
+ *     not vendored in the reference); aggregate semantics pinned by the
+ *     reference's golden vectors (tests/golden/, see oracle/gen_golden.py).
+ *
+ * The reference cannot be compiled here (Rust workspace; no rustc/cargo in
+ * the container), so parity is pinned by its golden test vectors instead:
+ * tests/test_oracle_golden.py checks this oracle against four of them.
+ *
+ * Pipeline-level semantics (watermark generation, batching) live in the test
+ * harness (arroyo_amd/pipeline.py), not here: this library is the operator.
+ */
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+#include <stdio.h>
+
+#include "../include/arroyo_amd_types.h"
+
+#define ORACLE_API __attribute__((visibility("default")))
+
+/* ------------------------------------------------------------------ */
+/* aggregate state: per entry, n_aggs x 2 u64 words.
+ * COUNT: w0=count.  SUM: w0=sum(i64).  MIN/MAX: w0=value.
+ * AVG: w0=count, w1=bits(double sum).                                  */
+
+static inline double bits_to_d(int64_t b) { double d; memcpy(&d, &b, 8); return d; }
+static inline int64_t d_to_bits(double d) { int64_t b; memcpy(&b, &d, 8); return b; }
+
+typedef struct {
+    int cap;          /* power of two */
+    int64_t n;
+    int64_t *keys;
+    uint8_t *used;
+    int64_t *st;      /* [cap][n_aggs][2] */
+} Table;
+
+typedef struct {
+    AmdWindowConfig cfg;
+    /* open bins (execs) and closed panes (TieredRecordBatchHolder),
+     * sorted by bin */
+    int n_open, cap_open;
+    uint64_t *open_bins; Table **open_tables;
+    int n_closed, cap_closed;
+    uint64_t *closed_bins; Table **closed_tables;
+    /* state-table bins (ExpiringTimeKeyView keys), sorted */
+    int n_table, cap_table;
+    uint64_t *table_bins;
+    /* sliding state machine (sliding_aggregating_window.rs:63-73) */
+    int state;               /* 0 NoData, 1 OnlyBufferedData, 2 InMemoryData */
+    uint64_t earliest, next;
+    int has_wm; uint64_t wm;  /* last present watermark */
+    /* output accumulator */
+    int64_t out_rows, out_cap;
+    int out_cols;
+    int64_t **out;            /* [out_cols][out_cap] */
+    char err[256];
+} Op;
+
+static uint64_t hash64(uint64_t x) {
+    /* splitmix64 finalizer */
+    x += 0x9e3779b97f4a7c15ULL;
+    x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ULL;
+    x = (x ^ (x >> 27)) * 0x94d049bb133111ebULL;
+    return x ^ (x >> 31);
+}
+
+static Table *table_new(int cap_log2, int n_aggs) {
+    Table *t = calloc(1, sizeof(Table));
+    t->cap = 1 << cap_log2;
+    t->keys = malloc((size_t)t->cap * 8);
+    t->used = calloc((size_t)t->cap, 1);
+    t->st = malloc((size_t)t->cap * n_aggs * 16);
+    return t;
+}
+
+static void table_free(Table *t) {
+    if (!t) return;
+    free(t->keys); free(t->used); free(t->st); free(t);
+}
+
+static void st_init(const AmdWindowConfig *c, int64_t *s) {
+    for (int i = 0; i < c->n_aggs; i++) {
+        switch (c->agg_ops[i]) {
+        case AMD_AGG_COUNT: s[2*i] = 0; break;
+        case AMD_AGG_SUM:   s[2*i] = 0; break;
+        case AMD_AGG_MIN:   s[2*i] = INT64_MAX; break;
+        case AMD_AGG_MAX:   s[2*i] = INT64_MIN; break;
+        case AMD_AGG_AVG:   s[2*i] = 0; s[2*i+1] = d_to_bits(0.0); break;
+        }
+    }
+}
+
+static void table_grow(const AmdWindowConfig *c, Table *t);
+
+static int64_t *table_slot(const AmdWindowConfig *c, Table *t, int64_t key) {
+    if (t->n * 10 >= (int64_t)t->cap * 7) table_grow(c, t);
+    uint64_t m = (uint64_t)t->cap - 1;
+    uint64_t i = hash64((uint64_t)key) & m;
+    while (t->used[i] && t->keys[i] != key) i = (i + 1) & m;
+    if (!t->used[i]) {
+        t->used[i] = 1; t->keys[i] = key; t->n++;
+        st_init(c, &t->st[i * c->n_aggs * 2]);
+    }
+    return &t->st[i * c->n_aggs * 2];
+}
+
+static void table_grow(const AmdWindowConfig *c, Table *t) {
+    int ocap = t->cap;
+    int64_t *ok = t->keys; uint8_t *ou = t->used; int64_t *os = t->st;
+    t->cap <<= 1; t->n = 0;
+    t->keys = malloc((size_t)t->cap * 8);
+    t->used = calloc((size_t)t->cap, 1);
+    t->st = malloc((size_t)t->cap * c->n_aggs * 16);
+    for (int i = 0; i < ocap; i++) {
+        if (!ou[i]) continue;
+        int64_t *s = table_slot(c, t, ok[i]);
+        memcpy(s, &os[(size_t)i * c->n_aggs * 2], (size_t)c->n_aggs * 16);
+        /* table_slot counted it as new and re-inited; restore by copy above */
+    }
+    free(ok); free(ou); free(os);
+}
+
+static void st_update(const AmdWindowConfig *c, int64_t *s,
+                      const int64_t *const *vcols, int64_t row) {
+    for (int i = 0; i < c->n_aggs; i++) {
+        int col = c->agg_col[i];
+        switch (c->agg_ops[i]) {
+        case AMD_AGG_COUNT: s[2*i]++; break;
+        case AMD_AGG_SUM:   s[2*i] += vcols[col][row]; break;
+        case AMD_AGG_MIN: { int64_t v = vcols[col][row]; if (v < s[2*i]) s[2*i] = v; } break;
+        case AMD_AGG_MAX: { int64_t v = vcols[col][row]; if (v > s[2*i]) s[2*i] = v; } break;
+        case AMD_AGG_AVG:
+            s[2*i]++;
+            s[2*i+1] = d_to_bits(bits_to_d(s[2*i+1]) + (double)vcols[col][row]);
+            break;
+        }
+    }
+}
+
+static void st_merge(const AmdWindowConfig *c, int64_t *a, const int64_t *b) {
+    for (int i = 0; i < c->n_aggs; i++) {
+        switch (c->agg_ops[i]) {
+        case AMD_AGG_COUNT: a[2*i] += b[2*i]; break;
+        case AMD_AGG_SUM:   a[2*i] += b[2*i]; break;
+        case AMD_AGG_MIN:   if (b[2*i] < a[2*i]) a[2*i] = b[2*i]; break;
+        case AMD_AGG_MAX:   if (b[2*i] > a[2*i]) a[2*i] = b[2*i]; break;
+        case AMD_AGG_AVG:
+            a[2*i] += b[2*i];
+            a[2*i+1] = d_to_bits(bits_to_d(a[2*i+1]) + bits_to_d(b[2*i+1]));
+            break;
+        }
+    }
+}
+
+/* sorted (bins,tables) vector helpers */
+static int vec_find(const uint64_t *bins, int n, uint64_t b) {
+    int lo = 0, hi = n;
+    while (lo < hi) { int mid = (lo + hi) / 2; if (bins[mid] < b) lo = mid + 1; else hi = mid; }
+    return lo; /* insertion point; bins[lo]==b if present */
+}
+
+static Table *panes_get(int *n, int *cap, uint64_t **bins, Table ***tables,
+                        uint64_t b, const AmdWindowConfig *c, int create) {
+    int i = vec_find(*bins, *n, b);
+    if (i < *n && (*bins)[i] == b) return (*tables)[i];
+    if (!create) return NULL;
+    if (*n == *cap) {
+        *cap = *cap ? *cap * 2 : 16;
+        *bins = realloc(*bins, (size_t)*cap * 8);
+        *tables = realloc(*tables, (size_t)*cap * sizeof(Table *));
+    }
+    memmove(*bins + i + 1, *bins + i, (size_t)(*n - i) * 8);
+    memmove(*tables + i + 1, *tables + i, (size_t)(*n - i) * sizeof(Table *));
+    (*bins)[i] = b;
+    Table *t = table_new(8, c->n_aggs);
+    (*tables)[i] = t;
+    (*n)++;
+    return t;
+}
+
+static Table *panes_remove(int *n, uint64_t *bins, Table **tables, uint64_t b) {
+    int i = vec_find(bins, *n, b);
+    if (i >= *n || bins[i] != b) return NULL;
+    Table *t = tables[i];
+    memmove(bins + i, bins + i, 0);
+    memmove(bins + i, bins + i + 1, (size_t)(*n - i - 1) * 8);
+    memmove(tables + i, tables + i + 1, (size_t)(*n - i - 1) * sizeof(Table *));
+    (*n)--;
+    return t;
+}
+
+static void tset_add(Op *o, uint64_t b) {
+    int i = vec_find(o->table_bins, o->n_table, b);
+    if (i < o->n_table && o->table_bins[i] == b) return;
+    if (o->n_table == o->cap_table) {
+        o->cap_table = o->cap_table ? o->cap_table * 2 : 16;
+        o->table_bins = realloc(o->table_bins, (size_t)o->cap_table * 8);
+    }
+    memmove(o->table_bins + i + 1, o->table_bins + i, (size_t)(o->n_table - i) * 8);
+    o->table_bins[i] = b;
+    o->n_table++;
+}
+
+static void tset_remove_exact(Op *o, uint64_t b) {
+    int i = vec_find(o->table_bins, o->n_table, b);
+    if (i < o->n_table && o->table_bins[i] == b) {
+        memmove(o->table_bins + i, o->table_bins + i + 1,
+                (size_t)(o->n_table - i - 1) * 8);
+        o->n_table--;
+    }
+}
+
+static void tset_retain_ge(Op *o, uint64_t cutoff) {
+    int i = vec_find(o->table_bins, o->n_table, cutoff);
+    if (i > 0) {
+        memmove(o->table_bins, o->table_bins + i, (size_t)(o->n_table - i) * 8);
+        o->n_table -= i;
+    }
+}
+
+static inline uint64_t bin_of(uint64_t ts, uint64_t w) {
+    return w ? ts - ts % w : ts;
+}
+
+/* output accumulation: columns [key?, aggs..., win_start, win_end, _ts] */
+static void out_reserve(Op *o, int64_t add) {
+    if (o->out_rows + add <= o->out_cap) return;
+    int64_t ncap = o->out_cap ? o->out_cap : 1024;
+    while (ncap < o->out_rows + add) ncap *= 2;
+    for (int i = 0; i < o->out_cols; i++)
+        o->out[i] = realloc(o->out[i], (size_t)ncap * 8);
+    o->out_cap = ncap;
+}
+
+static void emit_table(Op *o, Table *t, uint64_t ws, uint64_t we) {
+    const AmdWindowConfig *c = &o->cfg;
+    out_reserve(o, t->n);
+    for (int i = 0; i < t->cap; i++) {
+        if (!t->used[i]) continue;
+        int64_t r = o->out_rows++;
+        int col = 0;
+        if (c->n_keys) o->out[col++][r] = t->keys[i];
+        const int64_t *s = &t->st[(size_t)i * c->n_aggs * 2];
+        for (int a = 0; a < c->n_aggs; a++) {
+            if (c->agg_ops[a] == AMD_AGG_AVG) {
+                double v = s[2*a] ? bits_to_d(s[2*a+1]) / (double)s[2*a] : 0.0;
+                o->out[col++][r] = d_to_bits(v);
+            } else {
+                o->out[col++][r] = s[2*a];
+            }
+        }
+        o->out[col++][r] = (int64_t)ws;
+        o->out[col++][r] = (int64_t)we;
+        o->out[col++][r] = (int64_t)(we - 1);
+    }
+}
+
+/* advance(): sliding_aggregating_window.rs:115-210 */
+static void advance(Op *o) {
+    const AmdWindowConfig *c = &o->cfg;
+    uint64_t b = (o->state == 1) ? o->earliest : o->next;
+    uint64_t E = b + c->slide_nanos;
+
+    /* partial_table.flush(Some(bin_end)): retention(=width) cutoff :131 */
+    if (E >= c->width_nanos) tset_retain_ge(o, E - c->width_nanos);
+
+    Table *pane = panes_remove(&o->n_open, o->open_bins, o->open_tables, b);
+    if (pane) {
+        Table *tgt = panes_get(&o->n_closed, &o->cap_closed, &o->closed_bins,
+                               &o->closed_tables, b, c, 0);
+        if (!tgt) {
+            /* move the pane wholesale */
+            int i = vec_find(o->closed_bins, o->n_closed, b);
+            if (o->n_closed == o->cap_closed) {
+                o->cap_closed = o->cap_closed ? o->cap_closed * 2 : 16;
+                o->closed_bins = realloc(o->closed_bins, (size_t)o->cap_closed * 8);
+                o->closed_tables = realloc(o->closed_tables,
+                                           (size_t)o->cap_closed * sizeof(Table *));
+            }
+            memmove(o->closed_bins + i + 1, o->closed_bins + i,
+                    (size_t)(o->n_closed - i) * 8);
+            memmove(o->closed_tables + i + 1, o->closed_tables + i,
+                    (size_t)(o->n_closed - i) * sizeof(Table *));
+            o->closed_bins[i] = b;
+            o->closed_tables[i] = pane;
+            o->n_closed++;
+        } else {
+            for (int i = 0; i < pane->cap; i++)
+                if (pane->used[i])
+                    st_merge(c, table_slot(c, tgt, pane->keys[i]),
+                             &pane->st[(size_t)i * c->n_aggs * 2]);
+            table_free(pane);
+        }
+        tset_add(o, b); /* partial_table.insert :151 */
+    }
+    /* expire_timestamp(bin_end - width + slide), exact key :160 */
+    if (E + c->slide_nanos >= c->width_nanos)
+        tset_remove_exact(o, E + c->slide_nanos - c->width_nanos);
+
+    /* merge closed panes in [E-width, E) :161-196 */
+    uint64_t lo = (E >= c->width_nanos) ? E - c->width_nanos : 0;
+    Table *merged = table_new(8, c->n_aggs);
+    for (int p = 0; p < o->n_closed; p++) {
+        if (o->closed_bins[p] < lo || o->closed_bins[p] >= E) continue;
+        Table *t = o->closed_tables[p];
+        for (int i = 0; i < t->cap; i++)
+            if (t->used[i])
+                st_merge(c, table_slot(c, merged, t->keys[i]),
+                         &t->st[(size_t)i * c->n_aggs * 2]);
+    }
+    /* delete_before(bin_end + slide - width) :173-174 */
+    uint64_t del = (E + c->slide_nanos >= c->width_nanos)
+                       ? E + c->slide_nanos - c->width_nanos : 0;
+    int keep = 0;
+    for (int p = 0; p < o->n_closed; p++) {
+        if (o->closed_bins[p] >= del) {
+            o->closed_bins[keep] = o->closed_bins[p];
+            o->closed_tables[keep] = o->closed_tables[p];
+            keep++;
+        } else {
+            table_free(o->closed_tables[p]);
+        }
+    }
+    o->n_closed = keep;
+
+    emit_table(o, merged, E - c->width_nanos, E);
+    table_free(merged);
+
+    /* state transition :176-187 */
+    if (o->n_closed == 0) {
+        if (o->n_table > 0) {
+            o->state = 1;
+            o->earliest = bin_of(o->table_bins[0], c->slide_nanos);
+        } else {
+            o->state = 0;
+        }
+    } else {
+        o->state = 2;
+        o->next = E;
+    }
+}
+
+/* ------------------------------------------------------------------ */
+ORACLE_API void *oracle_create(const AmdWindowConfig *cfg) {
+    if (!cfg || cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS ||
+        cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->width_nanos == 0)
+        return NULL;
+    Op *o = calloc(1, sizeof(Op));
+    o->cfg = *cfg;
+    if (o->cfg.is_tumbling) o->cfg.slide_nanos = o->cfg.width_nanos;
+    o->out_cols = cfg->n_keys + cfg->n_aggs + 3;
+    o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
+    return o;
+}
+
+ORACLE_API const char *oracle_last_error(void *h) {
+    return h ? ((Op *)h)->err : "null handle";
+}
+
+ORACLE_API int oracle_process_batch(void *h, const int64_t *const *cols,
+                                    int32_t n_cols, int64_t n_rows) {
+    Op *o = h;
+    const AmdWindowConfig *c = &o->cfg;
+    if (n_cols != c->n_keys + c->n_value_cols + 1) {
+        snprintf(o->err, sizeof o->err, "expected %d cols, got %d",
+                 c->n_keys + c->n_value_cols + 1, n_cols);
+        return 1;
+    }
+    const int64_t *kcol = c->n_keys ? cols[0] : NULL;
+    const int64_t *const *vcols = cols + c->n_keys;
+    const int64_t *ts = cols[n_cols - 1];
+    uint64_t wmb = o->has_wm ? bin_of(o->wm, c->slide_nanos) : 0;
+    uint64_t cached_bin = UINT64_MAX; Table *cached = NULL;
+    for (int64_t r = 0; r < n_rows; r++) {
+        uint64_t b = bin_of((uint64_t)ts[r], c->slide_nanos);
+        if (o->has_wm && b < wmb) continue; /* late drop */
+        if (!c->is_tumbling) {
+            if (o->state == 0) { o->state = 1; o->earliest = b; }
+            else if (o->state == 1 && b < o->earliest) o->earliest = b;
+        }
+        if (b != cached_bin) {
+            cached = panes_get(&o->n_open, &o->cap_open, &o->open_bins,
+                               &o->open_tables, b, c, 1);
+            cached_bin = b;
+        }
+        int64_t key = kcol ? kcol[r] : 0;
+        st_update(c, table_slot(c, cached, key), vcols, r);
+    }
+    return 0;
+}
+
+ORACLE_API int oracle_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
+    Op *o = h;
+    const AmdWindowConfig *c = &o->cfg;
+    o->has_wm = 1;
+    o->wm = wm;
+    if (c->is_tumbling) {
+        /* tumbling_aggregating_window.rs:321-392 */
+        uint64_t wb = bin_of(wm, c->width_nanos);
+        while (o->n_open > 0 && o->open_bins[0] < wb) {
+            uint64_t b = o->open_bins[0];
+            Table *pane = panes_remove(&o->n_open, o->open_bins, o->open_tables, b);
+            emit_table(o, pane, b, b + c->width_nanos);
+            table_free(pane);
+        }
+    } else {
+        while (o->state != 0) {
+            uint64_t base = (o->state == 1) ? o->earliest : o->next;
+            if (!(base + c->slide_nanos <= bin_of(wm, c->slide_nanos))) break;
+            advance(o);
+        }
+    }
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = o->out_rows;
+        out->n_cols = o->out_cols;
+        out->cols = calloc((size_t)o->out_cols, sizeof(void *));
+        out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
+            if (o->out_rows)
+                memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
+        }
+        for (int a = 0; a < c->n_aggs; a++)
+            if (c->agg_ops[a] == AMD_AGG_AVG)
+                out->is_f64[c->n_keys + a] = 1;
+        o->out_rows = 0;
+    }
+    return 0;
+}
+
+/* handle_checkpoint (sliding :693-737): drain open bins' partial states into
+ * the state table.  Output columns: [key?, partial state words
+ * (1 per agg, 2 for AVG)..., bin _timestamp]. */
+ORACLE_API int oracle_checkpoint_drain(void *h, AmdOutBatch *out) {
+    Op *o = h;
+    const AmdWindowConfig *c = &o->cfg;
+    int swords = 0;
+    for (int a = 0; a < c->n_aggs; a++)
+        swords += (c->agg_ops[a] == AMD_AGG_AVG) ? 2 : 1;
+    int ncols = c->n_keys + swords + 1;
+    int64_t total = 0;
+    for (int p = 0; p < o->n_open; p++) total += o->open_tables[p]->n;
+    memset(out, 0, sizeof *out);
+    out->n_rows = total;
+    out->n_cols = ncols;
+    out->cols = calloc((size_t)ncols, sizeof(void *));
+    out->is_f64 = calloc((size_t)ncols, sizeof(int32_t));
+    for (int i = 0; i < ncols; i++)
+        out->cols[i] = malloc((size_t)(total ? total : 1) * 8);
+    {
+        int col = c->n_keys;
+        for (int a = 0; a < c->n_aggs; a++) {
+            if (c->agg_ops[a] == AMD_AGG_AVG) { out->is_f64[col + 1] = 1; col += 2; }
+            else col += 1;
+        }
+    }
+    int64_t r = 0;
+    for (int p = 0; p < o->n_open; p++) {
+        Table *t = o->open_tables[p];
+        uint64_t b = o->open_bins[p];
+        tset_add(o, b);
+        for (int i = 0; i < t->cap; i++) {
+            if (!t->used[i]) continue;
+            int col = 0;
+            if (c->n_keys) ((int64_t *)out->cols[col++])[r] = t->keys[i];
+            const int64_t *s = &t->st[(size_t)i * c->n_aggs * 2];
+            for (int a = 0; a < c->n_aggs; a++) {
+                ((int64_t *)out->cols[col++])[r] = s[2*a];
+                if (c->agg_ops[a] == AMD_AGG_AVG)
+                    ((int64_t *)out->cols[col++])[r] = s[2*a+1];
+            }
+            ((int64_t *)out->cols[col])[r] = (int64_t)b;
+            r++;
+        }
+    }
+    return 0;
+}
+
+ORACLE_API void oracle_free_out(AmdOutBatch *out) {
+    if (!out) return;
+    for (int i = 0; i < out->n_cols; i++) free(out->cols[i]);
+    free(out->cols);
+    free(out->is_f64);
+    memset(out, 0, sizeof *out);
+}
+
+ORACLE_API void oracle_destroy(void *h) {
+    Op *o = h;
+    if (!o) return;
+    for (int p = 0; p < o->n_open; p++) table_free(o->open_tables[p]);
+    for (int p = 0; p < o->n_closed; p++) table_free(o->closed_tables[p]);
+    free(o->open_bins); free(o->open_tables);
+    free(o->closed_bins); free(o->closed_tables);
+    free(o->table_bins);
+    for (int i = 0; i < o->out_cols; i++) free(o->out[i]);
+    free(o->out);
+    free(o);
+}
