@@ -1,0 +1,285 @@
+"""Measurement harness for the arroyo-amd hot path (BASELINE.json metric:
+rows/sec on nexmark q5 sliding-window aggregate).
+
+A "step" is one pass of the hot path over one 64K-row synthetic nexmark bid
+batch already resident in HBM: the fused bin+hash-aggregate update kernel
+(k_update_lds), plus the watermark-driven window firing (merge + compact
+kernels) at the reference's watermark cadence (1/s of event time,
+crates/arroyo-worker/src/arrow/watermark_generator.rs).  Outputs stay
+device-resident (the next pipeline stage's collector consumes them in place);
+the host-visible emission path is covered by tests, not timed here.
+
+Workload = BASELINE.json configs[1]: q5 hot-items, 10s width / 2s slide
+COUNT GROUP BY auction, 64K-row batches (configs[0] is the reference's
+CPU-runnable plumbing case; the others are parity-test cases).
+
+Usage: python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches this under torch.distributed.run, one rank per
+GPU; each rank processes its own 64K-row batch per step and the keyed
+shuffle (device partition kernel + RCCL all-to-all over xGMI) runs inside
+the timed region ("weak" scaling: per-GPU work fixed).
+
+Emits ONE JSON line from rank 0, including:
+  - roofline: dominant-kernel achieved GB/s (algorithmic bytes per launch /
+    HIP-event launch time, measured on the operator's own stream) vs the
+    8 TB/s HBM3E spec peak (/opt/skills/guides/MI355X_MICROARCH.md);
+    algorithmic bytes = 16 B/row (key i64 + _timestamp i64), see DESIGN.md.
+  - cpu_baseline: the CPU oracle (oracle/arroyo_oracle.c, a restatement of
+    the reference's operator — kind "port") timed on this box's host cores
+    over a bounded sample of the same workload, rank 0 at N=1 only.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from arroyo_amd import cabi, nexmark  # noqa: E402
+from arroyo_amd.pipeline import NS  # noqa: E402
+
+BATCH_ROWS = 65536
+EVENTS_PER_SEC = 1_000_000       # event-time rate of the synthetic stream
+BASE_BATCHES = 16                # distinct batches in the replay ring
+WIDTH_S, SLIDE_S = 10, 2
+LOG2_CAPACITY = 19               # ~130K distinct auctions per 2s pane
+RING_PANES = 16
+HBM_PEAK_GBPS = 8000.0           # spec peak (MI355X_MICROARCH.md)
+ALG_BYTES_PER_ROW = 16           # compulsory HBM read: auction i64 + ts i64
+CPU_SAMPLE_ROWS = 64_000_000
+
+
+def op_config(device):
+    return cabi.make_config(
+        width_ns=WIDTH_S * NS, slide_ns=SLIDE_S * NS, n_keys=1,
+        n_value_cols=0, aggs=[(cabi.COUNT, -1)],
+        log2_capacity=LOG2_CAPACITY, ring_panes=RING_PANES,
+        device=device, emit_to_host=False)
+
+
+def gen_stream(seed):
+    """BASE_BATCHES x BATCH_ROWS bid rows; returns (key, ts, span_ns)."""
+    n = BASE_BATCHES * BATCH_ROWS
+    key, ts = nexmark.bids(n, events_per_sec=EVENTS_PER_SEC, seed=seed)
+    # replay period: event time advanced by one full pass of the base stream
+    span = int(((n * nexmark.TOTAL_PROPORTION) // nexmark.BID_PROPORTION)
+               * NS // EVENTS_PER_SEC)
+    return key, ts, span
+
+
+class WatermarkClock:
+    """Reference watermark cadence: emit (max_ts - 1s lateness) at most once
+    per 1s of event time (watermark_generator.rs:150-196)."""
+
+    def __init__(self):
+        self.last = 0
+
+    def maybe(self, batch_max_ts):
+        if batch_max_ts - self.last > NS:
+            self.last = batch_max_ts
+            return batch_max_ts - NS
+        return None
+
+
+def run_gpu(args):
+    import torch
+
+    from arroyo_amd import gpu
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+    torch.cuda.set_device(local_rank)
+    dev = torch.device("cuda", local_rank)
+
+    key, ts, span = gen_stream(seed=42 + rank)
+    d_key = torch.from_numpy(key).to(dev)
+    d_ts = torch.from_numpy(ts).to(dev)
+    # per-batch views into the replay ring
+    views = [(d_key[i * BATCH_ROWS:(i + 1) * BATCH_ROWS],
+              d_ts[i * BATCH_ROWS:(i + 1) * BATCH_ROWS])
+             for i in range(BASE_BATCHES)]
+    # shuffle scratch (N>1): partitioned send buffers + receive buffers
+    if world > 1:
+        sk = torch.empty(BATCH_ROWS, dtype=torch.int64, device=dev)
+        st = torch.empty(BATCH_ROWS, dtype=torch.int64, device=dev)
+        rbuf_cap = (world + 1) * BATCH_ROWS  # hot keys can skew the exchange
+        rk = torch.empty(rbuf_cap, dtype=torch.int64, device=dev)
+        rt = torch.empty(rbuf_cap, dtype=torch.int64, device=dev)
+
+    op = gpu.make_op(op_config(local_rank))
+    wm_lib = gpu.lib()
+    batch_span = span // BASE_BATCHES
+
+    def one_step(step):
+        b = step % BASE_BATCHES
+        cycle = step // BASE_BATCHES
+        ts_off = cycle * span
+        bk, bt = views[b]
+        if world > 1:
+            counts = gpu.partition_device(
+                bk.data_ptr(), 0, bt.data_ptr(), BATCH_ROWS, world,
+                sk.data_ptr(), 0, st.data_ptr())
+            send_splits = [int(c) for c in counts]
+            tsizes = torch.tensor(send_splits, dtype=torch.int64, device=dev)
+            gathered = torch.empty(world * world, dtype=torch.int64,
+                                   device=dev)
+            dist.all_gather_into_tensor(gathered, tsizes)
+            recv_splits = [int(gathered[s * world + rank])
+                           for s in range(world)]
+            n_recv = sum(recv_splits)
+            dist.all_to_all_single(rk[:n_recv], sk,
+                                   output_split_sizes=recv_splits,
+                                   input_split_sizes=send_splits)
+            dist.all_to_all_single(rt[:n_recv], st,
+                                   output_split_sizes=recv_splits,
+                                   input_split_sizes=send_splits)
+            torch.cuda.current_stream().synchronize()
+            if n_recv:
+                op.process_batch_device([rk.data_ptr(), rt.data_ptr()],
+                                        n_recv, ts_off)
+        else:
+            op.process_batch_device([bk.data_ptr(), bt.data_ptr()],
+                                    BATCH_ROWS, ts_off)
+        # event-time clock is global across ranks: same watermark everywhere
+        max_ts = int(ts[(b + 1) * BATCH_ROWS - 1]) + ts_off
+        wm = clock.maybe(max_ts)
+        if wm is not None:
+            rc = wm_lib.arroyo_amd_handle_watermark(
+                op._h, ctypes.c_uint64(wm), None)
+            if rc != 0:
+                raise RuntimeError(op._fn["last_error"](op._h).decode())
+
+    clock = WatermarkClock()
+    torch.cuda.synchronize()
+    for s in range(args.warmup):
+        one_step(s)
+    torch.cuda.synchronize()
+    op.perf()  # reset kernel-time counters after warmup
+
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for s in range(args.warmup, args.warmup + args.steps):
+        one_step(s)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    if dist:
+        import torch.distributed as dist_mod
+        e = torch.tensor([elapsed], device=dev)
+        dist_mod.all_reduce(e, op=dist_mod.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    perf = op.perf()
+    op.close()
+    if dist:
+        dist.destroy_process_group()
+    return elapsed, perf, rank, world
+
+
+def cpu_baseline():
+    """Oracle (CPU restatement, kind 'port') rows/s on a bounded sample of
+    the same workload, single thread."""
+    import oracle
+    from arroyo_amd.pipeline import batches_from_columns
+
+    key, ts = nexmark.bids(CPU_SAMPLE_ROWS, events_per_sec=EVENTS_PER_SEC,
+                           seed=42)
+    op = oracle.make_op(op_config(device=0))
+    clock = WatermarkClock()
+    t0 = time.perf_counter()
+    for cols in batches_from_columns([key, ts], BATCH_ROWS):
+        op.process_batch(cols)
+        wm = clock.maybe(int(cols[-1][-1]))
+        if wm is not None:
+            op.handle_watermark(wm)
+    t1 = time.perf_counter()
+    op.close()
+    return {
+        "value": CPU_SAMPLE_ROWS / (t1 - t0),
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{CPU_SAMPLE_ROWS} rows ({t1 - t0:.1f}s) of the same "
+                  f"nexmark q5 stream, oracle/arroyo_oracle.c single-thread",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=400)
+    ap.add_argument("--warmup", type=int, default=100)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    elapsed, perf, rank, world = run_gpu(args)
+    if rank != 0:
+        return
+
+    n_gpus = world if world > 1 else args.gpus
+    total_rows = args.steps * BATCH_ROWS * n_gpus
+    value = total_rows / elapsed
+
+    launches = max(perf["launches"], 1)
+    avg_launch_ms = perf["update_ms"] / launches
+    rows_per_launch = perf["rows"] / launches
+    achieved_gbps = (rows_per_launch * ALG_BYTES_PER_ROW) / (
+        avg_launch_ms * 1e-3) / 1e9 if avg_launch_ms > 0 else 0.0
+
+    cpu = None
+    if world == 1 and not args.skip_cpu_baseline:
+        cpu = cpu_baseline()
+
+    print(json.dumps({
+        "metric": "nexmark_q5_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed * 1000 / args.steps,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": "nexmark q5 hot-items: 10s/2s sliding COUNT GROUP BY"
+                        " auction, 64K-row batches (BASELINE.json configs[1])",
+            "batch_rows": BATCH_ROWS,
+            "events_per_sec": EVENTS_PER_SEC,
+            "width_s": WIDTH_S,
+            "slide_s": SLIDE_S,
+            "parallelism": f"key-hash dp{n_gpus}",
+        },
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved_gbps,
+            "peak": HBM_PEAK_GBPS,
+            "unit": "GB/s",
+            "frac": achieved_gbps / HBM_PEAK_GBPS,
+            "traffic": None,
+            "kernel": "k_update_lds",
+            "avg_launch_us": avg_launch_ms * 1000,
+            "rows_per_launch": rows_per_launch,
+        },
+        "cpu_baseline": cpu,
+    }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,104 @@
+/* arroyo-amd C ABI: the drop-in boundary for the MI355X-native execution
+ * path of Arroyo's windowed-aggregate operators.
+ *
+ * Each entry point replaces one method of the reference's `ArrowOperator`
+ * trait (ArroyoSystems/arroyo, crates/arroyo-operator/src/operator.rs), with
+ * the same ownership and threading contract: one thread of execution per
+ * handle, operator receives owned immutable batches, emits owned batches.
+ * The Rust host would bind these over FFI (see INTEGRATION.md for the
+ * ready-to-paste binding) exactly where OperatorConstructor::with_config and
+ * the operator methods are invoked today:
+ *
+ *   arroyo_amd_create           <-> OperatorConstructor::with_config
+ *                                   (operator.rs:56-63; config contents from
+ *                                   api::SlidingWindowAggregateOperator,
+ *                                   sliding_aggregating_window.rs:449-528)
+ *   arroyo_amd_process_batch    <-> ArrowOperator::process_batch
+ *                                   (operator.rs:1183;
+ *                                   sliding_aggregating_window.rs:598-674)
+ *   arroyo_amd_handle_watermark <-> ArrowOperator::handle_watermark
+ *                                   (operator.rs:1208; sliding :676-691);
+ *                                   emitted batches = Collector::collect
+ *                                   calls (context.rs:491-494)
+ *   arroyo_amd_checkpoint_drain <-> ArrowOperator::handle_checkpoint
+ *                                   (operator.rs:1218; sliding :693-737)
+ *   arroyo_amd_restore          <-> ArrowOperator::on_start restore
+ *                                   (operator.rs:1167; sliding :556-595)
+ *   arroyo_amd_destroy          <-> operator drop
+ *   arroyo_amd_partition        <-> ArrowCollector::repartition +
+ *                                   server_for_hash_array
+ *                                   (context.rs:506-560,
+ *                                   arroyo-operator/src/lib.rs:30-41) --
+ *                                   device-side partitioning for the keyed
+ *                                   shuffle; the exchange itself is RCCL
+ *                                   all-to-all over xGMI at harness level.
+ *
+ * Batches are Arrow-layout columns (8-byte fixed-width values buffers, key
+ * columns first, `_timestamp` in nanoseconds last -- ArroyoSchema,
+ * arroyo-rpc/src/df.rs:24-30).  Timestamps/watermarks are u64 ns since the
+ * epoch; the end-of-stream watermark is UINT64_MAX.  Errors: non-zero return
+ * + arroyo_amd_last_error(handle).
+ */
+#ifndef ARROYO_AMD_H
+#define ARROYO_AMD_H
+
+#include "arroyo_amd_types.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Create a sliding/tumbling window-aggregate operator on cfg->device.
+ * Returns NULL on failure (message via arroyo_amd_last_error(NULL)).
+ * Requires a HIP device: there is no CPU fallback. */
+void *arroyo_amd_create(const AmdWindowConfig *cfg);
+
+/* Ingest one batch of host-memory columns
+ * [keys..., values..., _timestamp], each `n_rows` i64 values. */
+int arroyo_amd_process_batch(void *h, const int64_t *const *cols,
+                             int32_t n_cols, int64_t n_rows);
+
+/* Same, with columns already resident in device HBM; ts_offset is added to
+ * every timestamp (synthetic-stream ring replay). */
+int arroyo_amd_process_batch_device(void *h, const int64_t *const *dcols,
+                                    int32_t n_cols, int64_t n_rows,
+                                    uint64_t ts_offset);
+
+/* Advance the watermark; fires every window the reference would fire, in
+ * order, and returns the emitted rows (all fired windows concatenated;
+ * column order [key?, aggs..., window_start, window_end, _timestamp]).
+ * `out` may be NULL to defer collection to a later call. */
+int arroyo_amd_handle_watermark(void *h, uint64_t watermark_nanos,
+                                AmdOutBatch *out);
+
+/* Drain open panes' partial states for a checkpoint barrier.  Columns:
+ * [key?, partial state words (AVG takes 2)..., bin _timestamp]. */
+int arroyo_amd_checkpoint_drain(void *h, AmdOutBatch *out);
+
+/* Restore checkpointed partial states (layout as checkpoint_drain). */
+int arroyo_amd_restore(void *h, const int64_t *const *cols, int32_t n_cols,
+                       int64_t n_rows, int has_watermark,
+                       uint64_t watermark_nanos);
+
+void arroyo_amd_free_out(AmdOutBatch *out);
+void arroyo_amd_destroy(void *h);
+const char *arroyo_amd_last_error(void *h);
+
+/* Dominant-kernel timing for the measurement harness: total k_update time
+ * (HIP events on the operator's stream), rows processed, launches, and rows
+ * emitted device-side since the last call. */
+int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
+                    int64_t *launches, int64_t *emitted_device_rows);
+
+/* Device-side key-hash partitioning (shuffle): writes rows regrouped into
+ * per-partition contiguous segments and per-partition counts. */
+int arroyo_amd_partition(const int64_t *d_keys, const int64_t *d_vals,
+                         const int64_t *d_ts, int64_t n, uint32_t n_parts,
+                         int64_t *d_out_keys, int64_t *d_out_vals,
+                         int64_t *d_out_ts, uint64_t *h_counts);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* ARROYO_AMD_H */
