@@ -622,6 +622,7 @@ struct GpuOp {
     std::set<uint64_t> table_bins;        /* ExpiringTimeKeyView keys */
 
     int use_lds;
+    int force_blocks;
     /* perf counters for bench; events harvested lazily at sync points */
     std::vector<std::pair<hipEvent_t, hipEvent_t>> pending_ev;
     double   update_kernel_ms;
@@ -691,6 +692,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     o->state = 0;
     o->use_lds = 1;
     if (const char *e = getenv("ARROYO_AMD_LDS")) o->use_lds = atoi(e);
+    o->force_blocks = 0;
+    if (const char *e = getenv("ARROYO_AMD_BLOCKS")) o->force_blocks = atoi(e);
 
     if (hipSetDevice(cfg->device) != hipSuccess) {
         snprintf(g_err, sizeof g_err,
@@ -783,6 +786,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     A.agg = o->agg;
     int64_t want = (n_rows + 255) / 256;
     int blocks = (int)(want > 4096 ? 4096 : (want < 1 ? 1 : want));
+    if (o->force_blocks > 0) blocks = o->force_blocks;
     hipEvent_t t0, t1;
     hipEventCreate(&t0);
     hipEventCreate(&t1);
