@@ -196,6 +196,8 @@ struct UpdateArgs {
     uint64_t ts_offset;  /* added to ts (bench ring replay); 0 otherwise */
     DeviceRing ring;
     AggSpec agg;
+    int mode;  /* debug stage isolation: 0 full, 1 read+bin only,
+                  2 +tag protocol, 3 +table upsert (no state update) */
 };
 
 __device__ inline void fold_min_bin(uint64_t local_min, uint64_t *min_bin) {
@@ -220,6 +222,7 @@ k_update(UpdateArgs A) {
         uint64_t bin = t - t % A.slide;
         if (A.has_wm && bin < A.wm_bin) continue;           /* late drop */
         if (bin < local_min) local_min = bin;
+        if (A.mode == 1) continue;
         uint32_t p = (uint32_t)((bin / A.slide) & (A.ring.R - 1));
         uint64_t tag = A.ring.tag[p];
         if (tag != bin) {
@@ -231,6 +234,7 @@ k_update(UpdateArgs A) {
                 continue;
             }
         }
+        if (A.mode == 2) continue;
         int64_t key = A.key_col ? A.key_col[i] : 0;
         uint64_t *st;
         if (key == EMPTY_KEY) {
@@ -244,9 +248,25 @@ k_update(UpdateArgs A) {
             st = A.ring.state +
                  ((size_t)p * A.ring.C + (size_t)s) * A.agg.n_aggs * 2;
         }
+        if (A.mode == 3) continue;
         atomic_update(st, A.agg, vc, i);
     }
     fold_min_bin(local_min, A.ring.min_bin);
+}
+
+/* pure streaming-read calibration kernel: coalesced loads of both columns,
+ * one atomic per wave — the ceiling any update-kernel variant can reach. */
+__global__ void __launch_bounds__(256)
+k_stream_sum(const int64_t *a, const int64_t *b, int64_t n,
+             unsigned long long *out) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t acc = 0;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        acc += (uint64_t)a[i] + (uint64_t)b[i];
+    for (int off = 32; off; off >>= 1)
+        acc += (uint64_t)__shfl_down((long long)acc, off, 64);
+    if ((threadIdx.x & 63) == 0) atomicAdd(out, (unsigned long long)acc);
 }
 
 /* LDS-staged variant: each workgroup pre-aggregates its rows into an LDS
@@ -275,6 +295,7 @@ k_update_lds(UpdateArgs A) {
         uint64_t bin = t - t % A.slide;
         if (A.has_wm && bin < A.wm_bin) continue;
         if (bin < local_min) local_min = bin;
+        if (A.mode == 1) continue;
         uint32_t p = (uint32_t)((bin / A.slide) & (A.ring.R - 1));
         uint64_t tag = A.ring.tag[p];
         if (tag != bin) {
@@ -286,6 +307,7 @@ k_update_lds(UpdateArgs A) {
                 continue;
             }
         }
+        if (A.mode == 2) continue;
         int64_t key = A.key_col ? A.key_col[i] : 0;
         /* try the LDS table first (2 probes), fall through to global */
         bool done = false;
@@ -623,6 +645,7 @@ struct GpuOp {
 
     int use_lds;
     int force_blocks;
+    int kmode;
     /* perf counters for bench; events harvested lazily at sync points */
     std::vector<std::pair<hipEvent_t, hipEvent_t>> pending_ev;
     double   update_kernel_ms;
@@ -694,6 +717,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (const char *e = getenv("ARROYO_AMD_LDS")) o->use_lds = atoi(e);
     o->force_blocks = 0;
     if (const char *e = getenv("ARROYO_AMD_BLOCKS")) o->force_blocks = atoi(e);
+    o->kmode = 0;
+    if (const char *e = getenv("ARROYO_AMD_KMODE")) o->kmode = atoi(e);
 
     if (hipSetDevice(cfg->device) != hipSuccess) {
         snprintf(g_err, sizeof g_err,
@@ -784,6 +809,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     A.ts_offset = ts_offset;
     A.ring = o->ring;
     A.agg = o->agg;
+    A.mode = o->kmode;
     int64_t want = (n_rows + 255) / 256;
     int blocks = (int)(want > 4096 ? 4096 : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
@@ -1283,6 +1309,34 @@ API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
     o->launches = 0;
     o->emitted_device_rows = 0;
     return 0;
+}
+
+/* streaming-read bandwidth calibration (debug/measurement only) */
+API double arroyo_amd_stream_gbps(const void *d_a, const void *d_b, int64_t n,
+                                  int iters) {
+    unsigned long long *d_out;
+    if (hipMalloc((void **)&d_out, 8) != hipSuccess) return -1;
+    hipMemset(d_out, 0, 8);
+    int64_t want = (n + 255) / 256;
+    int blocks = (int)(want > 8192 ? 8192 : want);
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    hipLaunchKernelGGL(k_stream_sum, dim3(blocks), dim3(256), 0, 0,
+                       (const int64_t *)d_a, (const int64_t *)d_b, n, d_out);
+    hipEventRecord(t0, 0);
+    for (int i = 0; i < iters; i++)
+        hipLaunchKernelGGL(k_stream_sum, dim3(blocks), dim3(256), 0, 0,
+                           (const int64_t *)d_a, (const int64_t *)d_b, n,
+                           d_out);
+    hipEventRecord(t1, 0);
+    hipEventSynchronize(t1);
+    float ms = 0;
+    hipEventElapsedTime(&ms, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    hipFree(d_out);
+    return (double)n * 16.0 * iters / (ms * 1e-3) / 1e9;
 }
 
 /* K8 standalone: partition device-resident rows by key hash for the RCCL

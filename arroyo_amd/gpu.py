@@ -49,6 +49,9 @@ def lib():
         _lib.arroyo_amd_restore.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
             ctypes.c_int64, ctypes.c_int, ctypes.c_uint64]
+        _lib.arroyo_amd_stream_gbps.restype = ctypes.c_double
+        _lib.arroyo_amd_stream_gbps.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
         _lib.arroyo_amd_partition.restype = ctypes.c_int
         _lib.arroyo_amd_partition.argtypes = [ctypes.c_void_p] * 3 + [
             ctypes.c_int64, ctypes.c_uint32] + [ctypes.c_void_p] * 3 + [
