@@ -97,6 +97,15 @@ __host__ __device__ inline int64_t dec_max(uint64_t e) {
     return (int64_t)(e ^ 0x8000000000000000ULL);
 }
 
+/* floor-divide t by slide via the precomputed reciprocal; exact for all
+ * t < 2^64 (q from mulhi is in {q*, q*-1}; one fixup resolves it). */
+__device__ inline uint64_t div_slide(uint64_t t, uint64_t slide,
+                                     uint64_t inv) {
+    uint64_t q = __umul64hi(t, inv);
+    if (t - q * slide >= slide) q++;
+    return q;
+}
+
 __device__ inline uint64_t hash64(uint64_t x) {
     x += 0x9e3779b97f4a7c15ULL;
     x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ULL;
@@ -191,6 +200,8 @@ struct UpdateArgs {
     const int64_t *vcols[4];
     int64_t  n_rows;
     uint64_t slide;
+    uint64_t slide_inv;  /* floor(2^64 / slide): q = mulhi(t, inv) (+fixup)
+                            replaces the ~100-cycle software u64 modulo */
     uint64_t wm_bin;     /* late-drop cutoff bin; 0 if no watermark yet */
     int      has_wm;
     uint64_t ts_offset;  /* added to ts (bench ring replay); 0 otherwise */
@@ -201,14 +212,21 @@ struct UpdateArgs {
 };
 
 __device__ inline void fold_min_bin(uint64_t local_min, uint64_t *min_bin) {
-    /* one wavefront reduction + one atomic per wave, not one per row */
+    /* wavefront reduction, then an atomic ONLY if we would improve the
+     * current value: same-address atomics serialize at the L2 bank (~one
+     * per few ns), so an unconditional per-wave atomicMin alone was costing
+     * ~12 ns x waves per launch.  The L1-bypassing volatile read may be a
+     * touch stale, which only costs a rare extra atomic. */
     for (int off = 32; off; off >>= 1) {
         uint64_t v = (uint64_t)__shfl_down((long long)local_min, off, 64);
         if (v < local_min) local_min = v;
     }
-    if ((threadIdx.x & 63) == 0 && local_min != ~0ULL)
-        atomicMin((unsigned long long *)min_bin,
-                  (unsigned long long)local_min);
+    if ((threadIdx.x & 63) == 0 && local_min != ~0ULL) {
+        uint64_t cur = *(volatile unsigned long long *)min_bin;
+        if (local_min < cur)
+            atomicMin((unsigned long long *)min_bin,
+                      (unsigned long long)local_min);
+    }
 }
 
 __global__ void __launch_bounds__(256)
@@ -219,11 +237,12 @@ k_update(UpdateArgs A) {
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride) {
         uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
-        uint64_t bin = t - t % A.slide;
+        uint64_t q = div_slide(t, A.slide, A.slide_inv);
+        uint64_t bin = q * A.slide;
         if (A.has_wm && bin < A.wm_bin) continue;           /* late drop */
         if (bin < local_min) local_min = bin;
         if (A.mode == 1) continue;
-        uint32_t p = (uint32_t)((bin / A.slide) & (A.ring.R - 1));
+        uint32_t p = (uint32_t)(q & (A.ring.R - 1));
         uint64_t tag = A.ring.tag[p];
         if (tag != bin) {
             uint64_t old = atomicCAS((unsigned long long *)&A.ring.tag[p],
@@ -254,19 +273,29 @@ k_update(UpdateArgs A) {
     fold_min_bin(local_min, A.ring.min_bin);
 }
 
-/* pure streaming-read calibration kernel: coalesced loads of both columns,
- * one atomic per wave — the ceiling any update-kernel variant can reach. */
+/* pure streaming-read calibration kernel: 16 B/lane vector loads of both
+ * columns, LDS-reduced to one atomic per workgroup — the ceiling any
+ * update-kernel variant can reach. n must be even (rows; reads n/2 pairs). */
 __global__ void __launch_bounds__(256)
-k_stream_sum(const int64_t *a, const int64_t *b, int64_t n,
+k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
              unsigned long long *out) {
+    __shared__ unsigned long long red[4];
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t acc = 0;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-         i += stride)
-        acc += (uint64_t)a[i] + (uint64_t)b[i];
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n2;
+         i += stride) {
+        ulonglong2 va = a[i], vb = b[i];
+        acc += va.x + va.y + vb.x + vb.y;
+    }
     for (int off = 32; off; off >>= 1)
         acc += (uint64_t)__shfl_down((long long)acc, off, 64);
-    if ((threadIdx.x & 63) == 0) atomicAdd(out, (unsigned long long)acc);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long t = 0;
+        for (unsigned w = 0; w < blockDim.x / 64; w++) t += red[w];
+        atomicAdd(out, t);
+    }
 }
 
 /* LDS-staged variant: each workgroup pre-aggregates its rows into an LDS
@@ -292,11 +321,12 @@ k_update_lds(UpdateArgs A) {
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride) {
         uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
-        uint64_t bin = t - t % A.slide;
+        uint64_t q = div_slide(t, A.slide, A.slide_inv);
+        uint64_t bin = q * A.slide;
         if (A.has_wm && bin < A.wm_bin) continue;
         if (bin < local_min) local_min = bin;
         if (A.mode == 1) continue;
-        uint32_t p = (uint32_t)((bin / A.slide) & (A.ring.R - 1));
+        uint32_t p = (uint32_t)(q & (A.ring.R - 1));
         uint64_t tag = A.ring.tag[p];
         if (tag != bin) {
             uint64_t old = atomicCAS((unsigned long long *)&A.ring.tag[p],
@@ -804,6 +834,9 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     A.ts_col = dcols[c];
     A.n_rows = n_rows;
     A.slide = o->slide;
+    A.slide_inv = o->slide == 1
+                      ? ~0ULL
+                      : (uint64_t)((((unsigned __int128)1) << 64) / o->slide);
     A.has_wm = o->has_wm;
     A.wm_bin = o->has_wm ? o->wm - o->wm % o->slide : 0;
     A.ts_offset = ts_offset;
@@ -1317,18 +1350,20 @@ API double arroyo_amd_stream_gbps(const void *d_a, const void *d_b, int64_t n,
     unsigned long long *d_out;
     if (hipMalloc((void **)&d_out, 8) != hipSuccess) return -1;
     hipMemset(d_out, 0, 8);
-    int64_t want = (n + 255) / 256;
+    int64_t n2 = n / 2;
+    int64_t want = (n2 + 255) / 256;
     int blocks = (int)(want > 8192 ? 8192 : want);
     hipEvent_t t0, t1;
     hipEventCreate(&t0);
     hipEventCreate(&t1);
     hipLaunchKernelGGL(k_stream_sum, dim3(blocks), dim3(256), 0, 0,
-                       (const int64_t *)d_a, (const int64_t *)d_b, n, d_out);
+                       (const ulonglong2 *)d_a, (const ulonglong2 *)d_b, n2,
+                       d_out);
     hipEventRecord(t0, 0);
     for (int i = 0; i < iters; i++)
         hipLaunchKernelGGL(k_stream_sum, dim3(blocks), dim3(256), 0, 0,
-                           (const int64_t *)d_a, (const int64_t *)d_b, n,
-                           d_out);
+                           (const ulonglong2 *)d_a, (const ulonglong2 *)d_b,
+                           n2, d_out);
     hipEventRecord(t1, 0);
     hipEventSynchronize(t1);
     float ms = 0;

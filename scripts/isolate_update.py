@@ -66,7 +66,7 @@ def main():
                     bk, bt = views[bidx]
                     op.process_batch_device(
                         [bk.data_ptr(), bt.data_ptr()], rows, off)
-                    if mode == 0 and (s - last_wm[0]) * batch_span > NS:
+                    if (s - last_wm[0]) * batch_span > NS:
                         last_wm[0] = s
                         wm = int(ts[(bidx + 1) * rows - 1]) + off - NS
                         lib.arroyo_amd_handle_watermark(
