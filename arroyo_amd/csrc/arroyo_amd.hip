@@ -166,29 +166,58 @@ __device__ inline void atomic_merge(uint64_t *dst, const uint64_t *src,
 
 /* claim-or-find a key slot in an open-addressing table.
  * Keys transition EMPTY->k exactly once, so the plain-load fast path is
- * safe: a stale L1 read can only see EMPTY, and the CAS resolves it. */
-__device__ inline int64_t table_upsert(int64_t *keys, uint64_t *fill,
-                                       uint32_t C, int64_t key, int *err) {
+ * safe: a stale L1 read can only see EMPTY, and the CAS resolves it.
+ * Fullness is detected by probe length (expected probe count at <=7/8 load
+ * is single digits; a 256-probe chain means the table is effectively full).
+ * No per-insert fill counter: a same-address atomicAdd per insert was
+ * serializing at the L2 bank. */
+#define MAX_PROBES 256u
+
+__device__ inline int64_t table_upsert(int64_t *keys, uint32_t C, int64_t key,
+                                       int *err) {
     uint64_t m = C - 1;
     uint64_t i = hash64((uint64_t)key) & m;
-    for (uint32_t probes = 0; probes < C; probes++) {
+    uint32_t lim = C < MAX_PROBES ? C : MAX_PROBES;
+    for (uint32_t probes = 0; probes < lim; probes++) {
         int64_t k = keys[i];
         if (k == key) return (int64_t)i;
         if (k == EMPTY_KEY) {
             int64_t old = (int64_t)atomicCAS((unsigned long long *)&keys[i],
                                              (unsigned long long)EMPTY_KEY,
                                              (unsigned long long)key);
-            if (old == EMPTY_KEY) {
-                uint64_t f = atomicAdd((unsigned long long *)fill, 1ULL);
-                if ((f + 1) * 8 >= (uint64_t)C * 7) *err = ERR_TABLE_FULL;
-                return (int64_t)i;
-            }
-            if (old == key) return (int64_t)i;
+            if (old == EMPTY_KEY || old == key) return (int64_t)i;
         }
         i = (i + 1) & m;
     }
     *err = ERR_TABLE_FULL;
     return -1;
+}
+
+/* claim a pane's ring slot for `bin`; wave-cooperative callers dedup first.
+ * The volatile re-read (L2, bypassing a possibly-stale L1 line) keeps the
+ * CAS storm on a freshly-opened bin to the waves in flight at that moment. */
+__device__ inline void claim_tag(uint64_t *tp, uint64_t bin, int *err) {
+    uint64_t tag = *(volatile unsigned long long *)tp;
+    if (tag == bin) return;
+    uint64_t old = atomicCAS((unsigned long long *)tp,
+                             (unsigned long long)EMPTY_TAG,
+                             (unsigned long long)bin);
+    if (old != EMPTY_TAG && old != bin) *err = ERR_RING_CONFLICT;
+}
+
+/* wave-level tag protocol: rows in a wave are usually consecutive and share
+ * one bin, so one leader lane claims for everyone. */
+__device__ inline void claim_tag_wave(uint64_t *tags, uint32_t p, uint64_t bin,
+                                      uint64_t cached_tag, int *err) {
+    if (cached_tag == bin) return;
+    unsigned long long act = __ballot(1);
+    int leader = (int)(__ffsll((long long)act) - 1);
+    uint64_t b0 = (uint64_t)__shfl((long long)bin, leader, 64);
+    if (__all(bin == b0)) {
+        if ((int)(threadIdx.x & 63) == leader) claim_tag(&tags[p], bin, err);
+    } else {
+        claim_tag(&tags[p], bin, err);
+    }
 }
 
 /* ------------------------------------------------------------------ */
@@ -243,16 +272,7 @@ k_update(UpdateArgs A) {
         if (bin < local_min) local_min = bin;
         if (A.mode == 1) continue;
         uint32_t p = (uint32_t)(q & (A.ring.R - 1));
-        uint64_t tag = A.ring.tag[p];
-        if (tag != bin) {
-            uint64_t old = atomicCAS((unsigned long long *)&A.ring.tag[p],
-                                     (unsigned long long)EMPTY_TAG,
-                                     (unsigned long long)bin);
-            if (old != EMPTY_TAG && old != bin) {
-                *A.ring.err = ERR_RING_CONFLICT;
-                continue;
-            }
-        }
+        claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
         if (A.mode == 2) continue;
         int64_t key = A.key_col ? A.key_col[i] : 0;
         uint64_t *st;
@@ -261,8 +281,7 @@ k_update(UpdateArgs A) {
             st = A.ring.spec_state + (size_t)p * A.agg.n_aggs * 2;
         } else {
             int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-            int64_t s = table_upsert(keys, &A.ring.fill[p], A.ring.C, key,
-                                     A.ring.err);
+            int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
             if (s < 0) continue;
             st = A.ring.state +
                  ((size_t)p * A.ring.C + (size_t)s) * A.agg.n_aggs * 2;
@@ -327,16 +346,7 @@ k_update_lds(UpdateArgs A) {
         if (bin < local_min) local_min = bin;
         if (A.mode == 1) continue;
         uint32_t p = (uint32_t)(q & (A.ring.R - 1));
-        uint64_t tag = A.ring.tag[p];
-        if (tag != bin) {
-            uint64_t old = atomicCAS((unsigned long long *)&A.ring.tag[p],
-                                     (unsigned long long)EMPTY_TAG,
-                                     (unsigned long long)bin);
-            if (old != EMPTY_TAG && old != bin) {
-                *A.ring.err = ERR_RING_CONFLICT;
-                continue;
-            }
-        }
+        claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
         if (A.mode == 2) continue;
         int64_t key = A.key_col ? A.key_col[i] : 0;
         /* try the LDS table first (2 probes), fall through to global */
@@ -391,8 +401,7 @@ k_update_lds(UpdateArgs A) {
                 st = A.ring.spec_state + (size_t)p * na * 2;
             } else {
                 int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-                int64_t s = table_upsert(keys, &A.ring.fill[p], A.ring.C, key,
-                                         A.ring.err);
+                int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
                 if (s < 0) continue;
                 st = A.ring.state + ((size_t)p * A.ring.C + (size_t)s) * na * 2;
             }
@@ -407,8 +416,7 @@ k_update_lds(UpdateArgs A) {
         if (key == EMPTY_KEY) continue;
         uint32_t p = ls_pane[s];
         int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-        int64_t slot = table_upsert(keys, &A.ring.fill[p], A.ring.C, key,
-                                    A.ring.err);
+        int64_t slot = table_upsert(keys, A.ring.C, key, A.ring.err);
         if (slot < 0) continue;
         atomic_merge(A.ring.state +
                          ((size_t)p * A.ring.C + (size_t)slot) * na * 2,
@@ -442,7 +450,7 @@ k_merge(MergeArgs M) {
         size_t slot = i % M.ring.C;
         int64_t key = M.ring.keys[(size_t)p * M.ring.C + slot];
         if (key == EMPTY_KEY) continue;
-        int64_t d = table_upsert(M.m_keys, M.m_fill, M.CM, key, M.ring.err);
+        int64_t d = table_upsert(M.m_keys, M.CM, key, M.ring.err);
         if (d < 0) continue;
         atomic_merge(M.m_state + (size_t)d * M.agg.n_aggs * 2,
                      M.ring.state +
@@ -575,8 +583,7 @@ k_restore(RestoreArgs R) {
             st = R.ring.spec_state + (size_t)p * R.agg.n_aggs * 2;
         } else {
             int64_t *keys = R.ring.keys + (size_t)p * R.ring.C;
-            int64_t s = table_upsert(keys, &R.ring.fill[p], R.ring.C, key,
-                                     R.ring.err);
+            int64_t s = table_upsert(keys, R.ring.C, key, R.ring.err);
             if (s < 0) continue;
             st = R.ring.state +
                  ((size_t)p * R.ring.C + (size_t)s) * R.agg.n_aggs * 2;
