@@ -323,94 +323,85 @@ k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
  * ring.  Entries evicted on LDS collision fall through to global atomics. */
 #define LDS_SLOTS 1024   /* x (8B key + 4B pane + n_aggs*16B) */
 
-__global__ void __launch_bounds__(256)
-k_update_lds(UpdateArgs A) {
-    __shared__ int64_t  ls_key[LDS_SLOTS];
-    __shared__ uint32_t ls_pane[LDS_SLOTS];
-    extern __shared__ uint64_t ls_st[];   /* [LDS_SLOTS][n_aggs][2] */
+/* per-row body shared by the scalar and vectorized LDS kernels */
+__device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
+                                      uint32_t *ls_pane, uint64_t *ls_st,
+                                      int64_t row, uint64_t traw, int64_t key,
+                                      uint64_t &local_min) {
     const int na = A.agg.n_aggs;
-    for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
-        ls_key[i] = EMPTY_KEY;
-        for (int w = 0; w < na * 2; w++) ls_st[(size_t)i * na * 2 + w] = 0;
-    }
-    __syncthreads();
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
     const int64_t *const *vc = A.vcols;
-    uint64_t local_min = ~0ULL;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < A.n_rows; i += stride) {
-        uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
-        uint64_t q = div_slide(t, A.slide, A.slide_inv);
-        uint64_t bin = q * A.slide;
-        if (A.has_wm && bin < A.wm_bin) continue;
-        if (bin < local_min) local_min = bin;
-        if (A.mode == 1) continue;
-        uint32_t p = (uint32_t)(q & (A.ring.R - 1));
-        claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
-        if (A.mode == 2) continue;
-        int64_t key = A.key_col ? A.key_col[i] : 0;
-        /* try the LDS table first (2 probes), fall through to global */
-        bool done = false;
-        if (key != EMPTY_KEY) {
-            uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
-                         (LDS_SLOTS - 1);
-            for (int pr = 0; pr < 2 && !done; pr++) {
-                uint32_t s = (h + pr) & (LDS_SLOTS - 1);
-                int64_t k = ls_key[s];
-                if (k == EMPTY_KEY) {
-                    int64_t old = (int64_t)atomicCAS(
-                        (unsigned long long *)&ls_key[s],
-                        (unsigned long long)EMPTY_KEY, (unsigned long long)key);
-                    if (old == EMPTY_KEY) { ls_pane[s] = p; k = key; }
-                    else k = old;
-                }
-                if (k == key && ls_pane[s] == p) {
-                    uint64_t *st = ls_st + (size_t)s * na * 2;
-                    for (int a = 0; a < na; a++) {
-                        uint64_t *w = st + 2 * a;
-                        switch (A.agg.op[a]) {
-                        case AMD_AGG_COUNT:
-                            atomicAdd((unsigned long long *)w, 1ULL); break;
-                        case AMD_AGG_SUM:
-                            atomicAdd((unsigned long long *)w,
-                                      (unsigned long long)vc[A.agg.col[a]][i]);
-                            break;
-                        case AMD_AGG_MIN:
-                            atomicMax((unsigned long long *)w,
-                                      (unsigned long long)enc_min(vc[A.agg.col[a]][i]));
-                            break;
-                        case AMD_AGG_MAX:
-                            atomicMax((unsigned long long *)w,
-                                      (unsigned long long)enc_max(vc[A.agg.col[a]][i]));
-                            break;
-                        case AMD_AGG_AVG:
-                            atomicAdd((unsigned long long *)w, 1ULL);
-                            atomicAdd((double *)(w + 1),
-                                      (double)vc[A.agg.col[a]][i]);
-                            break;
-                        }
+    uint64_t t = traw + A.ts_offset;
+    uint64_t q = div_slide(t, A.slide, A.slide_inv);
+    uint64_t bin = q * A.slide;
+    if (A.has_wm && bin < A.wm_bin) return;
+    if (bin < local_min) local_min = bin;
+    if (A.mode == 1) return;
+    uint32_t p = (uint32_t)(q & (A.ring.R - 1));
+    claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
+    if (A.mode == 2) return;
+    /* try the LDS table first (2 probes), fall through to global */
+    bool done = false;
+    if (key != EMPTY_KEY) {
+        uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
+                     (LDS_SLOTS - 1);
+        for (int pr = 0; pr < 2 && !done; pr++) {
+            uint32_t s = (h + pr) & (LDS_SLOTS - 1);
+            int64_t k = ls_key[s];
+            if (k == EMPTY_KEY) {
+                int64_t old = (int64_t)atomicCAS(
+                    (unsigned long long *)&ls_key[s],
+                    (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+                if (old == EMPTY_KEY) { ls_pane[s] = p; k = key; }
+                else k = old;
+            }
+            if (k == key && ls_pane[s] == p) {
+                uint64_t *st = ls_st + (size_t)s * na * 2;
+                for (int a = 0; a < na; a++) {
+                    uint64_t *w = st + 2 * a;
+                    switch (A.agg.op[a]) {
+                    case AMD_AGG_COUNT:
+                        atomicAdd((unsigned long long *)w, 1ULL); break;
+                    case AMD_AGG_SUM:
+                        atomicAdd((unsigned long long *)w,
+                                  (unsigned long long)vc[A.agg.col[a]][row]);
+                        break;
+                    case AMD_AGG_MIN:
+                        atomicMax((unsigned long long *)w,
+                                  (unsigned long long)enc_min(vc[A.agg.col[a]][row]));
+                        break;
+                    case AMD_AGG_MAX:
+                        atomicMax((unsigned long long *)w,
+                                  (unsigned long long)enc_max(vc[A.agg.col[a]][row]));
+                        break;
+                    case AMD_AGG_AVG:
+                        atomicAdd((unsigned long long *)w, 1ULL);
+                        atomicAdd((double *)(w + 1),
+                                  (double)vc[A.agg.col[a]][row]);
+                        break;
                     }
-                    done = true;
                 }
+                done = true;
             }
-        }
-        if (!done) {
-            uint64_t *st;
-            if (key == EMPTY_KEY) {
-                atomicExch(&A.ring.spec_used[p], 1u);
-                st = A.ring.spec_state + (size_t)p * na * 2;
-            } else {
-                int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-                int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
-                if (s < 0) continue;
-                st = A.ring.state + ((size_t)p * A.ring.C + (size_t)s) * na * 2;
-            }
-            atomic_update(st, A.agg, vc, i);
         }
     }
-    fold_min_bin(local_min, A.ring.min_bin);
-    __syncthreads();
-    /* flush the LDS table into the ring */
+    if (!done) {
+        uint64_t *st;
+        if (key == EMPTY_KEY) {
+            atomicExch(&A.ring.spec_used[p], 1u);
+            st = A.ring.spec_state + (size_t)p * na * 2;
+        } else {
+            int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
+            int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
+            if (s < 0) return;
+            st = A.ring.state + ((size_t)p * A.ring.C + (size_t)s) * na * 2;
+        }
+        atomic_update(st, A.agg, vc, row);
+    }
+}
+
+__device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
+                                 uint32_t *ls_pane, uint64_t *ls_st) {
+    const int na = A.agg.n_aggs;
     for (int s = threadIdx.x; s < LDS_SLOTS; s += blockDim.x) {
         int64_t key = ls_key[s];
         if (key == EMPTY_KEY) continue;
@@ -422,6 +413,68 @@ k_update_lds(UpdateArgs A) {
                          ((size_t)p * A.ring.C + (size_t)slot) * na * 2,
                      ls_st + (size_t)s * na * 2, A.agg);
     }
+}
+
+__device__ inline void lds_init(const UpdateArgs &A, int64_t *ls_key,
+                                uint64_t *ls_st) {
+    const int na = A.agg.n_aggs;
+    for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+        ls_key[i] = EMPTY_KEY;
+        for (int w = 0; w < na * 2; w++) ls_st[(size_t)i * na * 2 + w] = 0;
+    }
+    __syncthreads();
+}
+
+__global__ void __launch_bounds__(256)
+k_update_lds(UpdateArgs A) {
+    __shared__ int64_t  ls_key[LDS_SLOTS];
+    __shared__ uint32_t ls_pane[LDS_SLOTS];
+    extern __shared__ uint64_t ls_st[];   /* [LDS_SLOTS][n_aggs][2] */
+    lds_init(A, ls_key, ls_st);
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < A.n_rows; i += stride)
+        lds_update_row(A, ls_key, ls_pane, ls_st, i, (uint64_t)A.ts_col[i],
+                       A.key_col ? A.key_col[i] : 0, local_min);
+    fold_min_bin(local_min, A.ring.min_bin);
+    __syncthreads();
+    lds_flush(A, ls_key, ls_pane, ls_st);
+}
+
+/* vectorized variant: 16 B/lane loads of the ts (and key) columns, two rows
+ * per thread per iteration -- requires 16 B-aligned column pointers (host
+ * checks).  Fewer, fatter waves: wave-dispatch cost was measurable at one
+ * 8 B load per thread. */
+__global__ void __launch_bounds__(256)
+k_update_lds_vec(UpdateArgs A) {
+    __shared__ int64_t  ls_key[LDS_SLOTS];
+    __shared__ uint32_t ls_pane[LDS_SLOTS];
+    extern __shared__ uint64_t ls_st[];
+    lds_init(A, ls_key, ls_st);
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL;
+    int64_t n2 = A.n_rows >> 1;
+    for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < n2;
+         v += stride) {
+        ulonglong2 tsv = ((const ulonglong2 *)A.ts_col)[v];
+        int64_t k0 = 0, k1 = 0;
+        if (A.key_col) {
+            ulonglong2 kv = ((const ulonglong2 *)A.key_col)[v];
+            k0 = (int64_t)kv.x;
+            k1 = (int64_t)kv.y;
+        }
+        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0, local_min);
+        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y, k1,
+                       local_min);
+    }
+    if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+        lds_update_row(A, ls_key, ls_pane, ls_st, A.n_rows - 1,
+                       (uint64_t)A.ts_col[A.n_rows - 1],
+                       A.key_col ? A.key_col[A.n_rows - 1] : 0, local_min);
+    fold_min_bin(local_min, A.ring.min_bin);
+    __syncthreads();
+    lds_flush(A, ls_key, ls_pane, ls_st);
 }
 
 /* ------------------------------------------------------------------ */
@@ -683,25 +736,30 @@ struct GpuOp {
     int use_lds;
     int force_blocks;
     int kmode;
-    /* perf counters for bench; events harvested lazily at sync points */
-    std::vector<std::pair<hipEvent_t, hipEvent_t>> pending_ev;
-    double   update_kernel_ms;
+    /* perf counters for bench; a fixed pool of reusable event pairs samples
+     * a subset of launches (create/destroy per launch was host overhead) */
+    std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pool;
+    std::vector<int> ev_inflight;  /* indices into ev_pool */
+    double   update_kernel_ms;     /* sum over SAMPLED launches */
+    int64_t  sampled_launches;
     int64_t  update_rows;
     int64_t  launches;
     int64_t  emitted_device_rows;
     char err_msg[512];
 };
 
+#define EV_POOL 64
+
 static void harvest_events(GpuOp *o) {
-    for (auto &pr : o->pending_ev) {
-        hipEventSynchronize(pr.second);
+    for (int idx : o->ev_inflight) {
+        hipEventSynchronize(o->ev_pool[idx].second);
         float ms = 0;
-        hipEventElapsedTime(&ms, pr.first, pr.second);
+        hipEventElapsedTime(&ms, o->ev_pool[idx].first,
+                            o->ev_pool[idx].second);
         o->update_kernel_ms += ms;
-        hipEventDestroy(pr.first);
-        hipEventDestroy(pr.second);
+        o->sampled_launches++;
     }
-    o->pending_ev.clear();
+    o->ev_inflight.clear();
 }
 
 static thread_local char g_err[512];
@@ -850,26 +908,48 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     A.ring = o->ring;
     A.agg = o->agg;
     A.mode = o->kmode;
-    int64_t want = (n_rows + 255) / 256;
-    int blocks = (int)(want > 4096 ? 4096 : (want < 1 ? 1 : want));
+    /* vectorized path needs 16B-aligned ts/key columns */
+    bool vec = o->use_lds && n_rows >= 2 &&
+               ((uintptr_t)A.ts_col & 15) == 0 &&
+               (!A.key_col || ((uintptr_t)A.key_col & 15) == 0);
+    int64_t units = vec ? (n_rows + 1) / 2 : n_rows;
+    int64_t want = (units + 255) / 256;
+    int blocks = (int)(want > 2048 ? 2048 : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
-    hipEvent_t t0, t1;
-    hipEventCreate(&t0);
-    hipEventCreate(&t1);
-    hipEventRecord(t0, o->stream);
-    if (o->use_lds) {
-        size_t shmem = (size_t)LDS_SLOTS * o->agg.n_aggs * 16;
+    /* sample kernel time on a subset of launches via a reusable event pool */
+    bool sample = (o->launches & 7) == 0;
+    if (sample && o->ev_pool.empty()) {
+        o->ev_pool.resize(EV_POOL);
+        for (auto &pr : o->ev_pool) {
+            hipEventCreate(&pr.first);
+            hipEventCreate(&pr.second);
+        }
+    }
+    if (sample && o->ev_inflight.size() >= EV_POOL) harvest_events(o);
+    int ev = -1;
+    if (sample) {
+        /* find a free pair: pool minus inflight (inflight cleared on
+         * harvest, so pool scan order is fine) */
+        ev = (int)(o->ev_inflight.size());
+        hipEventRecord(o->ev_pool[ev].first, o->stream);
+    }
+    size_t shmem = (size_t)LDS_SLOTS * o->agg.n_aggs * 16;
+    if (vec) {
+        hipLaunchKernelGGL(k_update_lds_vec, dim3(blocks), dim3(256), shmem,
+                           o->stream, A);
+    } else if (o->use_lds) {
         hipLaunchKernelGGL(k_update_lds, dim3(blocks), dim3(256), shmem,
                            o->stream, A);
     } else {
         hipLaunchKernelGGL(k_update, dim3(blocks), dim3(256), 0, o->stream, A);
     }
-    hipEventRecord(t1, o->stream);
+    if (sample) {
+        hipEventRecord(o->ev_pool[ev].second, o->stream);
+        o->ev_inflight.push_back(ev);
+    }
     HIP_CHECK(o, hipGetLastError());
-    o->pending_ev.emplace_back(t0, t1);
     o->update_rows += n_rows;
     o->launches++;
-    if (o->pending_ev.size() > 512) harvest_events(o);
     return 0;
 }
 
@@ -1312,6 +1392,10 @@ API void arroyo_amd_destroy(void *h) {
     if (!o) return;
     hipStreamSynchronize(o->stream);
     harvest_events(o);
+    for (auto &pr : o->ev_pool) {
+        hipEventDestroy(pr.first);
+        hipEventDestroy(pr.second);
+    }
     hipFree(o->ring.keys);
     hipFree(o->ring.state);
     hipFree(o->ring.tag);
@@ -1340,11 +1424,17 @@ API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
                         int64_t *launches, int64_t *emitted_device_rows) {
     GpuOp *o = (GpuOp *)h;
     harvest_events(o);
-    *update_ms = o->update_kernel_ms;
+    /* update_ms is extrapolated from the sampled launches so that
+     * update_ms / launches equals the sampled per-launch average */
+    *update_ms = o->sampled_launches
+                     ? o->update_kernel_ms *
+                           ((double)o->launches / (double)o->sampled_launches)
+                     : 0.0;
     *rows = o->update_rows;
     *launches = o->launches;
     *emitted_device_rows = o->emitted_device_rows;
     o->update_kernel_ms = 0;
+    o->sampled_launches = 0;
     o->update_rows = 0;
     o->launches = 0;
     o->emitted_device_rows = 0;
