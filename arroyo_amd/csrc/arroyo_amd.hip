@@ -736,6 +736,8 @@ struct GpuOp {
     int use_lds;
     int force_blocks;
     int kmode;
+    int use_vec;
+    int use_events;
     /* perf counters for bench; a fixed pool of reusable event pairs samples
      * a subset of launches (create/destroy per launch was host overhead) */
     std::vector<std::pair<hipEvent_t, hipEvent_t>> ev_pool;
@@ -814,6 +816,10 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (const char *e = getenv("ARROYO_AMD_BLOCKS")) o->force_blocks = atoi(e);
     o->kmode = 0;
     if (const char *e = getenv("ARROYO_AMD_KMODE")) o->kmode = atoi(e);
+    o->use_vec = 1;
+    if (const char *e = getenv("ARROYO_AMD_VEC")) o->use_vec = atoi(e);
+    o->use_events = 1;
+    if (const char *e = getenv("ARROYO_AMD_EVENTS")) o->use_events = atoi(e);
 
     if (hipSetDevice(cfg->device) != hipSuccess) {
         snprintf(g_err, sizeof g_err,
@@ -909,7 +915,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     A.agg = o->agg;
     A.mode = o->kmode;
     /* vectorized path needs 16B-aligned ts/key columns */
-    bool vec = o->use_lds && n_rows >= 2 &&
+    bool vec = o->use_lds && o->use_vec && n_rows >= 2 &&
                ((uintptr_t)A.ts_col & 15) == 0 &&
                (!A.key_col || ((uintptr_t)A.key_col & 15) == 0);
     int64_t units = vec ? (n_rows + 1) / 2 : n_rows;
@@ -917,7 +923,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     int blocks = (int)(want > 2048 ? 2048 : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
     /* sample kernel time on a subset of launches via a reusable event pool */
-    bool sample = (o->launches & 7) == 0;
+    bool sample = o->use_events && (o->launches & 7) == 0;
     if (sample && o->ev_pool.empty()) {
         o->ev_pool.resize(EV_POOL);
         for (auto &pr : o->ev_pool) {
