@@ -542,8 +542,7 @@ struct CompactArgs {
 };
 
 __device__ inline void emit_row(const CompactArgs &C, int64_t key,
-                                const uint64_t *st) {
-    int64_t r = (int64_t)atomicAdd(C.n_out, 1ULL);
+                                const uint64_t *st, int64_t r) {
     int col = 0;
     if (C.n_keys) C.out[col++][r] = key;
     for (int a = 0; a < C.agg.n_aggs; a++) {
@@ -585,15 +584,35 @@ __device__ inline void emit_row(const CompactArgs &C, int64_t key,
 
 __global__ void __launch_bounds__(256)
 k_compact(CompactArgs C) {
+    /* wave-aggregated output cursor: one atomicAdd per wave per iteration
+     * instead of one per emitted row (same-address atomics serialize) */
     size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < C.CM; i += stride) {
-        int64_t key = C.m_keys[i];
-        if (key == EMPTY_KEY) continue;
-        emit_row(C, key, C.m_state + i * C.agg.n_aggs * 2);
+    int lane = (int)(threadIdx.x & 63);
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;;
+         i += stride) {
+        bool in = i < C.CM;
+        if (!__ballot(in)) break;          /* whole wave past the end */
+        int64_t key = in ? C.m_keys[i] : EMPTY_KEY;
+        bool live = in && key != EMPTY_KEY;
+        unsigned long long m = __ballot(live);
+        if (m) {
+            int leader = (int)(__ffsll((long long)m) - 1);
+            unsigned long long base = 0;
+            if (lane == leader)
+                base = atomicAdd(C.n_out,
+                                 (unsigned long long)__popcll((long long)m));
+            base = (unsigned long long)__shfl((long long)base, leader, 64);
+            if (live) {
+                int64_t r = (int64_t)(base +
+                                      __popcll((long long)(m & ((1ULL << lane) - 1))));
+                emit_row(C, key, C.m_state + i * C.agg.n_aggs * 2, r);
+            }
+        }
+        if (!in) break;
     }
     if (blockIdx.x == 0 && threadIdx.x == 0 && *C.m_spec_used)
-        emit_row(C, EMPTY_KEY, C.m_spec_state);
+        emit_row(C, EMPTY_KEY, C.m_spec_state,
+                 (int64_t)atomicAdd(C.n_out, 1ULL));
 }
 
 /* restore checkpointed partial states: insert raw state rows into a pane. */
@@ -1006,6 +1025,34 @@ API int arroyo_amd_process_batch_device(void *h, const int64_t *const *dcols,
     }
     if (flush_staged(o)) return 1;
     return launch_update(o, dcols, n_rows, ts_offset);
+}
+
+/* replay harness path: submit `reps` consecutive device-resident batches in
+ * one call (one kernel launch per batch, ts_offset advancing by ts_step),
+ * so the per-batch Python/FFI overhead is off the measured path.  batches
+ * are laid out back-to-back in dcols: batch k = rows [k*n_rows, (k+1)*n_rows)
+ * when contiguous=1, else the same batch is replayed reps times. */
+API int arroyo_amd_process_batches_device(void *h,
+                                          const int64_t *const *dcols,
+                                          int32_t n_cols, int64_t n_rows,
+                                          int32_t reps, int32_t contiguous,
+                                          uint64_t ts_offset0,
+                                          uint64_t ts_step) {
+    GpuOp *o = (GpuOp *)h;
+    if (n_cols != o->n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->n_in_cols, n_cols);
+        return 1;
+    }
+    if (flush_staged(o)) return 1;
+    const int64_t *cols[16];
+    for (int k = 0; k < reps; k++) {
+        for (int c = 0; c < n_cols; c++)
+            cols[c] = dcols[c] + (contiguous ? (int64_t)k * n_rows : 0);
+        if (launch_update(o, cols, n_rows, ts_offset0 + (uint64_t)k * ts_step))
+            return 1;
+    }
+    return 0;
 }
 
 static int check_device_error(GpuOp *o) {

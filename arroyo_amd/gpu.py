@@ -49,6 +49,11 @@ def lib():
         _lib.arroyo_amd_restore.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
             ctypes.c_int64, ctypes.c_int, ctypes.c_uint64]
+        _lib.arroyo_amd_process_batches_device.restype = ctypes.c_int
+        _lib.arroyo_amd_process_batches_device.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64, ctypes.c_int32, ctypes.c_int32, ctypes.c_uint64,
+            ctypes.c_uint64]
         _lib.arroyo_amd_stream_gbps.restype = ctypes.c_double
         _lib.arroyo_amd_stream_gbps.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
@@ -70,6 +75,15 @@ class GpuWindowOp(WindowOp):
         arr = (ctypes.c_void_p * len(dev_ptrs))(*dev_ptrs)
         rc = lib().arroyo_amd_process_batch_device(
             self._h, arr, len(dev_ptrs), n_rows, ts_offset)
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def process_batches_device(self, dev_ptrs, n_rows, reps, contiguous=True,
+                               ts_offset0=0, ts_step=0):
+        arr = (ctypes.c_void_p * len(dev_ptrs))(*dev_ptrs)
+        rc = lib().arroyo_amd_process_batches_device(
+            self._h, arr, len(dev_ptrs), n_rows, reps,
+            1 if contiguous else 0, ts_offset0, ts_step)
         if rc != 0:
             raise RuntimeError(self._fn["last_error"](self._h).decode())
 

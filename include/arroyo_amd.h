@@ -64,6 +64,20 @@ int arroyo_amd_process_batch_device(void *h, const int64_t *const *dcols,
                                     int32_t n_cols, int64_t n_rows,
                                     uint64_t ts_offset);
 
+/* Submit `reps` device-resident batches in one call (one kernel launch per
+ * batch).  contiguous=1: batch k = rows [k*n_rows,(k+1)*n_rows) of dcols;
+ * contiguous=0: the same rows replayed.  ts_offset advances by ts_step per
+ * batch.  Keeps per-batch FFI overhead off the hot loop. */
+int arroyo_amd_process_batches_device(void *h, const int64_t *const *dcols,
+                                      int32_t n_cols, int64_t n_rows,
+                                      int32_t reps, int32_t contiguous,
+                                      uint64_t ts_offset0, uint64_t ts_step);
+
+/* Streaming-read bandwidth calibration over two n-row i64 device columns;
+ * returns GB/s (measurement tooling). */
+double arroyo_amd_stream_gbps(const void *d_a, const void *d_b, int64_t n,
+                              int iters);
+
 /* Advance the watermark; fires every window the reference would fire, in
  * order, and returns the emitted rows (all fired windows concatenated;
  * column order [key?, aggs..., window_start, window_end, _timestamp]).
