@@ -347,14 +347,22 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
         for (int pr = 0; pr < 2 && !done; pr++) {
             uint32_t s = (h + pr) & (LDS_SLOTS - 1);
             int64_t k = ls_key[s];
+            bool claimed = false;
             if (k == EMPTY_KEY) {
                 int64_t old = (int64_t)atomicCAS(
                     (unsigned long long *)&ls_key[s],
                     (unsigned long long)EMPTY_KEY, (unsigned long long)key);
-                if (old == EMPTY_KEY) { ls_pane[s] = p; k = key; }
+                if (old == EMPTY_KEY) {
+                    ls_pane[s] = p;
+                    k = key;
+                    claimed = true;
+                }
                 else k = old;
             }
-            if (k == key && ls_pane[s] == p) {
+            /* ls_pane still PANE_UNSET means another thread's claim is not
+             * fully visible yet: fall through to the global path rather
+             * than aggregating into an unknown pane */
+            if (k == key && (claimed || ls_pane[s] == p)) {
                 uint64_t *st = ls_st + (size_t)s * na * 2;
                 for (int a = 0; a < na; a++) {
                     uint64_t *w = st + 2 * a;
@@ -415,11 +423,14 @@ __device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
     }
 }
 
+#define PANE_UNSET 0xFFFFFFFFu
+
 __device__ inline void lds_init(const UpdateArgs &A, int64_t *ls_key,
-                                uint64_t *ls_st) {
+                                uint32_t *ls_pane, uint64_t *ls_st) {
     const int na = A.agg.n_aggs;
     for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
         ls_key[i] = EMPTY_KEY;
+        ls_pane[i] = PANE_UNSET;   /* sentinel: claim not yet visible */
         for (int w = 0; w < na * 2; w++) ls_st[(size_t)i * na * 2 + w] = 0;
     }
     __syncthreads();
@@ -430,7 +441,7 @@ k_update_lds(UpdateArgs A) {
     __shared__ int64_t  ls_key[LDS_SLOTS];
     __shared__ uint32_t ls_pane[LDS_SLOTS];
     extern __shared__ uint64_t ls_st[];   /* [LDS_SLOTS][n_aggs][2] */
-    lds_init(A, ls_key, ls_st);
+    lds_init(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -451,7 +462,7 @@ k_update_lds_vec(UpdateArgs A) {
     __shared__ int64_t  ls_key[LDS_SLOTS];
     __shared__ uint32_t ls_pane[LDS_SLOTS];
     extern __shared__ uint64_t ls_st[];
-    lds_init(A, ls_key, ls_st);
+    lds_init(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL;
     int64_t n2 = A.n_rows >> 1;
@@ -945,7 +956,8 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
                (!A.key_col || ((uintptr_t)A.key_col & 15) == 0);
     int64_t units = vec ? (n_rows + 1) / 2 : n_rows;
     int64_t want = (units + 255) / 256;
-    int blocks = (int)(want > 2048 ? 2048 : (want < 1 ? 1 : want));
+    /* 1024-block cap won the grid sweep at ~1M-row fused launches */
+    int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
     /* sample kernel time on a subset of launches via a reusable event pool */
     bool sample = o->use_events && (o->launches & 7) == 0;
