@@ -108,6 +108,125 @@ def _out_to_numpy(out):
     return cols
 
 
+class AmdJoinConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_keys", ctypes.c_int32),
+        ("n_left_vals", ctypes.c_int32),
+        ("n_right_vals", ctypes.c_int32),
+        ("log2_rows_cap", ctypes.c_uint32),
+        ("instants", ctypes.c_uint32),
+        ("log2_out_cap", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_join_config(n_keys=1, n_left_vals=0, n_right_vals=0,
+                     log2_rows_cap=15, instants=128, log2_out_cap=20,
+                     device=0, emit_to_host=True):
+    cfg = AmdJoinConfig()
+    cfg.n_keys = n_keys
+    cfg.n_left_vals = n_left_vals
+    cfg.n_right_vals = n_right_vals
+    cfg.log2_rows_cap = log2_rows_cap
+    cfg.instants = instants
+    cfg.log2_out_cap = log2_out_cap
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+def bind_join(lib, prefix):
+    fn = {}
+    g = lambda n: getattr(lib, prefix + "join_" + n)
+    fn["create"] = g("create")
+    fn["create"].restype = ctypes.c_void_p
+    fn["create"].argtypes = [ctypes.POINTER(AmdJoinConfig)]
+    fn["process_batch"] = g("process_batch")
+    fn["process_batch"].restype = ctypes.c_int
+    fn["process_batch"].argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                    ctypes.POINTER(ctypes.c_void_p),
+                                    ctypes.c_int32, ctypes.c_int64]
+    fn["handle_watermark"] = g("handle_watermark")
+    fn["handle_watermark"].restype = ctypes.c_int
+    fn["handle_watermark"].argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                       ctypes.POINTER(AmdOutBatch)]
+    fn["checkpoint_drain"] = g("checkpoint_drain")
+    fn["checkpoint_drain"].restype = ctypes.c_int
+    fn["checkpoint_drain"].argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                       ctypes.POINTER(AmdOutBatch)]
+    fn["free_out"] = getattr(lib, prefix + "free_out")
+    fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+    fn["destroy"] = g("destroy")
+    fn["destroy"].argtypes = [ctypes.c_void_p]
+    fn["last_error"] = g("last_error")
+    fn["last_error"].restype = ctypes.c_char_p
+    fn["last_error"].argtypes = [ctypes.c_void_p]
+    return fn
+
+
+class JoinOp:
+    """One instant-join operator instance behind the C ABI (left = side 0,
+    right = side 1), mirroring InstantJoin's ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/instant_join.rs)."""
+
+    LEFT, RIGHT = 0, 1
+
+    def __init__(self, lib, prefix, cfg):
+        self._fn = bind_join(lib, prefix)
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{prefix}join_create failed")
+
+    def process_batch(self, side, cols):
+        n_rows = len(cols[0]) if cols else 0
+        arr = (ctypes.c_void_p * len(cols))()
+        keep = []
+        for i, c in enumerate(cols):
+            c = np.ascontiguousarray(c, dtype=np.int64)
+            keep.append(c)
+            arr[i] = c.ctypes.data_as(ctypes.c_void_p).value
+        rc = self._fn["process_batch"](self._h, side, arr, len(cols), n_rows)
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        return keep
+
+    def handle_watermark(self, wm):
+        out = AmdOutBatch()
+        rc = self._fn["handle_watermark"](self._h, wm, ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def checkpoint_drain(self, side):
+        out = AmdOutBatch()
+        rc = self._fn["checkpoint_drain"](self._h, side, ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore(self, side, cols):
+        """on_start re-processes drained batches (instant_join.rs:205-230)."""
+        if cols and len(cols[0]):
+            self.process_batch(side, cols)
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
 class WindowOp:
     """One window-aggregate operator instance behind the C ABI.
 

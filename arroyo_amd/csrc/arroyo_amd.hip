@@ -1588,3 +1588,581 @@ API int arroyo_amd_partition(const int64_t *d_keys, const int64_t *d_vals,
     hipFree(d_counts);
     return e == hipSuccess ? 0 : 1;
 }
+
+/* ================================================================== */
+/* Instant (windowed stream-stream) join: MI355X-native equivalent of
+ * InstantJoin (crates/arroyo-worker/src/arrow/instant_join.rs) behind the
+ * arroyo_amd_join_* C ABI (include/arroyo_amd.h).
+ *
+ * Design (not a translation): instead of per-instant DataFusion exec
+ * streams fed through channels (instant_join.rs:86-107), both sides'
+ * rows land directly in a device-resident instant table: `instants` slots,
+ * each an append buffer per side (key + value columns, SoA).  A fused
+ * append kernel (K2-equivalent routing + buffering) claims the instant
+ * slot by CAS on its tag and appends with a wave-aggregated cursor.  On
+ * watermark the host fires instants < wm in timestamp order
+ * (instant_join.rs:265-281): per instant, a build kernel chains the left
+ * rows into a hash multimap (LDS-free; the map is small and L2-resident)
+ * and a probe kernel streams the right rows and emits matches (K6) --
+ * or, for n_keys == 0 (join on the window itself), a cross-product
+ * kernel with no atomics at all. */
+
+#define JERR_LATE      4
+#define JERR_INSTANTS  5
+#define JERR_ROWS_CAP  6
+#define JERR_OUT_CAP   7
+
+struct JAppendArgs {
+    const int64_t *cols[6];   /* key?, vals..., ts */
+    int32_t n_keys, n_vals;
+    int64_t n_rows;
+    uint64_t *tag;            /* [I] */
+    unsigned long long *cursor; /* [I] */
+    int64_t *key;             /* [I*cap] */
+    int64_t *vals;            /* [n_vals][I*cap] */
+    uint32_t I, cap;
+    int has_wm; uint64_t wm;
+    int *err;
+};
+
+/* claim-or-find the slot for instant t (open addressing over the tags) */
+__device__ inline int32_t claim_instant(uint64_t *tag, uint32_t I, uint64_t t,
+                                        int *err) {
+    uint32_t m = I - 1;
+    uint32_t j = (uint32_t)hash64(t) & m;
+    for (uint32_t probes = 0; probes < I; probes++) {
+        uint64_t cur = tag[j];
+        if (cur == t) return (int32_t)j;
+        if (cur == EMPTY_TAG) {
+            uint64_t old = atomicCAS((unsigned long long *)&tag[j],
+                                     (unsigned long long)EMPTY_TAG,
+                                     (unsigned long long)t);
+            if (old == EMPTY_TAG || old == t) return (int32_t)j;
+        }
+        j = (j + 1) & m;
+    }
+    *err = JERR_INSTANTS;
+    return -1;
+}
+
+__global__ void __launch_bounds__(256)
+k_join_append(JAppendArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t *ts = A.cols[A.n_keys + A.n_vals];
+    int lane = (int)(threadIdx.x & 63);
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < A.n_rows; i += stride) {
+        uint64_t t = (uint64_t)ts[i];
+        if (A.has_wm && t < A.wm) { *A.err = JERR_LATE; continue; }
+        /* wave-uniform fast path: one leader claims slot + cursor range */
+        unsigned long long act = __ballot(1);
+        int leader = (int)(__ffsll((long long)act) - 1);
+        uint64_t t0 = (uint64_t)__shfl((long long)t, leader, 64);
+        int32_t slot;
+        uint64_t idx;
+        if (__all(t == t0)) {
+            int cnt = __popcll((long long)act);
+            int rank = __popcll((long long)(act & ((1ULL << lane) - 1)));
+            unsigned long long base = 0;
+            int32_t s0 = -1;
+            if (lane == leader) {
+                s0 = claim_instant(A.tag, A.I, t, A.err);
+                if (s0 >= 0)
+                    base = atomicAdd(&A.cursor[s0], (unsigned long long)cnt);
+            }
+            slot = (int32_t)__shfl((int)s0, leader, 64);
+            base = (unsigned long long)__shfl((long long)base, leader, 64);
+            if (slot < 0) continue;
+            idx = base + (uint64_t)rank;
+        } else {
+            slot = claim_instant(A.tag, A.I, t, A.err);
+            if (slot < 0) continue;
+            idx = atomicAdd(&A.cursor[slot], 1ULL);
+        }
+        if (idx >= A.cap) { *A.err = JERR_ROWS_CAP; continue; }
+        size_t off = (size_t)slot * A.cap + idx;
+        int c = 0;
+        if (A.n_keys) A.key[off] = A.cols[c++][i];
+        for (int v = 0; v < A.n_vals; v++)
+            A.vals[(size_t)v * A.I * A.cap + off] = A.cols[c++][i];
+    }
+}
+
+/* build a chained hash multimap over one instant's left rows */
+struct JBuildArgs {
+    const int64_t *key;   /* left keys of this slot, [nl] */
+    int64_t nl;
+    int64_t *b_keys;      /* [H] open-addressing key table */
+    int32_t *b_head;      /* [H] chain heads (-1 empty) */
+    int32_t *b_next;      /* [capL] */
+    uint32_t H;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_join_build(JBuildArgs B) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < B.nl; i += stride) {
+        int64_t key = B.key[i];
+        int64_t s = table_upsert(B.b_keys, B.H, key, B.err);
+        if (s < 0) continue;
+        B.b_next[i] = atomicExch(&B.b_head[s], (int32_t)i);
+    }
+}
+
+struct JProbeArgs {
+    const int64_t *r_key;     /* right keys of this slot, [nr] */
+    const int64_t *l_vals;    /* [n_left_vals][I*cap] base of left side */
+    const int64_t *r_vals;    /* [n_right_vals][I*cap] base of right side */
+    size_t l_off, r_off;      /* slot*cap element offset */
+    size_t plane;             /* I*cap: stride between value planes */
+    int64_t nr;
+    const int64_t *b_keys;
+    const int32_t *b_head;
+    const int32_t *b_next;
+    uint32_t H;
+    int32_t n_keys, nlv, nrv;
+    uint64_t instant;
+    int64_t *out[16];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int *err;
+};
+
+__device__ inline void jemit_row(const JProbeArgs &P, int64_t key, int64_t li,
+                                 int64_t ri, int64_t r) {
+    if (r >= P.out_cap) { *P.err = JERR_OUT_CAP; return; }
+    int col = 0;
+    if (P.n_keys) P.out[col++][r] = key;
+    for (int v = 0; v < P.nlv; v++)
+        P.out[col++][r] = P.l_vals[(size_t)v * P.plane + P.l_off + li];
+    for (int v = 0; v < P.nrv; v++)
+        P.out[col++][r] = P.r_vals[(size_t)v * P.plane + P.r_off + ri];
+    P.out[col][r] = (int64_t)P.instant;
+}
+
+__global__ void __launch_bounds__(256)
+k_join_probe(JProbeArgs P) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint32_t m = P.H - 1;
+    for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         j < P.nr; j += stride) {
+        int64_t key = P.r_key[j];
+        uint64_t s = hash64((uint64_t)key) & m;
+        int32_t head = -1;
+        for (uint32_t probes = 0; probes < P.H; probes++) {
+            int64_t k = P.b_keys[s];
+            if (k == key) { head = P.b_head[s]; break; }
+            if (k == EMPTY_KEY) break;
+            s = (s + 1) & m;
+        }
+        for (int32_t li = head; li >= 0; li = P.b_next[li]) {
+            int64_t r = (int64_t)atomicAdd(P.n_out, 1ULL);
+            jemit_row(P, key, li, j, r);
+        }
+    }
+}
+
+/* n_keys == 0: cross product of the instant's sides, no atomics */
+__global__ void __launch_bounds__(256)
+k_join_cross(JProbeArgs P, int64_t nl) {
+    int64_t total = nl * P.nr;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < total; i += stride)
+        jemit_row(P, 0, i / P.nr, i % P.nr, i);
+}
+
+struct GpuJoin {
+    AmdJoinConfig cfg;
+    uint32_t I, cap, H;
+    int64_t out_cap;
+    uint64_t *tag;
+    unsigned long long *cursor[2];
+    int64_t *key[2];
+    int64_t *vals[2];          /* [n_vals][I*cap] */
+    int64_t *b_keys; int32_t *b_head; int32_t *b_next;
+    int64_t *d_out[16];
+    unsigned long long *d_n_out;
+    int *d_err;
+    int64_t *stg[2][8];        /* pinned host + device staging per side */
+    int64_t *stg_d[2][8];
+    int64_t stg_cap;
+    std::vector<std::vector<int64_t>> host_out;
+    int out_cols;
+    int has_wm; uint64_t wm;
+    int64_t emitted_device_rows;
+    hipStream_t stream;
+    char err_msg[512];
+};
+
+#define JHIP(o, call)                                                         \
+    do {                                                                      \
+        hipError_t _e = (call);                                               \
+        if (_e != hipSuccess) {                                               \
+            snprintf((o)->err_msg, sizeof (o)->err_msg, "%s:%d hip: %s",      \
+                     __FILE__, __LINE__, hipGetErrorString(_e));              \
+            return 1;                                                         \
+        }                                                                     \
+    } while (0)
+
+API void *arroyo_amd_join_create(const AmdJoinConfig *cfg) {
+    if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->n_left_vals < 0 ||
+        cfg->n_right_vals < 0 || cfg->n_left_vals > 4 ||
+        cfg->n_right_vals > 4) {
+        snprintf(g_err, sizeof g_err, "invalid join config");
+        return nullptr;
+    }
+    GpuJoin *o = new GpuJoin();
+    o->cfg = *cfg;
+    o->I = cfg->instants ? cfg->instants : 128;
+    o->cap = 1u << (cfg->log2_rows_cap ? cfg->log2_rows_cap : 15);
+    o->H = o->cap * 2;
+    o->out_cap = 1ll << (cfg->log2_out_cap ? cfg->log2_out_cap : 20);
+    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1;
+    if (hipSetDevice(cfg->device) != hipSuccess) {
+        snprintf(g_err, sizeof g_err,
+                 "hipSetDevice(%d) failed: no HIP device (no CPU fallback)",
+                 cfg->device);
+        delete o;
+        return nullptr;
+    }
+    hipError_t e;
+    auto fail = [&](const char *what, hipError_t e2) {
+        snprintf(g_err, sizeof g_err, "%s: %s", what, hipGetErrorString(e2));
+        delete o;
+        return nullptr;
+    };
+#define JALLOC(p, bytes)                                                      \
+    if ((e = hipMalloc((void **)&(p), (bytes))) != hipSuccess)                \
+        return fail(#p, e);
+    size_t plane = (size_t)o->I * o->cap;
+    JALLOC(o->tag, (size_t)o->I * 8);
+    for (int s = 0; s < 2; s++) {
+        int nv = s == 0 ? cfg->n_left_vals : cfg->n_right_vals;
+        JALLOC(o->cursor[s], (size_t)o->I * 8);
+        if (cfg->n_keys) JALLOC(o->key[s], plane * 8);
+        if (nv) JALLOC(o->vals[s], (size_t)nv * plane * 8);
+    }
+    JALLOC(o->b_keys, (size_t)o->H * 8);
+    JALLOC(o->b_head, (size_t)o->H * 4);
+    JALLOC(o->b_next, (size_t)o->cap * 4);
+    for (int i = 0; i < o->out_cols && i < 16; i++)
+        JALLOC(o->d_out[i], (size_t)o->out_cap * 8);
+    JALLOC(o->d_n_out, 8);
+    JALLOC(o->d_err, 4);
+#undef JALLOC
+    hipMemset(o->tag, 0xFF, (size_t)o->I * 8);
+    for (int s = 0; s < 2; s++)
+        hipMemset(o->cursor[s], 0, (size_t)o->I * 8);
+    hipMemset(o->d_err, 0, 4);
+    hipStreamCreate(&o->stream);
+    o->stg_cap = 1 << 20;
+    for (int s = 0; s < 2; s++) {
+        int nc = cfg->n_keys + (s == 0 ? cfg->n_left_vals : cfg->n_right_vals) + 1;
+        for (int c = 0; c < nc; c++) {
+            if (hipHostMalloc((void **)&o->stg[s][c], (size_t)o->stg_cap * 8) !=
+                    hipSuccess ||
+                hipMalloc((void **)&o->stg_d[s][c], (size_t)o->stg_cap * 8) !=
+                    hipSuccess) {
+                snprintf(g_err, sizeof g_err, "join staging alloc failed");
+                delete o;
+                return nullptr;
+            }
+        }
+    }
+    o->host_out.resize(o->out_cols);
+    return o;
+}
+
+API const char *arroyo_amd_join_last_error(void *h) {
+    return h ? ((GpuJoin *)h)->err_msg : g_err;
+}
+
+static int join_check_err(GpuJoin *o) {
+    int e = 0;
+    JHIP(o, hipMemcpyAsync(&e, o->d_err, 4, hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipStreamSynchronize(o->stream));
+    if (!e) return 0;
+    const char *msg = e == JERR_LATE      ? "batch timestamp before watermark"
+                      : e == JERR_INSTANTS ? "live-instant table full; raise instants"
+                      : e == JERR_ROWS_CAP ? "per-instant row buffer full; raise log2_rows_cap"
+                      : e == JERR_OUT_CAP  ? "output buffer full; raise log2_out_cap"
+                      : e == ERR_TABLE_FULL ? "join build table full"
+                                            : "device error";
+    snprintf(o->err_msg, sizeof o->err_msg, "%s", msg);
+    return 1;
+}
+
+static int join_append_device(GpuJoin *o, int side, const int64_t *const *dcols,
+                              int64_t n_rows) {
+    if (n_rows == 0) return 0;
+    JAppendArgs A = {};
+    int nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    for (int c = 0; c < o->cfg.n_keys + nv + 1; c++) A.cols[c] = dcols[c];
+    A.n_keys = o->cfg.n_keys;
+    A.n_vals = nv;
+    A.n_rows = n_rows;
+    A.tag = o->tag;
+    A.cursor = o->cursor[side];
+    A.key = o->key[side];
+    A.vals = o->vals[side];
+    A.I = o->I;
+    A.cap = o->cap;
+    A.has_wm = o->has_wm;
+    A.wm = o->wm;
+    A.err = o->d_err;
+    int64_t want = (n_rows + 255) / 256;
+    int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
+    hipLaunchKernelGGL(k_join_append, dim3(blocks), dim3(256), 0, o->stream, A);
+    JHIP(o, hipGetLastError());
+    return 0;
+}
+
+API int arroyo_amd_join_process_batch(void *h, int32_t side,
+                                      const int64_t *const *cols,
+                                      int32_t n_cols, int64_t n_rows) {
+    GpuJoin *o = (GpuJoin *)h;
+    int nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    if (n_cols != o->cfg.n_keys + nv + 1) {
+        snprintf(o->err_msg, sizeof o->err_msg, "side %d expects %d cols",
+                 side, o->cfg.n_keys + nv + 1);
+        return 1;
+    }
+    int64_t done = 0;
+    while (done < n_rows) {
+        int64_t take = n_rows - done;
+        if (take > o->stg_cap) take = o->stg_cap;
+        const int64_t *dcols[6];
+        for (int c = 0; c < n_cols; c++) {
+            memcpy(o->stg[side][c], cols[c] + done, (size_t)take * 8);
+            JHIP(o, hipMemcpyAsync(o->stg_d[side][c], o->stg[side][c],
+                                   (size_t)take * 8, hipMemcpyHostToDevice,
+                                   o->stream));
+            dcols[c] = o->stg_d[side][c];
+        }
+        if (join_append_device(o, side, dcols, take)) return 1;
+        /* staging reused next iteration: wait for this append to finish */
+        JHIP(o, hipStreamSynchronize(o->stream));
+        done += take;
+    }
+    return 0;
+}
+
+API int arroyo_amd_join_process_batch_device(void *h, int32_t side,
+                                             const int64_t *const *dcols,
+                                             int32_t n_cols, int64_t n_rows) {
+    GpuJoin *o = (GpuJoin *)h;
+    int nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    if (n_cols != o->cfg.n_keys + nv + 1) {
+        snprintf(o->err_msg, sizeof o->err_msg, "side %d expects %d cols",
+                 side, o->cfg.n_keys + nv + 1);
+        return 1;
+    }
+    return join_append_device(o, side, dcols, n_rows);
+}
+
+static int join_fire(GpuJoin *o, uint32_t slot, uint64_t instant, uint64_t nl,
+                     uint64_t nr) {
+    size_t plane = (size_t)o->I * o->cap;
+    JHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    JProbeArgs P = {};
+    P.r_key = o->cfg.n_keys ? o->key[1] + (size_t)slot * o->cap : nullptr;
+    P.l_vals = o->vals[0];
+    P.r_vals = o->vals[1];
+    P.l_off = (size_t)slot * o->cap;
+    P.r_off = (size_t)slot * o->cap;
+    P.plane = plane;
+    P.nr = (int64_t)nr;
+    P.b_keys = o->b_keys;
+    P.b_head = o->b_head;
+    P.b_next = o->b_next;
+    P.H = o->H;
+    P.n_keys = o->cfg.n_keys;
+    P.nlv = o->cfg.n_left_vals;
+    P.nrv = o->cfg.n_right_vals;
+    P.instant = instant;
+    for (int i = 0; i < o->out_cols && i < 16; i++) P.out[i] = o->d_out[i];
+    P.n_out = o->d_n_out;
+    P.out_cap = o->out_cap;
+    P.err = o->d_err;
+    unsigned long long n = 0;
+    if (o->cfg.n_keys == 0) {
+        int64_t total = (int64_t)nl * (int64_t)nr;
+        if (total) {
+            if (total > o->out_cap) {
+                snprintf(o->err_msg, sizeof o->err_msg,
+                         "output buffer full; raise log2_out_cap");
+                return 1;
+            }
+            int64_t want = (total + 255) / 256;
+            int blocks = (int)(want > 1024 ? 1024 : want);
+            hipLaunchKernelGGL(k_join_cross, dim3(blocks), dim3(256), 0,
+                               o->stream, P, (int64_t)nl);
+            JHIP(o, hipGetLastError());
+        }
+        n = (unsigned long long)((int64_t)nl * (int64_t)nr);
+    } else if (nl && nr) {
+        JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8, o->stream));
+        JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4, o->stream));
+        JBuildArgs B = {};
+        B.key = o->key[0] + (size_t)slot * o->cap;
+        B.nl = (int64_t)nl;
+        B.b_keys = o->b_keys;
+        B.b_head = o->b_head;
+        B.b_next = o->b_next;
+        B.H = o->H;
+        B.err = o->d_err;
+        int64_t want = ((int64_t)nl + 255) / 256;
+        int blocks = (int)(want > 1024 ? 1024 : want);
+        hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(256), 0, o->stream,
+                           B);
+        JHIP(o, hipGetLastError());
+        want = ((int64_t)nr + 255) / 256;
+        blocks = (int)(want > 1024 ? 1024 : want);
+        hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0, o->stream,
+                           P);
+        JHIP(o, hipGetLastError());
+        JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                               o->stream));
+    }
+    /* retire the instant slot */
+    JHIP(o, hipMemsetAsync(o->tag + slot, 0xFF, 8, o->stream));
+    JHIP(o, hipMemsetAsync(o->cursor[0] + slot, 0, 8, o->stream));
+    JHIP(o, hipMemsetAsync(o->cursor[1] + slot, 0, 8, o->stream));
+    JHIP(o, hipStreamSynchronize(o->stream));
+    if (n == 0) return 0;
+    if ((int64_t)n > o->out_cap) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "output buffer full; raise log2_out_cap");
+        return 1;
+    }
+    if (o->cfg.emit_to_host) {
+        for (int i = 0; i < o->out_cols; i++) {
+            size_t old = o->host_out[i].size();
+            o->host_out[i].resize(old + n);
+            JHIP(o, hipMemcpyAsync(o->host_out[i].data() + old, o->d_out[i],
+                                   (size_t)n * 8, hipMemcpyDeviceToHost,
+                                   o->stream));
+        }
+        JHIP(o, hipStreamSynchronize(o->stream));
+    } else {
+        o->emitted_device_rows += (int64_t)n;
+    }
+    return 0;
+}
+
+API int arroyo_amd_join_handle_watermark(void *h, uint64_t wm,
+                                         AmdOutBatch *out) {
+    GpuJoin *o = (GpuJoin *)h;
+    if (join_check_err(o)) return 1;
+    o->has_wm = 1;
+    o->wm = wm;
+    /* snapshot live instants + row counts */
+    std::vector<uint64_t> tags(o->I);
+    std::vector<unsigned long long> c0(o->I), c1(o->I);
+    JHIP(o, hipMemcpyAsync(tags.data(), o->tag, (size_t)o->I * 8,
+                           hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipMemcpyAsync(c0.data(), o->cursor[0], (size_t)o->I * 8,
+                           hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipMemcpyAsync(c1.data(), o->cursor[1], (size_t)o->I * 8,
+                           hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipStreamSynchronize(o->stream));
+    std::map<uint64_t, uint32_t> fired;  /* timestamp order, :265-281 */
+    for (uint32_t s = 0; s < o->I; s++)
+        if (tags[s] != EMPTY_TAG && tags[s] < wm) fired[tags[s]] = s;
+    for (auto &kv : fired)
+        if (join_fire(o, kv.second, kv.first, c0[kv.second], c1[kv.second]))
+            return 1;
+    if (out) {
+        memset(out, 0, sizeof *out);
+        int64_t n = o->host_out.empty() ? 0 : (int64_t)o->host_out[0].size();
+        out->n_rows = n;
+        out->n_cols = o->out_cols;
+        out->cols = (void **)calloc(o->out_cols, sizeof(void *));
+        out->is_f64 = (int32_t *)calloc(o->out_cols, sizeof(int32_t));
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+            if (n) memcpy(out->cols[i], o->host_out[i].data(), (size_t)n * 8);
+        }
+        for (auto &v : o->host_out) v.clear();
+    }
+    return 0;
+}
+
+/* drain one side's buffered rows: [key?, vals..., _timestamp] */
+API int arroyo_amd_join_checkpoint_drain(void *h, int32_t side,
+                                         AmdOutBatch *out) {
+    GpuJoin *o = (GpuJoin *)h;
+    if (join_check_err(o)) return 1;
+    int nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    int ncols = o->cfg.n_keys + nv + 1;
+    std::vector<uint64_t> tags(o->I);
+    std::vector<unsigned long long> cur(o->I);
+    JHIP(o, hipMemcpyAsync(tags.data(), o->tag, (size_t)o->I * 8,
+                           hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipMemcpyAsync(cur.data(), o->cursor[side], (size_t)o->I * 8,
+                           hipMemcpyDeviceToHost, o->stream));
+    JHIP(o, hipStreamSynchronize(o->stream));
+    int64_t total = 0;
+    for (uint32_t s = 0; s < o->I; s++)
+        if (tags[s] != EMPTY_TAG) total += (int64_t)cur[s];
+    memset(out, 0, sizeof *out);
+    out->n_rows = total;
+    out->n_cols = ncols;
+    out->cols = (void **)calloc(ncols, sizeof(void *));
+    out->is_f64 = (int32_t *)calloc(ncols, sizeof(int32_t));
+    for (int i = 0; i < ncols; i++)
+        out->cols[i] = malloc((size_t)(total ? total : 1) * 8);
+    size_t plane = (size_t)o->I * o->cap;
+    int64_t r = 0;
+    for (uint32_t s = 0; s < o->I; s++) {
+        if (tags[s] == EMPTY_TAG || cur[s] == 0) continue;
+        int64_t n = (int64_t)cur[s];
+        int col = 0;
+        if (o->cfg.n_keys) {
+            JHIP(o, hipMemcpyAsync((int64_t *)out->cols[col] + r,
+                                   o->key[side] + (size_t)s * o->cap,
+                                   (size_t)n * 8, hipMemcpyDeviceToHost,
+                                   o->stream));
+            col++;
+        }
+        for (int v = 0; v < nv; v++, col++)
+            JHIP(o, hipMemcpyAsync(
+                        (int64_t *)out->cols[col] + r,
+                        o->vals[side] + (size_t)v * plane + (size_t)s * o->cap,
+                        (size_t)n * 8, hipMemcpyDeviceToHost, o->stream));
+        JHIP(o, hipStreamSynchronize(o->stream));
+        for (int64_t j = 0; j < n; j++)
+            ((int64_t *)out->cols[ncols - 1])[r + j] = (int64_t)tags[s];
+        r += n;
+    }
+    return 0;
+}
+
+API void arroyo_amd_join_destroy(void *h) {
+    GpuJoin *o = (GpuJoin *)h;
+    if (!o) return;
+    hipStreamSynchronize(o->stream);
+    hipFree(o->tag);
+    for (int s = 0; s < 2; s++) {
+        hipFree(o->cursor[s]);
+        hipFree(o->key[s]);
+        hipFree(o->vals[s]);
+        int nc = o->cfg.n_keys +
+                 (s == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals) + 1;
+        for (int c = 0; c < nc; c++) {
+            hipHostFree(o->stg[s][c]);
+            hipFree(o->stg_d[s][c]);
+        }
+    }
+    hipFree(o->b_keys);
+    hipFree(o->b_head);
+    hipFree(o->b_next);
+    for (int i = 0; i < o->out_cols && i < 16; i++) hipFree(o->d_out[i]);
+    hipFree(o->d_n_out);
+    hipFree(o->d_err);
+    hipStreamDestroy(o->stream);
+    delete o;
+}

@@ -115,6 +115,11 @@ def make_op(cfg):
     return GpuWindowOp(cfg)
 
 
+def make_join_op(cfg):
+    from arroyo_amd.cabi import JoinOp
+    return JoinOp(lib(), "arroyo_amd_", cfg)
+
+
 def partition_device(d_keys, d_vals, d_ts, n, n_parts, d_out_keys, d_out_vals,
                      d_out_ts):
     counts = (ctypes.c_uint64 * n_parts)()

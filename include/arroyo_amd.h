@@ -111,6 +111,33 @@ int arroyo_amd_partition(const int64_t *d_keys, const int64_t *d_vals,
                          int64_t *d_out_keys, int64_t *d_out_vals,
                          int64_t *d_out_ts, uint64_t *h_counts);
 
+/* ---- instant (windowed stream-stream) join ----------------------------
+ * Replaces InstantJoin (crates/arroyo-worker/src/arrow/instant_join.rs)
+ * behind the same ArrowOperator surface:
+ *   join_create          <-> InstantJoinConstructor::with_config (:372-412)
+ *   join_process_batch   <-> process_batch_index left/right routing
+ *                            (:241-255 -> process_side :109-172); side 0 =
+ *                            left, 1 = right; cols [key?, vals, _timestamp]
+ *   join_handle_watermark<-> handle_watermark (:256-283): fires every
+ *                            instant < wm in timestamp order; out columns
+ *                            [key?, left vals..., right vals..., _timestamp]
+ *   join_checkpoint_drain<-> handle_checkpoint (:285-303): one side's
+ *                            buffered rows
+ *   join_restore          = process_batch (on_start re-processes the
+ *                            drained batches, :205-230)                  */
+void *arroyo_amd_join_create(const AmdJoinConfig *cfg);
+int arroyo_amd_join_process_batch(void *h, int32_t side,
+                                  const int64_t *const *cols, int32_t n_cols,
+                                  int64_t n_rows);
+int arroyo_amd_join_process_batch_device(void *h, int32_t side,
+                                         const int64_t *const *dcols,
+                                         int32_t n_cols, int64_t n_rows);
+int arroyo_amd_join_handle_watermark(void *h, uint64_t watermark_nanos,
+                                     AmdOutBatch *out);
+int arroyo_amd_join_checkpoint_drain(void *h, int32_t side, AmdOutBatch *out);
+void arroyo_amd_join_destroy(void *h);
+const char *arroyo_amd_join_last_error(void *h);
+
 #ifdef __cplusplus
 }
 #endif

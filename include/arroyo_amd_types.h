@@ -53,6 +53,29 @@ typedef struct {
     int32_t  emit_to_host;    /* GPU path: 1 = outputs copied to host memory */
 } AmdWindowConfig;
 
+/* Instant (windowed stream-stream) join configuration.  Mirrors
+ * api::JoinOperator as decoded by InstantJoinConstructor
+ * (crates/arroyo-worker/src/arrow/instant_join.rs:372-412): two keyed input
+ * schemas and a join plan.  The serialized DataFusion HashJoinExec is
+ * replaced by its information content for the supported shape: equi-join on
+ * the (single i64) key column within each exact `_timestamp` instant
+ * (rows are routed to per-instant execs, instant_join.rs:109-172; fired in
+ * timestamp order when the watermark passes, :256-283).  n_keys=0 means the
+ * join condition is the window/instant itself (e.g. windowed_inner_join.sql:
+ * ON dropoffs.window = pickups.window) and each instant emits the cross
+ * product of its sides.  Input batch columns per side: [key?, vals...,
+ * _timestamp]; output: [key?, left vals..., right vals..., _timestamp]. */
+typedef struct {
+    int32_t  n_keys;            /* 0 or 1 */
+    int32_t  n_left_vals;
+    int32_t  n_right_vals;
+    uint32_t log2_rows_cap;     /* per-instant per-side row capacity (GPU) */
+    uint32_t instants;          /* live-instant slots (GPU, power of two) */
+    uint32_t log2_out_cap;      /* output rows per fire (GPU) */
+    int32_t  device;
+    int32_t  emit_to_host;
+} AmdJoinConfig;
+
 /* Output batch, allocated by the callee; free with *_free_out.
  * Column order: [key (if n_keys)], agg outputs (one column per agg),
  * window_start, window_end, _timestamp.  All columns are 8-byte elements;

@@ -32,3 +32,8 @@ def lib():
 def make_op(cfg):
     from arroyo_amd.cabi import WindowOp
     return WindowOp(lib(), "oracle_", cfg)
+
+
+def make_join_op(cfg):
+    from arroyo_amd.cabi import JoinOp
+    return JoinOp(lib(), "oracle_", cfg)

@@ -518,3 +518,260 @@ ORACLE_API void oracle_destroy(void *h) {
     free(o->out);
     free(o);
 }
+
+/* ================================================================== */
+/* Instant (windowed stream-stream) join oracle.
+ *
+ * Restated from crates/arroyo-worker/src/arrow/instant_join.rs:
+ *   - rows are routed to a per-exact-timestamp exec (process_side :109-172:
+ *     batches spanning several instants are sort+partitioned; we route
+ *     row-by-row, same result);
+ *   - a batch whose min timestamp is behind the watermark is a fatal error
+ *     (:129-139 panics);
+ *   - on watermark, every instant < watermark fires in timestamp order
+ *     (BTreeMap pop_first loop, :265-281) and emits the join of its two
+ *     sides.  The join itself is the reference's HashJoinExec built on the
+ *     left side and probed with the right (LockedJoinPair,
+ *     arroyo-planner/src/physical.rs:177-268); here: equi-join on the i64
+ *     key, or the cross product when n_keys == 0 (join on the instant
+ *     itself, e.g. windowed_inner_join.sql ON a.window = b.window);
+ *   - checkpoint drains each side's buffered rows (handle_checkpoint
+ *     :285-303 flushes the left/right ExpiringTimeKeyTables);
+ *   - restore re-processes the drained batches (on_start :205-230).
+ */
+
+typedef struct {
+    int64_t cap, n;
+    int64_t **cols;          /* [n_cols][cap]; n_cols = n_keys + n_vals */
+    int n_cols;
+} RowBuf;
+
+typedef struct {
+    uint64_t instant;
+    RowBuf l, r;
+} JInstant;
+
+typedef struct {
+    AmdJoinConfig cfg;
+    int n_inst, cap_inst;
+    JInstant *inst;          /* sorted by instant */
+    int has_wm; uint64_t wm;
+    int out_cols;
+    int64_t out_rows, out_cap;
+    int64_t **out;
+    char err[256];
+} JOp;
+
+static void rowbuf_init(RowBuf *b, int n_cols) {
+    memset(b, 0, sizeof *b);
+    b->n_cols = n_cols;
+    b->cols = calloc((size_t)n_cols, sizeof(int64_t *));
+}
+
+static void rowbuf_free(RowBuf *b) {
+    for (int c = 0; c < b->n_cols; c++) free(b->cols[c]);
+    free(b->cols);
+}
+
+static void rowbuf_push(RowBuf *b, const int64_t *vals) {
+    if (b->n == b->cap) {
+        b->cap = b->cap ? b->cap * 2 : 64;
+        for (int c = 0; c < b->n_cols; c++)
+            b->cols[c] = realloc(b->cols[c], (size_t)b->cap * 8);
+    }
+    for (int c = 0; c < b->n_cols; c++) b->cols[c][b->n] = vals[c];
+    b->n++;
+}
+
+static JInstant *jinstant_get(JOp *o, uint64_t t) {
+    int lo = 0, hi = o->n_inst;
+    while (lo < hi) { int m = (lo + hi) / 2; if (o->inst[m].instant < t) lo = m + 1; else hi = m; }
+    if (lo < o->n_inst && o->inst[lo].instant == t) return &o->inst[lo];
+    if (o->n_inst == o->cap_inst) {
+        o->cap_inst = o->cap_inst ? o->cap_inst * 2 : 16;
+        o->inst = realloc(o->inst, (size_t)o->cap_inst * sizeof(JInstant));
+    }
+    memmove(o->inst + lo + 1, o->inst + lo,
+            (size_t)(o->n_inst - lo) * sizeof(JInstant));
+    o->inst[lo].instant = t;
+    rowbuf_init(&o->inst[lo].l, o->cfg.n_keys + o->cfg.n_left_vals);
+    rowbuf_init(&o->inst[lo].r, o->cfg.n_keys + o->cfg.n_right_vals);
+    o->n_inst++;
+    return &o->inst[lo];
+}
+
+ORACLE_API void *oracle_join_create(const AmdJoinConfig *cfg) {
+    if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 ||
+        cfg->n_left_vals < 0 || cfg->n_right_vals < 0)
+        return NULL;
+    JOp *o = calloc(1, sizeof(JOp));
+    o->cfg = *cfg;
+    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1;
+    o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
+    return o;
+}
+
+ORACLE_API const char *oracle_join_last_error(void *h) {
+    return h ? ((JOp *)h)->err : "null handle";
+}
+
+ORACLE_API int oracle_join_process_batch(void *h, int32_t side,
+                                         const int64_t *const *cols,
+                                         int32_t n_cols, int64_t n_rows) {
+    JOp *o = h;
+    const AmdJoinConfig *c = &o->cfg;
+    int nv = side == 0 ? c->n_left_vals : c->n_right_vals;
+    int want = c->n_keys + nv + 1;
+    if (n_cols != want) {
+        snprintf(o->err, sizeof o->err, "side %d expects %d cols, got %d",
+                 side, want, n_cols);
+        return 1;
+    }
+    const int64_t *ts = cols[n_cols - 1];
+    int64_t tmp[64];
+    for (int64_t r = 0; r < n_rows; r++) {
+        if (o->has_wm && (uint64_t)ts[r] < o->wm) {
+            /* instant_join.rs:129-139 panics on pre-watermark data */
+            snprintf(o->err, sizeof o->err,
+                     "batch with timestamp %lld before the watermark %llu",
+                     (long long)ts[r], (unsigned long long)o->wm);
+            return 1;
+        }
+        JInstant *in = jinstant_get(o, (uint64_t)ts[r]);
+        RowBuf *b = side == 0 ? &in->l : &in->r;
+        for (int cidx = 0; cidx < n_cols - 1; cidx++) tmp[cidx] = cols[cidx][r];
+        rowbuf_push(b, tmp);
+    }
+    return 0;
+}
+
+static void jout_reserve(JOp *o, int64_t add) {
+    if (o->out_rows + add <= o->out_cap) return;
+    int64_t ncap = o->out_cap ? o->out_cap : 1024;
+    while (ncap < o->out_rows + add) ncap *= 2;
+    for (int i = 0; i < o->out_cols; i++)
+        o->out[i] = realloc(o->out[i], (size_t)ncap * 8);
+    o->out_cap = ncap;
+}
+
+static void jemit(JOp *o, const JInstant *in, int64_t li, int64_t ri) {
+    const AmdJoinConfig *c = &o->cfg;
+    jout_reserve(o, 1);
+    int64_t r = o->out_rows++;
+    int col = 0;
+    if (c->n_keys) o->out[col++][r] = in->l.cols[0][li];
+    for (int v = 0; v < c->n_left_vals; v++)
+        o->out[col++][r] = in->l.cols[c->n_keys + v][li];
+    for (int v = 0; v < c->n_right_vals; v++)
+        o->out[col++][r] = in->r.cols[c->n_keys + v][ri];
+    o->out[col][r] = (int64_t)in->instant;
+}
+
+static void fire_instant(JOp *o, JInstant *in) {
+    const AmdJoinConfig *c = &o->cfg;
+    if (c->n_keys == 0) {
+        for (int64_t li = 0; li < in->l.n; li++)
+            for (int64_t ri = 0; ri < in->r.n; ri++)
+                jemit(o, in, li, ri);
+        return;
+    }
+    /* hash multimap over the left (build) side, probe with the right:
+     * HashJoinExec via LockedJoinPair, planner/physical.rs:177-268 */
+    if (in->l.n == 0 || in->r.n == 0) return;
+    int64_t H = 64;
+    while (H < in->l.n * 2) H <<= 1;
+    int64_t *head = malloc((size_t)H * 8);
+    int64_t *next = malloc((size_t)in->l.n * 8);
+    for (int64_t i = 0; i < H; i++) head[i] = -1;
+    for (int64_t li = 0; li < in->l.n; li++) {
+        uint64_t s = hash64((uint64_t)in->l.cols[0][li]) & (uint64_t)(H - 1);
+        next[li] = head[s];
+        head[s] = li;
+    }
+    for (int64_t ri = 0; ri < in->r.n; ri++) {
+        int64_t key = in->r.cols[0][ri];
+        uint64_t s = hash64((uint64_t)key) & (uint64_t)(H - 1);
+        for (int64_t li = head[s]; li >= 0; li = next[li])
+            if (in->l.cols[0][li] == key) jemit(o, in, li, ri);
+    }
+    free(head);
+    free(next);
+}
+
+static void jbuild_out(JOp *o, AmdOutBatch *out) {
+    memset(out, 0, sizeof *out);
+    out->n_rows = o->out_rows;
+    out->n_cols = o->out_cols;
+    out->cols = calloc((size_t)o->out_cols, sizeof(void *));
+    out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
+    for (int i = 0; i < o->out_cols; i++) {
+        out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
+        if (o->out_rows)
+            memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
+    }
+    o->out_rows = 0;
+}
+
+ORACLE_API int oracle_join_handle_watermark(void *h, uint64_t wm,
+                                            AmdOutBatch *out) {
+    JOp *o = h;
+    o->has_wm = 1;
+    o->wm = wm;
+    int fired = 0;
+    while (fired < o->n_inst && o->inst[fired].instant < wm) {
+        fire_instant(o, &o->inst[fired]);
+        rowbuf_free(&o->inst[fired].l);
+        rowbuf_free(&o->inst[fired].r);
+        fired++;
+    }
+    if (fired) {
+        memmove(o->inst, o->inst + fired,
+                (size_t)(o->n_inst - fired) * sizeof(JInstant));
+        o->n_inst -= fired;
+    }
+    if (out) jbuild_out(o, out);
+    return 0;
+}
+
+/* drain one side's buffered rows: [key?, vals..., _timestamp] */
+ORACLE_API int oracle_join_checkpoint_drain(void *h, int32_t side,
+                                            AmdOutBatch *out) {
+    JOp *o = h;
+    const AmdJoinConfig *c = &o->cfg;
+    int nv = side == 0 ? c->n_left_vals : c->n_right_vals;
+    int ncols = c->n_keys + nv + 1;
+    int64_t total = 0;
+    for (int i = 0; i < o->n_inst; i++)
+        total += side == 0 ? o->inst[i].l.n : o->inst[i].r.n;
+    memset(out, 0, sizeof *out);
+    out->n_rows = total;
+    out->n_cols = ncols;
+    out->cols = calloc((size_t)ncols, sizeof(void *));
+    out->is_f64 = calloc((size_t)ncols, sizeof(int32_t));
+    for (int i = 0; i < ncols; i++)
+        out->cols[i] = malloc((size_t)(total ? total : 1) * 8);
+    int64_t r = 0;
+    for (int i = 0; i < o->n_inst; i++) {
+        RowBuf *b = side == 0 ? &o->inst[i].l : &o->inst[i].r;
+        for (int64_t j = 0; j < b->n; j++) {
+            for (int cidx = 0; cidx < ncols - 1; cidx++)
+                ((int64_t *)out->cols[cidx])[r] = b->cols[cidx][j];
+            ((int64_t *)out->cols[ncols - 1])[r] = (int64_t)o->inst[i].instant;
+            r++;
+        }
+    }
+    return 0;
+}
+
+ORACLE_API void oracle_join_destroy(void *h) {
+    JOp *o = h;
+    if (!o) return;
+    for (int i = 0; i < o->n_inst; i++) {
+        rowbuf_free(&o->inst[i].l);
+        rowbuf_free(&o->inst[i].r);
+    }
+    free(o->inst);
+    for (int i = 0; i < o->out_cols; i++) free(o->out[i]);
+    free(o->out);
+    free(o);
+}
