@@ -1688,13 +1688,18 @@ k_join_append(JAppendArgs A) {
     }
 }
 
-/* build a chained hash multimap over one instant's left rows */
+/* build a chained hash multimap over one instant's left rows.  `base` is
+ * the slot's global row offset (slot*cap): chain links are global indices so
+ * several slots holding the same instant (possible after slot recycling — a
+ * retired slot's tag hole lets a later append re-claim an earlier probe
+ * position) can be built into ONE map and joined as one logical instant. */
 struct JBuildArgs {
     const int64_t *key;   /* left keys of this slot, [nl] */
     int64_t nl;
+    int64_t base;         /* slot*cap: global index of this slot's row 0 */
     int64_t *b_keys;      /* [H] open-addressing key table */
     int32_t *b_head;      /* [H] chain heads (-1 empty) */
-    int32_t *b_next;      /* [capL] */
+    int32_t *b_next;      /* [I*cap], indexed by global row index */
     uint32_t H;
     int *err;
 };
@@ -1707,7 +1712,8 @@ k_join_build(JBuildArgs B) {
         int64_t key = B.key[i];
         int64_t s = table_upsert(B.b_keys, B.H, key, B.err);
         if (s < 0) continue;
-        B.b_next[i] = atomicExch(&B.b_head[s], (int32_t)i);
+        B.b_next[B.base + i] =
+            atomicExch(&B.b_head[s], (int32_t)(B.base + i));
     }
 }
 
@@ -1764,14 +1770,16 @@ k_join_probe(JProbeArgs P) {
     }
 }
 
-/* n_keys == 0: cross product of the instant's sides, no atomics */
+/* n_keys == 0: cross product of the instant's sides, no atomics;
+ * `out_base` offsets the output rows so one logical instant spread across
+ * several (left slot, right slot) pairs writes disjoint output ranges */
 __global__ void __launch_bounds__(256)
-k_join_cross(JProbeArgs P, int64_t nl) {
+k_join_cross(JProbeArgs P, int64_t nl, int64_t out_base) {
     int64_t total = nl * P.nr;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < total; i += stride)
-        jemit_row(P, 0, i / P.nr, i % P.nr, i);
+        jemit_row(P, 0, i / P.nr, i % P.nr, out_base + i);
 }
 
 struct GpuJoin {
@@ -1847,7 +1855,7 @@ API void *arroyo_amd_join_create(const AmdJoinConfig *cfg) {
     }
     JALLOC(o->b_keys, (size_t)o->H * 8);
     JALLOC(o->b_head, (size_t)o->H * 4);
-    JALLOC(o->b_next, (size_t)o->cap * 4);
+    JALLOC(o->b_next, (size_t)o->I * o->cap * 4);
     for (int i = 0; i < o->out_cols && i < 16; i++)
         JALLOC(o->d_out[i], (size_t)o->out_cap * 8);
     JALLOC(o->d_n_out, 8);
@@ -1963,18 +1971,20 @@ API int arroyo_amd_join_process_batch_device(void *h, int32_t side,
     return join_append_device(o, side, dcols, n_rows);
 }
 
-static int join_fire(GpuJoin *o, uint32_t slot, uint64_t instant, uint64_t nl,
-                     uint64_t nr) {
+/* Fire one logical instant.  `slots` is every table slot holding this
+ * instant's rows — normally one, but slot recycling can split an instant
+ * across slots (see k_join_build comment); the group joins as a whole so no
+ * cross-slot match is lost. */
+static int join_fire(GpuJoin *o, uint64_t instant,
+                     const std::vector<uint32_t> &slots,
+                     const std::vector<unsigned long long> &c0,
+                     const std::vector<unsigned long long> &c1) {
     size_t plane = (size_t)o->I * o->cap;
     JHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
     JProbeArgs P = {};
-    P.r_key = o->cfg.n_keys ? o->key[1] + (size_t)slot * o->cap : nullptr;
     P.l_vals = o->vals[0];
     P.r_vals = o->vals[1];
-    P.l_off = (size_t)slot * o->cap;
-    P.r_off = (size_t)slot * o->cap;
     P.plane = plane;
-    P.nr = (int64_t)nr;
     P.b_keys = o->b_keys;
     P.b_head = o->b_head;
     P.b_next = o->b_next;
@@ -1987,50 +1997,77 @@ static int join_fire(GpuJoin *o, uint32_t slot, uint64_t instant, uint64_t nl,
     P.n_out = o->d_n_out;
     P.out_cap = o->out_cap;
     P.err = o->d_err;
+    unsigned long long nl = 0, nr = 0;
+    for (uint32_t s : slots) { nl += c0[s]; nr += c1[s]; }
     unsigned long long n = 0;
     if (o->cfg.n_keys == 0) {
         int64_t total = (int64_t)nl * (int64_t)nr;
-        if (total) {
-            if (total > o->out_cap) {
-                snprintf(o->err_msg, sizeof o->err_msg,
-                         "output buffer full; raise log2_out_cap");
-                return 1;
-            }
-            int64_t want = (total + 255) / 256;
-            int blocks = (int)(want > 1024 ? 1024 : want);
-            hipLaunchKernelGGL(k_join_cross, dim3(blocks), dim3(256), 0,
-                               o->stream, P, (int64_t)nl);
-            JHIP(o, hipGetLastError());
+        if (total > o->out_cap) {
+            snprintf(o->err_msg, sizeof o->err_msg,
+                     "output buffer full; raise log2_out_cap");
+            return 1;
         }
-        n = (unsigned long long)((int64_t)nl * (int64_t)nr);
+        int64_t base = 0;
+        for (uint32_t ls : slots) {
+            if (!c0[ls]) continue;
+            for (uint32_t rs : slots) {
+                if (!c1[rs]) continue;
+                int64_t pair = (int64_t)c0[ls] * (int64_t)c1[rs];
+                P.l_off = (size_t)ls * o->cap;
+                P.r_off = (size_t)rs * o->cap;
+                P.nr = (int64_t)c1[rs];
+                int64_t want = (pair + 255) / 256;
+                int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
+                hipLaunchKernelGGL(k_join_cross, dim3(blocks), dim3(256), 0,
+                                   o->stream, P, (int64_t)c0[ls], base);
+                JHIP(o, hipGetLastError());
+                base += pair;
+            }
+        }
+        n = (unsigned long long)total;
     } else if (nl && nr) {
         JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8, o->stream));
         JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4, o->stream));
-        JBuildArgs B = {};
-        B.key = o->key[0] + (size_t)slot * o->cap;
-        B.nl = (int64_t)nl;
-        B.b_keys = o->b_keys;
-        B.b_head = o->b_head;
-        B.b_next = o->b_next;
-        B.H = o->H;
-        B.err = o->d_err;
-        int64_t want = ((int64_t)nl + 255) / 256;
-        int blocks = (int)(want > 1024 ? 1024 : want);
-        hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(256), 0, o->stream,
-                           B);
-        JHIP(o, hipGetLastError());
-        want = ((int64_t)nr + 255) / 256;
-        blocks = (int)(want > 1024 ? 1024 : want);
-        hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0, o->stream,
-                           P);
-        JHIP(o, hipGetLastError());
+        for (uint32_t ls : slots) {
+            if (!c0[ls]) continue;
+            JBuildArgs B = {};
+            B.key = o->key[0] + (size_t)ls * o->cap;
+            B.nl = (int64_t)c0[ls];
+            B.base = (int64_t)ls * o->cap;
+            B.b_keys = o->b_keys;
+            B.b_head = o->b_head;
+            B.b_next = o->b_next;
+            B.H = o->H;
+            B.err = o->d_err;
+            int64_t want = (B.nl + 255) / 256;
+            int blocks = (int)(want > 1024 ? 1024 : want);
+            hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(256), 0,
+                               o->stream, B);
+            JHIP(o, hipGetLastError());
+        }
+        /* chain links are global row indices: probe reads left values at
+         * l_vals[v*plane + li] directly, so l_off = 0 */
+        P.l_off = 0;
+        for (uint32_t rs : slots) {
+            if (!c1[rs]) continue;
+            P.r_key = o->key[1] + (size_t)rs * o->cap;
+            P.r_off = (size_t)rs * o->cap;
+            P.nr = (int64_t)c1[rs];
+            int64_t want = (P.nr + 255) / 256;
+            int blocks = (int)(want > 1024 ? 1024 : want);
+            hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0,
+                               o->stream, P);
+            JHIP(o, hipGetLastError());
+        }
         JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                o->stream));
     }
-    /* retire the instant slot */
-    JHIP(o, hipMemsetAsync(o->tag + slot, 0xFF, 8, o->stream));
-    JHIP(o, hipMemsetAsync(o->cursor[0] + slot, 0, 8, o->stream));
-    JHIP(o, hipMemsetAsync(o->cursor[1] + slot, 0, 8, o->stream));
+    /* retire every slot of the instant */
+    for (uint32_t s : slots) {
+        JHIP(o, hipMemsetAsync(o->tag + s, 0xFF, 8, o->stream));
+        JHIP(o, hipMemsetAsync(o->cursor[0] + s, 0, 8, o->stream));
+        JHIP(o, hipMemsetAsync(o->cursor[1] + s, 0, 8, o->stream));
+    }
     JHIP(o, hipStreamSynchronize(o->stream));
     if (n == 0) return 0;
     if ((int64_t)n > o->out_cap) {
@@ -2069,11 +2106,13 @@ API int arroyo_amd_join_handle_watermark(void *h, uint64_t wm,
     JHIP(o, hipMemcpyAsync(c1.data(), o->cursor[1], (size_t)o->I * 8,
                            hipMemcpyDeviceToHost, o->stream));
     JHIP(o, hipStreamSynchronize(o->stream));
-    std::map<uint64_t, uint32_t> fired;  /* timestamp order, :265-281 */
+    /* timestamp order (instant_join.rs:265-281); an instant may occupy
+     * several slots after recycling — fire them as one group */
+    std::map<uint64_t, std::vector<uint32_t>> fired;
     for (uint32_t s = 0; s < o->I; s++)
-        if (tags[s] != EMPTY_TAG && tags[s] < wm) fired[tags[s]] = s;
+        if (tags[s] != EMPTY_TAG && tags[s] < wm) fired[tags[s]].push_back(s);
     for (auto &kv : fired)
-        if (join_fire(o, kv.second, kv.first, c0[kv.second], c1[kv.second]))
+        if (join_fire(o, kv.first, kv.second, c0, c1))
             return 1;
     if (out) {
         memset(out, 0, sizeof *out);
