@@ -284,3 +284,262 @@ class WindowOp:
             self.close()
         except Exception:
             pass
+
+
+class AmdSessionConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_keys", ctypes.c_int32),
+        ("n_value_cols", ctypes.c_int32),
+        ("n_aggs", ctypes.c_int32),
+        ("agg_ops", ctypes.c_int32 * 8),
+        ("agg_col", ctypes.c_int32 * 8),
+        ("gap_nanos", ctypes.c_uint64),
+        ("log2_capacity", ctypes.c_uint32),
+        ("max_sessions", ctypes.c_uint32),
+        ("log2_batch_capacity", ctypes.c_uint32),
+        ("log2_out_cap", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_session_config(gap_ns, aggs, n_keys=1, n_value_cols=0,
+                        log2_capacity=16, max_sessions=8,
+                        log2_batch_capacity=14, log2_out_cap=20,
+                        device=0, emit_to_host=True):
+    cfg = AmdSessionConfig()
+    cfg.n_keys = n_keys
+    cfg.n_value_cols = n_value_cols
+    cfg.n_aggs = len(aggs)
+    for i, (op, col) in enumerate(aggs):
+        cfg.agg_ops[i] = op
+        cfg.agg_col[i] = col
+    cfg.gap_nanos = gap_ns
+    cfg.log2_capacity = log2_capacity
+    cfg.max_sessions = max_sessions
+    cfg.log2_batch_capacity = log2_batch_capacity
+    cfg.log2_out_cap = log2_out_cap
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+class AmdExpJoinConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_keys", ctypes.c_int32),
+        ("n_left_vals", ctypes.c_int32),
+        ("n_right_vals", ctypes.c_int32),
+        ("ttl_nanos", ctypes.c_uint64),
+        ("log2_capacity", ctypes.c_uint32),
+        ("log2_rows_cap", ctypes.c_uint32),
+        ("log2_out_cap", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_expjoin_config(ttl_ns, n_left_vals=0, n_right_vals=0,
+                        log2_capacity=16, log2_rows_cap=20, log2_out_cap=20,
+                        device=0, emit_to_host=True):
+    cfg = AmdExpJoinConfig()
+    cfg.n_keys = 1
+    cfg.n_left_vals = n_left_vals
+    cfg.n_right_vals = n_right_vals
+    cfg.ttl_nanos = ttl_ns
+    cfg.log2_capacity = log2_capacity
+    cfg.log2_rows_cap = log2_rows_cap
+    cfg.log2_out_cap = log2_out_cap
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+def _cols_to_ptrs(cols):
+    """Contiguous int64 copies + a void* array over them."""
+    keep = [np.ascontiguousarray(c, dtype=np.int64) for c in cols]
+    arr = (ctypes.c_void_p * len(keep))(
+        *[c.ctypes.data_as(ctypes.c_void_p).value for c in keep])
+    return keep, arr
+
+
+class SessionOp:
+    """One session-window aggregate operator behind the C ABI, mirroring
+    SessionAggregatingWindowFunc's ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/session_aggregating_window.rs)."""
+
+    def __init__(self, lib, prefix, cfg):
+        p = prefix + "session_"
+        g = lambda n: getattr(lib, p + n)
+        self._fn = {}
+        self._fn["create"] = g("create")
+        self._fn["create"].restype = ctypes.c_void_p
+        self._fn["create"].argtypes = [ctypes.POINTER(AmdSessionConfig)]
+        self._fn["process_batch"] = g("process_batch")
+        self._fn["process_batch"].restype = ctypes.c_int
+        self._fn["process_batch"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64]
+        self._fn["handle_watermark"] = g("handle_watermark")
+        self._fn["handle_watermark"].restype = ctypes.c_int
+        self._fn["handle_watermark"].argtypes = [
+            ctypes.c_void_p, ctypes.c_uint64, ctypes.POINTER(AmdOutBatch)]
+        self._fn["checkpoint_drain"] = g("checkpoint_drain")
+        self._fn["checkpoint_drain"].restype = ctypes.c_int
+        self._fn["checkpoint_drain"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(AmdOutBatch)]
+        self._fn["restore"] = g("restore")
+        self._fn["restore"].restype = ctypes.c_int
+        self._fn["restore"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64]
+        self._fn["free_out"] = getattr(lib, prefix + "free_out")
+        self._fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+        self._fn["destroy"] = g("destroy")
+        self._fn["destroy"].argtypes = [ctypes.c_void_p]
+        self._fn["last_error"] = g("last_error")
+        self._fn["last_error"].restype = ctypes.c_char_p
+        self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{p}create failed: "
+                               f"{self._fn['last_error'](None)}")
+
+    def _check(self, rc):
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def process_batch(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["process_batch"](self._h, arr, len(keep),
+                                              n_rows))
+
+    def handle_watermark(self, wm):
+        out = AmdOutBatch()
+        self._check(self._fn["handle_watermark"](self._h, wm,
+                                                 ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def checkpoint_drain(self):
+        out = AmdOutBatch()
+        self._check(self._fn["checkpoint_drain"](self._h, ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["restore"](self._h, arr, len(keep), n_rows))
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class ExpJoinOp:
+    """One TTL'd (non-windowed) join operator behind the C ABI, mirroring
+    JoinWithExpiration's ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/join_with_expiration.rs)."""
+
+    LEFT, RIGHT = 0, 1
+
+    def __init__(self, lib, prefix, cfg):
+        p = prefix + "expjoin_"
+        g = lambda n: getattr(lib, p + n)
+        self._fn = {}
+        self._fn["create"] = g("create")
+        self._fn["create"].restype = ctypes.c_void_p
+        self._fn["create"].argtypes = [ctypes.POINTER(AmdExpJoinConfig)]
+        self._fn["process_batch"] = g("process_batch")
+        self._fn["process_batch"].restype = ctypes.c_int
+        self._fn["process_batch"].argtypes = [
+            ctypes.c_void_p, ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32, ctypes.c_int64,
+            ctypes.POINTER(AmdOutBatch)]
+        self._fn["handle_watermark"] = g("handle_watermark")
+        self._fn["handle_watermark"].restype = ctypes.c_int
+        self._fn["handle_watermark"].argtypes = [ctypes.c_void_p,
+                                                 ctypes.c_uint64]
+        self._fn["expire"] = g("expire")
+        self._fn["expire"].restype = ctypes.c_int
+        self._fn["expire"].argtypes = [ctypes.c_void_p]
+        self._fn["checkpoint_drain"] = g("checkpoint_drain")
+        self._fn["checkpoint_drain"].restype = ctypes.c_int
+        self._fn["checkpoint_drain"].argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(AmdOutBatch)]
+        self._fn["restore"] = g("restore")
+        self._fn["restore"].restype = ctypes.c_int
+        self._fn["restore"].argtypes = [
+            ctypes.c_void_p, ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32, ctypes.c_int64,
+            ctypes.c_int, ctypes.c_uint64]
+        self._fn["free_out"] = getattr(lib, prefix + "free_out")
+        self._fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+        self._fn["destroy"] = g("destroy")
+        self._fn["destroy"].argtypes = [ctypes.c_void_p]
+        self._fn["last_error"] = g("last_error")
+        self._fn["last_error"].restype = ctypes.c_char_p
+        self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{p}create failed: "
+                               f"{self._fn['last_error'](None)}")
+
+    def _check(self, rc):
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def process_batch(self, side, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        out = AmdOutBatch()
+        self._check(self._fn["process_batch"](self._h, side, arr, len(keep),
+                                              n_rows, ctypes.byref(out)))
+        res = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return res
+
+    def handle_watermark(self, wm):
+        self._check(self._fn["handle_watermark"](self._h, wm))
+
+    def expire(self):
+        self._check(self._fn["expire"](self._h))
+
+    def checkpoint_drain(self, side):
+        out = AmdOutBatch()
+        self._check(self._fn["checkpoint_drain"](self._h, side,
+                                                 ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore(self, side, cols, watermark=None):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["restore"](
+            self._h, side, arr, len(keep), n_rows,
+            1 if watermark is not None else 0,
+            watermark if watermark is not None else 0))
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

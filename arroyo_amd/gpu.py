@@ -129,3 +129,13 @@ def partition_device(d_keys, d_vals, d_ts, n, n_parts, d_out_keys, d_out_vals,
     if rc != 0:
         raise RuntimeError("arroyo_amd_partition failed")
     return list(counts)
+
+
+def make_session_op(cfg):
+    from arroyo_amd.cabi import SessionOp
+    return SessionOp(lib(), "arroyo_amd_", cfg)
+
+
+def make_expjoin_op(cfg):
+    from arroyo_amd.cabi import ExpJoinOp
+    return ExpJoinOp(lib(), "arroyo_amd_", cfg)

@@ -138,6 +138,82 @@ int arroyo_amd_join_checkpoint_drain(void *h, int32_t side, AmdOutBatch *out);
 void arroyo_amd_join_destroy(void *h);
 const char *arroyo_amd_join_last_error(void *h);
 
+/* ---- session (gap) window aggregate -----------------------------------
+ * Replaces SessionAggregatingWindowFunc
+ * (crates/arroyo-worker/src/arrow/session_aggregating_window.rs) behind the
+ * same ArrowOperator surface:
+ *   session_create           <-> SessionAggregatingWindowConstructor
+ *                                ::with_config (:707-772)
+ *   session_process_batch    <-> process_batch (:849-893): late rows
+ *                                (ts < watermark) silently dropped, rest
+ *                                buffered into per-key session state
+ *   session_handle_watermark <-> handle_watermark -> advance (:76-98,
+ *                                :894-903): fires every session with
+ *                                data_end + gap < watermark; out columns
+ *                                [key?, aggs..., window_start, window_end,
+ *                                _timestamp = window_end - 1]
+ *   session_checkpoint_drain <-> handle_checkpoint (:905-921): live session
+ *                                partial states [key?, state words...,
+ *                                data_start, data_end]
+ *   session_restore          <-> on_start (:803-846): reload drained
+ *                                partial sessions
+ */
+void *arroyo_amd_session_create(const AmdSessionConfig *cfg);
+int arroyo_amd_session_process_batch(void *h, const int64_t *const *cols,
+                                     int32_t n_cols, int64_t n_rows);
+int arroyo_amd_session_process_batch_device(void *h,
+                                            const int64_t *const *dcols,
+                                            int32_t n_cols, int64_t n_rows,
+                                            uint64_t ts_offset);
+int arroyo_amd_session_handle_watermark(void *h, uint64_t watermark_nanos,
+                                        AmdOutBatch *out);
+int arroyo_amd_session_checkpoint_drain(void *h, AmdOutBatch *out);
+int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
+                               int32_t n_cols, int64_t n_rows);
+void arroyo_amd_session_destroy(void *h);
+const char *arroyo_amd_session_last_error(void *h);
+
+/* ---- non-windowed (TTL'd) stream-stream join --------------------------
+ * Replaces JoinWithExpiration
+ * (crates/arroyo-worker/src/arrow/join_with_expiration.rs) behind the same
+ * ArrowOperator surface:
+ *   expjoin_create          <-> JoinWithExpirationConstructor::with_config
+ *                               (:213-267)
+ *   expjoin_process_batch   <-> process_batch_index (:162-180) ->
+ *                               process_left/process_right (:42-108):
+ *                               inserts the batch into its side's per-key
+ *                               state AND returns the joined output rows
+ *                               [key, left vals..., right vals...,
+ *                               _timestamp = max(l_ts, r_ts)] against the
+ *                               other side's stored rows
+ *   expjoin_handle_watermark:   records the watermark (live state never
+ *                               evicts during a run, table_manager.rs:533;
+ *                               TTL applies on restore)
+ *   expjoin_expire          :   explicit eviction of rows with
+ *                               ts < watermark - ttl (bounded-memory mode;
+ *                               matches what the reference's state layer
+ *                               drops across checkpoint/restore)
+ *   expjoin_checkpoint_drain<-> tables() flush: one side's stored rows
+ *   expjoin_restore         <-> on_start: reload rows, dropping those with
+ *                               ts < watermark - ttl (get_key_time_table
+ *                               restore filter)
+ */
+void *arroyo_amd_expjoin_create(const AmdExpJoinConfig *cfg);
+int arroyo_amd_expjoin_process_batch(void *h, int32_t side,
+                                     const int64_t *const *cols,
+                                     int32_t n_cols, int64_t n_rows,
+                                     AmdOutBatch *out);
+int arroyo_amd_expjoin_handle_watermark(void *h, uint64_t watermark_nanos);
+int arroyo_amd_expjoin_expire(void *h);
+int arroyo_amd_expjoin_checkpoint_drain(void *h, int32_t side,
+                                        AmdOutBatch *out);
+int arroyo_amd_expjoin_restore(void *h, int32_t side,
+                               const int64_t *const *cols, int32_t n_cols,
+                               int64_t n_rows, int has_watermark,
+                               uint64_t watermark_nanos);
+void arroyo_amd_expjoin_destroy(void *h);
+const char *arroyo_amd_expjoin_last_error(void *h);
+
 #ifdef __cplusplus
 }
 #endif

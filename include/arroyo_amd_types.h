@@ -76,6 +76,61 @@ typedef struct {
     int32_t  emit_to_host;
 } AmdJoinConfig;
 
+/* Session (gap) window aggregate configuration.  Mirrors
+ * api::SessionWindowAggregateOperator as decoded by
+ * SessionAggregatingWindowConstructor
+ * (crates/arroyo-worker/src/arrow/session_aggregating_window.rs:707-772):
+ * gap_micros + input schema (keys first) + final aggregation plan.  The
+ * serialized DataFusion plan is replaced by the explicit aggregate spec, as
+ * for AmdWindowConfig.  Semantics (restated in oracle/arroyo_oracle.c and
+ * arroyo_amd/csrc/session.hip): per key, rows sorted by time form maximal
+ * runs where each next timestamp is strictly within (previous max + gap)
+ * (ActiveSession::add_batch :424-495); a session fires when
+ * data_end + gap < watermark (KeyComputingHolder::watermark_update
+ * :559-608); output window = [min_ts, max_ts + gap), _timestamp = end - 1
+ * (to_record_batch :316-380); rows with ts < watermark are dropped
+ * (process_batch :849-874, gt_eq filter). */
+typedef struct {
+    int32_t  n_keys;            /* 0 or 1 */
+    int32_t  n_value_cols;
+    int32_t  n_aggs;
+    int32_t  agg_ops[AMD_MAX_AGGS];
+    int32_t  agg_col[AMD_MAX_AGGS];
+    uint64_t gap_nanos;
+    uint32_t log2_capacity;     /* key slots in the session store (GPU) */
+    uint32_t max_sessions;      /* live sessions held inline per key (GPU) */
+    uint32_t log2_batch_capacity; /* per-batch pre-aggregation table (GPU) */
+    uint32_t log2_out_cap;      /* output rows per watermark (GPU) */
+    int32_t  device;
+    int32_t  emit_to_host;
+} AmdSessionConfig;
+
+/* Non-windowed (TTL'd) stream-stream join configuration.  Mirrors
+ * api::JoinOperator as decoded by JoinWithExpirationConstructor
+ * (crates/arroyo-worker/src/arrow/join_with_expiration.rs:213-267): two
+ * keyed schemas + ttl_micros + join plan.  Semantics: each incoming batch
+ * is inserted into its side's per-key state (KeyTimeView,
+ * crates/arroyo-state/src/tables/expiring_time_key_map.rs:932-1050) and
+ * immediately joined against the OTHER side's stored rows for its keys
+ * (process_left/process_right :42-108) — inner equi-join on the i64 key,
+ * output _timestamp = max(left._timestamp, right._timestamp)
+ * (post_join_timestamp_projection,
+ * crates/arroyo-planner/src/plan/join.rs:121-191).  The live in-memory view
+ * never evicts during a run (TTL filters state only on restore,
+ * table_manager.rs:533-570 passes the watermark to get_view); restore drops
+ * rows with ts < watermark - ttl. */
+typedef struct {
+    int32_t  n_keys;            /* 1 (i64 equi-join key) */
+    int32_t  n_left_vals;
+    int32_t  n_right_vals;
+    uint64_t ttl_nanos;
+    uint32_t log2_capacity;     /* key slots per side (GPU) */
+    uint32_t log2_rows_cap;     /* stored-row pool per side (GPU) */
+    uint32_t log2_out_cap;      /* output rows per process_batch (GPU) */
+    int32_t  device;
+    int32_t  emit_to_host;
+} AmdExpJoinConfig;
+
 /* Output batch, allocated by the callee; free with *_free_out.
  * Column order: [key (if n_keys)], agg outputs (one column per agg),
  * window_start, window_end, _timestamp.  All columns are 8-byte elements;

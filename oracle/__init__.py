@@ -37,3 +37,13 @@ def make_op(cfg):
 def make_join_op(cfg):
     from arroyo_amd.cabi import JoinOp
     return JoinOp(lib(), "oracle_", cfg)
+
+
+def make_session_op(cfg):
+    from arroyo_amd.cabi import SessionOp
+    return SessionOp(lib(), "oracle_", cfg)
+
+
+def make_expjoin_op(cfg):
+    from arroyo_amd.cabi import ExpJoinOp
+    return ExpJoinOp(lib(), "oracle_", cfg)

@@ -69,7 +69,8 @@ def main():
 
     for name in ("sliding_window_end", "hourly_by_event_type",
                  "tight_watermark", "most_active_driver_last_hour",
-                 "windowed_inner_join"):
+                 "windowed_inner_join", "session_window",
+                 "global_session_window", "updating_inner_join"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)

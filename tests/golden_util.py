@@ -25,7 +25,12 @@ def fmt_ts(ns):
     t = dt.fromtimestamp(s, tz=timezone.utc)
     base = t.strftime("%Y-%m-%dT%H:%M:%S")
     if frac:
-        base += (".%09d" % frac).rstrip("0")
+        digits = "%09d" % frac
+        # the reference prints fractional seconds at milli/micro/nano
+        # precision (chrono %.f): strip trailing zeros in groups of three
+        while digits.endswith("000"):
+            digits = digits[:-3]
+        base += "." + digits
     return base
 
 

@@ -1,0 +1,175 @@
+"""Session (gap) window tests: the reference's own golden vectors
+(session_window.sql / global_session_window.sql over inputs/impulse.json) and
+fuzz against an independent numpy restatement.
+
+Reference semantics: crates/arroyo-worker/src/arrow/session_aggregating_window.rs
+(per-key maximal runs where each next ts is strictly within prev-max + gap;
+a session fires when data_end + gap < watermark; window = [min_ts,
+max_ts + gap), _timestamp = end - 1; late rows ts < watermark dropped)."""
+import numpy as np
+import pytest
+
+import oracle
+from arroyo_amd import cabi
+from tests.golden_util import NS, assert_rows_match, fmt_ts, load_golden, load_inputs
+
+U64MAX = 2**64 - 1
+GAP20 = 20 * NS
+
+
+def np_sessions(key, ts, gap):
+    """Independent restatement: per key, sort ts, split where the next ts is
+    >= running-max + gap; emit (key, count, start, end, end-1)."""
+    rows = []
+    for k in np.unique(key):
+        t = np.sort(ts[key == k])
+        lo = 0
+        for i in range(1, len(t) + 1):
+            if i == len(t) or t[i] >= t[i - 1] + gap:
+                rows.append((int(k), i - lo, int(t[lo]),
+                             int(t[i - 1] + gap), int(t[i - 1] + gap - 1)))
+                lo = i
+    return sorted(rows)
+
+
+def rows_of(cols):
+    if cols is None or len(cols) == 0 or len(cols[0]) == 0:
+        return []
+    return sorted(tuple(int(c[r]) for c in cols) for r in range(len(cols[0])))
+
+
+def impulse_cols():
+    d = load_inputs()["impulse"]
+    return (np.array(d["counter"], dtype=np.int64),
+            np.array(d["ts"], dtype=np.int64))
+
+
+def run_session_window_golden(make_op):
+    """session_window.sql: SESSION(20s) COUNT(*) GROUP BY user_id where
+    user_id = 0 if counter % 10 == 0 else counter."""
+    counter, ts = impulse_cols()
+    user = np.where(counter % 10 == 0, 0, counter).astype(np.int64)
+    op = make_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1, n_value_cols=0))
+    op.process_batch([user, ts])
+    out = op.handle_watermark(U64MAX)
+    op.close()
+    got = [{"user_id": int(k), "rows": int(n), "start": fmt_ts(int(s)),
+            "end": fmt_ts(int(e))}
+           for k, n, s, e, _t in zip(*out)]
+    assert_rows_match(got, load_golden("session_window"))
+
+
+def run_global_session_window_golden(make_op):
+    """global_session_window.sql: unkeyed SESSION(20s) COUNT(*)."""
+    counter, ts = impulse_cols()
+    op = make_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=0, n_value_cols=0))
+    op.process_batch([ts])
+    out = op.handle_watermark(U64MAX)
+    op.close()
+    got = [{"rows": int(n), "start": fmt_ts(int(s)), "end": fmt_ts(int(e))}
+           for n, s, e, _t in zip(*out)]
+    assert_rows_match(got, load_golden("global_session_window"))
+
+
+def test_session_window_golden_oracle():
+    run_session_window_golden(oracle.make_session_op)
+
+
+def test_global_session_window_golden_oracle():
+    run_global_session_window_golden(oracle.make_session_op)
+
+
+def stream_fuzz(make_op, n=4000, seed=3, gap_s=5, aggs=None, n_value_cols=0):
+    """Stream a key/ts stream through the op with periodic watermarks;
+    returns all emitted rows.  Input is watermark-clean (each batch only
+    has ts >= previous watermark) so results equal the batch restatement."""
+    rng = np.random.default_rng(seed)
+    t0 = 1_600_000_000 * NS
+    gap = gap_s * NS
+    # ragged event times: bursts separated by quiet periods > gap
+    ts = t0 + np.cumsum(rng.choice(
+        [NS // 10, NS // 2, 2 * NS, 7 * NS], size=n,
+        p=[0.55, 0.3, 0.1, 0.05]).astype(np.int64))
+    key = rng.integers(0, 37, size=n).astype(np.int64)
+    vals = [rng.integers(-50, 50, size=n).astype(np.int64)
+            for _ in range(n_value_cols)]
+    aggs = aggs or [(cabi.COUNT, -1)]
+    op = make_op(cabi.make_session_config(
+        gap, aggs, n_keys=1, n_value_cols=n_value_cols))
+    got = []
+    step = n // 7
+    for b in range(0, n, step):
+        sl = slice(b, min(b + step, n))
+        cols = [key[sl]] + [v[sl] for v in vals] + [ts[sl]]
+        op.process_batch(cols)
+        wm = int(ts[sl][len(ts[sl]) // 2])  # mid-batch watermark, no late rows
+        got += rows_of(op.handle_watermark(wm))
+    got += rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    return key, ts, vals, got
+
+
+def test_session_oracle_vs_numpy_fuzz():
+    key, ts, _vals, got = stream_fuzz(oracle.make_session_op)
+    want = [(k, n, s, e, t)
+            for k, n, s, e, t in np_sessions(key, ts, 5 * NS)]
+    assert sorted(got) == sorted(want)
+
+
+def test_session_oracle_multi_agg():
+    """SUM/MIN/MAX over a value column alongside COUNT."""
+    key, ts, vals, got = stream_fuzz(
+        oracle.make_session_op, n_value_cols=1,
+        aggs=[(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.MIN, 0), (cabi.MAX, 0)])
+    v = vals[0]
+    want = []
+    for k, n, s, e, t in np_sessions(key, ts, 5 * NS):
+        m = (key == k) & (ts >= s) & (ts < e)
+        want.append((k, n, int(v[m].sum()), int(v[m].min()), int(v[m].max()),
+                     s, e, t))
+    assert sorted(got) == sorted(want)
+
+
+def test_session_oracle_drops_late_rows():
+    op = oracle.make_session_op(cabi.make_session_config(
+        2 * NS, [(cabi.COUNT, -1)], n_keys=1))
+    t0 = 1_600_000_000 * NS
+    op.process_batch([np.array([7], dtype=np.int64),
+                      np.array([t0 + 10 * NS], dtype=np.int64)])
+    op.handle_watermark(t0 + 10 * NS)
+    # ts < watermark: silently dropped (session_aggregating_window.rs:849-874)
+    op.process_batch([np.array([7], dtype=np.int64),
+                      np.array([t0 + 9 * NS], dtype=np.int64)])
+    out = op.handle_watermark(U64MAX)
+    rows = rows_of(out)
+    assert rows == [(7, 1, t0 + 10 * NS, t0 + 12 * NS, t0 + 12 * NS - 1)]
+    op.close()
+
+
+def test_session_oracle_checkpoint_roundtrip():
+    counter, ts = impulse_cols()
+    user = np.where(counter % 10 == 0, 0, counter).astype(np.int64)
+    mid = len(user) // 2
+
+    a = oracle.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    a.process_batch([user[:mid], ts[:mid]])
+    drained = a.checkpoint_drain()
+    a.close()
+    assert len(drained[0]) == mid
+
+    b = oracle.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    b.restore(drained)
+    b.process_batch([user[mid:], ts[mid:]])
+    got = rows_of(b.handle_watermark(U64MAX))
+    b.close()
+
+    c = oracle.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    c.process_batch([user, ts])
+    want = rows_of(c.handle_watermark(U64MAX))
+    c.close()
+    assert got == want
