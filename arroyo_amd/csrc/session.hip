@@ -1,0 +1,862 @@
+/* arroyo-amd session (gap) window aggregate: MI355X-native (gfx950)
+ * equivalent of SessionAggregatingWindowFunc
+ * (crates/arroyo-worker/src/arrow/session_aggregating_window.rs) behind the
+ * arroyo_amd_session_* C ABI (include/arroyo_amd.h).
+ *
+ * MI355X-first design (NOT a translation of the reference's per-key
+ * BTreeMap-of-batches + per-key DataFusion exec streams):
+ *   - Per-batch pre-aggregation (k_sess_update): one fused kernel folds each
+ *     row into a per-batch open-addressing table keyed by the session key,
+ *     accumulating (min_ts, max_ts, aggregate states) with LDS-free HBM
+ *     atomics on memset-zero-identity encoded words.  Because a batch's
+ *     event-time span is < gap (enforced; the host splits wider batches into
+ *     gap/2 buckets), a key's rows within one batch always belong to ONE
+ *     session, so (min_ts, max_ts, states) is a valid partial session —
+ *     this replaces the reference's per-row BTreeMap insertion and
+ *     ActiveSession::add_batch scan (:424-520).
+ *   - Session store merge (k_sess_merge): one thread per DISTINCT key of the
+ *     batch (the pre-agg table's occupied slots) merges its partial session
+ *     into the device-resident session store: an open-addressing key table
+ *     whose slots hold up to max_sessions live sessions inline.  Two
+ *     sessions are one iff their [start, end+gap) closures touch
+ *     (P.start < A.end + gap && A.start < P.end + gap, both strict) — the
+ *     interval form of the reference's strictly-within-gap row rule.  One
+ *     thread per distinct key means no locks and no CAS loops on session
+ *     data; only the key-slot claim is a CAS.
+ *   - Watermark firing (k_sess_fire): one thread per key slot scans its
+ *     sessions and emits every session with data_end + gap < watermark
+ *     (KeyComputingHolder::watermark_update :559-608) through a global
+ *     output cursor; output columns [key?, finals..., window_start,
+ *     window_end = data_end + gap, _timestamp = window_end - 1]
+ *     (to_record_batch :316-380).  Key slots are never un-claimed (open
+ *     addressing with deletions would corrupt probe chains); an idle key
+ *     costs its slot only.
+ *   - Late rows (ts < watermark) are dropped in the update kernel
+ *     (process_batch :849-874, gt_eq filter) — silently, as the reference
+ *     does (unlike the window operators, where pre-watermark data is a
+ *     fatal error).
+ *
+ * Parity is pinned against oracle/arroyo_oracle.c (itself pinned against
+ * the reference's session_window / global_session_window golden vectors) by
+ * tests/test_session.py.
+ */
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include "../../include/arroyo_amd_types.h"
+
+#define API extern "C" __attribute__((visibility("default")))
+
+namespace sess {
+
+#define EMPTY_KEY (-1LL)
+#define SERR_TABLE_FULL 1
+#define SERR_BATCH_FULL 2
+#define SERR_SESSIONS   3
+#define SERR_OUT_CAP    4
+
+__host__ __device__ inline uint64_t enc_min(int64_t v) {
+    return ~(((uint64_t)v) ^ 0x8000000000000000ULL);
+}
+__host__ __device__ inline int64_t dec_min(uint64_t e) {
+    return (int64_t)((~e) ^ 0x8000000000000000ULL);
+}
+__host__ __device__ inline uint64_t enc_max(int64_t v) {
+    return ((uint64_t)v) ^ 0x8000000000000000ULL;
+}
+__host__ __device__ inline int64_t dec_max(uint64_t e) {
+    return (int64_t)(e ^ 0x8000000000000000ULL);
+}
+
+__device__ inline uint64_t hash64(uint64_t x) {
+    x += 0x9e3779b97f4a7c15ULL;
+    x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ULL;
+    x = (x ^ (x >> 27)) * 0x94d049bb133111ebULL;
+    return x ^ (x >> 31);
+}
+
+/* aggregate state: n_aggs x 2 u64 words, zero = identity (memset-clear):
+ * COUNT/SUM w0 add; MIN w0 = enc_min max-domain; MAX w0 = enc_max;
+ * AVG w0 = count, w1 = bits(f64 sum) folded with CAS-add. */
+struct AggSpec {
+    int32_t n_aggs;
+    int32_t op[AMD_MAX_AGGS];
+    int32_t col[AMD_MAX_AGGS];
+};
+
+__device__ inline void atomic_fold(uint64_t *st, const AggSpec &a,
+                                   const int64_t *const *vcols, int64_t r) {
+    for (int i = 0; i < a.n_aggs; i++) {
+        uint64_t *w = st + 2 * i;
+        int64_t v = a.col[i] >= 0 ? vcols[i][r] : 0;
+        switch (a.op[i]) {
+        case AMD_AGG_COUNT:
+            atomicAdd((unsigned long long *)w, 1ULL);
+            break;
+        case AMD_AGG_SUM:
+            atomicAdd((unsigned long long *)w, (unsigned long long)v);
+            break;
+        case AMD_AGG_MIN:
+            atomicMax((unsigned long long *)w,
+                      (unsigned long long)enc_min(v));
+            break;
+        case AMD_AGG_MAX:
+            atomicMax((unsigned long long *)w,
+                      (unsigned long long)enc_max(v));
+            break;
+        case AMD_AGG_AVG: {
+            atomicAdd((unsigned long long *)w, 1ULL);
+            double dv = (double)v;
+            unsigned long long old = w[1], assumed;
+            do {
+                assumed = old;
+                double cur;
+                memcpy(&cur, &assumed, 8);
+                cur += dv;
+                unsigned long long nv2;
+                memcpy(&nv2, &cur, 8);
+                old = atomicCAS((unsigned long long *)&w[1], assumed, nv2);
+            } while (old != assumed);
+            break;
+        }
+        }
+    }
+}
+
+/* merge partial state words into a session's state (single thread) */
+__device__ inline void state_merge(uint64_t *dst, const uint64_t *src,
+                                   const AggSpec &a) {
+    for (int i = 0; i < a.n_aggs; i++) {
+        switch (a.op[i]) {
+        case AMD_AGG_COUNT:
+        case AMD_AGG_SUM:
+            dst[2 * i] += src[2 * i];
+            break;
+        case AMD_AGG_MIN:
+        case AMD_AGG_MAX:
+            if (src[2 * i] > dst[2 * i]) dst[2 * i] = src[2 * i];
+            break;
+        case AMD_AGG_AVG: {
+            dst[2 * i] += src[2 * i];
+            double x, y;
+            memcpy(&x, &dst[2 * i + 1], 8);
+            memcpy(&y, &src[2 * i + 1], 8);
+            x += y;
+            memcpy(&dst[2 * i + 1], &x, 8);
+            break;
+        }
+        }
+    }
+}
+
+/* claim-or-find the open-addressing slot for `key`; spec slot = index C for
+ * key == EMPTY_KEY (the sentinel value is a legal key). */
+__device__ inline int64_t key_slot(int64_t *keys, uint32_t C, int64_t key,
+                                   int *err, int which_err) {
+    if (key == EMPTY_KEY) return (int64_t)C;
+    uint64_t m = C - 1;
+    uint64_t j = hash64((uint64_t)key) & m;
+    for (uint32_t probes = 0; probes < C; probes++) {
+        int64_t cur = keys[j];
+        if (cur == key) return (int64_t)j;
+        if (cur == EMPTY_KEY) {
+            int64_t old = (int64_t)atomicCAS(
+                (unsigned long long *)&keys[j],
+                (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+            if (old == EMPTY_KEY || old == key) return (int64_t)j;
+        }
+        j = (j + 1) & m;
+    }
+    *err = which_err;
+    return -1;
+}
+
+struct UpdateArgs {
+    const int64_t *cols[12];  /* [key?], vals..., ts */
+    int32_t n_keys, n_vals;
+    int64_t n_rows;
+    uint64_t ts_offset;
+    int has_wm; uint64_t wm;
+    /* per-batch pre-agg table, B slots + 1 spec */
+    int64_t *bkeys;
+    uint64_t *bst;            /* [B+1][2 + n_aggs*2]: min_ts,max_ts,aggs */
+    uint32_t B;
+    AggSpec agg;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_update(UpdateArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t *ts = A.cols[A.n_keys + A.n_vals];
+    int sw = 2 + 2 * A.agg.n_aggs;
+    const int64_t *vcols[AMD_MAX_AGGS];
+    for (int i = 0; i < A.agg.n_aggs; i++)
+        vcols[i] = A.agg.col[i] >= 0 ? A.cols[A.n_keys + A.agg.col[i]]
+                                     : nullptr;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < A.n_rows; r += stride) {
+        uint64_t t = (uint64_t)ts[r] + A.ts_offset;
+        if (A.has_wm && t < A.wm) continue;  /* late: silently dropped */
+        int64_t key = A.n_keys ? A.cols[0][r] : 0;
+        int64_t s = key_slot(A.bkeys, A.B, key, A.err, SERR_BATCH_FULL);
+        if (s < 0) continue;
+        uint64_t *st = A.bst + (size_t)s * sw;
+        atomicMax((unsigned long long *)&st[0],
+                  (unsigned long long)enc_min((int64_t)t));
+        atomicMax((unsigned long long *)&st[1],
+                  (unsigned long long)enc_max((int64_t)t));
+        atomic_fold(st + 2, A.agg, vcols, r);
+    }
+}
+
+struct Store {
+    int64_t *keys;     /* [C+1]; slot C = spec (key == EMPTY_KEY) */
+    uint32_t *ns;      /* [C+1] live sessions per key */
+    int64_t *s_start;  /* [(C+1)*MS] */
+    int64_t *s_end;    /* [(C+1)*MS] data_end */
+    uint64_t *s_st;    /* [(C+1)*MS][n_aggs*2] */
+    uint32_t C, MS;
+};
+
+struct MergeArgs {
+    /* batch pre-agg table */
+    const int64_t *bkeys;
+    const uint64_t *bst;
+    uint32_t B;
+    Store store;
+    uint64_t gap;
+    AggSpec agg;
+    int *err;
+};
+
+/* one thread per occupied pre-agg slot: merge the partial session into the
+ * store.  No two threads share a key, so session-list surgery is plain
+ * single-threaded code. */
+__device__ inline void merge_partial(const MergeArgs &M, int64_t key,
+                                     int64_t pmin, int64_t pmax,
+                                     const uint64_t *pst, int *err) {
+    const Store &S = M.store;
+    int64_t slot = key_slot(S.keys, S.C, key, err, SERR_TABLE_FULL);
+    if (slot < 0) return;
+    int sw = 2 * M.agg.n_aggs;
+    int64_t *ss = S.s_start + (size_t)slot * S.MS;
+    int64_t *se = S.s_end + (size_t)slot * S.MS;
+    uint64_t *sst = S.s_st + (size_t)slot * S.MS * sw;
+    uint32_t n = S.ns[slot];
+    /* absorb every stored session whose gap-closure touches the partial's */
+    int64_t cs = pmin, ce = pmax;
+    uint64_t acc[AMD_MAX_AGGS * 2];
+    for (int i = 0; i < sw; i++) acc[i] = pst[i];
+    uint32_t w = 0;
+    for (uint32_t i = 0; i < n; i++) {
+        if (cs < se[i] + (int64_t)M.gap && ss[i] < ce + (int64_t)M.gap) {
+            if (ss[i] < cs) cs = ss[i];
+            if (se[i] > ce) ce = se[i];
+            state_merge(acc, sst + (size_t)i * sw, M.agg);
+        } else {
+            if (w != i) {
+                ss[w] = ss[i];
+                se[w] = se[i];
+                for (int k = 0; k < sw; k++)
+                    sst[(size_t)w * sw + k] = sst[(size_t)i * sw + k];
+            }
+            w++;
+        }
+    }
+    if (w >= S.MS) { *err = SERR_SESSIONS; return; }
+    ss[w] = cs;
+    se[w] = ce;
+    for (int k = 0; k < sw; k++) sst[(size_t)w * sw + k] = acc[k];
+    S.ns[slot] = w + 1;
+}
+
+__global__ void __launch_bounds__(256)
+k_sess_merge(MergeArgs M) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int sw = 2 + 2 * M.agg.n_aggs;
+    for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         s <= (int64_t)M.B; s += stride) {
+        const uint64_t *st = M.bst + (size_t)s * sw;
+        if (st[0] == 0 && st[1] == 0) continue;      /* untouched slot */
+        int64_t key = s == (int64_t)M.B ? EMPTY_KEY : M.bkeys[s];
+        if (s != (int64_t)M.B && key == EMPTY_KEY) continue;
+        merge_partial(M, key, dec_min(st[0]), dec_max(st[1]), st + 2, M.err);
+    }
+}
+
+struct FireArgs {
+    Store store;
+    uint64_t gap, wm;
+    AggSpec agg;
+    int64_t *out[AMD_MAX_AGGS + 4];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int32_t n_keys;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_fire(FireArgs F) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const Store &S = F.store;
+    int sw = 2 * F.agg.n_aggs;
+    for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         slot <= (int64_t)S.C; slot += stride) {
+        uint32_t n = S.ns[slot];
+        if (n == 0) continue;
+        if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
+        int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        int64_t *ss = S.s_start + (size_t)slot * S.MS;
+        int64_t *se = S.s_end + (size_t)slot * S.MS;
+        uint64_t *sst = S.s_st + (size_t)slot * S.MS * sw;
+        uint32_t w = 0;
+        for (uint32_t i = 0; i < n; i++) {
+            uint64_t close = (uint64_t)(se[i] + (int64_t)F.gap);
+            if (close < F.wm) {
+                int64_t r = (int64_t)atomicAdd(F.n_out, 1ULL);
+                if (r >= F.out_cap) { *F.err = SERR_OUT_CAP; continue; }
+                int col = 0;
+                if (F.n_keys) F.out[col++][r] = key;
+                for (int a = 0; a < F.agg.n_aggs; a++, col++) {
+                    uint64_t w0 = sst[(size_t)i * sw + 2 * a];
+                    uint64_t w1 = sst[(size_t)i * sw + 2 * a + 1];
+                    int64_t v;
+                    switch (F.agg.op[a]) {
+                    case AMD_AGG_COUNT:
+                    case AMD_AGG_SUM: v = (int64_t)w0; break;
+                    case AMD_AGG_MIN: v = dec_min(w0); break;
+                    case AMD_AGG_MAX: v = dec_max(w0); break;
+                    default: {  /* AVG */
+                        double sum;
+                        memcpy(&sum, &w1, 8);
+                        double avg = sum / (double)w0;
+                        memcpy(&v, &avg, 8);
+                        break;
+                    }
+                    }
+                    F.out[col][r] = v;
+                }
+                F.out[col++][r] = ss[i];
+                F.out[col++][r] = (int64_t)close;
+                F.out[col][r] = (int64_t)close - 1;
+            } else {
+                if (w != i) {
+                    ss[w] = ss[i];
+                    se[w] = se[i];
+                    for (int k = 0; k < sw; k++)
+                        sst[(size_t)w * sw + k] = sst[(size_t)i * sw + k];
+                }
+                w++;
+            }
+        }
+        S.ns[slot] = w;
+    }
+}
+
+/* checkpoint drain: one thread per key slot appends its live sessions */
+struct DrainArgs {
+    Store store;
+    int64_t *out[AMD_MAX_AGGS * 2 + 3];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int32_t n_keys, n_aggs;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_drain(DrainArgs D) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const Store &S = D.store;
+    int sw = 2 * D.n_aggs;
+    for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         slot <= (int64_t)S.C; slot += stride) {
+        uint32_t n = S.ns[slot];
+        if (n == 0) continue;
+        if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
+        int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        int64_t base = (int64_t)atomicAdd(D.n_out, (unsigned long long)n);
+        if (base + n > D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
+        for (uint32_t i = 0; i < n; i++) {
+            int64_t r = base + i;
+            int col = 0;
+            if (D.n_keys) D.out[col++][r] = key;
+            for (int k = 0; k < sw; k++)
+                D.out[col++][r] =
+                    (int64_t)S.s_st[(size_t)slot * S.MS * sw +
+                                    (size_t)i * sw + k];
+            D.out[col++][r] = S.s_start[(size_t)slot * S.MS + i];
+            D.out[col][r] = S.s_end[(size_t)slot * S.MS + i];
+        }
+    }
+}
+
+/* restore: rows pre-grouped by key on the host; one thread per key group
+ * replays its partial sessions through the same merge path */
+struct RestoreArgs {
+    const int64_t *keys;      /* [n] (ignored when n_keys == 0) */
+    const uint64_t *st;       /* [n][n_aggs*2] */
+    const int64_t *start;     /* [n] */
+    const int64_t *end;       /* [n] */
+    const int64_t *group_off; /* [n_groups+1] */
+    int64_t n_groups;
+    Store store;
+    uint64_t gap;
+    AggSpec agg;
+    int32_t n_keys;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_restore(RestoreArgs R) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int sw = 2 * R.agg.n_aggs;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.n_groups; g += stride) {
+        for (int64_t i = R.group_off[g]; i < R.group_off[g + 1]; i++) {
+            int64_t key = R.n_keys ? R.keys[i] : 0;
+            merge_partial({nullptr, nullptr, 0, R.store, R.gap, R.agg,
+                           R.err},
+                          key, R.start[i], R.end[i],
+                          R.st + (size_t)i * sw, R.err);
+        }
+    }
+}
+
+}  // namespace sess
+
+using namespace sess;
+
+static char g_sess_err[256];
+
+struct GpuSession {
+    AmdSessionConfig cfg;
+    AggSpec agg;
+    Store store;
+    int64_t *bkeys;
+    uint64_t *bst;
+    uint32_t B;
+    int64_t *d_out[AMD_MAX_AGGS * 2 + 4];
+    unsigned long long *d_n_out;
+    int *d_err;
+    int64_t *stg_h[12], *stg_d[12];
+    int64_t stg_cap;
+    int n_in_cols, out_cols, drain_cols;
+    int64_t out_cap;
+    int has_wm; uint64_t wm;
+    hipStream_t stream;
+    char err_msg[512];
+};
+
+#define SHIP(o, call)                                                         \
+    do {                                                                      \
+        hipError_t _e = (call);                                               \
+        if (_e != hipSuccess) {                                               \
+            snprintf((o)->err_msg, sizeof (o)->err_msg, "%s:%d hip: %s",      \
+                     __FILE__, __LINE__, hipGetErrorString(_e));              \
+            return 1;                                                         \
+        }                                                                     \
+    } while (0)
+
+API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
+    if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->n_value_cols < 0 ||
+        cfg->n_value_cols > 8 || cfg->n_aggs < 1 ||
+        cfg->n_aggs > AMD_MAX_AGGS || cfg->gap_nanos == 0) {
+        snprintf(g_sess_err, sizeof g_sess_err, "invalid session config");
+        return nullptr;
+    }
+    GpuSession *o = new GpuSession();
+    o->cfg = *cfg;
+    o->agg.n_aggs = cfg->n_aggs;
+    for (int i = 0; i < cfg->n_aggs; i++) {
+        o->agg.op[i] = cfg->agg_ops[i];
+        o->agg.col[i] = cfg->agg_col[i];
+    }
+    o->store.C = 1u << (cfg->log2_capacity ? cfg->log2_capacity : 16);
+    o->store.MS = cfg->max_sessions ? cfg->max_sessions : 8;
+    o->B = 1u << (cfg->log2_batch_capacity ? cfg->log2_batch_capacity : 14);
+    o->out_cap = 1ll << (cfg->log2_out_cap ? cfg->log2_out_cap : 20);
+    o->n_in_cols = cfg->n_keys + cfg->n_value_cols + 1;
+    o->out_cols = cfg->n_keys + cfg->n_aggs + 3;
+    o->drain_cols = cfg->n_keys + 2 * cfg->n_aggs + 2;
+    if (hipSetDevice(cfg->device) != hipSuccess) {
+        snprintf(g_sess_err, sizeof g_sess_err,
+                 "hipSetDevice(%d) failed: no HIP device (no CPU fallback)",
+                 cfg->device);
+        delete o;
+        return nullptr;
+    }
+    hipError_t e;
+    auto fail = [&](const char *what, hipError_t e2) {
+        snprintf(g_sess_err, sizeof g_sess_err, "%s: %s", what,
+                 hipGetErrorString(e2));
+        delete o;
+        return nullptr;
+    };
+#define SALLOC(p, bytes)                                                      \
+    if ((e = hipMalloc((void **)&(p), (bytes))) != hipSuccess)                \
+        return fail(#p, e);
+    size_t C1 = (size_t)o->store.C + 1;
+    size_t MS = o->store.MS, sw = 2 * (size_t)cfg->n_aggs;
+    SALLOC(o->store.keys, C1 * 8);
+    SALLOC(o->store.ns, C1 * 4);
+    SALLOC(o->store.s_start, C1 * MS * 8);
+    SALLOC(o->store.s_end, C1 * MS * 8);
+    SALLOC(o->store.s_st, C1 * MS * sw * 8);
+    SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
+    SALLOC(o->bst, ((size_t)o->B + 1) * (2 + sw) * 8);
+    int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
+    for (int i = 0; i < max_out; i++)
+        SALLOC(o->d_out[i], (size_t)o->out_cap * 8);
+    SALLOC(o->d_n_out, 8);
+    SALLOC(o->d_err, 4);
+#undef SALLOC
+    hipMemset(o->store.keys, 0xFF, C1 * 8);
+    hipMemset(o->store.ns, 0, C1 * 4);
+    hipMemset(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8);
+    hipMemset(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8);
+    hipMemset(o->d_err, 0, 4);
+    hipStreamCreate(&o->stream);
+    o->stg_cap = 1 << 20;
+    for (int c = 0; c < o->n_in_cols; c++) {
+        if (hipHostMalloc((void **)&o->stg_h[c], (size_t)o->stg_cap * 8) !=
+                hipSuccess ||
+            hipMalloc((void **)&o->stg_d[c], (size_t)o->stg_cap * 8) !=
+                hipSuccess) {
+            snprintf(g_sess_err, sizeof g_sess_err,
+                     "session staging alloc failed");
+            delete o;
+            return nullptr;
+        }
+    }
+    return o;
+}
+
+API const char *arroyo_amd_session_last_error(void *h) {
+    return h ? ((GpuSession *)h)->err_msg : g_sess_err;
+}
+
+static int sess_check_err(GpuSession *o) {
+    int e = 0;
+    SHIP(o, hipMemcpyAsync(&e, o->d_err, 4, hipMemcpyDeviceToHost,
+                           o->stream));
+    SHIP(o, hipStreamSynchronize(o->stream));
+    if (!e) return 0;
+    const char *msg =
+        e == SERR_TABLE_FULL ? "session key table full; raise log2_capacity"
+        : e == SERR_BATCH_FULL
+            ? "per-batch table full; raise log2_batch_capacity"
+        : e == SERR_SESSIONS
+            ? "per-key live-session limit hit; raise max_sessions"
+        : e == SERR_OUT_CAP ? "output buffer full; raise log2_out_cap"
+                            : "device error";
+    snprintf(o->err_msg, sizeof o->err_msg, "%s", msg);
+    return 1;
+}
+
+static int grid_for(int64_t want_threads) {
+    int64_t want = (want_threads + 255) / 256;
+    return (int)(want > 4096 ? 4096 : (want < 1 ? 1 : want));
+}
+
+/* one sub-batch whose event-time span is < gap */
+static int sess_submit(GpuSession *o, const int64_t *const *dcols,
+                       int64_t n_rows, uint64_t ts_offset) {
+    if (n_rows == 0) return 0;
+    size_t sw = 2 * (size_t)o->cfg.n_aggs;
+    SHIP(o, hipMemsetAsync(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8,
+                           o->stream));
+    SHIP(o, hipMemsetAsync(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8,
+                           o->stream));
+    UpdateArgs A = {};
+    for (int c = 0; c < o->n_in_cols; c++) A.cols[c] = dcols[c];
+    A.n_keys = o->cfg.n_keys;
+    A.n_vals = o->cfg.n_value_cols;
+    A.n_rows = n_rows;
+    A.ts_offset = ts_offset;
+    A.has_wm = o->has_wm;
+    A.wm = o->wm;
+    A.bkeys = o->bkeys;
+    A.bst = o->bst;
+    A.B = o->B;
+    A.agg = o->agg;
+    A.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_update, dim3(grid_for(n_rows)), dim3(256), 0,
+                       o->stream, A);
+    SHIP(o, hipGetLastError());
+    MergeArgs M = {};
+    M.bkeys = o->bkeys;
+    M.bst = o->bst;
+    M.B = o->B;
+    M.store = o->store;
+    M.gap = o->cfg.gap_nanos;
+    M.agg = o->agg;
+    M.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_merge, dim3(grid_for((int64_t)o->B + 1)),
+                       dim3(256), 0, o->stream, M);
+    SHIP(o, hipGetLastError());
+    return 0;
+}
+
+API int arroyo_amd_session_process_batch_device(void *h,
+                                                const int64_t *const *dcols,
+                                                int32_t n_cols,
+                                                int64_t n_rows,
+                                                uint64_t ts_offset) {
+    GpuSession *o = (GpuSession *)h;
+    if (n_cols != o->n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->n_in_cols, n_cols);
+        return 1;
+    }
+    /* the caller owns the guarantee that the batch's event-time span is
+     * < gap on this path (bench-style streams: ms spans vs second gaps) */
+    return sess_submit(o, dcols, n_rows, ts_offset);
+}
+
+API int arroyo_amd_session_process_batch(void *h, const int64_t *const *cols,
+                                         int32_t n_cols, int64_t n_rows) {
+    GpuSession *o = (GpuSession *)h;
+    if (n_cols != o->n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->n_in_cols, n_cols);
+        return 1;
+    }
+    if (n_rows == 0) return 0;
+    const int64_t *ts = cols[n_cols - 1];
+    /* bucket rows by floor(ts / (gap/2)) so each sub-batch spans < gap;
+     * sub-batch order is irrelevant (the merge is order-independent) */
+    uint64_t bw = o->cfg.gap_nanos / 2;
+    if (bw == 0) bw = 1;
+    int64_t mn = ts[0], mx = ts[0];
+    for (int64_t r = 1; r < n_rows; r++) {
+        if (ts[r] < mn) mn = ts[r];
+        if (ts[r] > mx) mx = ts[r];
+    }
+    std::vector<int64_t> order(n_rows);
+    for (int64_t i = 0; i < n_rows; i++) order[i] = i;
+    bool split = (uint64_t)(mx - mn) >= o->cfg.gap_nanos;
+    if (split)
+        std::stable_sort(order.begin(), order.end(),
+                         [&](int64_t a, int64_t b) {
+                             return (uint64_t)ts[a] / bw <
+                                    (uint64_t)ts[b] / bw;
+                         });
+    int64_t done = 0;
+    while (done < n_rows) {
+        int64_t take = n_rows - done;
+        if (split) {
+            uint64_t b0 = (uint64_t)ts[order[done]] / bw;
+            take = 1;
+            while (done + take < n_rows &&
+                   (uint64_t)ts[order[done + take]] / bw == b0)
+                take++;
+        }
+        int64_t sent = 0;
+        while (sent < take) {
+            int64_t chunk = take - sent;
+            if (chunk > o->stg_cap) chunk = o->stg_cap;
+            const int64_t *dcols[12];
+            for (int c = 0; c < o->n_in_cols; c++) {
+                for (int64_t i = 0; i < chunk; i++)
+                    o->stg_h[c][i] = cols[c][order[done + sent + i]];
+                SHIP(o, hipMemcpyAsync(o->stg_d[c], o->stg_h[c],
+                                       (size_t)chunk * 8,
+                                       hipMemcpyHostToDevice, o->stream));
+                dcols[c] = o->stg_d[c];
+            }
+            if (sess_submit(o, dcols, chunk, 0)) return 1;
+            SHIP(o, hipStreamSynchronize(o->stream));
+            sent += chunk;
+        }
+        done += take;
+    }
+    return 0;
+}
+
+API int arroyo_amd_session_handle_watermark(void *h, uint64_t wm,
+                                            AmdOutBatch *out) {
+    GpuSession *o = (GpuSession *)h;
+    if (sess_check_err(o)) return 1;
+    o->has_wm = 1;
+    o->wm = wm;
+    SHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    FireArgs F = {};
+    F.store = o->store;
+    F.gap = o->cfg.gap_nanos;
+    F.wm = wm;
+    F.agg = o->agg;
+    for (int i = 0; i < o->out_cols; i++) F.out[i] = o->d_out[i];
+    F.n_out = o->d_n_out;
+    F.out_cap = o->out_cap;
+    F.n_keys = o->cfg.n_keys;
+    F.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_fire,
+                       dim3(grid_for((int64_t)o->store.C + 1)), dim3(256), 0,
+                       o->stream, F);
+    SHIP(o, hipGetLastError());
+    unsigned long long n = 0;
+    SHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                           o->stream));
+    SHIP(o, hipStreamSynchronize(o->stream));
+    if (sess_check_err(o)) return 1;
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = (int64_t)n;
+        out->n_cols = o->out_cols;
+        out->cols = (void **)calloc(o->out_cols, sizeof(void *));
+        out->is_f64 = (int32_t *)calloc(o->out_cols, sizeof(int32_t));
+        for (int a = 0; a < o->cfg.n_aggs; a++)
+            if (o->cfg.agg_ops[a] == AMD_AGG_AVG)
+                out->is_f64[o->cfg.n_keys + a] = 1;
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+            if (n)
+                SHIP(o, hipMemcpyAsync(out->cols[i], o->d_out[i],
+                                       (size_t)n * 8, hipMemcpyDeviceToHost,
+                                       o->stream));
+        }
+        SHIP(o, hipStreamSynchronize(o->stream));
+    }
+    return 0;
+}
+
+API int arroyo_amd_session_checkpoint_drain(void *h, AmdOutBatch *out) {
+    GpuSession *o = (GpuSession *)h;
+    if (sess_check_err(o)) return 1;
+    SHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    DrainArgs D = {};
+    D.store = o->store;
+    for (int i = 0; i < o->drain_cols; i++) D.out[i] = o->d_out[i];
+    D.n_out = o->d_n_out;
+    D.out_cap = o->out_cap;
+    D.n_keys = o->cfg.n_keys;
+    D.n_aggs = o->cfg.n_aggs;
+    D.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_drain,
+                       dim3(grid_for((int64_t)o->store.C + 1)), dim3(256), 0,
+                       o->stream, D);
+    SHIP(o, hipGetLastError());
+    unsigned long long n = 0;
+    SHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                           o->stream));
+    SHIP(o, hipStreamSynchronize(o->stream));
+    if (sess_check_err(o)) return 1;
+    memset(out, 0, sizeof *out);
+    out->n_rows = (int64_t)n;
+    out->n_cols = o->drain_cols;
+    out->cols = (void **)calloc(o->drain_cols, sizeof(void *));
+    out->is_f64 = (int32_t *)calloc(o->drain_cols, sizeof(int32_t));
+    for (int i = 0; i < o->drain_cols; i++) {
+        out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+        if (n)
+            SHIP(o, hipMemcpyAsync(out->cols[i], o->d_out[i], (size_t)n * 8,
+                                   hipMemcpyDeviceToHost, o->stream));
+    }
+    SHIP(o, hipStreamSynchronize(o->stream));
+    return 0;
+}
+
+API int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
+                                   int32_t n_cols, int64_t n_rows) {
+    GpuSession *o = (GpuSession *)h;
+    if (n_cols != o->drain_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "restore expects %d cols, got %d", o->drain_cols, n_cols);
+        return 1;
+    }
+    if (n_rows == 0) return 0;
+    int sw = 2 * o->cfg.n_aggs;
+    /* group rows by key host-side: one device thread per key group */
+    std::vector<int64_t> order(n_rows);
+    for (int64_t i = 0; i < n_rows; i++) order[i] = i;
+    const int64_t *keys = o->cfg.n_keys ? cols[0] : nullptr;
+    if (keys)
+        std::stable_sort(order.begin(), order.end(),
+                         [&](int64_t a, int64_t b) {
+                             return keys[a] < keys[b];
+                         });
+    std::vector<int64_t> h_keys(n_rows), h_start(n_rows), h_end(n_rows);
+    std::vector<uint64_t> h_st((size_t)n_rows * sw);
+    std::vector<int64_t> off;
+    off.push_back(0);
+    for (int64_t i = 0; i < n_rows; i++) {
+        int64_t r = order[i];
+        h_keys[i] = keys ? keys[r] : 0;
+        if (i && keys && h_keys[i] != h_keys[i - 1]) off.push_back(i);
+        for (int k = 0; k < sw; k++)
+            h_st[(size_t)i * sw + k] =
+                (uint64_t)cols[o->cfg.n_keys + k][r];
+        h_start[i] = cols[o->cfg.n_keys + sw][r];
+        h_end[i] = cols[o->cfg.n_keys + sw + 1][r];
+    }
+    off.push_back(n_rows);
+    int64_t n_groups = (int64_t)off.size() - 1;
+    int64_t *d_keys, *d_start, *d_end, *d_off;
+    uint64_t *d_st;
+    SHIP(o, hipMalloc((void **)&d_keys, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_start, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_end, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_st, (size_t)n_rows * sw * 8));
+    SHIP(o, hipMalloc((void **)&d_off, (size_t)(n_groups + 1) * 8));
+    SHIP(o, hipMemcpy(d_keys, h_keys.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_start, h_start.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_end, h_end.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_st, h_st.data(), (size_t)n_rows * sw * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_off, off.data(), (size_t)(n_groups + 1) * 8,
+                      hipMemcpyHostToDevice));
+    RestoreArgs R = {};
+    R.keys = d_keys;
+    R.st = d_st;
+    R.start = d_start;
+    R.end = d_end;
+    R.group_off = d_off;
+    R.n_groups = n_groups;
+    R.store = o->store;
+    R.gap = o->cfg.gap_nanos;
+    R.agg = o->agg;
+    R.n_keys = o->cfg.n_keys;
+    R.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_restore, dim3(grid_for(n_groups)), dim3(256),
+                       0, o->stream, R);
+    SHIP(o, hipGetLastError());
+    SHIP(o, hipStreamSynchronize(o->stream));
+    hipFree(d_keys);
+    hipFree(d_start);
+    hipFree(d_end);
+    hipFree(d_st);
+    hipFree(d_off);
+    return sess_check_err(o);
+}
+
+API void arroyo_amd_session_destroy(void *h) {
+    GpuSession *o = (GpuSession *)h;
+    if (!o) return;
+    hipStreamSynchronize(o->stream);
+    hipFree(o->store.keys);
+    hipFree(o->store.ns);
+    hipFree(o->store.s_start);
+    hipFree(o->store.s_end);
+    hipFree(o->store.s_st);
+    hipFree(o->bkeys);
+    hipFree(o->bst);
+    int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
+    for (int i = 0; i < max_out; i++) hipFree(o->d_out[i]);
+    hipFree(o->d_n_out);
+    hipFree(o->d_err);
+    for (int c = 0; c < o->n_in_cols; c++) {
+        hipHostFree(o->stg_h[c]);
+        hipFree(o->stg_d[c]);
+    }
+    hipStreamDestroy(o->stream);
+    delete o;
+}

@@ -14,17 +14,19 @@ from arroyo_amd.cabi import AmdOutBatch, WindowOp, _out_to_numpy
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libarroyo_amd.so")
-_SRC = os.path.join(_DIR, "csrc", "arroyo_amd.hip")
+_SRCS = [os.path.join(_DIR, "csrc", f)
+         for f in ("arroyo_amd.hip", "session.hip", "expjoin.hip")]
 
 _lib = None
 
 HIPCC_CMD = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-             "-shared", "-fvisibility=hidden", "-o", _SO, _SRC]
+             "-shared", "-fvisibility=hidden", "-parallel-jobs=3",
+             "-o", _SO] + _SRCS
 
 
 def build(force=False):
     if force or not os.path.exists(_SO) or \
-            os.path.getmtime(_SO) < os.path.getmtime(_SRC):
+            os.path.getmtime(_SO) < max(os.path.getmtime(s) for s in _SRCS):
         subprocess.run(HIPCC_CMD, check=True, cwd=_DIR,
                        capture_output=True, text=True)
     return _SO

@@ -143,3 +143,99 @@ def test_expjoin_oracle_restore_ttl_filter():
                                      dtype=np.int64)])
     assert rows_of(out) == [(2, t0 + 26 * NS)]
     op.close()
+
+
+# ---------------------------------------------------------------- GPU parity
+
+
+@pytest.mark.gpu
+def test_updating_inner_join_golden_gpu():
+    from arroyo_amd import gpu
+    run_updating_inner_join_golden(gpu.make_expjoin_op)
+
+
+@pytest.mark.gpu
+def test_expjoin_gpu_vs_oracle_fuzz():
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(41)
+    t0 = 1_600_000_000 * NS
+    cfgk = dict(n_left_vals=1, n_right_vals=2)
+    g = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    o = oracle.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    got, want = [], []
+    for step in range(6):
+        n = 400
+        k = rng.integers(0, 64, size=n).astype(np.int64)
+        v1 = rng.integers(0, 10**6, size=n).astype(np.int64)
+        v2 = rng.integers(0, 10**6, size=n).astype(np.int64)
+        ts = t0 + (step * 60 + np.sort(
+            rng.integers(0, 60, size=n))).astype(np.int64) * NS
+        side = step % 2
+        cols = [k, v1, ts] if side == 0 else [k, v1, v2, ts]
+        got += rows_of(g.process_batch(side, cols))
+        want += rows_of(o.process_batch(side, cols))
+    g.close()
+    o.close()
+    assert sorted(got) == sorted(want)
+    assert len(want) > 1000
+
+
+@pytest.mark.gpu
+def test_expjoin_gpu_expire_matches_oracle():
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(43)
+    t0 = 1_600_000_000 * NS
+    ttl = 30 * NS
+    g = gpu.make_expjoin_op(cabi.make_expjoin_config(ttl))
+    o = oracle.make_expjoin_op(cabi.make_expjoin_config(ttl))
+    got, want = [], []
+    for step in range(5):
+        n = 200
+        k = rng.integers(0, 32, size=n).astype(np.int64)
+        ts = t0 + (step * 20 + np.sort(
+            rng.integers(0, 20, size=n))).astype(np.int64) * NS
+        side = step % 2
+        for op, acc in ((g, got), (o, want)):
+            acc += rows_of(op.process_batch(side, [k, ts]))
+            op.handle_watermark(int(ts[-1]))
+            op.expire()
+    g.close()
+    o.close()
+    assert sorted(got) == sorted(want)
+
+
+@pytest.mark.gpu
+def test_expjoin_gpu_checkpoint_restore():
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(47)
+    t0 = 1_600_000_000 * NS
+    lk, lv, lt = gen_side(rng, 500, t0)
+    rk, rv, rt = gen_side(rng, 500, t0)
+    cfgk = dict(n_left_vals=1, n_right_vals=1)
+
+    a = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    a.process_batch(a.LEFT, [lk, lv, lt])
+    ld = a.checkpoint_drain(a.LEFT)
+    rd = a.checkpoint_drain(a.RIGHT)
+    a.close()
+    assert len(ld[0]) == 500 and len(rd[0]) == 0
+
+    b = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    b.restore(b.LEFT, ld)
+    got = rows_of(b.process_batch(b.RIGHT, [rk, rv, rt]))
+    b.close()
+    assert got == np_join(lk, lv, lt, rk, rv, rt)
+
+
+@pytest.mark.gpu
+def test_expjoin_gpu_key_minus_one():
+    from arroyo_amd import gpu
+    t0 = 1_600_000_000 * NS
+    op = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR))
+    op.process_batch(op.LEFT, [np.array([-1, 2], dtype=np.int64),
+                               np.array([t0, t0], dtype=np.int64)])
+    out = op.process_batch(op.RIGHT,
+                           [np.array([-1, -1, 3], dtype=np.int64),
+                            np.array([t0 + NS] * 3, dtype=np.int64)])
+    op.close()
+    assert rows_of(out) == [(-1, t0 + NS), (-1, t0 + NS)]

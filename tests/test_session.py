@@ -173,3 +173,107 @@ def test_session_oracle_checkpoint_roundtrip():
     want = rows_of(c.handle_watermark(U64MAX))
     c.close()
     assert got == want
+
+
+# ---------------------------------------------------------------- GPU parity
+
+
+@pytest.mark.gpu
+def test_session_window_golden_gpu():
+    from arroyo_amd import gpu
+    run_session_window_golden(gpu.make_session_op)
+
+
+@pytest.mark.gpu
+def test_global_session_window_golden_gpu():
+    from arroyo_amd import gpu
+    run_global_session_window_golden(gpu.make_session_op)
+
+
+@pytest.mark.gpu
+def test_session_gpu_vs_oracle_fuzz():
+    """Streamed fuzz: HIP session path vs the CPU oracle, bit-exact."""
+    from arroyo_amd import gpu
+    _k, _t, _v, got = stream_fuzz(gpu.make_session_op, n=20000, seed=9)
+    _k2, _t2, _v2, want = stream_fuzz(oracle.make_session_op, n=20000,
+                                      seed=9)
+    assert sorted(got) == sorted(want)
+    assert len(want) > 50
+
+
+@pytest.mark.gpu
+def test_session_gpu_multi_agg_vs_oracle():
+    from arroyo_amd import gpu
+    aggs = [(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.MIN, 0), (cabi.MAX, 0)]
+    _k, _t, _v, got = stream_fuzz(gpu.make_session_op, n=8000, seed=4,
+                                  aggs=aggs, n_value_cols=1)
+    _k2, _t2, _v2, want = stream_fuzz(oracle.make_session_op, n=8000, seed=4,
+                                      aggs=aggs, n_value_cols=1)
+    assert sorted(got) == sorted(want)
+
+
+@pytest.mark.gpu
+def test_session_gpu_wide_batch_split():
+    """A single batch spanning many gaps must still sessionize correctly
+    (the host splits it into gap/2 buckets)."""
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(31)
+    t0 = 1_600_000_000 * NS
+    gap = 2 * NS
+    n = 3000
+    ts = t0 + np.sort(rng.integers(0, 600 * NS, size=n)).astype(np.int64)
+    key = rng.integers(0, 5, size=n).astype(np.int64)
+    op = gpu.make_session_op(cabi.make_session_config(
+        gap, [(cabi.COUNT, -1)], n_keys=1))
+    op.process_batch([key, ts])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    want = [(k, c, s, e, t) for k, c, s, e, t in np_sessions(key, ts, gap)]
+    assert got == sorted(want)
+
+
+@pytest.mark.gpu
+def test_session_gpu_checkpoint_roundtrip():
+    from arroyo_amd import gpu
+    counter, ts = impulse_cols()
+    user = np.where(counter % 10 == 0, 0, counter).astype(np.int64)
+    mid = len(user) // 2
+
+    a = gpu.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    a.process_batch([user[:mid], ts[:mid]])
+    drained = a.checkpoint_drain()
+    a.close()
+    assert len(drained[0]) > 0
+
+    b = gpu.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    b.restore(drained)
+    b.process_batch([user[mid:], ts[mid:]])
+    got = rows_of(b.handle_watermark(U64MAX))
+    b.close()
+
+    c = oracle.make_session_op(cabi.make_session_config(
+        GAP20, [(cabi.COUNT, -1)], n_keys=1))
+    c.process_batch([user, ts])
+    want = rows_of(c.handle_watermark(U64MAX))
+    c.close()
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_session_gpu_key_minus_one():
+    """key == -1 collides with the empty-slot sentinel: special slot."""
+    from arroyo_amd import gpu
+    t0 = 1_600_000_000 * NS
+    key = np.array([-1, -1, 3, -1], dtype=np.int64)
+    ts = np.array([t0, t0 + NS, t0 + NS, t0 + 10 * NS], dtype=np.int64)
+    op = gpu.make_session_op(cabi.make_session_config(
+        2 * NS, [(cabi.COUNT, -1)], n_keys=1))
+    op.process_batch([key, ts])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    want = [(-1, 2, t0, t0 + 3 * NS, t0 + 3 * NS - 1),
+            (-1, 1, t0 + 10 * NS, t0 + 12 * NS, t0 + 12 * NS - 1),
+            (3, 1, t0 + NS, t0 + 3 * NS, t0 + 3 * NS - 1)]
+    assert got == sorted(want)
