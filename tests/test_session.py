@@ -81,7 +81,8 @@ def test_global_session_window_golden_oracle():
     run_global_session_window_golden(oracle.make_session_op)
 
 
-def stream_fuzz(make_op, n=4000, seed=3, gap_s=5, aggs=None, n_value_cols=0):
+def stream_fuzz(make_op, n=4000, seed=3, gap_s=5, aggs=None, n_value_cols=0,
+                **cfg_kw):
     """Stream a key/ts stream through the op with periodic watermarks;
     returns all emitted rows.  Input is watermark-clean (each batch only
     has ts >= previous watermark) so results equal the batch restatement."""
@@ -96,8 +97,12 @@ def stream_fuzz(make_op, n=4000, seed=3, gap_s=5, aggs=None, n_value_cols=0):
     vals = [rng.integers(-50, 50, size=n).astype(np.int64)
             for _ in range(n_value_cols)]
     aggs = aggs or [(cabi.COUNT, -1)]
+    # sparse watermark cadence (one per ~n/7 rows) leaves many unfired
+    # sessions per key live at once: size the inline session store for it
+    cfg_kw.setdefault("max_sessions", 256)
+    cfg_kw.setdefault("log2_capacity", 10)
     op = make_op(cabi.make_session_config(
-        gap, aggs, n_keys=1, n_value_cols=n_value_cols))
+        gap, aggs, n_keys=1, n_value_cols=n_value_cols, **cfg_kw))
     got = []
     step = n // 7
     for b in range(0, n, step):
@@ -224,7 +229,8 @@ def test_session_gpu_wide_batch_split():
     ts = t0 + np.sort(rng.integers(0, 600 * NS, size=n)).astype(np.int64)
     key = rng.integers(0, 5, size=n).astype(np.int64)
     op = gpu.make_session_op(cabi.make_session_config(
-        gap, [(cabi.COUNT, -1)], n_keys=1))
+        gap, [(cabi.COUNT, -1)], n_keys=1, max_sessions=512,
+        log2_capacity=8))
     op.process_batch([key, ts])
     got = rows_of(op.handle_watermark(U64MAX))
     op.close()
