@@ -69,11 +69,6 @@ struct DeviceRing {
     uint64_t *state;     /* [R][C][n_aggs][2] encoded */
     uint64_t *tag;       /* [R] bin nanos or EMPTY_TAG */
     uint64_t *fill;      /* [R] occupied-slot count */
-    /* claimed-slot lists: every CAS-won key slot is appended here, so the
-     * fire/retire path touches only occupied slots (O(distinct keys)) and
-     * never scans or memsets whole panes (O(capacity)) */
-    uint32_t *claimed;   /* [R][C] slot indices */
-    unsigned long long *ccnt; /* [R] */
     /* special entry per pane for an actual key == EMPTY_KEY */
     uint32_t *spec_used; /* [R] */
     uint64_t *spec_state;/* [R][n_aggs][2] */
@@ -198,34 +193,6 @@ __device__ inline int64_t table_upsert(int64_t *keys, uint32_t C, int64_t key,
     return -1;
 }
 
-/* upsert that records the claimed slot (CAS winner only -> each occupied
- * slot appears exactly once in the claimed list) */
-__device__ inline int64_t table_upsert_rec(int64_t *keys, uint32_t C,
-                                           int64_t key, int *err,
-                                           uint32_t *claimed,
-                                           unsigned long long *ccnt) {
-    uint64_t m = C - 1;
-    uint64_t i = hash64((uint64_t)key) & m;
-    uint32_t lim = C < MAX_PROBES ? C : MAX_PROBES;
-    for (uint32_t probes = 0; probes < lim; probes++) {
-        int64_t k = keys[i];
-        if (k == key) return (int64_t)i;
-        if (k == EMPTY_KEY) {
-            int64_t old = (int64_t)atomicCAS((unsigned long long *)&keys[i],
-                                             (unsigned long long)EMPTY_KEY,
-                                             (unsigned long long)key);
-            if (old == EMPTY_KEY) {
-                claimed[atomicAdd(ccnt, 1ULL)] = (uint32_t)i;
-                return (int64_t)i;
-            }
-            if (old == key) return (int64_t)i;
-        }
-        i = (i + 1) & m;
-    }
-    *err = ERR_TABLE_FULL;
-    return -1;
-}
-
 /* claim a pane's ring slot for `bin`; wave-cooperative callers dedup first.
  * The volatile re-read (L2, bypassing a possibly-stale L1 line) keeps the
  * CAS storm on a freshly-opened bin to the waves in flight at that moment. */
@@ -314,9 +281,7 @@ k_update(UpdateArgs A) {
             st = A.ring.spec_state + (size_t)p * A.agg.n_aggs * 2;
         } else {
             int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-            int64_t s = table_upsert_rec(keys, A.ring.C, key, A.ring.err,
-                                         A.ring.claimed + (size_t)p * A.ring.C,
-                                         &A.ring.ccnt[p]);
+            int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
             if (s < 0) continue;
             st = A.ring.state +
                  ((size_t)p * A.ring.C + (size_t)s) * A.agg.n_aggs * 2;
@@ -434,9 +399,7 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
             st = A.ring.spec_state + (size_t)p * na * 2;
         } else {
             int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-            int64_t s = table_upsert_rec(keys, A.ring.C, key, A.ring.err,
-                                         A.ring.claimed + (size_t)p * A.ring.C,
-                                         &A.ring.ccnt[p]);
+            int64_t s = table_upsert(keys, A.ring.C, key, A.ring.err);
             if (s < 0) return;
             st = A.ring.state + ((size_t)p * A.ring.C + (size_t)s) * na * 2;
         }
@@ -452,9 +415,7 @@ __device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
         if (key == EMPTY_KEY) continue;
         uint32_t p = ls_pane[s];
         int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
-        int64_t slot = table_upsert_rec(keys, A.ring.C, key, A.ring.err,
-                                        A.ring.claimed + (size_t)p * A.ring.C,
-                                        &A.ring.ccnt[p]);
+        int64_t slot = table_upsert(keys, A.ring.C, key, A.ring.err);
         if (slot < 0) continue;
         atomic_merge(A.ring.state +
                          ((size_t)p * A.ring.C + (size_t)slot) * na * 2,
@@ -536,32 +497,24 @@ struct MergeArgs {
     int64_t  *m_keys;     /* [CM] */
     uint64_t *m_state;    /* [CM][n_aggs][2] */
     uint64_t *m_fill;
-    uint32_t *m_claimed;  /* [CM] merge-table claimed list */
-    unsigned long long *m_ccnt;
     uint32_t *m_spec_used;
     uint64_t *m_spec_state;
     uint32_t  CM;
     int32_t   n_src;
     uint32_t  src[64];    /* ring slots to merge */
-    uint64_t  off[65];    /* exclusive prefix sums of claimed counts */
 };
 
-/* iterate the source panes' CLAIMED slots only (O(distinct keys), not
- * O(pane capacity)): each entry is a unique occupied slot */
 __global__ void __launch_bounds__(256)
 k_merge(MergeArgs M) {
-    size_t total = (size_t)M.off[M.n_src];
+    size_t total = (size_t)M.n_src * M.ring.C;
     size_t stride = (size_t)gridDim.x * blockDim.x;
     for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < total; i += stride) {
-        int pi = 0;
-        while (i >= M.off[pi + 1]) pi++;     /* n_src <= width/slide, tiny */
-        uint32_t p = M.src[pi];
-        size_t slot = M.ring.claimed[(size_t)p * M.ring.C +
-                                     (i - M.off[pi])];
+        uint32_t p = M.src[i / M.ring.C];
+        size_t slot = i % M.ring.C;
         int64_t key = M.ring.keys[(size_t)p * M.ring.C + slot];
-        int64_t d = table_upsert_rec(M.m_keys, M.CM, key, M.ring.err,
-                                     M.m_claimed, M.m_ccnt);
+        if (key == EMPTY_KEY) continue;
+        int64_t d = table_upsert(M.m_keys, M.CM, key, M.ring.err);
         if (d < 0) continue;
         atomic_merge(M.m_state + (size_t)d * M.agg.n_aggs * 2,
                      M.ring.state +
@@ -580,46 +533,6 @@ k_merge(MergeArgs M) {
     }
 }
 
-/* scatter-clear a table's claimed slots back to empty (replaces whole-pane
- * memsets: O(occupied), one launch clears slots + meta + cursor) */
-struct ScatterClearArgs {
-    int64_t  *keys;
-    uint64_t *state;
-    uint32_t *claimed;
-    unsigned long long *ccnt;
-    int32_t   na;
-    /* optional meta cleared by block 0 thread 0 (null to skip) */
-    uint64_t *tag;          /* -> EMPTY_TAG */
-    uint64_t *fill;         /* -> 0 */
-    uint32_t *spec_used;    /* -> 0 */
-    uint64_t *spec_state;   /* -> 0 (na*2 words) */
-    unsigned long long *n_out; /* -> 0 */
-};
-
-__global__ void __launch_bounds__(256)
-k_scatter_clear(ScatterClearArgs A) {
-    size_t total = (size_t)*A.ccnt;
-    size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < total; i += stride) {
-        uint32_t slot = A.claimed[i];
-        A.keys[slot] = EMPTY_KEY;
-        for (int w = 0; w < A.na * 2; w++)
-            A.state[(size_t)slot * A.na * 2 + w] = 0;
-    }
-    if (blockIdx.x == 0 && threadIdx.x == 0) {
-        if (A.tag) *A.tag = EMPTY_TAG;
-        if (A.fill) *A.fill = 0;
-        if (A.spec_used) *A.spec_used = 0;
-        if (A.spec_state)
-            for (int w = 0; w < A.na * 2; w++) A.spec_state[w] = 0;
-        if (A.n_out) *A.n_out = 0;
-    }
-    /* the claimed cursor itself is reset by the host with a tiny
-     * hipMemsetAsync AFTER this kernel: zeroing it here would race with
-     * blocks that have not read `total` yet */
-}
-
 /* ------------------------------------------------------------------ */
 /* K5: compact occupied merge-table slots into output columns.
  * Output column order: [key?], aggs..., window_start, window_end, _ts.  */
@@ -627,8 +540,6 @@ k_scatter_clear(ScatterClearArgs A) {
 struct CompactArgs {
     const int64_t  *m_keys;
     const uint64_t *m_state;
-    const uint32_t *m_claimed;
-    const unsigned long long *m_ccnt;
     const uint32_t *m_spec_used;
     const uint64_t *m_spec_state;
     uint32_t  CM;
@@ -684,25 +595,41 @@ __device__ inline void emit_row(const CompactArgs &C, int64_t key,
 
 __global__ void __launch_bounds__(256)
 k_compact(CompactArgs C) {
-    /* The merge table's claimed list IS the compaction: entry i is a unique
-     * occupied slot, so output row = list index.  No scan of the table's
-     * capacity and no cursor atomics at all. */
-    size_t total = (size_t)*C.m_ccnt;
+    /* Two passes with ONE global cursor atomic per block: same-address
+     * global atomics cost ~12 ns each across the chip, so per-row (651K) or
+     * even per-wave (16K) cursors dominated this kernel (~200 us).  Pass 1
+     * counts the block's live slots, one atomicAdd claims the block's output
+     * range, pass 2 emits at LDS-cursor positions (LDS atomics are cheap). */
+    __shared__ unsigned long long blk_base;
+    __shared__ unsigned int blk_cnt;
     size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < total; i += stride) {
-        uint32_t slot = C.m_claimed[i];
-        emit_row(C, C.m_keys[slot],
-                 C.m_state + (size_t)slot * C.agg.n_aggs * 2, (int64_t)i);
+    if (threadIdx.x == 0) blk_cnt = 0;
+    __syncthreads();
+    unsigned int mine = 0;
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < C.CM;
+         i += stride)
+        mine += (C.m_keys[i] != EMPTY_KEY);
+    for (int off = 32; off; off >>= 1)
+        mine += (unsigned)__shfl_down((int)mine, off, 64);
+    if ((threadIdx.x & 63) == 0 && mine) atomicAdd(&blk_cnt, mine);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        blk_base = blk_cnt ? atomicAdd(C.n_out,
+                                       (unsigned long long)blk_cnt)
+                           : 0;
+        blk_cnt = 0;
     }
-    if (blockIdx.x == 0 && threadIdx.x == 0) {
-        unsigned long long n = (unsigned long long)total;
-        if (*C.m_spec_used) {
-            emit_row(C, EMPTY_KEY, C.m_spec_state, (int64_t)n);
-            n++;
-        }
-        *C.n_out = n;
+    __syncthreads();
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < C.CM;
+         i += stride) {
+        int64_t key = C.m_keys[i];
+        if (key == EMPTY_KEY) continue;
+        int64_t r = (int64_t)(blk_base + atomicAdd(&blk_cnt, 1u));
+        emit_row(C, key, C.m_state + i * C.agg.n_aggs * 2, r);
     }
+    if (blockIdx.x == 0 && threadIdx.x == 0 && *C.m_spec_used)
+        emit_row(C, EMPTY_KEY, C.m_spec_state,
+                 (int64_t)atomicAdd(C.n_out, 1ULL));
 }
 
 /* restore checkpointed partial states: insert raw state rows into a pane. */
@@ -745,9 +672,7 @@ k_restore(RestoreArgs R) {
             st = R.ring.spec_state + (size_t)p * R.agg.n_aggs * 2;
         } else {
             int64_t *keys = R.ring.keys + (size_t)p * R.ring.C;
-            int64_t s = table_upsert_rec(keys, R.ring.C, key, R.ring.err,
-                                         R.ring.claimed + (size_t)p * R.ring.C,
-                                         &R.ring.ccnt[p]);
+            int64_t s = table_upsert(keys, R.ring.C, key, R.ring.err);
             if (s < 0) continue;
             st = R.ring.state +
                  ((size_t)p * R.ring.C + (size_t)s) * R.agg.n_aggs * 2;
@@ -821,8 +746,6 @@ struct GpuOp {
     DeviceRing ring;
     int64_t  *m_keys;
     uint64_t *m_state, *m_fill;
-    uint32_t *m_claimed;
-    unsigned long long *m_ccnt;
     uint32_t *m_spec_used;
     uint64_t *m_spec_state;
     uint32_t CM;
@@ -881,20 +804,16 @@ static thread_local char g_err[512];
 
 static int ring_retire(GpuOp *o, uint32_t slot) {
     size_t na = o->agg.n_aggs;
-    ScatterClearArgs A = {};
-    A.keys = o->ring.keys + (size_t)slot * o->ring.C;
-    A.state = o->ring.state + (size_t)slot * o->ring.C * na * 2;
-    A.claimed = o->ring.claimed + (size_t)slot * o->ring.C;
-    A.ccnt = &o->ring.ccnt[slot];
-    A.na = (int32_t)na;
-    A.tag = o->ring.tag + slot;
-    A.fill = o->ring.fill + slot;
-    A.spec_used = o->ring.spec_used + slot;
-    A.spec_state = o->ring.spec_state + (size_t)slot * na * 2;
-    hipLaunchKernelGGL(k_scatter_clear, dim3(256), dim3(256), 0, o->stream,
-                       A);
-    HIP_CHECK(o, hipGetLastError());
-    HIP_CHECK(o, hipMemsetAsync(&o->ring.ccnt[slot], 0, 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.keys + (size_t)slot * o->ring.C, 0xFF,
+                                (size_t)o->ring.C * 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.state +
+                                    (size_t)slot * o->ring.C * na * 2,
+                                0, (size_t)o->ring.C * na * 16, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.tag + slot, 0xFF, 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.fill + slot, 0, 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.spec_used + slot, 0, 4, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->ring.spec_state + (size_t)slot * na * 2, 0,
+                                na * 16, o->stream));
     return 0;
 }
 
@@ -960,8 +879,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     ALLOC(o->ring.state, (size_t)o->ring.R * o->ring.C * na * 16);
     ALLOC(o->ring.tag, (size_t)o->ring.R * 8);
     ALLOC(o->ring.fill, (size_t)o->ring.R * 8);
-    ALLOC(o->ring.claimed, (size_t)o->ring.R * o->ring.C * 4);
-    ALLOC(o->ring.ccnt, (size_t)o->ring.R * 8);
     ALLOC(o->ring.spec_used, (size_t)o->ring.R * 4);
     ALLOC(o->ring.spec_state, (size_t)o->ring.R * na * 16);
     ALLOC(o->ring.err, 4);
@@ -969,8 +886,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     ALLOC(o->m_keys, (size_t)o->CM * 8);
     ALLOC(o->m_state, (size_t)o->CM * na * 16);
     ALLOC(o->m_fill, 8);
-    ALLOC(o->m_claimed, (size_t)o->CM * 4);
-    ALLOC(o->m_ccnt, 8);
     ALLOC(o->m_spec_used, 4);
     ALLOC(o->m_spec_state, na * 16);
     o->out_rows_cap = (int64_t)o->CM + 1;
@@ -982,7 +897,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
     hipMemset(o->ring.tag, 0xFF, (size_t)o->ring.R * 8);
     hipMemset(o->ring.fill, 0, (size_t)o->ring.R * 8);
-    hipMemset(o->ring.ccnt, 0, (size_t)o->ring.R * 8);
     hipMemset(o->ring.spec_used, 0, (size_t)o->ring.R * 4);
     hipMemset(o->ring.spec_state, 0, (size_t)o->ring.R * na * 16);
     hipMemset(o->ring.err, 0, 4);
@@ -990,7 +904,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_keys, 0xFF, (size_t)o->CM * 8);
     hipMemset(o->m_state, 0, (size_t)o->CM * na * 16);
     hipMemset(o->m_fill, 0, 8);
-    hipMemset(o->m_ccnt, 0, 8);
     hipMemset(o->m_spec_used, 0, 4);
     hipMemset(o->m_spec_state, 0, na * 16);
     hipStreamCreate(&o->stream);
@@ -1209,47 +1122,34 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                        uint64_t ws, uint64_t we, int raw_states,
                        uint64_t bin_ts) {
     size_t na = o->agg.n_aggs;
-    /* invariant: the merge table is clean here (scatter-cleared after the
-     * previous fire; zeroed at create) -- no per-fire memsets at all */
-    size_t total_claims = 0;
+    HIP_CHECK(o, hipMemsetAsync(o->m_keys, 0xFF, (size_t)o->CM * 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->m_state, 0, (size_t)o->CM * na * 16,
+                                o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->m_fill, 0, 8, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->m_spec_used, 0, 4, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->m_spec_state, 0, na * 16, o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
     if (!src.empty()) {
-        /* claimed counts of the source panes: one tiny D2H copy */
-        std::vector<unsigned long long> ccnt(o->ring.R);
-        HIP_CHECK(o, hipMemcpyAsync(ccnt.data(), o->ring.ccnt,
-                                    (size_t)o->ring.R * 8,
-                                    hipMemcpyDeviceToHost, o->stream));
-        HIP_CHECK(o, hipStreamSynchronize(o->stream));
         MergeArgs M = {};
         M.ring = o->ring;
         M.agg = o->agg;
         M.m_keys = o->m_keys;
         M.m_state = o->m_state;
         M.m_fill = o->m_fill;
-        M.m_claimed = o->m_claimed;
-        M.m_ccnt = o->m_ccnt;
         M.m_spec_used = o->m_spec_used;
         M.m_spec_state = o->m_spec_state;
         M.CM = o->CM;
         M.n_src = (int)src.size();
-        M.off[0] = 0;
-        for (size_t i = 0; i < src.size() && i < 64; i++) {
-            M.src[i] = src[i];
-            M.off[i + 1] = M.off[i] + ccnt[src[i]];
-        }
-        total_claims = (size_t)M.off[src.size()];
-        if (total_claims) {
-            int blocks = (int)((total_claims + 255) / 256);
-            if (blocks > 4096) blocks = 4096;
-            hipLaunchKernelGGL(k_merge, dim3(blocks), dim3(256), 0,
-                               o->stream, M);
-            HIP_CHECK(o, hipGetLastError());
-        }
+        for (size_t i = 0; i < src.size() && i < 64; i++) M.src[i] = src[i];
+        size_t total = (size_t)M.n_src * o->ring.C;
+        int blocks = (int)((total + 255) / 256);
+        if (blocks > 4096) blocks = 4096;
+        hipLaunchKernelGGL(k_merge, dim3(blocks), dim3(256), 0, o->stream, M);
+        HIP_CHECK(o, hipGetLastError());
     }
     CompactArgs C = {};
     C.m_keys = o->m_keys;
     C.m_state = o->m_state;
-    C.m_claimed = o->m_claimed;
-    C.m_ccnt = o->m_ccnt;
     C.m_spec_used = o->m_spec_used;
     C.m_spec_state = o->m_spec_state;
     C.CM = o->CM;
@@ -1260,32 +1160,13 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
     C.win_end = we;
     for (int i = 0; i < o->n_out_alloc && i < 16; i++) C.out[i] = o->d_out[i];
     C.n_out = o->d_n_out;
-    /* merge-table occupancy <= total pane claims: size the grid by that */
-    int blocks = (int)((total_claims + 256) / 256);
-    if (blocks > 1024) blocks = 1024;
+    int blocks = (int)((o->CM + 255) / 256);
+    if (blocks > 1024) blocks = 1024;   /* 1 global cursor atomic per block */
     hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(256), 0, o->stream, C);
     HIP_CHECK(o, hipGetLastError());
     unsigned long long n = 0;
     HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                 o->stream));
-    /* restore the clean-table invariant (does not touch d_out, so it can
-     * overlap the host-side output handling below) */
-    {
-        ScatterClearArgs A = {};
-        A.keys = o->m_keys;
-        A.state = o->m_state;
-        A.claimed = o->m_claimed;
-        A.ccnt = o->m_ccnt;
-        A.na = (int32_t)na;
-        A.fill = o->m_fill;
-        A.spec_used = o->m_spec_used;
-        A.spec_state = o->m_spec_state;
-        A.n_out = o->d_n_out;
-        hipLaunchKernelGGL(k_scatter_clear, dim3(256), dim3(256), 0,
-                           o->stream, A);
-        HIP_CHECK(o, hipGetLastError());
-        HIP_CHECK(o, hipMemsetAsync(o->m_ccnt, 0, 8, o->stream));
-    }
     HIP_CHECK(o, hipStreamSynchronize(o->stream));
     if (n == 0) return 0;
     int ncols = raw_states ? 0 : o->out_cols;
@@ -1596,8 +1477,6 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->ring.state);
     hipFree(o->ring.tag);
     hipFree(o->ring.fill);
-    hipFree(o->ring.claimed);
-    hipFree(o->ring.ccnt);
     hipFree(o->ring.spec_used);
     hipFree(o->ring.spec_state);
     hipFree(o->ring.err);
@@ -1605,8 +1484,6 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->m_keys);
     hipFree(o->m_state);
     hipFree(o->m_fill);
-    hipFree(o->m_claimed);
-    hipFree(o->m_ccnt);
     hipFree(o->m_spec_used);
     hipFree(o->m_spec_state);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
