@@ -761,6 +761,13 @@ struct GpuOp {
     hipStream_t stream;
     Staged stg;
 
+    /* one contiguous device status block [err, min_bin, tags[R]] mirrored
+     * into pinned host memory: ONE copy + ONE sync per watermark instead of
+     * three separate round trips (check_device_error + min_bin +
+     * sync_open_panes were ~60 us of host latency per watermark) */
+    uint64_t *d_status;
+    uint64_t *h_status;   /* pinned */
+
     /* host state machine (sliding_aggregating_window.rs:63-73) */
     int state;               /* 0 NoData 1 OnlyBufferedData 2 InMemoryData */
     uint64_t earliest, next_start;
@@ -877,12 +884,16 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         return fail(#p, e);
     ALLOC(o->ring.keys, (size_t)o->ring.R * o->ring.C * 8);
     ALLOC(o->ring.state, (size_t)o->ring.R * o->ring.C * na * 16);
-    ALLOC(o->ring.tag, (size_t)o->ring.R * 8);
+    ALLOC(o->d_status, (2 + (size_t)o->ring.R) * 8);
+    o->ring.err = (int *)o->d_status;
+    o->ring.min_bin = o->d_status + 1;
+    o->ring.tag = o->d_status + 2;
+    if (hipHostMalloc((void **)&o->h_status,
+                      (2 + (size_t)o->ring.R) * 8) != hipSuccess)
+        return fail("h_status", hipErrorOutOfMemory);
     ALLOC(o->ring.fill, (size_t)o->ring.R * 8);
     ALLOC(o->ring.spec_used, (size_t)o->ring.R * 4);
     ALLOC(o->ring.spec_state, (size_t)o->ring.R * na * 16);
-    ALLOC(o->ring.err, 4);
-    ALLOC(o->ring.min_bin, 8);
     ALLOC(o->m_keys, (size_t)o->CM * 8);
     ALLOC(o->m_state, (size_t)o->CM * na * 16);
     ALLOC(o->m_fill, 8);
@@ -1079,11 +1090,13 @@ API int arroyo_amd_process_batches_device(void *h,
     return 0;
 }
 
+/* one round trip: pulls err + min_bin + pane tags into h_status */
 static int check_device_error(GpuOp *o) {
-    int e = 0;
-    HIP_CHECK(o, hipMemcpyAsync(&e, o->ring.err, 4, hipMemcpyDeviceToHost,
-                                o->stream));
+    HIP_CHECK(o, hipMemcpyAsync(o->h_status, o->d_status,
+                                (2 + (size_t)o->ring.R) * 8,
+                                hipMemcpyDeviceToHost, o->stream));
     HIP_CHECK(o, hipStreamSynchronize(o->stream));
+    int e = *(int *)o->h_status;
     if (e == ERR_RING_CONFLICT) {
         snprintf(o->err_msg, sizeof o->err_msg,
                  "pane ring conflict: more than ring_panes=%u live bins; "
@@ -1099,13 +1112,10 @@ static int check_device_error(GpuOp *o) {
     return 0;
 }
 
-/* refresh host view of open panes from the device ring tags */
+/* refresh host view of open panes from h_status (populated by
+ * check_device_error, which every entry path calls first) */
 static int sync_open_panes(GpuOp *o) {
-    std::vector<uint64_t> tags(o->ring.R);
-    HIP_CHECK(o, hipMemcpyAsync(tags.data(), o->ring.tag,
-                                (size_t)o->ring.R * 8, hipMemcpyDeviceToHost,
-                                o->stream));
-    HIP_CHECK(o, hipStreamSynchronize(o->stream));
+    const uint64_t *tags = o->h_status + 2;
     o->open.clear();
     for (uint32_t s = 0; s < o->ring.R; s++) {
         if (tags[s] == EMPTY_TAG) continue;
@@ -1294,8 +1304,7 @@ API int arroyo_amd_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
      * OnlyBufferedData `earliest` only advances while not InMemoryData;
      * arrivals during InMemoryData are intentionally not folded -- see
      * sliding_aggregating_window.rs:635-647) */
-    uint64_t minb = ~0ULL;
-    HIP_CHECK(o, hipMemcpy(&minb, o->ring.min_bin, 8, hipMemcpyDeviceToHost));
+    uint64_t minb = o->h_status[1];
     HIP_CHECK(o, hipMemsetAsync(o->ring.min_bin, 0xFF, 8, o->stream));
     if (!o->cfg.is_tumbling && minb != ~0ULL) {
         if (o->state == 0) {
@@ -1475,12 +1484,11 @@ API void arroyo_amd_destroy(void *h) {
     }
     hipFree(o->ring.keys);
     hipFree(o->ring.state);
-    hipFree(o->ring.tag);
+    hipFree(o->d_status);
+    hipHostFree(o->h_status);
     hipFree(o->ring.fill);
     hipFree(o->ring.spec_used);
     hipFree(o->ring.spec_state);
-    hipFree(o->ring.err);
-    hipFree(o->ring.min_bin);
     hipFree(o->m_keys);
     hipFree(o->m_state);
     hipFree(o->m_fill);
