@@ -632,6 +632,11 @@ k_compact(CompactArgs C) {
                  (int64_t)atomicAdd(C.n_out, 1ULL));
 }
 
+__global__ void k_accum(unsigned long long *acc,
+                        const unsigned long long *n) {
+    *acc += *n;
+}
+
 /* restore checkpointed partial states: insert raw state rows into a pane. */
 struct RestoreArgs {
     const int64_t *key_col;           /* null when unkeyed */
@@ -753,6 +758,7 @@ struct GpuOp {
     int64_t *d_out[16];       /* device output columns, CM+1 rows each */
     int n_out_alloc;
     unsigned long long *d_n_out;
+    unsigned long long *d_emitted;  /* running device-side emitted rows */
     int64_t out_rows_cap;
 
     /* host-accumulated emission (emit_to_host) */
@@ -903,6 +909,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_n_out, 8);
+    ALLOC(o->d_emitted, 8);
 #undef ALLOC
     hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
     hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
@@ -917,6 +924,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_fill, 0, 8);
     hipMemset(o->m_spec_used, 0, 4);
     hipMemset(o->m_spec_state, 0, na * 16);
+    hipMemset(o->d_emitted, 0, 8);
     hipStreamCreate(&o->stream);
     /* pinned staging: 1M rows; enough columns for input batches and for
      * restore's raw-state batches */
@@ -1174,6 +1182,15 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
     if (blocks > 1024) blocks = 1024;   /* 1 global cursor atomic per block */
     hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(256), 0, o->stream, C);
     HIP_CHECK(o, hipGetLastError());
+    if (!o->cfg.emit_to_host && !raw_states) {
+        /* device-resident emission: the next pipeline stage's collector
+         * consumes d_out in place.  Accounting stays on device so firing
+         * costs NO host round trip; perf() folds the accumulator in. */
+        hipLaunchKernelGGL(k_accum, dim3(1), dim3(1), 0, o->stream,
+                           o->d_emitted, o->d_n_out);
+        HIP_CHECK(o, hipGetLastError());
+        return 0;
+    }
     unsigned long long n = 0;
     HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                 o->stream));
@@ -1195,10 +1212,6 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                                         hipMemcpyDeviceToHost, o->stream));
         }
         HIP_CHECK(o, hipStreamSynchronize(o->stream));
-    } else {
-        /* device-resident emission: the collector of the next pipeline stage
-         * consumes d_out in place; only account rows */
-        o->emitted_device_rows += (int64_t)n;
     }
     return 0;
 }
@@ -1496,6 +1509,7 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->m_spec_state);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
     hipFree(o->d_n_out);
+    hipFree(o->d_emitted);
     for (int i = 0; i < o->stg.ncols; i++) {
         hipHostFree(o->stg.buf[i]);
         hipFree(o->stg.dbuf[i]);
@@ -1517,7 +1531,13 @@ API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
                      : 0.0;
     *rows = o->update_rows;
     *launches = o->launches;
-    *emitted_device_rows = o->emitted_device_rows;
+    /* fold in the device-side fire accumulator (no per-fire round trips) */
+    unsigned long long dev_emitted = 0;
+    HIP_CHECK(o, hipMemcpyAsync(&dev_emitted, o->d_emitted, 8,
+                                hipMemcpyDeviceToHost, o->stream));
+    HIP_CHECK(o, hipStreamSynchronize(o->stream));
+    HIP_CHECK(o, hipMemsetAsync(o->d_emitted, 0, 8, o->stream));
+    *emitted_device_rows = o->emitted_device_rows + (int64_t)dev_emitted;
     o->update_kernel_ms = 0;
     o->sampled_launches = 0;
     o->update_rows = 0;
