@@ -262,7 +262,7 @@ __global__ void __launch_bounds__(256)
 k_update(UpdateArgs A) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     const int64_t *const *vc = A.vcols;
-    uint64_t local_min = ~0ULL;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride) {
         uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
@@ -272,7 +272,10 @@ k_update(UpdateArgs A) {
         if (bin < local_min) local_min = bin;
         if (A.mode == 1) continue;
         uint32_t p = (uint32_t)(q & (A.ring.R - 1));
-        claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
+        if (bin != last_bin) {
+            claim_tag_wave(A.ring.tag, p, bin, EMPTY_TAG, A.ring.err);
+            last_bin = bin;
+        }
         if (A.mode == 2) continue;
         int64_t key = A.key_col ? A.key_col[i] : 0;
         uint64_t *st;
@@ -327,7 +330,8 @@ k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
 __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                                       uint32_t *ls_pane, uint64_t *ls_st,
                                       int64_t row, uint64_t traw, int64_t key,
-                                      uint64_t &local_min) {
+                                      uint64_t &local_min,
+                                      uint64_t &last_bin) {
     const int na = A.agg.n_aggs;
     const int64_t *const *vc = A.vcols;
     uint64_t t = traw + A.ts_offset;
@@ -337,14 +341,19 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
     if (bin < local_min) local_min = bin;
     if (A.mode == 1) return;
     uint32_t p = (uint32_t)(q & (A.ring.R - 1));
-    claim_tag_wave(A.ring.tag, p, bin, A.ring.tag[p], A.ring.err);
+    /* register-cached bin: the tag claim (and its global tag read) runs
+     * once per thread per bin transition, not once per row */
+    if (bin != last_bin) {
+        claim_tag_wave(A.ring.tag, p, bin, EMPTY_TAG, A.ring.err);
+        last_bin = bin;
+    }
     if (A.mode == 2) return;
     /* try the LDS table first (2 probes), fall through to global */
     bool done = false;
     if (key != EMPTY_KEY) {
         uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
                      (LDS_SLOTS - 1);
-        for (int pr = 0; pr < 2 && !done; pr++) {
+        for (int pr = 0; pr < 4 && !done; pr++) {
             uint32_t s = (h + pr) & (LDS_SLOTS - 1);
             int64_t k = ls_key[s];
             bool claimed = false;
@@ -443,11 +452,11 @@ k_update_lds(UpdateArgs A) {
     extern __shared__ uint64_t ls_st[];   /* [LDS_SLOTS][n_aggs][2] */
     lds_init(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    uint64_t local_min = ~0ULL;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride)
         lds_update_row(A, ls_key, ls_pane, ls_st, i, (uint64_t)A.ts_col[i],
-                       A.key_col ? A.key_col[i] : 0, local_min);
+                       A.key_col ? A.key_col[i] : 0, local_min, last_bin);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
     lds_flush(A, ls_key, ls_pane, ls_st);
@@ -464,7 +473,7 @@ k_update_lds_vec(UpdateArgs A) {
     extern __shared__ uint64_t ls_st[];
     lds_init(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    uint64_t local_min = ~0ULL;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     int64_t n2 = A.n_rows >> 1;
     for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < n2;
          v += stride) {
@@ -475,14 +484,16 @@ k_update_lds_vec(UpdateArgs A) {
             k0 = (int64_t)kv.x;
             k1 = (int64_t)kv.y;
         }
-        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0, local_min);
+        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0, local_min,
+                       last_bin);
         lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y, k1,
-                       local_min);
+                       local_min, last_bin);
     }
     if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
         lds_update_row(A, ls_key, ls_pane, ls_st, A.n_rows - 1,
                        (uint64_t)A.ts_col[A.n_rows - 1],
-                       A.key_col ? A.key_col[A.n_rows - 1] : 0, local_min);
+                       A.key_col ? A.key_col[A.n_rows - 1] : 0, local_min,
+                       last_bin);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
     lds_flush(A, ls_key, ls_pane, ls_st);
@@ -686,6 +697,17 @@ k_restore(RestoreArgs R) {
     }
 }
 
+/* clear one pane's scalar metadata in a single launch (replaces four
+ * 4-8 B memset launches at retire) */
+__global__ void k_retire_meta(uint64_t *tag, uint64_t *fill,
+                              uint32_t *spec_used, uint64_t *spec_state,
+                              int na2) {
+    *tag = EMPTY_TAG;
+    *fill = 0;
+    *spec_used = 0;
+    for (int w = 0; w < na2; w++) spec_state[w] = 0;
+}
+
 /* K8: shuffle partition ids -- hash(key) -> contiguous range owner,
  * matching server_for_hash (crates/arroyo-types/src/lib.rs:640-647). */
 __global__ void __launch_bounds__(256)
@@ -750,6 +772,10 @@ struct GpuOp {
 
     DeviceRing ring;
     int64_t  *m_keys;
+    /* m_state/m_fill/m_spec_used/m_spec_state/d_n_out live in ONE
+     * contiguous zero-blob so a fire clears them with a single memset */
+    uint64_t *m_zero_blob;
+    size_t    m_zero_bytes;
     uint64_t *m_state, *m_fill;
     uint32_t *m_spec_used;
     uint64_t *m_spec_state;
@@ -822,11 +848,12 @@ static int ring_retire(GpuOp *o, uint32_t slot) {
     HIP_CHECK(o, hipMemsetAsync(o->ring.state +
                                     (size_t)slot * o->ring.C * na * 2,
                                 0, (size_t)o->ring.C * na * 16, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->ring.tag + slot, 0xFF, 8, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->ring.fill + slot, 0, 8, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->ring.spec_used + slot, 0, 4, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->ring.spec_state + (size_t)slot * na * 2, 0,
-                                na * 16, o->stream));
+    hipLaunchKernelGGL(k_retire_meta, dim3(1), dim3(1), 0, o->stream,
+                       o->ring.tag + slot, o->ring.fill + slot,
+                       o->ring.spec_used + slot,
+                       o->ring.spec_state + (size_t)slot * na * 2,
+                       (int)(na * 2));
+    HIP_CHECK(o, hipGetLastError());
     return 0;
 }
 
@@ -901,14 +928,16 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     ALLOC(o->ring.spec_used, (size_t)o->ring.R * 4);
     ALLOC(o->ring.spec_state, (size_t)o->ring.R * na * 16);
     ALLOC(o->m_keys, (size_t)o->CM * 8);
-    ALLOC(o->m_state, (size_t)o->CM * na * 16);
-    ALLOC(o->m_fill, 8);
-    ALLOC(o->m_spec_used, 4);
-    ALLOC(o->m_spec_state, na * 16);
+    o->m_zero_bytes = (size_t)o->CM * na * 16 + 8 + 8 + na * 16 + 8;
+    ALLOC(o->m_zero_blob, o->m_zero_bytes);
+    o->m_state = o->m_zero_blob;
+    o->m_fill = o->m_zero_blob + (size_t)o->CM * na * 2;
+    o->m_spec_used = (uint32_t *)(o->m_fill + 1);
+    o->m_spec_state = o->m_fill + 2;
+    o->d_n_out = (unsigned long long *)(o->m_spec_state + na * 2);
     o->out_rows_cap = (int64_t)o->CM + 1;
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
-    ALLOC(o->d_n_out, 8);
     ALLOC(o->d_emitted, 8);
 #undef ALLOC
     hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
@@ -920,10 +949,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->ring.err, 0, 4);
     hipMemset(o->ring.min_bin, 0xFF, 8);
     hipMemset(o->m_keys, 0xFF, (size_t)o->CM * 8);
-    hipMemset(o->m_state, 0, (size_t)o->CM * na * 16);
-    hipMemset(o->m_fill, 0, 8);
-    hipMemset(o->m_spec_used, 0, 4);
-    hipMemset(o->m_spec_state, 0, na * 16);
+    hipMemset(o->m_zero_blob, 0, o->m_zero_bytes);
     hipMemset(o->d_emitted, 0, 8);
     hipStreamCreate(&o->stream);
     /* pinned staging: 1M rows; enough columns for input batches and for
@@ -1141,12 +1167,8 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                        uint64_t bin_ts) {
     size_t na = o->agg.n_aggs;
     HIP_CHECK(o, hipMemsetAsync(o->m_keys, 0xFF, (size_t)o->CM * 8, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->m_state, 0, (size_t)o->CM * na * 16,
+    HIP_CHECK(o, hipMemsetAsync(o->m_zero_blob, 0, o->m_zero_bytes,
                                 o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->m_fill, 0, 8, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->m_spec_used, 0, 4, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->m_spec_state, 0, na * 16, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
     if (!src.empty()) {
         MergeArgs M = {};
         M.ring = o->ring;
@@ -1503,12 +1525,8 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->ring.spec_used);
     hipFree(o->ring.spec_state);
     hipFree(o->m_keys);
-    hipFree(o->m_state);
-    hipFree(o->m_fill);
-    hipFree(o->m_spec_used);
-    hipFree(o->m_spec_state);
+    hipFree(o->m_zero_blob);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
-    hipFree(o->d_n_out);
     hipFree(o->d_emitted);
     for (int i = 0; i < o->stg.ncols; i++) {
         hipHostFree(o->stg.buf[i]);
