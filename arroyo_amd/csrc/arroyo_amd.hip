@@ -324,9 +324,10 @@ k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
  * table (cutting global atomic traffic on hot keys -- nexmark sends ~50% of
  * bids to one auction), then flushes distinct (pane,key) entries to the
  * ring.  Entries evicted on LDS collision fall through to global atomics. */
-#define LDS_SLOTS 1024   /* x (8B key + 4B pane + n_aggs*16B) */
+#define LDS_SLOTS 1024   /* default; x (8B key + 4B pane + n_aggs*16B) */
 
 /* per-row body shared by the scalar and vectorized LDS kernels */
+template <int SLOTS>
 __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                                       uint32_t *ls_pane, uint64_t *ls_st,
                                       int64_t row, uint64_t traw, int64_t key,
@@ -352,9 +353,9 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
     bool done = false;
     if (key != EMPTY_KEY) {
         uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
-                     (LDS_SLOTS - 1);
+                     (SLOTS - 1);
         for (int pr = 0; pr < 4 && !done; pr++) {
-            uint32_t s = (h + pr) & (LDS_SLOTS - 1);
+            uint32_t s = (h + pr) & (SLOTS - 1);
             int64_t k = ls_key[s];
             bool claimed = false;
             if (k == EMPTY_KEY) {
@@ -416,10 +417,11 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
     }
 }
 
+template <int SLOTS>
 __device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
                                  uint32_t *ls_pane, uint64_t *ls_st) {
     const int na = A.agg.n_aggs;
-    for (int s = threadIdx.x; s < LDS_SLOTS; s += blockDim.x) {
+    for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) {
         int64_t key = ls_key[s];
         if (key == EMPTY_KEY) continue;
         uint32_t p = ls_pane[s];
@@ -434,10 +436,11 @@ __device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
 
 #define PANE_UNSET 0xFFFFFFFFu
 
+template <int SLOTS>
 __device__ inline void lds_init(const UpdateArgs &A, int64_t *ls_key,
                                 uint32_t *ls_pane, uint64_t *ls_st) {
     const int na = A.agg.n_aggs;
-    for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
         ls_key[i] = EMPTY_KEY;
         ls_pane[i] = PANE_UNSET;   /* sentinel: claim not yet visible */
         for (int w = 0; w < na * 2; w++) ls_st[(size_t)i * na * 2 + w] = 0;
@@ -450,28 +453,31 @@ k_update_lds(UpdateArgs A) {
     __shared__ int64_t  ls_key[LDS_SLOTS];
     __shared__ uint32_t ls_pane[LDS_SLOTS];
     extern __shared__ uint64_t ls_st[];   /* [LDS_SLOTS][n_aggs][2] */
-    lds_init(A, ls_key, ls_pane, ls_st);
+    lds_init<LDS_SLOTS>(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride)
-        lds_update_row(A, ls_key, ls_pane, ls_st, i, (uint64_t)A.ts_col[i],
-                       A.key_col ? A.key_col[i] : 0, local_min, last_bin);
+        lds_update_row<LDS_SLOTS>(A, ls_key, ls_pane, ls_st, i,
+                                  (uint64_t)A.ts_col[i],
+                                  A.key_col ? A.key_col[i] : 0, local_min,
+                                  last_bin);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
-    lds_flush(A, ls_key, ls_pane, ls_st);
+    lds_flush<LDS_SLOTS>(A, ls_key, ls_pane, ls_st);
 }
 
 /* vectorized variant: 16 B/lane loads of the ts (and key) columns, two rows
  * per thread per iteration -- requires 16 B-aligned column pointers (host
  * checks).  Fewer, fatter waves: wave-dispatch cost was measurable at one
  * 8 B load per thread. */
+template <int SLOTS>
 __global__ void __launch_bounds__(256)
 k_update_lds_vec(UpdateArgs A) {
-    __shared__ int64_t  ls_key[LDS_SLOTS];
-    __shared__ uint32_t ls_pane[LDS_SLOTS];
+    __shared__ int64_t  ls_key[SLOTS];
+    __shared__ uint32_t ls_pane[SLOTS];
     extern __shared__ uint64_t ls_st[];
-    lds_init(A, ls_key, ls_pane, ls_st);
+    lds_init<SLOTS>(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     int64_t n2 = A.n_rows >> 1;
@@ -484,20 +490,23 @@ k_update_lds_vec(UpdateArgs A) {
             k0 = (int64_t)kv.x;
             k1 = (int64_t)kv.y;
         }
-        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0, local_min,
-                       last_bin);
-        lds_update_row(A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y, k1,
-                       local_min, last_bin);
+        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0,
+                              local_min, last_bin);
+        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y,
+                              k1, local_min, last_bin);
     }
     if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
-        lds_update_row(A, ls_key, ls_pane, ls_st, A.n_rows - 1,
-                       (uint64_t)A.ts_col[A.n_rows - 1],
-                       A.key_col ? A.key_col[A.n_rows - 1] : 0, local_min,
-                       last_bin);
+        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, A.n_rows - 1,
+                              (uint64_t)A.ts_col[A.n_rows - 1],
+                              A.key_col ? A.key_col[A.n_rows - 1] : 0,
+                              local_min, last_bin);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
-    lds_flush(A, ls_key, ls_pane, ls_st);
+    lds_flush<SLOTS>(A, ls_key, ls_pane, ls_st);
 }
+
+template __global__ void k_update_lds_vec<1024>(UpdateArgs);
+template __global__ void k_update_lds_vec<2048>(UpdateArgs);
 
 /* ------------------------------------------------------------------ */
 /* K4: merge source panes into the merge table.                        */
@@ -1001,8 +1010,9 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
                (!A.key_col || ((uintptr_t)A.key_col & 15) == 0);
     int64_t units = vec ? (n_rows + 1) / 2 : n_rows;
     int64_t want = (units + 255) / 256;
-    /* 1024-block cap won the grid sweep at ~1M-row fused launches */
-    int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
+    /* 512-block cap won the grid sweep at ~1M-row fused launches (fewer
+     * blocks -> fewer same-key flush contenders on the global table) */
+    int blocks = (int)(want > 512 ? 512 : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
     /* sample kernel time on a subset of launches via a reusable event pool */
     bool sample = o->use_events && (o->launches & 7) == 0;
@@ -1021,11 +1031,18 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         ev = (int)(o->ev_inflight.size());
         hipEventRecord(o->ev_pool[ev].first, o->stream);
     }
-    size_t shmem = (size_t)LDS_SLOTS * o->agg.n_aggs * 16;
+    int slots = 1024;
+    if (const char *e = getenv("ARROYO_AMD_LDS_SLOTS")) slots = atoi(e);
+    size_t shmem = (size_t)slots * o->agg.n_aggs * 16;
     if (vec) {
-        hipLaunchKernelGGL(k_update_lds_vec, dim3(blocks), dim3(256), shmem,
-                           o->stream, A);
+        if (slots >= 2048)
+            hipLaunchKernelGGL(k_update_lds_vec<2048>, dim3(blocks),
+                               dim3(256), shmem, o->stream, A);
+        else
+            hipLaunchKernelGGL(k_update_lds_vec<1024>, dim3(blocks),
+                               dim3(256), shmem, o->stream, A);
     } else if (o->use_lds) {
+        shmem = (size_t)LDS_SLOTS * o->agg.n_aggs * 16;
         hipLaunchKernelGGL(k_update_lds, dim3(blocks), dim3(256), shmem,
                            o->stream, A);
     } else {

@@ -42,13 +42,16 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from arroyo_amd import cabi, nexmark  # noqa: E402
 from arroyo_amd.pipeline import NS  # noqa: E402
 
+import os as _os
+
 BATCH_ROWS = 65536
 EVENTS_PER_SEC = 1_000_000       # event-time rate of the synthetic stream
 # replay ring sized to 512 MiB so launches read from HBM, not the 256 MiB
 # Infinity Cache (PMC FETCH_SIZE showed a 4-batch ring was fully L3-resident)
 BASE_BATCHES = 512
 WIDTH_S, SLIDE_S = 10, 2
-LOG2_CAPACITY = 19               # ~130K distinct auctions per 2s pane
+LOG2_CAPACITY = int(_os.environ.get("BENCH_LOG2_CAP", "19"))
+# ~130K distinct auctions per 2s pane -> default 2^19 slots (25% load)
 RING_PANES = 16
 HBM_PEAK_GBPS = 8000.0           # spec peak (MI355X_MICROARCH.md)
 ALG_BYTES_PER_ROW = 16           # compulsory HBM read: auction i64 + ts i64
