@@ -543,3 +543,124 @@ class ExpJoinOp:
             self.close()
         except Exception:
             pass
+
+
+COUNT_DISTINCT = 5
+
+
+class AmdUpdatingConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_keys", ctypes.c_int32),
+        ("n_value_cols", ctypes.c_int32),
+        ("n_aggs", ctypes.c_int32),
+        ("agg_ops", ctypes.c_int32 * 8),
+        ("agg_col", ctypes.c_int32 * 8),
+        ("log2_capacity", ctypes.c_uint32),
+        ("log2_nodes", ctypes.c_uint32),
+        ("log2_out_cap", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_updagg_config(aggs, n_keys=1, n_value_cols=0, log2_capacity=16,
+                       log2_nodes=20, log2_out_cap=20, device=0,
+                       emit_to_host=True):
+    cfg = AmdUpdatingConfig()
+    cfg.n_keys = n_keys
+    cfg.n_value_cols = n_value_cols
+    cfg.n_aggs = len(aggs)
+    for i, (op, col) in enumerate(aggs):
+        cfg.agg_ops[i] = op
+        cfg.agg_col[i] = col
+    cfg.log2_capacity = log2_capacity
+    cfg.log2_nodes = log2_nodes
+    cfg.log2_out_cap = log2_out_cap
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+class UpdAggOp:
+    """One updating (non-windowed) aggregate operator behind the C ABI,
+    mirroring IncrementalAggregatingFunc's ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/incremental_aggregator.rs)."""
+
+    def __init__(self, lib, prefix, cfg):
+        p = prefix + "updagg_"
+        g = lambda n: getattr(lib, p + n)
+        self._fn = {}
+        self._fn["create"] = g("create")
+        self._fn["create"].restype = ctypes.c_void_p
+        self._fn["create"].argtypes = [ctypes.POINTER(AmdUpdatingConfig)]
+        self._fn["process_batch"] = g("process_batch")
+        self._fn["process_batch"].restype = ctypes.c_int
+        self._fn["process_batch"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64]
+        self._fn["flush"] = g("flush")
+        self._fn["flush"].restype = ctypes.c_int
+        self._fn["flush"].argtypes = [ctypes.c_void_p,
+                                      ctypes.POINTER(AmdOutBatch)]
+        self._fn["checkpoint_drain"] = g("checkpoint_drain")
+        self._fn["checkpoint_drain"].restype = ctypes.c_int
+        self._fn["checkpoint_drain"].argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(AmdOutBatch)]
+        self._fn["restore"] = g("restore")
+        self._fn["restore"].restype = ctypes.c_int
+        self._fn["restore"].argtypes = [
+            ctypes.c_void_p, ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32, ctypes.c_int64]
+        self._fn["free_out"] = getattr(lib, prefix + "free_out")
+        self._fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+        self._fn["destroy"] = g("destroy")
+        self._fn["destroy"].argtypes = [ctypes.c_void_p]
+        self._fn["last_error"] = g("last_error")
+        self._fn["last_error"].restype = ctypes.c_char_p
+        self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{p}create failed")
+
+    def _check(self, rc):
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def process_batch(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["process_batch"](self._h, arr, len(keep),
+                                              n_rows))
+
+    def flush(self):
+        out = AmdOutBatch()
+        self._check(self._fn["flush"](self._h, ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def checkpoint_drain(self, which):
+        out = AmdOutBatch()
+        self._check(self._fn["checkpoint_drain"](self._h, which,
+                                                 ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore(self, which, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["restore"](self._h, which, arr, len(keep),
+                                        n_rows))
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

@@ -141,3 +141,8 @@ def make_session_op(cfg):
 def make_expjoin_op(cfg):
     from arroyo_amd.cabi import ExpJoinOp
     return ExpJoinOp(lib(), "arroyo_amd_", cfg)
+
+
+def make_updagg_op(cfg):
+    from arroyo_amd.cabi import UpdAggOp
+    return UpdAggOp(lib(), "arroyo_amd_", cfg)

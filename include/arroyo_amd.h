@@ -214,6 +214,38 @@ int arroyo_amd_expjoin_restore(void *h, int32_t side,
 void arroyo_amd_expjoin_destroy(void *h);
 const char *arroyo_amd_expjoin_last_error(void *h);
 
+/* ---- updating (non-windowed) aggregate --------------------------------
+ * Replaces IncrementalAggregatingFunc
+ * (crates/arroyo-worker/src/arrow/incremental_aggregator.rs) behind the
+ * same ArrowOperator surface:
+ *   updagg_create        <-> constructor (:1035-1193)
+ *   updagg_process_batch <-> process_batch (:931-949) ->
+ *                            keyed_aggregate/global_aggregate (:778-884);
+ *                            cols [key?, vals..., is_retract] where
+ *                            is_retract mirrors _updating_meta.is_retract
+ *                            (get_retracts :740-758)
+ *   updagg_flush         <-> handle_tick/on_close -> flush (:637-737):
+ *                            out columns [key?, agg outputs...,
+ *                            is_retract]
+ *   updagg_checkpoint_drain(which): which=0 scalar accumulator states
+ *                            (the "a" table, checkpoint_sliding :271-340),
+ *                            which=1 distinct-value multiset rows
+ *                            [key?, agg_index, value, net_count] (the "b"
+ *                            table, checkpoint_batch :342-418)
+ *   updagg_restore       <-> on_start -> initialize (:446-599)
+ */
+void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg);
+int arroyo_amd_updagg_process_batch(void *h, const int64_t *const *cols,
+                                    int32_t n_cols, int64_t n_rows);
+int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out);
+int arroyo_amd_updagg_checkpoint_drain(void *h, int32_t which,
+                                       AmdOutBatch *out);
+int arroyo_amd_updagg_restore(void *h, int32_t which,
+                              const int64_t *const *cols, int32_t n_cols,
+                              int64_t n_rows);
+void arroyo_amd_updagg_destroy(void *h);
+const char *arroyo_amd_updagg_last_error(void *h);
+
 #ifdef __cplusplus
 }
 #endif

@@ -33,7 +33,9 @@ enum AmdAggOp {
     AMD_AGG_SUM   = 1, /* partial: sum (i64)    final: sum            */
     AMD_AGG_MIN   = 2, /* partial: min          final: min            */
     AMD_AGG_MAX   = 3, /* partial: max          final: max            */
-    AMD_AGG_AVG   = 4  /* partial: (count,sum f64)  final: sum/divide */
+    AMD_AGG_AVG   = 4, /* partial: (count,sum f64)  final: sum/divide */
+    AMD_AGG_COUNT_DISTINCT = 5 /* updating aggregate only: exact count of
+                                  distinct values under append+retract */
 };
 
 #define AMD_MAX_AGGS 8
@@ -130,6 +132,38 @@ typedef struct {
     int32_t  device;
     int32_t  emit_to_host;
 } AmdExpJoinConfig;
+
+/* Updating (non-windowed) aggregate configuration.  Mirrors
+ * api::UpdatingAggregateOperator as decoded by
+ * IncrementalAggregatingFunc's constructor
+ * (crates/arroyo-worker/src/arrow/incremental_aggregator.rs:1035-1193):
+ * aggregate exprs + flush interval + ttl.  Semantics (restated in
+ * oracle/arroyo_oracle.c and arroyo_amd/csrc/updagg.hip): per key,
+ * retractable accumulators updated row by row (appends and, when the input
+ * carries _updating_meta.is_retract, retractions); on flush
+ * (handle_tick -> flush :637-737) every key touched since the last flush
+ * emits retract(previously emitted value) + append(new value) -- the
+ * retract omitted for first-time keys, the append omitted (retract only)
+ * when the key's rows have all been retracted, and the whole pair skipped
+ * when the value did not change.  COUNT/SUM/AVG retract by subtraction
+ * (the reference's Sliding accumulators); COUNT DISTINCT keeps an exact
+ * per-key value multiset (the reference's Batch accumulator,
+ * IncrementalState::Batch :84-174); MIN/MAX are append-only here (the
+ * reference re-aggregates a stored multiset on demand; unsupported-retract
+ * is a loud error).  Flush cadence is driven by the caller (the reference's
+ * tick timer). */
+typedef struct {
+    int32_t  n_keys;            /* 0 or 1 */
+    int32_t  n_value_cols;
+    int32_t  n_aggs;
+    int32_t  agg_ops[AMD_MAX_AGGS];
+    int32_t  agg_col[AMD_MAX_AGGS];
+    uint32_t log2_capacity;     /* key slots (GPU) */
+    uint32_t log2_nodes;        /* distinct-value node pool (GPU) */
+    uint32_t log2_out_cap;
+    int32_t  device;
+    int32_t  emit_to_host;
+} AmdUpdatingConfig;
 
 /* Output batch, allocated by the callee; free with *_free_out.
  * Column order: [key (if n_keys)], agg outputs (one column per agg),

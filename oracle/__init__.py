@@ -47,3 +47,8 @@ def make_session_op(cfg):
 def make_expjoin_op(cfg):
     from arroyo_amd.cabi import ExpJoinOp
     return ExpJoinOp(lib(), "oracle_", cfg)
+
+
+def make_updagg_op(cfg):
+    from arroyo_amd.cabi import UpdAggOp
+    return UpdAggOp(lib(), "oracle_", cfg)

@@ -1395,3 +1395,379 @@ ORACLE_API void oracle_expjoin_destroy(void *h) {
     free(o->out);
     free(o);
 }
+
+/* ================================================================== */
+/* Updating (non-windowed) aggregate oracle.
+ *
+ * Restated from crates/arroyo-worker/src/arrow/incremental_aggregator.rs:
+ *   - process_batch :931-949 routes to keyed_aggregate :826-884 /
+ *     global_aggregate :778-824: per row, the key's accumulators are
+ *     updated (or retracted when the batch's _updating_meta.is_retract is
+ *     set for the row, get_retracts :740-758);
+ *   - COUNT/SUM/AVG use retractable "Sliding" accumulators
+ *     (IncrementalState::Sliding :78-83 -- retract = subtract);
+ *     COUNT DISTINCT uses the "Batch" value-multiset accumulator
+ *     (IncrementalState::Batch :84-174: per-value refcount, evaluate counts
+ *     values with count > 0); MIN/MAX would also be Batch-type -- here they
+ *     are append-only and a retraction is a loud error (stated scope);
+ *   - flush :637-737 (driven by the caller, the reference's tick timer):
+ *     every key touched since the last flush emits
+ *     retract(value as of last flush) + append(current value); the retract
+ *     is omitted for keys first seen this interval (updated_keys value None,
+ *     keyed_aggregate :839-856), the append is omitted when all the key's
+ *     rows are retracted (null timestamp check :670-678 -- here: live row
+ *     count 0), and the pair is skipped when the value is unchanged
+ *     (:649-661).
+ */
+
+typedef struct {
+    int64_t cap, n;
+    int64_t *vals;            /* distinct values */
+    int64_t *cnt;             /* net refcounts */
+} UMultiset;
+
+typedef struct {
+    int64_t rows;             /* live row count (presence) */
+    int64_t st[AMD_MAX_AGGS * 2];
+    UMultiset ms[AMD_MAX_AGGS];   /* per COUNT_DISTINCT agg */
+    int emitted;
+    int64_t last[AMD_MAX_AGGS];   /* last emitted values (AVG: f64 bits) */
+    int changed;
+} UEntry;
+
+typedef struct {
+    AmdUpdatingConfig cfg;
+    int64_t map_cap, map_n;
+    int64_t *map_keys;
+    uint8_t *map_used;
+    UEntry *ent;
+    int out_cols;
+    int64_t out_rows, out_cap;
+    int64_t **out;
+    char err[256];
+} UOp;
+
+static void ugrow(UOp *o);
+
+static int64_t ukey_slot(UOp *o, int64_t key) {
+    if (o->map_n * 10 >= o->map_cap * 7) ugrow(o);
+    uint64_t m = (uint64_t)o->map_cap - 1;
+    uint64_t i = hash64((uint64_t)key) & m;
+    while (o->map_used[i] && o->map_keys[i] != key) i = (i + 1) & m;
+    if (!o->map_used[i]) {
+        o->map_used[i] = 1;
+        o->map_keys[i] = key;
+        o->map_n++;
+        UEntry *e = &o->ent[i];
+        memset(e, 0, sizeof *e);
+        for (int a = 0; a < o->cfg.n_aggs; a++) {
+            switch (o->cfg.agg_ops[a]) {
+            case AMD_AGG_MIN: e->st[2 * a] = INT64_MAX; break;
+            case AMD_AGG_MAX: e->st[2 * a] = INT64_MIN; break;
+            case AMD_AGG_AVG: e->st[2 * a + 1] = d_to_bits(0.0); break;
+            }
+        }
+    }
+    return (int64_t)i;
+}
+
+static void ugrow(UOp *o) {
+    int64_t ocap = o->map_cap;
+    int64_t *ok = o->map_keys; uint8_t *ou = o->map_used;
+    UEntry *oe = o->ent;
+    o->map_cap <<= 1; o->map_n = 0;
+    o->map_keys = malloc((size_t)o->map_cap * 8);
+    o->map_used = calloc((size_t)o->map_cap, 1);
+    o->ent = calloc((size_t)o->map_cap, sizeof(UEntry));
+    for (int64_t i = 0; i < ocap; i++) {
+        if (!ou[i]) continue;
+        int64_t s = ukey_slot(o, ok[i]);
+        o->ent[s] = oe[i];
+    }
+    free(ok); free(ou); free(oe);
+}
+
+ORACLE_API void *oracle_updagg_create(const AmdUpdatingConfig *cfg) {
+    if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->n_value_cols < 0 ||
+        cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS)
+        return NULL;
+    UOp *o = calloc(1, sizeof(UOp));
+    o->cfg = *cfg;
+    o->map_cap = 64;
+    o->map_keys = malloc((size_t)o->map_cap * 8);
+    o->map_used = calloc((size_t)o->map_cap, 1);
+    o->ent = calloc((size_t)o->map_cap, sizeof(UEntry));
+    o->out_cols = cfg->n_keys + cfg->n_aggs + 1;  /* + is_retract */
+    o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
+    return o;
+}
+
+ORACLE_API const char *oracle_updagg_last_error(void *h) {
+    return h ? ((UOp *)h)->err : "null handle / invalid config";
+}
+
+static void ums_add(UMultiset *m, int64_t v, int64_t d) {
+    for (int64_t i = 0; i < m->n; i++)
+        if (m->vals[i] == v) { m->cnt[i] += d; return; }
+    if (m->n == m->cap) {
+        m->cap = m->cap ? m->cap * 2 : 16;
+        m->vals = realloc(m->vals, (size_t)m->cap * 8);
+        m->cnt = realloc(m->cnt, (size_t)m->cap * 8);
+    }
+    m->vals[m->n] = v;
+    m->cnt[m->n++] = d;
+}
+
+ORACLE_API int oracle_updagg_process_batch(void *h,
+                                           const int64_t *const *cols,
+                                           int32_t n_cols, int64_t n_rows) {
+    UOp *o = h;
+    const AmdUpdatingConfig *c = &o->cfg;
+    int want = c->n_keys + c->n_value_cols + 1;
+    if (n_cols != want) {
+        snprintf(o->err, sizeof o->err, "expected %d cols, got %d", want,
+                 n_cols);
+        return 1;
+    }
+    const int64_t *retr = cols[n_cols - 1];
+    for (int64_t r = 0; r < n_rows; r++) {
+        int64_t key = c->n_keys ? cols[0][r] : 0;
+        int64_t s = ukey_slot(o, key);
+        UEntry *e = &o->ent[s];
+        int64_t d = retr[r] ? -1 : 1;
+        e->rows += d;
+        e->changed = 1;
+        for (int a = 0; a < c->n_aggs; a++) {
+            int64_t v = c->agg_col[a] >= 0
+                            ? cols[c->n_keys + c->agg_col[a]][r] : 0;
+            switch (c->agg_ops[a]) {
+            case AMD_AGG_COUNT: e->st[2 * a] += d; break;
+            case AMD_AGG_SUM:   e->st[2 * a] += d * v; break;
+            case AMD_AGG_AVG:
+                e->st[2 * a] += d;
+                e->st[2 * a + 1] =
+                    d_to_bits(bits_to_d(e->st[2 * a + 1]) + (double)d * v);
+                break;
+            case AMD_AGG_MIN:
+                if (d < 0) {
+                    snprintf(o->err, sizeof o->err,
+                             "MIN does not support retraction");
+                    return 1;
+                }
+                if (v < e->st[2 * a]) e->st[2 * a] = v;
+                break;
+            case AMD_AGG_MAX:
+                if (d < 0) {
+                    snprintf(o->err, sizeof o->err,
+                             "MAX does not support retraction");
+                    return 1;
+                }
+                if (v > e->st[2 * a]) e->st[2 * a] = v;
+                break;
+            case AMD_AGG_COUNT_DISTINCT:
+                ums_add(&e->ms[a], v, d);
+                break;
+            }
+        }
+    }
+    return 0;
+}
+
+static void ueval(const UOp *o, const UEntry *e, int64_t *out) {
+    const AmdUpdatingConfig *c = &o->cfg;
+    for (int a = 0; a < c->n_aggs; a++) {
+        switch (c->agg_ops[a]) {
+        case AMD_AGG_COUNT:
+        case AMD_AGG_SUM:
+        case AMD_AGG_MIN:
+        case AMD_AGG_MAX:
+            out[a] = e->st[2 * a];
+            break;
+        case AMD_AGG_AVG:
+            out[a] = d_to_bits(bits_to_d(e->st[2 * a + 1]) /
+                               (double)e->st[2 * a]);
+            break;
+        case AMD_AGG_COUNT_DISTINCT: {
+            int64_t n = 0;
+            for (int64_t i = 0; i < e->ms[a].n; i++)
+                if (e->ms[a].cnt[i] > 0) n++;
+            out[a] = n;
+            break;
+        }
+        }
+    }
+}
+
+static void uout_reserve(UOp *o, int64_t add) {
+    if (o->out_rows + add <= o->out_cap) return;
+    int64_t ncap = o->out_cap ? o->out_cap : 1024;
+    while (ncap < o->out_rows + add) ncap *= 2;
+    for (int i = 0; i < o->out_cols; i++)
+        o->out[i] = realloc(o->out[i], (size_t)ncap * 8);
+    o->out_cap = ncap;
+}
+
+static void uemit(UOp *o, int64_t key, const int64_t *vals, int retract) {
+    uout_reserve(o, 1);
+    int64_t r = o->out_rows++;
+    int col = 0;
+    if (o->cfg.n_keys) o->out[col++][r] = key;
+    for (int a = 0; a < o->cfg.n_aggs; a++) o->out[col++][r] = vals[a];
+    o->out[col][r] = retract;
+}
+
+ORACLE_API int oracle_updagg_flush(void *h, AmdOutBatch *out) {
+    UOp *o = h;
+    const AmdUpdatingConfig *c = &o->cfg;
+    for (int64_t s = 0; s < o->map_cap; s++) {
+        if (!o->map_used[s] || !o->ent[s].changed) continue;
+        UEntry *e = &o->ent[s];
+        e->changed = 0;
+        int64_t cur[AMD_MAX_AGGS];
+        if (e->rows > 0) ueval(o, e, cur);
+        if (e->emitted) {
+            if (e->rows > 0 &&
+                memcmp(cur, e->last, (size_t)c->n_aggs * 8) == 0)
+                continue;  /* unchanged: skip the retract/append pair */
+            uemit(o, o->map_keys[s], e->last, 1);
+            e->emitted = 0;
+        }
+        if (e->rows > 0) {
+            uemit(o, o->map_keys[s], cur, 0);
+            memcpy(e->last, cur, (size_t)c->n_aggs * 8);
+            e->emitted = 1;
+        }
+    }
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = o->out_rows;
+        out->n_cols = o->out_cols;
+        out->cols = calloc((size_t)o->out_cols, sizeof(void *));
+        out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
+        for (int a = 0; a < c->n_aggs; a++)
+            if (c->agg_ops[a] == AMD_AGG_AVG)
+                out->is_f64[c->n_keys + a] = 1;
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
+            if (o->out_rows)
+                memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
+        }
+        o->out_rows = 0;
+    }
+    return 0;
+}
+
+/* which=0: scalar accumulator states [key?, rows, st words..., emitted,
+ * last words...]; which=1: distinct-value multiset rows
+ * [key?, agg_index, value, net_count] */
+ORACLE_API int oracle_updagg_checkpoint_drain(void *h, int32_t which,
+                                              AmdOutBatch *out) {
+    UOp *o = h;
+    const AmdUpdatingConfig *c = &o->cfg;
+    int ncols;
+    int64_t total = 0;
+    if (which == 0) {
+        ncols = c->n_keys + 1 + 2 * c->n_aggs + 1 + c->n_aggs;
+        for (int64_t s = 0; s < o->map_cap; s++)
+            if (o->map_used[s]) total++;
+    } else {
+        ncols = c->n_keys + 3;
+        for (int64_t s = 0; s < o->map_cap; s++)
+            if (o->map_used[s])
+                for (int a = 0; a < c->n_aggs; a++)
+                    for (int64_t i = 0; i < o->ent[s].ms[a].n; i++)
+                        if (o->ent[s].ms[a].cnt[i] != 0) total++;
+    }
+    memset(out, 0, sizeof *out);
+    out->n_rows = total;
+    out->n_cols = ncols;
+    out->cols = calloc((size_t)ncols, sizeof(void *));
+    out->is_f64 = calloc((size_t)ncols, sizeof(int32_t));
+    for (int i = 0; i < ncols; i++)
+        out->cols[i] = malloc((size_t)(total ? total : 1) * 8);
+    int64_t r = 0;
+    for (int64_t s = 0; s < o->map_cap; s++) {
+        if (!o->map_used[s]) continue;
+        UEntry *e = &o->ent[s];
+        if (which == 0) {
+            int col = 0;
+            if (c->n_keys) ((int64_t *)out->cols[col++])[r] = o->map_keys[s];
+            ((int64_t *)out->cols[col++])[r] = e->rows;
+            for (int w = 0; w < 2 * c->n_aggs; w++)
+                ((int64_t *)out->cols[col++])[r] = e->st[w];
+            ((int64_t *)out->cols[col++])[r] = e->emitted;
+            for (int a = 0; a < c->n_aggs; a++)
+                ((int64_t *)out->cols[col++])[r] = e->last[a];
+            r++;
+        } else {
+            for (int a = 0; a < c->n_aggs; a++)
+                for (int64_t i = 0; i < e->ms[a].n; i++) {
+                    if (e->ms[a].cnt[i] == 0) continue;
+                    int col = 0;
+                    if (c->n_keys)
+                        ((int64_t *)out->cols[col++])[r] = o->map_keys[s];
+                    ((int64_t *)out->cols[col++])[r] = a;
+                    ((int64_t *)out->cols[col++])[r] = e->ms[a].vals[i];
+                    ((int64_t *)out->cols[col])[r] = e->ms[a].cnt[i];
+                    r++;
+                }
+        }
+    }
+    return 0;
+}
+
+ORACLE_API int oracle_updagg_restore(void *h, int32_t which,
+                                     const int64_t *const *cols,
+                                     int32_t n_cols, int64_t n_rows) {
+    UOp *o = h;
+    const AmdUpdatingConfig *c = &o->cfg;
+    if (which == 0) {
+        int want = c->n_keys + 1 + 2 * c->n_aggs + 1 + c->n_aggs;
+        if (n_cols != want) {
+            snprintf(o->err, sizeof o->err, "restore(0) expects %d cols",
+                     want);
+            return 1;
+        }
+        for (int64_t r = 0; r < n_rows; r++) {
+            int64_t s = ukey_slot(o, c->n_keys ? cols[0][r] : 0);
+            UEntry *e = &o->ent[s];
+            int col = c->n_keys;
+            e->rows = cols[col++][r];
+            for (int w = 0; w < 2 * c->n_aggs; w++)
+                e->st[w] = cols[col++][r];
+            e->emitted = (int)cols[col++][r];
+            for (int a = 0; a < c->n_aggs; a++)
+                e->last[a] = cols[col++][r];
+        }
+    } else {
+        int want = c->n_keys + 3;
+        if (n_cols != want) {
+            snprintf(o->err, sizeof o->err, "restore(1) expects %d cols",
+                     want);
+            return 1;
+        }
+        for (int64_t r = 0; r < n_rows; r++) {
+            int64_t s = ukey_slot(o, c->n_keys ? cols[0][r] : 0);
+            int a = (int)cols[c->n_keys][r];
+            ums_add(&o->ent[s].ms[a], cols[c->n_keys + 1][r],
+                    cols[c->n_keys + 2][r]);
+        }
+    }
+    return 0;
+}
+
+ORACLE_API void oracle_updagg_destroy(void *h) {
+    UOp *o = h;
+    if (!o) return;
+    for (int64_t s = 0; s < o->map_cap; s++) {
+        if (!o->map_used[s]) continue;
+        for (int a = 0; a < o->cfg.n_aggs; a++) {
+            free(o->ent[s].ms[a].vals);
+            free(o->ent[s].ms[a].cnt);
+        }
+    }
+    free(o->map_keys); free(o->map_used); free(o->ent);
+    for (int i = 0; i < o->out_cols; i++) free(o->out[i]);
+    free(o->out);
+    free(o);
+}

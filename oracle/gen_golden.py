@@ -64,16 +64,54 @@ def main():
             "event_type_dict": event_types,
         },
     }
-    with open(f"{OUT}/inputs.json", "w") as f:
-        json.dump(fixtures, f)
+
+    # Debezium updating-aggregate fixture: aggregate_updates.json is a
+    # debezium_json stream (ops c/u/d over orders with pk=id) feeding
+    # debezium_agg.sql's non-windowed GROUP BY (IncrementalAggregatingFunc).
+    # Encode to columns: op (0 append / 1 retract), product id, customer id,
+    # quantity -- an update becomes retract(before)+append(after).
+    agg_upd = load_rows(f"{REF}/inputs/aggregate_updates.json")
+    products, customers = {}, {}
+
+    def pid(name):
+        return products.setdefault(name, len(products))
+
+    def cid(name):
+        return customers.setdefault(name, len(customers))
+
+    ops, prods, custs, qtys = [], [], [], []
+
+    def emit(rec, retract):
+        ops.append(1 if retract else 0)
+        prods.append(pid(rec["product_name"]))
+        custs.append(cid(rec["customer_name"]))
+        qtys.append(rec["quantity"])
+
+    for r in agg_upd:
+        if r["op"] == "c":
+            emit(r["after"], False)
+        elif r["op"] == "u":
+            emit(r["before"], True)
+            emit(r["after"], False)
+        elif r["op"] == "d":
+            emit(r["before"], True)
+    fixtures["aggregate_updates"] = {
+        "op": ops, "product": prods, "customer": custs, "quantity": qtys,
+        "product_dict": [k for k, _ in
+                         sorted(products.items(), key=lambda kv: kv[1])],
+    }
 
     for name in ("sliding_window_end", "hourly_by_event_type",
                  "tight_watermark", "most_active_driver_last_hour",
                  "windowed_inner_join", "session_window",
-                 "global_session_window", "updating_inner_join"):
+                 "global_session_window", "updating_inner_join",
+                 "debezium_agg", "filter_updating_aggregates"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)
+
+    with open(f"{OUT}/inputs.json", "w") as f:
+        json.dump(fixtures, f)
 
     print("wrote fixtures to", os.path.abspath(OUT))
 
