@@ -15,12 +15,13 @@ from arroyo_amd.cabi import AmdOutBatch, WindowOp, _out_to_numpy
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libarroyo_amd.so")
 _SRCS = [os.path.join(_DIR, "csrc", f)
-         for f in ("arroyo_amd.hip", "session.hip", "expjoin.hip")]
+         for f in ("arroyo_amd.hip", "session.hip", "expjoin.hip",
+                   "updagg.hip")]
 
 _lib = None
 
 HIPCC_CMD = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-             "-shared", "-fvisibility=hidden", "-parallel-jobs=3",
+             "-shared", "-fvisibility=hidden", "-parallel-jobs=4",
              "-o", _SO] + _SRCS
 
 

@@ -217,3 +217,97 @@ def test_updagg_oracle_min_max_append_only():
     with pytest.raises(RuntimeError, match="retraction"):
         op.process_batch([k, v, np.array([1], dtype=np.int64)])
     op.close()
+
+
+# ---------------------------------------------------------------- GPU parity
+
+
+@pytest.mark.gpu
+def test_debezium_agg_golden_gpu():
+    from arroyo_amd import gpu
+    run_debezium_agg_golden(gpu.make_updagg_op)
+
+
+@pytest.mark.gpu
+def test_filter_updating_golden_gpu():
+    from arroyo_amd import gpu
+    run_filter_updating_golden(gpu.make_updagg_op)
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_vs_oracle_fuzz():
+    """Final merged state AND per-flush emission sets must match the oracle
+    (emission order within a flush differs; retract/append pairing and the
+    set of emitted rows must not)."""
+    from arroyo_amd import gpu
+    for seed in (5, 21):
+        _k, _v, _r, eg = updagg_fuzz(gpu.make_updagg_op, seed=seed, n=6000)
+        _k2, _v2, _r2, eo = updagg_fuzz(oracle.make_updagg_op, seed=seed,
+                                        n=6000)
+        assert merge_debezium(eg) == merge_debezium(eo)
+        assert sorted(eg) == sorted(eo)
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_delete_and_reappear():
+    from arroyo_amd import gpu
+    op = gpu.make_updagg_op(cabi.make_updagg_config(
+        [(cabi.COUNT, -1)], n_keys=1))
+    k = np.array([3], dtype=np.int64)
+    op.process_batch([k, np.array([0], dtype=np.int64)])
+    assert rows_of(op.flush()) == [(3, 1, 0)]
+    op.process_batch([k, np.array([1], dtype=np.int64)])
+    assert rows_of(op.flush()) == [(3, 1, 1)]
+    op.process_batch([k, np.array([0], dtype=np.int64)])
+    assert rows_of(op.flush()) == [(3, 1, 0)]
+    op.process_batch([k, np.array([0], dtype=np.int64)])
+    op.process_batch([k, np.array([1], dtype=np.int64)])
+    assert rows_of(op.flush()) == []
+    op.close()
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_checkpoint_roundtrip():
+    from arroyo_amd import gpu
+    key, val, retract, _ = updagg_fuzz(oracle.make_updagg_op, seed=8)
+    n = len(key)
+    mid = n // 2
+    cfg = lambda: cabi.make_updagg_config(
+        [(cabi.COUNT, -1), (cabi.COUNT_DISTINCT, 0), (cabi.SUM, 0)],
+        n_keys=1, n_value_cols=1)
+
+    a = gpu.make_updagg_op(cfg())
+    a.process_batch([key[:mid], val[:mid], retract[:mid]])
+    e0 = rows_of(a.flush())
+    d0 = a.checkpoint_drain(0)
+    d1 = a.checkpoint_drain(1)
+    a.close()
+    assert len(d0[0]) > 0
+
+    b = gpu.make_updagg_op(cfg())
+    b.restore(0, d0)
+    b.restore(1, d1)
+    b.process_batch([key[mid:], val[mid:], retract[mid:]])
+    e1 = rows_of(b.flush())
+    b.close()
+
+    c = oracle.make_updagg_op(cfg())
+    c.process_batch([key, val, retract])
+    ec = rows_of(c.flush())
+    c.close()
+    assert merge_debezium(e0 + e1) == merge_debezium(ec)
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_min_max_append_only():
+    from arroyo_amd import gpu
+    op = gpu.make_updagg_op(cabi.make_updagg_config(
+        [(cabi.MIN, 0), (cabi.MAX, 0)], n_keys=1, n_value_cols=1))
+    k = np.array([1], dtype=np.int64)
+    v = np.array([5], dtype=np.int64)
+    op.process_batch([k, v, np.array([0], dtype=np.int64)])
+    assert rows_of(op.flush()) == [(1, 5, 5, 0)]
+    op.process_batch([k, v, np.array([1], dtype=np.int64)])
+    with pytest.raises(RuntimeError, match="retraction"):
+        op.flush()
+    op.close()
