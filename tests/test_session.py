@@ -283,3 +283,66 @@ def test_session_gpu_key_minus_one():
             (-1, 1, t0 + 10 * NS, t0 + 12 * NS, t0 + 12 * NS - 1),
             (3, 1, t0 + NS, t0 + 3 * NS, t0 + 3 * NS - 1)]
     assert got == sorted(want)
+
+
+def zipf_keys(rng, n, key_space=10_000_000, s=1.0):
+    """Bounded Zipf(s) over `key_space` ranks (BASELINE.json config 5)."""
+    ranks = np.arange(1, key_space + 1, dtype=np.float64)
+    p = 1.0 / ranks ** s
+    p /= p.sum()
+    return rng.choice(key_space, size=n, p=p).astype(np.int64)
+
+
+@pytest.mark.gpu
+def test_session_config5_zipf_checkpoint_under_load():
+    """BASELINE.json config 5 shape at single-GPU scale: session windows over
+    a Zipf stream drawn from a 10M-key space, with a checkpoint drained and
+    restored mid-stream while data keeps flowing; results must equal the
+    oracle's uninterrupted run."""
+    from arroyo_amd import gpu
+
+    rng = np.random.default_rng(55)
+    n = 1_000_000
+    gap = 5 * NS
+    t0 = 1_600_000_000 * NS
+    key = zipf_keys(rng, n)
+    # ~64K rows per second of event time
+    ts = t0 + (np.arange(n, dtype=np.int64) * NS) // 65536
+    batch = 65536
+
+    def cfg():
+        return cabi.make_session_config(
+            gap, [(cabi.COUNT, -1)], n_keys=1, n_value_cols=0,
+            log2_capacity=21, max_sessions=8, log2_batch_capacity=17,
+            log2_out_cap=22)
+
+    # GPU run with a mid-stream checkpoint: drain into a fresh op
+    op = gpu.make_session_op(cfg())
+    got = []
+    wm = 0
+    for b in range(0, n, batch):
+        cols = [key[b:b + batch], ts[b:b + batch]]
+        op.process_batch(cols)
+        wm = int(ts[min(b + batch, n) - 1]) - NS
+        got += rows_of(op.handle_watermark(wm))
+        if b == (n // batch // 2) * batch:
+            drained = op.checkpoint_drain()
+            op.close()
+            op = gpu.make_session_op(cfg())
+            op.restore(drained)
+    got += rows_of(op.handle_watermark(U64MAX))
+    op.close()
+
+    # oracle, uninterrupted
+    o = oracle.make_session_op(cfg())
+    want = []
+    for b in range(0, n, batch):
+        o.process_batch([key[b:b + batch], ts[b:b + batch]])
+        want += rows_of(o.handle_watermark(
+            int(ts[min(b + batch, n) - 1]) - NS))
+    want += rows_of(o.handle_watermark(U64MAX))
+    o.close()
+
+    assert len(got) == len(want)
+    assert sorted(got) == sorted(want)
+    assert len(want) > 100_000  # cold Zipf keys produce many 1-row sessions
