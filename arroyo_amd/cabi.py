@@ -664,3 +664,116 @@ class UpdAggOp:
             self.close()
         except Exception:
             pass
+
+
+class AmdWindowFnConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_cols", ctypes.c_int32),
+        ("part_col", ctypes.c_int32),
+        ("n_order", ctypes.c_int32),
+        ("order_col", ctypes.c_int32 * 2),
+        ("order_desc", ctypes.c_int32 * 2),
+        ("limit", ctypes.c_int64),
+        ("log2_rows_cap", ctypes.c_uint32),
+        ("instants", ctypes.c_uint32),
+        ("log2_out_cap", ctypes.c_uint32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_windowfn_config(n_cols, part_col, order, limit=0, log2_rows_cap=15,
+                         instants=128, log2_out_cap=20, device=0,
+                         emit_to_host=True):
+    """order: list of (col, desc) pairs, max 2."""
+    cfg = AmdWindowFnConfig()
+    cfg.n_cols = n_cols
+    cfg.part_col = part_col
+    cfg.n_order = len(order)
+    for i, (col, desc) in enumerate(order):
+        cfg.order_col[i] = col
+        cfg.order_desc[i] = 1 if desc else 0
+    cfg.limit = limit
+    cfg.log2_rows_cap = log2_rows_cap
+    cfg.instants = instants
+    cfg.log2_out_cap = log2_out_cap
+    cfg.device = device
+    cfg.emit_to_host = 1 if emit_to_host else 0
+    return cfg
+
+
+class WindowFnOp:
+    """One ROW_NUMBER window-function operator behind the C ABI, mirroring
+    WindowFunctionOperator's ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/window_fn.rs)."""
+
+    def __init__(self, lib, prefix, cfg):
+        p = prefix + "windowfn_"
+        g = lambda n: getattr(lib, p + n)
+        self._fn = {}
+        self._fn["create"] = g("create")
+        self._fn["create"].restype = ctypes.c_void_p
+        self._fn["create"].argtypes = [ctypes.POINTER(AmdWindowFnConfig)]
+        self._fn["process_batch"] = g("process_batch")
+        self._fn["process_batch"].restype = ctypes.c_int
+        self._fn["process_batch"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64]
+        self._fn["handle_watermark"] = g("handle_watermark")
+        self._fn["handle_watermark"].restype = ctypes.c_int
+        self._fn["handle_watermark"].argtypes = [
+            ctypes.c_void_p, ctypes.c_uint64, ctypes.POINTER(AmdOutBatch)]
+        self._fn["checkpoint_drain"] = g("checkpoint_drain")
+        self._fn["checkpoint_drain"].restype = ctypes.c_int
+        self._fn["checkpoint_drain"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(AmdOutBatch)]
+        self._fn["free_out"] = getattr(lib, prefix + "free_out")
+        self._fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+        self._fn["destroy"] = g("destroy")
+        self._fn["destroy"].argtypes = [ctypes.c_void_p]
+        self._fn["last_error"] = g("last_error")
+        self._fn["last_error"].restype = ctypes.c_char_p
+        self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{p}create failed")
+
+    def _check(self, rc):
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def process_batch(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        self._check(self._fn["process_batch"](
+            self._h, arr, len(keep), len(keep[0]) if keep else 0))
+
+    def handle_watermark(self, wm):
+        out = AmdOutBatch()
+        self._check(self._fn["handle_watermark"](self._h, wm,
+                                                 ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def checkpoint_drain(self):
+        out = AmdOutBatch()
+        self._check(self._fn["checkpoint_drain"](self._h, ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore(self, cols):
+        if cols and len(cols[0]):
+            self.process_batch(cols)
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

@@ -246,6 +246,29 @@ int arroyo_amd_updagg_restore(void *h, int32_t which,
 void arroyo_amd_updagg_destroy(void *h);
 const char *arroyo_amd_updagg_last_error(void *h);
 
+/* ---- SQL window function (ROW_NUMBER per instant) ---------------------
+ * Replaces WindowFunctionOperator
+ * (crates/arroyo-worker/src/arrow/window_fn.rs) behind the same
+ * ArrowOperator surface:
+ *   windowfn_create           <-> WindowFunctionConstructor (:180-273)
+ *   windowfn_process_batch    <-> process_batch (:275-300): buffers rows
+ *                                 per exact instant; late rows silently
+ *                                 filtered (filter_and_split_batches)
+ *   windowfn_handle_watermark <-> handle_watermark (:220-246): fires
+ *                                 instants < wm in timestamp order; out
+ *                                 columns [input cols..., row_number]
+ *   windowfn_checkpoint_drain <-> handle_checkpoint (:302-324): buffered
+ *                                 rows; restore = process_batch
+ */
+void *arroyo_amd_windowfn_create(const AmdWindowFnConfig *cfg);
+int arroyo_amd_windowfn_process_batch(void *h, const int64_t *const *cols,
+                                      int32_t n_cols, int64_t n_rows);
+int arroyo_amd_windowfn_handle_watermark(void *h, uint64_t watermark_nanos,
+                                         AmdOutBatch *out);
+int arroyo_amd_windowfn_checkpoint_drain(void *h, AmdOutBatch *out);
+void arroyo_amd_windowfn_destroy(void *h);
+const char *arroyo_amd_windowfn_last_error(void *h);
+
 #ifdef __cplusplus
 }
 #endif

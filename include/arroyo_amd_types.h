@@ -165,6 +165,31 @@ typedef struct {
     int32_t  emit_to_host;
 } AmdUpdatingConfig;
 
+/* SQL window-function (ROW_NUMBER) configuration.  Mirrors
+ * api::WindowFunctionOperator as decoded by WindowFunctionConstructor
+ * (crates/arroyo-worker/src/arrow/window_fn.rs:180-273): a
+ * BoundedWindowAggExec run per exact `_timestamp` instant.  Semantics
+ * (filter_and_split_batches :52-93, handle_watermark :220-246): rows are
+ * buffered per instant (late rows ts < watermark silently filtered); when
+ * the watermark passes an instant it fires in timestamp order, computing
+ * ROW_NUMBER() OVER (PARTITION BY part_col ORDER BY order cols) per row;
+ * rows with row_number > limit are dropped (the reference plans the
+ * downstream filter separately; fusing it here saves materialising the
+ * full ranking).  Output columns: input columns + trailing row_number. */
+typedef struct {
+    int32_t  n_cols;          /* input columns incl. trailing _timestamp */
+    int32_t  part_col;        /* partition column index; -1 = whole instant */
+    int32_t  n_order;         /* 1 or 2 ORDER BY columns */
+    int32_t  order_col[2];
+    int32_t  order_desc[2];
+    int64_t  limit;           /* keep row_number <= limit; 0 = keep all */
+    uint32_t log2_rows_cap;   /* per-instant row capacity (GPU) */
+    uint32_t instants;        /* live-instant slots (GPU, power of two) */
+    uint32_t log2_out_cap;
+    int32_t  device;
+    int32_t  emit_to_host;
+} AmdWindowFnConfig;
+
 /* Output batch, allocated by the callee; free with *_free_out.
  * Column order: [key (if n_keys)], agg outputs (one column per agg),
  * window_start, window_end, _timestamp.  All columns are 8-byte elements;

@@ -52,3 +52,8 @@ def make_expjoin_op(cfg):
 def make_updagg_op(cfg):
     from arroyo_amd.cabi import UpdAggOp
     return UpdAggOp(lib(), "oracle_", cfg)
+
+
+def make_windowfn_op(cfg):
+    from arroyo_amd.cabi import WindowFnOp
+    return WindowFnOp(lib(), "oracle_", cfg)

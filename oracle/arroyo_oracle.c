@@ -1771,3 +1771,222 @@ ORACLE_API void oracle_updagg_destroy(void *h) {
     free(o->out);
     free(o);
 }
+
+/* ================================================================== */
+/* SQL window-function (ROW_NUMBER per instant) oracle.
+ *
+ * Restated from crates/arroyo-worker/src/arrow/window_fn.rs:
+ *   - process_batch :275-300 buffers rows per exact `_timestamp` instant,
+ *     silently filtering rows with ts < watermark
+ *     (filter_and_split_batches :52-93, filter_by_time);
+ *   - handle_watermark :220-246 fires every instant < watermark in
+ *     timestamp order through the BoundedWindowAggExec: here ROW_NUMBER()
+ *     OVER (PARTITION BY part_col ORDER BY order cols), ties resolved by
+ *     input order (DataFusion's stable sort);
+ *   - rows with row_number > limit are dropped (the reference's downstream
+ *     filter; limit 0 keeps all).
+ */
+
+typedef struct {
+    uint64_t instant;
+    int64_t cap, n;
+    int64_t **cols;           /* [n_cols-1][cap] (ts implied) */
+} WfInstant;
+
+typedef struct {
+    AmdWindowFnConfig cfg;
+    int n_inst, cap_inst;
+    WfInstant *inst;          /* sorted by instant */
+    int has_wm; uint64_t wm;
+    int out_cols;
+    int64_t out_rows, out_cap;
+    int64_t **out;
+    char err[256];
+} WfOp;
+
+static WfInstant *wf_get(WfOp *o, uint64_t t) {
+    int lo = 0, hi = o->n_inst;
+    while (lo < hi) { int m = (lo + hi) / 2; if (o->inst[m].instant < t) lo = m + 1; else hi = m; }
+    if (lo < o->n_inst && o->inst[lo].instant == t) return &o->inst[lo];
+    if (o->n_inst == o->cap_inst) {
+        o->cap_inst = o->cap_inst ? o->cap_inst * 2 : 16;
+        o->inst = realloc(o->inst, (size_t)o->cap_inst * sizeof(WfInstant));
+    }
+    memmove(o->inst + lo + 1, o->inst + lo,
+            (size_t)(o->n_inst - lo) * sizeof(WfInstant));
+    WfInstant *in = &o->inst[lo];
+    memset(in, 0, sizeof *in);
+    in->instant = t;
+    in->cols = calloc((size_t)o->cfg.n_cols - 1, sizeof(int64_t *));
+    o->n_inst++;
+    return in;
+}
+
+ORACLE_API void *oracle_windowfn_create(const AmdWindowFnConfig *cfg) {
+    if (!cfg || cfg->n_cols < 2 || cfg->n_cols > 12 || cfg->n_order < 1 ||
+        cfg->n_order > 2 || cfg->part_col >= cfg->n_cols - 1)
+        return NULL;
+    WfOp *o = calloc(1, sizeof(WfOp));
+    o->cfg = *cfg;
+    o->out_cols = cfg->n_cols + 1;
+    o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
+    return o;
+}
+
+ORACLE_API const char *oracle_windowfn_last_error(void *h) {
+    return h ? ((WfOp *)h)->err : "null handle / invalid config";
+}
+
+ORACLE_API int oracle_windowfn_process_batch(void *h,
+                                             const int64_t *const *cols,
+                                             int32_t n_cols,
+                                             int64_t n_rows) {
+    WfOp *o = h;
+    if (n_cols != o->cfg.n_cols) {
+        snprintf(o->err, sizeof o->err, "expected %d cols, got %d",
+                 o->cfg.n_cols, n_cols);
+        return 1;
+    }
+    const int64_t *ts = cols[n_cols - 1];
+    for (int64_t r = 0; r < n_rows; r++) {
+        /* late rows silently filtered (filter_by_time) */
+        if (o->has_wm && (uint64_t)ts[r] < o->wm) continue;
+        WfInstant *in = wf_get(o, (uint64_t)ts[r]);
+        if (in->n == in->cap) {
+            in->cap = in->cap ? in->cap * 2 : 64;
+            for (int c = 0; c < n_cols - 1; c++)
+                in->cols[c] = realloc(in->cols[c], (size_t)in->cap * 8);
+        }
+        for (int c = 0; c < n_cols - 1; c++) in->cols[c][in->n] = cols[c][r];
+        in->n++;
+    }
+    return 0;
+}
+
+static const WfOp *g_wf_sort_op;
+static const WfInstant *g_wf_sort_in;
+
+static int wf_cmp(const void *a, const void *b) {
+    int64_t i = *(const int64_t *)a, j = *(const int64_t *)b;
+    const AmdWindowFnConfig *c = &g_wf_sort_op->cfg;
+    const WfInstant *in = g_wf_sort_in;
+    if (c->part_col >= 0) {
+        int64_t pa = in->cols[c->part_col][i], pb = in->cols[c->part_col][j];
+        if (pa != pb) return pa < pb ? -1 : 1;
+    }
+    for (int k = 0; k < c->n_order; k++) {
+        int64_t va = in->cols[c->order_col[k]][i];
+        int64_t vb = in->cols[c->order_col[k]][j];
+        if (va != vb) {
+            int lt = va < vb ? -1 : 1;
+            return c->order_desc[k] ? -lt : lt;
+        }
+    }
+    return i < j ? -1 : i > j ? 1 : 0;  /* stable: input order */
+}
+
+static void wf_out_reserve(WfOp *o, int64_t add) {
+    if (o->out_rows + add <= o->out_cap) return;
+    int64_t ncap = o->out_cap ? o->out_cap : 1024;
+    while (ncap < o->out_rows + add) ncap *= 2;
+    for (int i = 0; i < o->out_cols; i++)
+        o->out[i] = realloc(o->out[i], (size_t)ncap * 8);
+    o->out_cap = ncap;
+}
+
+static void wf_fire(WfOp *o, WfInstant *in) {
+    const AmdWindowFnConfig *c = &o->cfg;
+    int64_t *idx = malloc((size_t)in->n * 8);
+    for (int64_t i = 0; i < in->n; i++) idx[i] = i;
+    g_wf_sort_op = o;
+    g_wf_sort_in = in;
+    qsort(idx, (size_t)in->n, 8, wf_cmp);
+    int64_t rn = 0;
+    for (int64_t i = 0; i < in->n; i++) {
+        if (i == 0 ||
+            (c->part_col >= 0 &&
+             in->cols[c->part_col][idx[i]] !=
+                 in->cols[c->part_col][idx[i - 1]]))
+            rn = 0;
+        rn++;
+        if (c->limit && rn > c->limit) continue;
+        wf_out_reserve(o, 1);
+        int64_t r = o->out_rows++;
+        for (int cc = 0; cc < c->n_cols - 1; cc++)
+            o->out[cc][r] = in->cols[cc][idx[i]];
+        o->out[c->n_cols - 1][r] = (int64_t)in->instant;
+        o->out[c->n_cols][r] = rn;
+    }
+    free(idx);
+}
+
+ORACLE_API int oracle_windowfn_handle_watermark(void *h, uint64_t wm,
+                                                AmdOutBatch *out) {
+    WfOp *o = h;
+    o->has_wm = 1;
+    o->wm = wm;
+    int fired = 0;
+    while (fired < o->n_inst && o->inst[fired].instant < wm) {
+        wf_fire(o, &o->inst[fired]);
+        for (int c = 0; c < o->cfg.n_cols - 1; c++)
+            free(o->inst[fired].cols[c]);
+        free(o->inst[fired].cols);
+        fired++;
+    }
+    if (fired) {
+        memmove(o->inst, o->inst + fired,
+                (size_t)(o->n_inst - fired) * sizeof(WfInstant));
+        o->n_inst -= fired;
+    }
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = o->out_rows;
+        out->n_cols = o->out_cols;
+        out->cols = calloc((size_t)o->out_cols, sizeof(void *));
+        out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
+            if (o->out_rows)
+                memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
+        }
+        o->out_rows = 0;
+    }
+    return 0;
+}
+
+ORACLE_API int oracle_windowfn_checkpoint_drain(void *h, AmdOutBatch *out) {
+    WfOp *o = h;
+    int ncols = o->cfg.n_cols;
+    int64_t total = 0;
+    for (int i = 0; i < o->n_inst; i++) total += o->inst[i].n;
+    memset(out, 0, sizeof *out);
+    out->n_rows = total;
+    out->n_cols = ncols;
+    out->cols = calloc((size_t)ncols, sizeof(void *));
+    out->is_f64 = calloc((size_t)ncols, sizeof(int32_t));
+    for (int i = 0; i < ncols; i++)
+        out->cols[i] = malloc((size_t)(total ? total : 1) * 8);
+    int64_t r = 0;
+    for (int i = 0; i < o->n_inst; i++) {
+        WfInstant *in = &o->inst[i];
+        for (int64_t j = 0; j < in->n; j++, r++) {
+            for (int c = 0; c < ncols - 1; c++)
+                ((int64_t *)out->cols[c])[r] = in->cols[c][j];
+            ((int64_t *)out->cols[ncols - 1])[r] = (int64_t)in->instant;
+        }
+    }
+    return 0;
+}
+
+ORACLE_API void oracle_windowfn_destroy(void *h) {
+    WfOp *o = h;
+    if (!o) return;
+    for (int i = 0; i < o->n_inst; i++) {
+        for (int c = 0; c < o->cfg.n_cols - 1; c++) free(o->inst[i].cols[c]);
+        free(o->inst[i].cols);
+    }
+    free(o->inst);
+    for (int i = 0; i < o->out_cols; i++) free(o->out[i]);
+    free(o->out);
+    free(o);
+}
