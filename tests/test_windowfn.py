@@ -33,7 +33,7 @@ def run_most_active_driver(make_window_op, make_windowfn_op):
 
     win = make_window_op(cabi.make_config(
         width_ns=HOUR, slide_ns=60 * NS, n_keys=1, n_value_cols=0,
-        aggs=[(cabi.COUNT, -1)], log2_capacity=14))
+        aggs=[(cabi.COUNT, -1)], log2_capacity=14, ring_panes=256))
     # window output columns: [driver, count, ws, we, _ts]
     wf = make_windowfn_op(cabi.make_windowfn_config(
         n_cols=5, part_col=2, order=[(1, True), (0, True)], limit=1,
