@@ -657,6 +657,204 @@ __global__ void k_accum(unsigned long long *acc,
     *acc += *n;
 }
 
+/* ------------------------------------------------------------------ */
+/* K4+K5 fused, hash-aligned: each workgroup owns a disjoint range of HOME
+ * slots (hash(key) & mask in [a, a+RANGE)).  It scans that range (+ the
+ * MAX_PROBES displacement overscan) of every source pane, dedupes and sums
+ * the owned keys in an LDS table, and emits final rows directly -- no
+ * global merge table, no state atomics, ONE output-cursor atomic per
+ * workgroup.  Used when n_aggs <= MF_MAX_AGGS (LDS budget) and not
+ * draining raw states; the legacy merge+compact path covers the rest. */
+#define MF_RANGE 256u
+#define MF_SLOTS 2048
+#define MF_MAX_AGGS 2
+#define ERR_MF_OVERFLOW 6
+
+struct MergeFusedArgs {
+    DeviceRing ring;
+    AggSpec agg;
+    int32_t n_src;
+    uint32_t src[64];
+    uint64_t win_start, win_end;
+    int32_t n_keys;
+    int64_t *out[16];
+    unsigned long long *n_out;
+};
+
+__global__ void __launch_bounds__(256)
+k_merge_fused(MergeFusedArgs M) {
+    __shared__ int64_t lkey[MF_SLOTS];
+    extern __shared__ uint64_t lst[];    /* [MF_SLOTS][n_aggs][2] */
+    __shared__ unsigned long long blk_base;
+    __shared__ unsigned int blk_cnt;
+    const int na = M.agg.n_aggs;
+    const uint32_t C = M.ring.C, mask = C - 1;
+    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        for (int w = 0; w < na * 2; w++) lst[(size_t)i * na * 2 + w] = 0;
+    }
+    if (threadIdx.x == 0) blk_cnt = 0;
+    __syncthreads();
+    uint32_t a = blockIdx.x * MF_RANGE;          /* gridDim.x == C/MF_RANGE */
+    uint32_t span = MF_RANGE + MAX_PROBES;       /* displacement overscan */
+    for (int p = 0; p < M.n_src; p++) {
+        const int64_t *keys = M.ring.keys + (size_t)M.src[p] * C;
+        const uint64_t *st = M.ring.state + (size_t)M.src[p] * C * na * 2;
+        for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
+            uint32_t idx = (a + t) & mask;
+            int64_t key = keys[idx];
+            if (key == EMPTY_KEY) continue;
+            uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
+            if (rel >= MF_RANGE) continue;       /* another WG owns it */
+            /* LDS upsert */
+            uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u) &
+                         (MF_SLOTS - 1);
+            int64_t slot = -1;
+            for (int pr = 0; pr < MF_SLOTS && slot < 0; pr++) {
+                uint32_t sidx = (h + pr) & (MF_SLOTS - 1);
+                int64_t k = lkey[sidx];
+                if (k == key) {
+                    slot = sidx;
+                } else if (k == EMPTY_KEY) {
+                    int64_t old = (int64_t)atomicCAS(
+                        (unsigned long long *)&lkey[sidx],
+                        (unsigned long long)EMPTY_KEY,
+                        (unsigned long long)key);
+                    if (old == EMPTY_KEY || old == key) slot = sidx;
+                }
+            }
+            if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; continue; }
+            uint64_t *d = lst + (size_t)slot * na * 2;
+            for (int ag = 0; ag < na; ag++) {
+                switch (M.agg.op[ag]) {
+                case AMD_AGG_COUNT:
+                case AMD_AGG_SUM:
+                    atomicAdd((unsigned long long *)&d[2 * ag],
+                              (unsigned long long)st[(size_t)idx * na * 2 +
+                                                     2 * ag]);
+                    break;
+                case AMD_AGG_MIN:
+                case AMD_AGG_MAX:
+                    atomicMax((unsigned long long *)&d[2 * ag],
+                              (unsigned long long)st[(size_t)idx * na * 2 +
+                                                     2 * ag]);
+                    break;
+                case AMD_AGG_AVG:
+                    atomicAdd((unsigned long long *)&d[2 * ag],
+                              (unsigned long long)st[(size_t)idx * na * 2 +
+                                                     2 * ag]);
+                    atomicAdd((double *)&d[2 * ag + 1],
+                              *(const double *)&st[(size_t)idx * na * 2 +
+                                                   2 * ag + 1]);
+                    break;
+                }
+            }
+        }
+    }
+    __syncthreads();
+    /* pass 1: count occupied LDS slots; one global cursor add per WG */
+    unsigned int mine = 0;
+    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x)
+        mine += (lkey[i] != EMPTY_KEY);
+    for (int off = 32; off; off >>= 1)
+        mine += (unsigned)__shfl_down((int)mine, off, 64);
+    if ((threadIdx.x & 63) == 0 && mine) atomicAdd(&blk_cnt, mine);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        blk_base = blk_cnt ? atomicAdd(M.n_out,
+                                       (unsigned long long)blk_cnt)
+                           : 0;
+        blk_cnt = 0;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x) {
+        int64_t key = lkey[i];
+        if (key == EMPTY_KEY) continue;
+        int64_t r = (int64_t)(blk_base + atomicAdd(&blk_cnt, 1u));
+        const uint64_t *d = lst + (size_t)i * na * 2;
+        int col = 0;
+        if (M.n_keys) M.out[col++][r] = key;
+        for (int ag = 0; ag < na; ag++) {
+            uint64_t w0 = d[2 * ag];
+            switch (M.agg.op[ag]) {
+            case AMD_AGG_COUNT:
+            case AMD_AGG_SUM: M.out[col++][r] = (int64_t)w0; break;
+            case AMD_AGG_MIN: M.out[col++][r] = dec_min(w0); break;
+            case AMD_AGG_MAX: M.out[col++][r] = dec_max(w0); break;
+            case AMD_AGG_AVG: {
+                double v = w0 ? (*(const double *)(d + 2 * ag + 1)) /
+                                    (double)(int64_t)w0
+                              : 0.0;
+                int64_t b;
+                memcpy(&b, &v, 8);
+                M.out[col++][r] = b;
+                break;
+            }
+            }
+        }
+        M.out[col++][r] = (int64_t)M.win_start;
+        M.out[col++][r] = (int64_t)M.win_end;
+        M.out[col][r] = (int64_t)(M.win_end - 1);
+    }
+    /* special (key == sentinel) entries: one extra row */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        uint64_t spec[2 * AMD_MAX_AGGS] = {};
+        int any = 0;
+        for (int p = 0; p < M.n_src; p++) {
+            uint32_t pp = M.src[p];
+            if (!M.ring.spec_used[pp]) continue;
+            any = 1;
+            const uint64_t *st =
+                M.ring.spec_state + (size_t)pp * na * 2;
+            for (int ag = 0; ag < na; ag++) {
+                switch (M.agg.op[ag]) {
+                case AMD_AGG_COUNT:
+                case AMD_AGG_SUM: spec[2 * ag] += st[2 * ag]; break;
+                case AMD_AGG_MIN:
+                case AMD_AGG_MAX:
+                    if (st[2 * ag] > spec[2 * ag]) spec[2 * ag] = st[2 * ag];
+                    break;
+                case AMD_AGG_AVG: {
+                    spec[2 * ag] += st[2 * ag];
+                    double x, y;
+                    memcpy(&x, &spec[2 * ag + 1], 8);
+                    memcpy(&y, &st[2 * ag + 1], 8);
+                    x += y;
+                    memcpy(&spec[2 * ag + 1], &x, 8);
+                    break;
+                }
+                }
+            }
+        }
+        if (any) {
+            int64_t r = (int64_t)atomicAdd(M.n_out, 1ULL);
+            int col = 0;
+            if (M.n_keys) M.out[col++][r] = EMPTY_KEY;
+            for (int ag = 0; ag < na; ag++) {
+                uint64_t w0 = spec[2 * ag];
+                switch (M.agg.op[ag]) {
+                case AMD_AGG_COUNT:
+                case AMD_AGG_SUM: M.out[col++][r] = (int64_t)w0; break;
+                case AMD_AGG_MIN: M.out[col++][r] = dec_min(w0); break;
+                case AMD_AGG_MAX: M.out[col++][r] = dec_max(w0); break;
+                case AMD_AGG_AVG: {
+                    double v = w0 ? (*(const double *)(spec + 2 * ag + 1)) /
+                                        (double)(int64_t)w0
+                                  : 0.0;
+                    int64_t b;
+                    memcpy(&b, &v, 8);
+                    M.out[col++][r] = b;
+                    break;
+                }
+                }
+            }
+            M.out[col++][r] = (int64_t)M.win_start;
+            M.out[col++][r] = (int64_t)M.win_end;
+            M.out[col][r] = (int64_t)(M.win_end - 1);
+        }
+    }
+}
+
 /* restore checkpointed partial states: insert raw state rows into a pane. */
 struct RestoreArgs {
     const int64_t *key_col;           /* null when unkeyed */
@@ -1160,6 +1358,16 @@ static int check_device_error(GpuOp *o) {
                  o->cfg.log2_capacity);
         return 1;
     }
+    if (e == ERR_MF_OVERFLOW) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "fused-merge LDS table overflow (pathological key "
+                 "clustering); raise log2_capacity");
+        return 1;
+    }
+    if (e) {
+        snprintf(o->err_msg, sizeof o->err_msg, "device error %d", e);
+        return 1;
+    }
     return 0;
 }
 
@@ -1183,6 +1391,52 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                        uint64_t ws, uint64_t we, int raw_states,
                        uint64_t bin_ts) {
     size_t na = o->agg.n_aggs;
+    /* fused hash-aligned path: one kernel, no merge table, no state
+     * atomics (see k_merge_fused); legacy merge+compact otherwise */
+    if (!raw_states && na <= MF_MAX_AGGS && o->ring.C >= MF_RANGE &&
+        src.size() <= 64) {
+        HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+        if (!src.empty()) {
+            MergeFusedArgs M = {};
+            M.ring = o->ring;
+            M.agg = o->agg;
+            M.n_src = (int)src.size();
+            for (size_t i = 0; i < src.size(); i++) M.src[i] = src[i];
+            M.win_start = ws;
+            M.win_end = we;
+            M.n_keys = o->cfg.n_keys;
+            for (int i = 0; i < o->n_out_alloc && i < 16; i++)
+                M.out[i] = o->d_out[i];
+            M.n_out = o->d_n_out;
+            size_t shmem = (size_t)MF_SLOTS * na * 16;
+            hipLaunchKernelGGL(k_merge_fused,
+                               dim3(o->ring.C / MF_RANGE), dim3(256), shmem,
+                               o->stream, M);
+            HIP_CHECK(o, hipGetLastError());
+        }
+        if (!o->cfg.emit_to_host) {
+            hipLaunchKernelGGL(k_accum, dim3(1), dim3(1), 0, o->stream,
+                               o->d_emitted, o->d_n_out);
+            HIP_CHECK(o, hipGetLastError());
+            return 0;
+        }
+        unsigned long long n = 0;
+        HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8,
+                                    hipMemcpyDeviceToHost, o->stream));
+        HIP_CHECK(o, hipStreamSynchronize(o->stream));
+        if (n == 0) return 0;
+        if ((size_t)o->out_cols > o->host_out.size())
+            o->host_out.resize(o->out_cols);
+        for (int i = 0; i < o->out_cols; i++) {
+            size_t old = o->host_out[i].size();
+            o->host_out[i].resize(old + n);
+            HIP_CHECK(o, hipMemcpyAsync(o->host_out[i].data() + old,
+                                        o->d_out[i], n * 8,
+                                        hipMemcpyDeviceToHost, o->stream));
+        }
+        HIP_CHECK(o, hipStreamSynchronize(o->stream));
+        return 0;
+    }
     HIP_CHECK(o, hipMemsetAsync(o->m_keys, 0xFF, (size_t)o->CM * 8, o->stream));
     HIP_CHECK(o, hipMemsetAsync(o->m_zero_blob, 0, o->m_zero_bytes,
                                 o->stream));
