@@ -681,15 +681,16 @@ struct MergeFusedArgs {
     unsigned long long *n_out;
 };
 
+template <int SLOTS>
 __global__ void __launch_bounds__(256)
 k_merge_fused(MergeFusedArgs M) {
-    __shared__ int64_t lkey[MF_SLOTS];
-    extern __shared__ uint64_t lst[];    /* [MF_SLOTS][n_aggs][2] */
+    __shared__ int64_t lkey[SLOTS];
+    extern __shared__ uint64_t lst[];    /* [SLOTS][n_aggs][2] */
     __shared__ unsigned long long blk_base;
     __shared__ unsigned int blk_cnt;
     const int na = M.agg.n_aggs;
     const uint32_t C = M.ring.C, mask = C - 1;
-    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x) {
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
         lkey[i] = EMPTY_KEY;
         for (int w = 0; w < na * 2; w++) lst[(size_t)i * na * 2 + w] = 0;
     }
@@ -708,10 +709,10 @@ k_merge_fused(MergeFusedArgs M) {
             if (rel >= MF_RANGE) continue;       /* another WG owns it */
             /* LDS upsert */
             uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u) &
-                         (MF_SLOTS - 1);
+                         (SLOTS - 1);
             int64_t slot = -1;
-            for (int pr = 0; pr < MF_SLOTS && slot < 0; pr++) {
-                uint32_t sidx = (h + pr) & (MF_SLOTS - 1);
+            for (int pr = 0; pr < SLOTS && slot < 0; pr++) {
+                uint32_t sidx = (h + pr) & (SLOTS - 1);
                 int64_t k = lkey[sidx];
                 if (k == key) {
                     slot = sidx;
@@ -754,7 +755,7 @@ k_merge_fused(MergeFusedArgs M) {
     __syncthreads();
     /* pass 1: count occupied LDS slots; one global cursor add per WG */
     unsigned int mine = 0;
-    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x)
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x)
         mine += (lkey[i] != EMPTY_KEY);
     for (int off = 32; off; off >>= 1)
         mine += (unsigned)__shfl_down((int)mine, off, 64);
@@ -767,7 +768,7 @@ k_merge_fused(MergeFusedArgs M) {
         blk_cnt = 0;
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < MF_SLOTS; i += blockDim.x) {
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
         int64_t key = lkey[i];
         if (key == EMPTY_KEY) continue;
         int64_t r = (int64_t)(blk_base + atomicAdd(&blk_cnt, 1u));
@@ -854,6 +855,9 @@ k_merge_fused(MergeFusedArgs M) {
         }
     }
 }
+
+template __global__ void k_merge_fused<1024>(MergeFusedArgs);
+template __global__ void k_merge_fused<2048>(MergeFusedArgs);
 
 /* restore checkpointed partial states: insert raw state rows into a pane. */
 struct RestoreArgs {
@@ -1408,10 +1412,21 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             for (int i = 0; i < o->n_out_alloc && i < 16; i++)
                 M.out[i] = o->d_out[i];
             M.n_out = o->d_n_out;
-            size_t shmem = (size_t)MF_SLOTS * na * 16;
-            hipLaunchKernelGGL(k_merge_fused,
-                               dim3(o->ring.C / MF_RANGE), dim3(256), shmem,
-                               o->stream, M);
+            /* 1024 slots won the sweep (3->4 workgroups per CU); typical
+             * occupancy is ~320 owned keys per 256-slot home range and
+             * overflow is a loud error, not silent corruption */
+            int mfs = 1024;
+            if (const char *ev = getenv("ARROYO_AMD_MF_SLOTS"))
+                mfs = atoi(ev);
+            size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) * na * 16;
+            if (mfs >= 2048)
+                hipLaunchKernelGGL(k_merge_fused<2048>,
+                                   dim3(o->ring.C / MF_RANGE), dim3(256),
+                                   shmem, o->stream, M);
+            else
+                hipLaunchKernelGGL(k_merge_fused<1024>,
+                                   dim3(o->ring.C / MF_RANGE), dim3(256),
+                                   shmem, o->stream, M);
             HIP_CHECK(o, hipGetLastError());
         }
         if (!o->cfg.emit_to_host) {
