@@ -1822,6 +1822,15 @@ API void arroyo_amd_destroy(void *h) {
     delete o;
 }
 
+/* synchronize the operator's internal stream: callers that reuse receive
+ * buffers (the N>1 bench overwrites its RCCL exchange buffers each step)
+ * must fence the consuming kernels first */
+API int arroyo_amd_sync(void *h) {
+    GpuOp *o = (GpuOp *)h;
+    HIP_CHECK(o, hipStreamSynchronize(o->stream));
+    return 0;
+}
+
 /* perf introspection for bench.py's roofline leg */
 API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
                         int64_t *launches, int64_t *emitted_device_rows) {

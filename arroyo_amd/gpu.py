@@ -60,6 +60,8 @@ def lib():
         _lib.arroyo_amd_stream_gbps.restype = ctypes.c_double
         _lib.arroyo_amd_stream_gbps.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
+        _lib.arroyo_amd_sync.restype = ctypes.c_int
+        _lib.arroyo_amd_sync.argtypes = [ctypes.c_void_p]
         _lib.arroyo_amd_partition.restype = ctypes.c_int
         _lib.arroyo_amd_partition.argtypes = [ctypes.c_void_p] * 3 + [
             ctypes.c_int64, ctypes.c_uint32] + [ctypes.c_void_p] * 3 + [

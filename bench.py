@@ -162,6 +162,9 @@ def run_gpu(args):
         ts_off = cycle * span
         bk, bt = views[b]
         if world > 1:
+            # fence the operator's stream before overwriting the exchange
+            # buffers it may still be consuming from the previous step
+            wm_lib.arroyo_amd_sync(op._h)
             counts = gpu.partition_device(
                 bk.data_ptr(), 0, bt.data_ptr(), BATCH_ROWS, world,
                 sk.data_ptr(), 0, st.data_ptr())
