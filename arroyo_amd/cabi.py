@@ -777,3 +777,110 @@ class WindowFnOp:
             self.close()
         except Exception:
             pass
+
+
+MOP_CONST, MOP_ADD, MOP_SUB, MOP_MUL, MOP_DIV, MOP_MOD = range(6)
+MOP_EQ, MOP_NE, MOP_LT, MOP_LE, MOP_GT, MOP_GE = range(6, 12)
+MOP_AND, MOP_OR, MOP_NOT, MOP_I2F, MOP_F2I = range(12, 17)
+MOP_FADD, MOP_FSUB, MOP_FMUL, MOP_FDIV = range(17, 21)
+
+
+class AmdMapInstr(ctypes.Structure):
+    _fields_ = [("op", ctypes.c_int32), ("a", ctypes.c_int32),
+                ("b", ctypes.c_int32), ("dst", ctypes.c_int32),
+                ("imm", ctypes.c_int64)]
+
+
+class AmdMapConfig(ctypes.Structure):
+    _fields_ = [
+        ("n_in_cols", ctypes.c_int32),
+        ("n_prog", ctypes.c_int32),
+        ("prog", AmdMapInstr * 64),
+        ("n_out", ctypes.c_int32),
+        ("out_reg", ctypes.c_int32 * 16),
+        ("out_is_f64", ctypes.c_int32 * 16),
+        ("filter_reg", ctypes.c_int32),
+        ("device", ctypes.c_int32),
+        ("emit_to_host", ctypes.c_int32),
+    ]
+
+
+def make_map_config(n_in_cols, prog, out_reg, out_is_f64=None,
+                    filter_reg=-1, device=0):
+    """prog: list of (op, a, b, dst[, imm]) tuples.  imm may be a float for
+    MOP_CONST feeding f64 ops (stored as the f64 bit pattern)."""
+    import struct
+    cfg = AmdMapConfig()
+    cfg.n_in_cols = n_in_cols
+    cfg.n_prog = len(prog)
+    for i, ins in enumerate(prog):
+        op, a, b, dst = ins[:4]
+        imm = ins[4] if len(ins) > 4 else 0
+        if isinstance(imm, float):
+            imm = struct.unpack("<q", struct.pack("<d", imm))[0]
+        cfg.prog[i].op = op
+        cfg.prog[i].a = a
+        cfg.prog[i].b = b
+        cfg.prog[i].dst = dst
+        cfg.prog[i].imm = imm
+    cfg.n_out = len(out_reg)
+    for i, r in enumerate(out_reg):
+        cfg.out_reg[i] = r
+        cfg.out_is_f64[i] = 1 if (out_is_f64 and out_is_f64[i]) else 0
+    cfg.filter_reg = filter_reg
+    cfg.device = device
+    cfg.emit_to_host = 1
+    return cfg
+
+
+class MapOp:
+    """One stateless map/filter/projection operator behind the C ABI,
+    mirroring the expression operators' ArrowOperator surface
+    (crates/arroyo-worker/src/arrow/mod.rs:48-243)."""
+
+    def __init__(self, lib, prefix, cfg):
+        p = prefix + "map_"
+        g = lambda n: getattr(lib, p + n)
+        self._fn = {}
+        self._fn["create"] = g("create")
+        self._fn["create"].restype = ctypes.c_void_p
+        self._fn["create"].argtypes = [ctypes.POINTER(AmdMapConfig)]
+        self._fn["process_batch"] = g("process_batch")
+        self._fn["process_batch"].restype = ctypes.c_int
+        self._fn["process_batch"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64, ctypes.POINTER(AmdOutBatch)]
+        self._fn["free_out"] = getattr(lib, prefix + "free_out")
+        self._fn["free_out"].argtypes = [ctypes.POINTER(AmdOutBatch)]
+        self._fn["destroy"] = g("destroy")
+        self._fn["destroy"].argtypes = [ctypes.c_void_p]
+        self._fn["last_error"] = g("last_error")
+        self._fn["last_error"].restype = ctypes.c_char_p
+        self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self.cfg = cfg
+        self._h = self._fn["create"](ctypes.byref(cfg))
+        if not self._h:
+            raise RuntimeError(f"{p}create failed")
+
+    def process_batch(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        out = AmdOutBatch()
+        rc = self._fn["process_batch"](self._h, arr, len(keep),
+                                       len(keep[0]) if keep else 0,
+                                       ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        res = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return res
+
+    def close(self):
+        if self._h:
+            self._fn["destroy"](self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

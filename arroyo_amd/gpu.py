@@ -16,12 +16,12 @@ _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libarroyo_amd.so")
 _SRCS = [os.path.join(_DIR, "csrc", f)
          for f in ("arroyo_amd.hip", "session.hip", "expjoin.hip",
-                   "updagg.hip", "windowfn.hip")]
+                   "updagg.hip", "windowfn.hip", "mapop.hip")]
 
 _lib = None
 
 HIPCC_CMD = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-             "-shared", "-fvisibility=hidden", "-parallel-jobs=5",
+             "-shared", "-fvisibility=hidden", "-parallel-jobs=6",
              "-o", _SO] + _SRCS
 
 
@@ -154,3 +154,8 @@ def make_updagg_op(cfg):
 def make_windowfn_op(cfg):
     from arroyo_amd.cabi import WindowFnOp
     return WindowFnOp(lib(), "arroyo_amd_", cfg)
+
+
+def make_map_op(cfg):
+    from arroyo_amd.cabi import MapOp
+    return MapOp(lib(), "arroyo_amd_", cfg)

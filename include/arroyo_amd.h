@@ -269,6 +269,23 @@ int arroyo_amd_windowfn_checkpoint_drain(void *h, AmdOutBatch *out);
 void arroyo_amd_windowfn_destroy(void *h);
 const char *arroyo_amd_windowfn_last_error(void *h);
 
+/* ---- stateless map / filter / projection ------------------------------
+ * Replaces ValueExecutionOperator / ProjectionOperator /
+ * KeyExecutionOperator (crates/arroyo-worker/src/arrow/mod.rs:48-243):
+ *   map_create        <-> the operators' constructors (:70-97, :127-178,
+ *                         :213-243): the serialized expression plan becomes
+ *                         the register program in AmdMapConfig
+ *   map_process_batch <-> process_batch (:56-68, :112-125, :196-211):
+ *                         stateless, emits the projected (and filtered,
+ *                         order-preserving) rows immediately
+ * No state: nothing to checkpoint (the reference's tables() are empty). */
+void *arroyo_amd_map_create(const AmdMapConfig *cfg);
+int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
+                                 int32_t n_cols, int64_t n_rows,
+                                 AmdOutBatch *out);
+void arroyo_amd_map_destroy(void *h);
+const char *arroyo_amd_map_last_error(void *h);
+
 #ifdef __cplusplus
 }
 #endif

@@ -190,6 +190,51 @@ typedef struct {
     int32_t  emit_to_host;
 } AmdWindowFnConfig;
 
+/* Stateless map/filter/projection configuration.  Replaces the reference's
+ * expression operators (crates/arroyo-worker/src/arrow/mod.rs):
+ * ValueExecutionOperator :48, ProjectionOperator :99, KeyExecutionOperator
+ * :180 -- each runs a serialized DataFusion expression plan over every
+ * batch.  Here the expression plan's information content is a small
+ * register program evaluated per row: registers r0..r(n_in_cols-1) are
+ * preloaded with the input columns, instructions write higher registers,
+ * `out_reg` names the emitted columns (KeyExecutionOperator = key exprs
+ * first), and `filter_reg` (when >= 0) keeps rows whose register is
+ * nonzero, preserving row order like the reference's filter kernels.
+ * Integer division/modulo by zero is a loud error (DataFusion errors too);
+ * f64 ops follow IEEE.  i64 values and f64 bit patterns share the
+ * register file; AMD_MOP_I2F / F2I convert. */
+enum AmdMapOp {
+    AMD_MOP_CONST = 0,  /* dst = imm */
+    AMD_MOP_ADD, AMD_MOP_SUB, AMD_MOP_MUL, AMD_MOP_DIV, AMD_MOP_MOD,
+    AMD_MOP_EQ, AMD_MOP_NE, AMD_MOP_LT, AMD_MOP_LE, AMD_MOP_GT, AMD_MOP_GE,
+    AMD_MOP_AND, AMD_MOP_OR, AMD_MOP_NOT,
+    AMD_MOP_I2F, AMD_MOP_F2I,    /* value <-> f64 bit pattern (truncating) */
+    AMD_MOP_FADD, AMD_MOP_FSUB, AMD_MOP_FMUL, AMD_MOP_FDIV
+};
+
+#define AMD_MAP_MAX_PROG 64
+#define AMD_MAP_MAX_REGS 32
+#define AMD_MAP_MAX_OUT  16
+
+typedef struct {
+    int32_t op;     /* AmdMapOp */
+    int32_t a, b;   /* source registers (b unused for unary/CONST) */
+    int32_t dst;
+    int64_t imm;    /* CONST value (f64 ops: bit pattern) */
+} AmdMapInstr;
+
+typedef struct {
+    int32_t n_in_cols;          /* input columns incl. trailing _timestamp */
+    int32_t n_prog;
+    AmdMapInstr prog[AMD_MAP_MAX_PROG];
+    int32_t n_out;
+    int32_t out_reg[AMD_MAP_MAX_OUT];
+    int32_t out_is_f64[AMD_MAP_MAX_OUT];
+    int32_t filter_reg;         /* -1 = no filter */
+    int32_t device;
+    int32_t emit_to_host;
+} AmdMapConfig;
+
 /* Output batch, allocated by the callee; free with *_free_out.
  * Column order: [key (if n_keys)], agg outputs (one column per agg),
  * window_start, window_end, _timestamp.  All columns are 8-byte elements;

@@ -57,3 +57,8 @@ def make_updagg_op(cfg):
 def make_windowfn_op(cfg):
     from arroyo_amd.cabi import WindowFnOp
     return WindowFnOp(lib(), "oracle_", cfg)
+
+
+def make_map_op(cfg):
+    from arroyo_amd.cabi import MapOp
+    return MapOp(lib(), "oracle_", cfg)

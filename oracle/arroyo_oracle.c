@@ -1990,3 +1990,116 @@ ORACLE_API void oracle_windowfn_destroy(void *h) {
     free(o->out);
     free(o);
 }
+
+/* ================================================================== */
+/* Stateless map/filter/projection oracle.
+ *
+ * Restated from crates/arroyo-worker/src/arrow/mod.rs: the expression
+ * operators evaluate a DataFusion plan per batch
+ * (StatelessPhysicalExecutor::process_batch :245-290) and emit the result
+ * immediately; filters preserve row order.  The register program carries
+ * the plan's information content for the supported expression set. */
+
+typedef struct {
+    AmdMapConfig cfg;
+    char err[256];
+} MOp;
+
+ORACLE_API void *oracle_map_create(const AmdMapConfig *cfg) {
+    if (!cfg || cfg->n_in_cols < 1 || cfg->n_in_cols > AMD_MAP_MAX_REGS ||
+        cfg->n_prog < 0 || cfg->n_prog > AMD_MAP_MAX_PROG ||
+        cfg->n_out < 1 || cfg->n_out > AMD_MAP_MAX_OUT)
+        return NULL;
+    for (int i = 0; i < cfg->n_prog; i++)
+        if (cfg->prog[i].dst < 0 || cfg->prog[i].dst >= AMD_MAP_MAX_REGS ||
+            cfg->prog[i].a < 0 || cfg->prog[i].a >= AMD_MAP_MAX_REGS ||
+            cfg->prog[i].b < 0 || cfg->prog[i].b >= AMD_MAP_MAX_REGS)
+            return NULL;
+    MOp *o = calloc(1, sizeof(MOp));
+    o->cfg = *cfg;
+    return o;
+}
+
+ORACLE_API const char *oracle_map_last_error(void *h) {
+    return h ? ((MOp *)h)->err : "null handle / invalid config";
+}
+
+static int map_eval_row(const AmdMapConfig *c, const int64_t *const *cols,
+                        int64_t r, int64_t *regs, char *err) {
+    for (int i = 0; i < c->n_in_cols; i++) regs[i] = cols[i][r];
+    for (int i = 0; i < c->n_prog; i++) {
+        const AmdMapInstr *in = &c->prog[i];
+        int64_t a = regs[in->a], b = regs[in->b], v = 0;
+        double fa = bits_to_d(a), fb = bits_to_d(b);
+        switch (in->op) {
+        case AMD_MOP_CONST: v = in->imm; break;
+        case AMD_MOP_ADD: v = a + b; break;
+        case AMD_MOP_SUB: v = a - b; break;
+        case AMD_MOP_MUL: v = a * b; break;
+        case AMD_MOP_DIV:
+            if (b == 0) {
+                snprintf(err, 256, "division by zero");
+                return 1;
+            }
+            v = a / b;
+            break;
+        case AMD_MOP_MOD:
+            if (b == 0) {
+                snprintf(err, 256, "division by zero");
+                return 1;
+            }
+            v = a % b;
+            break;
+        case AMD_MOP_EQ: v = a == b; break;
+        case AMD_MOP_NE: v = a != b; break;
+        case AMD_MOP_LT: v = a < b; break;
+        case AMD_MOP_LE: v = a <= b; break;
+        case AMD_MOP_GT: v = a > b; break;
+        case AMD_MOP_GE: v = a >= b; break;
+        case AMD_MOP_AND: v = (a != 0) && (b != 0); break;
+        case AMD_MOP_OR: v = (a != 0) || (b != 0); break;
+        case AMD_MOP_NOT: v = a == 0; break;
+        case AMD_MOP_I2F: v = d_to_bits((double)a); break;
+        case AMD_MOP_F2I: v = (int64_t)fa; break;
+        case AMD_MOP_FADD: v = d_to_bits(fa + fb); break;
+        case AMD_MOP_FSUB: v = d_to_bits(fa - fb); break;
+        case AMD_MOP_FMUL: v = d_to_bits(fa * fb); break;
+        case AMD_MOP_FDIV: v = d_to_bits(fa / fb); break;
+        }
+        regs[in->dst] = v;
+    }
+    return 0;
+}
+
+ORACLE_API int oracle_map_process_batch(void *h, const int64_t *const *cols,
+                                        int32_t n_cols, int64_t n_rows,
+                                        AmdOutBatch *out) {
+    MOp *o = h;
+    const AmdMapConfig *c = &o->cfg;
+    if (n_cols != c->n_in_cols) {
+        snprintf(o->err, sizeof o->err, "expected %d cols, got %d",
+                 c->n_in_cols, n_cols);
+        return 1;
+    }
+    memset(out, 0, sizeof *out);
+    out->n_cols = c->n_out;
+    out->cols = calloc((size_t)c->n_out, sizeof(void *));
+    out->is_f64 = calloc((size_t)c->n_out, sizeof(int32_t));
+    for (int i = 0; i < c->n_out; i++) {
+        out->cols[i] = malloc((size_t)(n_rows ? n_rows : 1) * 8);
+        out->is_f64[i] = c->out_is_f64[i];
+    }
+    int64_t w = 0;
+    int64_t regs[AMD_MAP_MAX_REGS] = {0};
+    for (int64_t r = 0; r < n_rows; r++) {
+        if (map_eval_row(c, cols, r, regs, o->err)) return 1;
+        if (c->filter_reg >= 0 && regs[c->filter_reg] == 0) continue;
+        for (int i = 0; i < c->n_out; i++)
+            ((int64_t *)out->cols[i])[w] = regs[c->out_reg[i]];
+        w++;
+    }
+    out->n_rows = w;
+    return 0;
+}
+
+ORACLE_API void oracle_map_destroy(void *h) { free(h); }
