@@ -602,6 +602,10 @@ class UpdAggOp:
         self._fn["flush"].restype = ctypes.c_int
         self._fn["flush"].argtypes = [ctypes.c_void_p,
                                       ctypes.POINTER(AmdOutBatch)]
+        self._fn["expire"] = g("expire")
+        self._fn["expire"].restype = ctypes.c_int
+        self._fn["expire"].argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                       ctypes.POINTER(AmdOutBatch)]
         self._fn["checkpoint_drain"] = g("checkpoint_drain")
         self._fn["checkpoint_drain"].restype = ctypes.c_int
         self._fn["checkpoint_drain"].argtypes = [
@@ -636,6 +640,14 @@ class UpdAggOp:
     def flush(self):
         out = AmdOutBatch()
         self._check(self._fn["flush"](self._h, ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def expire(self, idle_flushes):
+        out = AmdOutBatch()
+        self._check(self._fn["expire"](self._h, idle_flushes,
+                                       ctypes.byref(out)))
         cols = _out_to_numpy(out)
         self._fn["free_out"](ctypes.byref(out))
         return cols

@@ -358,6 +358,53 @@ k_updagg_flush(FlushArgs F) {
     }
 }
 
+/* TTL eviction (the reference's UpdatingCache::time_out, epoch-based here;
+ * see oracle).  Evicted keys emit retract(last) and reset in place: the
+ * key slot stays claimed (open addressing), its distinct-value chain nodes
+ * are orphaned in the append-only pool (bounded by log2_nodes; a loud
+ * pool-full error, not silent corruption). */
+struct ExpireArgs {
+    UStore store;
+    AggSpec agg;
+    uint32_t cur_epoch;
+    uint32_t idle;
+    int32_t n_keys;
+    int64_t *out[AMD_MAX_AGGS + 2];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_updagg_expire(ExpireArgs E) {
+    const UStore &S = E.store;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int na = E.agg.n_aggs;
+    for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         slot <= (int64_t)S.C; slot += stride) {
+        if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
+        uint32_t te = S.epoch[slot];
+        if (te == 0 || E.cur_epoch - te <= E.idle) continue;
+        if (!S.emitted[slot] && S.rows[slot] == 0) continue;
+        if (S.emitted[slot]) {
+            int64_t r = (int64_t)atomicAdd(E.n_out, 1ULL);
+            if (r >= E.out_cap) { *E.err = UERR_OUT_CAP; continue; }
+            int col = 0;
+            if (E.n_keys)
+                E.out[col++][r] =
+                    slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+            for (int a = 0; a < na; a++)
+                E.out[col++][r] = S.last[(size_t)slot * na + a];
+            E.out[col][r] = 1;
+        }
+        S.emitted[slot] = 0;
+        S.rows[slot] = 0;
+        for (int w = 0; w < 2 * na; w++)
+            S.st[(size_t)slot * na * 2 + w] = 0;
+        for (int a = 0; a < na; a++) S.head[(size_t)slot * 8 + a] = -1;
+    }
+}
+
 /* checkpoint drain which=0: scalar rows
  * [key?, rows, st words..., emitted, last...]; which=1: multiset rows
  * [key?, agg_index, value, net_count] */
@@ -668,6 +715,48 @@ API int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out) {
         for (int a = 0; a < o->cfg.n_aggs; a++)
             if (o->cfg.agg_ops[a] == AMD_AGG_AVG)
                 out->is_f64[o->cfg.n_keys + a] = 1;
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+            if (n)
+                UHIP(o, hipMemcpyAsync(out->cols[i], o->d_out[i],
+                                       (size_t)n * 8, hipMemcpyDeviceToHost,
+                                       o->stream));
+        }
+        UHIP(o, hipStreamSynchronize(o->stream));
+    }
+    return 0;
+}
+
+API int arroyo_amd_updagg_expire(void *h, int64_t idle_flushes,
+                                 AmdOutBatch *out) {
+    GpuUpdAgg *o = (GpuUpdAgg *)h;
+    if (ua_check_err(o)) return 1;
+    UHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    ExpireArgs E = {};
+    E.store = o->store;
+    E.agg = o->agg;
+    E.cur_epoch = o->cur_epoch;
+    E.idle = (uint32_t)idle_flushes;
+    E.n_keys = o->cfg.n_keys;
+    for (int i = 0; i < o->out_cols; i++) E.out[i] = o->d_out[i];
+    E.n_out = o->d_n_out;
+    E.out_cap = o->out_cap;
+    E.err = o->d_err;
+    hipLaunchKernelGGL(k_updagg_expire,
+                       dim3(ua_grid((int64_t)o->store.C + 1)), dim3(256), 0,
+                       o->stream, E);
+    UHIP(o, hipGetLastError());
+    unsigned long long n = 0;
+    UHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                           o->stream));
+    UHIP(o, hipStreamSynchronize(o->stream));
+    if (ua_check_err(o)) return 1;
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = (int64_t)n;
+        out->n_cols = o->out_cols;
+        out->cols = (void **)calloc(o->out_cols, sizeof(void *));
+        out->is_f64 = (int32_t *)calloc(o->out_cols, sizeof(int32_t));
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
             if (n)

@@ -238,6 +238,10 @@ void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg);
 int arroyo_amd_updagg_process_batch(void *h, const int64_t *const *cols,
                                     int32_t n_cols, int64_t n_rows);
 int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out);
+/* TTL eviction of keys idle for more than `idle_flushes` flush epochs
+ * (the reference's UpdatingCache::time_out, wall-clock there); emits the
+ * retract rows */
+int arroyo_amd_updagg_expire(void *h, int64_t idle_flushes, AmdOutBatch *out);
 int arroyo_amd_updagg_checkpoint_drain(void *h, int32_t which,
                                        AmdOutBatch *out);
 int arroyo_amd_updagg_restore(void *h, int32_t which,

@@ -1433,6 +1433,7 @@ typedef struct {
     int emitted;
     int64_t last[AMD_MAX_AGGS];   /* last emitted values (AVG: f64 bits) */
     int changed;
+    int64_t touched_epoch;        /* last flush epoch with activity */
 } UEntry;
 
 typedef struct {
@@ -1444,6 +1445,7 @@ typedef struct {
     int out_cols;
     int64_t out_rows, out_cap;
     int64_t **out;
+    int64_t epoch;
     char err[256];
 } UOp;
 
@@ -1537,6 +1539,7 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
         int64_t d = retr[r] ? -1 : 1;
         e->rows += d;
         e->changed = 1;
+        e->touched_epoch = o->epoch;
         for (int a = 0; a < c->n_aggs; a++) {
             int64_t v = c->agg_col[a] >= 0
                             ? cols[c->n_keys + c->agg_col[a]][r] : 0;
@@ -1638,6 +1641,7 @@ ORACLE_API int oracle_updagg_flush(void *h, AmdOutBatch *out) {
             e->emitted = 1;
         }
     }
+    o->epoch++;
     if (out) {
         memset(out, 0, sizeof *out);
         out->n_rows = o->out_rows;
@@ -1647,6 +1651,51 @@ ORACLE_API int oracle_updagg_flush(void *h, AmdOutBatch *out) {
         for (int a = 0; a < c->n_aggs; a++)
             if (c->agg_ops[a] == AMD_AGG_AVG)
                 out->is_f64[c->n_keys + a] = 1;
+        for (int i = 0; i < o->out_cols; i++) {
+            out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
+            if (o->out_rows)
+                memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
+        }
+        o->out_rows = 0;
+    }
+    return 0;
+}
+
+/* TTL eviction: the reference's UpdatingCache::time_out (updating_cache.rs)
+ * retracts and removes keys idle longer than the ttl during flush, measured
+ * in wall-clock time; here the cutoff is flush epochs (idle_flushes) so
+ * tests are deterministic.  Evicted keys emit retract(last emitted) and
+ * their state resets (a re-arriving key starts fresh, as in the
+ * reference's remove()). */
+ORACLE_API int oracle_updagg_expire(void *h, int64_t idle_flushes,
+                                    AmdOutBatch *out) {
+    UOp *o = h;
+    const AmdUpdatingConfig *c = &o->cfg;
+    for (int64_t s = 0; s < o->map_cap; s++) {
+        if (!o->map_used[s]) continue;
+        UEntry *e = &o->ent[s];
+        if (o->epoch - e->touched_epoch <= idle_flushes) continue;
+        if (e->emitted) uemit(o, o->map_keys[s], e->last, 1);
+        for (int a = 0; a < c->n_aggs; a++) {
+            free(e->ms[a].vals);
+            free(e->ms[a].cnt);
+        }
+        int64_t te = e->touched_epoch;
+        memset(e, 0, sizeof *e);
+        e->touched_epoch = te;  /* stays idle unless touched again */
+        for (int a = 0; a < c->n_aggs; a++) {
+            switch (c->agg_ops[a]) {
+            case AMD_AGG_MIN: e->st[2 * a] = INT64_MAX; break;
+            case AMD_AGG_MAX: e->st[2 * a] = INT64_MIN; break;
+            }
+        }
+    }
+    if (out) {
+        memset(out, 0, sizeof *out);
+        out->n_rows = o->out_rows;
+        out->n_cols = o->out_cols;
+        out->cols = calloc((size_t)o->out_cols, sizeof(void *));
+        out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
             if (o->out_rows)

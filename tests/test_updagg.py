@@ -311,3 +311,38 @@ def test_updagg_gpu_min_max_append_only():
     with pytest.raises(RuntimeError, match="retraction"):
         op.flush()
     op.close()
+
+
+def ttl_scenario(make_op):
+    """key 1 active every flush; key 2 goes idle and is TTL-evicted (the
+    reference's UpdatingCache::time_out -> retract), then reappears fresh."""
+    op = make_op(cabi.make_updagg_config([(cabi.COUNT, -1)], n_keys=1))
+    z = np.array([0], dtype=np.int64)
+    both = np.array([1, 2], dtype=np.int64)
+    one = np.array([1], dtype=np.int64)
+    out = []
+    op.process_batch([both, np.zeros(2, dtype=np.int64)])
+    out += rows_of(op.flush())
+    for _ in range(3):
+        op.process_batch([one, z])
+        out += rows_of(op.flush())
+    out += rows_of(op.expire(2))   # key 2 idle for 3 > 2 flushes: retract
+    op.process_batch([np.array([2], dtype=np.int64), z])
+    out += rows_of(op.flush())     # reappears as a fresh count of 1
+    op.close()
+    return out
+
+
+def test_updagg_oracle_ttl_expire():
+    out = ttl_scenario(oracle.make_updagg_op)
+    assert merge_debezium(out) == {1: (4,), 2: (1,)}
+    # the eviction emitted exactly one retract for key 2's old value
+    assert (2, 1, 1) in out
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_ttl_expire():
+    from arroyo_amd import gpu
+    got = ttl_scenario(gpu.make_updagg_op)
+    want = ttl_scenario(oracle.make_updagg_op)
+    assert sorted(got) == sorted(want)
