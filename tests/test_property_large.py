@@ -102,3 +102,55 @@ def test_q5_device_resident_path_reconciles():
         m = (bins >= ws) & (bins < we)
         assert int(wcnt.sum()) == int(m.sum())
         assert len(wkey) == len(np.unique(key[m]))
+
+
+def test_q5_fused_watermark_submission_identical():
+    """Submitting two watermark periods of rows in one pass and then the two
+    watermarks (the bench's BENCH_WM_FUSE batching) must produce the same
+    windows as strict per-period submission: a post-watermark row can never
+    land in a pane that watermark's windows read."""
+    import torch
+
+    from arroyo_amd import gpu
+
+    n = 2_097_152
+    key, ts = nexmark.bids(n, events_per_sec=1_000_000, seed=77)
+    dev = torch.device("cuda", 0)
+    d_key = torch.from_numpy(key).to(dev)
+    d_ts = torch.from_numpy(ts).to(dev)
+    batch = 65536
+
+    def run(fuse):
+        op = gpu.make_op(cabi.make_config(
+            width_ns=WIDTH, slide_ns=SLIDE, n_keys=1, n_value_cols=0,
+            aggs=[(cabi.COUNT, -1)], log2_capacity=19, ring_panes=16,
+            emit_to_host=True))
+        outs = []
+        pending = []
+        wm_last = 0
+        for b in range(n // batch):
+            op.process_batches_device(
+                [d_key.data_ptr() + b * batch * 8,
+                 d_ts.data_ptr() + b * batch * 8], batch, 1,
+                contiguous=True)
+            mx = int(ts[(b + 1) * batch - 1])
+            if mx - wm_last > NS:
+                wm_last = mx
+                pending.append(mx - NS)
+            if len(pending) >= fuse:
+                for wm in pending:
+                    outs.append(op.handle_watermark(wm))
+                pending = []
+        for wm in pending:
+            outs.append(op.handle_watermark(wm))
+        outs.append(op.handle_watermark(U64MAX))
+        op.close()
+        rows = []
+        for cols in outs:
+            if cols is None or len(cols) == 0 or len(cols[0]) == 0:
+                continue
+            rows += [tuple(int(c[i]) for c in cols)
+                     for i in range(len(cols[0]))]
+        return sorted(rows)
+
+    assert run(1) == run(2) == run(4)
