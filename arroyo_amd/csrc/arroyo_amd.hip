@@ -1212,10 +1212,12 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
                (!A.key_col || ((uintptr_t)A.key_col & 15) == 0);
     int64_t units = vec ? (n_rows + 1) / 2 : n_rows;
     int64_t want = (units + 255) / 256;
-    /* 640-block cap won the grid sweep at ~1M-row fused launches (2.5
-     * blocks/CU; fewer blocks -> fewer same-key flush contenders on the
-     * global table, more -> better tail occupancy) */
-    int blocks = (int)(want > 640 ? 640 : (want < 1 ? 1 : want));
+    /* grid-sweep optima: 640 blocks at ~0.5M units (1M-row launches), 896
+     * at ~1M units (2M-row fused launches).  Fewer blocks -> fewer
+     * same-key flush contenders on the global table; more -> better tail
+     * occupancy at larger launches. */
+    int cap = units >= (640 << 10) ? 896 : 640;
+    int blocks = (int)(want > cap ? cap : (want < 1 ? 1 : want));
     if (o->force_blocks > 0) blocks = o->force_blocks;
     /* sample kernel time on a subset of launches via a reusable event pool */
     bool sample = o->use_events && (o->launches & 7) == 0;
