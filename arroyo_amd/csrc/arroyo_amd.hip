@@ -40,6 +40,7 @@
  * pinned against the reference's golden vectors) by tests/test_gpu_parity.py.
  */
 #include <hip/hip_runtime.h>
+#include <hipcub/hipcub.hpp>
 
 #include <cstdio>
 #include <cstdlib>
@@ -507,6 +508,157 @@ k_update_lds_vec(UpdateArgs A) {
 
 template __global__ void k_update_lds_vec<1024>(UpdateArgs);
 template __global__ void k_update_lds_vec<2048>(UpdateArgs);
+
+/* ------------------------------------------------------------------ */
+/* Radix-partitioned update path (ARROYO_AMD_RADIX=1, keyed
+ * n_value_cols==0 launches): instead of every block hammering the shared
+ * pane tables, rows are first partitioned by the top 8 bits of hash(key)
+ * (histogram -> exclusive scan -> scatter, all streaming), then ONE block
+ * per bucket aggregates its rows -- every key belongs to exactly one
+ * block, so the LDS table sees no cross-block duplication and the pane
+ * table sees one uncontended upsert per distinct (key, bin) per launch.
+ * Trades ~32 B/row of extra streaming traffic for the elimination of
+ * cross-block atomic contention.                                      */
+#define RDX_BUCKETS 256
+#define RDX_SLOTS 4096           /* LDS table of the per-bucket aggregator */
+
+__global__ void __launch_bounds__(256)
+k_radix_hist(const int64_t *key_col, int64_t n_rows,
+             unsigned int *hist /* [RDX_BUCKETS][gridDim] */) {
+    __shared__ unsigned int cnt[RDX_BUCKETS];
+    for (int i = threadIdx.x; i < RDX_BUCKETS; i += blockDim.x) cnt[i] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n_rows; i += stride)
+        atomicAdd(&cnt[hash64((uint64_t)key_col[i]) >> 56], 1u);
+    __syncthreads();
+    for (int i = threadIdx.x; i < RDX_BUCKETS; i += blockDim.x)
+        hist[(size_t)i * gridDim.x + blockIdx.x] = cnt[i];
+}
+
+__global__ void __launch_bounds__(256)
+k_radix_scatter(const int64_t *key_col, const int64_t *ts_col,
+                int64_t n_rows, uint64_t ts_offset,
+                const unsigned int *hist, int64_t *out_key,
+                int64_t *out_ts) {
+    __shared__ unsigned int cur[RDX_BUCKETS];
+    for (int i = threadIdx.x; i < RDX_BUCKETS; i += blockDim.x)
+        cur[i] = hist[(size_t)i * gridDim.x + blockIdx.x];
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n_rows; i += stride) {
+        int64_t k = key_col[i];
+        unsigned int pos = atomicAdd(&cur[hash64((uint64_t)k) >> 56], 1u);
+        out_key[pos] = k;
+        out_ts[pos] = ts_col[i] + (int64_t)ts_offset;
+    }
+}
+
+/* one block per bucket: rows [bstart, bend) of the scratch are this
+ * block's exclusive key set */
+struct RadixAggArgs {
+    const int64_t *key;
+    const int64_t *ts;
+    const unsigned int *bucket_base;  /* hist after scan: [b][0] = start */
+    int hist_blocks;
+    int64_t n_rows;
+    uint64_t slide, slide_inv;
+    uint64_t wm_bin;
+    int has_wm;
+    DeviceRing ring;
+    AggSpec agg;
+};
+
+__global__ void __launch_bounds__(256)
+k_radix_agg(RadixAggArgs A) {
+    __shared__ int64_t  lkey[RDX_SLOTS];
+    __shared__ uint32_t lpane[RDX_SLOTS];
+    __shared__ uint64_t lcnt[RDX_SLOTS];   /* COUNT-only state (n_aggs==1) */
+    for (int i = threadIdx.x; i < RDX_SLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        lpane[i] = PANE_UNSET;
+        lcnt[i] = 0;
+    }
+    __syncthreads();
+    int b = blockIdx.x;
+    int64_t lo = A.bucket_base[(size_t)b * A.hist_blocks];
+    int64_t hi = b + 1 < RDX_BUCKETS
+                     ? (int64_t)A.bucket_base[(size_t)(b + 1) * A.hist_blocks]
+                     : A.n_rows;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint64_t t = (uint64_t)A.ts[i];
+        uint64_t q = div_slide(t, A.slide, A.slide_inv);
+        uint64_t bin = q * A.slide;
+        if (A.has_wm && bin < A.wm_bin) continue;
+        if (bin < local_min) local_min = bin;
+        uint32_t p = (uint32_t)(q & (A.ring.R - 1));
+        if (bin != last_bin) {
+            claim_tag_wave(A.ring.tag, p, bin, EMPTY_TAG, A.ring.err);
+            last_bin = bin;
+        }
+        int64_t key = A.key[i];
+        if (key == EMPTY_KEY) {
+            atomicExch(&A.ring.spec_used[p], 1u);
+            atomicAdd((unsigned long long *)
+                          &A.ring.spec_state[(size_t)p * A.agg.n_aggs * 2],
+                      1ULL);
+            continue;
+        }
+        uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
+                     (RDX_SLOTS - 1);
+        bool done = false;
+        for (int pr = 0; pr < 8 && !done; pr++) {
+            uint32_t sl = (h + pr) & (RDX_SLOTS - 1);
+            int64_t k = lkey[sl];
+            bool claimed = false;
+            if (k == EMPTY_KEY) {
+                int64_t old = (int64_t)atomicCAS(
+                    (unsigned long long *)&lkey[sl],
+                    (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+                if (old == EMPTY_KEY) {
+                    lpane[sl] = p;
+                    k = key;
+                    claimed = true;
+                } else {
+                    k = old;
+                }
+            }
+            if (k == key && (claimed || lpane[sl] == p)) {
+                atomicAdd((unsigned long long *)&lcnt[sl], 1ULL);
+                done = true;
+            }
+        }
+        if (!done) {
+            /* LDS full for this probe window: straight to the pane table
+             * (uncontended -- this block owns the key) */
+            int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
+            int64_t sl = table_upsert(keys, A.ring.C, key, A.ring.err);
+            if (sl >= 0)
+                atomicAdd((unsigned long long *)
+                              &A.ring.state[((size_t)p * A.ring.C +
+                                             (size_t)sl) *
+                                            A.agg.n_aggs * 2],
+                          1ULL);
+        }
+    }
+    fold_min_bin(local_min, A.ring.min_bin);
+    __syncthreads();
+    for (int i = threadIdx.x; i < RDX_SLOTS; i += blockDim.x) {
+        int64_t key = lkey[i];
+        if (key == EMPTY_KEY) continue;
+        uint32_t p = lpane[i];
+        int64_t *keys = A.ring.keys + (size_t)p * A.ring.C;
+        int64_t sl = table_upsert(keys, A.ring.C, key, A.ring.err);
+        if (sl < 0) continue;
+        atomicAdd((unsigned long long *)
+                      &A.ring.state[((size_t)p * A.ring.C + (size_t)sl) *
+                                    A.agg.n_aggs * 2],
+                  (unsigned long long)lcnt[i]);
+    }
+}
 
 /* ------------------------------------------------------------------ */
 /* K4: merge source panes into the merge table.                        */
@@ -1020,6 +1172,12 @@ struct GpuOp {
     std::set<uint64_t> table_bins;        /* ExpiringTimeKeyView keys */
 
     int use_lds;
+    int use_radix;             /* ARROYO_AMD_RADIX=1: partitioned update */
+    int64_t *rdx_key, *rdx_ts; /* scatter scratch (lazily sized) */
+    unsigned int *rdx_hist;
+    void *rdx_tmp;
+    size_t rdx_tmp_bytes;
+    int64_t rdx_cap;
     int force_blocks;
     int kmode;
     int use_vec;
@@ -1099,6 +1257,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     o->state = 0;
     o->use_lds = 1;
     if (const char *e = getenv("ARROYO_AMD_LDS")) o->use_lds = atoi(e);
+    o->use_radix = 0;
+    if (const char *e = getenv("ARROYO_AMD_RADIX")) o->use_radix = atoi(e);
     o->force_blocks = 0;
     if (const char *e = getenv("ARROYO_AMD_BLOCKS")) o->force_blocks = atoi(e);
     o->kmode = 0;
@@ -1186,6 +1346,62 @@ API const char *arroyo_amd_last_error(void *h) {
     return h ? ((GpuOp *)h)->err_msg : g_err;
 }
 
+#define RDX_HIST_BLOCKS 640
+
+static int launch_update_radix(GpuOp *o, const int64_t *key_col,
+                               const int64_t *ts_col, int64_t n_rows,
+                               uint64_t ts_offset) {
+    if (n_rows > o->rdx_cap) {
+        hipFree(o->rdx_key);
+        hipFree(o->rdx_ts);
+        hipFree(o->rdx_hist);
+        hipFree(o->rdx_tmp);
+        o->rdx_cap = n_rows + (n_rows >> 2);
+        HIP_CHECK(o, hipMalloc((void **)&o->rdx_key,
+                               (size_t)o->rdx_cap * 8));
+        HIP_CHECK(o, hipMalloc((void **)&o->rdx_ts,
+                               (size_t)o->rdx_cap * 8));
+        HIP_CHECK(o, hipMalloc((void **)&o->rdx_hist,
+                               (size_t)RDX_BUCKETS * RDX_HIST_BLOCKS * 4));
+        o->rdx_tmp_bytes = 0;
+        hipcub::DeviceScan::ExclusiveSum(
+            nullptr, o->rdx_tmp_bytes, o->rdx_hist, o->rdx_hist,
+            RDX_BUCKETS * RDX_HIST_BLOCKS);
+        HIP_CHECK(o, hipMalloc(&o->rdx_tmp,
+                               o->rdx_tmp_bytes ? o->rdx_tmp_bytes : 1));
+    }
+    hipLaunchKernelGGL(k_radix_hist, dim3(RDX_HIST_BLOCKS), dim3(256), 0,
+                       o->stream, key_col, n_rows, o->rdx_hist);
+    HIP_CHECK(o, hipGetLastError());
+    size_t tmp = o->rdx_tmp_bytes;
+    hipcub::DeviceScan::ExclusiveSum(o->rdx_tmp, tmp, o->rdx_hist,
+                                     o->rdx_hist,
+                                     RDX_BUCKETS * RDX_HIST_BLOCKS,
+                                     o->stream);
+    hipLaunchKernelGGL(k_radix_scatter, dim3(RDX_HIST_BLOCKS), dim3(256), 0,
+                       o->stream, key_col, ts_col, n_rows, ts_offset,
+                       o->rdx_hist, o->rdx_key, o->rdx_ts);
+    HIP_CHECK(o, hipGetLastError());
+    RadixAggArgs A = {};
+    A.key = o->rdx_key;
+    A.ts = o->rdx_ts;
+    A.bucket_base = o->rdx_hist;
+    A.hist_blocks = RDX_HIST_BLOCKS;
+    A.n_rows = n_rows;
+    A.slide = o->slide;
+    A.slide_inv = o->slide == 1
+                      ? ~0ULL
+                      : (uint64_t)((((unsigned __int128)1) << 64) / o->slide);
+    A.wm_bin = o->has_wm ? o->wm - o->wm % o->slide : 0;
+    A.has_wm = o->has_wm;
+    A.ring = o->ring;
+    A.agg = o->agg;
+    hipLaunchKernelGGL(k_radix_agg, dim3(RDX_BUCKETS), dim3(256), 0,
+                       o->stream, A);
+    HIP_CHECK(o, hipGetLastError());
+    return 0;
+}
+
 /* launch the update kernel over device-resident columns */
 static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
                          uint64_t ts_offset) {
@@ -1238,6 +1454,20 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     }
     int slots = 1024;
     if (const char *e = getenv("ARROYO_AMD_LDS_SLOTS")) slots = atoi(e);
+    bool radix = o->use_radix && o->cfg.n_keys == 1 &&
+                 o->cfg.n_value_cols == 0 && o->agg.n_aggs == 1 &&
+                 o->agg.op[0] == AMD_AGG_COUNT && n_rows >= (1 << 18);
+    if (radix) {
+        if (launch_update_radix(o, dcols[0], dcols[1], n_rows, ts_offset))
+            return 1;
+        if (sample) {
+            hipEventRecord(o->ev_pool[ev].second, o->stream);
+            o->ev_inflight.push_back(ev);
+        }
+        o->update_rows += n_rows;
+        o->launches++;
+        return 0;
+    }
     size_t shmem = (size_t)slots * o->agg.n_aggs * 16;
     if (vec) {
         if (slots >= 2048)
@@ -1817,6 +2047,10 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->m_zero_blob);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
     hipFree(o->d_emitted);
+    hipFree(o->rdx_key);
+    hipFree(o->rdx_ts);
+    hipFree(o->rdx_hist);
+    hipFree(o->rdx_tmp);
     for (int i = 0; i < o->stg.ncols; i++) {
         hipHostFree(o->stg.buf[i]);
         hipFree(o->stg.dbuf[i]);
