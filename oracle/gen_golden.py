@@ -105,7 +105,8 @@ def main():
                  "tight_watermark", "most_active_driver_last_hour",
                  "windowed_inner_join", "session_window",
                  "global_session_window", "updating_inner_join",
-                 "debezium_agg", "filter_updating_aggregates"):
+                 "debezium_agg", "filter_updating_aggregates",
+                 "aggregates", "grouped_aggregates"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)

@@ -346,3 +346,64 @@ def test_updagg_gpu_ttl_expire():
     got = ttl_scenario(gpu.make_updagg_op)
     want = ttl_scenario(oracle.make_updagg_op)
     assert sorted(got) == sorted(want)
+
+
+def run_aggregates_golden(make_op, make_map_op=None):
+    """aggregates.sql / grouped_aggregates.sql: MIN/MAX/SUM/COUNT/AVG(counter)
+    over impulse, global and GROUP BY counter % 5 (the key expression runs
+    through the map operator, KeyExecutionOperator's role)."""
+    d = load_inputs()["impulse"]
+    counter = np.array(d["counter"], dtype=np.int64)
+    zeros = np.zeros(len(counter), dtype=np.int64)
+    aggs = [(cabi.MIN, 0), (cabi.MAX, 0), (cabi.SUM, 0), (cabi.COUNT, -1),
+            (cabi.AVG, 0)]
+
+    # global
+    op = make_op(cabi.make_updagg_config(aggs, n_keys=0, n_value_cols=1))
+    op.process_batch([counter, zeros])
+    out = op.flush()
+    op.close()
+    rows = [r for r in zip(*[c.tolist() for c in out]) if not r[-1]]
+    assert len(rows) == 1
+    mn, mx, sm, cnt, avg, _ = rows[0]
+    g = load_golden("aggregates")[0]["after"]
+    assert (mn, mx, sm, cnt) == (g["min"], g["max"], g["sum"], g["count"])
+    assert abs(avg - g["avg"]) < 1e-9
+
+    # grouped: key = counter % 5 computed by the map operator
+    if make_map_op is not None:
+        mp = make_map_op(cabi.make_map_config(
+            n_in_cols=2,
+            prog=[(cabi.MOP_CONST, 0, 0, 2, 5), (cabi.MOP_MOD, 0, 2, 3)],
+            out_reg=[3, 0, 1]))
+        keyed = mp.process_batch([counter, zeros])
+        mp.close()
+    else:
+        keyed = [counter % 5, counter, zeros]
+    op = make_op(cabi.make_updagg_config(aggs, n_keys=1, n_value_cols=1))
+    op.process_batch([c.astype(np.int64) for c in keyed])
+    out = op.flush()
+    op.close()
+    got = {}
+    for r in zip(*[c.tolist() for c in out]):
+        if not r[-1]:
+            got[int(r[0])] = r[1:-1]
+    want = {g["after"]["counter_mod"]: g["after"]
+            for g in load_golden("grouped_aggregates")}
+    assert set(got) == set(want)
+    for k, (mn, mx, sm, cnt, avg) in got.items():
+        w = want[k]
+        assert (mn, mx, sm, cnt) == (w["min"], w["max"], w["sum"],
+                                     w["count"])
+        assert abs(avg - w["avg"]) < 1e-9
+
+
+def test_aggregates_goldens_oracle():
+    import oracle as om
+    run_aggregates_golden(om.make_updagg_op, om.make_map_op)
+
+
+@pytest.mark.gpu
+def test_aggregates_goldens_gpu():
+    from arroyo_amd import gpu
+    run_aggregates_golden(gpu.make_updagg_op, gpu.make_map_op)
