@@ -74,28 +74,30 @@ def _rank_main(rank, world, port, result_q):
     dist.destroy_process_group()
 
 
-def test_two_rank_shuffle_matches_single():
+def _run_world(world, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29371
-    procs = [ctx.Process(target=_rank_main, args=(r, WORLD, port, q))
-             for r in range(WORLD)]
+    procs = [ctx.Process(target=_rank_main, args=(r, world, port, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(WORLD):
+    for _ in range(world):
         rank, rows = q.get(timeout=300)
         results[rank] = rows
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
 
-    # no key may appear on both ranks
-    keys0 = {r[0] for r in results[0]}
-    keys1 = {r[0] for r in results[1]}
-    assert not (keys0 & keys1)
+    # key-hash range partitioning: no key may appear on two ranks
+    for a in range(world):
+        for b in range(a + 1, world):
+            assert not ({r[0] for r in results[a]} &
+                        {r[0] for r in results[b]})
 
-    merged = results[0] | results[1]
+    merged = set()
+    for rows in results.values():
+        merged |= rows
 
     cols = nexmark.bids(100_000, events_per_sec=20_000)
     kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=0,
@@ -108,3 +110,13 @@ def test_two_rank_shuffle_matches_single():
         want.add(tuple(int(c[r]) for c in want_cols))
     op.close()
     assert merged == want
+
+
+def test_two_rank_shuffle_matches_single():
+    _run_world(2, 29371)
+
+
+def test_shuffle_window_parity_world4():
+    """Same sharded-vs-single parity at world_size=4 (the q7 config's 1->8
+    GPU scaling path; gloo here, RCCL on the GPU boxes)."""
+    _run_world(4, 29377)
