@@ -128,6 +128,33 @@ __device__ inline void atomic_fold(uint64_t *st, const AggSpec &a,
     }
 }
 
+/* atomically merge encoded partial words (LDS flush -> global table) */
+__device__ inline void state_merge_atomic(uint64_t *dst, const uint64_t *src,
+                                          const AggSpec &a) {
+    for (int i = 0; i < a.n_aggs; i++) {
+        switch (a.op[i]) {
+        case AMD_AGG_COUNT:
+        case AMD_AGG_SUM:
+            atomicAdd((unsigned long long *)&dst[2 * i],
+                      (unsigned long long)src[2 * i]);
+            break;
+        case AMD_AGG_MIN:
+        case AMD_AGG_MAX:
+            atomicMax((unsigned long long *)&dst[2 * i],
+                      (unsigned long long)src[2 * i]);
+            break;
+        case AMD_AGG_AVG: {
+            atomicAdd((unsigned long long *)&dst[2 * i],
+                      (unsigned long long)src[2 * i]);
+            double v;
+            memcpy(&v, &src[2 * i + 1], 8);
+            atomicAdd((double *)&dst[2 * i + 1], v);
+            break;
+        }
+        }
+    }
+}
+
 /* merge partial state words into a session's state (single thread) */
 __device__ inline void state_merge(uint64_t *dst, const uint64_t *src,
                                    const AggSpec &a) {
@@ -190,11 +217,25 @@ struct UpdateArgs {
     int *err;
 };
 
+/* LDS-staged: a per-workgroup table absorbs hot keys (Zipf streams send a
+ * large share of a batch's rows to one key, and serialized same-address
+ * HBM atomics were ~70 us of a 64K-row batch) before flushing distinct
+ * entries into the per-batch global table.  Same pattern as the window
+ * path's k_update_lds.  LDS entry = key + (min,max,aggs) encoded words. */
+#define SESS_LDS_SLOTS 512
+
 __global__ void __launch_bounds__(256)
 k_sess_update(UpdateArgs A) {
+    __shared__ int64_t lkey[SESS_LDS_SLOTS];
+    extern __shared__ uint64_t lst[];    /* [SLOTS][2 + n_aggs*2] */
+    int sw = 2 + 2 * A.agg.n_aggs;
+    for (int i = threadIdx.x; i < SESS_LDS_SLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        for (int w = 0; w < sw; w++) lst[(size_t)i * sw + w] = 0;
+    }
+    __syncthreads();
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     const int64_t *ts = A.cols[A.n_keys + A.n_vals];
-    int sw = 2 + 2 * A.agg.n_aggs;
     const int64_t *vcols[AMD_MAX_AGGS];
     for (int i = 0; i < A.agg.n_aggs; i++)
         vcols[i] = A.agg.col[i] >= 0 ? A.cols[A.n_keys + A.agg.col[i]]
@@ -204,16 +245,61 @@ k_sess_update(UpdateArgs A) {
         uint64_t t = (uint64_t)ts[r] + A.ts_offset;
         if (A.has_wm && t < A.wm) continue;  /* late: silently dropped */
         int64_t key = A.n_keys ? A.cols[0][r] : 0;
+        /* LDS first (hot keys collapse here); EMPTY_KEY sentinel collides
+         * with a real key of -1: that rare key goes straight to global */
+        bool done = false;
+        if (key != EMPTY_KEY) {
+            uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u) &
+                         (SESS_LDS_SLOTS - 1);
+            for (int pr = 0; pr < 4 && !done; pr++) {
+                uint32_t sl = (h + pr) & (SESS_LDS_SLOTS - 1);
+                int64_t k = lkey[sl];
+                if (k == EMPTY_KEY) {
+                    int64_t old = (int64_t)atomicCAS(
+                        (unsigned long long *)&lkey[sl],
+                        (unsigned long long)EMPTY_KEY,
+                        (unsigned long long)key);
+                    k = old == EMPTY_KEY ? key : old;
+                }
+                if (k == key) {
+                    uint64_t *st = lst + (size_t)sl * sw;
+                    atomicMax((unsigned long long *)&st[0],
+                              (unsigned long long)enc_min((int64_t)t));
+                    atomicMax((unsigned long long *)&st[1],
+                              (unsigned long long)enc_max((int64_t)t));
+                    atomic_fold(st + 2, A.agg, vcols, r);
+                    done = true;
+                }
+            }
+        }
+        if (!done) {
+            int64_t s = key_slot(A.bkeys, A.B, key, A.err, SERR_BATCH_FULL);
+            if (s < 0) continue;
+            uint64_t *st = A.bst + (size_t)s * sw;
+            atomicMax((unsigned long long *)&st[0],
+                      (unsigned long long)enc_min((int64_t)t));
+            atomicMax((unsigned long long *)&st[1],
+                      (unsigned long long)enc_max((int64_t)t));
+            atomic_fold(st + 2, A.agg, vcols, r);
+        }
+    }
+    __syncthreads();
+    /* flush LDS entries into the per-batch global table */
+    for (int i = threadIdx.x; i < SESS_LDS_SLOTS; i += blockDim.x) {
+        int64_t key = lkey[i];
+        if (key == EMPTY_KEY) continue;
         int64_t s = key_slot(A.bkeys, A.B, key, A.err, SERR_BATCH_FULL);
         if (s < 0) continue;
-        uint64_t *st = A.bst + (size_t)s * sw;
-        atomicMax((unsigned long long *)&st[0],
-                  (unsigned long long)enc_min((int64_t)t));
-        atomicMax((unsigned long long *)&st[1],
-                  (unsigned long long)enc_max((int64_t)t));
-        atomic_fold(st + 2, A.agg, vcols, r);
+        uint64_t *dst = A.bst + (size_t)s * sw;
+        const uint64_t *src = lst + (size_t)i * sw;
+        atomicMax((unsigned long long *)&dst[0],
+                  (unsigned long long)src[0]);
+        atomicMax((unsigned long long *)&dst[1],
+                  (unsigned long long)src[1]);
+        state_merge_atomic(dst + 2, src + 2, A.agg);
     }
 }
+
 
 struct Store {
     int64_t *keys;     /* [C+1]; slot C = spec (key == EMPTY_KEY) */
@@ -586,8 +672,9 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     A.B = o->B;
     A.agg = o->agg;
     A.err = o->d_err;
-    hipLaunchKernelGGL(k_sess_update, dim3(grid_for(n_rows)), dim3(256), 0,
-                       o->stream, A);
+    size_t shmem = (size_t)SESS_LDS_SLOTS * (2 + 2 * o->cfg.n_aggs) * 8;
+    hipLaunchKernelGGL(k_sess_update, dim3(grid_for(n_rows)), dim3(256),
+                       shmem, o->stream, A);
     SHIP(o, hipGetLastError());
     MergeArgs M = {};
     M.bkeys = o->bkeys;
