@@ -327,13 +327,20 @@ k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
  * ring.  Entries evicted on LDS collision fall through to global atomics. */
 #define LDS_SLOTS 1024   /* default; x (8B key + 4B pane + n_aggs*16B) */
 
-/* per-row body shared by the scalar and vectorized LDS kernels */
-template <int SLOTS>
+#define PANE_UNSET 0xFFFFFFFFu
+
+/* per-row body shared by the scalar and vectorized LDS kernels.
+ * COUNT_ONLY specializes the hot q5 shape (single COUNT(*)) away from the
+ * aggregate-spec loop; last_key/last_slot cache skips hash+probe when
+ * consecutive rows repeat a key (the nexmark hot auction makes runs
+ * common). */
+template <int SLOTS, bool COUNT_ONLY>
 __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                                       uint32_t *ls_pane, uint64_t *ls_st,
                                       int64_t row, uint64_t traw, int64_t key,
                                       uint64_t &local_min,
-                                      uint64_t &last_bin) {
+                                      uint64_t &last_bin, int64_t &last_key,
+                                      uint32_t &last_slot) {
     const int na = A.agg.n_aggs;
     const int64_t *const *vc = A.vcols;
     uint64_t t = traw + A.ts_offset;
@@ -350,7 +357,18 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
         last_bin = bin;
     }
     if (A.mode == 2) return;
-    /* try the LDS table first (2 probes), fall through to global */
+    const int na2 = COUNT_ONLY ? 1 : na;
+    (void)na2;
+    /* same (key, pane) as the previous row: reuse the cached LDS slot */
+    if (key == last_key && last_slot != PANE_UNSET &&
+        ls_pane[last_slot] == p) {
+        if (COUNT_ONLY) {
+            atomicAdd((unsigned long long *)
+                          (ls_st + (size_t)last_slot * 2), 1ULL);
+            return;
+        }
+    }
+    /* try the LDS table first, fall through to global */
     bool done = false;
     if (key != EMPTY_KEY) {
         uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
@@ -375,6 +393,13 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
              * than aggregating into an unknown pane */
             if (k == key && (claimed || ls_pane[s] == p)) {
                 uint64_t *st = ls_st + (size_t)s * na * 2;
+                if (COUNT_ONLY) {
+                    atomicAdd((unsigned long long *)st, 1ULL);
+                    last_key = key;
+                    last_slot = s;
+                    done = true;
+                    break;
+                }
                 for (int a = 0; a < na; a++) {
                     uint64_t *w = st + 2 * a;
                     switch (A.agg.op[a]) {
@@ -435,8 +460,6 @@ __device__ inline void lds_flush(const UpdateArgs &A, int64_t *ls_key,
     }
 }
 
-#define PANE_UNSET 0xFFFFFFFFu
-
 template <int SLOTS>
 __device__ inline void lds_init(const UpdateArgs &A, int64_t *ls_key,
                                 uint32_t *ls_pane, uint64_t *ls_st) {
@@ -457,12 +480,15 @@ k_update_lds(UpdateArgs A) {
     lds_init<LDS_SLOTS>(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_slot = PANE_UNSET;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < A.n_rows; i += stride)
-        lds_update_row<LDS_SLOTS>(A, ls_key, ls_pane, ls_st, i,
-                                  (uint64_t)A.ts_col[i],
-                                  A.key_col ? A.key_col[i] : 0, local_min,
-                                  last_bin);
+        lds_update_row<LDS_SLOTS, false>(A, ls_key, ls_pane, ls_st, i,
+                                         (uint64_t)A.ts_col[i],
+                                         A.key_col ? A.key_col[i] : 0,
+                                         local_min, last_bin, last_key,
+                                         last_slot);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
     lds_flush<LDS_SLOTS>(A, ls_key, ls_pane, ls_st);
@@ -472,7 +498,7 @@ k_update_lds(UpdateArgs A) {
  * per thread per iteration -- requires 16 B-aligned column pointers (host
  * checks).  Fewer, fatter waves: wave-dispatch cost was measurable at one
  * 8 B load per thread. */
-template <int SLOTS>
+template <int SLOTS, bool COUNT_ONLY>
 __global__ void __launch_bounds__(256)
 k_update_lds_vec(UpdateArgs A) {
     __shared__ int64_t  ls_key[SLOTS];
@@ -481,6 +507,8 @@ k_update_lds_vec(UpdateArgs A) {
     lds_init<SLOTS>(A, ls_key, ls_pane, ls_st);
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_slot = PANE_UNSET;
     int64_t n2 = A.n_rows >> 1;
     for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < n2;
          v += stride) {
@@ -491,23 +519,30 @@ k_update_lds_vec(UpdateArgs A) {
             k0 = (int64_t)kv.x;
             k1 = (int64_t)kv.y;
         }
-        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0,
-                              local_min, last_bin);
-        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y,
-                              k1, local_min, last_bin);
+        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st, 2 * v,
+                                          tsv.x, k0, local_min, last_bin,
+                                          last_key, last_slot);
+        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st,
+                                          2 * v + 1, tsv.y, k1, local_min,
+                                          last_bin, last_key, last_slot);
     }
     if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
-        lds_update_row<SLOTS>(A, ls_key, ls_pane, ls_st, A.n_rows - 1,
-                              (uint64_t)A.ts_col[A.n_rows - 1],
-                              A.key_col ? A.key_col[A.n_rows - 1] : 0,
-                              local_min, last_bin);
+        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st,
+                                          A.n_rows - 1,
+                                          (uint64_t)A.ts_col[A.n_rows - 1],
+                                          A.key_col ? A.key_col[A.n_rows - 1]
+                                                    : 0,
+                                          local_min, last_bin, last_key,
+                                          last_slot);
     fold_min_bin(local_min, A.ring.min_bin);
     __syncthreads();
     lds_flush<SLOTS>(A, ls_key, ls_pane, ls_st);
 }
 
-template __global__ void k_update_lds_vec<1024>(UpdateArgs);
-template __global__ void k_update_lds_vec<2048>(UpdateArgs);
+template __global__ void k_update_lds_vec<1024, false>(UpdateArgs);
+template __global__ void k_update_lds_vec<2048, false>(UpdateArgs);
+template __global__ void k_update_lds_vec<1024, true>(UpdateArgs);
+template __global__ void k_update_lds_vec<2048, true>(UpdateArgs);
 
 /* ------------------------------------------------------------------ */
 /* Radix-partitioned update path (ARROYO_AMD_RADIX=1, keyed
@@ -1469,13 +1504,19 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         return 0;
     }
     size_t shmem = (size_t)slots * o->agg.n_aggs * 16;
+    bool count_only = o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
+                      o->kmode == 0;
     if (vec) {
         if (slots >= 2048)
-            hipLaunchKernelGGL(k_update_lds_vec<2048>, dim3(blocks),
-                               dim3(256), shmem, o->stream, A);
+            hipLaunchKernelGGL((count_only
+                                    ? k_update_lds_vec<2048, true>
+                                    : k_update_lds_vec<2048, false>),
+                               dim3(blocks), dim3(256), shmem, o->stream, A);
         else
-            hipLaunchKernelGGL(k_update_lds_vec<1024>, dim3(blocks),
-                               dim3(256), shmem, o->stream, A);
+            hipLaunchKernelGGL((count_only
+                                    ? k_update_lds_vec<1024, true>
+                                    : k_update_lds_vec<1024, false>),
+                               dim3(blocks), dim3(256), shmem, o->stream, A);
     } else if (o->use_lds) {
         shmem = (size_t)LDS_SLOTS * o->agg.n_aggs * 16;
         hipLaunchKernelGGL(k_update_lds, dim3(blocks), dim3(256), shmem,
