@@ -298,6 +298,8 @@ class AmdSessionConfig(ctypes.Structure):
         ("max_sessions", ctypes.c_uint32),
         ("log2_batch_capacity", ctypes.c_uint32),
         ("log2_out_cap", ctypes.c_uint32),
+        ("log2_distinct", ctypes.c_uint32),
+        ("log2_cd_regions", ctypes.c_uint32),
         ("device", ctypes.c_int32),
         ("emit_to_host", ctypes.c_int32),
     ]
@@ -306,6 +308,7 @@ class AmdSessionConfig(ctypes.Structure):
 def make_session_config(gap_ns, aggs, n_keys=1, n_value_cols=0,
                         log2_capacity=16, max_sessions=8,
                         log2_batch_capacity=14, log2_out_cap=20,
+                        log2_distinct=10, log2_cd_regions=18,
                         device=0, emit_to_host=True):
     cfg = AmdSessionConfig()
     cfg.n_keys = n_keys
@@ -319,6 +322,8 @@ def make_session_config(gap_ns, aggs, n_keys=1, n_value_cols=0,
     cfg.max_sessions = max_sessions
     cfg.log2_batch_capacity = log2_batch_capacity
     cfg.log2_out_cap = log2_out_cap
+    cfg.log2_distinct = log2_distinct
+    cfg.log2_cd_regions = log2_cd_regions
     cfg.device = device
     cfg.emit_to_host = 1 if emit_to_host else 0
     return cfg
@@ -371,6 +376,7 @@ class SessionOp:
         p = prefix + "session_"
         g = lambda n: getattr(lib, p + n)
         self._fn = {}
+        self._fn["_lib_prefix"] = (lib, prefix)
         self._fn["create"] = g("create")
         self._fn["create"].restype = ctypes.c_void_p
         self._fn["create"].argtypes = [ctypes.POINTER(AmdSessionConfig)]
@@ -434,6 +440,35 @@ class SessionOp:
         keep, arr = _cols_to_ptrs(cols)
         n_rows = len(keep[0]) if keep else 0
         self._check(self._fn["restore"](self._h, arr, len(keep), n_rows))
+
+    def _bind_values(self):
+        # GPU-only extension (the oracle drains raw rows instead)
+        if "drain_values" not in self._fn:
+            g = self._fn["_lib_prefix"]
+            dv = getattr(g[0], g[1] + "session_drain_values")
+            dv.restype = ctypes.c_int
+            dv.argtypes = [ctypes.c_void_p, ctypes.POINTER(AmdOutBatch)]
+            rv = getattr(g[0], g[1] + "session_restore_values")
+            rv.restype = ctypes.c_int
+            rv.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p),
+                           ctypes.c_int32, ctypes.c_int64]
+            self._fn["drain_values"] = dv
+            self._fn["restore_values"] = rv
+
+    def drain_values(self):
+        self._bind_values()
+        out = AmdOutBatch()
+        self._check(self._fn["drain_values"](self._h, ctypes.byref(out)))
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def restore_values(self, cols):
+        self._bind_values()
+        keep, arr = _cols_to_ptrs(cols)
+        n_rows = len(keep[0]) if keep else 0
+        self._check(self._fn["restore_values"](self._h, arr, len(keep),
+                                               n_rows))
 
     def close(self):
         if self._h:

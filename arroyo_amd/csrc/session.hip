@@ -124,8 +124,85 @@ __device__ inline void atomic_fold(uint64_t *st, const AggSpec &a,
             } while (old != assumed);
             break;
         }
+        case AMD_AGG_COUNT_DISTINCT:
+            /* handled by the batch value chain, not the scalar words */
+            break;
         }
     }
+}
+
+/* -------- per-session exact distinct sets (COUNT DISTINCT) ----------
+ * Single-writer hash regions: only the one phase-2/restore thread that
+ * owns a key ever touches its sessions' regions, so inserts are plain
+ * stores.  Region layout: [D/64 bitmap words][D value slots]; regions are
+ * pool-allocated and zero-initialised once (never reused -- fired
+ * sessions orphan theirs; bounded by log2_cd_regions, loud error). */
+#define SERR_CD_FULL    5
+#define SERR_CD_REGIONS 6
+#define SERR_BV_POOL    7
+
+struct CdPool {
+    uint64_t *regions;        /* [n_regions][D/64 + D] */
+    unsigned long long *rcur;
+    int64_t n_regions;
+    uint32_t D;
+};
+
+__device__ inline uint64_t cd_region_words(uint32_t D) {
+    return D / 64 + D;
+}
+
+__device__ inline int64_t cd_alloc(const CdPool &P, int *err) {
+    int64_t idx = (int64_t)atomicAdd(P.rcur, 1ULL);
+    if (idx >= P.n_regions) { *err = SERR_CD_REGIONS; return -1; }
+    return idx;
+}
+
+/* insert v into region r (single writer); returns 1 if newly inserted */
+__device__ inline int cd_insert(const CdPool &P, int64_t r, int64_t v,
+                                int *err) {
+    uint64_t *bm = P.regions + (size_t)r * cd_region_words(P.D);
+    int64_t *vals = (int64_t *)(bm + P.D / 64);
+    uint32_t h = (uint32_t)hash64((uint64_t)v) & (P.D - 1);
+    for (uint32_t pr = 0; pr < P.D; pr++) {
+        uint32_t sl = (h + pr) & (P.D - 1);
+        uint64_t bit = 1ULL << (sl & 63);
+        if (!(bm[sl >> 6] & bit)) {
+            bm[sl >> 6] |= bit;
+            vals[sl] = v;
+            return 1;
+        }
+        if (vals[sl] == v) return 0;
+    }
+    *err = SERR_CD_FULL;
+    return 0;
+}
+
+/* merge region src into the session state acc (w0 count, w1 region+1) */
+__device__ inline void cd_merge_into(const CdPool &P, uint64_t *acc,
+                                     uint64_t src_count, uint64_t src_reg1,
+                                     int *err) {
+    if (src_reg1 == 0) return;
+    if (acc[1] == 0) {
+        acc[0] = src_count;
+        acc[1] = src_reg1;
+        return;
+    }
+    /* reinsert the smaller set into the larger */
+    uint64_t big_reg1 = acc[1], big_cnt = acc[0];
+    uint64_t small_reg1 = src_reg1, small_cnt = src_count;
+    if (small_cnt > big_cnt) {
+        big_reg1 = src_reg1; big_cnt = src_count;
+        small_reg1 = acc[1]; small_cnt = acc[0];
+    }
+    int64_t sr = (int64_t)small_reg1 - 1, br = (int64_t)big_reg1 - 1;
+    const uint64_t *bm = P.regions + (size_t)sr * cd_region_words(P.D);
+    const int64_t *vals = (const int64_t *)(bm + P.D / 64);
+    for (uint32_t sl = 0; sl < P.D; sl++)
+        if (bm[sl >> 6] & (1ULL << (sl & 63)))
+            big_cnt += cd_insert(P, br, vals[sl], err);
+    acc[0] = big_cnt;
+    acc[1] = big_reg1;
 }
 
 /* atomically merge encoded partial words (LDS flush -> global table) */
@@ -214,6 +291,13 @@ struct UpdateArgs {
     uint64_t *bst;            /* [B+1][2 + n_aggs*2]: min_ts,max_ts,aggs */
     uint32_t B;
     AggSpec agg;
+    /* COUNT DISTINCT: batch value chain (raw multiset per batch key; the
+     * single-writer phase-2 thread dedupes into the session's region) */
+    int32_t cd_agg;           /* agg index, -1 when absent */
+    int64_t *bv_val;
+    int32_t *bv_next;
+    unsigned long long *bv_cur;
+    int64_t bv_cap;
     int *err;
 };
 
@@ -246,9 +330,11 @@ k_sess_update(UpdateArgs A) {
         if (A.has_wm && t < A.wm) continue;  /* late: silently dropped */
         int64_t key = A.n_keys ? A.cols[0][r] : 0;
         /* LDS first (hot keys collapse here); EMPTY_KEY sentinel collides
-         * with a real key of -1: that rare key goes straight to global */
+         * with a real key of -1: that rare key goes straight to global.
+         * COUNT DISTINCT rows need the global slot for the value chain, so
+         * LDS staging is bypassed. */
         bool done = false;
-        if (key != EMPTY_KEY) {
+        if (key != EMPTY_KEY && A.cd_agg < 0) {
             uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u) &
                          (SESS_LDS_SLOTS - 1);
             for (int pr = 0; pr < 4 && !done; pr++) {
@@ -281,6 +367,16 @@ k_sess_update(UpdateArgs A) {
             atomicMax((unsigned long long *)&st[1],
                       (unsigned long long)enc_max((int64_t)t));
             atomic_fold(st + 2, A.agg, vcols, r);
+            if (A.cd_agg >= 0) {
+                int64_t v = A.cols[A.n_keys + A.agg.col[A.cd_agg]][r];
+                int64_t idx = (int64_t)atomicAdd(A.bv_cur, 1ULL);
+                if (idx >= A.bv_cap) { *A.err = SERR_BV_POOL; continue; }
+                A.bv_val[idx] = v;
+                A.bv_next[idx] = (int32_t)(
+                    (int64_t)atomicExch(
+                        (unsigned long long *)&st[2 + 2 * A.cd_agg],
+                        (unsigned long long)(idx + 1)) - 1);
+            }
         }
     }
     __syncthreads();
@@ -318,6 +414,10 @@ struct MergeArgs {
     Store store;
     uint64_t gap;
     AggSpec agg;
+    int32_t cd_agg;
+    CdPool cd;
+    const int64_t *bv_val;
+    const int32_t *bv_next;
     int *err;
 };
 
@@ -339,12 +439,30 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     int64_t cs = pmin, ce = pmax;
     uint64_t acc[AMD_MAX_AGGS * 2];
     for (int i = 0; i < sw; i++) acc[i] = pst[i];
+    if (M.cd_agg >= 0) {
+        /* the partial's CD word is the batch chain head, folded in below;
+         * acc starts with no region */
+        acc[2 * M.cd_agg] = 0;
+        acc[2 * M.cd_agg + 1] = 0;
+    }
     uint32_t w = 0;
     for (uint32_t i = 0; i < n; i++) {
         if (cs < se[i] + (int64_t)M.gap && ss[i] < ce + (int64_t)M.gap) {
             if (ss[i] < cs) cs = ss[i];
             if (se[i] > ce) ce = se[i];
-            state_merge(acc, sst + (size_t)i * sw, M.agg);
+            if (M.cd_agg >= 0) {
+                uint64_t *aw = acc + 2 * M.cd_agg;
+                const uint64_t *sv = sst + (size_t)i * sw + 2 * M.cd_agg;
+                cd_merge_into(M.cd, aw, sv[0], sv[1], err);
+                /* scalar merge must skip the CD words: zero them on the
+                 * source copy path by merging around */
+                uint64_t saved0 = aw[0], saved1 = aw[1];
+                state_merge(acc, sst + (size_t)i * sw, M.agg);
+                aw[0] = saved0;
+                aw[1] = saved1;
+            } else {
+                state_merge(acc, sst + (size_t)i * sw, M.agg);
+            }
         } else {
             if (w != i) {
                 ss[w] = ss[i];
@@ -356,6 +474,18 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
         }
     }
     if (w >= S.MS) { *err = SERR_SESSIONS; return; }
+    if (M.cd_agg >= 0 && pst[2 * M.cd_agg] != 0) {
+        /* fold the batch's raw value chain into the merged session's set */
+        uint64_t *aw = acc + 2 * M.cd_agg;
+        if (aw[1] == 0) {
+            int64_t r = cd_alloc(M.cd, err);
+            if (r < 0) return;
+            aw[1] = (uint64_t)(r + 1);
+        }
+        for (int32_t j = (int32_t)((int64_t)pst[2 * M.cd_agg] - 1); j >= 0;
+             j = M.bv_next[j])
+            aw[0] += cd_insert(M.cd, (int64_t)aw[1] - 1, M.bv_val[j], err);
+    }
     ss[w] = cs;
     se[w] = ce;
     for (int k = 0; k < sw; k++) sst[(size_t)w * sw + k] = acc[k];
@@ -415,7 +545,8 @@ k_sess_fire(FireArgs F) {
                     int64_t v;
                     switch (F.agg.op[a]) {
                     case AMD_AGG_COUNT:
-                    case AMD_AGG_SUM: v = (int64_t)w0; break;
+                    case AMD_AGG_SUM:
+                    case AMD_AGG_COUNT_DISTINCT: v = (int64_t)w0; break;
                     case AMD_AGG_MIN: v = dec_min(w0); break;
                     case AMD_AGG_MAX: v = dec_max(w0); break;
                     default: {  /* AVG */
@@ -495,6 +626,8 @@ struct RestoreArgs {
     uint64_t gap;
     AggSpec agg;
     int32_t n_keys;
+    int32_t cd_agg;
+    CdPool cd;
     int *err;
 };
 
@@ -502,14 +635,110 @@ __global__ void __launch_bounds__(256)
 k_sess_restore(RestoreArgs R) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int sw = 2 * R.agg.n_aggs;
+    MergeArgs M = {};
+    M.store = R.store;
+    M.gap = R.gap;
+    M.agg = R.agg;
+    M.cd_agg = R.cd_agg;
+    M.cd = R.cd;
+    M.err = R.err;
     for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < R.n_groups; g += stride) {
         for (int64_t i = R.group_off[g]; i < R.group_off[g + 1]; i++) {
             int64_t key = R.n_keys ? R.keys[i] : 0;
-            merge_partial({nullptr, nullptr, 0, R.store, R.gap, R.agg,
-                           R.err},
-                          key, R.start[i], R.end[i],
+            merge_partial(M, key, R.start[i], R.end[i],
                           R.st + (size_t)i * sw, R.err);
+        }
+    }
+}
+
+/* ---- COUNT DISTINCT checkpoint value stream ----
+ * drain: one row per live (session, value): [key?, session_start, value];
+ * restore: host-grouped by key, one thread per key re-inserts (the
+ * restored sessions' CD words start zeroed; counts rebuild here). */
+struct CdDrainArgs {
+    Store store;
+    AggSpec agg;
+    int32_t cd_agg, n_keys;
+    CdPool cd;
+    int64_t *out[3];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_drain_values(CdDrainArgs D) {
+    const Store &S = D.store;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int sw = 2 * D.agg.n_aggs;
+    for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         slot <= (int64_t)S.C; slot += stride) {
+        uint32_t n = S.ns[slot];
+        if (n == 0) continue;
+        if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
+        int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        for (uint32_t i = 0; i < n; i++) {
+            uint64_t reg1 =
+                S.s_st[(size_t)slot * S.MS * sw + (size_t)i * sw +
+                       2 * D.cd_agg + 1];
+            if (reg1 == 0) continue;
+            const uint64_t *bm =
+                D.cd.regions + (size_t)(reg1 - 1) * cd_region_words(D.cd.D);
+            const int64_t *vals = (const int64_t *)(bm + D.cd.D / 64);
+            for (uint32_t sl = 0; sl < D.cd.D; sl++) {
+                if (!(bm[sl >> 6] & (1ULL << (sl & 63)))) continue;
+                int64_t r = (int64_t)atomicAdd(D.n_out, 1ULL);
+                if (r >= D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
+                int col = 0;
+                if (D.n_keys) D.out[col++][r] = key;
+                D.out[col++][r] = S.s_start[(size_t)slot * S.MS + i];
+                D.out[col][r] = vals[sl];
+            }
+        }
+    }
+}
+
+struct CdRestoreArgs {
+    const int64_t *keys;
+    const int64_t *start;
+    const int64_t *value;
+    const int64_t *group_off;
+    int64_t n_groups;
+    Store store;
+    AggSpec agg;
+    int32_t cd_agg, n_keys;
+    CdPool cd;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_restore_values(CdRestoreArgs R) {
+    const Store &S = R.store;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int sw = 2 * R.agg.n_aggs;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.n_groups; g += stride) {
+        for (int64_t i = R.group_off[g]; i < R.group_off[g + 1]; i++) {
+            int64_t key = R.n_keys ? R.keys[i] : 0;
+            int64_t slot = key_slot(S.keys, S.C, key, R.err,
+                                    SERR_TABLE_FULL);
+            if (slot < 0) continue;
+            uint32_t n = S.ns[slot];
+            for (uint32_t si = 0; si < n; si++) {
+                if (S.s_start[(size_t)slot * S.MS + si] != R.start[i])
+                    continue;
+                uint64_t *w = &S.s_st[(size_t)slot * S.MS * sw +
+                                      (size_t)si * sw + 2 * R.cd_agg];
+                if (w[1] == 0) {
+                    int64_t r = cd_alloc(R.cd, R.err);
+                    if (r < 0) break;
+                    w[1] = (uint64_t)(r + 1);
+                }
+                w[0] += cd_insert(R.cd, (int64_t)w[1] - 1, R.value[i],
+                                  R.err);
+                break;
+            }
         }
     }
 }
@@ -527,6 +756,12 @@ struct GpuSession {
     int64_t *bkeys;
     uint64_t *bst;
     uint32_t B;
+    /* COUNT DISTINCT state */
+    int32_t cd_agg;            /* -1 when absent */
+    CdPool cd;
+    int64_t *bv_val;
+    int32_t *bv_next;
+    unsigned long long *bv_cur;
     int64_t *d_out[AMD_MAX_AGGS * 2 + 4];
     unsigned long long *d_n_out;
     int *d_err;
@@ -556,7 +791,19 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
         snprintf(g_sess_err, sizeof g_sess_err, "invalid session config");
         return nullptr;
     }
+    int cd_agg = -1;
+    for (int a = 0; a < cfg->n_aggs; a++)
+        if (cfg->agg_ops[a] == AMD_AGG_COUNT_DISTINCT) {
+            if (cd_agg >= 0 || cfg->agg_col[a] < 0) {
+                snprintf(g_sess_err, sizeof g_sess_err,
+                         "at most one COUNT DISTINCT aggregate, over a "
+                         "value column");
+                return nullptr;
+            }
+            cd_agg = a;
+        }
     GpuSession *o = new GpuSession();
+    o->cd_agg = cd_agg;
     o->cfg = *cfg;
     o->agg.n_aggs = cfg->n_aggs;
     for (int i = 0; i < cfg->n_aggs; i++) {
@@ -601,6 +848,21 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
         SALLOC(o->d_out[i], (size_t)o->out_cap * 8);
     SALLOC(o->d_n_out, 8);
     SALLOC(o->d_err, 4);
+    if (cd_agg >= 0) {
+        o->cd.D = 1u << (cfg->log2_distinct ? cfg->log2_distinct : 10);
+        o->cd.n_regions =
+            1ll << (cfg->log2_cd_regions ? cfg->log2_cd_regions : 14);
+        size_t words = (size_t)(o->cd.D / 64 + o->cd.D);
+        SALLOC(o->cd.regions, (size_t)o->cd.n_regions * words * 8);
+        SALLOC(o->cd.rcur, 8);
+        SALLOC(o->bv_val, (size_t)(1 << 20) * 8);
+        SALLOC(o->bv_next, (size_t)(1 << 20) * 4);
+        SALLOC(o->bv_cur, 8);
+        hipMemset(o->cd.regions, 0,
+                  (size_t)o->cd.n_regions * words * 8);
+        hipMemset(o->cd.rcur, 0, 8);
+        hipMemset(o->bv_cur, 0, 8);
+    }
 #undef SALLOC
     hipMemset(o->store.keys, 0xFF, C1 * 8);
     hipMemset(o->store.ns, 0, C1 * 4);
@@ -640,6 +902,11 @@ static int sess_check_err(GpuSession *o) {
         : e == SERR_SESSIONS
             ? "per-key live-session limit hit; raise max_sessions"
         : e == SERR_OUT_CAP ? "output buffer full; raise log2_out_cap"
+        : e == SERR_CD_FULL
+            ? "per-session distinct set full; raise log2_distinct"
+        : e == SERR_CD_REGIONS
+            ? "distinct-set region pool full; raise log2_cd_regions"
+        : e == SERR_BV_POOL ? "batch value-chain pool full"
                             : "device error";
     snprintf(o->err_msg, sizeof o->err_msg, "%s", msg);
     return 1;
@@ -671,6 +938,14 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     A.bst = o->bst;
     A.B = o->B;
     A.agg = o->agg;
+    A.cd_agg = o->cd_agg;
+    if (o->cd_agg >= 0) {
+        SHIP(o, hipMemsetAsync(o->bv_cur, 0, 8, o->stream));
+        A.bv_val = o->bv_val;
+        A.bv_next = o->bv_next;
+        A.bv_cur = o->bv_cur;
+        A.bv_cap = 1 << 20;
+    }
     A.err = o->d_err;
     size_t shmem = (size_t)SESS_LDS_SLOTS * (2 + 2 * o->cfg.n_aggs) * 8;
     hipLaunchKernelGGL(k_sess_update, dim3(grid_for(n_rows)), dim3(256),
@@ -683,6 +958,10 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     M.store = o->store;
     M.gap = o->cfg.gap_nanos;
     M.agg = o->agg;
+    M.cd_agg = o->cd_agg;
+    M.cd = o->cd;
+    M.bv_val = o->bv_val;
+    M.bv_next = o->bv_next;
     M.err = o->d_err;
     hipLaunchKernelGGL(k_sess_merge, dim3(grid_for((int64_t)o->B + 1)),
                        dim3(256), 0, o->stream, M);
@@ -879,6 +1158,12 @@ API int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
         for (int k = 0; k < sw; k++)
             h_st[(size_t)i * sw + k] =
                 (uint64_t)cols[o->cfg.n_keys + k][r];
+        if (o->cd_agg >= 0) {
+            /* drained CD words are (count, region id) from the OLD op;
+             * zero them -- restore_values rebuilds count and region */
+            h_st[(size_t)i * sw + 2 * o->cd_agg] = 0;
+            h_st[(size_t)i * sw + 2 * o->cd_agg + 1] = 0;
+        }
         h_start[i] = cols[o->cfg.n_keys + sw][r];
         h_end[i] = cols[o->cfg.n_keys + sw + 1][r];
     }
@@ -912,6 +1197,8 @@ API int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
     R.gap = o->cfg.gap_nanos;
     R.agg = o->agg;
     R.n_keys = o->cfg.n_keys;
+    R.cd_agg = o->cd_agg;
+    R.cd = o->cd;
     R.err = o->d_err;
     hipLaunchKernelGGL(k_sess_restore, dim3(grid_for(n_groups)), dim3(256),
                        0, o->stream, R);
@@ -921,6 +1208,117 @@ API int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
     hipFree(d_start);
     hipFree(d_end);
     hipFree(d_st);
+    hipFree(d_off);
+    return sess_check_err(o);
+}
+
+API int arroyo_amd_session_drain_values(void *h, AmdOutBatch *out) {
+    GpuSession *o = (GpuSession *)h;
+    if (sess_check_err(o)) return 1;
+    int ncols = o->cfg.n_keys + 2;
+    memset(out, 0, sizeof *out);
+    out->n_cols = ncols;
+    out->cols = (void **)calloc(ncols, sizeof(void *));
+    out->is_f64 = (int32_t *)calloc(ncols, sizeof(int32_t));
+    if (o->cd_agg < 0) {
+        for (int i = 0; i < ncols; i++) out->cols[i] = malloc(8);
+        return 0;
+    }
+    SHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    CdDrainArgs D = {};
+    D.store = o->store;
+    D.agg = o->agg;
+    D.cd_agg = o->cd_agg;
+    D.n_keys = o->cfg.n_keys;
+    D.cd = o->cd;
+    for (int i = 0; i < ncols; i++) D.out[i] = o->d_out[i];
+    D.n_out = o->d_n_out;
+    D.out_cap = o->out_cap;
+    D.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_drain_values,
+                       dim3(grid_for((int64_t)o->store.C + 1)), dim3(256), 0,
+                       o->stream, D);
+    SHIP(o, hipGetLastError());
+    unsigned long long n = 0;
+    SHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                           o->stream));
+    SHIP(o, hipStreamSynchronize(o->stream));
+    if (sess_check_err(o)) return 1;
+    out->n_rows = (int64_t)n;
+    for (int i = 0; i < ncols; i++) {
+        out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+        if (n)
+            SHIP(o, hipMemcpyAsync(out->cols[i], o->d_out[i], (size_t)n * 8,
+                                   hipMemcpyDeviceToHost, o->stream));
+    }
+    SHIP(o, hipStreamSynchronize(o->stream));
+    return 0;
+}
+
+API int arroyo_amd_session_restore_values(void *h,
+                                          const int64_t *const *cols,
+                                          int32_t n_cols, int64_t n_rows) {
+    GpuSession *o = (GpuSession *)h;
+    int want = o->cfg.n_keys + 2;
+    if (o->cd_agg < 0 || n_cols != want) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "restore_values expects %d cols and a COUNT DISTINCT "
+                 "aggregate", want);
+        return o->cd_agg < 0 && n_rows == 0 ? 0 : 1;
+    }
+    if (n_rows == 0) return 0;
+    std::vector<int64_t> order(n_rows);
+    for (int64_t i = 0; i < n_rows; i++) order[i] = i;
+    const int64_t *keys = o->cfg.n_keys ? cols[0] : nullptr;
+    if (keys)
+        std::stable_sort(order.begin(), order.end(),
+                         [&](int64_t a, int64_t b) {
+                             return keys[a] < keys[b];
+                         });
+    std::vector<int64_t> h_keys(n_rows), h_start(n_rows), h_val(n_rows);
+    std::vector<int64_t> off;
+    off.push_back(0);
+    for (int64_t i = 0; i < n_rows; i++) {
+        int64_t r = order[i];
+        h_keys[i] = keys ? keys[r] : 0;
+        if (i && keys && h_keys[i] != h_keys[i - 1]) off.push_back(i);
+        h_start[i] = cols[o->cfg.n_keys][r];
+        h_val[i] = cols[o->cfg.n_keys + 1][r];
+    }
+    off.push_back(n_rows);
+    int64_t n_groups = (int64_t)off.size() - 1;
+    int64_t *d_keys, *d_start, *d_val, *d_off;
+    SHIP(o, hipMalloc((void **)&d_keys, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_start, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_val, (size_t)n_rows * 8));
+    SHIP(o, hipMalloc((void **)&d_off, (size_t)(n_groups + 1) * 8));
+    SHIP(o, hipMemcpy(d_keys, h_keys.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_start, h_start.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_val, h_val.data(), (size_t)n_rows * 8,
+                      hipMemcpyHostToDevice));
+    SHIP(o, hipMemcpy(d_off, off.data(), (size_t)(n_groups + 1) * 8,
+                      hipMemcpyHostToDevice));
+    CdRestoreArgs R = {};
+    R.keys = d_keys;
+    R.start = d_start;
+    R.value = d_val;
+    R.group_off = d_off;
+    R.n_groups = n_groups;
+    R.store = o->store;
+    R.agg = o->agg;
+    R.cd_agg = o->cd_agg;
+    R.n_keys = o->cfg.n_keys;
+    R.cd = o->cd;
+    R.err = o->d_err;
+    hipLaunchKernelGGL(k_sess_restore_values, dim3(grid_for(n_groups)),
+                       dim3(256), 0, o->stream, R);
+    SHIP(o, hipGetLastError());
+    SHIP(o, hipStreamSynchronize(o->stream));
+    hipFree(d_keys);
+    hipFree(d_start);
+    hipFree(d_val);
     hipFree(d_off);
     return sess_check_err(o);
 }
@@ -940,6 +1338,13 @@ API void arroyo_amd_session_destroy(void *h) {
     for (int i = 0; i < max_out; i++) hipFree(o->d_out[i]);
     hipFree(o->d_n_out);
     hipFree(o->d_err);
+    if (o->cd_agg >= 0) {
+        hipFree(o->cd.regions);
+        hipFree(o->cd.rcur);
+        hipFree(o->bv_val);
+        hipFree(o->bv_next);
+        hipFree(o->bv_cur);
+    }
     for (int c = 0; c < o->n_in_cols; c++) {
         hipHostFree(o->stg_h[c]);
         hipFree(o->stg_d[c]);

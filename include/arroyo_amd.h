@@ -168,6 +168,12 @@ int arroyo_amd_session_process_batch_device(void *h,
 int arroyo_amd_session_handle_watermark(void *h, uint64_t watermark_nanos,
                                         AmdOutBatch *out);
 int arroyo_amd_session_checkpoint_drain(void *h, AmdOutBatch *out);
+/* COUNT DISTINCT value stream for checkpoints: one row per live
+ * (session, value): [key?, session_start, value]; restore after
+ * session_restore (which zeroes the CD words it rebuilds) */
+int arroyo_amd_session_drain_values(void *h, AmdOutBatch *out);
+int arroyo_amd_session_restore_values(void *h, const int64_t *const *cols,
+                                      int32_t n_cols, int64_t n_rows);
 int arroyo_amd_session_restore(void *h, const int64_t *const *cols,
                                int32_t n_cols, int64_t n_rows);
 void arroyo_amd_session_destroy(void *h);

@@ -96,13 +96,18 @@ typedef struct {
     int32_t  n_keys;            /* 0 or 1 */
     int32_t  n_value_cols;
     int32_t  n_aggs;
-    int32_t  agg_ops[AMD_MAX_AGGS];
+    int32_t  agg_ops[AMD_MAX_AGGS];  /* incl. AMD_AGG_COUNT_DISTINCT (one
+                                        per op on the GPU path): exact
+                                        per-session distinct count via
+                                        single-writer hash regions */
     int32_t  agg_col[AMD_MAX_AGGS];
     uint64_t gap_nanos;
     uint32_t log2_capacity;     /* key slots in the session store (GPU) */
     uint32_t max_sessions;      /* live sessions held inline per key (GPU) */
     uint32_t log2_batch_capacity; /* per-batch pre-aggregation table (GPU) */
     uint32_t log2_out_cap;      /* output rows per watermark (GPU) */
+    uint32_t log2_distinct;     /* per-session distinct-set capacity (GPU) */
+    uint32_t log2_cd_regions;   /* distinct-set region pool (GPU) */
     int32_t  device;
     int32_t  emit_to_host;
 } AmdSessionConfig;

@@ -948,6 +948,21 @@ static void semit(SOp *o, int64_t key, SRows *b, const int64_t *order,
         case AMD_AGG_AVG:
             o->out[col][r] = d_to_bits(acc_d / (double)(hi - lo));
             break;
+        case AMD_AGG_COUNT_DISTINCT: {
+            /* exact distinct over the session's rows (the reference runs
+             * its final aggregate over the buffered rows at close) */
+            int64_t n = hi - lo;
+            int64_t *vals = malloc((size_t)n * 8);
+            for (int64_t i = lo; i < hi; i++)
+                vals[i - lo] = b->vals[vc][order[i]];
+            qsort(vals, (size_t)n, 8, cmp_i64);
+            int64_t d = 0;
+            for (int64_t i = 0; i < n; i++)
+                if (i == 0 || vals[i] != vals[i - 1]) d++;
+            free(vals);
+            o->out[col][r] = d;
+            break;
+        }
         }
     }
     int64_t end = data_end + (int64_t)c->gap_nanos;

@@ -346,3 +346,89 @@ def test_session_config5_zipf_checkpoint_under_load():
     assert len(got) == len(want)
     assert sorted(got) == sorted(want)
     assert len(want) > 100_000  # cold Zipf keys produce many 1-row sessions
+
+
+def np_session_count_distinct(key, val, ts, gap):
+    rows = []
+    for k in np.unique(key):
+        m = key == k
+        t = ts[m]
+        v = val[m]
+        o = np.argsort(t, kind="stable")
+        t, v = t[o], v[o]
+        lo = 0
+        for i in range(1, len(t) + 1):
+            if i == len(t) or t[i] >= t[i - 1] + gap:
+                rows.append((int(k), len(np.unique(v[lo:i])), int(t[lo]),
+                             int(t[i - 1] + gap), int(t[i - 1] + gap - 1)))
+                lo = i
+    return sorted(rows)
+
+
+def cd_stream(n=20_000, seed=21, gap_s=5):
+    rng = np.random.default_rng(seed)
+    t0 = 1_600_000_000 * NS
+    ts = t0 + np.cumsum(rng.choice(
+        [NS // 10, NS // 2, 2 * NS, 7 * NS], size=n,
+        p=[0.55, 0.3, 0.1, 0.05]).astype(np.int64))
+    key = rng.integers(0, 29, size=n).astype(np.int64)
+    val = rng.integers(0, 200, size=n).astype(np.int64)
+    return key, val, ts, gap_s * NS
+
+
+def run_cd(make_op, key, val, ts, gap, checkpoint_at=None, **cfg_kw):
+    cfg_kw.setdefault("max_sessions", 256)
+    cfg_kw.setdefault("log2_capacity", 10)
+    op = make_op(cabi.make_session_config(
+        gap, [(cabi.COUNT_DISTINCT, 0)], n_keys=1, n_value_cols=1, **cfg_kw))
+    got = []
+    n = len(ts)
+    step = n // 7
+    for i, b in enumerate(range(0, n, step)):
+        sl = slice(b, min(b + step, n))
+        op.process_batch([key[sl], val[sl], ts[sl]])
+        got += rows_of(op.handle_watermark(int(ts[sl][len(ts[sl]) // 2])))
+        if checkpoint_at is not None and i == checkpoint_at:
+            drained = op.checkpoint_drain()
+            values = op.drain_values()
+            op.close()
+            op = make_op(cabi.make_session_config(
+                gap, [(cabi.COUNT_DISTINCT, 0)], n_keys=1, n_value_cols=1,
+                **cfg_kw))
+            op.restore(drained)
+            op.restore_values(values)
+    got += rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    return got
+
+
+def test_session_count_distinct_oracle_vs_numpy():
+    key, val, ts, gap = cd_stream()
+    op = oracle.make_session_op(cabi.make_session_config(
+        gap, [(cabi.COUNT_DISTINCT, 0)], n_keys=1, n_value_cols=1))
+    got = []
+    n = len(ts)
+    step = n // 7
+    for b in range(0, n, step):
+        sl = slice(b, min(b + step, n))
+        op.process_batch([key[sl], val[sl], ts[sl]])
+        got += rows_of(op.handle_watermark(int(ts[sl][len(ts[sl]) // 2])))
+    got += rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    assert sorted(got) == np_session_count_distinct(key, val, ts, gap)
+
+
+@pytest.mark.gpu
+def test_session_count_distinct_gpu_vs_numpy():
+    from arroyo_amd import gpu
+    key, val, ts, gap = cd_stream(n=60_000)
+    got = run_cd(gpu.make_session_op, key, val, ts, gap)
+    assert sorted(got) == np_session_count_distinct(key, val, ts, gap)
+
+
+@pytest.mark.gpu
+def test_session_count_distinct_gpu_checkpoint_roundtrip():
+    from arroyo_amd import gpu
+    key, val, ts, gap = cd_stream(n=30_000, seed=31)
+    got = run_cd(gpu.make_session_op, key, val, ts, gap, checkpoint_at=3)
+    assert sorted(got) == np_session_count_distinct(key, val, ts, gap)
