@@ -377,7 +377,8 @@ def cd_stream(n=20_000, seed=21, gap_s=5):
 
 
 def run_cd(make_op, key, val, ts, gap, checkpoint_at=None, **cfg_kw):
-    cfg_kw.setdefault("max_sessions", 256)
+    # sparse watermarks (one per ~n/7 rows) leave many live sessions per key
+    cfg_kw.setdefault("max_sessions", 1024)
     cfg_kw.setdefault("log2_capacity", 10)
     op = make_op(cabi.make_session_config(
         gap, [(cabi.COUNT_DISTINCT, 0)], n_keys=1, n_value_cols=1, **cfg_kw))
