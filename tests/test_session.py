@@ -295,10 +295,11 @@ def zipf_keys(rng, n, key_space=10_000_000, s=1.0):
 
 @pytest.mark.gpu
 def test_session_config5_zipf_checkpoint_under_load():
-    """BASELINE.json config 5 shape at single-GPU scale: session windows over
-    a Zipf stream drawn from a 10M-key space, with a checkpoint drained and
-    restored mid-stream while data keeps flowing; results must equal the
-    oracle's uninterrupted run."""
+    """BASELINE.json config 5 verbatim at single-GPU scale: session-window
+    COUNT DISTINCT over a Zipf stream drawn from a 10M-key space, with a
+    checkpoint (sessions + distinct-value sets) drained and restored
+    mid-stream while data keeps flowing; results must equal the oracle's
+    uninterrupted run."""
     from arroyo_amd import gpu
 
     rng = np.random.default_rng(55)
@@ -306,30 +307,33 @@ def test_session_config5_zipf_checkpoint_under_load():
     gap = 5 * NS
     t0 = 1_600_000_000 * NS
     key = zipf_keys(rng, n)
+    item = rng.integers(0, 1000, size=n).astype(np.int64)
     # ~64K rows per second of event time
     ts = t0 + (np.arange(n, dtype=np.int64) * NS) // 65536
     batch = 65536
 
     def cfg():
         return cabi.make_session_config(
-            gap, [(cabi.COUNT, -1)], n_keys=1, n_value_cols=0,
+            gap, [(cabi.COUNT_DISTINCT, 0)], n_keys=1, n_value_cols=1,
             log2_capacity=21, max_sessions=8, log2_batch_capacity=17,
-            log2_out_cap=22)
+            log2_out_cap=22, log2_distinct=11, log2_cd_regions=21)
 
     # GPU run with a mid-stream checkpoint: drain into a fresh op
     op = gpu.make_session_op(cfg())
     got = []
     wm = 0
     for b in range(0, n, batch):
-        cols = [key[b:b + batch], ts[b:b + batch]]
+        cols = [key[b:b + batch], item[b:b + batch], ts[b:b + batch]]
         op.process_batch(cols)
         wm = int(ts[min(b + batch, n) - 1]) - NS
         got += rows_of(op.handle_watermark(wm))
         if b == (n // batch // 2) * batch:
             drained = op.checkpoint_drain()
+            values = op.drain_values()
             op.close()
             op = gpu.make_session_op(cfg())
             op.restore(drained)
+            op.restore_values(values)
     got += rows_of(op.handle_watermark(U64MAX))
     op.close()
 
@@ -337,7 +341,8 @@ def test_session_config5_zipf_checkpoint_under_load():
     o = oracle.make_session_op(cfg())
     want = []
     for b in range(0, n, batch):
-        o.process_batch([key[b:b + batch], ts[b:b + batch]])
+        o.process_batch([key[b:b + batch], item[b:b + batch],
+                         ts[b:b + batch]])
         want += rows_of(o.handle_watermark(
             int(ts[min(b + batch, n) - 1]) - NS))
     want += rows_of(o.handle_watermark(U64MAX))
