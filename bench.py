@@ -98,6 +98,10 @@ def run_gpu(args):
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    # functional testing of the N>1 path on a 1-GPU box: all ranks share
+    # device 0 (RCCL permitting); never set in real runs
+    if os.environ.get("BENCH_FORCE_DEV") is not None:
+        local_rank = int(os.environ["BENCH_FORCE_DEV"])
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
