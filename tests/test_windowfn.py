@@ -194,3 +194,23 @@ def test_windowfn_gpu_checkpoint_roundtrip():
     want = rows_of(o.handle_watermark(U64MAX))
     o.close()
     assert got == want
+
+
+@pytest.mark.gpu
+def test_windowfn_gpu_large_instant():
+    """One 500K-row instant: exercises the hipCUB sort pipeline at scale."""
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(6)
+    t0 = 1_600_000_000 * NS
+    n = 500_000
+    part = rng.integers(0, 1000, size=n).astype(np.int64)
+    v = rng.integers(0, 10**6, size=n).astype(np.int64)
+    ts = np.full(n, t0, dtype=np.int64)
+    op = gpu.make_windowfn_op(cabi.make_windowfn_config(
+        n_cols=3, part_col=0, order=[(1, True)], limit=3,
+        log2_rows_cap=19, instants=64, log2_out_cap=20))
+    op.process_batch([part, v, ts])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    want = np_row_numbers([part, v, ts], 0, [(1, True)], 3)
+    assert got == want
