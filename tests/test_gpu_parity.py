@@ -268,3 +268,22 @@ def test_partition_kernel_matches_host():
                           ts[host_pid == p].tolist()))
         assert seg == wantseg
         off += counts[p]
+
+
+def test_nondivisible_width_slide():
+    """hop(3s, 10s): width not a multiple of slide — windows span 3-4 panes
+    with a partial leading pane; bit-exact vs oracle."""
+    cols = nexmark.bids(300_000, events_per_sec=50_000, seed=17)
+    got, want = run_both(cols, width_ns=10 * NS, slide_ns=3 * NS, n_keys=1,
+                         n_value_cols=0, aggs=[(cabi.COUNT, -1)],
+                         log2_capacity=14, ring_panes=32)
+    assert_parity(got, want)
+
+
+def test_width_equals_slide():
+    """hop(10s, 10s) behaves like tumbling through the sliding machinery."""
+    cols = nexmark.bids(200_000, events_per_sec=50_000, seed=19)
+    got, want = run_both(cols, width_ns=10 * NS, slide_ns=10 * NS, n_keys=1,
+                         n_value_cols=0, aggs=[(cabi.COUNT, -1)],
+                         log2_capacity=14, ring_panes=16)
+    assert_parity(got, want)
