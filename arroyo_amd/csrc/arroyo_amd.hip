@@ -1270,6 +1270,14 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         snprintf(g_err, sizeof g_err, "invalid config");
         return nullptr;
     }
+    if (!cfg->is_tumbling && cfg->slide_nanos != 0 &&
+        cfg->width_nanos % cfg->slide_nanos != 0) {
+        /* the reference's planner rejects this (arroyo-planner/src/lib.rs
+         * :644: "hop() width currently must be a multiple of slide") */
+        snprintf(g_err, sizeof g_err,
+                 "hop() width must be a multiple of slide");
+        return nullptr;
+    }
     GpuOp *o = new GpuOp();
     o->cfg = *cfg;
     if (o->cfg.is_tumbling) o->cfg.slide_nanos = o->cfg.width_nanos;

@@ -366,6 +366,14 @@ ORACLE_API void *oracle_create(const AmdWindowConfig *cfg) {
     if (!cfg || cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS ||
         cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->width_nanos == 0)
         return NULL;
+    /* the reference's planner rejects non-divisible hop parameters
+     * ("hop() width currently must be a multiple of slide",
+     * arroyo-planner/src/lib.rs:644); the operator's behavior there is
+     * unreachable (and its earliest-bin state machine would not
+     * terminate), so reject them here too */
+    if (!cfg->is_tumbling && cfg->slide_nanos != 0 &&
+        cfg->width_nanos % cfg->slide_nanos != 0)
+        return NULL;
     Op *o = calloc(1, sizeof(Op));
     o->cfg = *cfg;
     if (o->cfg.is_tumbling) o->cfg.slide_nanos = o->cfg.width_nanos;

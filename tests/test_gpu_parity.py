@@ -270,14 +270,18 @@ def test_partition_kernel_matches_host():
         off += counts[p]
 
 
-def test_nondivisible_width_slide():
-    """hop(3s, 10s): width not a multiple of slide — windows span 3-4 panes
-    with a partial leading pane; bit-exact vs oracle."""
-    cols = nexmark.bids(300_000, events_per_sec=50_000, seed=17)
-    got, want = run_both(cols, width_ns=10 * NS, slide_ns=3 * NS, n_keys=1,
-                         n_value_cols=0, aggs=[(cabi.COUNT, -1)],
-                         log2_capacity=14, ring_panes=32)
-    assert_parity(got, want)
+def test_nondivisible_width_slide_rejected():
+    """hop(3s, 10s): the reference's planner rejects width not a multiple
+    of slide (arroyo-planner/src/lib.rs:644) — and the operator's
+    earliest-bin state machine would not terminate there — so creation
+    fails loudly on both paths."""
+    with pytest.raises(Exception):
+        gpu_op(width_ns=10 * NS, slide_ns=3 * NS, n_keys=1, n_value_cols=0,
+               aggs=[(cabi.COUNT, -1)], log2_capacity=14, ring_panes=32)
+    with pytest.raises(Exception):
+        oracle_op(width_ns=10 * NS, slide_ns=3 * NS, n_keys=1,
+                  n_value_cols=0, aggs=[(cabi.COUNT, -1)],
+                  log2_capacity=14)
 
 
 def test_width_equals_slide():
