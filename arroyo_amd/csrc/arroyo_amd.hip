@@ -1278,8 +1278,15 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
                  "hop() width must be a multiple of slide");
         return nullptr;
     }
+    if (!cfg->is_tumbling && cfg->slide_nanos == 0) {
+        snprintf(g_err, sizeof g_err, "sliding window needs slide > 0");
+        return nullptr;
+    }
     GpuOp *o = new GpuOp();
     o->cfg = *cfg;
+    /* hop(x, x) is a tumble window (arroyo-planner/src/lib.rs:649-651) */
+    if (!o->cfg.is_tumbling && o->cfg.slide_nanos == o->cfg.width_nanos)
+        o->cfg.is_tumbling = 1;
     if (o->cfg.is_tumbling) o->cfg.slide_nanos = o->cfg.width_nanos;
     o->width = o->cfg.width_nanos;
     o->slide = o->cfg.slide_nanos;

@@ -374,8 +374,13 @@ ORACLE_API void *oracle_create(const AmdWindowConfig *cfg) {
     if (!cfg->is_tumbling && cfg->slide_nanos != 0 &&
         cfg->width_nanos % cfg->slide_nanos != 0)
         return NULL;
+    if (!cfg->is_tumbling && cfg->slide_nanos == 0)
+        return NULL;
     Op *o = calloc(1, sizeof(Op));
     o->cfg = *cfg;
+    /* hop(x, x) is a tumble window (arroyo-planner/src/lib.rs:649-651) */
+    if (!o->cfg.is_tumbling && o->cfg.slide_nanos == o->cfg.width_nanos)
+        o->cfg.is_tumbling = 1;
     if (o->cfg.is_tumbling) o->cfg.slide_nanos = o->cfg.width_nanos;
     o->out_cols = cfg->n_keys + cfg->n_aggs + 3;
     o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
