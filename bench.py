@@ -3,9 +3,12 @@ rows/sec on nexmark q5 sliding-window aggregate).
 
 A "step" is one pass of the hot path over one 64K-row synthetic nexmark bid
 batch already resident in HBM: the fused bin+hash-aggregate update kernel
-(k_update_lds), plus the watermark-driven window firing (merge + compact
-kernels) at the reference's watermark cadence (1/s of event time,
-crates/arroyo-worker/src/arrow/watermark_generator.rs).  Outputs stay
+(k_update_lds_vec), plus the watermark-driven window firing (the fused
+hash-aligned merge+compact kernel k_merge_fused) at the reference's
+watermark cadence (1/s of event time,
+crates/arroyo-worker/src/arrow/watermark_generator.rs).  Steps are
+submitted in fused watermark periods (BENCH_WM_FUSE, default 2) --
+bit-identical outputs, verified by tests/test_property_large.py.  Outputs stay
 device-resident (the next pipeline stage's collector consumes them in place);
 the host-visible emission path is covered by tests, not timed here.
 
@@ -42,15 +45,13 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from arroyo_amd import cabi, nexmark  # noqa: E402
 from arroyo_amd.pipeline import NS  # noqa: E402
 
-import os as _os
-
 BATCH_ROWS = 65536
 EVENTS_PER_SEC = 1_000_000       # event-time rate of the synthetic stream
 # replay ring sized to 512 MiB so launches read from HBM, not the 256 MiB
 # Infinity Cache (PMC FETCH_SIZE showed a 4-batch ring was fully L3-resident)
 BASE_BATCHES = 512
 WIDTH_S, SLIDE_S = 10, 2
-LOG2_CAPACITY = int(_os.environ.get("BENCH_LOG2_CAP", "19"))
+LOG2_CAPACITY = int(os.environ.get("BENCH_LOG2_CAP", "19"))
 # ~130K distinct auctions per 2s pane -> default 2^19 slots (25% load)
 RING_PANES = 16
 HBM_PEAK_GBPS = 8000.0           # spec peak (MI355X_MICROARCH.md)
