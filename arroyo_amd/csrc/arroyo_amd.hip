@@ -27,10 +27,12 @@
  *     domain), so pane retirement (K9) is hipMemsetAsync and inserts need no
  *     init handshake: any lane may atomically fold its row into a slot the
  *     moment the key CAS lands.
- *   - Watermark advance (K4+K5) merges the width/slide live panes into a
- *     merge table (k_merge) and compacts occupied slots into output columns
- *     (k_compact) with a single atomic cursor; outputs stay device-resident
- *     unless emit_to_host is set.
+ *   - Watermark advance (K4+K5): the fused hash-aligned kernel
+ *     k_merge_fused (one workgroup per home-slot range, LDS dedup across
+ *     panes, direct emission — no merge table, no state atomics) for
+ *     narrow aggregate sets; the legacy merge-table k_merge + k_compact
+ *     pair for wide sets and checkpoint drains.  Outputs stay
+ *     device-resident unless emit_to_host is set.
  *   - The watermark/firing state machine stays on the host (it is
  *     control-rate), replicated statement-for-statement from the reference
  *     (see host section below); the only per-watermark device traffic is a

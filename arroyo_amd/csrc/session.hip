@@ -7,8 +7,9 @@
  * BTreeMap-of-batches + per-key DataFusion exec streams):
  *   - Per-batch pre-aggregation (k_sess_update): one fused kernel folds each
  *     row into a per-batch open-addressing table keyed by the session key,
- *     accumulating (min_ts, max_ts, aggregate states) with LDS-free HBM
- *     atomics on memset-zero-identity encoded words.  Because a batch's
+ *     accumulating (min_ts, max_ts, aggregate states) on memset-zero-
+ *     identity encoded words, with an LDS staging table absorbing hot-key
+ *     atomics first.  Because a batch's
  *     event-time span is < gap (enforced; the host splits wider batches into
  *     gap/2 buckets), a key's rows within one batch always belong to ONE
  *     session, so (min_ts, max_ts, states) is a valid partial session —
@@ -35,6 +36,9 @@
  *     (process_batch :849-874, gt_eq filter) — silently, as the reference
  *     does (unlike the window operators, where pre-watermark data is a
  *     fatal error).
+ *   - COUNT DISTINCT keeps an exact per-session value set in single-writer
+ *     hash regions (see the cd_* primitives below): phase 1 appends raw
+ *     values to a per-batch chain, the per-key phase-2 thread dedupes.
  *
  * Parity is pinned against oracle/arroyo_oracle.c (itself pinned against
  * the reference's session_window / global_session_window golden vectors) by
