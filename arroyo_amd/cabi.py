@@ -509,6 +509,7 @@ class ExpJoinOp:
         self._fn["expire"] = g("expire")
         self._fn["expire"].restype = ctypes.c_int
         self._fn["expire"].argtypes = [ctypes.c_void_p]
+        self._fn["_lib_prefix"] = (lib, prefix)
         self._fn["checkpoint_drain"] = g("checkpoint_drain")
         self._fn["checkpoint_drain"].restype = ctypes.c_int
         self._fn["checkpoint_drain"].argtypes = [
@@ -548,6 +549,39 @@ class ExpJoinOp:
 
     def handle_watermark(self, wm):
         self._check(self._fn["handle_watermark"](self._h, wm))
+
+    def _bind_device(self):
+        # GPU-only extension (the oracle has no device-resident path)
+        if "process_batch_device" not in self._fn:
+            lib, prefix = self._fn["_lib_prefix"]
+            pd = getattr(lib, prefix + "expjoin_process_batch_device")
+            pd.restype = ctypes.c_int
+            pd.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                           ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+                           ctypes.c_int64]
+            co = getattr(lib, prefix + "expjoin_collect")
+            co.restype = ctypes.c_int
+            co.argtypes = [ctypes.c_void_p, ctypes.POINTER(AmdOutBatch)]
+            self._fn["process_batch_device"] = pd
+            self._fn["collect"] = co
+
+    def process_batch_device(self, side, dptrs, n_rows):
+        """Device-resident ingest: dptrs are raw device addresses of the
+        same [key, vals..., ts] columns process_batch takes.  Matches
+        accumulate on the device until collect()."""
+        self._bind_device()
+        arr = (ctypes.c_void_p * len(dptrs))(*dptrs)
+        self._check(self._fn["process_batch_device"](
+            self._h, side, arr, len(dptrs), n_rows))
+
+    def collect(self):
+        """Drain device-accumulated match rows to host numpy columns."""
+        self._bind_device()
+        out = AmdOutBatch()
+        self._check(self._fn["collect"](self._h, ctypes.byref(out)))
+        res = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return res
 
     def expire(self):
         self._check(self._fn["expire"](self._h))
@@ -641,6 +675,7 @@ class UpdAggOp:
         self._fn["expire"].restype = ctypes.c_int
         self._fn["expire"].argtypes = [ctypes.c_void_p, ctypes.c_int64,
                                        ctypes.POINTER(AmdOutBatch)]
+        self._fn["_lib_prefix"] = (lib, prefix)
         self._fn["checkpoint_drain"] = g("checkpoint_drain")
         self._fn["checkpoint_drain"].restype = ctypes.c_int
         self._fn["checkpoint_drain"].argtypes = [
@@ -671,6 +706,21 @@ class UpdAggOp:
         n_rows = len(keep[0]) if keep else 0
         self._check(self._fn["process_batch"](self._h, arr, len(keep),
                                               n_rows))
+
+    def process_batch_device(self, dptrs, n_rows):
+        """Device-resident ingest (GPU-only extension): dptrs are raw
+        device addresses of the [keys..., vals..., is_retract] columns."""
+        if "process_batch_device" not in self._fn:
+            lib, prefix = self._fn["_lib_prefix"]
+            pd = getattr(lib, prefix + "updagg_process_batch_device")
+            pd.restype = ctypes.c_int
+            pd.argtypes = [ctypes.c_void_p,
+                           ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+                           ctypes.c_int64]
+            self._fn["process_batch_device"] = pd
+        arr = (ctypes.c_void_p * len(dptrs))(*dptrs)
+        self._check(self._fn["process_batch_device"](self._h, arr,
+                                                     len(dptrs), n_rows))
 
     def flush(self):
         out = AmdOutBatch()

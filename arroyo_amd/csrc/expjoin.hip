@@ -479,6 +479,47 @@ API int arroyo_amd_expjoin_process_batch(void *h, int32_t side,
     return 0;
 }
 
+/* device-resident ingest: probes + inserts without host staging; matched
+ * rows stay in d_out (the n_out cursor is NOT reset here so several
+ * device batches may accumulate until expjoin_collect drains them) */
+API int arroyo_amd_expjoin_process_batch_device(void *h, int32_t side,
+                                                const int64_t *const *dcols,
+                                                int32_t n_cols,
+                                                int64_t n_rows) {
+    GpuExpJoin *o = (GpuExpJoin *)h;
+    int32_t nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    if (n_cols != 1 + nv + 1) {
+        snprintf(o->err_msg, sizeof o->err_msg, "side %d expects %d cols",
+                 side, 1 + nv + 1);
+        return 1;
+    }
+    return ej_ingest(o, side, dcols, n_rows, 1);
+}
+
+/* drain the accumulated device-side match rows to the host */
+API int arroyo_amd_expjoin_collect(void *h, AmdOutBatch *out) {
+    GpuExpJoin *o = (GpuExpJoin *)h;
+    unsigned long long n = 0;
+    EJHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                            o->stream));
+    EJHIP(o, hipStreamSynchronize(o->stream));
+    if (ej_check_err(o)) return 1;
+    memset(out, 0, sizeof *out);
+    out->n_rows = (int64_t)n;
+    out->n_cols = o->out_cols;
+    out->cols = (void **)calloc(o->out_cols, sizeof(void *));
+    out->is_f64 = (int32_t *)calloc(o->out_cols, sizeof(int32_t));
+    for (int i = 0; i < o->out_cols; i++) {
+        out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+        if (n)
+            EJHIP(o, hipMemcpyAsync(out->cols[i], o->d_out[i], (size_t)n * 8,
+                                    hipMemcpyDeviceToHost, o->stream));
+    }
+    EJHIP(o, hipStreamSynchronize(o->stream));
+    EJHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    return 0;
+}
+
 API int arroyo_amd_expjoin_handle_watermark(void *h, uint64_t wm) {
     GpuExpJoin *o = (GpuExpJoin *)h;
     o->has_wm = 1;

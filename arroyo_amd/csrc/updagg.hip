@@ -683,6 +683,31 @@ API int arroyo_amd_updagg_process_batch(void *h, const int64_t *const *cols,
     return 0;
 }
 
+API int arroyo_amd_updagg_process_batch_device(void *h,
+                                               const int64_t *const *dcols,
+                                               int32_t n_cols,
+                                               int64_t n_rows) {
+    GpuUpdAgg *o = (GpuUpdAgg *)h;
+    if (n_cols != o->n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->n_in_cols, n_cols);
+        return 1;
+    }
+    UpdateArgs A = {};
+    for (int c = 0; c < n_cols; c++) A.cols[c] = dcols[c];
+    A.n_keys = o->cfg.n_keys;
+    A.n_vals = o->cfg.n_value_cols;
+    A.n_rows = n_rows;
+    A.store = o->store;
+    A.agg = o->agg;
+    A.cur_epoch = o->cur_epoch;
+    A.err = o->d_err;
+    hipLaunchKernelGGL(k_updagg_update, dim3(ua_grid(n_rows)), dim3(256), 0,
+                       o->stream, A);
+    UHIP(o, hipGetLastError());
+    return 0;
+}
+
 API int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out) {
     GpuUpdAgg *o = (GpuUpdAgg *)h;
     if (ua_check_err(o)) return 1;

@@ -209,6 +209,10 @@ int arroyo_amd_expjoin_process_batch(void *h, int32_t side,
                                      const int64_t *const *cols,
                                      int32_t n_cols, int64_t n_rows,
                                      AmdOutBatch *out);
+int arroyo_amd_expjoin_process_batch_device(void *h, int32_t side,
+                                            const int64_t *const *dcols,
+                                            int32_t n_cols, int64_t n_rows);
+int arroyo_amd_expjoin_collect(void *h, AmdOutBatch *out);
 int arroyo_amd_expjoin_handle_watermark(void *h, uint64_t watermark_nanos);
 int arroyo_amd_expjoin_expire(void *h);
 int arroyo_amd_expjoin_checkpoint_drain(void *h, int32_t side,
@@ -243,6 +247,9 @@ const char *arroyo_amd_expjoin_last_error(void *h);
 void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg);
 int arroyo_amd_updagg_process_batch(void *h, const int64_t *const *cols,
                                     int32_t n_cols, int64_t n_rows);
+int arroyo_amd_updagg_process_batch_device(void *h,
+                                           const int64_t *const *dcols,
+                                           int32_t n_cols, int64_t n_rows);
 int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out);
 /* TTL eviction of keys idle for more than `idle_flushes` flush epochs
  * (the reference's UpdatingCache::time_out, wall-clock there); emits the

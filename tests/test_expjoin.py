@@ -239,3 +239,40 @@ def test_expjoin_gpu_key_minus_one():
                             np.array([t0 + NS] * 3, dtype=np.int64)])
     op.close()
     assert rows_of(out) == [(-1, t0 + NS), (-1, t0 + NS)]
+
+
+@pytest.mark.gpu
+def test_expjoin_gpu_device_resident_matches_host():
+    """process_batch_device + collect (device-resident pipeline surface,
+    no host staging) must produce the same match set as the host path."""
+    import torch
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(53)
+    t0 = 1_600_000_000 * NS
+    cfgk = dict(n_left_vals=1, n_right_vals=1)
+    d = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    h = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    dev = torch.device("cuda", 0)
+    want = []
+    keep = []  # keep device tensors alive until collect
+    for step in range(6):
+        n = 300
+        k = rng.integers(0, 48, size=n).astype(np.int64)
+        v = rng.integers(0, 10**6, size=n).astype(np.int64)
+        ts = t0 + (step * 60 + np.sort(
+            rng.integers(0, 60, size=n))).astype(np.int64) * NS
+        side = step % 2
+        want += rows_of(h.process_batch(side, [k, v, ts]))
+        tk = torch.from_numpy(k).to(dev)
+        tv = torch.from_numpy(v).to(dev)
+        tt = torch.from_numpy(ts).to(dev)
+        keep += [tk, tv, tt]
+        d.process_batch_device(side, [tk.data_ptr(), tv.data_ptr(),
+                                      tt.data_ptr()], n)
+    got = rows_of(d.collect())
+    # a second collect after draining must return nothing
+    assert rows_of(d.collect()) == []
+    d.close()
+    h.close()
+    assert sorted(got) == sorted(want)
+    assert len(want) > 300

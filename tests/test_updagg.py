@@ -447,3 +447,38 @@ def test_every_aggregate_subset_oracle():
 def test_every_aggregate_subset_gpu():
     from arroyo_amd import gpu
     run_every_aggregate_subset(gpu.make_updagg_op)
+
+
+@pytest.mark.gpu
+def test_updagg_gpu_device_resident_matches_host():
+    """process_batch_device (device-resident pipeline surface, no host
+    staging) must leave the same merged state as the host ingest path."""
+    import torch
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(59)
+    n = 5000
+    key = rng.integers(0, 64, size=n).astype(np.int64)
+    val = rng.integers(0, 100, size=n).astype(np.int64)
+    retract = np.zeros(n, dtype=np.int64)
+    aggs = [(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.COUNT_DISTINCT, 0)]
+    h = gpu.make_updagg_op(cabi.make_updagg_config(aggs, n_keys=1,
+                                                   n_value_cols=1))
+    d = gpu.make_updagg_op(cabi.make_updagg_config(aggs, n_keys=1,
+                                                   n_value_cols=1))
+    dev = torch.device("cuda", 0)
+    got, want = [], []
+    for lo in range(0, n, 1000):
+        sl = slice(lo, lo + 1000)
+        h.process_batch([key[sl], val[sl], retract[sl]])
+        want += rows_of(h.flush())
+        tk = torch.from_numpy(key[sl].copy()).to(dev)
+        tv = torch.from_numpy(val[sl].copy()).to(dev)
+        tr = torch.from_numpy(retract[sl].copy()).to(dev)
+        d.process_batch_device([tk.data_ptr(), tv.data_ptr(),
+                                tr.data_ptr()], sl.stop - sl.start)
+        torch.cuda.synchronize()
+        got += rows_of(d.flush())
+    h.close()
+    d.close()
+    assert merge_debezium(got) == merge_debezium(want)
+    assert sorted(got) == sorted(want)
