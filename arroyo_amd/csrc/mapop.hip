@@ -214,10 +214,8 @@ static int map_grid(int64_t want) {
     return (int)(w > 2048 ? 2048 : (w < 1 ? 1 : w));
 }
 
-API int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
-                                     int32_t n_cols, int64_t n_rows,
-                                     AmdOutBatch *out) {
-    GpuMap *o = (GpuMap *)h;
+static int map_run(GpuMap *o, const int64_t *const *cols, int32_t n_cols,
+                   int64_t n_rows, AmdOutBatch *out) {
     const AmdMapConfig &c = o->cfg;
     if (n_cols != c.n_in_cols) {
         snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
@@ -307,6 +305,24 @@ API int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
     for (int i = 0; i < c.n_out; i++)
         if (!out->cols[i]) out->cols[i] = malloc(8);
     return 0;
+}
+
+API int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
+                                     int32_t n_cols, int64_t n_rows,
+                                     AmdOutBatch *out) {
+    GpuMap *o = (GpuMap *)h;
+    memset(out, 0, sizeof *out); /* so early-validation errors leave a
+                                  * well-defined (empty) out */
+    int rc = map_run(o, cols, n_cols, n_rows, out);
+    if (rc && out->cols) {
+        /* error exits (device error, HIP failure) must not leak the
+         * partially filled output the caller will never free */
+        for (int i = 0; i < out->n_cols; i++) free(out->cols[i]);
+        free(out->cols);
+        free(out->is_f64);
+        memset(out, 0, sizeof *out);
+    }
+    return rc;
 }
 
 API void arroyo_amd_map_destroy(void *h) {

@@ -127,6 +127,10 @@ def bench_updagg_device():
 
 
 if __name__ == "__main__":
+    # torch's bundled HIP runtime must initialize before the op library's
+    # (system-ROCm) runtime touches the device, or torch.cuda breaks
+    import torch
+    torch.zeros(1, device="cuda")
     bench_expjoin()
     bench_updagg()
     bench_expjoin_device()

@@ -267,6 +267,9 @@ def test_expjoin_gpu_device_resident_matches_host():
         tv = torch.from_numpy(v).to(dev)
         tt = torch.from_numpy(ts).to(dev)
         keep += [tk, tv, tt]
+        # H2D copies run on torch's stream, the op on its own: fence before
+        # handing the pointers over
+        torch.cuda.synchronize()
         d.process_batch_device(side, [tk.data_ptr(), tv.data_ptr(),
                                       tt.data_ptr()], n)
     got = rows_of(d.collect())

@@ -474,9 +474,10 @@ def test_updagg_gpu_device_resident_matches_host():
         tk = torch.from_numpy(key[sl].copy()).to(dev)
         tv = torch.from_numpy(val[sl].copy()).to(dev)
         tr = torch.from_numpy(retract[sl].copy()).to(dev)
+        # fence torch's H2D stream before the op's stream reads the data
+        torch.cuda.synchronize()
         d.process_batch_device([tk.data_ptr(), tv.data_ptr(),
                                 tr.data_ptr()], sl.stop - sl.start)
-        torch.cuda.synchronize()
         got += rows_of(d.flush())
     h.close()
     d.close()
