@@ -71,7 +71,6 @@ struct DeviceRing {
     int64_t  *keys;      /* [R][C] */
     uint64_t *state;     /* [R][C][n_aggs][2] encoded */
     uint64_t *tag;       /* [R] bin nanos or EMPTY_TAG */
-    uint64_t *fill;      /* [R] occupied-slot count */
     /* special entry per pane for an actual key == EMPTY_KEY */
     uint32_t *spec_used; /* [R] */
     uint64_t *spec_state;/* [R][n_aggs][2] */
@@ -705,7 +704,6 @@ struct MergeArgs {
     AggSpec agg;
     int64_t  *m_keys;     /* [CM] */
     uint64_t *m_state;    /* [CM][n_aggs][2] */
-    uint64_t *m_fill;
     uint32_t *m_spec_used;
     uint64_t *m_spec_state;
     uint32_t  CM;
@@ -1099,11 +1097,9 @@ k_restore(RestoreArgs R) {
 
 /* clear one pane's scalar metadata in a single launch (replaces four
  * 4-8 B memset launches at retire) */
-__global__ void k_retire_meta(uint64_t *tag, uint64_t *fill,
-                              uint32_t *spec_used, uint64_t *spec_state,
-                              int na2) {
+__global__ void k_retire_meta(uint64_t *tag, uint32_t *spec_used,
+                              uint64_t *spec_state, int na2) {
     *tag = EMPTY_TAG;
-    *fill = 0;
     *spec_used = 0;
     for (int w = 0; w < na2; w++) spec_state[w] = 0;
 }
@@ -1176,7 +1172,7 @@ struct GpuOp {
      * contiguous zero-blob so a fire clears them with a single memset */
     uint64_t *m_zero_blob;
     size_t    m_zero_bytes;
-    uint64_t *m_state, *m_fill;
+    uint64_t *m_state;
     uint32_t *m_spec_used;
     uint64_t *m_spec_state;
     uint32_t CM;
@@ -1255,8 +1251,7 @@ static int ring_retire(GpuOp *o, uint32_t slot) {
                                     (size_t)slot * o->ring.C * na * 2,
                                 0, (size_t)o->ring.C * na * 16, o->stream));
     hipLaunchKernelGGL(k_retire_meta, dim3(1), dim3(1), 0, o->stream,
-                       o->ring.tag + slot, o->ring.fill + slot,
-                       o->ring.spec_used + slot,
+                       o->ring.tag + slot, o->ring.spec_used + slot,
                        o->ring.spec_state + (size_t)slot * na * 2,
                        (int)(na * 2));
     HIP_CHECK(o, hipGetLastError());
@@ -1347,16 +1342,15 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (hipHostMalloc((void **)&o->h_status,
                       (2 + (size_t)o->ring.R) * 8) != hipSuccess)
         return fail("h_status", hipErrorOutOfMemory);
-    ALLOC(o->ring.fill, (size_t)o->ring.R * 8);
     ALLOC(o->ring.spec_used, (size_t)o->ring.R * 4);
     ALLOC(o->ring.spec_state, (size_t)o->ring.R * na * 16);
     ALLOC(o->m_keys, (size_t)o->CM * 8);
-    o->m_zero_bytes = (size_t)o->CM * na * 16 + 8 + 8 + na * 16 + 8;
+    o->m_zero_bytes = (size_t)o->CM * na * 16 + 8 + na * 16 + 8;
     ALLOC(o->m_zero_blob, o->m_zero_bytes);
     o->m_state = o->m_zero_blob;
-    o->m_fill = o->m_zero_blob + (size_t)o->CM * na * 2;
-    o->m_spec_used = (uint32_t *)(o->m_fill + 1);
-    o->m_spec_state = o->m_fill + 2;
+    o->m_spec_used =
+        (uint32_t *)(o->m_zero_blob + (size_t)o->CM * na * 2);
+    o->m_spec_state = o->m_zero_blob + (size_t)o->CM * na * 2 + 1;
     o->d_n_out = (unsigned long long *)(o->m_spec_state + na * 2);
     o->out_rows_cap = (int64_t)o->CM + 1;
     for (int i = 0; i < o->n_out_alloc; i++)
@@ -1366,7 +1360,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
     hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
     hipMemset(o->ring.tag, 0xFF, (size_t)o->ring.R * 8);
-    hipMemset(o->ring.fill, 0, (size_t)o->ring.R * 8);
     hipMemset(o->ring.spec_used, 0, (size_t)o->ring.R * 4);
     hipMemset(o->ring.spec_state, 0, (size_t)o->ring.R * na * 16);
     hipMemset(o->ring.err, 0, 4);
@@ -1752,7 +1745,6 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
         M.agg = o->agg;
         M.m_keys = o->m_keys;
         M.m_state = o->m_state;
-        M.m_fill = o->m_fill;
         M.m_spec_used = o->m_spec_used;
         M.m_spec_state = o->m_spec_state;
         M.CM = o->CM;
@@ -2098,7 +2090,6 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->ring.state);
     hipFree(o->d_status);
     hipHostFree(o->h_status);
-    hipFree(o->ring.fill);
     hipFree(o->ring.spec_used);
     hipFree(o->ring.spec_state);
     hipFree(o->m_keys);
