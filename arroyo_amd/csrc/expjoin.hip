@@ -328,6 +328,10 @@ API void *arroyo_amd_expjoin_create(const AmdExpJoinConfig *cfg) {
         return nullptr;
     }
     hipMemset(o->d_err, 0, 4);
+    /* the device-resident ingest path accumulates matches from a cursor
+     * that must start at zero (the host path resets it per batch, which
+     * masked a reused allocation's stale value here) */
+    hipMemset(o->d_n_out, 0, 8);
     hipStreamCreate(&o->stream);
     o->stg_cap = 1 << 20;
     int max_in = 1 + (cfg->n_left_vals > cfg->n_right_vals
