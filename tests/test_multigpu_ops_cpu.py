@@ -1,0 +1,152 @@
+"""Multi-rank sharding semantics for the NON-window stateful operators
+(gloo, CPU): TTL join and updating aggregate behind the same keyed shuffle
+as the window path (tests/test_multigpu_cpu.py).
+
+Both operators partition by key exactly like the windowed aggregates: every
+key's state lives wholly on one shard (contiguous key-hash ranges,
+server_for_hash, crates/arroyo-types/src/lib.rs:640-647), joins only match
+equal keys and updating aggregates group by key, so the sharded run needs no
+data-path collective and must reproduce the single-instance result exactly
+(SURVEY.md §8e).  The per-shard operator here is the C oracle; on GPU the
+same layout runs one HIP operator per rank with the exchange over RCCL.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import oracle
+from arroyo_amd import cabi
+from arroyo_amd.shuffle import shuffle_columns
+
+NS = 10**9
+HOUR = 3600 * NS
+T0 = 1_600_000_000 * NS
+WORLD = 2
+
+
+def _rows(cols):
+    if cols is None or len(cols) == 0 or len(cols[0]) == 0:
+        return []
+    return [tuple(int(c[r]) for c in cols) for r in range(len(cols[0]))]
+
+
+def _gen_stream(seed, n):
+    rng = np.random.default_rng(seed)
+    key = rng.integers(0, 40, size=n).astype(np.int64)
+    val = rng.integers(0, 10**6, size=n).astype(np.int64)
+    ts = T0 + np.sort(rng.integers(0, 600, size=n)).astype(np.int64) * NS
+    return key, val, ts
+
+
+def _expjoin_run(op, batches):
+    got = []
+    for side, cols in batches:
+        got += _rows(op.process_batch(side, cols))
+    op.close()
+    return sorted(got)
+
+
+def _expjoin_rank_main(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+        24 * HOUR, n_left_vals=1, n_right_vals=1))
+    got = []
+    for step in range(6):
+        key, val, ts = _gen_stream(100 + step, 600)
+        side = step % 2
+        # this rank's slice of the source stream (parallel source subtasks)
+        mine = np.arange(len(key)) % world == rank
+        cols = shuffle_columns([key[mine], val[mine], ts[mine]], world)
+        if len(cols[0]):
+            got += _rows(op.process_batch(side, cols))
+    op.close()
+    result_q.put((rank, sorted(got)))
+    dist.destroy_process_group()
+
+
+def test_expjoin_two_rank_shuffle_matches_single():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_expjoin_rank_main,
+                         args=(r, WORLD, 29381, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, rows = q.get(timeout=300)
+        results[rank] = rows
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    # key-disjointness: no join key may appear on two ranks
+    assert not ({r[0] for r in results[0]} & {r[0] for r in results[1]})
+
+    single = oracle.make_expjoin_op(cabi.make_expjoin_config(
+        24 * HOUR, n_left_vals=1, n_right_vals=1))
+    batches = []
+    for step in range(6):
+        key, val, ts = _gen_stream(100 + step, 600)
+        batches.append((step % 2, [key, val, ts]))
+    want = _expjoin_run(single, batches)
+    assert sorted(results[0] + results[1]) == want
+    assert len(want) > 1000
+
+
+def _updagg_rank_main(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    op = oracle.make_updagg_op(cabi.make_updagg_config(
+        [(cabi.COUNT, -1), (cabi.SUM, 0)], n_keys=1, n_value_cols=1))
+    got = []
+    for step in range(5):
+        key, val, _ts = _gen_stream(200 + step, 800)
+        retract = np.zeros(len(key), dtype=np.int64)
+        mine = np.arange(len(key)) % world == rank
+        cols = shuffle_columns([key[mine], val[mine], retract[mine]], world)
+        if len(cols[0]):
+            op.process_batch(cols)
+        got += _rows(op.flush())
+    op.close()
+    result_q.put((rank, sorted(got)))
+    dist.destroy_process_group()
+
+
+def test_updagg_two_rank_shuffle_matches_single():
+    """Per-flush emission streams merge to the single-instance stream: each
+    key's retract/append history lives wholly on its owner rank, and flush
+    cadence is global (every batch), so the merged per-key final rows and
+    the full emission multiset must both match."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_updagg_rank_main,
+                         args=(r, WORLD, 29383, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, rows = q.get(timeout=300)
+        results[rank] = rows
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    assert not ({r[0] for r in results[0]} & {r[0] for r in results[1]})
+
+    single = oracle.make_updagg_op(cabi.make_updagg_config(
+        [(cabi.COUNT, -1), (cabi.SUM, 0)], n_keys=1, n_value_cols=1))
+    want = []
+    for step in range(5):
+        key, val, _ts = _gen_stream(200 + step, 800)
+        retract = np.zeros(len(key), dtype=np.int64)
+        single.process_batch([key, val, retract])
+        want += _rows(single.flush())
+    single.close()
+    assert sorted(results[0] + results[1]) == sorted(want)
+    assert len(want) > 100
