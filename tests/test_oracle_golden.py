@@ -138,3 +138,38 @@ def test_c_oracle_vs_pyoracle_fuzz(seed, tumbling):
         want.add((k[0], *[int(x) for x in fin], ws, we, ots))
     assert got == want
     cop.close()
+
+
+@pytest.mark.parametrize("seed", [11, 13])
+def test_c_oracle_avg_fuzz(seed):
+    """Same property fuzz with AVG in the agg set (f64 output column).
+    The C oracle and the pure-Python restatement both accumulate the AVG
+    partial as an IEEE double in arrival order, so values match to 1e-12
+    relative (1e-9 is the stated cross-device bar)."""
+    rng = np.random.default_rng(seed)
+    n = 2000
+    width, slide = 10 * NS, 2 * NS
+    base = np.cumsum(rng.integers(0, 50_000_000, size=n))
+    ts = (100 * NS + base).astype(np.int64)
+    key = rng.integers(0, 23, size=n).astype(np.int64)
+    val = rng.integers(-1000, 1000, size=n).astype(np.int64)
+    aggs = [(cabi.COUNT, -1), (cabi.AVG, 0)]
+    cop = make_oracle(width_ns=width, slide_ns=slide, n_keys=1,
+                      n_value_cols=1, aggs=aggs)
+    outs = run_stream(cop, batches_from_columns([key, val, ts], 89), NS)
+    cols = concat_outputs(outs)
+    got = {}
+    for r in range(len(cols[0])):
+        # key, count, avg(f64), window_start, window_end, ts
+        kk = (int(cols[0][r]), int(cols[1][r]), int(cols[3][r]),
+              int(cols[4][r]), int(cols[5][r]))
+        got[kk] = float(cols[2][r])
+    want = {}
+    for (k, fin, ws, we, ots) in _pyoracle_run(
+            width, slide, False, True, aggs, [key, val, ts], NS, 89):
+        want[(k[0], int(fin[0]), ws, we, ots)] = float(fin[1])
+    assert set(got) == set(want)
+    for kk, av in want.items():
+        assert got[kk] == pytest.approx(av, rel=1e-12)
+    assert len(want) > 500
+    cop.close()
