@@ -831,6 +831,7 @@ class WindowFnOp:
         self._fn["last_error"] = g("last_error")
         self._fn["last_error"].restype = ctypes.c_char_p
         self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self._fn["_lib_prefix"] = (lib, prefix)
         self.cfg = cfg
         self._h = self._fn["create"](ctypes.byref(cfg))
         if not self._h:
@@ -844,6 +845,21 @@ class WindowFnOp:
         keep, arr = _cols_to_ptrs(cols)
         self._check(self._fn["process_batch"](
             self._h, arr, len(keep), len(keep[0]) if keep else 0))
+
+    def process_batch_device(self, dptrs, n_rows):
+        """Device-resident ingest (GPU-only extension): dptrs are raw
+        device addresses of the same columns process_batch takes."""
+        if "process_batch_device" not in self._fn:
+            lib, prefix = self._fn["_lib_prefix"]
+            pd = getattr(lib, prefix + "windowfn_process_batch_device")
+            pd.restype = ctypes.c_int
+            pd.argtypes = [ctypes.c_void_p,
+                           ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+                           ctypes.c_int64]
+            self._fn["process_batch_device"] = pd
+        arr = (ctypes.c_void_p * len(dptrs))(*dptrs)
+        self._check(self._fn["process_batch_device"](self._h, arr,
+                                                     len(dptrs), n_rows))
 
     def handle_watermark(self, wm):
         out = AmdOutBatch()
@@ -954,10 +970,35 @@ class MapOp:
         self._fn["last_error"] = g("last_error")
         self._fn["last_error"].restype = ctypes.c_char_p
         self._fn["last_error"].argtypes = [ctypes.c_void_p]
+        self._fn["_lib_prefix"] = (lib, prefix)
         self.cfg = cfg
         self._h = self._fn["create"](ctypes.byref(cfg))
         if not self._h:
             raise RuntimeError(f"{p}create failed")
+
+    def process_batch_device(self, dptrs, n_rows):
+        """Device-resident map/filter (GPU-only extension): input device
+        addresses in, (device output addresses, surviving row count) out.
+        The returned addresses are owned by the operator and valid until
+        the next process_batch* call."""
+        if "process_batch_device" not in self._fn:
+            lib, prefix = self._fn["_lib_prefix"]
+            pd = getattr(lib, prefix + "map_process_batch_device")
+            pd.restype = ctypes.c_int
+            pd.argtypes = [ctypes.c_void_p,
+                           ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+                           ctypes.c_int64, ctypes.POINTER(ctypes.c_void_p),
+                           ctypes.POINTER(ctypes.c_int64)]
+            self._fn["process_batch_device"] = pd
+        arr = (ctypes.c_void_p * len(dptrs))(*dptrs)
+        outp = (ctypes.c_void_p * self.cfg.n_out)()
+        n_out = ctypes.c_int64(0)
+        rc = self._fn["process_batch_device"](self._h, arr, len(dptrs),
+                                              n_rows, outp,
+                                              ctypes.byref(n_out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        return [outp[i] for i in range(self.cfg.n_out)], int(n_out.value)
 
     def process_batch(self, cols):
         keep, arr = _cols_to_ptrs(cols)

@@ -307,6 +307,81 @@ static int map_run(GpuMap *o, const int64_t *const *cols, int32_t n_cols,
     return 0;
 }
 
+/* device-resident map/filter: input columns already in HBM; output stays
+ * in HBM.  Fills d_out_cols[i] with device addresses owned by the
+ * operator (valid until the next process_batch* call) and *n_out_rows
+ * with the surviving row count.  One launch, so n_rows must fit the
+ * operator's chunk capacity (the host path chunks instead). */
+API int arroyo_amd_map_process_batch_device(void *h,
+                                            const int64_t *const *dcols,
+                                            int32_t n_cols, int64_t n_rows,
+                                            const int64_t **d_out_cols,
+                                            int64_t *n_out_rows) {
+    GpuMap *o = (GpuMap *)h;
+    const AmdMapConfig &c = o->cfg;
+    if (n_cols != c.n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 c.n_in_cols, n_cols);
+        return 1;
+    }
+    if (n_rows > o->cap) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "device batch of %lld rows exceeds capacity %lld",
+                 (long long)n_rows, (long long)o->cap);
+        return 1;
+    }
+    MapArgs A = {};
+    for (int cc = 0; cc < n_cols; cc++) A.cols[cc] = dcols[cc];
+    A.n_in = n_cols;
+    A.n_rows = n_rows;
+    A.cfg = c;
+    A.keep = c.filter_reg >= 0 ? o->d_keep : nullptr;
+    for (int i = 0; i < c.n_out; i++) A.vals[i] = o->d_vals[i];
+    A.err = o->d_err;
+    hipLaunchKernelGGL(k_map_eval, dim3(map_grid(n_rows)), dim3(256), 0,
+                       o->stream, A);
+    MHIP(o, hipGetLastError());
+    int64_t n_keep = n_rows;
+    if (c.filter_reg >= 0 && n_rows > 0) {
+        size_t tmp = o->cub_bytes;
+        hipcub::DeviceScan::ExclusiveSum(o->cub_tmp, tmp, o->d_keep,
+                                         o->d_pos, (int)n_rows, o->stream);
+        hipLaunchKernelGGL(
+            k_map_scatter, dim3(map_grid(n_rows)), dim3(256), 0, o->stream,
+            o->d_keep, o->d_pos, o->d_vals[0],
+            c.n_out > 1 ? o->d_vals[1] : nullptr,
+            c.n_out > 2 ? o->d_vals[2] : nullptr,
+            c.n_out > 3 ? o->d_vals[3] : nullptr,
+            c.n_out > 4 ? o->d_vals[4] : nullptr,
+            c.n_out > 5 ? o->d_vals[5] : nullptr, o->d_out[0],
+            c.n_out > 1 ? o->d_out[1] : nullptr,
+            c.n_out > 2 ? o->d_out[2] : nullptr,
+            c.n_out > 3 ? o->d_out[3] : nullptr,
+            c.n_out > 4 ? o->d_out[4] : nullptr,
+            c.n_out > 5 ? o->d_out[5] : nullptr, c.n_out, n_rows);
+        MHIP(o, hipGetLastError());
+        int64_t tail[2] = {0, 0};
+        MHIP(o, hipMemcpyAsync(&tail[0], o->d_pos + (n_rows - 1), 8,
+                               hipMemcpyDeviceToHost, o->stream));
+        MHIP(o, hipMemcpyAsync(&tail[1], o->d_keep + (n_rows - 1), 8,
+                               hipMemcpyDeviceToHost, o->stream));
+        MHIP(o, hipStreamSynchronize(o->stream));
+        n_keep = tail[0] + tail[1];
+    }
+    int derr = 0;
+    MHIP(o, hipMemcpyAsync(&derr, o->d_err, 4, hipMemcpyDeviceToHost,
+                           o->stream));
+    MHIP(o, hipStreamSynchronize(o->stream));
+    if (derr) {
+        snprintf(o->err_msg, sizeof o->err_msg, "division by zero");
+        return 1;
+    }
+    for (int i = 0; i < c.n_out; i++)
+        d_out_cols[i] = c.filter_reg >= 0 ? o->d_out[i] : o->d_vals[i];
+    *n_out_rows = n_keep;
+    return 0;
+}
+
 API int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
                                      int32_t n_cols, int64_t n_rows,
                                      AmdOutBatch *out) {

@@ -392,6 +392,38 @@ API int arroyo_amd_windowfn_process_batch(void *h, const int64_t *const *cols,
     return 0;
 }
 
+/* device-resident ingest: rows already in HBM append straight into the
+ * instant table (no host staging); same arrival-sequence stamping */
+API int arroyo_amd_windowfn_process_batch_device(void *h,
+                                                 const int64_t *const *dcols,
+                                                 int32_t n_cols,
+                                                 int64_t n_rows) {
+    GpuWindowFn *o = (GpuWindowFn *)h;
+    if (n_cols != o->cfg.n_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->cfg.n_cols, n_cols);
+        return 1;
+    }
+    WfAppendArgs A = {};
+    for (int c = 0; c < n_cols; c++) A.cols[c] = dcols[c];
+    A.n_cols = n_cols;
+    A.n_rows = n_rows;
+    A.tag = o->tag;
+    A.cursor = o->cursor;
+    A.planes = o->planes;
+    A.I = o->I;
+    A.cap = o->cap;
+    A.has_wm = o->has_wm;
+    A.wm = o->wm;
+    A.seq_base = o->seq;
+    o->seq += n_rows;
+    A.err = o->d_err;
+    hipLaunchKernelGGL(k_wf_append, dim3(wf_grid(n_rows)), dim3(256), 0,
+                       o->stream, A);
+    WHIP(o, hipGetLastError());
+    return 0;
+}
+
 static int wf_fire(GpuWindowFn *o, uint64_t instant,
                    const std::vector<uint32_t> &slots,
                    const std::vector<unsigned long long> &cnt) {

@@ -280,6 +280,9 @@ const char *arroyo_amd_updagg_last_error(void *h);
 void *arroyo_amd_windowfn_create(const AmdWindowFnConfig *cfg);
 int arroyo_amd_windowfn_process_batch(void *h, const int64_t *const *cols,
                                       int32_t n_cols, int64_t n_rows);
+int arroyo_amd_windowfn_process_batch_device(void *h,
+                                             const int64_t *const *dcols,
+                                             int32_t n_cols, int64_t n_rows);
 int arroyo_amd_windowfn_handle_watermark(void *h, uint64_t watermark_nanos,
                                          AmdOutBatch *out);
 int arroyo_amd_windowfn_checkpoint_drain(void *h, AmdOutBatch *out);
@@ -300,6 +303,10 @@ void *arroyo_amd_map_create(const AmdMapConfig *cfg);
 int arroyo_amd_map_process_batch(void *h, const int64_t *const *cols,
                                  int32_t n_cols, int64_t n_rows,
                                  AmdOutBatch *out);
+int arroyo_amd_map_process_batch_device(void *h, const int64_t *const *dcols,
+                                        int32_t n_cols, int64_t n_rows,
+                                        const int64_t **d_out_cols,
+                                        int64_t *n_out_rows);
 void arroyo_amd_map_destroy(void *h);
 const char *arroyo_amd_map_last_error(void *h);
 

@@ -214,3 +214,35 @@ def test_windowfn_gpu_large_instant():
     op.close()
     want = np_row_numbers([part, v, ts], 0, [(1, True)], 3)
     assert got == want
+
+
+@pytest.mark.gpu
+def test_windowfn_gpu_device_resident_matches_host():
+    """process_batch_device (device-resident ingest surface) must leave the
+    instant table in the same state as the host path: identical fired rows,
+    including the arrival-sequence tiebreak (sequence stamping is shared)."""
+    import torch
+    from arroyo_amd import gpu
+    rng = np.random.default_rng(61)
+    t0 = 1_600_000_000 * NS
+    n = 3000
+    instants = t0 + np.arange(20, dtype=np.int64) * NS
+    part = rng.integers(0, 5, size=n).astype(np.int64)
+    v1 = rng.integers(0, 40, size=n).astype(np.int64)
+    ts = np.sort(rng.choice(instants, size=n)).astype(np.int64)
+    cols = [part, v1, ts]
+    kw = dict(n_cols=3, part_col=0, order=[(1, False)], log2_rows_cap=12,
+              instants=64)
+    h = gpu.make_windowfn_op(cabi.make_windowfn_config(**kw))
+    d = gpu.make_windowfn_op(cabi.make_windowfn_config(**kw))
+    dev = torch.device("cuda", 0)
+    h.process_batch(cols)
+    tens = [torch.from_numpy(c.copy()).to(dev) for c in cols]
+    torch.cuda.synchronize()
+    d.process_batch_device([t.data_ptr() for t in tens], n)
+    got = sorted(rows_of(d.handle_watermark(U64MAX)))
+    want = sorted(rows_of(h.handle_watermark(U64MAX)))
+    h.close()
+    d.close()
+    assert got == want
+    assert len(want) == n
