@@ -820,6 +820,11 @@ class WindowFnOp:
         self._fn["handle_watermark"].restype = ctypes.c_int
         self._fn["handle_watermark"].argtypes = [
             ctypes.c_void_p, ctypes.c_uint64, ctypes.POINTER(AmdOutBatch)]
+        self._fn["restore"] = g("restore")
+        self._fn["restore"].restype = ctypes.c_int
+        self._fn["restore"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32,
+            ctypes.c_int64]
         self._fn["checkpoint_drain"] = g("checkpoint_drain")
         self._fn["checkpoint_drain"].restype = ctypes.c_int
         self._fn["checkpoint_drain"].argtypes = [
@@ -860,6 +865,11 @@ class WindowFnOp:
         arr = (ctypes.c_void_p * len(dptrs))(*dptrs)
         self._check(self._fn["process_batch_device"](self._h, arr,
                                                      len(dptrs), n_rows))
+
+    def restore(self, cols):
+        keep, arr = _cols_to_ptrs(cols)
+        self._check(self._fn["restore"](
+            self._h, arr, len(keep), len(keep[0]) if keep else 0))
 
     def handle_watermark(self, wm):
         out = AmdOutBatch()

@@ -576,6 +576,15 @@ API int arroyo_amd_windowfn_handle_watermark(void *h, uint64_t wm,
     return 0;
 }
 
+/* restore = re-ingest the drained rows (state IS the raw buffered rows;
+ * drain preserves per-instant arrival order so fresh sequence stamps
+ * reproduce the ROW_NUMBER tiebreak), mirroring the reference's
+ * buffered-instant table restore (window_fn.rs ExpiringTimeKeyTable). */
+API int arroyo_amd_windowfn_restore(void *h, const int64_t *const *cols,
+                                    int32_t n_cols, int64_t n_rows) {
+    return arroyo_amd_windowfn_process_batch(h, cols, n_cols, n_rows);
+}
+
 API int arroyo_amd_windowfn_checkpoint_drain(void *h, AmdOutBatch *out) {
     GpuWindowFn *o = (GpuWindowFn *)h;
     if (wf_check_err(o)) return 1;

@@ -283,6 +283,8 @@ int arroyo_amd_windowfn_process_batch(void *h, const int64_t *const *cols,
 int arroyo_amd_windowfn_process_batch_device(void *h,
                                              const int64_t *const *dcols,
                                              int32_t n_cols, int64_t n_rows);
+int arroyo_amd_windowfn_restore(void *h, const int64_t *const *cols,
+                                int32_t n_cols, int64_t n_rows);
 int arroyo_amd_windowfn_handle_watermark(void *h, uint64_t watermark_nanos,
                                          AmdOutBatch *out);
 int arroyo_amd_windowfn_checkpoint_drain(void *h, AmdOutBatch *out);

@@ -2055,6 +2055,16 @@ ORACLE_API int oracle_windowfn_checkpoint_drain(void *h, AmdOutBatch *out) {
     return 0;
 }
 
+/* restore = re-ingest the drained rows: windowfn state IS the raw
+ * buffered rows (drain preserves per-instant arrival order, so the
+ * ROW_NUMBER arrival-sequence tiebreak is reproduced), mirroring
+ * WindowFunctionOperator's table restore of buffered instants
+ * (crates/arroyo-worker/src/arrow/window_fn.rs ExpiringTimeKeyTable). */
+ORACLE_API int oracle_windowfn_restore(void *h, const int64_t *const *cols,
+                                       int32_t n_cols, int64_t n_rows) {
+    return oracle_windowfn_process_batch(h, cols, n_cols, n_rows);
+}
+
 ORACLE_API void oracle_windowfn_destroy(void *h) {
     WfOp *o = h;
     if (!o) return;

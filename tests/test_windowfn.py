@@ -246,3 +246,50 @@ def test_windowfn_gpu_device_resident_matches_host():
     d.close()
     assert got == want
     assert len(want) == n
+
+
+def _wf_roundtrip(make_op):
+    """Checkpoint/restore roundtrip: drain a half-ingested op, restore into
+    a fresh one, finish the stream, and match the uninterrupted run
+    (including arrival-order ROW_NUMBER ties: drain preserves per-instant
+    arrival order)."""
+    rng = np.random.default_rng(71)
+    t0 = 1_600_000_000 * NS
+    n = 3000
+    instants = t0 + np.arange(16, dtype=np.int64) * NS
+    part = rng.integers(0, 5, size=n).astype(np.int64)
+    v1 = rng.integers(0, 8, size=n).astype(np.int64)  # many ties
+    ts = np.sort(rng.choice(instants, size=n)).astype(np.int64)
+    cols = [part, v1, ts]
+    kw = dict(n_cols=3, part_col=0, order=[(1, False)], log2_rows_cap=12,
+              instants=64)
+
+    a = make_op(cabi.make_windowfn_config(**kw))
+    half = n // 2
+    a.process_batch([c[:half] for c in cols])
+    drained = a.checkpoint_drain()
+    a.close()
+    assert len(drained[0]) == half
+
+    b = make_op(cabi.make_windowfn_config(**kw))
+    b.restore(drained)
+    b.process_batch([c[half:] for c in cols])
+    got = sorted(rows_of(b.handle_watermark(U64MAX)))
+    b.close()
+
+    c_ = make_op(cabi.make_windowfn_config(**kw))
+    c_.process_batch(cols)
+    want = sorted(rows_of(c_.handle_watermark(U64MAX)))
+    c_.close()
+    assert got == want
+    assert len(want) == n
+
+
+def test_windowfn_oracle_checkpoint_restore_roundtrip():
+    _wf_roundtrip(oracle.make_windowfn_op)
+
+
+@pytest.mark.gpu
+def test_windowfn_gpu_checkpoint_restore_roundtrip():
+    from arroyo_amd import gpu
+    _wf_roundtrip(gpu.make_windowfn_op)
