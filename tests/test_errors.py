@@ -1,0 +1,106 @@
+"""API error paths: invalid configs are rejected at create and argument
+mismatches return errors through last_error — never crashes or silent
+acceptance.  Mirrors the reference's planner/operator validation behavior
+(e.g. arroyo-planner/src/lib.rs:644 rejects hop widths not divisible by the
+slide).  CPU (oracle) versions here; the GPU library shares the same
+validation code paths (create-time checks run before any HIP call)."""
+import numpy as np
+import pytest
+
+import oracle
+from arroyo_amd import cabi
+
+NS = 10**9
+
+
+def test_window_create_rejects_zero_width():
+    with pytest.raises(RuntimeError):
+        oracle.make_op(cabi.make_config(
+            width_ns=0, slide_ns=NS, n_keys=1, aggs=[(cabi.COUNT, -1)]))
+
+
+def test_window_create_rejects_nondivisible_hop():
+    # planner: "hop() width must be evenly divisible by the slide"
+    with pytest.raises(RuntimeError):
+        oracle.make_op(cabi.make_config(
+            width_ns=10 * NS, slide_ns=3 * NS, n_keys=1,
+            aggs=[(cabi.COUNT, -1)]))
+
+
+def test_window_create_rejects_zero_slide():
+    with pytest.raises(RuntimeError):
+        oracle.make_op(cabi.make_config(
+            width_ns=10 * NS, slide_ns=0, n_keys=1,
+            aggs=[(cabi.COUNT, -1)]))
+
+
+def test_window_process_batch_wrong_cols():
+    op = oracle.make_op(cabi.make_config(
+        width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=1,
+        aggs=[(cabi.SUM, 0)]))
+    with pytest.raises(RuntimeError):
+        # expects [key, value, ts]; hand only [key, ts]
+        op.process_batch([np.zeros(4, dtype=np.int64),
+                          np.zeros(4, dtype=np.int64)])
+    op.close()
+
+
+def test_session_create_rejects_zero_gap():
+    with pytest.raises(RuntimeError):
+        oracle.make_session_op(cabi.make_session_config(
+            0, [(cabi.COUNT, -1)], n_keys=1))
+
+
+def test_session_create_rejects_no_aggs():
+    with pytest.raises(RuntimeError):
+        oracle.make_session_op(cabi.make_session_config(
+            5 * NS, [], n_keys=1))
+
+
+def test_expjoin_create_rejects_zero_ttl():
+    with pytest.raises(RuntimeError):
+        oracle.make_expjoin_op(cabi.make_expjoin_config(0))
+
+
+def test_expjoin_wrong_side_cols():
+    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+        3600 * NS, n_left_vals=2, n_right_vals=0))
+    with pytest.raises(RuntimeError):
+        # left expects [key, v0, v1, ts]
+        op.process_batch(op.LEFT, [np.zeros(2, dtype=np.int64),
+                                   np.zeros(2, dtype=np.int64)])
+    op.close()
+
+
+def test_mapop_create_rejects_bad_register():
+    with pytest.raises(RuntimeError):
+        oracle.make_map_op(cabi.make_map_config(
+            n_in_cols=1, prog=[(cabi.MOP_ADD, 0, 31, 40)], out_reg=[0]))
+
+
+def test_mapop_div_by_zero_is_error():
+    op = oracle.make_map_op(cabi.make_map_config(
+        n_in_cols=2, prog=[(cabi.MOP_DIV, 0, 1, 2)], out_reg=[2]))
+    with pytest.raises(RuntimeError):
+        op.process_batch([np.array([4], dtype=np.int64),
+                          np.array([0], dtype=np.int64)])
+    op.close()
+
+
+def test_windowfn_wrong_cols():
+    op = oracle.make_windowfn_op(cabi.make_windowfn_config(
+        n_cols=3, part_col=0, order=[(1, False)]))
+    with pytest.raises(RuntimeError):
+        op.process_batch([np.zeros(2, dtype=np.int64),
+                          np.zeros(2, dtype=np.int64)])
+    op.close()
+
+
+def test_updagg_wrong_cols():
+    op = oracle.make_updagg_op(cabi.make_updagg_config(
+        [(cabi.COUNT, -1)], n_keys=1, n_value_cols=1))
+    with pytest.raises(RuntimeError):
+        # expects [key, value, is_retract]
+        op.process_batch([np.zeros(3, dtype=np.int64),
+                          np.zeros(3, dtype=np.int64)])
+    op.close()
