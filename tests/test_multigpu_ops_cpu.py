@@ -150,3 +150,132 @@ def test_updagg_two_rank_shuffle_matches_single():
     single.close()
     assert sorted(results[0] + results[1]) == sorted(want)
     assert len(want) > 100
+
+
+def _session_rank_main(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    op = oracle.make_session_op(cabi.make_session_config(
+        5 * NS, [(cabi.COUNT, -1), (cabi.SUM, 0)], n_keys=1,
+        n_value_cols=1))
+    got = []
+    for step in range(6):
+        key, val, ts = _gen_stream(300 + step, 700)
+        # bursty: cluster timestamps so sessions form and close
+        ts = T0 + (step * 120 + (ts - T0) // (20 * NS)) * NS
+        mine = np.arange(len(key)) % world == rank
+        cols = shuffle_columns([key[mine], val[mine], ts[mine]], world)
+        if len(cols[0]):
+            op.process_batch(cols)
+        got += _rows(op.handle_watermark(int(ts.max()) - NS))
+    got += _rows(op.handle_watermark(2**64 - 1))
+    op.close()
+    result_q.put((rank, sorted(got)))
+    dist.destroy_process_group()
+
+
+def test_session_two_rank_shuffle_matches_single():
+    """Session windows shard by key like everything else: all of a key's
+    rows land on its owner rank, so session extents and aggregates are
+    computed whole there and the merged firings equal the single run."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_session_rank_main,
+                         args=(r, WORLD, 29387, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, rows = q.get(timeout=300)
+        results[rank] = rows
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    assert not ({r[0] for r in results[0]} & {r[0] for r in results[1]})
+
+    single = oracle.make_session_op(cabi.make_session_config(
+        5 * NS, [(cabi.COUNT, -1), (cabi.SUM, 0)], n_keys=1,
+        n_value_cols=1))
+    want = []
+    for step in range(6):
+        key, val, ts = _gen_stream(300 + step, 700)
+        ts = T0 + (step * 120 + (ts - T0) // (20 * NS)) * NS
+        single.process_batch([key, val, ts])
+        want += _rows(single.handle_watermark(int(ts.max()) - NS))
+    want += _rows(single.handle_watermark(2**64 - 1))
+    single.close()
+    assert sorted(results[0] + results[1]) == sorted(want)
+    assert len(want) > 50
+
+
+def _join_quantize(step, ts):
+    """30s instants within a step, steps in disjoint 900s blocks so no
+    batch falls behind the previous step's watermark (the instant join
+    errors on pre-watermark rows, matching the reference's buffering
+    contract)."""
+    return T0 + (step * 900 + ((ts - T0) // (30 * NS)) * 30) * NS
+
+
+def _join_rank_main(rank, world, port, result_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        op = oracle.make_join_op(cabi.make_join_config(
+            n_keys=1, n_left_vals=1, n_right_vals=1))
+        got = []
+        for step in range(4):
+            lk, lv, lt = _gen_stream(400 + step, 400)
+            rk, rv, rt = _gen_stream(500 + step, 400)
+            lt = _join_quantize(step, lt)
+            rt = _join_quantize(step, rt)
+            for side, (k, v, t) in ((0, (lk, lv, lt)), (1, (rk, rv, rt))):
+                mine = np.arange(len(k)) % world == rank
+                cols = shuffle_columns([k[mine], v[mine], t[mine]], world)
+                if len(cols[0]):
+                    op.process_batch(side, cols)
+            got += _rows(op.handle_watermark(
+                int(max(lt.max(), rt.max())) + NS))
+        op.close()
+        result_q.put((rank, sorted(got)))
+        dist.destroy_process_group()
+    except Exception as e:  # surface child failures instead of hanging
+        result_q.put((rank, e))
+        raise
+
+
+def test_instant_join_two_rank_shuffle_matches_single():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_join_rank_main,
+                         args=(r, WORLD, 29389, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, rows = q.get(timeout=120)
+        assert not isinstance(rows, Exception), rows
+        results[rank] = rows
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    assert not ({r[0] for r in results[0]} & {r[0] for r in results[1]})
+
+    single = oracle.make_join_op(cabi.make_join_config(
+        n_keys=1, n_left_vals=1, n_right_vals=1))
+    want = []
+    for step in range(4):
+        lk, lv, lt = _gen_stream(400 + step, 400)
+        rk, rv, rt = _gen_stream(500 + step, 400)
+        lt = _join_quantize(step, lt)
+        rt = _join_quantize(step, rt)
+        single.process_batch(0, [lk, lv, lt])
+        single.process_batch(1, [rk, rv, rt])
+        want += _rows(single.handle_watermark(
+            int(max(lt.max(), rt.max())) + NS))
+    single.close()
+    assert sorted(results[0] + results[1]) == sorted(want)
+    assert len(want) > 500
