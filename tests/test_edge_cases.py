@@ -114,9 +114,35 @@ def check_map_edges(m):
     mp.close()
 
 
+def check_watermark_idempotence(m):
+    """Re-delivering the same watermark (WatermarkHolder can forward the
+    min repeatedly) fires nothing new, and a REGRESSED watermark is a
+    no-op: already-fired panes never re-fire or error."""
+    op = m.make_op(cabi.make_config(width_ns=10 * NS, slide_ns=2 * NS,
+                                    aggs=[(cabi.COUNT, -1)], n_keys=1,
+                                    log2_capacity=12))
+    op.process_batch([one(7), one(T0 + 20 * NS)])
+    out = op.handle_watermark(T0 + 40 * NS)
+    assert len(out[0]) == 5
+    for wm in (T0 + 40 * NS, T0 + 10 * NS):  # repeat, then regress
+        out = op.handle_watermark(wm)
+        assert out is None or len(out[0]) == 0
+    op.close()
+
+    sp = m.make_session_op(cabi.make_session_config(
+        2 * NS, [(cabi.COUNT, -1)], n_keys=1))
+    sp.process_batch([one(9), one(T0 + NS)])
+    out = sp.handle_watermark(T0 + 60 * NS)
+    assert len(out[0]) == 1
+    for wm in (T0 + 60 * NS, T0 + 5 * NS):
+        out = sp.handle_watermark(wm)
+        assert out is None or len(out[0]) == 0
+    sp.close()
+
+
 ALL = [check_window_edges, check_join_edges, check_session_edges,
        check_expjoin_edges, check_updagg_edges, check_windowfn_edges,
-       check_map_edges]
+       check_map_edges, check_watermark_idempotence]
 
 
 @pytest.mark.parametrize("check", ALL, ids=lambda f: f.__name__)
