@@ -330,12 +330,31 @@ k_stream_sum(const ulonglong2 *a, const ulonglong2 *b, int64_t n2,
 
 #define PANE_UNSET 0xFFFFFFFFu
 
+/* wavefront same-slot combining (ARROYO_AMD_WAVECMB=1, COUNT-only path):
+ * group the wave's lanes by resolved LDS slot and issue ONE atomicAdd of
+ * the group's popcount per distinct slot, instead of up to 64 serialized
+ * same-address LDS atomics when a hot key dominates the wave.  The loop
+ * runs (distinct slots in wave) iterations of ballot+shfl. */
+__device__ inline void wave_count_combine(int slot, uint64_t *ls_st) {
+    unsigned long long act = __ballot(slot >= 0);
+    const unsigned lane = threadIdx.x & 63u;
+    while (act) {
+        int lead = __ffsll((unsigned long long)act) - 1;
+        int ls = __shfl(slot, lead, 64);
+        unsigned long long grp = __ballot(slot == ls) & act;
+        if (slot == ls && lane == (unsigned)lead)
+            atomicAdd((unsigned long long *)(ls_st + (size_t)ls * 2),
+                      (unsigned long long)__popcll(grp));
+        act &= ~grp;
+    }
+}
+
 /* per-row body shared by the scalar and vectorized LDS kernels.
  * COUNT_ONLY specializes the hot q5 shape (single COUNT(*)) away from the
  * aggregate-spec loop; last_key/last_slot cache skips hash+probe when
  * consecutive rows repeat a key (the nexmark hot auction makes runs
  * common). */
-template <int SLOTS, bool COUNT_ONLY>
+template <int SLOTS, bool COUNT_ONLY, bool WAVECMB = false>
 __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                                       uint32_t *ls_pane, uint64_t *ls_st,
                                       int64_t row, uint64_t traw, int64_t key,
@@ -360,18 +379,23 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
     if (A.mode == 2) return;
     const int na2 = COUNT_ONLY ? 1 : na;
     (void)na2;
+    int cslot = -1;  /* WAVECMB: resolved LDS slot, combined below */
     /* same (key, pane) as the previous row: reuse the cached LDS slot */
     if (key == last_key && last_slot != PANE_UNSET &&
         ls_pane[last_slot] == p) {
         if (COUNT_ONLY) {
-            atomicAdd((unsigned long long *)
-                          (ls_st + (size_t)last_slot * 2), 1ULL);
-            return;
+            if (WAVECMB) {
+                cslot = (int)last_slot;
+            } else {
+                atomicAdd((unsigned long long *)
+                              (ls_st + (size_t)last_slot * 2), 1ULL);
+                return;
+            }
         }
     }
     /* try the LDS table first, fall through to global */
     bool done = false;
-    if (key != EMPTY_KEY) {
+    if (!(WAVECMB && cslot >= 0) && key != EMPTY_KEY) {
         uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e37u + p) &
                      (SLOTS - 1);
         for (int pr = 0; pr < 4 && !done; pr++) {
@@ -395,7 +419,11 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
             if (k == key && (claimed || ls_pane[s] == p)) {
                 uint64_t *st = ls_st + (size_t)s * na * 2;
                 if (COUNT_ONLY) {
-                    atomicAdd((unsigned long long *)st, 1ULL);
+                    if (WAVECMB) {
+                        cslot = (int)s;
+                    } else {
+                        atomicAdd((unsigned long long *)st, 1ULL);
+                    }
                     last_key = key;
                     last_slot = s;
                     done = true;
@@ -428,6 +456,10 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                 done = true;
             }
         }
+    }
+    if (WAVECMB && COUNT_ONLY) {
+        wave_count_combine(cslot, ls_st);
+        if (cslot >= 0) return;
     }
     if (!done) {
         uint64_t *st;
@@ -499,7 +531,7 @@ k_update_lds(UpdateArgs A) {
  * per thread per iteration -- requires 16 B-aligned column pointers (host
  * checks).  Fewer, fatter waves: wave-dispatch cost was measurable at one
  * 8 B load per thread. */
-template <int SLOTS, bool COUNT_ONLY>
+template <int SLOTS, bool COUNT_ONLY, bool WAVECMB = false>
 __global__ void __launch_bounds__(256)
 k_update_lds_vec(UpdateArgs A) {
     __shared__ int64_t  ls_key[SLOTS];
@@ -520,16 +552,16 @@ k_update_lds_vec(UpdateArgs A) {
             k0 = (int64_t)kv.x;
             k1 = (int64_t)kv.y;
         }
-        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st, 2 * v,
-                                          tsv.x, k0, local_min, last_bin,
-                                          last_key, last_slot);
-        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st,
-                                          2 * v + 1, tsv.y, k1, local_min,
-                                          last_bin, last_key, last_slot);
+        lds_update_row<SLOTS, COUNT_ONLY, WAVECMB>(
+            A, ls_key, ls_pane, ls_st, 2 * v, tsv.x, k0, local_min,
+            last_bin, last_key, last_slot);
+        lds_update_row<SLOTS, COUNT_ONLY, WAVECMB>(
+            A, ls_key, ls_pane, ls_st, 2 * v + 1, tsv.y, k1, local_min,
+            last_bin, last_key, last_slot);
     }
     if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
-        lds_update_row<SLOTS, COUNT_ONLY>(A, ls_key, ls_pane, ls_st,
-                                          A.n_rows - 1,
+        lds_update_row<SLOTS, COUNT_ONLY, WAVECMB>(A, ls_key, ls_pane,
+                                          ls_st, A.n_rows - 1,
                                           (uint64_t)A.ts_col[A.n_rows - 1],
                                           A.key_col ? A.key_col[A.n_rows - 1]
                                                     : 0,
@@ -544,6 +576,8 @@ template __global__ void k_update_lds_vec<1024, false>(UpdateArgs);
 template __global__ void k_update_lds_vec<2048, false>(UpdateArgs);
 template __global__ void k_update_lds_vec<1024, true>(UpdateArgs);
 template __global__ void k_update_lds_vec<2048, true>(UpdateArgs);
+template __global__ void k_update_lds_vec<1024, true, true>(UpdateArgs);
+template __global__ void k_update_lds_vec<2048, true, true>(UpdateArgs);
 
 /* ------------------------------------------------------------------ */
 /* Radix-partitioned update path (ARROYO_AMD_RADIX=1, keyed
@@ -1214,6 +1248,7 @@ struct GpuOp {
     int force_blocks;
     int kmode;
     int use_vec;
+    int use_wavecmb = 0;
     int use_events;
     /* perf counters for bench; a fixed pool of reusable event pairs samples
      * a subset of launches (create/destroy per launch was host overhead) */
@@ -1312,6 +1347,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (const char *e = getenv("ARROYO_AMD_KMODE")) o->kmode = atoi(e);
     o->use_vec = 1;
     if (const char *e = getenv("ARROYO_AMD_VEC")) o->use_vec = atoi(e);
+    if (const char *e = getenv("ARROYO_AMD_WAVECMB")) o->use_wavecmb = atoi(e);
     o->use_events = 1;
     if (const char *e = getenv("ARROYO_AMD_EVENTS")) o->use_events = atoi(e);
 
@@ -1517,13 +1553,16 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     bool count_only = o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
                       o->kmode == 0;
     if (vec) {
+        bool cmb = count_only && o->use_wavecmb;
         if (slots >= 2048)
-            hipLaunchKernelGGL((count_only
+            hipLaunchKernelGGL((cmb ? k_update_lds_vec<2048, true, true>
+                                : count_only
                                     ? k_update_lds_vec<2048, true>
                                     : k_update_lds_vec<2048, false>),
                                dim3(blocks), dim3(256), shmem, o->stream, A);
         else
-            hipLaunchKernelGGL((count_only
+            hipLaunchKernelGGL((cmb ? k_update_lds_vec<1024, true, true>
+                                : count_only
                                     ? k_update_lds_vec<1024, true>
                                     : k_update_lds_vec<1024, false>),
                                dim3(blocks), dim3(256), shmem, o->stream, A);
