@@ -106,7 +106,8 @@ def main():
                  "windowed_inner_join", "session_window",
                  "global_session_window", "updating_inner_join",
                  "debezium_agg", "filter_updating_aggregates",
-                 "aggregates", "grouped_aggregates", "every_aggregate"):
+                 "aggregates", "grouped_aggregates", "every_aggregate",
+                 "month_loose_watermark"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)

@@ -21,9 +21,16 @@ def test_window_create_rejects_zero_width():
 
 def test_window_create_rejects_nondivisible_hop():
     # planner: "hop() width must be evenly divisible by the slide"
+    # (arroyo-planner/src/lib.rs:642-648; the reference pins this itself:
+    # most_active_driver_last_hour_unaligned.sql is a --fail test expecting
+    # "hop() width 3600s currently must be a multiple of slide 2400s")
     with pytest.raises(RuntimeError):
         oracle.make_op(cabi.make_config(
             width_ns=10 * NS, slide_ns=3 * NS, n_keys=1,
+            aggs=[(cabi.COUNT, -1)]))
+    with pytest.raises(RuntimeError):  # the reference's exact fail case
+        oracle.make_op(cabi.make_config(
+            width_ns=3600 * NS, slide_ns=2400 * NS, n_keys=1,
             aggs=[(cabi.COUNT, -1)]))
 
 

@@ -161,6 +161,17 @@ def test_golden_vectors_on_gpu():
     assert_rows_match(got, load_golden("hourly_by_event_type"))
     g.close()
 
+    # month_loose_watermark.sql: tumble(30 days) unkeyed COUNT, 1-minute
+    # watermark lateness
+    g = gpu_op(width_ns=30 * 86400 * NS, slide_ns=0, is_tumbling=True,
+               n_keys=0, n_value_cols=0, aggs=[(cabi.COUNT, -1)],
+               log2_capacity=10)
+    outs = run_stream(g, batches_from_columns([ts], 32), 60 * NS)
+    cnt, ws, we, _ = concat_outputs(outs)
+    got = [{"month": fmt_ts(s), "count": int(c)} for c, s in zip(cnt, ws)]
+    assert_rows_match(got, load_golden("month_loose_watermark"))
+    g.close()
+
 
 def test_checkpoint_drain_restore_roundtrip():
     """Drain mid-stream, restore into a fresh operator, finish the stream;

@@ -67,6 +67,21 @@ def test_tight_watermark():
     op.close()
 
 
+def test_month_loose_watermark():
+    # month_loose_watermark.sql: tumble(30 days) unkeyed COUNT over cars
+    # with `watermark AS (timestamp - INTERVAL '1 minute')` -- one 30-day
+    # bin (epoch-aligned: 2023-08-21 = 653 x 30d) holding all 7932 rows.
+    inp = load_inputs()["cars"]
+    ts = np.array(inp["ts"], dtype=np.int64)
+    op = make_oracle(width_ns=30 * 86400 * NS, slide_ns=0, is_tumbling=True,
+                     n_keys=0, n_value_cols=0, aggs=[(cabi.COUNT, -1)])
+    outs = run_stream(op, batches_from_columns([ts], 32), 60 * NS)
+    cnt, ws, we, _ = concat_outputs(outs)
+    got = [{"month": fmt_ts(s), "count": int(c)} for c, s in zip(cnt, ws)]
+    assert_rows_match(got, load_golden("month_loose_watermark"))
+    op.close()
+
+
 def test_most_active_driver_last_hour():
     # most_active_driver_last_hour.sql: hop(1min, 1h) COUNT GROUP BY
     # driver_id over cars.json, watermark = ts - 1h; the ROW_NUMBER()=1
