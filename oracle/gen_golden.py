@@ -107,7 +107,8 @@ def main():
                  "global_session_window", "updating_inner_join",
                  "debezium_agg", "filter_updating_aggregates",
                  "aggregates", "grouped_aggregates", "every_aggregate",
-                 "month_loose_watermark"):
+                 "month_loose_watermark", "reinvoke_window_function",
+                 "active_drivers"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)

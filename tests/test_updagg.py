@@ -483,3 +483,57 @@ def test_updagg_gpu_device_resident_matches_host():
     d.close()
     assert merge_debezium(got) == merge_debezium(want)
     assert sorted(got) == sorted(want)
+
+
+def run_active_drivers_golden(make_op, flush_every=9):
+    """active_drivers.sql: nested updating aggregates —
+    count(*) per driver -> HAVING count(*) > 85 -> count(*) of survivors.
+    The HAVING filter on the updating stream passes appends/retracts of
+    rows whose count exceeds 85 straight through, so the outer COUNT
+    tracks the live set exactly.  Pinned to the reference's Debezium
+    golden merged by its --pk=drivers upsert semantics."""
+    d = load_inputs()["cars"]
+    driver = np.array(d["driver_id"], dtype=np.int64)
+    n = len(driver)
+    zeros = np.zeros(n, dtype=np.int64)
+
+    inner = make_op(cabi.make_updagg_config([(cabi.COUNT, -1)], n_keys=1,
+                                            n_value_cols=0))
+    outer = make_op(cabi.make_updagg_config([(cabi.COUNT, -1)], n_keys=0,
+                                            n_value_cols=0))
+    out_emissions = []
+    step = max(1, n // flush_every)
+    for b in range(0, n, step):
+        sl = slice(b, min(b + step, n))
+        inner.process_batch([driver[sl], zeros[sl]])
+        fed = [r[-1] for r in rows_of(inner.flush()) if r[1] > 85]
+        if fed:
+            outer.process_batch([np.array(fed, dtype=np.int64)])
+        out_emissions += rows_of(outer.flush())
+    inner.close()
+    outer.close()
+
+    state = merge_debezium(out_emissions, n_keys=0)
+    got = [v[0] for v in state.values()]
+
+    # replay the golden's Debezium stream with its pk=drivers upserts
+    live = []
+    for r in load_golden("active_drivers"):
+        if r["op"] == "c":
+            live.append(r["after"]["drivers"])
+        elif r["op"] == "u":
+            live.remove(r["before"]["drivers"])
+            live.append(r["after"]["drivers"])
+        elif r["op"] == "d":
+            live.remove(r["before"]["drivers"])
+    assert got == live and len(live) == 1
+
+
+def test_active_drivers_golden_oracle():
+    run_active_drivers_golden(oracle.make_updagg_op)
+
+
+@pytest.mark.gpu
+def test_active_drivers_golden_gpu():
+    from arroyo_amd import gpu
+    run_active_drivers_golden(gpu.make_updagg_op)
