@@ -118,12 +118,16 @@ class AmdJoinConfig(ctypes.Structure):
         ("log2_out_cap", ctypes.c_uint32),
         ("device", ctypes.c_int32),
         ("emit_to_host", ctypes.c_int32),
+        ("join_type", ctypes.c_int32),
     ]
+
+
+JOIN_INNER, JOIN_LEFT, JOIN_RIGHT, JOIN_FULL = 0, 1, 2, 3
 
 
 def make_join_config(n_keys=1, n_left_vals=0, n_right_vals=0,
                      log2_rows_cap=15, instants=128, log2_out_cap=20,
-                     device=0, emit_to_host=True):
+                     device=0, emit_to_host=True, join_type=JOIN_INNER):
     cfg = AmdJoinConfig()
     cfg.n_keys = n_keys
     cfg.n_left_vals = n_left_vals
@@ -133,6 +137,7 @@ def make_join_config(n_keys=1, n_left_vals=0, n_right_vals=0,
     cfg.log2_out_cap = log2_out_cap
     cfg.device = device
     cfg.emit_to_host = 1 if emit_to_host else 0
+    cfg.join_type = join_type
     return cfg
 
 

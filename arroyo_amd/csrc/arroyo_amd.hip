@@ -2385,6 +2385,7 @@ k_join_build(JBuildArgs B) {
 
 struct JProbeArgs {
     const int64_t *r_key;     /* right keys of this slot, [nr] */
+    const int64_t *l_key;     /* left key plane base (global row index) */
     const int64_t *l_vals;    /* [n_left_vals][I*cap] base of left side */
     const int64_t *r_vals;    /* [n_right_vals][I*cap] base of right side */
     size_t l_off, r_off;      /* slot*cap element offset */
@@ -2395,6 +2396,8 @@ struct JProbeArgs {
     const int32_t *b_next;
     uint32_t H;
     int32_t n_keys, nlv, nrv;
+    int32_t jt;               /* AMD_JOIN_* */
+    uint32_t *l_hit;          /* [plane/32] matched-left bitmap (outer) */
     uint64_t instant;
     int64_t *out[16];
     unsigned long long *n_out;
@@ -2402,22 +2405,33 @@ struct JProbeArgs {
     int *err;
 };
 
+/* li/ri == -1: that side absent (outer join) -> zero-filled values,
+ * presence flag 0.  li is a GLOBAL row index when the chain map is in
+ * play (l_off = 0), slot-relative otherwise. */
 __device__ inline void jemit_row(const JProbeArgs &P, int64_t key, int64_t li,
                                  int64_t ri, int64_t r) {
     if (r >= P.out_cap) { *P.err = JERR_OUT_CAP; return; }
     int col = 0;
     if (P.n_keys) P.out[col++][r] = key;
     for (int v = 0; v < P.nlv; v++)
-        P.out[col++][r] = P.l_vals[(size_t)v * P.plane + P.l_off + li];
+        P.out[col++][r] =
+            li >= 0 ? P.l_vals[(size_t)v * P.plane + P.l_off + li] : 0;
     for (int v = 0; v < P.nrv; v++)
-        P.out[col++][r] = P.r_vals[(size_t)v * P.plane + P.r_off + ri];
-    P.out[col][r] = (int64_t)P.instant;
+        P.out[col++][r] =
+            ri >= 0 ? P.r_vals[(size_t)v * P.plane + P.r_off + ri] : 0;
+    P.out[col++][r] = (int64_t)P.instant;
+    if (P.jt != AMD_JOIN_INNER) {
+        P.out[col++][r] = li >= 0;
+        P.out[col][r] = ri >= 0;
+    }
 }
 
 __global__ void __launch_bounds__(256)
 k_join_probe(JProbeArgs P) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint32_t m = P.H - 1;
+    const bool emit_r = P.jt == AMD_JOIN_RIGHT || P.jt == AMD_JOIN_FULL;
+    const bool track_l = P.l_hit != nullptr;
     for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          j < P.nr; j += stride) {
         int64_t key = P.r_key[j];
@@ -2429,10 +2443,34 @@ k_join_probe(JProbeArgs P) {
             if (k == EMPTY_KEY) break;
             s = (s + 1) & m;
         }
+        bool hit = false;
         for (int32_t li = head; li >= 0; li = P.b_next[li]) {
             int64_t r = (int64_t)atomicAdd(P.n_out, 1ULL);
             jemit_row(P, key, li, j, r);
+            hit = true;
+            if (track_l)
+                atomicOr(&P.l_hit[(uint32_t)li >> 5], 1u << ((uint32_t)li & 31u));
         }
+        if (!hit && emit_r) {
+            int64_t r = (int64_t)atomicAdd(P.n_out, 1ULL);
+            jemit_row(P, key, -1, j, r);
+        }
+    }
+}
+
+/* outer-join left pass: emit every left row of one slot whose matched bit
+ * is unset ([base, base+nl) global row indices); with a zeroed bitmap this
+ * doubles as the pad-every-left-row kernel for empty-right instants */
+__global__ void __launch_bounds__(256)
+k_join_unmatched_left(JProbeArgs P, int64_t base, int64_t nl) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < nl; i += stride) {
+        int64_t li = base + i;
+        if (P.l_hit[(uint32_t)li >> 5] & (1u << ((uint32_t)li & 31u)))
+            continue;
+        int64_t r = (int64_t)atomicAdd(P.n_out, 1ULL);
+        jemit_row(P, P.n_keys ? P.l_key[li] : 0, li, -1, r);
     }
 }
 
@@ -2457,6 +2495,7 @@ struct GpuJoin {
     int64_t *key[2];
     int64_t *vals[2];          /* [n_vals][I*cap] */
     int64_t *b_keys; int32_t *b_head; int32_t *b_next;
+    uint32_t *l_hit;           /* [I*cap/32] matched-left bitmap (outer) */
     int64_t *d_out[16];
     unsigned long long *d_n_out;
     int *d_err;
@@ -2484,7 +2523,8 @@ struct GpuJoin {
 API void *arroyo_amd_join_create(const AmdJoinConfig *cfg) {
     if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->n_left_vals < 0 ||
         cfg->n_right_vals < 0 || cfg->n_left_vals > 4 ||
-        cfg->n_right_vals > 4) {
+        cfg->n_right_vals > 4 || cfg->join_type < 0 ||
+        cfg->join_type > AMD_JOIN_FULL) {
         snprintf(g_err, sizeof g_err, "invalid join config");
         return nullptr;
     }
@@ -2494,7 +2534,8 @@ API void *arroyo_amd_join_create(const AmdJoinConfig *cfg) {
     o->cap = 1u << (cfg->log2_rows_cap ? cfg->log2_rows_cap : 15);
     o->H = o->cap * 2;
     o->out_cap = 1ll << (cfg->log2_out_cap ? cfg->log2_out_cap : 20);
-    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1;
+    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1 +
+                  (cfg->join_type != AMD_JOIN_INNER ? 2 : 0);
     if (hipSetDevice(cfg->device) != hipSuccess) {
         snprintf(g_err, sizeof g_err,
                  "hipSetDevice(%d) failed: no HIP device (no CPU fallback)",
@@ -2522,6 +2563,8 @@ API void *arroyo_amd_join_create(const AmdJoinConfig *cfg) {
     JALLOC(o->b_keys, (size_t)o->H * 8);
     JALLOC(o->b_head, (size_t)o->H * 4);
     JALLOC(o->b_next, (size_t)o->I * o->cap * 4);
+    if (cfg->join_type == AMD_JOIN_LEFT || cfg->join_type == AMD_JOIN_FULL)
+        JALLOC(o->l_hit, plane / 32 * 4 + 4);
     for (int i = 0; i < o->out_cols && i < 16; i++)
         JALLOC(o->d_out[i], (size_t)o->out_cap * 8);
     JALLOC(o->d_n_out, 8);
@@ -2658,42 +2701,100 @@ static int join_fire(GpuJoin *o, uint64_t instant,
     P.n_keys = o->cfg.n_keys;
     P.nlv = o->cfg.n_left_vals;
     P.nrv = o->cfg.n_right_vals;
+    P.jt = o->cfg.join_type;
+    P.l_key = o->key[0];
     P.instant = instant;
     for (int i = 0; i < o->out_cols && i < 16; i++) P.out[i] = o->d_out[i];
     P.n_out = o->d_n_out;
     P.out_cap = o->out_cap;
     P.err = o->d_err;
+    const bool emit_l =
+        P.jt == AMD_JOIN_LEFT || P.jt == AMD_JOIN_FULL;
+    const bool emit_r =
+        P.jt == AMD_JOIN_RIGHT || P.jt == AMD_JOIN_FULL;
     unsigned long long nl = 0, nr = 0;
     for (uint32_t s : slots) { nl += c0[s]; nr += c1[s]; }
     unsigned long long n = 0;
-    if (o->cfg.n_keys == 0) {
-        int64_t total = (int64_t)nl * (int64_t)nr;
-        if (total > o->out_cap) {
-            snprintf(o->err_msg, sizeof o->err_msg,
-                     "output buffer full; raise log2_out_cap");
-            return 1;
-        }
-        int64_t base = 0;
+    auto pad_left_slots = [&](bool use_bitmap) -> int {
+        /* emit left rows whose matched bit is unset; with use_bitmap false
+         * the bitmap is zeroed first so every row emits (empty right) */
         for (uint32_t ls : slots) {
             if (!c0[ls]) continue;
+            int64_t base = (int64_t)ls * o->cap;
+            if (!use_bitmap)
+                JHIP(o, hipMemsetAsync(o->l_hit + base / 32, 0,
+                                       (size_t)o->cap / 32 * 4, o->stream));
+            JProbeArgs Q = P;
+            Q.l_off = 0;
+            Q.l_hit = o->l_hit;
+            int64_t want = ((int64_t)c0[ls] + 255) / 256;
+            int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
+            hipLaunchKernelGGL(k_join_unmatched_left, dim3(blocks), dim3(256),
+                               0, o->stream, Q, base, (int64_t)c0[ls]);
+            JHIP(o, hipGetLastError());
+        }
+        return 0;
+    };
+    if (o->cfg.n_keys == 0) {
+        if (nl && nr) {
+            int64_t total = (int64_t)nl * (int64_t)nr;
+            if (total > o->out_cap) {
+                snprintf(o->err_msg, sizeof o->err_msg,
+                         "output buffer full; raise log2_out_cap");
+                return 1;
+            }
+            int64_t base = 0;
+            for (uint32_t ls : slots) {
+                if (!c0[ls]) continue;
+                for (uint32_t rs : slots) {
+                    if (!c1[rs]) continue;
+                    int64_t pair = (int64_t)c0[ls] * (int64_t)c1[rs];
+                    P.l_off = (size_t)ls * o->cap;
+                    P.r_off = (size_t)rs * o->cap;
+                    P.nr = (int64_t)c1[rs];
+                    int64_t want = (pair + 255) / 256;
+                    int blocks =
+                        (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
+                    hipLaunchKernelGGL(k_join_cross, dim3(blocks), dim3(256),
+                                       0, o->stream, P, (int64_t)c0[ls], base);
+                    JHIP(o, hipGetLastError());
+                    base += pair;
+                }
+            }
+            n = (unsigned long long)total;
+        } else if (nl && emit_l) {
+            if (pad_left_slots(false)) return 1;
+            JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                                   o->stream));
+        } else if (nr && emit_r) {
+            /* probe with an empty build side: every right row unmatched */
+            JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8,
+                                   o->stream));
+            JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4,
+                                   o->stream));
+            P.l_off = 0;
             for (uint32_t rs : slots) {
                 if (!c1[rs]) continue;
-                int64_t pair = (int64_t)c0[ls] * (int64_t)c1[rs];
-                P.l_off = (size_t)ls * o->cap;
+                P.r_key = o->key[1] + (size_t)rs * o->cap;
                 P.r_off = (size_t)rs * o->cap;
                 P.nr = (int64_t)c1[rs];
-                int64_t want = (pair + 255) / 256;
-                int blocks = (int)(want > 1024 ? 1024 : (want < 1 ? 1 : want));
-                hipLaunchKernelGGL(k_join_cross, dim3(blocks), dim3(256), 0,
-                                   o->stream, P, (int64_t)c0[ls], base);
+                int64_t want = (P.nr + 255) / 256;
+                int blocks = (int)(want > 1024 ? 1024 : want);
+                hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0,
+                                   o->stream, P);
                 JHIP(o, hipGetLastError());
-                base += pair;
             }
+            JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                                   o->stream));
         }
-        n = (unsigned long long)total;
-    } else if (nl && nr) {
-        JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8, o->stream));
-        JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4, o->stream));
+    } else if ((nl && nr) || (nl && emit_l) || (nr && emit_r)) {
+        bool probes = nr && (nl || emit_r);
+        if (probes || nl) {
+            JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8,
+                                   o->stream));
+            JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4,
+                                   o->stream));
+        }
         for (uint32_t ls : slots) {
             if (!c0[ls]) continue;
             JBuildArgs B = {};
@@ -2711,20 +2812,30 @@ static int join_fire(GpuJoin *o, uint64_t instant,
                                o->stream, B);
             JHIP(o, hipGetLastError());
         }
+        if (emit_l)
+            for (uint32_t ls : slots) {
+                if (!c0[ls]) continue;
+                JHIP(o, hipMemsetAsync(
+                            o->l_hit + (size_t)ls * o->cap / 32, 0,
+                            (size_t)o->cap / 32 * 4, o->stream));
+            }
         /* chain links are global row indices: probe reads left values at
          * l_vals[v*plane + li] directly, so l_off = 0 */
         P.l_off = 0;
-        for (uint32_t rs : slots) {
-            if (!c1[rs]) continue;
-            P.r_key = o->key[1] + (size_t)rs * o->cap;
-            P.r_off = (size_t)rs * o->cap;
-            P.nr = (int64_t)c1[rs];
-            int64_t want = (P.nr + 255) / 256;
-            int blocks = (int)(want > 1024 ? 1024 : want);
-            hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0,
-                               o->stream, P);
-            JHIP(o, hipGetLastError());
-        }
+        P.l_hit = emit_l ? o->l_hit : nullptr;
+        if (probes)
+            for (uint32_t rs : slots) {
+                if (!c1[rs]) continue;
+                P.r_key = o->key[1] + (size_t)rs * o->cap;
+                P.r_off = (size_t)rs * o->cap;
+                P.nr = (int64_t)c1[rs];
+                int64_t want = (P.nr + 255) / 256;
+                int blocks = (int)(want > 1024 ? 1024 : want);
+                hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0,
+                                   o->stream, P);
+                JHIP(o, hipGetLastError());
+            }
+        if (emit_l && nl && pad_left_slots(true)) return 1;
         JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                o->stream));
     }
@@ -2865,6 +2976,7 @@ API void arroyo_amd_join_destroy(void *h) {
     hipFree(o->b_keys);
     hipFree(o->b_head);
     hipFree(o->b_next);
+    hipFree(o->l_hit);
     for (int i = 0; i < o->out_cols && i < 16; i++) hipFree(o->d_out[i]);
     hipFree(o->d_n_out);
     hipFree(o->d_err);

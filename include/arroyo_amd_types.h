@@ -66,7 +66,16 @@ typedef struct {
  * join condition is the window/instant itself (e.g. windowed_inner_join.sql:
  * ON dropoffs.window = pickups.window) and each instant emits the cross
  * product of its sides.  Input batch columns per side: [key?, vals...,
- * _timestamp]; output: [key?, left vals..., right vals..., _timestamp]. */
+ * _timestamp]; output: [key?, left vals..., right vals..., _timestamp].
+ * join_type selects the reference's JoinType (planner plan/join.rs maps
+ * LEFT/RIGHT/FULL onto the same per-instant HashJoinExec): non-inner
+ * output gains two trailing presence columns [left_present, right_present]
+ * (the C ABI's stand-in for Arrow validity bitmaps; absent side's value
+ * columns are zero-filled).  Unmatched rows emit once, at instant fire. */
+#define AMD_JOIN_INNER 0
+#define AMD_JOIN_LEFT  1
+#define AMD_JOIN_RIGHT 2
+#define AMD_JOIN_FULL  3
 typedef struct {
     int32_t  n_keys;            /* 0 or 1 */
     int32_t  n_left_vals;
@@ -76,6 +85,7 @@ typedef struct {
     uint32_t log2_out_cap;      /* output rows per fire (GPU) */
     int32_t  device;
     int32_t  emit_to_host;
+    int32_t  join_type;         /* AMD_JOIN_* (default inner) */
 } AmdJoinConfig;
 
 /* Session (gap) window aggregate configuration.  Mirrors

@@ -615,11 +615,14 @@ static JInstant *jinstant_get(JOp *o, uint64_t t) {
 
 ORACLE_API void *oracle_join_create(const AmdJoinConfig *cfg) {
     if (!cfg || cfg->n_keys < 0 || cfg->n_keys > 1 ||
-        cfg->n_left_vals < 0 || cfg->n_right_vals < 0)
+        cfg->n_left_vals < 0 || cfg->n_right_vals < 0 ||
+        cfg->join_type < 0 || cfg->join_type > AMD_JOIN_FULL)
         return NULL;
     JOp *o = calloc(1, sizeof(JOp));
     o->cfg = *cfg;
-    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1;
+    /* non-inner joins append [left_present, right_present] columns */
+    o->out_cols = cfg->n_keys + cfg->n_left_vals + cfg->n_right_vals + 1 +
+                  (cfg->join_type != AMD_JOIN_INNER ? 2 : 0);
     o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
     return o;
 }
@@ -667,34 +670,61 @@ static void jout_reserve(JOp *o, int64_t add) {
     o->out_cap = ncap;
 }
 
+/* li/ri == -1: that side is absent (outer join); its value columns are
+ * zero-filled and its presence flag 0 */
 static void jemit(JOp *o, const JInstant *in, int64_t li, int64_t ri) {
     const AmdJoinConfig *c = &o->cfg;
     jout_reserve(o, 1);
     int64_t r = o->out_rows++;
     int col = 0;
-    if (c->n_keys) o->out[col++][r] = in->l.cols[0][li];
+    if (c->n_keys)
+        o->out[col++][r] = li >= 0 ? in->l.cols[0][li] : in->r.cols[0][ri];
     for (int v = 0; v < c->n_left_vals; v++)
-        o->out[col++][r] = in->l.cols[c->n_keys + v][li];
+        o->out[col++][r] = li >= 0 ? in->l.cols[c->n_keys + v][li] : 0;
     for (int v = 0; v < c->n_right_vals; v++)
-        o->out[col++][r] = in->r.cols[c->n_keys + v][ri];
-    o->out[col][r] = (int64_t)in->instant;
+        o->out[col++][r] = ri >= 0 ? in->r.cols[c->n_keys + v][ri] : 0;
+    o->out[col++][r] = (int64_t)in->instant;
+    if (c->join_type != AMD_JOIN_INNER) {
+        o->out[col++][r] = li >= 0;
+        o->out[col][r] = ri >= 0;
+    }
 }
 
 static void fire_instant(JOp *o, JInstant *in) {
     const AmdJoinConfig *c = &o->cfg;
+    int jt = c->join_type;
+    int emit_l = jt == AMD_JOIN_LEFT || jt == AMD_JOIN_FULL;
+    int emit_r = jt == AMD_JOIN_RIGHT || jt == AMD_JOIN_FULL;
     if (c->n_keys == 0) {
-        for (int64_t li = 0; li < in->l.n; li++)
-            for (int64_t ri = 0; ri < in->r.n; ri++)
-                jemit(o, in, li, ri);
+        /* window-condition join: cross product; a row is unmatched only
+         * when the other side of the instant is empty */
+        if (in->l.n && in->r.n) {
+            for (int64_t li = 0; li < in->l.n; li++)
+                for (int64_t ri = 0; ri < in->r.n; ri++)
+                    jemit(o, in, li, ri);
+        } else if (in->l.n && emit_l) {
+            for (int64_t li = 0; li < in->l.n; li++) jemit(o, in, li, -1);
+        } else if (in->r.n && emit_r) {
+            for (int64_t ri = 0; ri < in->r.n; ri++) jemit(o, in, -1, ri);
+        }
         return;
     }
     /* hash multimap over the left (build) side, probe with the right:
-     * HashJoinExec via LockedJoinPair, planner/physical.rs:177-268 */
-    if (in->l.n == 0 || in->r.n == 0) return;
+     * HashJoinExec via LockedJoinPair, planner/physical.rs:177-268;
+     * JoinType Left/Right/Full additionally emits null-padded rows for
+     * the unmatched side (plan/join.rs join_type passthrough) */
+    if (in->l.n == 0 || in->r.n == 0) {
+        if (in->l.n && emit_l)
+            for (int64_t li = 0; li < in->l.n; li++) jemit(o, in, li, -1);
+        if (in->r.n && emit_r)
+            for (int64_t ri = 0; ri < in->r.n; ri++) jemit(o, in, -1, ri);
+        return;
+    }
     int64_t H = 64;
     while (H < in->l.n * 2) H <<= 1;
     int64_t *head = malloc((size_t)H * 8);
     int64_t *next = malloc((size_t)in->l.n * 8);
+    char *l_hit = calloc((size_t)in->l.n, 1);
     for (int64_t i = 0; i < H; i++) head[i] = -1;
     for (int64_t li = 0; li < in->l.n; li++) {
         uint64_t s = hash64((uint64_t)in->l.cols[0][li]) & (uint64_t)(H - 1);
@@ -704,11 +734,21 @@ static void fire_instant(JOp *o, JInstant *in) {
     for (int64_t ri = 0; ri < in->r.n; ri++) {
         int64_t key = in->r.cols[0][ri];
         uint64_t s = hash64((uint64_t)key) & (uint64_t)(H - 1);
+        int hit = 0;
         for (int64_t li = head[s]; li >= 0; li = next[li])
-            if (in->l.cols[0][li] == key) jemit(o, in, li, ri);
+            if (in->l.cols[0][li] == key) {
+                jemit(o, in, li, ri);
+                l_hit[li] = 1;
+                hit = 1;
+            }
+        if (!hit && emit_r) jemit(o, in, -1, ri);
     }
+    if (emit_l)
+        for (int64_t li = 0; li < in->l.n; li++)
+            if (!l_hit[li]) jemit(o, in, li, -1);
     free(head);
     free(next);
+    free(l_hit);
 }
 
 static void jbuild_out(JOp *o, AmdOutBatch *out) {

@@ -108,7 +108,7 @@ def main():
                  "debezium_agg", "filter_updating_aggregates",
                  "aggregates", "grouped_aggregates", "every_aggregate",
                  "month_loose_watermark", "reinvoke_window_function",
-                 "active_drivers"):
+                 "active_drivers", "windowed_outer_join"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)

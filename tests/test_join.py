@@ -273,3 +273,122 @@ def test_join_gpu_rejects_late_rows():
         # device-side error is surfaced at the next watermark
         op.handle_watermark(t0 + 4 * NS)
     op.close()
+
+
+def np_outer_join(lk, lv, lt, rk, rv, rt, join_type):
+    """numpy/python restatement of per-instant LEFT/RIGHT/FULL join with
+    presence flags (missing side zero-filled)."""
+    rows = []
+    matched_l = set()
+    for j in range(len(rk)):
+        hit = False
+        for i in range(len(lk)):
+            if lt[i] == rt[j] and lk[i] == rk[j]:
+                rows.append((int(lk[i]), int(lv[i]), int(rv[j]),
+                             int(lt[i]), 1, 1))
+                matched_l.add(i)
+                hit = True
+        if not hit and join_type in (2, 3):
+            rows.append((int(rk[j]), 0, int(rv[j]), int(rt[j]), 0, 1))
+    if join_type in (1, 3):
+        for i in range(len(lk)):
+            if i not in matched_l:
+                rows.append((int(lk[i]), int(lv[i]), 0, int(lt[i]), 1, 0))
+    return sorted(rows)
+
+
+def outer_join_scenario(make_op, join_type):
+    rng = np.random.default_rng(83 + join_type)
+    t0 = 1_600_000_000 * NS
+    instants = t0 + np.arange(8, dtype=np.int64) * NS
+    def side(seed, n):
+        r = np.random.default_rng(seed)
+        k = r.integers(0, 24, size=n).astype(np.int64)
+        v = r.integers(0, 10**6, size=n).astype(np.int64)
+        t = np.sort(r.choice(instants, size=n)).astype(np.int64)
+        return k, v, t
+    lk, lv, lt = side(7 + join_type, 400)
+    rk, rv, rt = side(8 + join_type, 350)
+    op = make_op(cabi.make_join_config(n_keys=1, n_left_vals=1,
+                                       n_right_vals=1, join_type=join_type))
+    op.process_batch(op.LEFT, [lk, lv, lt])
+    op.process_batch(op.RIGHT, [rk, rv, rt])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    return got, np_outer_join(lk, lv, lt, rk, rv, rt, join_type)
+
+
+@pytest.mark.parametrize("join_type", [1, 2, 3],
+                         ids=["left", "right", "full"])
+def test_outer_join_oracle_vs_numpy(join_type):
+    """JoinType Left/Right/Full on the instant join (planner plan/join.rs
+    passes the SQL join type through to the per-instant HashJoinExec):
+    unmatched rows emit once at instant fire with zero-filled values and
+    presence flags for the absent side."""
+    got, want = outer_join_scenario(oracle.make_join_op, join_type)
+    assert got == want
+    assert any(r[4] == 0 or r[5] == 0 for r in want)  # pads exercised
+
+
+def test_outer_join_oracle_one_sided_instants():
+    """Instants where one side is entirely empty: FULL emits every row of
+    the present side padded; INNER emits nothing."""
+    t0 = 1_600_000_000 * NS
+    for join_type, expect in ((0, []), (3, [(5, 11, 0, t0, 1, 0)])):
+        op = oracle.make_join_op(cabi.make_join_config(
+            n_keys=1, n_left_vals=1, n_right_vals=1, join_type=join_type))
+        op.process_batch(op.LEFT,
+                         [np.array([5], dtype=np.int64),
+                          np.array([11], dtype=np.int64),
+                          np.array([t0], dtype=np.int64)])
+        got = rows_of(op.handle_watermark(U64MAX))
+        op.close()
+        assert got == expect
+
+
+def test_outer_join_oracle_window_condition():
+    """n_keys=0 (join ON the window itself): cross product when both sides
+    present; FULL pads the present side when the other is empty."""
+    t0 = 1_600_000_000 * NS
+    op = oracle.make_join_op(cabi.make_join_config(
+        n_keys=0, n_left_vals=1, n_right_vals=1, join_type=3))
+    # instant t0: both sides; instant t0+1s: left only; t0+2s: right only
+    op.process_batch(op.LEFT,
+                     [np.array([10, 20], dtype=np.int64),
+                      np.array([t0, t0 + NS], dtype=np.int64)])
+    op.process_batch(op.RIGHT,
+                     [np.array([30, 40], dtype=np.int64),
+                      np.array([t0, t0 + 2 * NS], dtype=np.int64)])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    assert got == sorted([(10, 30, t0, 1, 1),
+                          (20, 0, t0 + NS, 1, 0),
+                          (0, 40, t0 + 2 * NS, 0, 1)])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("join_type", [1, 2, 3],
+                         ids=["left", "right", "full"])
+def test_outer_join_gpu_vs_numpy(join_type):
+    from arroyo_amd import gpu
+    got, want = outer_join_scenario(gpu.make_join_op, join_type)
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_outer_join_gpu_window_condition_and_one_sided():
+    from arroyo_amd import gpu
+    t0 = 1_600_000_000 * NS
+    op = gpu.make_join_op(cabi.make_join_config(
+        n_keys=0, n_left_vals=1, n_right_vals=1, join_type=3))
+    op.process_batch(op.LEFT,
+                     [np.array([10, 20], dtype=np.int64),
+                      np.array([t0, t0 + NS], dtype=np.int64)])
+    op.process_batch(op.RIGHT,
+                     [np.array([30, 40], dtype=np.int64),
+                      np.array([t0, t0 + 2 * NS], dtype=np.int64)])
+    got = rows_of(op.handle_watermark(U64MAX))
+    op.close()
+    assert got == sorted([(10, 30, t0, 1, 1),
+                          (20, 0, t0 + NS, 1, 0),
+                          (0, 40, t0 + 2 * NS, 0, 1)])
