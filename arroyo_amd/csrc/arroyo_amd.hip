@@ -2458,6 +2458,19 @@ k_join_probe(JProbeArgs P) {
     }
 }
 
+/* pad-every-right-row kernel: window-condition (n_keys == 0) outer joins
+ * with an empty left side have no key plane to probe, so emit the right
+ * rows directly (slot-relative ri against P.r_off) */
+__global__ void __launch_bounds__(256)
+k_join_pad_right(JProbeArgs P) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t ri = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         ri < P.nr; ri += stride) {
+        int64_t r = (int64_t)atomicAdd(P.n_out, 1ULL);
+        jemit_row(P, 0, -1, ri, r);
+    }
+}
+
 /* outer-join left pass: emit every left row of one slot whose matched bit
  * is unset ([base, base+nl) global row indices); with a zeroed bitmap this
  * doubles as the pad-every-left-row kernel for empty-right instants */
@@ -2767,21 +2780,16 @@ static int join_fire(GpuJoin *o, uint64_t instant,
             JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                    o->stream));
         } else if (nr && emit_r) {
-            /* probe with an empty build side: every right row unmatched */
-            JHIP(o, hipMemsetAsync(o->b_keys, 0xFF, (size_t)o->H * 8,
-                                   o->stream));
-            JHIP(o, hipMemsetAsync(o->b_head, 0xFF, (size_t)o->H * 4,
-                                   o->stream));
-            P.l_off = 0;
+            /* no key plane exists when n_keys == 0: pad the right rows
+             * directly instead of probing */
             for (uint32_t rs : slots) {
                 if (!c1[rs]) continue;
-                P.r_key = o->key[1] + (size_t)rs * o->cap;
                 P.r_off = (size_t)rs * o->cap;
                 P.nr = (int64_t)c1[rs];
                 int64_t want = (P.nr + 255) / 256;
                 int blocks = (int)(want > 1024 ? 1024 : want);
-                hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(256), 0,
-                                   o->stream, P);
+                hipLaunchKernelGGL(k_join_pad_right, dim3(blocks), dim3(256),
+                                   0, o->stream, P);
                 JHIP(o, hipGetLastError());
             }
             JHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
