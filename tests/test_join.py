@@ -392,3 +392,75 @@ def test_outer_join_gpu_window_condition_and_one_sided():
     assert got == sorted([(10, 30, t0, 1, 1),
                           (20, 0, t0 + NS, 1, 0),
                           (0, 40, t0 + 2 * NS, 0, 1)])
+
+
+@pytest.mark.parametrize("join_type", [1, 2, 3],
+                         ids=["left", "right", "full"])
+def test_outer_join_oracle_multiwatermark_fuzz(join_type):
+    """Outer joins across several watermarks with skewed sides (some
+    instants left-only or right-only): each instant fires exactly once,
+    with pads decided by that instant's final contents."""
+    rng = np.random.default_rng(91 + join_type)
+    t0 = 1_600_000_000 * NS
+    instants = t0 + np.arange(30, dtype=np.int64) * NS
+    # left skews early instants, right skews late: many one-sided instants
+    lk = rng.integers(0, 12, size=300).astype(np.int64)
+    lv = rng.integers(0, 10**6, size=300).astype(np.int64)
+    lt = rng.choice(instants[:20], size=300).astype(np.int64)
+    rk = rng.integers(0, 12, size=300).astype(np.int64)
+    rv = rng.integers(0, 10**6, size=300).astype(np.int64)
+    rt = rng.choice(instants[10:], size=300).astype(np.int64)
+    # three rounds, each feeding the rows in [prev_wm, wm)
+    op = oracle.make_join_op(cabi.make_join_config(
+        n_keys=1, n_left_vals=1, n_right_vals=1, join_type=join_type))
+    got = []
+    prev = 0
+    for wm in (int(t0 + 10 * NS), int(t0 + 20 * NS), U64MAX):
+        lm = (lt >= prev) & (lt < wm)
+        rm = (rt >= prev) & (rt < wm)
+        op.process_batch(op.LEFT, [lk[lm], lv[lm], lt[lm]])
+        op.process_batch(op.RIGHT, [rk[rm], rv[rm], rt[rm]])
+        got += rows_of(op.handle_watermark(wm))
+        prev = wm
+    op.close()
+    want = np_outer_join(lk, lv, lt, rk, rv, rt, join_type)
+    assert sorted(got) == want
+    assert any(r[4] == 0 for r in want) or any(r[5] == 0 for r in want)
+
+
+def test_outer_join_oracle_checkpoint_roundtrip():
+    """Drain both sides of a half-fed FULL join, restore into a fresh op,
+    finish, and match the uninterrupted run (join_type orthogonal to the
+    raw-row drain format)."""
+    rng = np.random.default_rng(97)
+    t0 = 1_600_000_000 * NS
+    instants = t0 + np.arange(10, dtype=np.int64) * NS
+    def gen(seed, n):
+        r = np.random.default_rng(seed)
+        return (r.integers(0, 10, size=n).astype(np.int64),
+                r.integers(0, 1000, size=n).astype(np.int64),
+                np.sort(r.choice(instants, size=n)).astype(np.int64))
+    lk, lv, lt = gen(1, 200)
+    rk, rv, rt = gen(2, 150)
+    cfgk = dict(n_keys=1, n_left_vals=1, n_right_vals=1,
+                join_type=cabi.JOIN_FULL)
+
+    a = oracle.make_join_op(cabi.make_join_config(**cfgk))
+    a.process_batch(a.LEFT, [lk, lv, lt])
+    ld = a.checkpoint_drain(a.LEFT)
+    rd = a.checkpoint_drain(a.RIGHT)
+    a.close()
+
+    b = oracle.make_join_op(cabi.make_join_config(**cfgk))
+    b.restore(b.LEFT, ld)
+    b.restore(b.RIGHT, rd)
+    b.process_batch(b.RIGHT, [rk, rv, rt])
+    got = rows_of(b.handle_watermark(U64MAX))
+    b.close()
+
+    c = oracle.make_join_op(cabi.make_join_config(**cfgk))
+    c.process_batch(c.LEFT, [lk, lv, lt])
+    c.process_batch(c.RIGHT, [rk, rv, rt])
+    want = rows_of(c.handle_watermark(U64MAX))
+    c.close()
+    assert got == want
