@@ -345,12 +345,15 @@ class AmdExpJoinConfig(ctypes.Structure):
         ("log2_out_cap", ctypes.c_uint32),
         ("device", ctypes.c_int32),
         ("emit_to_host", ctypes.c_int32),
+        ("join_type", ctypes.c_int32),
+        ("updating", ctypes.c_int32),
     ]
 
 
 def make_expjoin_config(ttl_ns, n_left_vals=0, n_right_vals=0,
                         log2_capacity=16, log2_rows_cap=20, log2_out_cap=20,
-                        device=0, emit_to_host=True):
+                        device=0, emit_to_host=True, join_type=0,
+                        updating=False):
     cfg = AmdExpJoinConfig()
     cfg.n_keys = 1
     cfg.n_left_vals = n_left_vals
@@ -361,6 +364,8 @@ def make_expjoin_config(ttl_ns, n_left_vals=0, n_right_vals=0,
     cfg.log2_out_cap = log2_out_cap
     cfg.device = device
     cfg.emit_to_host = 1 if emit_to_host else 0
+    cfg.join_type = join_type
+    cfg.updating = 1 if updating else 0
     return cfg
 
 

@@ -296,6 +296,17 @@ API void *arroyo_amd_expjoin_create(const AmdExpJoinConfig *cfg) {
         snprintf(g_ej_err, sizeof g_ej_err, "invalid expjoin config");
         return nullptr;
     }
+    if (cfg->join_type != 0 || cfg->updating != 0) {
+        /* semantics pinned by the CPU oracle (oracle/arroyo_oracle.c
+         * expjoin_insert) and the updating_*_join goldens; the HIP path
+         * for updating/outer TTL joins is round-2 work -- fail loudly
+         * rather than silently joining inner/append-only */
+        snprintf(g_ej_err, sizeof g_ej_err,
+                 "updating/outer TTL join not yet on the GPU path "
+                 "(join_type=%d updating=%d)",
+                 cfg->join_type, cfg->updating);
+        return nullptr;
+    }
     GpuExpJoin *o = new GpuExpJoin();
     o->cfg = *cfg;
     o->out_cols = 1 + cfg->n_left_vals + cfg->n_right_vals + 1;

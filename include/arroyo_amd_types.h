@@ -135,7 +135,19 @@ typedef struct {
  * crates/arroyo-planner/src/plan/join.rs:121-191).  The live in-memory view
  * never evicts during a run (TTL filters state only on restore,
  * table_manager.rs:533-570 passes the watermark to get_view); restore drops
- * rows with ts < watermark - ttl. */
+ * rows with ts < watermark - ttl.
+ *
+ * join_type (AMD_JOIN_*) selects the reference's JoinType for updating
+ * joins (the planner passes the SQL join type into the DataFusion join
+ * inside JoinWithExpiration, plan/join.rs:326-379); non-inner output gains
+ * trailing [left_present, right_present] columns (Arrow-validity stand-in)
+ * and, because a later match must retract an earlier null-padded row, a
+ * trailing is_retract column.  updating=1 means inputs carry retractions:
+ * per-side layout becomes [key, vals..., is_retract, _timestamp], the
+ * output gains the trailing is_retract column, and emissions maintain the
+ * join incrementally (append/retract pairs as either side's multiset
+ * changes).  CPU oracle only in round 1: the GPU library rejects
+ * join_type != 0 or updating != 0 at create (no silent fallback). */
 typedef struct {
     int32_t  n_keys;            /* 1 (i64 equi-join key) */
     int32_t  n_left_vals;
@@ -146,6 +158,8 @@ typedef struct {
     uint32_t log2_out_cap;      /* output rows per process_batch (GPU) */
     int32_t  device;
     int32_t  emit_to_host;
+    int32_t  join_type;         /* AMD_JOIN_* (default inner) */
+    int32_t  updating;          /* 1 = inputs carry is_retract */
 } AmdExpJoinConfig;
 
 /* Updating (non-windowed) aggregate configuration.  Mirrors

@@ -1253,7 +1253,8 @@ static void erows_push(ERows *b, int nv, const int64_t *const *cols,
 
 ORACLE_API void *oracle_expjoin_create(const AmdExpJoinConfig *cfg) {
     if (!cfg || cfg->n_keys != 1 || cfg->n_left_vals < 0 ||
-        cfg->n_right_vals < 0 || cfg->ttl_nanos == 0)
+        cfg->n_right_vals < 0 || cfg->ttl_nanos == 0 ||
+        cfg->join_type < 0 || cfg->join_type > AMD_JOIN_FULL)
         return NULL;
     EOp *o = calloc(1, sizeof(EOp));
     o->cfg = *cfg;
@@ -1265,7 +1266,12 @@ ORACLE_API void *oracle_expjoin_create(const AmdExpJoinConfig *cfg) {
         sd->map_used = calloc((size_t)sd->map_cap, 1);
         sd->rows = calloc((size_t)sd->map_cap, sizeof(ERows));
     }
-    o->out_cols = 1 + cfg->n_left_vals + cfg->n_right_vals + 1;
+    /* non-inner: + [left_present, right_present]; non-inner or updating:
+     * + trailing is_retract (a later match retracts the earlier
+     * null-padded row) */
+    o->out_cols = 1 + cfg->n_left_vals + cfg->n_right_vals + 1 +
+                  (cfg->join_type != AMD_JOIN_INNER ? 2 : 0) +
+                  (cfg->join_type != AMD_JOIN_INNER || cfg->updating ? 1 : 0);
     o->out = calloc((size_t)o->out_cols, sizeof(int64_t *));
     return o;
 }
@@ -1297,55 +1303,146 @@ static void ebuild_out(EOp *o, AmdOutBatch *out) {
     o->out_rows = 0;
 }
 
+/* emit one output row; lv/rv NULL = that side absent (outer join) */
+static void eemit2(EOp *o, int64_t key, const int64_t *lv, int64_t lts,
+                   const int64_t *rv, int64_t rts, int retract) {
+    const AmdExpJoinConfig *c = &o->cfg;
+    eout_reserve(o, 1);
+    int64_t rr = o->out_rows++;
+    int col = 0;
+    o->out[col++][rr] = key;
+    for (int v = 0; v < c->n_left_vals; v++)
+        o->out[col++][rr] = lv ? lv[v] : 0;
+    for (int v = 0; v < c->n_right_vals; v++)
+        o->out[col++][rr] = rv ? rv[v] : 0;
+    o->out[col++][rr] = lv ? (rv ? (lts > rts ? lts : rts) : lts) : rts;
+    if (c->join_type != AMD_JOIN_INNER) {
+        o->out[col++][rr] = lv != NULL;
+        o->out[col++][rr] = rv != NULL;
+    }
+    if (c->join_type != AMD_JOIN_INNER || c->updating)
+        o->out[col][rr] = retract;
+}
+
+/* pair emission with arrival-side routing: `mine` is the incoming row's
+ * values, `ob` a stored row of the other side */
+static void eemit_pair(EOp *o, int32_t side, int64_t key,
+                       const int64_t *mine, int64_t mts, const ERows *ob,
+                       int64_t j, int retract) {
+    int64_t ovals[64];
+    ESide *other = &o->side[1 - side];
+    for (int v = 0; v < other->nv; v++) ovals[v] = ob->vals[v][j];
+    if (side == 0)
+        eemit2(o, key, mine, mts, ovals, ob->ts[j], retract);
+    else
+        eemit2(o, key, ovals, ob->ts[j], mine, mts, retract);
+}
+
+/* null-padded emission for a stored row of side `s` */
+static void eemit_null(EOp *o, int32_t s, int64_t key, const ERows *b,
+                       int64_t j, int retract) {
+    int64_t vals[64];
+    for (int v = 0; v < o->side[s].nv; v++) vals[v] = b->vals[v][j];
+    if (s == 0)
+        eemit2(o, key, vals, b->ts[j], NULL, 0, retract);
+    else
+        eemit2(o, key, NULL, 0, vals, b->ts[j], retract);
+}
+
+static ERows *eside_find(ESide *sd, int64_t key) {
+    uint64_t m = (uint64_t)sd->map_cap - 1;
+    uint64_t i = hash64((uint64_t)key) & m;
+    while (sd->map_used[i] && sd->map_keys[i] != key) i = (i + 1) & m;
+    return sd->map_used[i] ? &sd->rows[i] : NULL;
+}
+
 static int expjoin_insert(EOp *o, int32_t side, const int64_t *const *cols,
                           int32_t n_cols, int64_t n_rows, int emit,
                           AmdOutBatch *out) {
     const AmdExpJoinConfig *c = &o->cfg;
     int nv = side == 0 ? c->n_left_vals : c->n_right_vals;
-    int ov = side == 0 ? c->n_right_vals : c->n_left_vals;
-    int want = 1 + nv + 1;
+    int upd = c->updating ? 1 : 0;
+    int jt = c->join_type;
+    int want = 1 + nv + upd + 1;
     if (n_cols != want) {
         snprintf(o->err, sizeof o->err, "side %d expects %d cols, got %d",
                  side, want, n_cols);
         return 1;
     }
+    /* which null-padded rows this join type emits (plan/join.rs join_type
+     * into the DataFusion join: LEFT pads unmatched left rows, etc.) */
+    int null_own = jt == AMD_JOIN_FULL ||
+                   (side == 0 ? jt == AMD_JOIN_LEFT : jt == AMD_JOIN_RIGHT);
+    int null_other = jt == AMD_JOIN_FULL ||
+                     (side == 0 ? jt == AMD_JOIN_RIGHT : jt == AMD_JOIN_LEFT);
     const int64_t *ts = cols[n_cols - 1];
-    ESide *other = &o->side[1 - side];
+    const int64_t *retr = upd ? cols[1 + nv] : NULL;
     for (int64_t r = 0; r < n_rows; r++) {
         int64_t key = cols[0][r];
-        if (emit) {
-            /* probe the other side's stored rows BEFORE inserting this row
-             * (same-side rows never join each other) */
-            uint64_t m = (uint64_t)other->map_cap - 1;
-            uint64_t i = hash64((uint64_t)key) & m;
-            while (other->map_used[i] && other->map_keys[i] != key)
-                i = (i + 1) & m;
-            if (other->map_used[i]) {
-                ERows *ob = &other->rows[i];
-                for (int64_t j = 0; j < ob->n; j++) {
-                    eout_reserve(o, 1);
-                    int64_t rr = o->out_rows++;
-                    int col = 0;
-                    o->out[col++][rr] = key;
-                    /* output order is [left vals, right vals] regardless of
-                     * arrival side */
-                    if (side == 0) {
-                        for (int v = 0; v < nv; v++)
-                            o->out[col++][rr] = cols[1 + v][r];
-                        for (int v = 0; v < ov; v++)
-                            o->out[col++][rr] = ob->vals[v][j];
-                    } else {
-                        for (int v = 0; v < ov; v++)
-                            o->out[col++][rr] = ob->vals[v][j];
-                        for (int v = 0; v < nv; v++)
-                            o->out[col++][rr] = cols[1 + v][r];
-                    }
-                    int64_t mt = ts[r] > ob->ts[j] ? ts[r] : ob->ts[j];
-                    o->out[col][rr] = mt;
+        int64_t mine[64];
+        for (int v = 0; v < nv; v++) mine[v] = cols[1 + v][r];
+        ERows *ob = eside_find(&o->side[1 - side], key);
+        int64_t on = ob ? ob->n : 0;
+        if (!upd || !retr[r]) {
+            /* append */
+            ERows *own = eside_slot(&o->side[side], key);
+            if (emit) {
+                if (on > 0) {
+                    for (int64_t j = 0; j < on; j++)
+                        eemit_pair(o, side, key, mine, ts[r], ob, j, 0);
+                    if (null_other && own->n == 0)
+                        /* the other side's rows were unmatched until now:
+                         * retract their null-padded emissions */
+                        for (int64_t j = 0; j < on; j++)
+                            eemit_null(o, 1 - side, key, ob, j, 1);
+                } else if (null_own) {
+                    eemit2(o, key, side == 0 ? mine : NULL,
+                           side == 0 ? ts[r] : 0,
+                           side == 0 ? NULL : mine,
+                           side == 0 ? 0 : ts[r], 0);
                 }
             }
+            erows_push(own, nv, cols, r, ts[r]);
+        } else {
+            /* retract: remove one stored row with equal values */
+            ERows *own = eside_find(&o->side[side], key);
+            int64_t idx = -1;
+            if (own)
+                for (int64_t j = 0; j < own->n && idx < 0; j++) {
+                    int eq = 1;
+                    for (int v = 0; v < nv && eq; v++)
+                        eq = own->vals[v][j] == mine[v];
+                    if (eq) idx = j;
+                }
+            if (idx < 0) {
+                snprintf(o->err, sizeof o->err,
+                         "retract of unknown row for key %lld",
+                         (long long)key);
+                return 1;
+            }
+            int64_t sts = own->ts[idx];
+            if (emit) {
+                if (on > 0) {
+                    for (int64_t j = 0; j < on; j++)
+                        eemit_pair(o, side, key, mine, sts, ob, j, 1);
+                    if (null_other && own->n == 1)
+                        /* the other side's rows lose their last match:
+                         * their null-padded rows come back */
+                        for (int64_t j = 0; j < on; j++)
+                            eemit_null(o, 1 - side, key, ob, j, 0);
+                } else if (null_own) {
+                    eemit2(o, key, side == 0 ? mine : NULL,
+                           side == 0 ? sts : 0,
+                           side == 0 ? NULL : mine,
+                           side == 0 ? 0 : sts, 1);
+                }
+            }
+            /* remove idx (order within the multiset is not observable) */
+            for (int v = 0; v < nv; v++)
+                own->vals[v][idx] = own->vals[v][own->n - 1];
+            own->ts[idx] = own->ts[own->n - 1];
+            own->n--;
         }
-        erows_push(eside_slot(&o->side[side], key), nv, cols, r, ts[r]);
     }
     if (out) ebuild_out(o, out);
     return 0;

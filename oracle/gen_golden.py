@@ -108,7 +108,9 @@ def main():
                  "debezium_agg", "filter_updating_aggregates",
                  "aggregates", "grouped_aggregates", "every_aggregate",
                  "month_loose_watermark", "reinvoke_window_function",
-                 "active_drivers", "windowed_outer_join"):
+                 "active_drivers", "windowed_outer_join",
+                 "updating_left_join", "updating_right_join",
+                 "updating_full_join"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)
