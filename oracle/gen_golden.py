@@ -110,7 +110,7 @@ def main():
                  "month_loose_watermark", "reinvoke_window_function",
                  "active_drivers", "windowed_outer_join",
                  "updating_left_join", "updating_right_join",
-                 "updating_full_join"):
+                 "updating_full_join", "offset_impulse_join"):
         rows = load_rows(f"{REF}/golden_outputs/{name}.json")
         with open(f"{OUT}/{name}.golden.json", "w") as f:
             json.dump(rows, f)
