@@ -116,8 +116,9 @@ def stream_fuzz(make_op, n=4000, seed=3, gap_s=5, aggs=None, n_value_cols=0,
     return key, ts, vals, got
 
 
-def test_session_oracle_vs_numpy_fuzz():
-    key, ts, _vals, got = stream_fuzz(oracle.make_session_op)
+@pytest.mark.parametrize("seed", [3, 17, 71])
+def test_session_oracle_vs_numpy_fuzz(seed):
+    key, ts, _vals, got = stream_fuzz(oracle.make_session_op, seed=seed)
     want = [(k, n, s, e, t)
             for k, n, s, e, t in np_sessions(key, ts, 5 * NS)]
     assert sorted(got) == sorted(want)

@@ -126,9 +126,10 @@ def wf_fuzz(make_op, seed=13, n=4000, limit=0):
     return sorted(got), sorted(want)
 
 
+@pytest.mark.parametrize("seed", [13, 29])
 @pytest.mark.parametrize("limit", [0, 3])
-def test_windowfn_oracle_vs_numpy_fuzz(limit):
-    got, want = wf_fuzz(oracle.make_windowfn_op, limit=limit)
+def test_windowfn_oracle_vs_numpy_fuzz(limit, seed):
+    got, want = wf_fuzz(oracle.make_windowfn_op, limit=limit, seed=seed)
     assert got == want
     assert len(want) > 100
 
