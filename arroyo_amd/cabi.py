@@ -6,7 +6,6 @@ Generic over the implementing library: the product HIP library
 drive both through :class:`WindowOp` and compare.
 """
 import ctypes
-import os
 
 import numpy as np
 

@@ -10,7 +10,7 @@ import subprocess
 
 import numpy as np
 
-from arroyo_amd.cabi import AmdOutBatch, WindowOp, _out_to_numpy
+from arroyo_amd.cabi import WindowOp
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libarroyo_amd.so")

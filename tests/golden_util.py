@@ -4,7 +4,6 @@ import json
 import os
 from datetime import datetime as dt, timezone
 
-import numpy as np
 
 GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
 NS = 10**9

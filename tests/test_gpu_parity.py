@@ -130,8 +130,6 @@ def test_key_minus_one_special_slot():
 def test_golden_vectors_on_gpu():
     """The reference's own golden scenarios through the HIP path (batch 32,
     as in the reference smoke tests)."""
-    import json
-
     from tests.golden_util import (assert_rows_match, fmt_ts, load_golden,
                                    load_inputs)
     inp = load_inputs()["impulse"]

@@ -14,7 +14,6 @@ gloo with the C oracle as the per-shard operator.
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

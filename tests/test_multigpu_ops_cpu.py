@@ -13,7 +13,6 @@ same layout runs one HIP operator per rank with the exchange over RCCL.
 import os
 
 import numpy as np
-import pytest
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
