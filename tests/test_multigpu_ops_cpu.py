@@ -278,3 +278,64 @@ def test_instant_join_two_rank_shuffle_matches_single():
     single.close()
     assert sorted(results[0] + results[1]) == sorted(want)
     assert len(want) > 500
+
+
+def _windowfn_rank_main(rank, world, port, result_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        op = oracle.make_windowfn_op(cabi.make_windowfn_config(
+            n_cols=3, part_col=0, order=[(1, False)]))
+        got = []
+        for step in range(4):
+            part, val, ts = _gen_stream(600 + step, 500)
+            ts = T0 + (step * 30 + ((ts - T0) // (60 * NS)) * 3) * NS
+            mine = np.arange(len(part)) % world == rank
+            cols = shuffle_columns([part[mine], val[mine], ts[mine]], world)
+            if len(cols[0]):
+                op.process_batch(cols)
+            got += _rows(op.handle_watermark(int(ts.max()) + NS))
+        op.close()
+        result_q.put((rank, sorted(got)))
+        dist.destroy_process_group()
+    except Exception as e:
+        result_q.put((rank, e))
+        raise
+
+
+def test_windowfn_two_rank_shuffle_matches_single():
+    """ROW_NUMBER shards by its PARTITION BY column: a partition's rows all
+    land on one rank, so per-partition ranking is computed whole there.
+    Tie order inside a partition follows arrival sequence, which the keyed
+    shuffle preserves per source (rows of one partition arrive from one
+    logical upstream order here), so the merged rows equal the single
+    run's."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_windowfn_rank_main,
+                         args=(r, WORLD, 29393, q)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, rows = q.get(timeout=120)
+        assert not isinstance(rows, Exception), rows
+        results[rank] = rows
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    assert not ({r[0] for r in results[0]} & {r[0] for r in results[1]})
+
+    single = oracle.make_windowfn_op(cabi.make_windowfn_config(
+        n_cols=3, part_col=0, order=[(1, False)]))
+    want = []
+    for step in range(4):
+        part, val, ts = _gen_stream(600 + step, 500)
+        ts = T0 + (step * 30 + ((ts - T0) // (60 * NS)) * 3) * NS
+        single.process_batch([part, val, ts])
+        want += _rows(single.handle_watermark(int(ts.max()) + NS))
+    single.close()
+    assert sorted(results[0] + results[1]) == sorted(want)
+    assert len(want) > 1000
