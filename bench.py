@@ -295,11 +295,16 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
+    if args.gpus > 1 and int(os.environ.get("WORLD_SIZE", "1")) == 1:
+        raise SystemExit(
+            "--gpus N>1 must run under torch.distributed.run with "
+            "--nproc-per-node N (one rank per GPU); a single-process run "
+            "would misreport whole-job throughput")
     elapsed, perf, rank, world = run_gpu(args)
     if rank != 0:
         return
 
-    n_gpus = world if world > 1 else args.gpus
+    n_gpus = world
     total_rows = args.steps * BATCH_ROWS * n_gpus
     value = total_rows / elapsed
 
