@@ -78,6 +78,11 @@ int arroyo_amd_process_batches_device(void *h, const int64_t *const *dcols,
 double arroyo_amd_stream_gbps(const void *d_a, const void *d_b, int64_t n,
                               int iters);
 
+/* Synchronize the operator's internal stream: callers that reuse
+ * device buffers handed to process_batch_device (e.g. the N>1 bench's
+ * RCCL exchange buffers) must fence the consuming kernels first. */
+int arroyo_amd_sync(void *h);
+
 /* Advance the watermark; fires every window the reference would fire, in
  * order, and returns the emitted rows (all fired windows concatenated;
  * column order [key?, aggs..., window_start, window_end, _timestamp]).
