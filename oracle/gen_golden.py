@@ -24,7 +24,9 @@ import os
 from datetime import datetime as dt, timezone
 
 REF = "/root/reference/crates/arroyo-sql-testing"
-OUT = os.path.join(os.path.dirname(__file__), "..", "tests", "golden")
+OUT = os.environ.get(
+    "GOLDEN_OUT",
+    os.path.join(os.path.dirname(__file__), "..", "tests", "golden"))
 
 NS = 10**9
 

@@ -3,6 +3,8 @@ golden test vectors (tests/golden/, from arroyo-sql-testing).  These four
 queries exercise the exact operators on the hot path: sliding (q5 window
 parameters hop 2s/10s), keyed sliding (hop 1min/1h GROUP BY driver_id),
 tumbling keyed/unkeyed, and both watermark-lateness variants."""
+import os
+
 import numpy as np
 import pytest
 
@@ -188,3 +190,25 @@ def test_c_oracle_avg_fuzz(seed):
         assert got[kk] == pytest.approx(av, rel=1e-12)
     assert len(want) > 500
     cop.close()
+
+
+@pytest.mark.skipif(not os.path.isdir("/root/reference"),
+                    reason="reference checkout absent (GPU box)")
+def test_golden_fixtures_reproducible(tmp_path):
+    """Re-running the extraction against /root/reference must reproduce the
+    committed fixtures bit-for-bit (guards against fixture drift)."""
+    import json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, GOLDEN_OUT=str(tmp_path))
+    subprocess.run([sys.executable, os.path.join(repo, "oracle",
+                                                 "gen_golden.py")],
+                   check=True, env=env, capture_output=True)
+    committed = os.path.join(repo, "tests", "golden")
+    names = sorted(os.listdir(committed))
+    assert names == sorted(os.listdir(tmp_path))
+    for n in names:
+        with open(os.path.join(committed, n)) as f1, \
+                open(tmp_path / n) as f2:
+            assert json.load(f1) == json.load(f2), n
