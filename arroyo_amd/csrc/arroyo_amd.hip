@@ -68,8 +68,12 @@
 enum WordOp { W_ADD_I64 = 0, W_MAX_U64 = 1, W_ADD_F64 = 2, W_NONE = 3 };
 
 struct DeviceRing {
-    int64_t  *keys;      /* [R][C] */
-    uint64_t *state;     /* [R][C][n_aggs][2] encoded */
+    int64_t  *keys;      /* [R][C] (unpacked layout; null when packed) */
+    uint64_t *state;     /* [R][C][n_aggs][2] encoded (null when packed) */
+    uint64_t *slots;     /* [R][C][2] = {key, count} AoS (packed layout:
+                            single-COUNT keyed shape; one 64B line holds
+                            both probe key and its state word, halving the
+                            random-access line touches of the hot kernel) */
     uint64_t *tag;       /* [R] bin nanos or EMPTY_TAG */
     /* special entry per pane for an actual key == EMPTY_KEY */
     uint32_t *spec_used; /* [R] */
@@ -78,6 +82,7 @@ struct DeviceRing {
     uint64_t *min_bin;   /* running min of non-late bins (state machine) */
     uint32_t  C;         /* slots per pane, power of two */
     uint32_t  R;         /* panes in ring, power of two */
+    int       packed;    /* 1: use `slots`, keys/state are null */
 };
 
 struct AggSpec {
@@ -580,6 +585,171 @@ template __global__ void k_update_lds_vec<1024, true, true>(UpdateArgs);
 template __global__ void k_update_lds_vec<2048, true, true>(UpdateArgs);
 
 /* ------------------------------------------------------------------ */
+/* Packed-table update path (default for the keyed single-COUNT shape —
+ * the q5 headline).  Round-2 redesign from measurement: the LDS-staged
+ * kernel was latency-bound, not bandwidth-bound (PMC traffic ~2.9x
+ * compulsory but only ~7% of peak) — the cost was ~1M+ random global
+ * upserts per launch (per-block LDS flush x 640 blocks + cold-key
+ * fallthrough), each touching TWO cache lines (key array + state array)
+ * with a returning atomic.  This path instead:
+ *   - stores the pane table AoS ({key, count} in one 16B slot): the probe
+ *     load brings the count word in the same 64B line, and the count
+ *     atomicAdd's result is unused so it compiles to a no-return
+ *     global_atomic_add (fire-and-forget; no latency chain);
+ *   - uses NO LDS staging and therefore has NO per-block flush: the
+ *     nexmark hot key is wave-uniform (consecutive rows -> one wave), so
+ *     two ballot samples (first/last pending lane) find the hot group and
+ *     its leader issues ONE atomicAdd(popcount) for the whole group;
+ *     cold keys (~unique; zero reuse, which is why LDS staging bought
+ *     nothing for them) go straight to the packed table;
+ *   - keeps a per-thread (key,pane)->slot register cache for
+ *     intra-thread runs (the vectorized 2-rows-per-iteration pairs). */
+
+__device__ inline uint64_t *packed_upsert(uint64_t *slots, uint32_t C,
+                                          int64_t key, int *err) {
+    uint64_t m = C - 1;
+    uint64_t i = hash64((uint64_t)key) & m;
+    uint32_t lim = C < MAX_PROBES ? C : MAX_PROBES;
+    for (uint32_t probes = 0; probes < lim; probes++) {
+        uint64_t *s = slots + i * 2;
+        int64_t k = (int64_t)s[0];
+        if (k == key) return s + 1;
+        if (k == EMPTY_KEY) {
+            int64_t old = (int64_t)atomicCAS((unsigned long long *)s,
+                                             (unsigned long long)EMPTY_KEY,
+                                             (unsigned long long)key);
+            if (old == EMPTY_KEY || old == key) return s + 1;
+        }
+        i = (i + 1) & m;
+    }
+    *err = ERR_TABLE_FULL;
+    return nullptr;
+}
+
+#define HOT_MIN 4   /* combine a wave group only when >= this many lanes */
+
+/* one row of the packed path; called by every active lane together (the
+ * ballots inside require it), rows that drop out carry need=false */
+__device__ inline void packed_row(const UpdateArgs &A, uint64_t traw,
+                                  int64_t key, uint64_t &local_min,
+                                  uint64_t &last_bin, int64_t &last_key,
+                                  uint32_t &last_pane, uint64_t *&last_cnt) {
+    uint64_t t = traw + A.ts_offset;
+    uint64_t q = div_slide(t, A.slide, A.slide_inv);
+    uint64_t bin = q * A.slide;
+    bool alive = !(A.has_wm && bin < A.wm_bin);      /* late drop */
+    if (alive && bin < local_min) local_min = bin;
+    uint32_t p = (uint32_t)(q & (A.ring.R - 1));
+    if (alive && bin != last_bin) {
+        claim_tag_wave(A.ring.tag, p, bin, EMPTY_TAG, A.ring.err);
+        last_bin = bin;
+    }
+    bool need = alive;
+    if (need && key == EMPTY_KEY) {                  /* sentinel-valued key */
+        atomicExch(&A.ring.spec_used[p], 1u);
+        atomicAdd((unsigned long long *)
+                      (A.ring.spec_state + (size_t)p * 2), 1ULL);
+        need = false;
+    }
+    /* intra-thread run cache (vec pairs, bursty keys) */
+    if (need && key == last_key && p == last_pane && last_cnt) {
+        atomicAdd((unsigned long long *)last_cnt, 1ULL);
+        need = false;
+    }
+    /* wave-hot combine: sample the first and last pending lanes; if >=
+     * HOT_MIN lanes share that (key, bin), the group leader does one
+     * upsert + one atomicAdd(count_of_group) */
+    unsigned long long pend = __ballot(need);
+    const int lane = (int)(threadIdx.x & 63);
+    for (int s = 0; s < 2 && pend; s++) {
+        int sl = s == 0 ? (int)(__ffsll((long long)pend) - 1)
+                        : 63 - (int)__clzll((long long)pend);
+        int64_t k0 = (int64_t)__shfl((long long)key, sl, 64);
+        uint64_t b0 = (uint64_t)__shfl((long long)bin, sl, 64);
+        uint32_t pp = (uint32_t)__shfl((int)p, sl, 64);
+        bool mine = need && key == k0 && bin == b0;
+        unsigned long long grp = __ballot(mine);
+        int cnt = __popcll((long long)grp);
+        if (cnt >= HOT_MIN) {
+            if (mine && lane == (int)(__ffsll((long long)grp) - 1)) {
+                uint64_t *c = packed_upsert(
+                    A.ring.slots + (size_t)pp * A.ring.C * 2, A.ring.C, k0,
+                    A.ring.err);
+                if (c) {
+                    atomicAdd((unsigned long long *)c,
+                              (unsigned long long)cnt);
+                    last_key = k0;
+                    last_pane = pp;
+                    last_cnt = c;
+                }
+            }
+            if (mine) need = false;
+            pend &= ~grp;
+        }
+    }
+    if (need) {
+        uint64_t *c = packed_upsert(A.ring.slots + (size_t)p * A.ring.C * 2,
+                                    A.ring.C, key, A.ring.err);
+        if (c) {
+            atomicAdd((unsigned long long *)c, 1ULL);
+            last_key = key;
+            last_pane = p;
+            last_cnt = c;
+        }
+    }
+}
+
+template <bool VEC>
+__global__ void __launch_bounds__(256)
+k_update_packed(UpdateArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_pane = PANE_UNSET;
+    uint64_t *last_cnt = nullptr;
+    if (VEC) {
+        int64_t n2 = A.n_rows >> 1;
+        for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+             v < n2; v += stride) {
+            ulonglong2 tsv = ((const ulonglong2 *)A.ts_col)[v];
+            ulonglong2 kv = ((const ulonglong2 *)A.key_col)[v];
+            packed_row(A, tsv.x, (int64_t)kv.x, local_min, last_bin,
+                       last_key, last_pane, last_cnt);
+            packed_row(A, tsv.y, (int64_t)kv.y, local_min, last_bin,
+                       last_key, last_pane, last_cnt);
+        }
+        if ((A.n_rows & 1) && blockIdx.x == 0 && (threadIdx.x & 63) == 0 &&
+            threadIdx.x == 0)
+            packed_row(A, (uint64_t)A.ts_col[A.n_rows - 1],
+                       A.key_col[A.n_rows - 1], local_min, last_bin,
+                       last_key, last_pane, last_cnt);
+    } else {
+        for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+             i < A.n_rows; i += stride)
+            packed_row(A, (uint64_t)A.ts_col[i], A.key_col[i], local_min,
+                       last_bin, last_key, last_pane, last_cnt);
+    }
+    fold_min_bin(local_min, A.ring.min_bin);
+}
+
+template __global__ void k_update_packed<true>(UpdateArgs);
+template __global__ void k_update_packed<false>(UpdateArgs);
+
+/* packed pane clear: {EMPTY_KEY, 0} per slot (a single memset cannot set
+ * the two words differently, and count must start at 0) */
+__global__ void __launch_bounds__(256)
+k_retire_packed(uint64_t *slots, int64_t n_slots) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n_slots; i += stride) {
+        ulonglong2 v;
+        v.x = (unsigned long long)EMPTY_KEY;
+        v.y = 0;
+        ((ulonglong2 *)slots)[i] = v;
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* Radix-partitioned update path (ARROYO_AMD_RADIX=1, keyed
  * n_value_cols==0 launches): instead of every block hammering the shared
  * pane tables, rows are first partitioned by the top 8 bits of hash(key)
@@ -753,6 +923,18 @@ k_merge(MergeArgs M) {
          i < total; i += stride) {
         uint32_t p = M.src[i / M.ring.C];
         size_t slot = i % M.ring.C;
+        if (M.ring.packed) {
+            const uint64_t *s =
+                M.ring.slots + ((size_t)p * M.ring.C + slot) * 2;
+            int64_t key = (int64_t)s[0];
+            if (key == EMPTY_KEY) continue;
+            int64_t d = table_upsert(M.m_keys, M.CM, key, M.ring.err);
+            if (d < 0) continue;
+            uint64_t src_state[2] = {s[1], 0};
+            atomic_merge(M.m_state + (size_t)d * M.agg.n_aggs * 2,
+                         src_state, M.agg);
+            continue;
+        }
         int64_t key = M.ring.keys[(size_t)p * M.ring.C + slot];
         if (key == EMPTY_KEY) continue;
         int64_t d = table_upsert(M.m_keys, M.CM, key, M.ring.err);
@@ -902,7 +1084,7 @@ struct MergeFusedArgs {
     unsigned long long *n_out;
 };
 
-template <int SLOTS>
+template <int SLOTS, bool PACKED = false>
 __global__ void __launch_bounds__(256)
 k_merge_fused(MergeFusedArgs M) {
     __shared__ int64_t lkey[SLOTS];
@@ -920,11 +1102,23 @@ k_merge_fused(MergeFusedArgs M) {
     uint32_t a = blockIdx.x * MF_RANGE;          /* gridDim.x == C/MF_RANGE */
     uint32_t span = MF_RANGE + MAX_PROBES;       /* displacement overscan */
     for (int p = 0; p < M.n_src; p++) {
-        const int64_t *keys = M.ring.keys + (size_t)M.src[p] * C;
-        const uint64_t *st = M.ring.state + (size_t)M.src[p] * C * na * 2;
+        const int64_t *keys =
+            PACKED ? nullptr : M.ring.keys + (size_t)M.src[p] * C;
+        const uint64_t *st =
+            PACKED ? nullptr : M.ring.state + (size_t)M.src[p] * C * na * 2;
+        const uint64_t *pslots =
+            PACKED ? M.ring.slots + (size_t)M.src[p] * C * 2 : nullptr;
         for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
             uint32_t idx = (a + t) & mask;
-            int64_t key = keys[idx];
+            uint64_t pw0 = 0;
+            int64_t key;
+            if (PACKED) {
+                ulonglong2 sv = ((const ulonglong2 *)pslots)[idx];
+                key = (int64_t)sv.x;
+                pw0 = sv.y;
+            } else {
+                key = keys[idx];
+            }
             if (key == EMPTY_KEY) continue;
             uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
             if (rel >= MF_RANGE) continue;       /* another WG owns it */
@@ -947,6 +1141,11 @@ k_merge_fused(MergeFusedArgs M) {
             }
             if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; continue; }
             uint64_t *d = lst + (size_t)slot * na * 2;
+            if (PACKED) {   /* single COUNT state, read above */
+                atomicAdd((unsigned long long *)&d[0],
+                          (unsigned long long)pw0);
+                continue;
+            }
             for (int ag = 0; ag < na; ag++) {
                 switch (M.agg.op[ag]) {
                 case AMD_AGG_COUNT:
@@ -1079,6 +1278,8 @@ k_merge_fused(MergeFusedArgs M) {
 
 template __global__ void k_merge_fused<1024>(MergeFusedArgs);
 template __global__ void k_merge_fused<2048>(MergeFusedArgs);
+template __global__ void k_merge_fused<1024, true>(MergeFusedArgs);
+template __global__ void k_merge_fused<2048, true>(MergeFusedArgs);
 
 /* restore checkpointed partial states: insert raw state rows into a pane. */
 struct RestoreArgs {
@@ -1118,6 +1319,14 @@ k_restore(RestoreArgs R) {
         if (key == EMPTY_KEY) {
             atomicExch(&R.ring.spec_used[p], 1u);
             st = R.ring.spec_state + (size_t)p * R.agg.n_aggs * 2;
+        } else if (R.ring.packed) {
+            uint64_t *c = packed_upsert(
+                R.ring.slots + (size_t)p * R.ring.C * 2, R.ring.C, key,
+                R.ring.err);
+            if (c)
+                atomicAdd((unsigned long long *)c,
+                          (unsigned long long)enc[0]);
+            continue;
         } else {
             int64_t *keys = R.ring.keys + (size_t)p * R.ring.C;
             int64_t s = table_upsert(keys, R.ring.C, key, R.ring.err);
@@ -1280,11 +1489,21 @@ static thread_local char g_err[512];
 
 static int ring_retire(GpuOp *o, uint32_t slot) {
     size_t na = o->agg.n_aggs;
-    HIP_CHECK(o, hipMemsetAsync(o->ring.keys + (size_t)slot * o->ring.C, 0xFF,
-                                (size_t)o->ring.C * 8, o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->ring.state +
-                                    (size_t)slot * o->ring.C * na * 2,
-                                0, (size_t)o->ring.C * na * 16, o->stream));
+    if (o->ring.packed) {
+        int blocks = (int)((o->ring.C + 255) / 256);
+        if (blocks > 1024) blocks = 1024;
+        hipLaunchKernelGGL(k_retire_packed, dim3(blocks), dim3(256), 0,
+                           o->stream,
+                           o->ring.slots + (size_t)slot * o->ring.C * 2,
+                           (int64_t)o->ring.C);
+    } else {
+        HIP_CHECK(o, hipMemsetAsync(o->ring.keys + (size_t)slot * o->ring.C,
+                                    0xFF, (size_t)o->ring.C * 8, o->stream));
+        HIP_CHECK(o, hipMemsetAsync(o->ring.state +
+                                        (size_t)slot * o->ring.C * na * 2,
+                                    0, (size_t)o->ring.C * na * 16,
+                                    o->stream));
+    }
     hipLaunchKernelGGL(k_retire_meta, dim3(1), dim3(1), 0, o->stream,
                        o->ring.tag + slot, o->ring.spec_used + slot,
                        o->ring.spec_state + (size_t)slot * na * 2,
@@ -1365,12 +1584,26 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         delete o;
         return nullptr;
     };
+    /* packed AoS table for the keyed single-COUNT shape (the q5 headline):
+     * {key, count} per 16 B slot; see k_update_packed */
+    o->ring.packed = (o->cfg.n_keys == 1 && o->agg.n_aggs == 1 &&
+                      o->agg.op[0] == AMD_AGG_COUNT && !o->use_radix &&
+                      o->kmode == 0);
+    if (const char *ev = getenv("ARROYO_AMD_PACKED"))
+        if (!atoi(ev)) o->ring.packed = 0;
     hipError_t e;
 #define ALLOC(p, bytes)                                                      \
     if ((e = hipMalloc((void **)&(p), (bytes))) != hipSuccess)               \
         return fail(#p, e);
-    ALLOC(o->ring.keys, (size_t)o->ring.R * o->ring.C * 8);
-    ALLOC(o->ring.state, (size_t)o->ring.R * o->ring.C * na * 16);
+    if (o->ring.packed) {
+        ALLOC(o->ring.slots, (size_t)o->ring.R * o->ring.C * 16);
+        o->ring.keys = nullptr;
+        o->ring.state = nullptr;
+    } else {
+        ALLOC(o->ring.keys, (size_t)o->ring.R * o->ring.C * 8);
+        ALLOC(o->ring.state, (size_t)o->ring.R * o->ring.C * na * 16);
+        o->ring.slots = nullptr;
+    }
     ALLOC(o->d_status, (2 + (size_t)o->ring.R) * 8);
     o->ring.err = (int *)o->d_status;
     o->ring.min_bin = o->d_status + 1;
@@ -1393,8 +1626,14 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
 #undef ALLOC
-    hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
-    hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
+    if (o->ring.packed) {
+        int64_t n_slots = (int64_t)o->ring.R * o->ring.C;
+        hipLaunchKernelGGL(k_retire_packed, dim3(1024), dim3(256), 0, 0,
+                           o->ring.slots, n_slots);
+    } else {
+        hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
+        hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
+    }
     hipMemset(o->ring.tag, 0xFF, (size_t)o->ring.R * 8);
     hipMemset(o->ring.spec_used, 0, (size_t)o->ring.R * 4);
     hipMemset(o->ring.spec_state, 0, (size_t)o->ring.R * na * 16);
@@ -1532,6 +1771,30 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
          * harvest, so pool scan order is fine) */
         ev = (int)(o->ev_inflight.size());
         hipEventRecord(o->ev_pool[ev].first, o->stream);
+    }
+    if (o->ring.packed) {
+        bool pvec = n_rows >= 2 && ((uintptr_t)A.ts_col & 15) == 0 &&
+                    ((uintptr_t)A.key_col & 15) == 0;
+        int64_t punits = pvec ? (n_rows + 1) / 2 : n_rows;
+        int64_t pwant = (punits + 255) / 256;
+        int pcap = 1024;
+        if (const char *ev2 = getenv("ARROYO_AMD_PBLOCKS")) pcap = atoi(ev2);
+        int pblocks = (int)(pwant > pcap ? pcap : (pwant < 1 ? 1 : pwant));
+        if (o->force_blocks > 0) pblocks = o->force_blocks;
+        if (pvec)
+            hipLaunchKernelGGL(k_update_packed<true>, dim3(pblocks),
+                               dim3(256), 0, o->stream, A);
+        else
+            hipLaunchKernelGGL(k_update_packed<false>, dim3(pblocks),
+                               dim3(256), 0, o->stream, A);
+        if (sample) {
+            hipEventRecord(o->ev_pool[ev].second, o->stream);
+            o->ev_inflight.push_back(ev);
+        }
+        HIP_CHECK(o, hipGetLastError());
+        o->update_rows += n_rows;
+        o->launches++;
+        return 0;
     }
     int slots = 1024;
     if (const char *e = getenv("ARROYO_AMD_LDS_SLOTS")) slots = atoi(e);
@@ -1742,7 +2005,16 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             if (const char *ev = getenv("ARROYO_AMD_MF_SLOTS"))
                 mfs = atoi(ev);
             size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) * na * 16;
-            if (mfs >= 2048)
+            if (o->ring.packed) {
+                if (mfs >= 2048)
+                    hipLaunchKernelGGL((k_merge_fused<2048, true>),
+                                       dim3(o->ring.C / MF_RANGE), dim3(256),
+                                       shmem, o->stream, M);
+                else
+                    hipLaunchKernelGGL((k_merge_fused<1024, true>),
+                                       dim3(o->ring.C / MF_RANGE), dim3(256),
+                                       shmem, o->stream, M);
+            } else if (mfs >= 2048)
                 hipLaunchKernelGGL(k_merge_fused<2048>,
                                    dim3(o->ring.C / MF_RANGE), dim3(256),
                                    shmem, o->stream, M);
@@ -1913,8 +2185,10 @@ static int build_out(GpuOp *o, AmdOutBatch *out, int raw_states) {
     out->n_cols = ncols;
     out->cols = (void **)calloc(ncols, sizeof(void *));
     out->is_f64 = (int32_t *)calloc(ncols, sizeof(int32_t));
+    if (!out->cols || !out->is_f64) goto oom;
     for (int i = 0; i < ncols; i++) {
         out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
+        if (!out->cols[i]) goto oom;
         if (n) memcpy(out->cols[i], o->host_out[i].data(), (size_t)n * 8);
     }
     if (!raw_states) {
@@ -1934,6 +2208,15 @@ static int build_out(GpuOp *o, AmdOutBatch *out, int raw_states) {
     }
     for (auto &v : o->host_out) v.clear();
     return 0;
+oom:
+    /* loud error per the library convention, never a null-deref */
+    if (out->cols)
+        for (int i = 0; i < ncols; i++) free(out->cols[i]);
+    free(out->cols);
+    free(out->is_f64);
+    memset(out, 0, sizeof *out);
+    snprintf(o->err_msg, sizeof o->err_msg, "build_out: host allocation failed");
+    return 1;
 }
 
 API int arroyo_amd_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
@@ -2110,7 +2393,12 @@ API int arroyo_amd_restore(void *h, const int64_t *const *cols,
 
 API void arroyo_amd_free_out(AmdOutBatch *out) {
     if (!out) return;
-    for (int i = 0; i < out->n_cols; i++) free(out->cols[i]);
+    for (int i = 0; i < out->n_cols; i++) {
+        if (out->on_device)
+            hipFree(out->cols[i]);   /* device-resident emission (ABI doc) */
+        else
+            free(out->cols[i]);
+    }
     free(out->cols);
     free(out->is_f64);
     memset(out, 0, sizeof *out);
@@ -2127,6 +2415,7 @@ API void arroyo_amd_destroy(void *h) {
     }
     hipFree(o->ring.keys);
     hipFree(o->ring.state);
+    hipFree(o->ring.slots);
     hipFree(o->d_status);
     hipHostFree(o->h_status);
     hipFree(o->ring.spec_used);

@@ -639,6 +639,13 @@ API int arroyo_amd_expjoin_restore(void *h, int32_t side,
     if (ej_copy_in(o, side, cols, n_cols, n_rows, 0,
                    cutoff ? cutoff : 0))
         return 1;
+    /* carry the restored watermark so a post-restore expire() uses it
+     * (documented restore semantics; was a silent no-op until the next
+     * handle_watermark) */
+    if (has_watermark) {
+        o->has_wm = 1;
+        if (watermark_nanos > o->wm) o->wm = watermark_nanos;
+    }
     return ej_check_err(o);
 }
 

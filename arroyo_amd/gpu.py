@@ -28,8 +28,11 @@ HIPCC_CMD = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
 def build(force=False):
     if force or not os.path.exists(_SO) or \
             os.path.getmtime(_SO) < max(os.path.getmtime(s) for s in _SRCS):
-        subprocess.run(HIPCC_CMD, check=True, cwd=_DIR,
-                       capture_output=True, text=True)
+        proc = subprocess.run(HIPCC_CMD, cwd=_DIR,
+                              capture_output=True, text=True)
+        if proc.returncode != 0:
+            raise RuntimeError(
+                f"hipcc failed (rc {proc.returncode}):\n{proc.stderr}")
     return _SO
 
 
@@ -109,8 +112,11 @@ class GpuWindowOp(WindowOp):
         rows = ctypes.c_int64()
         launches = ctypes.c_int64()
         emitted = ctypes.c_int64()
-        lib().arroyo_amd_perf(self._h, ctypes.byref(ms), ctypes.byref(rows),
-                              ctypes.byref(launches), ctypes.byref(emitted))
+        rc = lib().arroyo_amd_perf(self._h, ctypes.byref(ms),
+                                   ctypes.byref(rows), ctypes.byref(launches),
+                                   ctypes.byref(emitted))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
         return {"update_ms": ms.value, "rows": rows.value,
                 "launches": launches.value,
                 "emitted_device_rows": emitted.value}
