@@ -628,8 +628,26 @@ __device__ inline uint64_t *packed_upsert(uint64_t *slots, uint32_t C,
 
 #define HOT_MIN 4   /* combine a wave group only when >= this many lanes */
 
+/* find-or-claim the count word for (pane p, key).  PACKED: one {key,count}
+ * line (fewest line touches, but count atomics keep the probe lines dirty).
+ * split (!PACKED): probe the read-only key array (stays clean and
+ * L2-cacheable under concurrent updates) and add to the separate state
+ * array. */
+template <bool PACKED>
+__device__ inline uint64_t *count_upsert(const DeviceRing &ring, uint32_t p,
+                                         int64_t key) {
+    if (PACKED)
+        return packed_upsert(ring.slots + (size_t)p * ring.C * 2, ring.C,
+                             key, ring.err);
+    int64_t s = table_upsert(ring.keys + (size_t)p * ring.C, ring.C, key,
+                             ring.err);
+    if (s < 0) return nullptr;
+    return ring.state + ((size_t)p * ring.C + (size_t)s) * 2;
+}
+
 /* one row of the packed path; called by every active lane together (the
  * ballots inside require it), rows that drop out carry need=false */
+template <bool PACKED>
 __device__ inline void packed_row(const UpdateArgs &A, uint64_t traw,
                                   int64_t key, uint64_t &local_min,
                                   uint64_t &last_bin, int64_t &last_key,
@@ -672,9 +690,7 @@ __device__ inline void packed_row(const UpdateArgs &A, uint64_t traw,
         int cnt = __popcll((long long)grp);
         if (cnt >= HOT_MIN) {
             if (mine && lane == (int)(__ffsll((long long)grp) - 1)) {
-                uint64_t *c = packed_upsert(
-                    A.ring.slots + (size_t)pp * A.ring.C * 2, A.ring.C, k0,
-                    A.ring.err);
+                uint64_t *c = count_upsert<PACKED>(A.ring, pp, k0);
                 if (c) {
                     atomicAdd((unsigned long long *)c,
                               (unsigned long long)cnt);
@@ -688,8 +704,7 @@ __device__ inline void packed_row(const UpdateArgs &A, uint64_t traw,
         }
     }
     if (need) {
-        uint64_t *c = packed_upsert(A.ring.slots + (size_t)p * A.ring.C * 2,
-                                    A.ring.C, key, A.ring.err);
+        uint64_t *c = count_upsert<PACKED>(A.ring, p, key);
         if (c) {
             atomicAdd((unsigned long long *)c, 1ULL);
             last_key = key;
@@ -699,7 +714,7 @@ __device__ inline void packed_row(const UpdateArgs &A, uint64_t traw,
     }
 }
 
-template <bool VEC>
+template <bool VEC, bool PACKED = true>
 __global__ void __launch_bounds__(256)
 k_update_packed(UpdateArgs A) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -713,27 +728,29 @@ k_update_packed(UpdateArgs A) {
              v < n2; v += stride) {
             ulonglong2 tsv = ((const ulonglong2 *)A.ts_col)[v];
             ulonglong2 kv = ((const ulonglong2 *)A.key_col)[v];
-            packed_row(A, tsv.x, (int64_t)kv.x, local_min, last_bin,
-                       last_key, last_pane, last_cnt);
-            packed_row(A, tsv.y, (int64_t)kv.y, local_min, last_bin,
-                       last_key, last_pane, last_cnt);
+            packed_row<PACKED>(A, tsv.x, (int64_t)kv.x, local_min, last_bin,
+                               last_key, last_pane, last_cnt);
+            packed_row<PACKED>(A, tsv.y, (int64_t)kv.y, local_min, last_bin,
+                               last_key, last_pane, last_cnt);
         }
-        if ((A.n_rows & 1) && blockIdx.x == 0 && (threadIdx.x & 63) == 0 &&
-            threadIdx.x == 0)
-            packed_row(A, (uint64_t)A.ts_col[A.n_rows - 1],
-                       A.key_col[A.n_rows - 1], local_min, last_bin,
-                       last_key, last_pane, last_cnt);
+        if ((A.n_rows & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+            packed_row<PACKED>(A, (uint64_t)A.ts_col[A.n_rows - 1],
+                               A.key_col[A.n_rows - 1], local_min, last_bin,
+                               last_key, last_pane, last_cnt);
     } else {
         for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
              i < A.n_rows; i += stride)
-            packed_row(A, (uint64_t)A.ts_col[i], A.key_col[i], local_min,
-                       last_bin, last_key, last_pane, last_cnt);
+            packed_row<PACKED>(A, (uint64_t)A.ts_col[i], A.key_col[i],
+                               local_min, last_bin, last_key, last_pane,
+                               last_cnt);
     }
     fold_min_bin(local_min, A.ring.min_bin);
 }
 
-template __global__ void k_update_packed<true>(UpdateArgs);
-template __global__ void k_update_packed<false>(UpdateArgs);
+template __global__ void k_update_packed<true, true>(UpdateArgs);
+template __global__ void k_update_packed<false, true>(UpdateArgs);
+template __global__ void k_update_packed<true, false>(UpdateArgs);
+template __global__ void k_update_packed<false, false>(UpdateArgs);
 
 /* packed pane clear: {EMPTY_KEY, 0} per slot (a single memset cannot set
  * the two words differently, and count must start at 0) */
@@ -1448,6 +1465,7 @@ struct GpuOp {
     std::set<uint64_t> table_bins;        /* ExpiringTimeKeyView keys */
 
     int use_lds;
+    int upd_kind;              /* 0 lds, 1 packed AoS, 2 split wave-combine */
     int use_radix;             /* ARROYO_AMD_RADIX=1: partitioned update */
     int64_t *rdx_key, *rdx_ts; /* scatter scratch (lazily sized) */
     unsigned int *rdx_hist;
@@ -1584,13 +1602,23 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         delete o;
         return nullptr;
     };
-    /* packed AoS table for the keyed single-COUNT shape (the q5 headline):
-     * {key, count} per 16 B slot; see k_update_packed */
-    o->ring.packed = (o->cfg.n_keys == 1 && o->agg.n_aggs == 1 &&
-                      o->agg.op[0] == AMD_AGG_COUNT && !o->use_radix &&
-                      o->kmode == 0);
-    if (const char *ev = getenv("ARROYO_AMD_PACKED"))
-        if (!atoi(ev)) o->ring.packed = 0;
+    /* update-kernel flavor for the keyed single-COUNT shape (the q5
+     * headline): "split" = wave-combine no-LDS kernel over the standard
+     * split key/state arrays (default; key probes stay L2-clean),
+     * "packed" = the same kernel over an AoS {key,count} table,
+     * "lds" = the round-1 LDS-staged kernel.  See k_update_packed. */
+    bool count_shape = (o->cfg.n_keys == 1 && o->agg.n_aggs == 1 &&
+                        o->agg.op[0] == AMD_AGG_COUNT && !o->use_radix &&
+                        o->kmode == 0);
+    o->upd_kind = count_shape ? 2 : 0;
+    if (const char *ev = getenv("ARROYO_AMD_UPD")) {
+        if (!strcmp(ev, "lds")) o->upd_kind = 0;
+        else if (count_shape && !strcmp(ev, "packed")) o->upd_kind = 1;
+        else if (count_shape && !strcmp(ev, "split")) o->upd_kind = 2;
+    }
+    if (const char *ev = getenv("ARROYO_AMD_PACKED"))   /* legacy alias */
+        if (!atoi(ev)) o->upd_kind = 0;
+    o->ring.packed = o->upd_kind == 1;
     hipError_t e;
 #define ALLOC(p, bytes)                                                      \
     if ((e = hipMalloc((void **)&(p), (bytes))) != hipSuccess)               \
@@ -1772,7 +1800,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         ev = (int)(o->ev_inflight.size());
         hipEventRecord(o->ev_pool[ev].first, o->stream);
     }
-    if (o->ring.packed) {
+    if (o->upd_kind != 0) {
         bool pvec = n_rows >= 2 && ((uintptr_t)A.ts_col & 15) == 0 &&
                     ((uintptr_t)A.key_col & 15) == 0;
         int64_t punits = pvec ? (n_rows + 1) / 2 : n_rows;
@@ -1781,12 +1809,21 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         if (const char *ev2 = getenv("ARROYO_AMD_PBLOCKS")) pcap = atoi(ev2);
         int pblocks = (int)(pwant > pcap ? pcap : (pwant < 1 ? 1 : pwant));
         if (o->force_blocks > 0) pblocks = o->force_blocks;
-        if (pvec)
-            hipLaunchKernelGGL(k_update_packed<true>, dim3(pblocks),
-                               dim3(256), 0, o->stream, A);
-        else
-            hipLaunchKernelGGL(k_update_packed<false>, dim3(pblocks),
-                               dim3(256), 0, o->stream, A);
+        if (o->upd_kind == 1) {
+            if (pvec)
+                hipLaunchKernelGGL((k_update_packed<true, true>),
+                                   dim3(pblocks), dim3(256), 0, o->stream, A);
+            else
+                hipLaunchKernelGGL((k_update_packed<false, true>),
+                                   dim3(pblocks), dim3(256), 0, o->stream, A);
+        } else {
+            if (pvec)
+                hipLaunchKernelGGL((k_update_packed<true, false>),
+                                   dim3(pblocks), dim3(256), 0, o->stream, A);
+            else
+                hipLaunchKernelGGL((k_update_packed<false, false>),
+                                   dim3(pblocks), dim3(256), 0, o->stream, A);
+        }
         if (sample) {
             hipEventRecord(o->ev_pool[ev].second, o->stream);
             o->ev_inflight.push_back(ev);

@@ -299,6 +299,19 @@ def _oracle_run(key, ts):
     op.close()
 
 
+def _oracle_shard_worker(key, ts, min_rows, q):
+    """N-core leg worker: self-timed steady-state passes over its shard
+    (process start/import/alloc overhead excluded — the measured quantity
+    is per-core operator throughput, matching the reference's long-running
+    subtask model)."""
+    passes = max(1, -(-min_rows // max(len(key), 1)))
+    t0 = time.perf_counter()
+    for _ in range(passes):
+        _oracle_run(key, ts)
+    t1 = time.perf_counter()
+    q.put((passes * len(key), t1 - t0))
+
+
 def physical_cores():
     try:
         import psutil
@@ -325,34 +338,41 @@ def cpu_baseline():
     one_core = CPU_SAMPLE_ROWS / (t1 - t0)
 
     ncores = physical_cores()
-    elapsed_n = None
+    value_n = one_core
     if ncores > 1:
         from arroyo_amd.shuffle import partition_ids
         pid = partition_ids(key, ncores)
-        shards = []
+        q = mp.Queue()
+        # each worker replays its shard until it has done >= min_rows of
+        # work, self-timing only the operator loop (not fork/import), so
+        # the N-core figure is steady-state per-core throughput x N
+        min_rows = max(2_000_000, CPU_SAMPLE_ROWS // ncores)
+        procs = []
         for p in range(ncores):
             m = pid == p
-            shards.append((key[m], ts[m]))
-        procs = [mp.Process(target=_oracle_run, args=s) for s in shards]
-        tn0 = time.perf_counter()
+            procs.append(mp.Process(target=_oracle_shard_worker,
+                                    args=(key[m], ts[m], min_rows, q)))
         for pr in procs:
             pr.start()
+        rates = []
+        for _ in procs:
+            rows_done, secs = q.get()
+            rates.append(rows_done / secs)
         for pr in procs:
             pr.join()
-        tn1 = time.perf_counter()
-        elapsed_n = tn1 - tn0
-    value_n = CPU_SAMPLE_ROWS / elapsed_n if elapsed_n else one_core
+        # whole-job rate with every shard running concurrently = the
+        # slowest per-shard rate x N (shards are balanced by key hash)
+        value_n = min(rates) * ncores
     return {
         "value": value_n,
         "unit": "rows/s",
-        "cores": ncores if elapsed_n else 1,
+        "cores": ncores,
         "kind": "port",
         "value_1core": one_core,
         "sample": f"{CPU_SAMPLE_ROWS} rows of the same nexmark q5 stream, "
                   f"oracle/arroyo_oracle.c: single-thread {t1 - t0:.1f}s; "
-                  f"key-hash-sharded across {ncores} physical cores "
-                  f"{elapsed_n:.1f}s" if elapsed_n else
-                  f"{CPU_SAMPLE_ROWS} rows, single-thread {t1 - t0:.1f}s",
+                  f"N-core = min over {ncores} key-hash shards of "
+                  f"self-timed steady-state rate x {ncores}",
     }
 
 
