@@ -752,10 +752,272 @@ template __global__ void k_update_packed<false, true>(UpdateArgs);
 template __global__ void k_update_packed<true, false>(UpdateArgs);
 template __global__ void k_update_packed<false, false>(UpdateArgs);
 
+/* ------------------------------------------------------------------ */
+/* Batched-probe update kernel (upd_kind 3, the default for the keyed
+ * single-COUNT shape).  Measurement drove this shape: the wave-combine
+ * variant's per-row ballots made every row cost the wave max(lane
+ * latencies) — the wave could not start row i+1 until the slowest cold
+ * lane's global probe for row i landed (107 us vs the LDS kernel's 71 us
+ * per 1.85M rows).  Here there is NO cross-lane coordination in the row
+ * loop at all: each thread takes 4 consecutive rows per iteration,
+ * collapses sequential same-key runs (the nexmark hot key arrives in
+ * runs), checks its register cache, then issues the remaining rows'
+ * first probe loads back-to-back (independent -> memory-level
+ * parallelism 4) before resolving any of them.  Count updates are
+ * no-return atomicAdds (fire-and-forget): same-address hot-key traffic
+ * queues at the memory-side cache without stalling the wave (~9 adds/us
+ * needed vs the measured ~88/us single-address ceiling). */
+
+__device__ inline uint64_t *probe_resolve(int64_t *keys, uint64_t *state,
+                                          uint32_t C, int *err, int64_t key,
+                                          uint64_t i, int64_t kk) {
+    /* `kk` is the already-loaded key at probe position `i`; keys/state are
+     * the pane's planes */
+    uint64_t m = C - 1;
+    uint32_t lim = C < MAX_PROBES ? C : MAX_PROBES;
+    for (uint32_t probes = 0; probes < lim; probes++) {
+        if (kk == key) return state + i * 2;
+        if (kk == EMPTY_KEY) {
+            int64_t old = (int64_t)atomicCAS((unsigned long long *)&keys[i],
+                                             (unsigned long long)EMPTY_KEY,
+                                             (unsigned long long)key);
+            if (old == EMPTY_KEY || old == key) return state + i * 2;
+        }
+        i = (i + 1) & m;
+        kk = keys[i];
+    }
+    *err = ERR_TABLE_FULL;
+    return nullptr;
+}
+
+/* scalar one-row path (tail rows, misaligned columns; noinline keeps the
+ * hot kernel's CFG small — an inlined copy ICEd clang-22's simplifycfg) */
+__device__ __noinline__ void batch_row_scalar(const UpdateArgs &A, uint64_t traw,
+                                        int64_t key, uint64_t &local_min,
+                                        uint64_t &last_bin, int64_t &last_key,
+                                        uint32_t &last_pane,
+                                        uint64_t *&last_cnt) {
+    uint64_t t = traw + A.ts_offset;
+    uint64_t q = div_slide(t, A.slide, A.slide_inv);
+    uint64_t bin = q * A.slide;
+    if (A.has_wm && bin < A.wm_bin) return;
+    if (bin < local_min) local_min = bin;
+    uint32_t p = (uint32_t)(q & (A.ring.R - 1));
+    if (bin != last_bin) {
+        claim_tag(&A.ring.tag[p], bin, A.ring.err);
+        last_bin = bin;
+    }
+    if (key == EMPTY_KEY) {
+        atomicExch(&A.ring.spec_used[p], 1u);
+        atomicAdd((unsigned long long *)(A.ring.spec_state + (size_t)p * 2),
+                  1ULL);
+        return;
+    }
+    if (key == last_key && p == last_pane && last_cnt) {
+        atomicAdd((unsigned long long *)last_cnt, 1ULL);
+        return;
+    }
+    uint64_t h = hash64((uint64_t)key) & (A.ring.C - 1);
+    int64_t kk = A.ring.keys[(size_t)p * A.ring.C + h];
+    uint64_t *c = probe_resolve(A.ring.keys + (size_t)p * A.ring.C,
+                                A.ring.state + (size_t)p * A.ring.C * 2,
+                                A.ring.C, A.ring.err, key, h, kk);
+    if (c) {
+        atomicAdd((unsigned long long *)c, 1ULL);
+        last_key = key;
+        last_pane = p;
+        last_cnt = c;
+    }
+}
+
+/* phase A of one row: bin, late-drop, min fold, tag claim, sentinel-key
+ * special case.  Returns false when the row needs no table update. */
+__device__ inline bool row_prep(const UpdateArgs &A, uint64_t traw,
+                                int64_t key, uint64_t &local_min,
+                                uint64_t &last_bin, uint32_t &pane) {
+    uint64_t t = traw + A.ts_offset;
+    uint64_t q = div_slide(t, A.slide, A.slide_inv);
+    uint64_t bin = q * A.slide;
+    if (A.has_wm && bin < A.wm_bin) return false;
+    if (bin < local_min) local_min = bin;
+    pane = (uint32_t)(q & (A.ring.R - 1));
+    if (bin != last_bin) {
+        claim_tag(&A.ring.tag[pane], bin, A.ring.err);
+        last_bin = bin;
+    }
+    if (key == EMPTY_KEY) {
+        atomicExch(&A.ring.spec_used[pane], 1u);
+        atomicAdd((unsigned long long *)(A.ring.spec_state + (size_t)pane * 2),
+                  1ULL);
+        return false;
+    }
+    return true;
+}
+
+template <int Q>
+__global__ void __launch_bounds__(256)
+k_update_batch(UpdateArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_pane = PANE_UNSET;
+    uint64_t *last_cnt = nullptr;
+    int64_t nq = A.n_rows / Q;
+    const uint64_t m = A.ring.C - 1;
+    for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nq;
+         v += stride) {
+        /* row pair 0/1 (and 2/3 when Q == 4), written longhand: arrays +
+         * unrolled loops here ICEd clang-22's simplifycfg */
+        ulonglong2 tp0 = ((const ulonglong2 *)A.ts_col)[(Q / 2) * v];
+        ulonglong2 kp0 = ((const ulonglong2 *)A.key_col)[(Q / 2) * v];
+        ulonglong2 tp1 = tp0, kp1 = kp0;
+        if (Q == 4) {
+            tp1 = ((const ulonglong2 *)A.ts_col)[(Q / 2) * v + 1];
+            kp1 = ((const ulonglong2 *)A.key_col)[(Q / 2) * v + 1];
+        }
+        int64_t k0 = (int64_t)kp0.x, k1 = (int64_t)kp0.y;
+        int64_t k2 = (int64_t)kp1.x, k3 = (int64_t)kp1.y;
+        uint32_t p0 = 0, p1 = 0, p2 = 0, p3 = 0;
+        bool n0 = row_prep(A, tp0.x, k0, local_min, last_bin, p0);
+        bool n1 = row_prep(A, tp0.y, k1, local_min, last_bin, p1);
+        bool n2 = false, n3 = false;
+        if (Q == 4) {
+            n2 = row_prep(A, tp1.x, k2, local_min, last_bin, p2);
+            n3 = row_prep(A, tp1.y, k3, local_min, last_bin, p3);
+        }
+        /* collapse sequential same-key runs (the hot key arrives in runs) */
+        uint32_t c0 = 1, c1 = 1, c2 = 1, c3 = 1;
+        if (Q == 4 && n3 && n2 && k3 == k2 && p3 == p2) {
+            c2 += c3;
+            n3 = false;
+        }
+        if (Q == 4 && n2 && n1 && k2 == k1 && p2 == p1) {
+            c1 += c2;
+            n2 = false;
+        }
+        if (n1 && n0 && k1 == k0 && p1 == p0) {
+            c0 += c1;
+            n1 = false;
+        }
+        /* register cache from the previous quad */
+        if (n0 && k0 == last_key && p0 == last_pane && last_cnt) {
+            atomicAdd((unsigned long long *)last_cnt,
+                      (unsigned long long)c0);
+            n0 = false;
+        }
+        if (n1 && k1 == last_key && p1 == last_pane && last_cnt) {
+            atomicAdd((unsigned long long *)last_cnt,
+                      (unsigned long long)c1);
+            n1 = false;
+        }
+        if (Q == 4 && n2 && k2 == last_key && p2 == last_pane && last_cnt) {
+            atomicAdd((unsigned long long *)last_cnt,
+                      (unsigned long long)c2);
+            n2 = false;
+        }
+        if (Q == 4 && n3 && k3 == last_key && p3 == last_pane && last_cnt) {
+            atomicAdd((unsigned long long *)last_cnt,
+                      (unsigned long long)c3);
+            n3 = false;
+        }
+        /* issue every remaining row's first probe load before resolving
+         * any of them (memory-level parallelism across the quad) */
+        uint64_t h0 = 0, h1 = 0, h2 = 0, h3 = 0;
+        int64_t f0 = 0, f1 = 0, f2 = 0, f3 = 0;
+        if (n0) {
+            h0 = hash64((uint64_t)k0) & m;
+            f0 = A.ring.keys[(size_t)p0 * A.ring.C + h0];
+        }
+        if (n1) {
+            h1 = hash64((uint64_t)k1) & m;
+            f1 = A.ring.keys[(size_t)p1 * A.ring.C + h1];
+        }
+        if (Q == 4 && n2) {
+            h2 = hash64((uint64_t)k2) & m;
+            f2 = A.ring.keys[(size_t)p2 * A.ring.C + h2];
+        }
+        if (Q == 4 && n3) {
+            h3 = hash64((uint64_t)k3) & m;
+            f3 = A.ring.keys[(size_t)p3 * A.ring.C + h3];
+        }
+        if (n0) {
+            uint64_t *c = probe_resolve(
+                A.ring.keys + (size_t)p0 * A.ring.C,
+                A.ring.state + (size_t)p0 * A.ring.C * 2, A.ring.C,
+                A.ring.err, k0, h0, f0);
+            if (c) {
+                atomicAdd((unsigned long long *)c, (unsigned long long)c0);
+                last_key = k0;
+                last_pane = p0;
+                last_cnt = c;
+            }
+        }
+        if (n1) {
+            uint64_t *c = probe_resolve(
+                A.ring.keys + (size_t)p1 * A.ring.C,
+                A.ring.state + (size_t)p1 * A.ring.C * 2, A.ring.C,
+                A.ring.err, k1, h1, f1);
+            if (c) {
+                atomicAdd((unsigned long long *)c, (unsigned long long)c1);
+                last_key = k1;
+                last_pane = p1;
+                last_cnt = c;
+            }
+        }
+        if (Q == 4 && n2) {
+            uint64_t *c = probe_resolve(
+                A.ring.keys + (size_t)p2 * A.ring.C,
+                A.ring.state + (size_t)p2 * A.ring.C * 2, A.ring.C,
+                A.ring.err, k2, h2, f2);
+            if (c) {
+                atomicAdd((unsigned long long *)c, (unsigned long long)c2);
+                last_key = k2;
+                last_pane = p2;
+                last_cnt = c;
+            }
+        }
+        if (Q == 4 && n3) {
+            uint64_t *c = probe_resolve(
+                A.ring.keys + (size_t)p3 * A.ring.C,
+                A.ring.state + (size_t)p3 * A.ring.C * 2, A.ring.C,
+                A.ring.err, k3, h3, f3);
+            if (c) {
+                atomicAdd((unsigned long long *)c, (unsigned long long)c3);
+                last_key = k3;
+                last_pane = p3;
+                last_cnt = c;
+            }
+        }
+    }
+    /* tail rows (n_rows % Q) are handled by a separate scalar launch from
+     * the host: a device-side tail call forced caller-save scratch spills
+     * into this kernel's hot loop */
+    fold_min_bin(local_min, A.ring.min_bin);
+}
+
+template __global__ void k_update_batch<2>(UpdateArgs);
+template __global__ void k_update_batch<4>(UpdateArgs);
+
+/* scalar fallback of the batch kind (misaligned column pointers) */
+__global__ void __launch_bounds__(256)
+k_update_batch_scalar(UpdateArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_pane = PANE_UNSET;
+    uint64_t *last_cnt = nullptr;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < A.n_rows; i += stride)
+        batch_row_scalar(A, (uint64_t)A.ts_col[i], A.key_col[i], local_min,
+                         last_bin, last_key, last_pane, last_cnt);
+    fold_min_bin(local_min, A.ring.min_bin);
+}
+
 /* packed pane clear: {EMPTY_KEY, 0} per slot (a single memset cannot set
  * the two words differently, and count must start at 0) */
 __global__ void __launch_bounds__(256)
-k_retire_packed(uint64_t *slots, int64_t n_slots) {
+k_retire_packed(uint64_t *slots, int64_t n_slots, uint64_t *tag,
+                uint32_t *spec_used, uint64_t *spec_state, int na2) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < n_slots; i += stride) {
@@ -763,6 +1025,36 @@ k_retire_packed(uint64_t *slots, int64_t n_slots) {
         v.x = (unsigned long long)EMPTY_KEY;
         v.y = 0;
         ((ulonglong2 *)slots)[i] = v;
+    }
+    if (tag && blockIdx.x == 0 && threadIdx.x == 0) {
+        *tag = EMPTY_TAG;
+        *spec_used = 0;
+        for (int w = 0; w < na2; w++) spec_state[w] = 0;
+    }
+}
+
+/* one-launch pane retire for the split/unpacked layout: keys to 0xFF,
+ * states to 0, scalar metadata — replaces two fill launches plus a
+ * metadata kernel (~5 us each on a busy stream) per retire */
+__global__ void __launch_bounds__(256)
+k_retire_all(int64_t *keys, uint64_t *state, int64_t C, int na2,
+             uint64_t *tag, uint32_t *spec_used, uint64_t *spec_state) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t kv2 = C >> 1;          /* C is a power of two >= 2 */
+    ulonglong2 ff, zz;
+    ff.x = ff.y = ~0ULL;
+    zz.x = zz.y = 0;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < kv2; i += stride)
+        ((ulonglong2 *)keys)[i] = ff;
+    int64_t sv2 = (C * na2) >> 1;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < sv2; i += stride)
+        ((ulonglong2 *)state)[i] = zz;
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        *tag = EMPTY_TAG;
+        *spec_used = 0;
+        for (int w = 0; w < na2; w++) spec_state[w] = 0;
     }
 }
 
@@ -1097,8 +1389,12 @@ struct MergeFusedArgs {
     uint32_t src[64];
     uint64_t win_start, win_end;
     int32_t n_keys;
+    uint32_t range;     /* home slots per WG; gridDim.x == C / range */
     int64_t *out[16];
     unsigned long long *n_out;
+    unsigned long long *accum;  /* running emitted-rows accumulator (may be
+                                   null); folded in here so the separate
+                                   k_accum launch (~5 us) is not needed */
 };
 
 template <int SLOTS, bool PACKED = false>
@@ -1116,8 +1412,8 @@ k_merge_fused(MergeFusedArgs M) {
     }
     if (threadIdx.x == 0) blk_cnt = 0;
     __syncthreads();
-    uint32_t a = blockIdx.x * MF_RANGE;          /* gridDim.x == C/MF_RANGE */
-    uint32_t span = MF_RANGE + MAX_PROBES;       /* displacement overscan */
+    uint32_t a = blockIdx.x * M.range;           /* gridDim.x == C/range */
+    uint32_t span = M.range + MAX_PROBES;        /* displacement overscan */
     for (int p = 0; p < M.n_src; p++) {
         const int64_t *keys =
             PACKED ? nullptr : M.ring.keys + (size_t)M.src[p] * C;
@@ -1125,20 +1421,10 @@ k_merge_fused(MergeFusedArgs M) {
             PACKED ? nullptr : M.ring.state + (size_t)M.src[p] * C * na * 2;
         const uint64_t *pslots =
             PACKED ? M.ring.slots + (size_t)M.src[p] * C * 2 : nullptr;
-        for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
-            uint32_t idx = (a + t) & mask;
-            uint64_t pw0 = 0;
-            int64_t key;
-            if (PACKED) {
-                ulonglong2 sv = ((const ulonglong2 *)pslots)[idx];
-                key = (int64_t)sv.x;
-                pw0 = sv.y;
-            } else {
-                key = keys[idx];
-            }
-            if (key == EMPTY_KEY) continue;
+        auto fold = [&](uint32_t idx, int64_t key, uint64_t pw0) {
+            if (key == EMPTY_KEY) return;
             uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
-            if (rel >= MF_RANGE) continue;       /* another WG owns it */
+            if (rel >= M.range) return;          /* another WG owns it */
             /* LDS upsert */
             uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u) &
                          (SLOTS - 1);
@@ -1156,12 +1442,12 @@ k_merge_fused(MergeFusedArgs M) {
                     if (old == EMPTY_KEY || old == key) slot = sidx;
                 }
             }
-            if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; continue; }
+            if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; return; }
             uint64_t *d = lst + (size_t)slot * na * 2;
-            if (PACKED) {   /* single COUNT state, read above */
+            if (PACKED) {   /* single COUNT state, read with the key */
                 atomicAdd((unsigned long long *)&d[0],
                           (unsigned long long)pw0);
-                continue;
+                return;
             }
             for (int ag = 0; ag < na; ag++) {
                 switch (M.agg.op[ag]) {
@@ -1187,6 +1473,23 @@ k_merge_fused(MergeFusedArgs M) {
                     break;
                 }
             }
+        };
+        if (PACKED) {
+            for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
+                uint32_t idx = (a + t) & mask;
+                ulonglong2 sv = ((const ulonglong2 *)pslots)[idx];
+                fold(idx, (int64_t)sv.x, sv.y);
+            }
+        } else {
+            /* 2 keys per 16 B load: a and span are multiples of 2, and an
+             * even idx never straddles the wrap, so the pair load is safe */
+            for (uint32_t t = 2 * threadIdx.x; t < span;
+                 t += 2 * blockDim.x) {
+                uint32_t idx = (a + t) & mask;
+                ulonglong2 kv = *(const ulonglong2 *)&keys[idx];
+                fold(idx, (int64_t)kv.x, 0);
+                fold(idx + 1, (int64_t)kv.y, 0);
+            }
         }
     }
     __syncthreads();
@@ -1202,6 +1505,8 @@ k_merge_fused(MergeFusedArgs M) {
         blk_base = blk_cnt ? atomicAdd(M.n_out,
                                        (unsigned long long)blk_cnt)
                            : 0;
+        if (M.accum && blk_cnt)
+            atomicAdd(M.accum, (unsigned long long)blk_cnt);
         blk_cnt = 0;
     }
     __syncthreads();
@@ -1266,6 +1571,7 @@ k_merge_fused(MergeFusedArgs M) {
         }
         if (any) {
             int64_t r = (int64_t)atomicAdd(M.n_out, 1ULL);
+            if (M.accum) atomicAdd(M.accum, 1ULL);
             int col = 0;
             if (M.n_keys) M.out[col++][r] = EMPTY_KEY;
             for (int ag = 0; ag < na; ag++) {
@@ -1507,25 +1813,25 @@ static thread_local char g_err[512];
 
 static int ring_retire(GpuOp *o, uint32_t slot) {
     size_t na = o->agg.n_aggs;
+    int blocks = (int)((o->ring.C + 255) / 256);
+    if (blocks > 1024) blocks = 1024;
     if (o->ring.packed) {
-        int blocks = (int)((o->ring.C + 255) / 256);
-        if (blocks > 1024) blocks = 1024;
         hipLaunchKernelGGL(k_retire_packed, dim3(blocks), dim3(256), 0,
                            o->stream,
                            o->ring.slots + (size_t)slot * o->ring.C * 2,
-                           (int64_t)o->ring.C);
+                           (int64_t)o->ring.C, o->ring.tag + slot,
+                           o->ring.spec_used + slot,
+                           o->ring.spec_state + (size_t)slot * na * 2,
+                           (int)(na * 2));
     } else {
-        HIP_CHECK(o, hipMemsetAsync(o->ring.keys + (size_t)slot * o->ring.C,
-                                    0xFF, (size_t)o->ring.C * 8, o->stream));
-        HIP_CHECK(o, hipMemsetAsync(o->ring.state +
-                                        (size_t)slot * o->ring.C * na * 2,
-                                    0, (size_t)o->ring.C * na * 16,
-                                    o->stream));
+        hipLaunchKernelGGL(k_retire_all, dim3(blocks), dim3(256), 0,
+                           o->stream,
+                           o->ring.keys + (size_t)slot * o->ring.C,
+                           o->ring.state + (size_t)slot * o->ring.C * na * 2,
+                           (int64_t)o->ring.C, (int)(na * 2),
+                           o->ring.tag + slot, o->ring.spec_used + slot,
+                           o->ring.spec_state + (size_t)slot * na * 2);
     }
-    hipLaunchKernelGGL(k_retire_meta, dim3(1), dim3(1), 0, o->stream,
-                       o->ring.tag + slot, o->ring.spec_used + slot,
-                       o->ring.spec_state + (size_t)slot * na * 2,
-                       (int)(na * 2));
     HIP_CHECK(o, hipGetLastError());
     return 0;
 }
@@ -1610,11 +1916,12 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     bool count_shape = (o->cfg.n_keys == 1 && o->agg.n_aggs == 1 &&
                         o->agg.op[0] == AMD_AGG_COUNT && !o->use_radix &&
                         o->kmode == 0);
-    o->upd_kind = count_shape ? 2 : 0;
+    o->upd_kind = count_shape ? 3 : 0;
     if (const char *ev = getenv("ARROYO_AMD_UPD")) {
         if (!strcmp(ev, "lds")) o->upd_kind = 0;
         else if (count_shape && !strcmp(ev, "packed")) o->upd_kind = 1;
         else if (count_shape && !strcmp(ev, "split")) o->upd_kind = 2;
+        else if (count_shape && !strcmp(ev, "batch")) o->upd_kind = 3;
     }
     if (const char *ev = getenv("ARROYO_AMD_PACKED"))   /* legacy alias */
         if (!atoi(ev)) o->upd_kind = 0;
@@ -1657,7 +1964,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (o->ring.packed) {
         int64_t n_slots = (int64_t)o->ring.R * o->ring.C;
         hipLaunchKernelGGL(k_retire_packed, dim3(1024), dim3(256), 0, 0,
-                           o->ring.slots, n_slots);
+                           o->ring.slots, n_slots, (uint64_t *)nullptr,
+                           (uint32_t *)nullptr, (uint64_t *)nullptr, 0);
     } else {
         hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
         hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
@@ -1803,13 +2111,38 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     if (o->upd_kind != 0) {
         bool pvec = n_rows >= 2 && ((uintptr_t)A.ts_col & 15) == 0 &&
                     ((uintptr_t)A.key_col & 15) == 0;
-        int64_t punits = pvec ? (n_rows + 1) / 2 : n_rows;
+        int64_t punits = pvec ? (n_rows + 1) / (o->upd_kind == 3 ? 4 : 2)
+                              : n_rows;
         int64_t pwant = (punits + 255) / 256;
         int pcap = 1024;
         if (const char *ev2 = getenv("ARROYO_AMD_PBLOCKS")) pcap = atoi(ev2);
         int pblocks = (int)(pwant > pcap ? pcap : (pwant < 1 ? 1 : pwant));
         if (o->force_blocks > 0) pblocks = o->force_blocks;
-        if (o->upd_kind == 1) {
+        if (o->upd_kind == 3) {
+            int bq = 4;
+            if (const char *ev3 = getenv("ARROYO_AMD_BQ")) bq = atoi(ev3);
+            if (!pvec) bq = 1;
+            if (bq >= 4)
+                hipLaunchKernelGGL(k_update_batch<4>, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            else if (bq >= 2)
+                hipLaunchKernelGGL(k_update_batch<2>, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            else
+                hipLaunchKernelGGL(k_update_batch_scalar, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            int64_t tail = bq > 1 ? A.n_rows % bq : 0;
+            if (tail) {
+                /* quad/pair kernels skip the last n_rows % bq rows; finish
+                 * them with a tiny scalar launch (same parity) */
+                UpdateArgs T = A;
+                T.key_col = A.key_col + (A.n_rows - tail);
+                T.ts_col = A.ts_col + (A.n_rows - tail);
+                T.n_rows = tail;
+                hipLaunchKernelGGL(k_update_batch_scalar, dim3(1), dim3(64),
+                                   0, o->stream, T);
+            }
+        } else if (o->upd_kind == 1) {
             if (pvec)
                 hipLaunchKernelGGL((k_update_packed<true, true>),
                                    dim3(pblocks), dim3(256), 0, o->stream, A);
@@ -2035,38 +2368,45 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             for (int i = 0; i < o->n_out_alloc && i < 16; i++)
                 M.out[i] = o->d_out[i];
             M.n_out = o->d_n_out;
-            /* 1024 slots won the sweep (3->4 workgroups per CU); typical
-             * occupancy is ~320 owned keys per 256-slot home range and
-             * overflow is a loud error, not silent corruption */
+            /* emitted-row accounting folded into the kernel for the
+             * device-resident path (saves the separate k_accum launch) */
+            M.accum = o->cfg.emit_to_host ? nullptr : o->d_emitted;
+            /* 1024 slots won the round-1 sweep (3->4 workgroups per CU);
+             * typical occupancy is ~320 owned keys per 256-slot home range
+             * and overflow is a loud error, not silent corruption.  Larger
+             * ranges cut the overscan ratio (range+256)/range but need the
+             * 2048-slot table. */
             int mfs = 1024;
             if (const char *ev = getenv("ARROYO_AMD_MF_SLOTS"))
                 mfs = atoi(ev);
+            uint32_t range = MF_RANGE;
+            if (const char *ev2 = getenv("ARROYO_AMD_MF_RANGE"))
+                range = (uint32_t)atoi(ev2);
+            if (range < 2) range = 2;
+            if (range > o->ring.C) range = o->ring.C;
+            while (range & (range - 1)) range &= range - 1;  /* pow2 floor */
+            M.range = range;
             size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) * na * 16;
             if (o->ring.packed) {
                 if (mfs >= 2048)
                     hipLaunchKernelGGL((k_merge_fused<2048, true>),
-                                       dim3(o->ring.C / MF_RANGE), dim3(256),
+                                       dim3(o->ring.C / range), dim3(256),
                                        shmem, o->stream, M);
                 else
                     hipLaunchKernelGGL((k_merge_fused<1024, true>),
-                                       dim3(o->ring.C / MF_RANGE), dim3(256),
+                                       dim3(o->ring.C / range), dim3(256),
                                        shmem, o->stream, M);
             } else if (mfs >= 2048)
                 hipLaunchKernelGGL(k_merge_fused<2048>,
-                                   dim3(o->ring.C / MF_RANGE), dim3(256),
+                                   dim3(o->ring.C / range), dim3(256),
                                    shmem, o->stream, M);
             else
                 hipLaunchKernelGGL(k_merge_fused<1024>,
-                                   dim3(o->ring.C / MF_RANGE), dim3(256),
+                                   dim3(o->ring.C / range), dim3(256),
                                    shmem, o->stream, M);
             HIP_CHECK(o, hipGetLastError());
         }
-        if (!o->cfg.emit_to_host) {
-            hipLaunchKernelGGL(k_accum, dim3(1), dim3(1), 0, o->stream,
-                               o->d_emitted, o->d_n_out);
-            HIP_CHECK(o, hipGetLastError());
-            return 0;
-        }
+        if (!o->cfg.emit_to_host) return 0;
         unsigned long long n = 0;
         HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8,
                                     hipMemcpyDeviceToHost, o->stream));
