@@ -854,9 +854,50 @@ __device__ inline bool row_prep(const UpdateArgs &A, uint64_t traw,
     return true;
 }
 
+/* tiny per-block hot cache (direct-mapped LDS): absorbs the wave-uniform
+ * nexmark hot key, whose same-address global atomics otherwise serialize
+ * the whole chip (measured: 200-1000 us/launch, worse with more blocks).
+ * 256 slots -> the flush is only 256 upserts per block (the 1024-slot
+ * staging table's flush was the old LDS kernel's main global cost).
+ * Returns true when the row was absorbed. */
+#define BATCH_LSLOTS 256
+__device__ inline bool lds_hot_try(int64_t *ls_key, uint32_t *ls_pane,
+                                   unsigned long long *ls_cnt, int64_t key,
+                                   uint32_t pane, uint32_t cnt) {
+    uint32_t s = (uint32_t)hash64((uint64_t)key * 2654435761u + pane) &
+                 (BATCH_LSLOTS - 1);
+    int64_t lk = ls_key[s];
+    if (lk == EMPTY_KEY) {
+        int64_t old = (int64_t)atomicCAS((unsigned long long *)&ls_key[s],
+                                         (unsigned long long)EMPTY_KEY,
+                                         (unsigned long long)key);
+        if (old == EMPTY_KEY) {
+            ls_pane[s] = pane;
+            atomicAdd(&ls_cnt[s], (unsigned long long)cnt);
+            return true;
+        }
+        lk = old;
+    }
+    /* a stale PANE_UNSET read sends the row to the global path — safe */
+    if (lk == key && ls_pane[s] == pane) {
+        atomicAdd(&ls_cnt[s], (unsigned long long)cnt);
+        return true;
+    }
+    return false;
+}
+
 template <int Q>
 __global__ void __launch_bounds__(256)
 k_update_batch(UpdateArgs A) {
+    __shared__ int64_t ls_key[BATCH_LSLOTS];
+    __shared__ uint32_t ls_pane[BATCH_LSLOTS];
+    __shared__ unsigned long long ls_cnt[BATCH_LSLOTS];
+    for (int i = threadIdx.x; i < BATCH_LSLOTS; i += blockDim.x) {
+        ls_key[i] = EMPTY_KEY;
+        ls_pane[i] = PANE_UNSET;
+        ls_cnt[i] = 0;
+    }
+    __syncthreads();
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
     int64_t last_key = EMPTY_KEY;
@@ -920,6 +961,15 @@ k_update_batch(UpdateArgs A) {
                       (unsigned long long)c3);
             n3 = false;
         }
+        /* per-block hot cache (LDS) */
+        if (n0 && lds_hot_try(ls_key, ls_pane, ls_cnt, k0, p0, c0))
+            n0 = false;
+        if (n1 && lds_hot_try(ls_key, ls_pane, ls_cnt, k1, p1, c1))
+            n1 = false;
+        if (Q == 4 && n2 && lds_hot_try(ls_key, ls_pane, ls_cnt, k2, p2, c2))
+            n2 = false;
+        if (Q == 4 && n3 && lds_hot_try(ls_key, ls_pane, ls_cnt, k3, p3, c3))
+            n3 = false;
         /* issue every remaining row's first probe load before resolving
          * any of them (memory-level parallelism across the quad) */
         uint64_t h0 = 0, h1 = 0, h2 = 0, h3 = 0;
@@ -993,6 +1043,19 @@ k_update_batch(UpdateArgs A) {
      * the host: a device-side tail call forced caller-save scratch spills
      * into this kernel's hot loop */
     fold_min_bin(local_min, A.ring.min_bin);
+    /* flush the hot cache: <= BATCH_LSLOTS upserts per block */
+    __syncthreads();
+    for (int s = threadIdx.x; s < BATCH_LSLOTS; s += blockDim.x) {
+        int64_t key = ls_key[s];
+        if (key == EMPTY_KEY) continue;
+        uint32_t p = ls_pane[s];
+        uint64_t h = hash64((uint64_t)key) & m;
+        int64_t kk = A.ring.keys[(size_t)p * A.ring.C + h];
+        uint64_t *c = probe_resolve(A.ring.keys + (size_t)p * A.ring.C,
+                                    A.ring.state + (size_t)p * A.ring.C * 2,
+                                    A.ring.C, A.ring.err, key, h, kk);
+        if (c) atomicAdd((unsigned long long *)c, ls_cnt[s]);
+    }
 }
 
 template __global__ void k_update_batch<2>(UpdateArgs);
