@@ -842,7 +842,11 @@ __device__ inline bool row_prep(const UpdateArgs &A, uint64_t traw,
     if (bin < local_min) local_min = bin;
     pane = (uint32_t)(q & (A.ring.R - 1));
     if (bin != last_bin) {
-        claim_tag(&A.ring.tag[pane], bin, A.ring.err);
+        /* one leader lane claims for the wave: every thread's FIRST row
+         * takes this path, and a per-thread volatile tag read serialized
+         * the whole grid on one address (measured: launch time grew
+         * linearly with block count) */
+        claim_tag_wave(A.ring.tag, pane, bin, EMPTY_TAG, A.ring.err);
         last_bin = bin;
     }
     if (key == EMPTY_KEY) {
