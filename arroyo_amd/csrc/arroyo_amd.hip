@@ -1065,6 +1065,116 @@ k_update_batch(UpdateArgs A) {
 template __global__ void k_update_batch<2>(UpdateArgs);
 template __global__ void k_update_batch<4>(UpdateArgs);
 
+/* array/loop form of the same kernel for wider quads (Q = 8): with the
+ * helpers taking primitive arguments the ICE that forced the longhand
+ * form above does not trigger.  87% of the narrow kernel's wave time is
+ * parked on memory waits (SQ_WAIT_ANY, profiles/r02e) — wider quads put
+ * more independent probe loads in flight per thread. */
+template <int Q>
+__global__ void __launch_bounds__(256)
+k_update_batch_n(UpdateArgs A) {
+    __shared__ int64_t ls_key[BATCH_LSLOTS];
+    __shared__ uint32_t ls_pane[BATCH_LSLOTS];
+    __shared__ unsigned long long ls_cnt[BATCH_LSLOTS];
+    for (int i = threadIdx.x; i < BATCH_LSLOTS; i += blockDim.x) {
+        ls_key[i] = EMPTY_KEY;
+        ls_pane[i] = PANE_UNSET;
+        ls_cnt[i] = 0;
+    }
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    int64_t last_key = EMPTY_KEY;
+    uint32_t last_pane = PANE_UNSET;
+    uint64_t *last_cnt = nullptr;
+    int64_t nq = A.n_rows / Q;
+    const uint64_t m = A.ring.C - 1;
+    for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nq;
+         v += stride) {
+        int64_t key[Q];
+        uint64_t traw[Q];
+#pragma unroll
+        for (int j = 0; j < Q / 2; j++) {
+            ulonglong2 tp = ((const ulonglong2 *)A.ts_col)[(Q / 2) * v + j];
+            ulonglong2 kp = ((const ulonglong2 *)A.key_col)[(Q / 2) * v + j];
+            traw[2 * j] = tp.x;
+            traw[2 * j + 1] = tp.y;
+            key[2 * j] = (int64_t)kp.x;
+            key[2 * j + 1] = (int64_t)kp.y;
+        }
+        bool need[Q];
+        uint32_t pane[Q];
+        uint32_t cnt[Q];
+#pragma unroll
+        for (int j = 0; j < Q; j++) {
+            pane[j] = 0;
+            need[j] = row_prep(A, traw[j], key[j], local_min, last_bin,
+                               pane[j]);
+            cnt[j] = 1;
+        }
+#pragma unroll
+        for (int j = Q - 1; j >= 1; j--)
+            if (need[j] && need[j - 1] && key[j] == key[j - 1] &&
+                pane[j] == pane[j - 1]) {
+                cnt[j - 1] += cnt[j];
+                need[j] = false;
+            }
+#pragma unroll
+        for (int j = 0; j < Q; j++)
+            if (need[j] && key[j] == last_key && pane[j] == last_pane &&
+                last_cnt) {
+                atomicAdd((unsigned long long *)last_cnt,
+                          (unsigned long long)cnt[j]);
+                need[j] = false;
+            }
+#pragma unroll
+        for (int j = 0; j < Q; j++)
+            if (need[j] &&
+                lds_hot_try(ls_key, ls_pane, ls_cnt, key[j], pane[j],
+                            cnt[j]))
+                need[j] = false;
+        uint64_t h[Q];
+        int64_t firstk[Q];
+#pragma unroll
+        for (int j = 0; j < Q; j++)
+            if (need[j]) {
+                h[j] = hash64((uint64_t)key[j]) & m;
+                firstk[j] = A.ring.keys[(size_t)pane[j] * A.ring.C + h[j]];
+            }
+#pragma unroll
+        for (int j = 0; j < Q; j++) {
+            if (!need[j]) continue;
+            uint64_t *c = probe_resolve(
+                A.ring.keys + (size_t)pane[j] * A.ring.C,
+                A.ring.state + (size_t)pane[j] * A.ring.C * 2, A.ring.C,
+                A.ring.err, key[j], h[j], firstk[j]);
+            if (c) {
+                atomicAdd((unsigned long long *)c,
+                          (unsigned long long)cnt[j]);
+                last_key = key[j];
+                last_pane = pane[j];
+                last_cnt = c;
+            }
+        }
+    }
+    fold_min_bin(local_min, A.ring.min_bin);
+    __syncthreads();
+    for (int s = threadIdx.x; s < BATCH_LSLOTS; s += blockDim.x) {
+        int64_t key = ls_key[s];
+        if (key == EMPTY_KEY) continue;
+        uint32_t p = ls_pane[s];
+        uint64_t h = hash64((uint64_t)key) & m;
+        int64_t kk = A.ring.keys[(size_t)p * A.ring.C + h];
+        uint64_t *c = probe_resolve(A.ring.keys + (size_t)p * A.ring.C,
+                                    A.ring.state + (size_t)p * A.ring.C * 2,
+                                    A.ring.C, A.ring.err, key, h, kk);
+        if (c) atomicAdd((unsigned long long *)c, ls_cnt[s]);
+    }
+}
+
+template __global__ void k_update_batch_n<8>(UpdateArgs);
+template __global__ void k_update_batch_n<16>(UpdateArgs);
+
 /* scalar fallback of the batch kind (misaligned column pointers) */
 __global__ void __launch_bounds__(256)
 k_update_batch_scalar(UpdateArgs A) {
@@ -2189,7 +2299,13 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
             int bq = 4;
             if (const char *ev3 = getenv("ARROYO_AMD_BQ")) bq = atoi(ev3);
             if (!pvec) bq = 1;
-            if (bq >= 4)
+            if (bq >= 16)
+                hipLaunchKernelGGL(k_update_batch_n<16>, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            else if (bq >= 8)
+                hipLaunchKernelGGL(k_update_batch_n<8>, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            else if (bq >= 4)
                 hipLaunchKernelGGL(k_update_batch<4>, dim3(pblocks),
                                    dim3(256), 0, o->stream, A);
             else if (bq >= 2)
