@@ -1175,6 +1175,157 @@ k_update_batch_n(UpdateArgs A) {
 template __global__ void k_update_batch_n<8>(UpdateArgs);
 template __global__ void k_update_batch_n<16>(UpdateArgs);
 
+/* ------------------------------------------------------------------ */
+/* Radix-regroup update path (upd_kind 4).  The batched kernel is bound
+ * by random-access THROUGHPUT: ~50 MB/launch of 64B-granule table RMWs
+ * (cold keys have no block-local reuse — their ~7 occurrences land in
+ * different blocks).  This path converts that into streaming:
+ *   pass 1  histogram rows into 1024 hash-buckets, per (bucket, block)
+ *   scan    exclusive sum (hipCUB) -> every block's private output range
+ *   pass 2  scatter (key, pane) to bucket segments — LDS cursors only,
+ *           no global atomics
+ *   pass 3  one block per bucket aggregates its segment in LDS with
+ *           EXCLUSIVE ownership of its keys, then writes each distinct
+ *           (key, pane) once to the pane table
+ * Round 1 tried a 256-bucket version and measured 6x slower; the fixes
+ * here: 1024 buckets (the changing hot key spreads over ~1200 distinct
+ * keys per launch, so buckets stay balanced), per-(bucket,block)
+ * reservations instead of global scatter cursors, and the aggregation
+ * table sized so a bucket's ~130 distinct keys never overflow. */
+#define RDX2_LOG_B 10
+#define RDX2_B (1u << RDX2_LOG_B)
+#define RDX2_HB 1024               /* hist/scatter grid blocks */
+#define RDX2_ASLOTS 2048           /* aggregation LDS slots per bucket */
+
+struct Rdx2Args {
+    const int64_t *key_col;
+    const int64_t *ts_col;
+    int64_t n_rows;
+    uint64_t slide, slide_inv;
+    uint64_t wm_bin;
+    int has_wm;
+    uint64_t ts_offset;
+    DeviceRing ring;
+    uint32_t *hist;       /* [RDX2_B * RDX2_HB + 1] */
+    int64_t *skey;        /* [cap] scatter output */
+    uint32_t *spane;      /* [cap] */
+};
+
+__global__ void __launch_bounds__(256)
+k_rdx2_hist(Rdx2Args A) {
+    __shared__ uint32_t cnt[RDX2_B];
+    for (int i = threadIdx.x; i < (int)RDX2_B; i += blockDim.x) cnt[i] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint64_t local_min = ~0ULL, last_bin = EMPTY_TAG;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < A.n_rows; i += stride) {
+        uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
+        uint64_t q = div_slide(t, A.slide, A.slide_inv);
+        uint64_t bin = q * A.slide;
+        if (A.has_wm && bin < A.wm_bin) continue;
+        if (bin < local_min) local_min = bin;
+        uint32_t p = (uint32_t)(q & (A.ring.R - 1));
+        if (bin != last_bin) {
+            claim_tag_wave(A.ring.tag, p, bin, EMPTY_TAG, A.ring.err);
+            last_bin = bin;
+        }
+        int64_t key = A.key_col[i];
+        if (key == EMPTY_KEY) {   /* sentinel: handled here, not scattered */
+            atomicExch(&A.ring.spec_used[p], 1u);
+            atomicAdd((unsigned long long *)
+                          (A.ring.spec_state + (size_t)p * 2), 1ULL);
+            continue;
+        }
+        atomicAdd(&cnt[(uint32_t)(hash64((uint64_t)key) >>
+                                  (64 - RDX2_LOG_B))], 1u);
+    }
+    fold_min_bin(local_min, A.ring.min_bin);
+    __syncthreads();
+    for (int i = threadIdx.x; i < (int)RDX2_B; i += blockDim.x)
+        A.hist[(size_t)i * RDX2_HB + blockIdx.x] = cnt[i];
+}
+
+__global__ void __launch_bounds__(256)
+k_rdx2_scatter(Rdx2Args A) {
+    __shared__ uint32_t cur[RDX2_B];
+    for (int i = threadIdx.x; i < (int)RDX2_B; i += blockDim.x)
+        cur[i] = A.hist[(size_t)i * RDX2_HB + blockIdx.x];
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < A.n_rows; i += stride) {
+        uint64_t t = (uint64_t)A.ts_col[i] + A.ts_offset;
+        uint64_t q = div_slide(t, A.slide, A.slide_inv);
+        uint64_t bin = q * A.slide;
+        if (A.has_wm && bin < A.wm_bin) continue;
+        int64_t key = A.key_col[i];
+        if (key == EMPTY_KEY) continue;
+        uint32_t b = (uint32_t)(hash64((uint64_t)key) >> (64 - RDX2_LOG_B));
+        uint32_t pos = atomicAdd(&cur[b], 1u);
+        A.skey[pos] = key;
+        A.spane[pos] = (uint32_t)(q & (A.ring.R - 1));
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_rdx2_agg(Rdx2Args A) {
+    __shared__ int64_t lkey[RDX2_ASLOTS];
+    __shared__ uint32_t lpane[RDX2_ASLOTS];
+    __shared__ unsigned long long lcnt[RDX2_ASLOTS];
+    for (int i = threadIdx.x; i < RDX2_ASLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        lpane[i] = PANE_UNSET;
+        lcnt[i] = 0;
+    }
+    __syncthreads();
+    uint32_t b = blockIdx.x;             /* gridDim.x == RDX2_B */
+    int64_t lo = A.hist[(size_t)b * RDX2_HB];
+    int64_t hi = A.hist[(size_t)(b + 1) * RDX2_HB];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        int64_t key = A.skey[i];
+        uint32_t p = A.spane[i];
+        uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u + p) &
+                     (RDX2_ASLOTS - 1);
+        for (int pr = 0; pr < RDX2_ASLOTS; pr++) {
+            uint32_t s = (h + pr) & (RDX2_ASLOTS - 1);
+            int64_t k = lkey[s];
+            bool claimed = false;
+            if (k == EMPTY_KEY) {
+                int64_t old = (int64_t)atomicCAS(
+                    (unsigned long long *)&lkey[s],
+                    (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+                if (old == EMPTY_KEY) {
+                    lpane[s] = p;
+                    claimed = true;
+                    k = key;
+                } else {
+                    k = old;
+                }
+            }
+            if (k == key && (claimed || lpane[s] == p)) {
+                atomicAdd(&lcnt[s], 1ULL);
+                h = ~0u;   /* found */
+                break;
+            }
+        }
+        if (h != ~0u) *A.ring.err = ERR_TABLE_FULL;  /* bucket overflow */
+    }
+    __syncthreads();
+    /* exclusive ownership: each distinct (key, pane) written once */
+    for (int s = threadIdx.x; s < RDX2_ASLOTS; s += blockDim.x) {
+        int64_t key = lkey[s];
+        if (key == EMPTY_KEY) continue;
+        uint32_t p = lpane[s];
+        uint64_t h = hash64((uint64_t)key) & (A.ring.C - 1);
+        int64_t kk = A.ring.keys[(size_t)p * A.ring.C + h];
+        uint64_t *c = probe_resolve(A.ring.keys + (size_t)p * A.ring.C,
+                                    A.ring.state + (size_t)p * A.ring.C * 2,
+                                    A.ring.C, A.ring.err, key, h, kk);
+        if (c) atomicAdd((unsigned long long *)c, lcnt[s]);
+    }
+}
+
 /* scalar fallback of the batch kind (misaligned column pointers) */
 __global__ void __launch_bounds__(256)
 k_update_batch_scalar(UpdateArgs A) {
@@ -1948,7 +2099,14 @@ struct GpuOp {
     std::set<uint64_t> table_bins;        /* ExpiringTimeKeyView keys */
 
     int use_lds;
-    int upd_kind;              /* 0 lds, 1 packed AoS, 2 split wave-combine */
+    int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
+                       3 batched-probe, 4 radix-regroup */
+    int64_t *rdx2_skey;
+    uint32_t *rdx2_spane;
+    uint32_t *rdx2_hist;
+    void *rdx2_tmp;
+    size_t rdx2_tmp_bytes;
+    int64_t rdx2_cap;
     int use_radix;             /* ARROYO_AMD_RADIX=1: partitioned update */
     int64_t *rdx_key, *rdx_ts; /* scatter scratch (lazily sized) */
     unsigned int *rdx_hist;
@@ -2099,6 +2257,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         else if (count_shape && !strcmp(ev, "packed")) o->upd_kind = 1;
         else if (count_shape && !strcmp(ev, "split")) o->upd_kind = 2;
         else if (count_shape && !strcmp(ev, "batch")) o->upd_kind = 3;
+        else if (count_shape && !strcmp(ev, "rdx2")) o->upd_kind = 4;
     }
     if (const char *ev = getenv("ARROYO_AMD_PACKED"))   /* legacy alias */
         if (!atoi(ev)) o->upd_kind = 0;
@@ -2284,6 +2443,61 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
          * harvest, so pool scan order is fine) */
         ev = (int)(o->ev_inflight.size());
         hipEventRecord(o->ev_pool[ev].first, o->stream);
+    }
+    if (o->upd_kind == 4) {
+        if (n_rows > o->rdx2_cap) {
+            hipFree(o->rdx2_skey);
+            hipFree(o->rdx2_spane);
+            o->rdx2_cap = n_rows + (n_rows >> 2);
+            HIP_CHECK(o, hipMalloc((void **)&o->rdx2_skey,
+                                   (size_t)o->rdx2_cap * 8));
+            HIP_CHECK(o, hipMalloc((void **)&o->rdx2_spane,
+                                   (size_t)o->rdx2_cap * 4));
+            if (!o->rdx2_hist) {
+                HIP_CHECK(o, hipMalloc((void **)&o->rdx2_hist,
+                                       ((size_t)RDX2_B * RDX2_HB + 1) * 4));
+                o->rdx2_tmp_bytes = 0;
+                hipcub::DeviceScan::ExclusiveSum(
+                    nullptr, o->rdx2_tmp_bytes, o->rdx2_hist, o->rdx2_hist,
+                    RDX2_B * RDX2_HB + 1);
+                HIP_CHECK(o, hipMalloc(&o->rdx2_tmp,
+                                       o->rdx2_tmp_bytes ? o->rdx2_tmp_bytes
+                                                         : 1));
+            }
+        }
+        Rdx2Args R = {};
+        R.key_col = A.key_col;
+        R.ts_col = A.ts_col;
+        R.n_rows = n_rows;
+        R.slide = A.slide;
+        R.slide_inv = A.slide_inv;
+        R.wm_bin = A.wm_bin;
+        R.has_wm = A.has_wm;
+        R.ts_offset = A.ts_offset;
+        R.ring = o->ring;
+        R.hist = o->rdx2_hist;
+        R.skey = o->rdx2_skey;
+        R.spane = o->rdx2_spane;
+        HIP_CHECK(o, hipMemsetAsync(o->rdx2_hist + (size_t)RDX2_B * RDX2_HB,
+                                    0, 4, o->stream));
+        hipLaunchKernelGGL(k_rdx2_hist, dim3(RDX2_HB), dim3(256), 0,
+                           o->stream, R);
+        size_t tmp = o->rdx2_tmp_bytes;
+        hipcub::DeviceScan::ExclusiveSum(o->rdx2_tmp, tmp, o->rdx2_hist,
+                                         o->rdx2_hist, RDX2_B * RDX2_HB + 1,
+                                         o->stream);
+        hipLaunchKernelGGL(k_rdx2_scatter, dim3(RDX2_HB), dim3(256), 0,
+                           o->stream, R);
+        hipLaunchKernelGGL(k_rdx2_agg, dim3(RDX2_B), dim3(256), 0,
+                           o->stream, R);
+        HIP_CHECK(o, hipGetLastError());
+        if (sample) {
+            hipEventRecord(o->ev_pool[ev].second, o->stream);
+            o->ev_inflight.push_back(ev);
+        }
+        o->update_rows += n_rows;
+        o->launches++;
+        return 0;
     }
     if (o->upd_kind != 0) {
         bool pvec = n_rows >= 2 && ((uintptr_t)A.ts_col & 15) == 0 &&
@@ -2988,6 +3202,10 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
     hipFree(o->rdx_tmp);
+    hipFree(o->rdx2_skey);
+    hipFree(o->rdx2_spane);
+    hipFree(o->rdx2_hist);
+    hipFree(o->rdx2_tmp);
     for (int i = 0; i < o->stg.ncols; i++) {
         hipHostFree(o->stg.buf[i]);
         hipFree(o->stg.dbuf[i]);
