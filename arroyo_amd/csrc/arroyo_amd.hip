@@ -2505,12 +2505,12 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         int64_t punits = pvec ? (n_rows + 1) / (o->upd_kind == 3 ? 4 : 2)
                               : n_rows;
         int64_t pwant = (punits + 255) / 256;
-        int pcap = 1024;
+        int pcap = 2048;
         if (const char *ev2 = getenv("ARROYO_AMD_PBLOCKS")) pcap = atoi(ev2);
         int pblocks = (int)(pwant > pcap ? pcap : (pwant < 1 ? 1 : pwant));
         if (o->force_blocks > 0) pblocks = o->force_blocks;
         if (o->upd_kind == 3) {
-            int bq = 4;
+            int bq = 8;
             if (const char *ev3 = getenv("ARROYO_AMD_BQ")) bq = atoi(ev3);
             if (!pvec) bq = 1;
             if (bq >= 16)

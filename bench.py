@@ -3,11 +3,13 @@ rows/sec on nexmark q5 sliding-window aggregate).
 
 A "step" is one pass of the hot path over one 64K-row synthetic nexmark bid
 batch already resident in HBM: the fused bin+hash-aggregate update kernel
-(k_update_lds_vec), plus the watermark-driven window firing (the fused
+(k_update_batch_n, quad-batched probes with a per-block LDS hot cache;
+ARROYO_AMD_UPD selects the measured alternates), plus the
+watermark-driven window firing (the fused
 hash-aligned merge+compact kernel k_merge_fused) at the reference's
 watermark cadence (1/s of event time,
 crates/arroyo-worker/src/arrow/watermark_generator.rs).  Steps are
-submitted in fused watermark periods (BENCH_WM_FUSE, default 2) --
+submitted in fused watermark periods (BENCH_WM_FUSE, default 4) --
 bit-identical outputs, verified by tests/test_property_large.py.  Outputs stay
 device-resident (the next pipeline stage's collector consumes them in place);
 the host-visible emission path is covered by tests, not timed here.
@@ -155,10 +157,9 @@ def run_gpu(args):
     # emit those watermarks in order -- outputs are bit-identical (covered
     # by tests/test_property_large.py), only emission latency grows by
     # WM_FUSE-1 periods of event time.  Larger launches amortize the
-    # ~10 us fixed cost (ramp + LDS init) of the update kernel; the sweep
-    # put the optimum at 2 (beyond that the per-block LDS table spans too
-    # many panes and pre-aggregation hit rate drops).
-    wm_fuse = int(os.environ.get("BENCH_WM_FUSE", "2"))
+    # ~7 us fixed cost (ramp + launch latency) of the update kernel
+    # (round-2 sweeps: optimum 4 for the batched kernel).
+    wm_fuse = int(os.environ.get("BENCH_WM_FUSE", "4"))
 
     def run_span(s_begin, n_steps):
         """single-GPU fast path: multi-batch submits split at fused
