@@ -32,6 +32,7 @@
  * tests/test_expjoin.py.
  */
 #include <hip/hip_runtime.h>
+#include <hipcub/hipcub.hpp>
 
 #include <cstdio>
 #include <cstdlib>
@@ -48,6 +49,7 @@ namespace ejoin {
 #define EJERR_TABLE_FULL 1
 #define EJERR_POOL_FULL  2
 #define EJERR_OUT_CAP    3
+#define EJERR_RETRACT    4
 
 __device__ inline uint64_t hash64(uint64_t x) {
     x += 0x9e3779b97f4a7c15ULL;
@@ -63,6 +65,7 @@ struct SideMap {
     int32_t *next;      /* [R] */
     int64_t *vals;      /* [nv][R] */
     int64_t *ts;        /* [R] */
+    int32_t *cnt;       /* [C+1] live rows per key (updating/outer modes) */
     unsigned long long *cursor; /* pool allocation */
     uint32_t C;
     int64_t R;
@@ -170,6 +173,7 @@ k_ej_insert(InsertArgs I) {
             I.own.vals[(size_t)v * I.own.R + idx] = I.cols[1 + v][r];
         I.own.ts[idx] = ts[r];
         I.own.next[idx] = atomicExch(&I.own.head[slot], (int32_t)idx);
+        atomicAdd(&I.own.cnt[slot], 1);
     }
 }
 
@@ -189,6 +193,7 @@ k_ej_compact(CompactArgs A) {
          slot <= (int64_t)A.src.C; slot += stride) {
         if (slot < (int64_t)A.src.C && A.src.keys[slot] == EMPTY_KEY)
             continue;
+        int32_t kept = 0;
         for (int32_t j = A.src.head[slot]; j >= 0; j = A.src.next[j]) {
             if ((uint64_t)A.src.ts[j] < A.cutoff) continue;
             int64_t idx = (int64_t)atomicAdd(A.dst.cursor, 1ULL);
@@ -198,7 +203,9 @@ k_ej_compact(CompactArgs A) {
                     A.src.vals[(size_t)v * A.src.R + j];
             A.dst.ts[idx] = A.src.ts[j];
             A.dst.next[idx] = atomicExch(&A.dst.head[slot], (int32_t)idx);
+            kept++;
         }
+        A.dst.cnt[slot] = kept;
     }
 }
 
@@ -231,6 +238,226 @@ k_ej_drain(EDrainArgs D) {
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* Updating (retraction-carrying) and LEFT/RIGHT/FULL outer modes
+ * (join_with_expiration.rs:29-131 with the planner's join_type,
+ * plan/join.rs:326-379).  Semantics restated from the CPU oracle
+ * (oracle/arroyo_oracle.c expjoin_insert), which is pinned by the
+ * reference's updating_left/right/full_join goldens: a key's first match
+ * retracts its earlier null-padded row; a retraction removing a key's
+ * last match brings the null rows back.
+ *
+ * Emission order within a key is semantically significant (own-count
+ * transitions decide null-row retractions), so the batch is grouped by
+ * key — claim the own-side slot per row, radix-sort (slot << 32 | row)
+ * — and ONE thread replays each key's rows in row order.  Keys are
+ * independent; the other side's map is read-only during a batch, so
+ * there are no cross-thread races. */
+
+__global__ void __launch_bounds__(256)
+k_ej_slots(const int64_t *keys_col, int64_t n_rows, SideMap own,
+           uint64_t *sortkey, int *err) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < n_rows; r += stride) {
+        int64_t slot = claim_slot(own, keys_col[r], err);
+        if (slot < 0) slot = own.C;  /* table full: err already set */
+        sortkey[r] = ((uint64_t)slot << 32) | (uint32_t)r;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_ej_segs(const uint64_t *sorted, int64_t n, uint32_t *segs,
+          unsigned int *n_segs) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        if (i == 0 || (sorted[i] >> 32) != (sorted[i - 1] >> 32))
+            segs[atomicAdd(n_segs, 1u)] = (uint32_t)i;
+}
+
+/* one output row; lv/rv null => that side absent (oracle eemit2) */
+__device__ inline void ej_emit(int64_t *const *out,
+                               unsigned long long *n_out, int64_t out_cap,
+                               int *err, int nlv, int nrv, int jt, int upd,
+                               int64_t key, const int64_t *lv, int64_t lts,
+                               const int64_t *rv, int64_t rts, int retract) {
+    int64_t o = (int64_t)atomicAdd(n_out, 1ULL);
+    if (o >= out_cap) { *err = EJERR_OUT_CAP; return; }
+    int col = 0;
+    out[col++][o] = key;
+    for (int v = 0; v < nlv; v++) out[col++][o] = lv ? lv[v] : 0;
+    for (int v = 0; v < nrv; v++) out[col++][o] = rv ? rv[v] : 0;
+    out[col++][o] = lv ? (rv ? (lts > rts ? lts : rts) : lts) : rts;
+    if (jt != AMD_JOIN_INNER) {
+        out[col++][o] = lv != nullptr;
+        out[col++][o] = rv != nullptr;
+    }
+    if (jt != AMD_JOIN_INNER || upd) out[col][o] = retract;
+}
+
+struct EjUpdArgs {
+    const int64_t *cols[12];   /* key, vals..., [is_retract,] ts */
+    int32_t nv, upd, jt, side, nlv, nrv;
+    int64_t n_rows;
+    const uint64_t *sorted;
+    const uint32_t *segs;
+    const unsigned int *n_segs;
+    SideMap own, other;
+    int64_t *out[20];
+    unsigned long long *n_out;
+    int64_t out_cap;
+    int *err;
+};
+
+__global__ void __launch_bounds__(256)
+k_ej_upd(EjUpdArgs U) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t *ts_col = U.cols[1 + U.nv + U.upd];
+    const int64_t *retr_col = U.upd ? U.cols[1 + U.nv] : nullptr;
+    const int null_own =
+        U.jt == AMD_JOIN_FULL ||
+        (U.side == 0 ? U.jt == AMD_JOIN_LEFT : U.jt == AMD_JOIN_RIGHT);
+    const int null_other =
+        U.jt == AMD_JOIN_FULL ||
+        (U.side == 0 ? U.jt == AMD_JOIN_RIGHT : U.jt == AMD_JOIN_LEFT);
+    const unsigned int nseg = *U.n_segs;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < (int64_t)nseg; g += stride) {
+        int64_t i0 = U.segs[g];
+        const int64_t slot = (int64_t)(U.sorted[i0] >> 32);
+        int32_t cnt_own = U.own.cnt[slot];
+        for (int64_t i = i0;
+             i < U.n_rows && (int64_t)(U.sorted[i] >> 32) == slot; i++) {
+            int64_t r = (int64_t)(uint32_t)U.sorted[i];
+            int64_t key = U.cols[0][r];
+            int64_t ts = ts_col[r];
+            int64_t mine[8];
+            for (int v = 0; v < U.nv; v++) mine[v] = U.cols[1 + v][r];
+            /* other side: read-only this batch */
+            int64_t oslot = find_slot(U.other, key);
+            int32_t on = oslot >= 0 ? U.other.cnt[oslot] : 0;
+            int32_t ohead = (oslot >= 0 && on > 0) ? U.other.head[oslot] : -1;
+            int64_t ovals[8];
+            const int retr = retr_col ? (int)retr_col[r] : 0;
+            if (!retr) {
+                if (on > 0) {
+                    for (int32_t j = ohead; j >= 0; j = U.other.next[j]) {
+                        for (int v = 0; v < U.other.nv; v++)
+                            ovals[v] =
+                                U.other.vals[(size_t)v * U.other.R + j];
+                        int64_t ots = U.other.ts[j];
+                        if (U.side == 0)
+                            ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                    U.nlv, U.nrv, U.jt, U.upd, key, mine,
+                                    ts, ovals, ots, 0);
+                        else
+                            ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                    U.nlv, U.nrv, U.jt, U.upd, key, ovals,
+                                    ots, mine, ts, 0);
+                    }
+                    if (null_other && cnt_own == 0)
+                        /* the other side's rows were unmatched until now:
+                         * retract their null-padded emissions */
+                        for (int32_t j = ohead; j >= 0; j = U.other.next[j]) {
+                            for (int v = 0; v < U.other.nv; v++)
+                                ovals[v] =
+                                    U.other.vals[(size_t)v * U.other.R + j];
+                            int64_t ots = U.other.ts[j];
+                            if (U.side == 0)
+                                ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                        U.nlv, U.nrv, U.jt, U.upd, key,
+                                        nullptr, 0, ovals, ots, 1);
+                            else
+                                ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                        U.nlv, U.nrv, U.jt, U.upd, key,
+                                        ovals, ots, nullptr, 0, 1);
+                        }
+                } else if (null_own) {
+                    if (U.side == 0)
+                        ej_emit(U.out, U.n_out, U.out_cap, U.err, U.nlv,
+                                U.nrv, U.jt, U.upd, key, mine, ts, nullptr,
+                                0, 0);
+                    else
+                        ej_emit(U.out, U.n_out, U.out_cap, U.err, U.nlv,
+                                U.nrv, U.jt, U.upd, key, nullptr, 0, mine,
+                                ts, 0);
+                }
+                /* append to own chain (single writer for this slot) */
+                int64_t idx = (int64_t)atomicAdd(U.own.cursor, 1ULL);
+                if (idx >= U.own.R) { *U.err = EJERR_POOL_FULL; continue; }
+                for (int v = 0; v < U.nv; v++)
+                    U.own.vals[(size_t)v * U.own.R + idx] = mine[v];
+                U.own.ts[idx] = ts;
+                U.own.next[idx] = U.own.head[slot];
+                U.own.head[slot] = (int32_t)idx;
+                cnt_own++;
+            } else {
+                /* retract: unlink the OLDEST value-equal stored row (the
+                 * chain is LIFO, so the last match along the walk) */
+                int32_t prev = -1, match = -1, mprev = -1;
+                for (int32_t j = U.own.head[slot]; j >= 0;
+                     prev = j, j = U.own.next[j]) {
+                    int eq = 1;
+                    for (int v = 0; v < U.nv && eq; v++)
+                        eq = U.own.vals[(size_t)v * U.own.R + j] == mine[v];
+                    if (eq) { match = j; mprev = prev; }
+                }
+                if (match < 0) { *U.err = EJERR_RETRACT; continue; }
+                int64_t sts = U.own.ts[match];
+                if (on > 0) {
+                    for (int32_t j = ohead; j >= 0; j = U.other.next[j]) {
+                        for (int v = 0; v < U.other.nv; v++)
+                            ovals[v] =
+                                U.other.vals[(size_t)v * U.other.R + j];
+                        int64_t ots = U.other.ts[j];
+                        if (U.side == 0)
+                            ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                    U.nlv, U.nrv, U.jt, U.upd, key, mine,
+                                    sts, ovals, ots, 1);
+                        else
+                            ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                    U.nlv, U.nrv, U.jt, U.upd, key, ovals,
+                                    ots, mine, sts, 1);
+                    }
+                    if (null_other && cnt_own == 1)
+                        /* the other side loses its last match: null rows
+                         * come back */
+                        for (int32_t j = ohead; j >= 0; j = U.other.next[j]) {
+                            for (int v = 0; v < U.other.nv; v++)
+                                ovals[v] =
+                                    U.other.vals[(size_t)v * U.other.R + j];
+                            int64_t ots = U.other.ts[j];
+                            if (U.side == 0)
+                                ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                        U.nlv, U.nrv, U.jt, U.upd, key,
+                                        nullptr, 0, ovals, ots, 0);
+                            else
+                                ej_emit(U.out, U.n_out, U.out_cap, U.err,
+                                        U.nlv, U.nrv, U.jt, U.upd, key,
+                                        ovals, ots, nullptr, 0, 0);
+                        }
+                } else if (null_own) {
+                    if (U.side == 0)
+                        ej_emit(U.out, U.n_out, U.out_cap, U.err, U.nlv,
+                                U.nrv, U.jt, U.upd, key, mine, sts, nullptr,
+                                0, 1);
+                    else
+                        ej_emit(U.out, U.n_out, U.out_cap, U.err, U.nlv,
+                                U.nrv, U.jt, U.upd, key, nullptr, 0, mine,
+                                sts, 1);
+                }
+                if (mprev < 0)
+                    U.own.head[slot] = U.own.next[match];
+                else
+                    U.own.next[mprev] = U.own.next[match];
+                cnt_own--;
+            }
+        }
+        U.own.cnt[slot] = cnt_own;
+    }
+}
+
 }  // namespace ejoin
 
 using namespace ejoin;
@@ -240,6 +467,13 @@ static char g_ej_err[256];
 struct GpuExpJoin {
     AmdExpJoinConfig cfg;
     SideMap side[2];
+    /* updating/outer-mode scratch: per-batch (slot<<32|row) sort + key
+     * segments (see k_ej_upd) */
+    uint64_t *d_sortin = nullptr, *d_sortout = nullptr;
+    uint32_t *d_segs = nullptr;
+    unsigned int *d_nseg = nullptr;
+    void *d_ejtmp = nullptr;
+    size_t ejtmp_bytes = 0;
     int64_t *d_out[16];
     unsigned long long *d_n_out;
     int *d_err;
@@ -273,9 +507,11 @@ static int ej_alloc_side(GpuExpJoin *o, SideMap *S, uint32_t C, int64_t R,
     EJHIP(o, hipMalloc((void **)&S->vals,
                        (size_t)(nv ? nv : 1) * (size_t)R * 8));
     EJHIP(o, hipMalloc((void **)&S->ts, (size_t)R * 8));
+    EJHIP(o, hipMalloc((void **)&S->cnt, ((size_t)C + 1) * 4));
     EJHIP(o, hipMalloc((void **)&S->cursor, 8));
     EJHIP(o, hipMemset(S->keys, 0xFF, (size_t)C * 8));
     EJHIP(o, hipMemset(S->head, 0xFF, ((size_t)C + 1) * 4));
+    EJHIP(o, hipMemset(S->cnt, 0, ((size_t)C + 1) * 4));
     EJHIP(o, hipMemset(S->cursor, 0, 8));
     return 0;
 }
@@ -286,6 +522,7 @@ static void ej_free_side(SideMap *S) {
     hipFree(S->next);
     hipFree(S->vals);
     hipFree(S->ts);
+    hipFree(S->cnt);
     hipFree(S->cursor);
 }
 
@@ -296,20 +533,17 @@ API void *arroyo_amd_expjoin_create(const AmdExpJoinConfig *cfg) {
         snprintf(g_ej_err, sizeof g_ej_err, "invalid expjoin config");
         return nullptr;
     }
-    if (cfg->join_type != 0 || cfg->updating != 0) {
-        /* semantics pinned by the CPU oracle (oracle/arroyo_oracle.c
-         * expjoin_insert) and the updating_*_join goldens; the HIP path
-         * for updating/outer TTL joins is round-2 work -- fail loudly
-         * rather than silently joining inner/append-only */
-        snprintf(g_ej_err, sizeof g_ej_err,
-                 "updating/outer TTL join not yet on the GPU path "
-                 "(join_type=%d updating=%d)",
-                 cfg->join_type, cfg->updating);
+    if (cfg->join_type < 0 || cfg->join_type > AMD_JOIN_FULL) {
+        snprintf(g_ej_err, sizeof g_ej_err, "invalid join_type");
         return nullptr;
     }
     GpuExpJoin *o = new GpuExpJoin();
     o->cfg = *cfg;
-    o->out_cols = 1 + cfg->n_left_vals + cfg->n_right_vals + 1;
+    /* non-inner: + [left_present, right_present]; non-inner or updating:
+     * + trailing is_retract (matches oracle_expjoin_create) */
+    o->out_cols = 1 + cfg->n_left_vals + cfg->n_right_vals + 1 +
+                  (cfg->join_type != AMD_JOIN_INNER ? 2 : 0) +
+                  (cfg->join_type != AMD_JOIN_INNER || cfg->updating ? 1 : 0);
     o->out_cap = 1ll << (cfg->log2_out_cap ? cfg->log2_out_cap : 20);
     if (hipSetDevice(cfg->device) != hipSuccess) {
         snprintf(g_ej_err, sizeof g_ej_err,
@@ -347,7 +581,8 @@ API void *arroyo_amd_expjoin_create(const AmdExpJoinConfig *cfg) {
     o->stg_cap = 1 << 20;
     int max_in = 1 + (cfg->n_left_vals > cfg->n_right_vals
                           ? cfg->n_left_vals
-                          : cfg->n_right_vals) + 1;
+                          : cfg->n_right_vals) +
+                 (cfg->updating ? 1 : 0) + 1;
     for (int c = 0; c < max_in; c++) {
         if (hipHostMalloc((void **)&o->stg_h[c], (size_t)o->stg_cap * 8) !=
                 hipSuccess ||
@@ -377,6 +612,7 @@ static int ej_check_err(GpuExpJoin *o) {
         : e == EJERR_POOL_FULL
             ? "stored-row pool full; raise log2_rows_cap or expire()"
         : e == EJERR_OUT_CAP ? "output buffer full; raise log2_out_cap"
+        : e == EJERR_RETRACT ? "retract of unknown row"
                              : "device error";
     snprintf(o->err_msg, sizeof o->err_msg, "%s", msg);
     return 1;
@@ -387,9 +623,65 @@ static int ej_grid(int64_t want_threads) {
     return (int)(want > 4096 ? 4096 : (want < 1 ? 1 : want));
 }
 
+/* updating/outer ingest: group the batch by key, then replay each key's
+ * rows in order with one thread (k_ej_upd) */
+static int ej_ingest_upd(GpuExpJoin *o, int32_t side,
+                         const int64_t *const *dcols, int64_t n_rows) {
+    int32_t nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    if (!o->d_sortin) {
+        EJHIP(o, hipMalloc((void **)&o->d_sortin, (size_t)o->stg_cap * 8));
+        EJHIP(o, hipMalloc((void **)&o->d_sortout, (size_t)o->stg_cap * 8));
+        EJHIP(o, hipMalloc((void **)&o->d_segs, (size_t)o->stg_cap * 4));
+        EJHIP(o, hipMalloc((void **)&o->d_nseg, 4));
+        o->ejtmp_bytes = 0;
+        hipcub::DeviceRadixSort::SortKeys(nullptr, o->ejtmp_bytes,
+                                          o->d_sortin, o->d_sortout,
+                                          (int)o->stg_cap);
+        EJHIP(o, hipMalloc(&o->d_ejtmp, o->ejtmp_bytes ? o->ejtmp_bytes : 1));
+    }
+    hipLaunchKernelGGL(k_ej_slots, dim3(ej_grid(n_rows)), dim3(256), 0,
+                       o->stream, dcols[0], n_rows, o->side[side],
+                       o->d_sortin, o->d_err);
+    EJHIP(o, hipGetLastError());
+    size_t tmp = o->ejtmp_bytes;
+    hipcub::DeviceRadixSort::SortKeys(o->d_ejtmp, tmp, o->d_sortin,
+                                      o->d_sortout, (int)n_rows, 0, 64,
+                                      o->stream);
+    EJHIP(o, hipMemsetAsync(o->d_nseg, 0, 4, o->stream));
+    hipLaunchKernelGGL(k_ej_segs, dim3(ej_grid(n_rows)), dim3(256), 0,
+                       o->stream, o->d_sortout, n_rows, o->d_segs,
+                       o->d_nseg);
+    EJHIP(o, hipGetLastError());
+    EjUpdArgs U = {};
+    for (int c = 0; c < 1 + nv + (o->cfg.updating ? 1 : 0) + 1; c++)
+        U.cols[c] = dcols[c];
+    U.nv = nv;
+    U.upd = o->cfg.updating ? 1 : 0;
+    U.jt = o->cfg.join_type;
+    U.side = side;
+    U.nlv = o->cfg.n_left_vals;
+    U.nrv = o->cfg.n_right_vals;
+    U.n_rows = n_rows;
+    U.sorted = o->d_sortout;
+    U.segs = o->d_segs;
+    U.n_segs = o->d_nseg;
+    U.own = o->side[side];
+    U.other = o->side[1 - side];
+    for (int i = 0; i < o->out_cols; i++) U.out[i] = o->d_out[i];
+    U.n_out = o->d_n_out;
+    U.out_cap = o->out_cap;
+    U.err = o->d_err;
+    hipLaunchKernelGGL(k_ej_upd, dim3(ej_grid(n_rows)), dim3(256), 0,
+                       o->stream, U);
+    EJHIP(o, hipGetLastError());
+    return 0;
+}
+
 static int ej_ingest(GpuExpJoin *o, int32_t side, const int64_t *const *dcols,
                      int64_t n_rows, int emit) {
     int32_t nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
+    if (emit && (o->cfg.join_type != AMD_JOIN_INNER || o->cfg.updating))
+        return ej_ingest_upd(o, side, dcols, n_rows);
     if (emit) {
         ProbeArgs P = {};
         for (int c = 0; c < 1 + nv + 1; c++) P.cols[c] = dcols[c];
@@ -464,9 +756,10 @@ API int arroyo_amd_expjoin_process_batch(void *h, int32_t side,
                                          AmdOutBatch *out) {
     GpuExpJoin *o = (GpuExpJoin *)h;
     int32_t nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
-    if (n_cols != 1 + nv + 1) {
+    int want = 1 + nv + (o->cfg.updating ? 1 : 0) + 1;
+    if (n_cols != want) {
         snprintf(o->err_msg, sizeof o->err_msg, "side %d expects %d cols",
-                 side, 1 + nv + 1);
+                 side, want);
         return 1;
     }
     EJHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
@@ -503,9 +796,10 @@ API int arroyo_amd_expjoin_process_batch_device(void *h, int32_t side,
                                                 int64_t n_rows) {
     GpuExpJoin *o = (GpuExpJoin *)h;
     int32_t nv = side == 0 ? o->cfg.n_left_vals : o->cfg.n_right_vals;
-    if (n_cols != 1 + nv + 1) {
+    int want = 1 + nv + (o->cfg.updating ? 1 : 0) + 1;
+    if (n_cols != want) {
         snprintf(o->err_msg, sizeof o->err_msg, "side %d expects %d cols",
-                 side, 1 + nv + 1);
+                 side, want);
         return 1;
     }
     return ej_ingest(o, side, dcols, n_rows, 1);
@@ -558,8 +852,11 @@ API int arroyo_amd_expjoin_expire(void *h) {
         EJHIP(o, hipMalloc((void **)&fresh.vals,
                            (size_t)(S->nv ? S->nv : 1) * (size_t)S->R * 8));
         EJHIP(o, hipMalloc((void **)&fresh.ts, (size_t)S->R * 8));
+        EJHIP(o, hipMalloc((void **)&fresh.cnt, ((size_t)S->C + 1) * 4));
         EJHIP(o, hipMalloc((void **)&fresh.cursor, 8));
         EJHIP(o, hipMemsetAsync(fresh.head, 0xFF, ((size_t)S->C + 1) * 4,
+                                o->stream));
+        EJHIP(o, hipMemsetAsync(fresh.cnt, 0, ((size_t)S->C + 1) * 4,
                                 o->stream));
         EJHIP(o, hipMemsetAsync(fresh.cursor, 0, 8, o->stream));
         CompactArgs A = {};
@@ -575,11 +872,13 @@ API int arroyo_amd_expjoin_expire(void *h) {
         hipFree(S->next);
         hipFree(S->vals);
         hipFree(S->ts);
+        hipFree(S->cnt);
         hipFree(S->cursor);
         S->head = fresh.head;
         S->next = fresh.next;
         S->vals = fresh.vals;
         S->ts = fresh.ts;
+        S->cnt = fresh.cnt;
         S->cursor = fresh.cursor;
     }
     return ej_check_err(o);
@@ -654,6 +953,11 @@ API void arroyo_amd_expjoin_destroy(void *h) {
     if (!o) return;
     hipStreamSynchronize(o->stream);
     for (int s = 0; s < 2; s++) ej_free_side(&o->side[s]);
+    hipFree(o->d_sortin);
+    hipFree(o->d_sortout);
+    hipFree(o->d_segs);
+    hipFree(o->d_nseg);
+    hipFree(o->d_ejtmp);
     for (int i = 0; i < o->out_cols; i++) hipFree(o->d_out[i]);
     hipFree(o->d_n_out);
     hipFree(o->d_err);

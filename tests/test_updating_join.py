@@ -5,9 +5,10 @@ Reference semantics: the planner passes the SQL join type into the
 DataFusion join inside JoinWithExpiration (plan/join.rs:326-379); with an
 updating input the join is maintained incrementally — a new match retracts
 the earlier null-padded row, a retraction that removes a key's last match
-brings the null-padded rows back.  CPU oracle only in round 1
-(oracle/arroyo_oracle.c expjoin_insert); the GPU library rejects these
-configs loudly (tests below).
+brings the null-padded rows back.  Round 2: the HIP path (k_ej_upd,
+arroyo_amd/csrc/expjoin.hip — per-key ordered replay after a
+radix-grouped batch sort) runs the same tests under -m gpu via the
+`ej_factory` fixture.
 """
 import numpy as np
 import pytest
@@ -15,6 +16,17 @@ import pytest
 import oracle
 from arroyo_amd import cabi
 from tests.golden_util import NS, load_golden, load_inputs
+
+
+@pytest.fixture(params=["oracle",
+                        pytest.param("gpu", marks=pytest.mark.gpu)])
+def ej_factory(request):
+    """Runs each test against the CPU oracle and (under -m gpu) the HIP
+    library through the same C ABI."""
+    if request.param == "oracle":
+        return oracle.make_expjoin_op
+    from arroyo_amd import gpu
+    return gpu.make_expjoin_op
 
 HOUR = 3600 * NS
 T0 = 1_600_000_000 * NS
@@ -41,10 +53,10 @@ def fold(emissions):
     return live
 
 
-def test_updating_left_join_scenario():
+def test_updating_left_join_scenario(ej_factory):
     """Hand-checked LEFT sequence: null row -> match retracts it -> right
     retraction brings it back."""
-    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+    op = ej_factory(cabi.make_expjoin_config(
         24 * HOUR, n_left_vals=0, n_right_vals=1,
         join_type=cabi.JOIN_LEFT, updating=True))
     a = np.array
@@ -65,9 +77,9 @@ def test_updating_left_join_scenario():
     op.close()
 
 
-def test_updating_full_join_scenario():
+def test_updating_full_join_scenario(ej_factory):
     """FULL: both sides pad; a match retracts both null rows."""
-    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+    op = ej_factory(cabi.make_expjoin_config(
         24 * HOUR, n_left_vals=1, n_right_vals=1,
         join_type=cabi.JOIN_FULL, updating=True))
     a = np.array
@@ -86,10 +98,10 @@ def test_updating_full_join_scenario():
     op.close()
 
 
-def test_updating_inner_retraction_propagates():
+def test_updating_inner_retraction_propagates(ej_factory):
     """INNER with updating inputs: retracting a stored right row retracts
     the pairs it participated in (no null rows)."""
-    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+    op = ej_factory(cabi.make_expjoin_config(
         24 * HOUR, n_left_vals=0, n_right_vals=1, updating=True))
     a = np.array
     assert rows_of(op.process_batch(op.LEFT,
@@ -103,7 +115,7 @@ def test_updating_inner_retraction_propagates():
     op.close()
 
 
-def run_updating_join_golden(join_type, golden):
+def run_updating_join_golden(ej_factory, join_type, golden):
     """updating_{left,right,full}_join.sql: impulse JOIN (counter % 2,
     count(*) WHERE counter < 3 GROUP BY 1) ON counter = right_count WHERE
     counter < 3, debezium sink.  The right side is an updating aggregate
@@ -116,7 +128,7 @@ def run_updating_join_golden(join_type, golden):
 
     agg = oracle.make_updagg_op(cabi.make_updagg_config(
         [(cabi.COUNT, -1)], n_keys=1, n_value_cols=0))
-    join = oracle.make_expjoin_op(cabi.make_expjoin_config(
+    join = ej_factory(cabi.make_expjoin_config(
         24 * HOUR, n_left_vals=0, n_right_vals=1, join_type=join_type,
         updating=True))
     ems = []
@@ -162,26 +174,59 @@ def run_updating_join_golden(join_type, golden):
     assert got == want
 
 
-def test_updating_left_join_golden():
-    run_updating_join_golden(cabi.JOIN_LEFT, "updating_left_join")
+def test_updating_left_join_golden(ej_factory):
+    run_updating_join_golden(ej_factory, cabi.JOIN_LEFT,
+                             "updating_left_join")
 
 
-def test_updating_right_join_golden():
-    run_updating_join_golden(cabi.JOIN_RIGHT, "updating_right_join")
+def test_updating_right_join_golden(ej_factory):
+    run_updating_join_golden(ej_factory, cabi.JOIN_RIGHT,
+                             "updating_right_join")
 
 
-def test_updating_full_join_golden():
-    run_updating_join_golden(cabi.JOIN_FULL, "updating_full_join")
+def test_updating_full_join_golden(ej_factory):
+    run_updating_join_golden(ej_factory, cabi.JOIN_FULL,
+                             "updating_full_join")
 
 
 @pytest.mark.gpu
-def test_gpu_rejects_updating_outer_expjoin():
-    """The HIP path must refuse these configs loudly (no silent
-    inner/append-only fallback) until its round-2 implementation."""
+@pytest.mark.parametrize("join_type", [0, 1, 2, 3],
+                         ids=["inner", "left", "right", "full"])
+def test_gpu_matches_oracle_updating_batches(join_type):
+    """Multi-row mixed append/retract batches: the HIP path's emission
+    multiset per batch equals the oracle's (within-batch emission order
+    is unspecified across keys)."""
     from arroyo_amd import gpu
-    for kw in (dict(join_type=cabi.JOIN_LEFT), dict(updating=True)):
-        with pytest.raises(RuntimeError):
-            gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **kw))
+    rng = np.random.default_rng(7 + join_type)
+    cfg = dict(n_left_vals=1, n_right_vals=1, join_type=join_type,
+               updating=True)
+    o1 = oracle.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfg))
+    o2 = gpu.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfg))
+    live = [{}, {}]   # (key,val) -> count, to generate valid retracts
+    for step in range(30):
+        side = int(rng.integers(0, 2))
+        n = int(rng.integers(1, 64))
+        keys, vals, retr, tss = [], [], [], []
+        for _ in range(n):
+            stored = [kv for kv, c in live[side].items() if c > 0]
+            if stored and rng.random() < 0.35:
+                k, v = stored[int(rng.integers(0, len(stored)))]
+                live[side][(k, v)] -= 1
+                keys.append(k); vals.append(v); retr.append(1)
+            else:
+                k = int(rng.integers(0, 8)); v = int(rng.integers(0, 4))
+                live[side][(k, v)] = live[side].get((k, v), 0) + 1
+                keys.append(k); vals.append(v); retr.append(0)
+            tss.append(T0 + (k * 7 + v) * NS)
+        cols = [np.array(keys, dtype=np.int64),
+                np.array(vals, dtype=np.int64),
+                np.array(retr, dtype=np.int64),
+                np.array(tss, dtype=np.int64)]
+        e1 = sorted(rows_of(o1.process_batch(side, cols)))
+        e2 = sorted(rows_of(o2.process_batch(side, cols)))
+        assert e1 == e2, f"step {step} side {side}"
+    o1.close()
+    o2.close()
 
 
 def np_join_of_multisets(left, right, join_type, nlv, nrv):
@@ -216,12 +261,12 @@ def np_join_of_multisets(left, right, join_type, nlv, nrv):
 @pytest.mark.parametrize("seed", [103, 211, 307])
 @pytest.mark.parametrize("join_type", [1, 2, 3],
                          ids=["left", "right", "full"])
-def test_updating_join_fuzz_fold_invariant(join_type, seed):
+def test_updating_join_fuzz_fold_invariant(ej_factory, join_type, seed):
     """Random append/retract interleavings on both sides: the folded
     emission stream must equal the brute-force join of the two final live
     multisets, for every prefix cadence."""
     rng = np.random.default_rng(seed + join_type)
-    op = oracle.make_expjoin_op(cabi.make_expjoin_config(
+    op = ej_factory(cabi.make_expjoin_config(
         24 * HOUR, n_left_vals=1, n_right_vals=1, join_type=join_type,
         updating=True))
     live = [[], []]  # per-side live rows (key, (val,), ts)
@@ -256,14 +301,14 @@ def test_updating_join_fuzz_fold_invariant(join_type, seed):
     op.close()
 
 
-def test_updating_join_checkpoint_roundtrip():
+def test_updating_join_checkpoint_roundtrip(ej_factory):
     """Drain/restore carries the net live multiset: a restored LEFT join
     continues emitting correct retract/append pairs for rows stored before
     the checkpoint."""
     a = np.array
     cfgk = dict(n_left_vals=0, n_right_vals=1, join_type=cabi.JOIN_LEFT,
                 updating=True)
-    op1 = oracle.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    op1 = ej_factory(cabi.make_expjoin_config(24 * HOUR, **cfgk))
     # left 7 live and unmatched at checkpoint time
     ems = rows_of(op1.process_batch(op1.LEFT, [a([7]), a([0]), a([T0])]))
     assert ems == [(7, 0, T0, 1, 0, 0)]
@@ -272,7 +317,7 @@ def test_updating_join_checkpoint_roundtrip():
     op1.close()
     assert len(ld[0]) == 1 and len(rd[0]) == 0
 
-    op2 = oracle.make_expjoin_op(cabi.make_expjoin_config(24 * HOUR, **cfgk))
+    op2 = ej_factory(cabi.make_expjoin_config(24 * HOUR, **cfgk))
     op2.restore(op2.LEFT, ld)
     op2.restore(op2.RIGHT, rd)
     # first match after restore retracts the pre-checkpoint null row
