@@ -243,39 +243,43 @@ def run_gpu(args):
     tw1 = time.perf_counter()
     op.perf()  # reset kernel-time counters after warmup
 
-    # enforce a minimum timed duration regardless of --steps: estimate the
-    # per-step time from the warmup and repeat the requested step count
-    # until the timed region covers >= MIN_TIMED_S.  All ranks must agree
-    # on the multiplier (collectives per step), so take the max across
-    # ranks before the barrier.
-    est_step = (tw1 - tw0) / max(args.warmup, 1)
-    mult = max(1, int(-(-MIN_TIMED_S // max(est_step * args.steps, 1e-9))))
-    if dist:
-        m = torch.tensor([mult], device=dev)
-        dist.all_reduce(m, op=dist.ReduceOp.MAX)
-        mult = int(m.item())
-    timed_steps = args.steps * mult
-
+    # enforce a minimum timed duration regardless of --steps: run the
+    # requested step count, then keep extending (doubling) the timed
+    # region until it covers >= MIN_TIMED_S of wall time.  All ranks agree
+    # on each extension (max-elapsed all-reduce before the decision), so
+    # the per-step collectives stay in lockstep.
+    del tw0, tw1
     wm_fires[0] = 0
-    if dist:
-        dist.barrier()
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    if world == 1:
-        run_span(args.warmup, timed_steps)
-    else:
-        for s in range(args.warmup, args.warmup + timed_steps):
-            one_step(s)
-    torch.cuda.synchronize()
-    if dist:
-        dist.barrier()
-    t1 = time.perf_counter()
-    elapsed = t1 - t0
-    if dist:
-        import torch.distributed as dist_mod
-        e = torch.tensor([elapsed], device=dev)
-        dist_mod.all_reduce(e, op=dist_mod.ReduceOp.MAX)
-        elapsed = float(e.item())
+    elapsed = 0.0
+    timed_steps = 0
+    chunk = args.steps
+    s_cursor = args.warmup
+    while True:
+        if dist:
+            dist.barrier()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        if world == 1:
+            run_span(s_cursor, chunk)
+        else:
+            for s in range(s_cursor, s_cursor + chunk):
+                one_step(s)
+        torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+        t1 = time.perf_counter()
+        s_cursor += chunk
+        timed_steps += chunk
+        elapsed += t1 - t0
+        e_all = elapsed
+        if dist:
+            e = torch.tensor([elapsed], device=dev)
+            dist.all_reduce(e, op=dist.ReduceOp.MAX)
+            e_all = float(e.item())
+        if e_all >= MIN_TIMED_S:
+            elapsed = e_all
+            break
+        chunk = max(chunk, timed_steps)   # double the region each pass
 
     perf = op.perf()
     perf["wm_fires_timed"] = wm_fires[0]
@@ -475,7 +479,7 @@ def main():
             "unit": "GB/s",
             "frac": achieved_gbps / HBM_PEAK_GBPS,
             "traffic": None,
-            "kernel": "k_update_packed",
+            "kernel": "k_update_batch",
             "avg_launch_us": avg_launch_ms * 1000,
             "rows_per_launch": rows_per_launch,
         },
