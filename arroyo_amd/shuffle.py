@@ -44,6 +44,14 @@ def partition_ids(keys, n_parts):
     return (h // range_size).astype(np.int64)
 
 
+def recv_splits_of(gathered_flat, world, rank):
+    """Decode an all-gathered [world x world] send-split matrix (row =
+    sender, column = destination) into this rank's receive splits, indexed
+    by source rank.  Shared by bench.py's RCCL exchange and the
+    single-process bookkeeping test."""
+    return [int(gathered_flat[src * world + rank]) for src in range(world)]
+
+
 def exchange_sizes(send_counts, group=None):
     """All-gather the per-destination row counts; returns recv_counts[src]."""
     world = dist.get_world_size(group)

@@ -205,9 +205,15 @@ def run_gpu(args):
             gathered = torch.empty(world * world, dtype=torch.int64,
                                    device=dev)
             dist.all_gather_into_tensor(gathered, tsizes)
-            recv_splits = [int(gathered[s * world + rank])
-                           for s in range(world)]
+            from arroyo_amd.shuffle import recv_splits_of
+            recv_splits = recv_splits_of(gathered.cpu().tolist(), world,
+                                         rank)
             n_recv = sum(recv_splits)
+            if n_recv > rbuf_cap:
+                raise RuntimeError(
+                    f"rank {rank}: skewed exchange overflows the receive "
+                    f"buffer ({n_recv} rows > cap {rbuf_cap}); raise the "
+                    f"rbuf_cap headroom")
             dist.all_to_all_single(rk[:n_recv], sk,
                                    output_split_sizes=recv_splits,
                                    input_split_sizes=send_splits)

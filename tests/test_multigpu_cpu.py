@@ -14,6 +14,7 @@ gloo with the C oracle as the per-shard operator.
 import os
 
 import numpy as np
+import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -26,7 +27,7 @@ from arroyo_amd.shuffle import shuffle_columns
 WORLD = 2
 
 
-def _rank_main(rank, world, port, result_q):
+def _rank_main(rank, world, port, result_q, op_kind="oracle"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -36,7 +37,14 @@ def _rank_main(rank, world, port, result_q):
     key, ts = cols
     kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=0,
               aggs=[(cabi.COUNT, -1)])
-    op = oracle.make_op(cabi.make_config(**kw))
+    if op_kind == "gpu":
+        # every rank's shard op runs on the same physical GPU (device 0):
+        # legal — handles are independent — and exactly the composed
+        # N-rank dataflow (gloo exchange + HIP operator per shard)
+        from arroyo_amd import gpu
+        op = gpu.make_op(cabi.make_config(**kw, emit_to_host=True))
+    else:
+        op = oracle.make_op(cabi.make_config(**kw))
 
     outs = []
     from arroyo_amd.pipeline import U64MAX, WatermarkGen
@@ -73,10 +81,10 @@ def _rank_main(rank, world, port, result_q):
     dist.destroy_process_group()
 
 
-def _run_world(world, port):
+def _run_world(world, port, op_kind="oracle"):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main, args=(r, world, port, q))
+    procs = [ctx.Process(target=_rank_main, args=(r, world, port, q, op_kind))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -119,3 +127,18 @@ def test_shuffle_window_parity_world4():
     """Same sharded-vs-single parity at world_size=4 (the q7 config's 1->8
     GPU scaling path; gloo here, RCCL on the GPU boxes)."""
     _run_world(4, 29377)
+
+
+@pytest.mark.gpu
+def test_two_rank_shuffle_hip_ops_match_single():
+    """The composed N>1 path with the HIP operator per shard: two gloo
+    ranks exchange by key-hash range and each feeds its shard through the
+    GPU window op on one device; the merged emissions equal the
+    single-instance oracle run.  (An 8-GPU RCCL run swaps gloo for
+    all_to_all_single over xGMI — same partitioning, same operator.)"""
+    _run_world(2, 29383, op_kind="gpu")
+
+
+@pytest.mark.gpu
+def test_four_rank_shuffle_hip_ops_match_single():
+    _run_world(4, 29389, op_kind="gpu")

@@ -339,3 +339,24 @@ def test_windowfn_two_rank_shuffle_matches_single():
     single.close()
     assert sorted(results[0] + results[1]) == sorted(want)
     assert len(want) > 1000
+
+
+def test_exchange_split_bookkeeping_simulation():
+    """Single-process simulation of the N>1 bench's split-size math: for a
+    random [world x world] send matrix (row = sender, col = destination),
+    every rank's decoded receive splits must match the senders' columns,
+    and totals must conserve rows."""
+    import numpy as np
+
+    from arroyo_amd.shuffle import recv_splits_of
+
+    rng = np.random.default_rng(5)
+    for world in (2, 4, 8):
+        sends = rng.integers(0, 70_000, size=(world, world))
+        gathered = [int(x) for x in sends.reshape(-1)]  # all_gather order
+        total_recv = 0
+        for rank in range(world):
+            rs = recv_splits_of(gathered, world, rank)
+            assert rs == [int(sends[src][rank]) for src in range(world)]
+            total_recv += sum(rs)
+        assert total_recv == int(sends.sum())
