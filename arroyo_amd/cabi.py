@@ -39,6 +39,7 @@ class AmdOutBatch(ctypes.Structure):
         ("cols", ctypes.POINTER(ctypes.c_void_p)),
         ("is_f64", ctypes.POINTER(ctypes.c_int32)),
         ("on_device", ctypes.c_int32),
+        ("validity", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
     ]
 
 
@@ -90,6 +91,27 @@ def bind(lib, prefix):
     fn["last_error"].restype = ctypes.c_char_p
     fn["last_error"].argtypes = [ctypes.c_void_p]
     return fn
+
+
+def out_validity(out):
+    """Decode the Arrow validity bitmaps of an AmdOutBatch into per-column
+    numpy bool masks (None = column all-valid)."""
+    n = out.n_rows
+    masks = []
+    if not out.validity:
+        return [None] * out.n_cols
+    for i in range(out.n_cols):
+        bm = out.validity[i]
+        if not bm:
+            masks.append(None)
+            continue
+        nbytes = (n + 7) // 8
+        raw = np.frombuffer(
+            bytes(ctypes.cast(bm, ctypes.POINTER(
+                ctypes.c_uint8 * max(nbytes, 1))).contents),
+            dtype=np.uint8)
+        masks.append(np.unpackbits(raw, bitorder="little")[:n].astype(bool))
+    return masks
 
 
 def _out_to_numpy(out):

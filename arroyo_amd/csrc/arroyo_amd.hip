@@ -3173,6 +3173,10 @@ API void arroyo_amd_free_out(AmdOutBatch *out) {
         else
             free(out->cols[i]);
     }
+    if (out->validity) {
+        for (int i = 0; i < out->n_cols; i++) free(out->validity[i]);
+        free(out->validity);
+    }
     free(out->cols);
     free(out->is_f64);
     memset(out, 0, sizeof *out);
@@ -3942,6 +3946,31 @@ static int join_fire(GpuJoin *o, uint64_t instant,
     return 0;
 }
 
+
+/* Arrow validity bitmaps for null-padded value columns (LSB bit order,
+ * see AmdOutBatch doc): left vals valid where the left-presence column is
+ * set, right vals likewise.  The presence columns stay in the output. */
+static void fill_join_validity(AmdOutBatch *out, int base_l, int nlv,
+                               int nrv, int lp_col, int rp_col) {
+    int64_t n = out->n_rows;
+    out->validity = (uint8_t **)calloc(out->n_cols, sizeof(uint8_t *));
+    if (!out->validity || n == 0) return;
+    size_t nbytes = (size_t)((n + 7) / 8);
+    for (int g = 0; g < 2; g++) {
+        int base = g == 0 ? base_l : base_l + nlv;
+        int cnt = g == 0 ? nlv : nrv;
+        const int64_t *pres =
+            (const int64_t *)out->cols[g == 0 ? lp_col : rp_col];
+        for (int c = base; c < base + cnt; c++) {
+            uint8_t *bm = (uint8_t *)calloc(nbytes, 1);
+            if (!bm) continue;
+            for (int64_t r = 0; r < n; r++)
+                if (pres[r]) bm[r >> 3] |= (uint8_t)(1u << (r & 7));
+            out->validity[c] = bm;
+        }
+    }
+}
+
 API int arroyo_amd_join_handle_watermark(void *h, uint64_t wm,
                                          AmdOutBatch *out) {
     GpuJoin *o = (GpuJoin *)h;
@@ -3976,6 +4005,13 @@ API int arroyo_amd_join_handle_watermark(void *h, uint64_t wm,
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
             if (n) memcpy(out->cols[i], o->host_out[i].data(), (size_t)n * 8);
+        }
+        if (o->cfg.join_type != AMD_JOIN_INNER) {
+            /* [key?, lvals, rvals, instant, lp, rp] */
+            int base_l = o->cfg.n_keys;
+            int inst = base_l + o->cfg.n_left_vals + o->cfg.n_right_vals;
+            fill_join_validity(out, base_l, o->cfg.n_left_vals,
+                               o->cfg.n_right_vals, inst + 1, inst + 2);
         }
         for (auto &v : o->host_out) v.clear();
     }

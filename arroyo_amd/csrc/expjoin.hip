@@ -601,6 +601,31 @@ API const char *arroyo_amd_expjoin_last_error(void *h) {
     return h ? ((GpuExpJoin *)h)->err_msg : g_ej_err;
 }
 
+/* Arrow validity bitmaps for the updating/outer output's null-padded
+ * value columns ([key, lvals, rvals, ts, lp, rp, retract]) */
+static void ej_fill_validity(GpuExpJoin *o, AmdOutBatch *out) {
+    if (o->cfg.join_type == AMD_JOIN_INNER) return;
+    int nlv = o->cfg.n_left_vals, nrv = o->cfg.n_right_vals;
+    int lp = 1 + nlv + nrv + 1;
+    int64_t n = out->n_rows;
+    out->validity = (uint8_t **)calloc(out->n_cols, sizeof(uint8_t *));
+    if (!out->validity || n == 0) return;
+    size_t nbytes = (size_t)((n + 7) / 8);
+    for (int g = 0; g < 2; g++) {
+        int base = g == 0 ? 1 : 1 + nlv;
+        int cnt = g == 0 ? nlv : nrv;
+        const int64_t *pres = (const int64_t *)out->cols[g == 0 ? lp
+                                                                : lp + 1];
+        for (int c = base; c < base + cnt; c++) {
+            uint8_t *bm = (uint8_t *)calloc(nbytes, 1);
+            if (!bm) continue;
+            for (int64_t r = 0; r < n; r++)
+                if (pres[r]) bm[r >> 3] |= (uint8_t)(1u << (r & 7));
+            out->validity[c] = bm;
+        }
+    }
+}
+
 static int ej_check_err(GpuExpJoin *o) {
     int e = 0;
     EJHIP(o, hipMemcpyAsync(&e, o->d_err, 4, hipMemcpyDeviceToHost,
@@ -783,6 +808,7 @@ API int arroyo_amd_expjoin_process_batch(void *h, int32_t side,
                                         o->stream));
         }
         EJHIP(o, hipStreamSynchronize(o->stream));
+        ej_fill_validity(o, out);
     }
     return 0;
 }
@@ -825,6 +851,7 @@ API int arroyo_amd_expjoin_collect(void *h, AmdOutBatch *out) {
                                     hipMemcpyDeviceToHost, o->stream));
     }
     EJHIP(o, hipStreamSynchronize(o->stream));
+    ej_fill_validity(o, out);
     EJHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
     return 0;
 }

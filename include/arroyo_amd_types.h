@@ -268,13 +268,22 @@ typedef struct {
  * Column order: [key (if n_keys)], agg outputs (one column per agg),
  * window_start, window_end, _timestamp.  All columns are 8-byte elements;
  * is_f64[i] marks double columns (AVG outputs).  on_device=1 means `cols`
- * are HIP device pointers. */
+ * are HIP device pointers.
+ *
+ * validity: Arrow-layout per-column validity bitmaps (LSB order: bit
+ * (r & 7) of byte (r >> 3) set = row r valid), or NULL when every column
+ * is fully valid.  validity[c] == NULL marks column c all-valid.  The
+ * outer-join operators set bitmaps on the null-padded value columns (the
+ * Arrow RecordBatch contract of ArroyoSchema, crates/arroyo-rpc/src/
+ * df.rs:24-30); the presence columns remain alongside for callers that
+ * prefer flags. */
 typedef struct {
     int64_t   n_rows;
     int32_t   n_cols;
     void    **cols;
     int32_t  *is_f64;
     int32_t   on_device;
+    uint8_t **validity;
 } AmdOutBatch;
 
 #ifdef __cplusplus

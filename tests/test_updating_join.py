@@ -326,3 +326,58 @@ def test_updating_join_checkpoint_roundtrip(ej_factory):
     op2.close()
     assert sorted(out) == sorted([(7, 0, T0, 1, 0, 1),
                                   (7, 42, T0 + NS, 1, 1, 0)])
+
+
+@pytest.mark.gpu
+def test_gpu_outer_join_validity_bitmaps():
+    """The C ABI carries Arrow validity bitmaps (LSB bit order) for the
+    null-padded value columns; the bitmap must equal the presence flags,
+    which are themselves oracle-pinned (df.rs:24-30 RecordBatch
+    contract)."""
+    import ctypes
+
+    from arroyo_amd import gpu
+    from arroyo_amd.cabi import AmdOutBatch, _out_to_numpy, out_validity
+
+    a = np.array
+    cfg = cabi.make_expjoin_config(24 * HOUR, n_left_vals=1, n_right_vals=1,
+                                   join_type=cabi.JOIN_FULL, updating=True)
+    op = gpu.make_expjoin_op(cfg)
+    emissions = []
+    batches = [
+        (op.LEFT, [a([3]), a([10]), a([0]), a([T0])]),
+        (op.RIGHT, [a([4]), a([20]), a([0]), a([T0])]),
+        (op.RIGHT, [a([3]), a([30]), a([0]), a([T0 + NS])]),
+        (op.RIGHT, [a([3]), a([30]), a([1]), a([T0 + NS])]),
+    ]
+    for side, cols in batches:
+        cols = [np.ascontiguousarray(c, dtype=np.int64) for c in cols]
+        n = len(cols[0])
+        arr = (ctypes.c_void_p * len(cols))(
+            *[c.ctypes.data_as(ctypes.c_void_p).value for c in cols])
+        out = AmdOutBatch()
+        rc = op._fn["process_batch"](op._h, side, arr, len(cols), n,
+                                     ctypes.byref(out))
+        assert rc == 0, op._fn["last_error"](op._h).decode()
+        got = _out_to_numpy(out)
+        masks = out_validity(out)
+        op._fn["free_out"](ctypes.byref(out))
+        if not len(got[0]):
+            continue
+        # layout: [key, lval, rval, ts, lp, rp, retract]
+        lp, rp = got[4].astype(bool), got[5].astype(bool)
+        assert masks[1] is not None and masks[2] is not None
+        assert (masks[1] == lp).all(), "left-value validity != presence"
+        assert (masks[2] == rp).all(), "right-value validity != presence"
+        for c in (0, 3, 4, 5, 6):
+            assert masks[c] is None  # key/ts/flags always valid
+        emissions.extend(
+            tuple(int(c[r]) for c in got) for r in range(len(got[0])))
+    op.close()
+    # sanity: the emission fold matches the oracle for the same sequence
+    o2 = oracle.make_expjoin_op(cfg)
+    ems2 = []
+    for side, cols in batches:
+        ems2 += rows_of(o2.process_batch(side, cols))
+    o2.close()
+    assert fold(emissions) == fold(ems2)
