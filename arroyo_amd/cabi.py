@@ -29,6 +29,7 @@ class AmdWindowConfig(ctypes.Structure):
         ("ring_panes", ctypes.c_uint32),
         ("device", ctypes.c_int32),
         ("emit_to_host", ctypes.c_int32),
+        ("val_is_f64", ctypes.c_int32 * 8),
     ]
 
 
@@ -45,8 +46,9 @@ class AmdOutBatch(ctypes.Structure):
 
 def make_config(width_ns, slide_ns, aggs, n_keys=1, n_value_cols=0,
                 is_tumbling=False, log2_capacity=20, ring_panes=64,
-                device=0, emit_to_host=True):
-    """aggs: list of (op, value_col_index); value_col_index -1 for COUNT(*)."""
+                device=0, emit_to_host=True, val_is_f64=()):
+    """aggs: list of (op, value_col_index); value_col_index -1 for COUNT(*).
+    val_is_f64: indices of value columns holding f64 bit patterns."""
     cfg = AmdWindowConfig()
     cfg.width_nanos = width_ns
     cfg.slide_nanos = width_ns if is_tumbling else slide_ns
@@ -57,6 +59,8 @@ def make_config(width_ns, slide_ns, aggs, n_keys=1, n_value_cols=0,
         cfg.agg_ops[i] = op
         cfg.agg_col[i] = col
     cfg.n_value_cols = n_value_cols
+    for v in val_is_f64:
+        cfg.val_is_f64[v] = 1
     cfg.log2_capacity = log2_capacity
     cfg.ring_panes = ring_panes
     cfg.device = device

@@ -89,6 +89,8 @@ struct AggSpec {
     int32_t n_aggs;
     int32_t op[AMD_MAX_AGGS];
     int32_t col[AMD_MAX_AGGS];
+    int32_t isf[AMD_MAX_AGGS];  /* value column is f64 (bit pattern in the
+                                   i64 plane); COUNT/AVG-count unaffected */
 };
 
 __host__ __device__ inline uint64_t enc_min(int64_t v) {
@@ -102,6 +104,48 @@ __host__ __device__ inline uint64_t enc_max(int64_t v) {
 }
 __host__ __device__ inline int64_t dec_max(uint64_t e) {
     return (int64_t)(e ^ 0x8000000000000000ULL);
+}
+
+/* order-preserving f64 <-> u64 with the all-zero bit pattern as identity:
+ * flip maps ordered (non-NaN) doubles onto (0x000F.., 0xFFF0..] of u64, so
+ * a memset-zero state is below every real encoding (MAX) and, negated,
+ * above every real one (MIN) — same memset-retirable property as the i64
+ * encodes. */
+__host__ __device__ inline uint64_t f64flip(uint64_t b) {
+    return (b & 0x8000000000000000ULL) ? ~b : (b | 0x8000000000000000ULL);
+}
+__host__ __device__ inline uint64_t f64unflip(uint64_t e) {
+    return (e & 0x8000000000000000ULL) ? (e & ~0x8000000000000000ULL) : ~e;
+}
+__host__ __device__ inline uint64_t enc_minf(int64_t bits) {
+    return ~f64flip((uint64_t)bits);
+}
+__host__ __device__ inline int64_t dec_minf(uint64_t e) {
+    return (int64_t)f64unflip(~e);
+}
+__host__ __device__ inline uint64_t enc_maxf(int64_t bits) {
+    return f64flip((uint64_t)bits);
+}
+__host__ __device__ inline int64_t dec_maxf(uint64_t e) {
+    return (int64_t)f64unflip(e);
+}
+
+/* encode one raw (i64 or f64-bits) value into the op's state domain */
+__host__ __device__ inline uint64_t enc_word(int op, int isf, int64_t raw) {
+    switch (op) {
+    case AMD_AGG_MIN: return isf ? enc_minf(raw) : enc_min(raw);
+    case AMD_AGG_MAX: return isf ? enc_maxf(raw) : enc_max(raw);
+    default: return (uint64_t)raw;    /* COUNT/SUM/AVG words are raw */
+    }
+}
+
+/* decode one state word back to the raw value domain */
+__host__ __device__ inline int64_t dec_word(int op, int isf, uint64_t e) {
+    switch (op) {
+    case AMD_AGG_MIN: return isf ? dec_minf(e) : dec_min(e);
+    case AMD_AGG_MAX: return isf ? dec_maxf(e) : dec_max(e);
+    default: return (int64_t)e;
+    }
 }
 
 /* floor-divide t by slide via the precomputed reciprocal; exact for all
@@ -130,17 +174,32 @@ __device__ inline void atomic_update(uint64_t *st, const AggSpec a,
             atomicAdd((unsigned long long *)w, 1ULL);
             break;
         case AMD_AGG_SUM:
-            atomicAdd((unsigned long long *)w, (unsigned long long)vcols[a.col[i]][r]);
+            if (a.isf[i]) {
+                int64_t b = vcols[a.col[i]][r];
+                double v;
+                memcpy(&v, &b, 8);
+                atomicAdd((double *)w, v);
+            } else {
+                atomicAdd((unsigned long long *)w,
+                          (unsigned long long)vcols[a.col[i]][r]);
+            }
             break;
         case AMD_AGG_MIN:
-            atomicMax((unsigned long long *)w, (unsigned long long)enc_min(vcols[a.col[i]][r]));
-            break;
         case AMD_AGG_MAX:
-            atomicMax((unsigned long long *)w, (unsigned long long)enc_max(vcols[a.col[i]][r]));
+            atomicMax((unsigned long long *)w,
+                      (unsigned long long)enc_word(a.op[i], a.isf[i],
+                                                   vcols[a.col[i]][r]));
             break;
         case AMD_AGG_AVG:
             atomicAdd((unsigned long long *)w, 1ULL);
-            atomicAdd((double *)(w + 1), (double)vcols[a.col[i]][r]);
+            if (a.isf[i]) {
+                int64_t b = vcols[a.col[i]][r];
+                double v;
+                memcpy(&v, &b, 8);
+                atomicAdd((double *)(w + 1), v);
+            } else {
+                atomicAdd((double *)(w + 1), (double)vcols[a.col[i]][r]);
+            }
             break;
         }
     }
@@ -152,9 +211,16 @@ __device__ inline void atomic_merge(uint64_t *dst, const uint64_t *src,
     for (int i = 0; i < a.n_aggs; i++) {
         switch (a.op[i]) {
         case AMD_AGG_COUNT:
-        case AMD_AGG_SUM:
             atomicAdd((unsigned long long *)(dst + 2 * i),
                       (unsigned long long)src[2 * i]);
+            break;
+        case AMD_AGG_SUM:
+            if (a.isf[i])
+                atomicAdd((double *)(dst + 2 * i),
+                          *(const double *)(src + 2 * i));
+            else
+                atomicAdd((unsigned long long *)(dst + 2 * i),
+                          (unsigned long long)src[2 * i]);
             break;
         case AMD_AGG_MIN:
         case AMD_AGG_MAX:
@@ -434,30 +500,7 @@ __device__ inline void lds_update_row(const UpdateArgs &A, int64_t *ls_key,
                     done = true;
                     break;
                 }
-                for (int a = 0; a < na; a++) {
-                    uint64_t *w = st + 2 * a;
-                    switch (A.agg.op[a]) {
-                    case AMD_AGG_COUNT:
-                        atomicAdd((unsigned long long *)w, 1ULL); break;
-                    case AMD_AGG_SUM:
-                        atomicAdd((unsigned long long *)w,
-                                  (unsigned long long)vc[A.agg.col[a]][row]);
-                        break;
-                    case AMD_AGG_MIN:
-                        atomicMax((unsigned long long *)w,
-                                  (unsigned long long)enc_min(vc[A.agg.col[a]][row]));
-                        break;
-                    case AMD_AGG_MAX:
-                        atomicMax((unsigned long long *)w,
-                                  (unsigned long long)enc_max(vc[A.agg.col[a]][row]));
-                        break;
-                    case AMD_AGG_AVG:
-                        atomicAdd((unsigned long long *)w, 1ULL);
-                        atomicAdd((double *)(w + 1),
-                                  (double)vc[A.agg.col[a]][row]);
-                        break;
-                    }
-                }
+                atomic_update(st, A.agg, vc, row);
                 done = true;
             }
         }
@@ -1621,13 +1664,11 @@ __device__ inline void emit_row(const CompactArgs &C, int64_t key,
         switch (C.agg.op[a]) {
         case AMD_AGG_COUNT:
         case AMD_AGG_SUM:
-            C.out[col++][r] = (int64_t)w0;
+            C.out[col++][r] = (int64_t)w0;   /* SUM f64: raw double bits */
             break;
         case AMD_AGG_MIN:
-            C.out[col++][r] = dec_min(w0);
-            break;
         case AMD_AGG_MAX:
-            C.out[col++][r] = dec_max(w0);
+            C.out[col++][r] = dec_word(C.agg.op[a], C.agg.isf[a], w0);
             break;
         case AMD_AGG_AVG:
             if (C.raw_states) {
@@ -1777,30 +1818,7 @@ k_merge_fused(MergeFusedArgs M) {
                           (unsigned long long)pw0);
                 return;
             }
-            for (int ag = 0; ag < na; ag++) {
-                switch (M.agg.op[ag]) {
-                case AMD_AGG_COUNT:
-                case AMD_AGG_SUM:
-                    atomicAdd((unsigned long long *)&d[2 * ag],
-                              (unsigned long long)st[(size_t)idx * na * 2 +
-                                                     2 * ag]);
-                    break;
-                case AMD_AGG_MIN:
-                case AMD_AGG_MAX:
-                    atomicMax((unsigned long long *)&d[2 * ag],
-                              (unsigned long long)st[(size_t)idx * na * 2 +
-                                                     2 * ag]);
-                    break;
-                case AMD_AGG_AVG:
-                    atomicAdd((unsigned long long *)&d[2 * ag],
-                              (unsigned long long)st[(size_t)idx * na * 2 +
-                                                     2 * ag]);
-                    atomicAdd((double *)&d[2 * ag + 1],
-                              *(const double *)&st[(size_t)idx * na * 2 +
-                                                   2 * ag + 1]);
-                    break;
-                }
-            }
+            atomic_merge(d, st + (size_t)idx * na * 2, M.agg);
         };
         if (PACKED) {
             for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
@@ -1850,8 +1868,10 @@ k_merge_fused(MergeFusedArgs M) {
             switch (M.agg.op[ag]) {
             case AMD_AGG_COUNT:
             case AMD_AGG_SUM: M.out[col++][r] = (int64_t)w0; break;
-            case AMD_AGG_MIN: M.out[col++][r] = dec_min(w0); break;
-            case AMD_AGG_MAX: M.out[col++][r] = dec_max(w0); break;
+            case AMD_AGG_MIN:
+            case AMD_AGG_MAX:
+                M.out[col++][r] = dec_word(M.agg.op[ag], M.agg.isf[ag], w0);
+                break;
             case AMD_AGG_AVG: {
                 double v = w0 ? (*(const double *)(d + 2 * ag + 1)) /
                                     (double)(int64_t)w0
@@ -1880,7 +1900,19 @@ k_merge_fused(MergeFusedArgs M) {
             for (int ag = 0; ag < na; ag++) {
                 switch (M.agg.op[ag]) {
                 case AMD_AGG_COUNT:
-                case AMD_AGG_SUM: spec[2 * ag] += st[2 * ag]; break;
+                    spec[2 * ag] += st[2 * ag];
+                    break;
+                case AMD_AGG_SUM:
+                    if (M.agg.isf[ag]) {
+                        double x, y;
+                        memcpy(&x, &spec[2 * ag], 8);
+                        memcpy(&y, &st[2 * ag], 8);
+                        x += y;
+                        memcpy(&spec[2 * ag], &x, 8);
+                    } else {
+                        spec[2 * ag] += st[2 * ag];
+                    }
+                    break;
                 case AMD_AGG_MIN:
                 case AMD_AGG_MAX:
                     if (st[2 * ag] > spec[2 * ag]) spec[2 * ag] = st[2 * ag];
@@ -1907,8 +1939,11 @@ k_merge_fused(MergeFusedArgs M) {
                 switch (M.agg.op[ag]) {
                 case AMD_AGG_COUNT:
                 case AMD_AGG_SUM: M.out[col++][r] = (int64_t)w0; break;
-                case AMD_AGG_MIN: M.out[col++][r] = dec_min(w0); break;
-                case AMD_AGG_MAX: M.out[col++][r] = dec_max(w0); break;
+                case AMD_AGG_MIN:
+                case AMD_AGG_MAX:
+                    M.out[col++][r] = dec_word(M.agg.op[ag], M.agg.isf[ag],
+                                               w0);
+                    break;
                 case AMD_AGG_AVG: {
                     double v = w0 ? (*(const double *)(spec + 2 * ag + 1)) /
                                         (double)(int64_t)w0
@@ -1953,16 +1988,9 @@ k_restore(RestoreArgs R) {
         int c = 0;
         for (int a = 0; a < R.agg.n_aggs; a++) {
             int64_t w0 = R.scols[c][i];
-            switch (R.agg.op[a]) {
-            case AMD_AGG_COUNT:
-            case AMD_AGG_SUM: enc[2 * a] = (uint64_t)w0; break;
-            case AMD_AGG_MIN: enc[2 * a] = enc_min(w0); break;
-            case AMD_AGG_MAX: enc[2 * a] = enc_max(w0); break;
-            case AMD_AGG_AVG:
-                enc[2 * a] = (uint64_t)w0;
+            enc[2 * a] = enc_word(R.agg.op[a], R.agg.isf[a], w0);
+            if (R.agg.op[a] == AMD_AGG_AVG)
                 enc[2 * a + 1] = (uint64_t)R.scols[c + 1][i];
-                break;
-            }
             c += R.swords[a];
         }
         uint32_t p = R.pane;
@@ -2204,6 +2232,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < cfg->n_aggs; i++) {
         o->agg.op[i] = cfg->agg_ops[i];
         o->agg.col[i] = cfg->agg_col[i];
+        o->agg.isf[i] = cfg->agg_col[i] >= 0 &&
+                        cfg->val_is_f64[cfg->agg_col[i]];
     }
     o->n_in_cols = cfg->n_keys + cfg->n_value_cols + 1;
     o->out_cols = cfg->n_keys + cfg->n_aggs + 3;
@@ -2967,7 +2997,7 @@ static int build_out(GpuOp *o, AmdOutBatch *out, int raw_states) {
     }
     if (!raw_states) {
         for (int a = 0; a < o->agg.n_aggs; a++)
-            if (o->agg.op[a] == AMD_AGG_AVG)
+            if (o->agg.op[a] == AMD_AGG_AVG || o->agg.isf[a])
                 out->is_f64[o->cfg.n_keys + a] = 1;
     } else {
         int col = o->cfg.n_keys;
@@ -2976,6 +3006,7 @@ static int build_out(GpuOp *o, AmdOutBatch *out, int raw_states) {
                 out->is_f64[col + 1] = 1;
                 col += 2;
             } else {
+                if (o->agg.isf[a]) out->is_f64[col] = 1;
                 col += 1;
             }
         }

@@ -48,11 +48,15 @@ typedef struct {
     int32_t  n_aggs;
     int32_t  agg_ops[AMD_MAX_AGGS];
     int32_t  agg_col[AMD_MAX_AGGS];  /* value-column index; -1 for COUNT(*) */
-    int32_t  n_value_cols;    /* i64 value columns between keys and _timestamp */
+    int32_t  n_value_cols;    /* value columns between keys and _timestamp */
     uint32_t log2_capacity;   /* hash slots per pane (GPU path) */
     uint32_t ring_panes;      /* live-pane ring size (GPU path, power of 2) */
     int32_t  device;          /* HIP device ordinal (GPU path) */
     int32_t  emit_to_host;    /* GPU path: 1 = outputs copied to host memory */
+    int32_t  val_is_f64[8];   /* value column v carries f64 bit patterns in
+                                 its i64 plane (Arrow Float64 column);
+                                 SUM/MIN/MAX/AVG over it aggregate as
+                                 doubles and the output column is f64 */
 } AmdWindowConfig;
 
 /* Instant (windowed stream-stream) join configuration.  Mirrors

@@ -17,7 +17,7 @@ def build(force=False):
             os.path.getmtime(_SO) < os.path.getmtime(_SRC):
         subprocess.run(
             ["gcc", "-O2", "-g", "-shared", "-fPIC", "-fvisibility=hidden",
-             "-o", _SO, _SRC],
+             "-o", _SO, _SRC, "-lm"],
             check=True, cwd=_DIR)
     return _SO
 

@@ -33,6 +33,7 @@
 #include <stdint.h>
 #include <stdlib.h>
 #include <string.h>
+#include <math.h>
 #include <stdio.h>
 
 #include "../include/arroyo_amd_types.h"
@@ -99,13 +100,23 @@ static void table_free(Table *t) {
     free(t->keys); free(t->used); free(t->st); free(t);
 }
 
+/* value column of agg i is f64 (bit patterns in the i64 plane) */
+static int agg_isf(const AmdWindowConfig *c, int i) {
+    return c->agg_col[i] >= 0 && c->val_is_f64[c->agg_col[i]];
+}
+
 static void st_init(const AmdWindowConfig *c, int64_t *s) {
     for (int i = 0; i < c->n_aggs; i++) {
+        int f = agg_isf(c, i);
         switch (c->agg_ops[i]) {
         case AMD_AGG_COUNT: s[2*i] = 0; break;
-        case AMD_AGG_SUM:   s[2*i] = 0; break;
-        case AMD_AGG_MIN:   s[2*i] = INT64_MAX; break;
-        case AMD_AGG_MAX:   s[2*i] = INT64_MIN; break;
+        case AMD_AGG_SUM:   s[2*i] = f ? d_to_bits(0.0) : 0; break;
+        case AMD_AGG_MIN:
+            s[2*i] = f ? d_to_bits(INFINITY) : INT64_MAX;
+            break;
+        case AMD_AGG_MAX:
+            s[2*i] = f ? d_to_bits(-INFINITY) : INT64_MIN;
+            break;
         case AMD_AGG_AVG:   s[2*i] = 0; s[2*i+1] = d_to_bits(0.0); break;
         }
     }
@@ -145,14 +156,31 @@ static void st_update(const AmdWindowConfig *c, int64_t *s,
                       const int64_t *const *vcols, int64_t row) {
     for (int i = 0; i < c->n_aggs; i++) {
         int col = c->agg_col[i];
+        int f = agg_isf(c, i);
         switch (c->agg_ops[i]) {
         case AMD_AGG_COUNT: s[2*i]++; break;
-        case AMD_AGG_SUM:   s[2*i] += vcols[col][row]; break;
-        case AMD_AGG_MIN: { int64_t v = vcols[col][row]; if (v < s[2*i]) s[2*i] = v; } break;
-        case AMD_AGG_MAX: { int64_t v = vcols[col][row]; if (v > s[2*i]) s[2*i] = v; } break;
+        case AMD_AGG_SUM:
+            if (f)
+                s[2*i] = d_to_bits(bits_to_d(s[2*i]) +
+                                   bits_to_d(vcols[col][row]));
+            else
+                s[2*i] += vcols[col][row];
+            break;
+        case AMD_AGG_MIN: {
+            int64_t v = vcols[col][row];
+            if (f ? bits_to_d(v) < bits_to_d(s[2*i]) : v < s[2*i])
+                s[2*i] = v;
+        } break;
+        case AMD_AGG_MAX: {
+            int64_t v = vcols[col][row];
+            if (f ? bits_to_d(v) > bits_to_d(s[2*i]) : v > s[2*i])
+                s[2*i] = v;
+        } break;
         case AMD_AGG_AVG:
             s[2*i]++;
-            s[2*i+1] = d_to_bits(bits_to_d(s[2*i+1]) + (double)vcols[col][row]);
+            s[2*i+1] = d_to_bits(bits_to_d(s[2*i+1]) +
+                                 (f ? bits_to_d(vcols[col][row])
+                                    : (double)vcols[col][row]));
             break;
         }
     }
@@ -160,11 +188,23 @@ static void st_update(const AmdWindowConfig *c, int64_t *s,
 
 static void st_merge(const AmdWindowConfig *c, int64_t *a, const int64_t *b) {
     for (int i = 0; i < c->n_aggs; i++) {
+        int f = agg_isf(c, i);
         switch (c->agg_ops[i]) {
         case AMD_AGG_COUNT: a[2*i] += b[2*i]; break;
-        case AMD_AGG_SUM:   a[2*i] += b[2*i]; break;
-        case AMD_AGG_MIN:   if (b[2*i] < a[2*i]) a[2*i] = b[2*i]; break;
-        case AMD_AGG_MAX:   if (b[2*i] > a[2*i]) a[2*i] = b[2*i]; break;
+        case AMD_AGG_SUM:
+            if (f)
+                a[2*i] = d_to_bits(bits_to_d(a[2*i]) + bits_to_d(b[2*i]));
+            else
+                a[2*i] += b[2*i];
+            break;
+        case AMD_AGG_MIN:
+            if (f ? bits_to_d(b[2*i]) < bits_to_d(a[2*i]) : b[2*i] < a[2*i])
+                a[2*i] = b[2*i];
+            break;
+        case AMD_AGG_MAX:
+            if (f ? bits_to_d(b[2*i]) > bits_to_d(a[2*i]) : b[2*i] > a[2*i])
+                a[2*i] = b[2*i];
+            break;
         case AMD_AGG_AVG:
             a[2*i] += b[2*i];
             a[2*i+1] = d_to_bits(bits_to_d(a[2*i+1]) + bits_to_d(b[2*i+1]));
@@ -456,7 +496,7 @@ ORACLE_API int oracle_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
                 memcpy(out->cols[i], o->out[i], (size_t)o->out_rows * 8);
         }
         for (int a = 0; a < c->n_aggs; a++)
-            if (c->agg_ops[a] == AMD_AGG_AVG)
+            if (c->agg_ops[a] == AMD_AGG_AVG || agg_isf(c, a))
                 out->is_f64[c->n_keys + a] = 1;
         o->out_rows = 0;
     }
@@ -486,7 +526,7 @@ ORACLE_API int oracle_checkpoint_drain(void *h, AmdOutBatch *out) {
         int col = c->n_keys;
         for (int a = 0; a < c->n_aggs; a++) {
             if (c->agg_ops[a] == AMD_AGG_AVG) { out->is_f64[col + 1] = 1; col += 2; }
-            else col += 1;
+            else { if (agg_isf(c, a)) out->is_f64[col] = 1; col += 1; }
         }
     }
     int64_t r = 0;
