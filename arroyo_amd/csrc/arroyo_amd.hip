@@ -1581,6 +1581,123 @@ k_radix_agg(RadixAggArgs A) {
 }
 
 /* ------------------------------------------------------------------ */
+/* Multi-column keys (n_keys 2..4): a device key dictionary normalizes
+ * each composite key to a 64-bit id (its dictionary slot), and the
+ * single-key engine runs unchanged on the ids; emissions decode ids back
+ * to the original tuples.  This mirrors the reference's own key
+ * normalization — it converts every key tuple to Arrow Row bytes before
+ * the state map (expiring_time_key_map.rs:1008-1049) — with the
+ * normalization done once, on device.  Exactness: slots are claimed by
+ * digest CAS but verified by full-tuple compare; a digest collision just
+ * probes on.  The dictionary never evicts (ids must stay stable for live
+ * panes); capacity is 4x the pane table and exhaustion is a loud error. */
+#define ERR_DICT_FULL 7
+#define ERR_DICT_SPIN 8
+
+struct DictEncArgs {
+    const int64_t *kcols[4];
+    int32_t nk;
+    int64_t n_rows;
+    int64_t *digest;     /* [D]; EMPTY_KEY = free (stored digests have the
+                            top bit cleared, so they never equal -1) */
+    int64_t *dkeys;      /* [D][nk] */
+    uint32_t *ready;     /* [D] tuple words published */
+    int64_t *ids;        /* [n_rows] out */
+    uint32_t D;
+    int *err;
+};
+
+__device__ inline uint64_t dev_hash_tuple(const int64_t *key, int nk) {
+    uint64_t h = hash64((uint64_t)key[0]);
+    for (int k = 1; k < nk; k++) h = hash64(h ^ (uint64_t)key[k]);
+    return h;
+}
+
+__global__ void __launch_bounds__(256)
+k_dict_encode(DictEncArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const uint64_t m = A.D - 1;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < A.n_rows; r += stride) {
+        int64_t key[4];
+        for (int k = 0; k < A.nk; k++) key[k] = A.kcols[k][r];
+        uint64_t h = dev_hash_tuple(key, A.nk);
+        int64_t dg = (int64_t)(h & 0x7FFFFFFFFFFFFFFFULL);
+        bool done = false;
+        /* outer retry loop instead of a divergent spin-wait: a lane that
+         * finds a claimed-but-unpublished slot backs off so the wave
+         * reconverges and the writer lane's stores can execute */
+        for (int tries = 0; tries < (1 << 20) && !done; tries++) {
+            uint64_t i = h & m;
+            for (uint32_t probes = 0; probes < A.D; probes++) {
+                int64_t cur = A.digest[i];
+                if (cur == EMPTY_KEY) {
+                    int64_t old = (int64_t)atomicCAS(
+                        (unsigned long long *)&A.digest[i],
+                        (unsigned long long)EMPTY_KEY,
+                        (unsigned long long)dg);
+                    if (old == EMPTY_KEY) {
+                        for (int k = 0; k < A.nk; k++)
+                            A.dkeys[i * A.nk + k] = key[k];
+                        __threadfence();
+                        atomicExch(&A.ready[i], 1u);
+                        A.ids[r] = (int64_t)i;
+                        done = true;
+                        break;
+                    }
+                    cur = old;
+                }
+                if (cur == dg) {
+                    if (!atomicAdd(&A.ready[i], 0u))
+                        break;   /* writer not published yet: retry row */
+                    bool eq = true;
+                    for (int k = 0; k < A.nk && eq; k++)
+                        eq = A.dkeys[i * A.nk + k] == key[k];
+                    if (eq) {
+                        A.ids[r] = (int64_t)i;
+                        done = true;
+                        break;
+                    }
+                    /* digest collision of a different tuple: probe on */
+                }
+                i = (i + 1) & m;
+                if (probes + 1 == A.D) {
+                    *A.err = ERR_DICT_FULL;
+                    A.ids[r] = 0;
+                    done = true;
+                    break;
+                }
+            }
+        }
+        if (!done) {
+            *A.err = ERR_DICT_SPIN;
+            A.ids[r] = 0;
+        }
+    }
+}
+
+/* expand an id column back to the original key tuple columns */
+struct DictDecArgs {
+    const int64_t *ids;
+    const unsigned long long *n;   /* device row count */
+    const int64_t *dkeys;
+    int32_t nk;
+    int64_t *out[4];
+};
+
+__global__ void __launch_bounds__(256)
+k_dict_decode(DictDecArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t n = (int64_t)*A.n;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
+        int64_t id = A.ids[r];
+        for (int k = 0; k < A.nk; k++)
+            A.out[k][r] = A.dkeys[id * A.nk + k];
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* K4: merge source panes into the merge table.                        */
 
 struct MergeArgs {
@@ -2129,6 +2246,15 @@ struct GpuOp {
     int use_lds;
     int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
                        3 batched-probe, 4 radix-regroup */
+    /* multi-column keys: device dictionary (see k_dict_encode) */
+    int mk;                     /* 1 when cfg.n_keys >= 2 */
+    int64_t *d_dict_digest;
+    int64_t *d_dict_keys;
+    uint32_t *d_dict_ready;
+    uint32_t dict_D;
+    int64_t *d_keyid_in;
+    int64_t keyid_in_cap;
+    int64_t *d_keyid_out;
     int64_t *rdx2_skey;
     uint32_t *rdx2_spane;
     uint32_t *rdx2_hist;
@@ -2203,7 +2329,7 @@ static int flush_staged(GpuOp *o);
 
 API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (!cfg || cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS ||
-        cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->width_nanos == 0 ||
+        cfg->n_keys < 0 || cfg->n_keys > 4 || cfg->width_nanos == 0 ||
         cfg->n_value_cols > 4) {
         snprintf(g_err, sizeof g_err, "invalid config");
         return nullptr;
@@ -2326,6 +2452,18 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
+    o->mk = o->cfg.n_keys >= 2;
+    if (o->mk) {
+        /* dictionary 4x the pane table; ids must stay stable, so no
+         * eviction — exhaustion is a loud ERR_DICT_FULL */
+        o->dict_D = o->ring.C << 2;
+        ALLOC(o->d_dict_digest, (size_t)o->dict_D * 8);
+        ALLOC(o->d_dict_keys, (size_t)o->dict_D * o->cfg.n_keys * 8);
+        ALLOC(o->d_dict_ready, (size_t)o->dict_D * 4);
+        ALLOC(o->d_keyid_out, (size_t)o->out_rows_cap * 8);
+        hipMemset(o->d_dict_digest, 0xFF, (size_t)o->dict_D * 8);
+        hipMemset(o->d_dict_ready, 0, (size_t)o->dict_D * 4);
+    }
 #undef ALLOC
     if (o->ring.packed) {
         int64_t n_slots = (int64_t)o->ring.R * o->ring.C;
@@ -2476,7 +2614,12 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     }
     if (o->upd_kind == 4) {
         if (n_rows > o->rdx2_cap) {
-            hipFree(o->rdx2_skey);
+            hipFree(o->d_dict_digest);
+    hipFree(o->d_dict_keys);
+    hipFree(o->d_dict_ready);
+    hipFree(o->d_keyid_in);
+    hipFree(o->d_keyid_out);
+    hipFree(o->rdx2_skey);
             hipFree(o->rdx2_spane);
             o->rdx2_cap = n_rows + (n_rows >> 2);
             HIP_CHECK(o, hipMalloc((void **)&o->rdx2_skey,
@@ -2643,6 +2786,44 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     return 0;
 }
 
+/* normalize a batch's composite keys to dictionary ids (device cols) */
+static int mk_encode(GpuOp *o, const int64_t *const *kcols, int64_t n_rows) {
+    if (n_rows > o->keyid_in_cap) {
+        hipFree(o->d_keyid_in);
+        o->keyid_in_cap = n_rows + (n_rows >> 2);
+        HIP_CHECK(o, hipMalloc((void **)&o->d_keyid_in,
+                               (size_t)o->keyid_in_cap * 8));
+    }
+    DictEncArgs E = {};
+    for (int k = 0; k < o->cfg.n_keys; k++) E.kcols[k] = kcols[k];
+    E.nk = o->cfg.n_keys;
+    E.n_rows = n_rows;
+    E.digest = o->d_dict_digest;
+    E.dkeys = o->d_dict_keys;
+    E.ready = o->d_dict_ready;
+    E.ids = o->d_keyid_in;
+    E.D = o->dict_D;
+    E.err = o->ring.err;
+    int64_t want = (n_rows + 255) / 256;
+    int blocks = (int)(want > 2048 ? 2048 : (want < 1 ? 1 : want));
+    hipLaunchKernelGGL(k_dict_encode, dim3(blocks), dim3(256), 0, o->stream,
+                       E);
+    HIP_CHECK(o, hipGetLastError());
+    return 0;
+}
+
+/* build the single-key engine view of a user batch: [ids, vals..., ts] */
+static int mk_engine_cols(GpuOp *o, const int64_t *const *dcols,
+                          int64_t n_rows, const int64_t **eng) {
+    if (mk_encode(o, dcols, n_rows)) return 1;
+    int c = 0;
+    eng[c++] = o->d_keyid_in;
+    for (int v = 0; v < o->cfg.n_value_cols; v++)
+        eng[c++] = dcols[o->cfg.n_keys + v];
+    eng[c] = dcols[o->cfg.n_keys + o->cfg.n_value_cols];
+    return 0;
+}
+
 static int flush_staged(GpuOp *o) {
     if (o->stg.n == 0) return 0;
     for (int i = 0; i < o->n_in_cols; i++)
@@ -2650,8 +2831,17 @@ static int flush_staged(GpuOp *o) {
                                     (size_t)o->stg.n * 8,
                                     hipMemcpyHostToDevice, o->stream));
     const int64_t *dcols[6];
-    for (int i = 0; i < o->n_in_cols; i++) dcols[i] = o->stg.dbuf[i];
-    int rc = launch_update(o, dcols, o->stg.n, 0);
+    int rc;
+    if (o->mk) {
+        const int64_t *eng[6];
+        if (mk_engine_cols(o, (const int64_t *const *)o->stg.dbuf, o->stg.n,
+                           eng))
+            return 1;
+        rc = launch_update(o, eng, o->stg.n, 0);
+    } else {
+        for (int i = 0; i < o->n_in_cols; i++) dcols[i] = o->stg.dbuf[i];
+        rc = launch_update(o, dcols, o->stg.n, 0);
+    }
     o->stg.n = 0;
     return rc;
 }
@@ -2689,6 +2879,11 @@ API int arroyo_amd_process_batch_device(void *h, const int64_t *const *dcols,
         return 1;
     }
     if (flush_staged(o)) return 1;
+    if (o->mk) {
+        const int64_t *eng[6];
+        if (mk_engine_cols(o, dcols, n_rows, eng)) return 1;
+        return launch_update(o, eng, n_rows, ts_offset);
+    }
     return launch_update(o, dcols, n_rows, ts_offset);
 }
 
@@ -2714,12 +2909,26 @@ API int arroyo_amd_process_batches_device(void *h,
      * adjacent memory: fuse them into a single launch (same computation,
      * ~15x the rows per launch -- the reference likewise drains its queue
      * into the per-bin execs in bulk) */
-    if (contiguous && ts_step == 0)
+    if (contiguous && ts_step == 0) {
+        if (o->mk) {
+            const int64_t *eng[6];
+            if (mk_engine_cols(o, dcols, n_rows * reps, eng)) return 1;
+            return launch_update(o, eng, n_rows * reps, ts_offset0);
+        }
         return launch_update(o, dcols, n_rows * reps, ts_offset0);
+    }
     const int64_t *cols[16];
     for (int k = 0; k < reps; k++) {
         for (int c = 0; c < n_cols; c++)
             cols[c] = dcols[c] + (contiguous ? (int64_t)k * n_rows : 0);
+        if (o->mk) {
+            const int64_t *eng[6];
+            if (mk_engine_cols(o, cols, n_rows, eng)) return 1;
+            if (launch_update(o, eng, n_rows,
+                              ts_offset0 + (uint64_t)k * ts_step))
+                return 1;
+            continue;
+        }
         if (launch_update(o, cols, n_rows, ts_offset0 + (uint64_t)k * ts_step))
             return 1;
     }
@@ -2749,6 +2958,17 @@ static int check_device_error(GpuOp *o) {
         snprintf(o->err_msg, sizeof o->err_msg,
                  "fused-merge LDS table overflow (pathological key "
                  "clustering); raise log2_capacity");
+        return 1;
+    }
+    if (e == ERR_DICT_FULL) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "key dictionary full (capacity %u composite keys over the "
+                 "operator lifetime); raise log2_capacity", o->dict_D);
+        return 1;
+    }
+    if (e == ERR_DICT_SPIN) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "key dictionary publish wait exhausted (device anomaly)");
         return 1;
     }
     if (e) {
@@ -2791,9 +3011,18 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             for (size_t i = 0; i < src.size(); i++) M.src[i] = src[i];
             M.win_start = ws;
             M.win_end = we;
-            M.n_keys = o->cfg.n_keys;
-            for (int i = 0; i < o->n_out_alloc && i < 16; i++)
-                M.out[i] = o->d_out[i];
+            M.n_keys = o->cfg.n_keys ? 1 : 0;
+            if (o->mk) {
+                /* engine emits the dictionary id; decode expands it into
+                 * the k key columns afterwards */
+                M.out[0] = o->d_keyid_out;
+                for (int i = 1; i < o->n_out_alloc - o->cfg.n_keys + 1 &&
+                                i < 16; i++)
+                    M.out[i] = o->d_out[o->cfg.n_keys - 1 + i];
+            } else {
+                for (int i = 0; i < o->n_out_alloc && i < 16; i++)
+                    M.out[i] = o->d_out[i];
+            }
             M.n_out = o->d_n_out;
             /* emitted-row accounting folded into the kernel for the
              * device-resident path (saves the separate k_accum launch) */
@@ -2832,6 +3061,17 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                                    dim3(o->ring.C / range), dim3(256),
                                    shmem, o->stream, M);
             HIP_CHECK(o, hipGetLastError());
+            if (o->mk) {
+                DictDecArgs DD = {};
+                DD.ids = o->d_keyid_out;
+                DD.n = o->d_n_out;
+                DD.dkeys = o->d_dict_keys;
+                DD.nk = o->cfg.n_keys;
+                for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = o->d_out[k];
+                hipLaunchKernelGGL(k_dict_decode, dim3(1024), dim3(256), 0,
+                                   o->stream, DD);
+                HIP_CHECK(o, hipGetLastError());
+            }
         }
         if (!o->cfg.emit_to_host) return 0;
         unsigned long long n = 0;
@@ -2878,16 +3118,34 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
     C.m_spec_state = o->m_spec_state;
     C.CM = o->CM;
     C.agg = o->agg;
-    C.n_keys = o->cfg.n_keys;
+    C.n_keys = o->cfg.n_keys ? 1 : 0;
     C.raw_states = raw_states;
     C.win_start = raw_states ? bin_ts : ws;
     C.win_end = we;
-    for (int i = 0; i < o->n_out_alloc && i < 16; i++) C.out[i] = o->d_out[i];
+    if (o->mk) {
+        C.out[0] = o->d_keyid_out;
+        for (int i = 1; i < o->n_out_alloc - o->cfg.n_keys + 1 && i < 16; i++)
+            C.out[i] = o->d_out[o->cfg.n_keys - 1 + i];
+    } else {
+        for (int i = 0; i < o->n_out_alloc && i < 16; i++)
+            C.out[i] = o->d_out[i];
+    }
     C.n_out = o->d_n_out;
     int blocks = (int)((o->CM + 255) / 256);
     if (blocks > 1024) blocks = 1024;   /* 1 global cursor atomic per block */
     hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(256), 0, o->stream, C);
     HIP_CHECK(o, hipGetLastError());
+    if (o->mk) {
+        DictDecArgs DD = {};
+        DD.ids = o->d_keyid_out;
+        DD.n = o->d_n_out;
+        DD.dkeys = o->d_dict_keys;
+        DD.nk = o->cfg.n_keys;
+        for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = o->d_out[k];
+        hipLaunchKernelGGL(k_dict_decode, dim3(1024), dim3(256), 0,
+                           o->stream, DD);
+        HIP_CHECK(o, hipGetLastError());
+    }
     if (!o->cfg.emit_to_host && !raw_states) {
         /* device-resident emission: the next pipeline stage's collector
          * consumes d_out in place.  Accounting stays on device so firing
@@ -3158,7 +3416,14 @@ API int arroyo_amd_restore(void *h, const int64_t *const *cols,
         }
         RestoreArgs R = {};
         int c2 = 0;
-        R.key_col = o->cfg.n_keys ? o->stg.dbuf[c2++] : nullptr;
+        if (o->mk) {
+            if (mk_encode(o, (const int64_t *const *)o->stg.dbuf, n))
+                return 1;
+            R.key_col = o->d_keyid_in;
+            c2 = o->cfg.n_keys;
+        } else {
+            R.key_col = o->cfg.n_keys ? o->stg.dbuf[c2++] : nullptr;
+        }
         for (int s = 0; s < swords_total; s++) R.scols[s] = o->stg.dbuf[c2++];
         memcpy(R.swords, swords, sizeof swords);
         R.n_rows = n;
@@ -3237,6 +3502,11 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
     hipFree(o->rdx_tmp);
+    hipFree(o->d_dict_digest);
+    hipFree(o->d_dict_keys);
+    hipFree(o->d_dict_ready);
+    hipFree(o->d_keyid_in);
+    hipFree(o->d_keyid_out);
     hipFree(o->rdx2_skey);
     hipFree(o->rdx2_spane);
     hipFree(o->rdx2_hist);

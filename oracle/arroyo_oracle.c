@@ -86,10 +86,15 @@ static uint64_t hash64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
-static Table *table_new(int cap_log2, int n_aggs) {
+/* internal key width: tuples of up to 4 i64 key columns (ArroyoSchema
+ * key_indices, crates/arroyo-rpc/src/df.rs:24-30); unkeyed ops store one
+ * zero word */
+#define NKI(c) ((c)->n_keys > 0 ? (c)->n_keys : 1)
+
+static Table *table_new(int cap_log2, int n_aggs, int nk) {
     Table *t = calloc(1, sizeof(Table));
     t->cap = 1 << cap_log2;
-    t->keys = malloc((size_t)t->cap * 8);
+    t->keys = malloc((size_t)t->cap * nk * 8);
     t->used = calloc((size_t)t->cap, 1);
     t->st = malloc((size_t)t->cap * n_aggs * 16);
     return t;
@@ -124,28 +129,47 @@ static void st_init(const AmdWindowConfig *c, int64_t *s) {
 
 static void table_grow(const AmdWindowConfig *c, Table *t);
 
-static int64_t *table_slot(const AmdWindowConfig *c, Table *t, int64_t key) {
+static uint64_t hash_tuple(const int64_t *key, int nk) {
+    uint64_t h = hash64((uint64_t)key[0]);
+    for (int k = 1; k < nk; k++)
+        h = hash64(h ^ (uint64_t)key[k]);
+    return h;
+}
+
+static int tuple_eq(const int64_t *a, const int64_t *b, int nk) {
+    for (int k = 0; k < nk; k++)
+        if (a[k] != b[k]) return 0;
+    return 1;
+}
+
+static int64_t *table_slot(const AmdWindowConfig *c, Table *t,
+                           const int64_t *key) {
+    int nk = NKI(c);
     if (t->n * 10 >= (int64_t)t->cap * 7) table_grow(c, t);
     uint64_t m = (uint64_t)t->cap - 1;
-    uint64_t i = hash64((uint64_t)key) & m;
-    while (t->used[i] && t->keys[i] != key) i = (i + 1) & m;
+    uint64_t i = hash_tuple(key, nk) & m;
+    while (t->used[i] && !tuple_eq(&t->keys[i * nk], key, nk))
+        i = (i + 1) & m;
     if (!t->used[i]) {
-        t->used[i] = 1; t->keys[i] = key; t->n++;
+        t->used[i] = 1;
+        memcpy(&t->keys[i * nk], key, (size_t)nk * 8);
+        t->n++;
         st_init(c, &t->st[i * c->n_aggs * 2]);
     }
     return &t->st[i * c->n_aggs * 2];
 }
 
 static void table_grow(const AmdWindowConfig *c, Table *t) {
+    int nk = NKI(c);
     int ocap = t->cap;
     int64_t *ok = t->keys; uint8_t *ou = t->used; int64_t *os = t->st;
     t->cap <<= 1; t->n = 0;
-    t->keys = malloc((size_t)t->cap * 8);
+    t->keys = malloc((size_t)t->cap * nk * 8);
     t->used = calloc((size_t)t->cap, 1);
     t->st = malloc((size_t)t->cap * c->n_aggs * 16);
     for (int i = 0; i < ocap; i++) {
         if (!ou[i]) continue;
-        int64_t *s = table_slot(c, t, ok[i]);
+        int64_t *s = table_slot(c, t, &ok[(size_t)i * nk]);
         memcpy(s, &os[(size_t)i * c->n_aggs * 2], (size_t)c->n_aggs * 16);
         /* table_slot counted it as new and re-inited; restore by copy above */
     }
@@ -233,7 +257,7 @@ static Table *panes_get(int *n, int *cap, uint64_t **bins, Table ***tables,
     memmove(*bins + i + 1, *bins + i, (size_t)(*n - i) * 8);
     memmove(*tables + i + 1, *tables + i, (size_t)(*n - i) * sizeof(Table *));
     (*bins)[i] = b;
-    Table *t = table_new(8, c->n_aggs);
+    Table *t = table_new(8, c->n_aggs, NKI(c));
     (*tables)[i] = t;
     (*n)++;
     return t;
@@ -300,7 +324,8 @@ static void emit_table(Op *o, Table *t, uint64_t ws, uint64_t we) {
         if (!t->used[i]) continue;
         int64_t r = o->out_rows++;
         int col = 0;
-        if (c->n_keys) o->out[col++][r] = t->keys[i];
+        for (int kk = 0; kk < c->n_keys; kk++)
+            o->out[col++][r] = t->keys[(size_t)i * NKI(c) + kk];
         const int64_t *s = &t->st[(size_t)i * c->n_aggs * 2];
         for (int a = 0; a < c->n_aggs; a++) {
             if (c->agg_ops[a] == AMD_AGG_AVG) {
@@ -348,7 +373,8 @@ static void advance(Op *o) {
         } else {
             for (int i = 0; i < pane->cap; i++)
                 if (pane->used[i])
-                    st_merge(c, table_slot(c, tgt, pane->keys[i]),
+                    st_merge(c, table_slot(c, tgt,
+                                           &pane->keys[(size_t)i * NKI(c)]),
                              &pane->st[(size_t)i * c->n_aggs * 2]);
             table_free(pane);
         }
@@ -360,13 +386,14 @@ static void advance(Op *o) {
 
     /* merge closed panes in [E-width, E) :161-196 */
     uint64_t lo = (E >= c->width_nanos) ? E - c->width_nanos : 0;
-    Table *merged = table_new(8, c->n_aggs);
+    Table *merged = table_new(8, c->n_aggs, NKI(c));
     for (int p = 0; p < o->n_closed; p++) {
         if (o->closed_bins[p] < lo || o->closed_bins[p] >= E) continue;
         Table *t = o->closed_tables[p];
         for (int i = 0; i < t->cap; i++)
             if (t->used[i])
-                st_merge(c, table_slot(c, merged, t->keys[i]),
+                st_merge(c, table_slot(c, merged,
+                                       &t->keys[(size_t)i * NKI(c)]),
                          &t->st[(size_t)i * c->n_aggs * 2]);
     }
     /* delete_before(bin_end + slide - width) :173-174 */
@@ -404,7 +431,7 @@ static void advance(Op *o) {
 /* ------------------------------------------------------------------ */
 ORACLE_API void *oracle_create(const AmdWindowConfig *cfg) {
     if (!cfg || cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS ||
-        cfg->n_keys < 0 || cfg->n_keys > 1 || cfg->width_nanos == 0)
+        cfg->n_keys < 0 || cfg->n_keys > 4 || cfg->width_nanos == 0)
         return NULL;
     /* the reference's planner rejects non-divisible hop parameters
      * ("hop() width currently must be a multiple of slide",
@@ -440,7 +467,6 @@ ORACLE_API int oracle_process_batch(void *h, const int64_t *const *cols,
                  c->n_keys + c->n_value_cols + 1, n_cols);
         return 1;
     }
-    const int64_t *kcol = c->n_keys ? cols[0] : NULL;
     const int64_t *const *vcols = cols + c->n_keys;
     const int64_t *ts = cols[n_cols - 1];
     uint64_t wmb = o->has_wm ? bin_of(o->wm, c->slide_nanos) : 0;
@@ -457,7 +483,8 @@ ORACLE_API int oracle_process_batch(void *h, const int64_t *const *cols,
                                &o->open_tables, b, c, 1);
             cached_bin = b;
         }
-        int64_t key = kcol ? kcol[r] : 0;
+        int64_t key[4] = {0, 0, 0, 0};
+        for (int kk = 0; kk < c->n_keys; kk++) key[kk] = cols[kk][r];
         st_update(c, table_slot(c, cached, key), vcols, r);
     }
     return 0;
@@ -537,7 +564,9 @@ ORACLE_API int oracle_checkpoint_drain(void *h, AmdOutBatch *out) {
         for (int i = 0; i < t->cap; i++) {
             if (!t->used[i]) continue;
             int col = 0;
-            if (c->n_keys) ((int64_t *)out->cols[col++])[r] = t->keys[i];
+            for (int kk = 0; kk < c->n_keys; kk++)
+                ((int64_t *)out->cols[col++])[r] =
+                    t->keys[(size_t)i * NKI(c) + kk];
             const int64_t *s = &t->st[(size_t)i * c->n_aggs * 2];
             for (int a = 0; a < c->n_aggs; a++) {
                 ((int64_t *)out->cols[col++])[r] = s[2*a];
