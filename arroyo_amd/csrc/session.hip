@@ -403,11 +403,16 @@ k_sess_update(UpdateArgs A) {
 
 struct Store {
     int64_t *keys;     /* [C+1]; slot C = spec (key == EMPTY_KEY) */
-    uint32_t *ns;      /* [C+1] live sessions per key */
-    int64_t *s_start;  /* [(C+1)*MS] */
-    int64_t *s_end;    /* [(C+1)*MS] data_end */
-    uint64_t *s_st;    /* [(C+1)*MS][n_aggs*2] */
+    /* per-slot interleaved record: [ns, (start, end, st[sw]) x MS].  One
+     * 64B line covers ns + the whole first session for sw <= 4, so the
+     * phase-2 merge touches ~2 random lines per key (key probe + record)
+     * instead of the 5 the split ns/s_start/s_end/s_st planes cost —
+     * round 1 measured the store random walk as the residual bottleneck
+     * (profiles/r01_session_note.md). */
+    int64_t *recs;     /* [(C+1)][rec_w] */
     uint32_t C, MS;
+    uint32_t sess_w;   /* words per session = 2 + 2*n_aggs */
+    uint32_t rec_w;    /* words per record  = 1 + MS*sess_w */
 };
 
 struct MergeArgs {
@@ -435,10 +440,8 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     int64_t slot = key_slot(S.keys, S.C, key, err, SERR_TABLE_FULL);
     if (slot < 0) return;
     int sw = 2 * M.agg.n_aggs;
-    int64_t *ss = S.s_start + (size_t)slot * S.MS;
-    int64_t *se = S.s_end + (size_t)slot * S.MS;
-    uint64_t *sst = S.s_st + (size_t)slot * S.MS * sw;
-    uint32_t n = S.ns[slot];
+    int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+    uint32_t n = (uint32_t)rec[0];
     /* absorb every stored session whose gap-closure touches the partial's */
     int64_t cs = pmin, ce = pmax;
     uint64_t acc[AMD_MAX_AGGS * 2];
@@ -451,28 +454,28 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     }
     uint32_t w = 0;
     for (uint32_t i = 0; i < n; i++) {
-        if (cs < se[i] + (int64_t)M.gap && ss[i] < ce + (int64_t)M.gap) {
-            if (ss[i] < cs) cs = ss[i];
-            if (se[i] > ce) ce = se[i];
+        int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+        const uint64_t *sist = (const uint64_t *)(si + 2);
+        if (cs < si[1] + (int64_t)M.gap && si[0] < ce + (int64_t)M.gap) {
+            if (si[0] < cs) cs = si[0];
+            if (si[1] > ce) ce = si[1];
             if (M.cd_agg >= 0) {
                 uint64_t *aw = acc + 2 * M.cd_agg;
-                const uint64_t *sv = sst + (size_t)i * sw + 2 * M.cd_agg;
+                const uint64_t *sv = sist + 2 * M.cd_agg;
                 cd_merge_into(M.cd, aw, sv[0], sv[1], err);
                 /* scalar merge must skip the CD words: zero them on the
                  * source copy path by merging around */
                 uint64_t saved0 = aw[0], saved1 = aw[1];
-                state_merge(acc, sst + (size_t)i * sw, M.agg);
+                state_merge(acc, sist, M.agg);
                 aw[0] = saved0;
                 aw[1] = saved1;
             } else {
-                state_merge(acc, sst + (size_t)i * sw, M.agg);
+                state_merge(acc, sist, M.agg);
             }
         } else {
             if (w != i) {
-                ss[w] = ss[i];
-                se[w] = se[i];
-                for (int k = 0; k < sw; k++)
-                    sst[(size_t)w * sw + k] = sst[(size_t)i * sw + k];
+                int64_t *sd = rec + 1 + (size_t)w * S.sess_w;
+                for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
             }
             w++;
         }
@@ -490,10 +493,11 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
              j = M.bv_next[j])
             aw[0] += cd_insert(M.cd, (int64_t)aw[1] - 1, M.bv_val[j], err);
     }
-    ss[w] = cs;
-    se[w] = ce;
-    for (int k = 0; k < sw; k++) sst[(size_t)w * sw + k] = acc[k];
-    S.ns[slot] = w + 1;
+    int64_t *wr = rec + 1 + (size_t)w * S.sess_w;
+    wr[0] = cs;
+    wr[1] = ce;
+    for (int k = 0; k < sw; k++) ((uint64_t *)(wr + 2))[k] = acc[k];
+    rec[0] = w + 1;
 }
 
 __global__ void __launch_bounds__(256)
@@ -528,24 +532,24 @@ k_sess_fire(FireArgs F) {
     int sw = 2 * F.agg.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        uint32_t n = S.ns[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+        uint32_t n = (uint32_t)rec[0];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
-        int64_t *ss = S.s_start + (size_t)slot * S.MS;
-        int64_t *se = S.s_end + (size_t)slot * S.MS;
-        uint64_t *sst = S.s_st + (size_t)slot * S.MS * sw;
         uint32_t w = 0;
         for (uint32_t i = 0; i < n; i++) {
-            uint64_t close = (uint64_t)(se[i] + (int64_t)F.gap);
+            int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+            const uint64_t *sist = (const uint64_t *)(si + 2);
+            uint64_t close = (uint64_t)(si[1] + (int64_t)F.gap);
             if (close < F.wm) {
                 int64_t r = (int64_t)atomicAdd(F.n_out, 1ULL);
                 if (r >= F.out_cap) { *F.err = SERR_OUT_CAP; continue; }
                 int col = 0;
                 if (F.n_keys) F.out[col++][r] = key;
                 for (int a = 0; a < F.agg.n_aggs; a++, col++) {
-                    uint64_t w0 = sst[(size_t)i * sw + 2 * a];
-                    uint64_t w1 = sst[(size_t)i * sw + 2 * a + 1];
+                    uint64_t w0 = sist[2 * a];
+                    uint64_t w1 = sist[2 * a + 1];
                     int64_t v;
                     switch (F.agg.op[a]) {
                     case AMD_AGG_COUNT:
@@ -563,20 +567,18 @@ k_sess_fire(FireArgs F) {
                     }
                     F.out[col][r] = v;
                 }
-                F.out[col++][r] = ss[i];
+                F.out[col++][r] = si[0];
                 F.out[col++][r] = (int64_t)close;
                 F.out[col][r] = (int64_t)close - 1;
             } else {
                 if (w != i) {
-                    ss[w] = ss[i];
-                    se[w] = se[i];
-                    for (int k = 0; k < sw; k++)
-                        sst[(size_t)w * sw + k] = sst[(size_t)i * sw + k];
+                    int64_t *sd = rec + 1 + (size_t)w * S.sess_w;
+                    for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
                 }
                 w++;
             }
         }
-        S.ns[slot] = w;
+        rec[0] = w;
     }
 }
 
@@ -597,22 +599,22 @@ k_sess_drain(DrainArgs D) {
     int sw = 2 * D.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        uint32_t n = S.ns[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+        uint32_t n = (uint32_t)rec[0];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
         int64_t base = (int64_t)atomicAdd(D.n_out, (unsigned long long)n);
         if (base + n > D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
         for (uint32_t i = 0; i < n; i++) {
+            const int64_t *si = rec + 1 + (size_t)i * S.sess_w;
             int64_t r = base + i;
             int col = 0;
             if (D.n_keys) D.out[col++][r] = key;
             for (int k = 0; k < sw; k++)
-                D.out[col++][r] =
-                    (int64_t)S.s_st[(size_t)slot * S.MS * sw +
-                                    (size_t)i * sw + k];
-            D.out[col++][r] = S.s_start[(size_t)slot * S.MS + i];
-            D.out[col][r] = S.s_end[(size_t)slot * S.MS + i];
+                D.out[col++][r] = si[2 + k];
+            D.out[col++][r] = si[0];
+            D.out[col][r] = si[1];
         }
     }
 }
@@ -678,14 +680,14 @@ k_sess_drain_values(CdDrainArgs D) {
     int sw = 2 * D.agg.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        uint32_t n = S.ns[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+        uint32_t n = (uint32_t)rec[0];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
         for (uint32_t i = 0; i < n; i++) {
-            uint64_t reg1 =
-                S.s_st[(size_t)slot * S.MS * sw + (size_t)i * sw +
-                       2 * D.cd_agg + 1];
+            const int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+            uint64_t reg1 = (uint64_t)si[2 + 2 * D.cd_agg + 1];
             if (reg1 == 0) continue;
             const uint64_t *bm =
                 D.cd.regions + (size_t)(reg1 - 1) * cd_region_words(D.cd.D);
@@ -696,7 +698,7 @@ k_sess_drain_values(CdDrainArgs D) {
                 if (r >= D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
                 int col = 0;
                 if (D.n_keys) D.out[col++][r] = key;
-                D.out[col++][r] = S.s_start[(size_t)slot * S.MS + i];
+                D.out[col++][r] = si[0];
                 D.out[col][r] = vals[sl];
             }
         }
@@ -728,12 +730,13 @@ k_sess_restore_values(CdRestoreArgs R) {
             int64_t slot = key_slot(S.keys, S.C, key, R.err,
                                     SERR_TABLE_FULL);
             if (slot < 0) continue;
-            uint32_t n = S.ns[slot];
+            int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+            uint32_t n = (uint32_t)rec[0];
             for (uint32_t si = 0; si < n; si++) {
-                if (S.s_start[(size_t)slot * S.MS + si] != R.start[i])
+                int64_t *sp = rec + 1 + (size_t)si * S.sess_w;
+                if (sp[0] != R.start[i])
                     continue;
-                uint64_t *w = &S.s_st[(size_t)slot * S.MS * sw +
-                                      (size_t)si * sw + 2 * R.cd_agg];
+                uint64_t *w = (uint64_t *)(sp + 2) + 2 * R.cd_agg;
                 if (w[1] == 0) {
                     int64_t r = cd_alloc(R.cd, R.err);
                     if (r < 0) break;
@@ -840,11 +843,10 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
         return fail(#p, e);
     size_t C1 = (size_t)o->store.C + 1;
     size_t MS = o->store.MS, sw = 2 * (size_t)cfg->n_aggs;
+    o->store.sess_w = (uint32_t)(2 + sw);
+    o->store.rec_w = (uint32_t)(1 + MS * (2 + sw));
     SALLOC(o->store.keys, C1 * 8);
-    SALLOC(o->store.ns, C1 * 4);
-    SALLOC(o->store.s_start, C1 * MS * 8);
-    SALLOC(o->store.s_end, C1 * MS * 8);
-    SALLOC(o->store.s_st, C1 * MS * sw * 8);
+    SALLOC(o->store.recs, C1 * o->store.rec_w * 8);
     SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
     SALLOC(o->bst, ((size_t)o->B + 1) * (2 + sw) * 8);
     int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
@@ -869,7 +871,7 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     }
 #undef SALLOC
     hipMemset(o->store.keys, 0xFF, C1 * 8);
-    hipMemset(o->store.ns, 0, C1 * 4);
+    hipMemset(o->store.recs, 0, C1 * o->store.rec_w * 8);
     hipMemset(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8);
     hipMemset(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8);
     hipMemset(o->d_err, 0, 4);
@@ -1332,10 +1334,7 @@ API void arroyo_amd_session_destroy(void *h) {
     if (!o) return;
     hipStreamSynchronize(o->stream);
     hipFree(o->store.keys);
-    hipFree(o->store.ns);
-    hipFree(o->store.s_start);
-    hipFree(o->store.s_end);
-    hipFree(o->store.s_st);
+    hipFree(o->store.recs);
     hipFree(o->bkeys);
     hipFree(o->bst);
     int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
