@@ -844,7 +844,9 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     size_t C1 = (size_t)o->store.C + 1;
     size_t MS = o->store.MS, sw = 2 * (size_t)cfg->n_aggs;
     o->store.sess_w = (uint32_t)(2 + sw);
-    o->store.rec_w = (uint32_t)(1 + MS * (2 + sw));
+    /* pad records to a 64B multiple so ns + the first session never
+     * straddle a line */
+    o->store.rec_w = (uint32_t)((1 + MS * (2 + sw) + 7) & ~7u);
     SALLOC(o->store.keys, C1 * 8);
     SALLOC(o->store.recs, C1 * o->store.rec_w * 8);
     SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
