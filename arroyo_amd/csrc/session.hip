@@ -403,16 +403,18 @@ k_sess_update(UpdateArgs A) {
 
 struct Store {
     int64_t *keys;     /* [C+1]; slot C = spec (key == EMPTY_KEY) */
-    /* per-slot interleaved record: [ns, (start, end, st[sw]) x MS].  One
-     * 64B line covers ns + the whole first session for sw <= 4, so the
-     * phase-2 merge touches ~2 random lines per key (key probe + record)
-     * instead of the 5 the split ns/s_start/s_end/s_st planes cost —
-     * round 1 measured the store random walk as the residual bottleneck
-     * (profiles/r01_session_note.md). */
+    /* Occupancy stays a DENSE u32 plane (the watermark fire scans every
+     * slot: strided reads through the records cost ~1 GB of line traffic
+     * per fire, measured 765 us — the dense plane scans at full
+     * bandwidth).  Live sessions sit in a per-slot interleaved record
+     * [(start, end, st[sw]) x MS] so the phase-2 merge touches ~2 random
+     * lines per key (key probe + record) instead of the 5 the split
+     * planes cost (profiles/r01_session_note.md). */
+    uint32_t *ns;      /* [C+1] live sessions per key (dense) */
     int64_t *recs;     /* [(C+1)][rec_w] */
     uint32_t C, MS;
     uint32_t sess_w;   /* words per session = 2 + 2*n_aggs */
-    uint32_t rec_w;    /* words per record  = 1 + MS*sess_w */
+    uint32_t rec_w;    /* words per record  = MS*sess_w padded to 64B */
 };
 
 struct MergeArgs {
@@ -441,7 +443,7 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     if (slot < 0) return;
     int sw = 2 * M.agg.n_aggs;
     int64_t *rec = S.recs + (size_t)slot * S.rec_w;
-    uint32_t n = (uint32_t)rec[0];
+    uint32_t n = S.ns[slot];
     /* absorb every stored session whose gap-closure touches the partial's */
     int64_t cs = pmin, ce = pmax;
     uint64_t acc[AMD_MAX_AGGS * 2];
@@ -454,7 +456,7 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     }
     uint32_t w = 0;
     for (uint32_t i = 0; i < n; i++) {
-        int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+        int64_t *si = rec + (size_t)i * S.sess_w;
         const uint64_t *sist = (const uint64_t *)(si + 2);
         if (cs < si[1] + (int64_t)M.gap && si[0] < ce + (int64_t)M.gap) {
             if (si[0] < cs) cs = si[0];
@@ -474,7 +476,7 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
             }
         } else {
             if (w != i) {
-                int64_t *sd = rec + 1 + (size_t)w * S.sess_w;
+                int64_t *sd = rec + (size_t)w * S.sess_w;
                 for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
             }
             w++;
@@ -493,11 +495,11 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
              j = M.bv_next[j])
             aw[0] += cd_insert(M.cd, (int64_t)aw[1] - 1, M.bv_val[j], err);
     }
-    int64_t *wr = rec + 1 + (size_t)w * S.sess_w;
+    int64_t *wr = rec + (size_t)w * S.sess_w;
     wr[0] = cs;
     wr[1] = ce;
     for (int k = 0; k < sw; k++) ((uint64_t *)(wr + 2))[k] = acc[k];
-    rec[0] = w + 1;
+    S.ns[slot] = w + 1;
 }
 
 __global__ void __launch_bounds__(256)
@@ -532,14 +534,14 @@ k_sess_fire(FireArgs F) {
     int sw = 2 * F.agg.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
-        uint32_t n = (uint32_t)rec[0];
+        uint32_t n = S.ns[slot];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         uint32_t w = 0;
         for (uint32_t i = 0; i < n; i++) {
-            int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+            int64_t *si = rec + (size_t)i * S.sess_w;
             const uint64_t *sist = (const uint64_t *)(si + 2);
             uint64_t close = (uint64_t)(si[1] + (int64_t)F.gap);
             if (close < F.wm) {
@@ -572,13 +574,13 @@ k_sess_fire(FireArgs F) {
                 F.out[col][r] = (int64_t)close - 1;
             } else {
                 if (w != i) {
-                    int64_t *sd = rec + 1 + (size_t)w * S.sess_w;
+                    int64_t *sd = rec + (size_t)w * S.sess_w;
                     for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
                 }
                 w++;
             }
         }
-        rec[0] = w;
+        if (w != n) S.ns[slot] = w;
     }
 }
 
@@ -599,15 +601,15 @@ k_sess_drain(DrainArgs D) {
     int sw = 2 * D.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
-        uint32_t n = (uint32_t)rec[0];
+        uint32_t n = S.ns[slot];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        const int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         int64_t base = (int64_t)atomicAdd(D.n_out, (unsigned long long)n);
         if (base + n > D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
         for (uint32_t i = 0; i < n; i++) {
-            const int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+            const int64_t *si = rec + (size_t)i * S.sess_w;
             int64_t r = base + i;
             int col = 0;
             if (D.n_keys) D.out[col++][r] = key;
@@ -680,13 +682,13 @@ k_sess_drain_values(CdDrainArgs D) {
     int sw = 2 * D.agg.n_aggs;
     for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          slot <= (int64_t)S.C; slot += stride) {
-        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
-        uint32_t n = (uint32_t)rec[0];
+        uint32_t n = S.ns[slot];
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        const int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         for (uint32_t i = 0; i < n; i++) {
-            const int64_t *si = rec + 1 + (size_t)i * S.sess_w;
+            const int64_t *si = rec + (size_t)i * S.sess_w;
             uint64_t reg1 = (uint64_t)si[2 + 2 * D.cd_agg + 1];
             if (reg1 == 0) continue;
             const uint64_t *bm =
@@ -731,9 +733,9 @@ k_sess_restore_values(CdRestoreArgs R) {
                                     SERR_TABLE_FULL);
             if (slot < 0) continue;
             int64_t *rec = S.recs + (size_t)slot * S.rec_w;
-            uint32_t n = (uint32_t)rec[0];
+            uint32_t n = S.ns[slot];
             for (uint32_t si = 0; si < n; si++) {
-                int64_t *sp = rec + 1 + (size_t)si * S.sess_w;
+                int64_t *sp = rec + (size_t)si * S.sess_w;
                 if (sp[0] != R.start[i])
                     continue;
                 uint64_t *w = (uint64_t *)(sp + 2) + 2 * R.cd_agg;
@@ -844,10 +846,11 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     size_t C1 = (size_t)o->store.C + 1;
     size_t MS = o->store.MS, sw = 2 * (size_t)cfg->n_aggs;
     o->store.sess_w = (uint32_t)(2 + sw);
-    /* pad records to a 64B multiple so ns + the first session never
-     * straddle a line */
-    o->store.rec_w = (uint32_t)((1 + MS * (2 + sw) + 7) & ~7u);
+    /* pad records to a 64B multiple so the first session never straddles
+     * a line */
+    o->store.rec_w = (uint32_t)((MS * (2 + sw) + 7) & ~7u);
     SALLOC(o->store.keys, C1 * 8);
+    SALLOC(o->store.ns, C1 * 4);
     SALLOC(o->store.recs, C1 * o->store.rec_w * 8);
     SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
     SALLOC(o->bst, ((size_t)o->B + 1) * (2 + sw) * 8);
@@ -873,6 +876,7 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     }
 #undef SALLOC
     hipMemset(o->store.keys, 0xFF, C1 * 8);
+    hipMemset(o->store.ns, 0, C1 * 4);
     hipMemset(o->store.recs, 0, C1 * o->store.rec_w * 8);
     hipMemset(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8);
     hipMemset(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8);
@@ -1336,6 +1340,7 @@ API void arroyo_amd_session_destroy(void *h) {
     if (!o) return;
     hipStreamSynchronize(o->stream);
     hipFree(o->store.keys);
+    hipFree(o->store.ns);
     hipFree(o->store.recs);
     hipFree(o->bkeys);
     hipFree(o->bst);
