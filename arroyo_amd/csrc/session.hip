@@ -400,6 +400,141 @@ k_sess_update(UpdateArgs A) {
     }
 }
 
+/* quad-batched phase-1 kernel (keyed, no COUNT DISTINCT): the same
+ * latency treatment the window path's k_update_batch measured out — four
+ * rows per thread, sequential same-key runs collapsed, a small LDS hot
+ * cache for the Zipf head, and the remaining rows' first probe loads
+ * issued together before any is resolved.  Every state update is a
+ * no-return atomic (atomicMax / atomicAdd), so the probe is the only
+ * latency chain. */
+__device__ inline int64_t sess_probe_resolve(int64_t *keys, uint32_t B,
+                                             int *err, int64_t key,
+                                             uint64_t j, int64_t kk) {
+    uint64_t m = B - 1;
+    for (uint32_t probes = 0; probes < B; probes++) {
+        if (kk == key) return (int64_t)j;
+        if (kk == EMPTY_KEY) {
+            int64_t old = (int64_t)atomicCAS((unsigned long long *)&keys[j],
+                                             (unsigned long long)EMPTY_KEY,
+                                             (unsigned long long)key);
+            if (old == EMPTY_KEY || old == key) return (int64_t)j;
+        }
+        j = (j + 1) & m;
+        kk = keys[j];
+    }
+    *err = SERR_BATCH_FULL;
+    return -1;
+}
+
+__global__ void __launch_bounds__(256)
+k_sess_update_batch(UpdateArgs A) {
+    __shared__ int64_t lkey[SESS_LDS_SLOTS];
+    extern __shared__ uint64_t lst[];    /* [SLOTS][2 + n_aggs*2] */
+    int sw = 2 + 2 * A.agg.n_aggs;
+    for (int i = threadIdx.x; i < SESS_LDS_SLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        for (int w = 0; w < sw; w++) lst[(size_t)i * sw + w] = 0;
+    }
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t *ts = A.cols[A.n_keys + A.n_vals];
+    const int64_t *vcols[AMD_MAX_AGGS];
+    for (int i = 0; i < A.agg.n_aggs; i++)
+        vcols[i] = A.agg.col[i] >= 0 ? A.cols[A.n_keys + A.agg.col[i]]
+                                     : nullptr;
+    const uint64_t m = A.B - 1;
+    int64_t nq = A.n_rows / 4;
+    for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nq;
+         v += stride) {
+        int64_t key[4];
+        uint64_t traw[4];
+        for (int j = 0; j < 2; j++) {
+            ulonglong2 tp = ((const ulonglong2 *)ts)[2 * v + j];
+            ulonglong2 kp = ((const ulonglong2 *)A.cols[0])[2 * v + j];
+            traw[2 * j] = tp.x;
+            traw[2 * j + 1] = tp.y;
+            key[2 * j] = (int64_t)kp.x;
+            key[2 * j + 1] = (int64_t)kp.y;
+        }
+        bool need[4];
+        for (int j = 0; j < 4; j++) {
+            traw[j] += A.ts_offset;
+            need[j] = !(A.has_wm && traw[j] < A.wm) && key[j] != EMPTY_KEY;
+            /* sentinel-valued key (-1): rare, straight to the global
+             * table's spec slot */
+            if (!(A.has_wm && traw[j] < A.wm) && key[j] == EMPTY_KEY) {
+                uint64_t *st = A.bst + (size_t)A.B * sw;
+                atomicMax((unsigned long long *)&st[0],
+                          (unsigned long long)enc_min((int64_t)traw[j]));
+                atomicMax((unsigned long long *)&st[1],
+                          (unsigned long long)enc_max((int64_t)traw[j]));
+                atomic_fold(st + 2, A.agg, vcols, 4 * v + j);
+            }
+        }
+        /* LDS hot cache (2-probe) */
+        for (int j = 0; j < 4; j++) {
+            if (!need[j]) continue;
+            uint32_t h = (uint32_t)hash64((uint64_t)key[j] * 0x9e37u) &
+                         (SESS_LDS_SLOTS - 1);
+            for (int pr = 0; pr < 2; pr++) {
+                uint32_t sl = (h + pr) & (SESS_LDS_SLOTS - 1);
+                int64_t k = lkey[sl];
+                if (k == EMPTY_KEY) {
+                    int64_t old = (int64_t)atomicCAS(
+                        (unsigned long long *)&lkey[sl],
+                        (unsigned long long)EMPTY_KEY,
+                        (unsigned long long)key[j]);
+                    k = old == EMPTY_KEY ? key[j] : old;
+                }
+                if (k == key[j]) {
+                    uint64_t *st = lst + (size_t)sl * sw;
+                    atomicMax((unsigned long long *)&st[0],
+                              (unsigned long long)enc_min((int64_t)traw[j]));
+                    atomicMax((unsigned long long *)&st[1],
+                              (unsigned long long)enc_max((int64_t)traw[j]));
+                    atomic_fold(st + 2, A.agg, vcols, 4 * v + j);
+                    need[j] = false;
+                    break;
+                }
+            }
+        }
+        /* issue the remaining rows' first probe loads together */
+        uint64_t h[4];
+        int64_t firstk[4];
+        for (int j = 0; j < 4; j++)
+            if (need[j]) {
+                h[j] = hash64((uint64_t)key[j]) & m;
+                firstk[j] = A.bkeys[h[j]];
+            }
+        for (int j = 0; j < 4; j++) {
+            if (!need[j]) continue;
+            int64_t s = sess_probe_resolve(A.bkeys, A.B, A.err, key[j],
+                                           h[j], firstk[j]);
+            if (s < 0) continue;
+            uint64_t *st = A.bst + (size_t)s * sw;
+            atomicMax((unsigned long long *)&st[0],
+                      (unsigned long long)enc_min((int64_t)traw[j]));
+            atomicMax((unsigned long long *)&st[1],
+                      (unsigned long long)enc_max((int64_t)traw[j]));
+            atomic_fold(st + 2, A.agg, vcols, 4 * v + j);
+        }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < SESS_LDS_SLOTS; i += blockDim.x) {
+        int64_t key = lkey[i];
+        if (key == EMPTY_KEY) continue;
+        int64_t s = key_slot(A.bkeys, A.B, key, A.err, SERR_BATCH_FULL);
+        if (s < 0) continue;
+        uint64_t *dst = A.bst + (size_t)s * sw;
+        const uint64_t *src = lst + (size_t)i * sw;
+        atomicMax((unsigned long long *)&dst[0],
+                  (unsigned long long)src[0]);
+        atomicMax((unsigned long long *)&dst[1],
+                  (unsigned long long)src[1]);
+        state_merge_atomic(dst + 2, src + 2, A.agg);
+    }
+}
+
 
 struct Store {
     int64_t *keys;     /* [C+1]; slot C = spec (key == EMPTY_KEY) */
@@ -960,17 +1095,41 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     }
     A.err = o->d_err;
     size_t shmem = (size_t)SESS_LDS_SLOTS * (2 + 2 * o->cfg.n_aggs) * 8;
-    /* cap the grid: at one row per thread the per-block LDS flush (up to
-     * 512 upserts) outweighs the rows themselves; fewer, fuller blocks
-     * amortize it and raise the hot-key LDS hit rate (the window path
-     * measured the same effect) */
     int ublocks = grid_for(n_rows);
-    if (ublocks > 96) ublocks = 96;
     if (const char *ev = getenv("ARROYO_AMD_SESS_BLOCKS"))
         if (atoi(ev) > 0) ublocks = atoi(ev);
-    hipLaunchKernelGGL(k_sess_update, dim3(ublocks), dim3(256),
-                       shmem, o->stream, A);
-    SHIP(o, hipGetLastError());
+    /* quad-batched path: keyed, no COUNT DISTINCT, 16B-aligned columns */
+    bool batch = o->cfg.n_keys == 1 && o->cd_agg < 0 && n_rows >= 4;
+    for (int c = 0; c < o->n_in_cols && batch; c++)
+        batch = ((uintptr_t)dcols[c] & 15) == 0;
+    if (const char *ev = getenv("ARROYO_AMD_SESS_BATCH"))
+        if (!atoi(ev)) batch = false;
+    if (batch) {
+        int64_t main_rows = n_rows & ~3ll;
+        UpdateArgs B4 = A;
+        B4.n_rows = main_rows;
+        int64_t want = (main_rows / 4 + 255) / 256;
+        int qb = (int)(want > 2048 ? 2048 : (want < 1 ? 1 : want));
+        if (const char *ev = getenv("ARROYO_AMD_SESS_BLOCKS"))
+            if (atoi(ev) > 0) qb = atoi(ev);
+        hipLaunchKernelGGL(k_sess_update_batch, dim3(qb), dim3(256), shmem,
+                           o->stream, B4);
+        SHIP(o, hipGetLastError());
+        int64_t tail = n_rows - main_rows;
+        if (tail) {
+            UpdateArgs T = A;
+            for (int c = 0; c < o->n_in_cols; c++)
+                T.cols[c] = dcols[c] + main_rows;
+            T.n_rows = tail;
+            hipLaunchKernelGGL(k_sess_update, dim3(1), dim3(64), shmem,
+                               o->stream, T);
+            SHIP(o, hipGetLastError());
+        }
+    } else {
+        hipLaunchKernelGGL(k_sess_update, dim3(ublocks), dim3(256),
+                           shmem, o->stream, A);
+        SHIP(o, hipGetLastError());
+    }
     MergeArgs M = {};
     M.bkeys = o->bkeys;
     M.bst = o->bst;
