@@ -553,9 +553,10 @@ struct Store {
 };
 
 struct MergeArgs {
-    /* batch pre-agg table */
-    const int64_t *bkeys;
-    const uint64_t *bst;
+    /* batch pre-agg table (cleared inline by k_sess_merge) */
+    int64_t *bkeys;
+    uint64_t *bst;
+    unsigned long long *bv_cur;   /* reset with the table (may be null) */
     uint32_t B;
     Store store;
     uint64_t gap;
@@ -643,12 +644,18 @@ k_sess_merge(MergeArgs M) {
     int sw = 2 + 2 * M.agg.n_aggs;
     for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          s <= (int64_t)M.B; s += stride) {
-        const uint64_t *st = M.bst + (size_t)s * sw;
+        uint64_t *st = M.bst + (size_t)s * sw;
         if (st[0] == 0 && st[1] == 0) continue;      /* untouched slot */
         int64_t key = s == (int64_t)M.B ? EMPTY_KEY : M.bkeys[s];
         if (s != (int64_t)M.B && key == EMPTY_KEY) continue;
         merge_partial(M, key, dec_min(st[0]), dec_max(st[1]), st + 2, M.err);
+        /* clear the slot inline (the lines are already dirty): the next
+         * batch starts from an empty table without the two 5 MB fill
+         * launches per batch this scan used to require */
+        for (int w = 0; w < sw; w++) st[w] = 0;
+        if (s != (int64_t)M.B) M.bkeys[s] = EMPTY_KEY;
     }
+    if (M.bv_cur && blockIdx.x == 0 && threadIdx.x == 0) *M.bv_cur = 0;
 }
 
 struct FireArgs {
@@ -1068,11 +1075,7 @@ static int grid_for(int64_t want_threads) {
 static int sess_submit(GpuSession *o, const int64_t *const *dcols,
                        int64_t n_rows, uint64_t ts_offset) {
     if (n_rows == 0) return 0;
-    size_t sw = 2 * (size_t)o->cfg.n_aggs;
-    SHIP(o, hipMemsetAsync(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8,
-                           o->stream));
-    SHIP(o, hipMemsetAsync(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8,
-                           o->stream));
+    /* no per-batch clears: k_sess_merge leaves the table empty behind it */
     UpdateArgs A = {};
     for (int c = 0; c < o->n_in_cols; c++) A.cols[c] = dcols[c];
     A.n_keys = o->cfg.n_keys;
@@ -1087,7 +1090,6 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     A.agg = o->agg;
     A.cd_agg = o->cd_agg;
     if (o->cd_agg >= 0) {
-        SHIP(o, hipMemsetAsync(o->bv_cur, 0, 8, o->stream));
         A.bv_val = o->bv_val;
         A.bv_next = o->bv_next;
         A.bv_cur = o->bv_cur;
@@ -1141,6 +1143,7 @@ static int sess_submit(GpuSession *o, const int64_t *const *dcols,
     M.cd = o->cd;
     M.bv_val = o->bv_val;
     M.bv_next = o->bv_next;
+    M.bv_cur = o->bv_cur;
     M.err = o->d_err;
     hipLaunchKernelGGL(k_sess_merge, dim3(grid_for((int64_t)o->B + 1)),
                        dim3(256), 0, o->stream, M);
