@@ -1881,7 +1881,71 @@ struct MergeFusedArgs {
     unsigned long long *accum;  /* running emitted-rows accumulator (may be
                                    null); folded in here so the separate
                                    k_accum launch (~5 us) is not needed */
+    /* closed-pane index (CPI): each closed pane compacted ONCE into
+     * home-range-grouped dense {key, state} entries, so the five fires
+     * that read it scan only occupied entries instead of the whole
+     * sparse table + displacement overscan (the fire path was 52% of
+     * GPU time on the sparse scan).  use_cpi covers n_src <= 16. */
+    int32_t use_cpi;
+    const uint64_t *cpi_entries[16];   /* [n_entries][1 + 2*na] */
+    const uint32_t *cpi_off[16];       /* [C/range + 1] */
 };
+
+/* CPI build: group a closed pane's occupied slots by HOME range
+ * (hash(key) & mask) / range — the ownership the fused merge partitions
+ * by — with a count/scan/scatter pass.  Runs once per pane close
+ * (control-rate). */
+struct CpiBuildArgs {
+    const int64_t *keys;      /* pane planes in the ring */
+    const uint64_t *state;
+    uint32_t C;
+    int32_t na;
+    uint32_t range;
+    uint32_t *off;            /* [NR+1]; zeroed before count */
+    uint32_t *cur;            /* [NR] scatter cursors */
+    uint64_t *entries;        /* [C][1 + 2*na] */
+};
+
+__global__ void __launch_bounds__(256)
+k_cpi_count(CpiBuildArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint32_t mask = A.C - 1;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)A.C; i += stride) {
+        int64_t key = A.keys[i];
+        if (key == EMPTY_KEY) continue;
+        uint32_t home = (uint32_t)hash64((uint64_t)key) & mask;
+        atomicAdd(&A.off[home / A.range + 1], 1u);
+    }
+}
+
+__global__ void k_cpi_scan(uint32_t *off, uint32_t *cur, uint32_t nr) {
+    /* single thread: NR+1 is ~2K entries, control-rate */
+    uint32_t acc = 0;
+    for (uint32_t r = 0; r <= nr; r++) {
+        acc += off[r];
+        off[r] = acc;
+        if (r < nr) cur[r] = acc;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_cpi_scatter(CpiBuildArgs A) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    uint32_t mask = A.C - 1;
+    int ew = 1 + 2 * A.na;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < (int64_t)A.C; i += stride) {
+        int64_t key = A.keys[i];
+        if (key == EMPTY_KEY) continue;
+        uint32_t home = (uint32_t)hash64((uint64_t)key) & mask;
+        uint32_t pos = atomicAdd(&A.cur[home / A.range], 1u);
+        uint64_t *e = A.entries + (size_t)pos * ew;
+        e[0] = (uint64_t)key;
+        for (int w = 0; w < 2 * A.na; w++)
+            e[1 + w] = A.state[(size_t)i * A.na * 2 + w];
+    }
+}
 
 template <int SLOTS, bool PACKED = false>
 __global__ void __launch_bounds__(256)
@@ -1907,10 +1971,13 @@ k_merge_fused(MergeFusedArgs M) {
             PACKED ? nullptr : M.ring.state + (size_t)M.src[p] * C * na * 2;
         const uint64_t *pslots =
             PACKED ? M.ring.slots + (size_t)M.src[p] * C * 2 : nullptr;
-        auto fold = [&](uint32_t idx, int64_t key, uint64_t pw0) {
+        auto fold = [&](int64_t key, const uint64_t *stv, uint64_t pw0,
+                        bool own) {
             if (key == EMPTY_KEY) return;
-            uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
-            if (rel >= M.range) return;          /* another WG owns it */
+            if (!own) {
+                uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
+                if (rel >= M.range) return;      /* another WG owns it */
+            }
             /* LDS upsert */
             uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u) &
                          (SLOTS - 1);
@@ -1935,13 +2002,24 @@ k_merge_fused(MergeFusedArgs M) {
                           (unsigned long long)pw0);
                 return;
             }
-            atomic_merge(d, st + (size_t)idx * na * 2, M.agg);
+            atomic_merge(d, stv, M.agg);
         };
-        if (PACKED) {
+        if (!PACKED && M.use_cpi) {
+            /* dense home-range-grouped entries: only occupied slots read,
+             * ownership established at build time */
+            const uint32_t *off = M.cpi_off[p];
+            const uint64_t *ent = M.cpi_entries[p];
+            const int ew = 1 + 2 * na;
+            uint32_t lo = off[blockIdx.x], hi = off[blockIdx.x + 1];
+            for (uint32_t t = lo + threadIdx.x; t < hi; t += blockDim.x) {
+                const uint64_t *e = ent + (size_t)t * ew;
+                fold((int64_t)e[0], e + 1, 0, true);
+            }
+        } else if (PACKED) {
             for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
                 uint32_t idx = (a + t) & mask;
                 ulonglong2 sv = ((const ulonglong2 *)pslots)[idx];
-                fold(idx, (int64_t)sv.x, sv.y);
+                fold((int64_t)sv.x, nullptr, sv.y, false);
             }
         } else {
             /* 2 keys per 16 B load: a and span are multiples of 2, and an
@@ -1950,8 +2028,9 @@ k_merge_fused(MergeFusedArgs M) {
                  t += 2 * blockDim.x) {
                 uint32_t idx = (a + t) & mask;
                 ulonglong2 kv = *(const ulonglong2 *)&keys[idx];
-                fold(idx, (int64_t)kv.x, 0);
-                fold(idx + 1, (int64_t)kv.y, 0);
+                fold((int64_t)kv.x, st + (size_t)idx * na * 2, 0, false);
+                fold((int64_t)kv.y, st + ((size_t)idx + 1) * na * 2, 0,
+                     false);
             }
         }
     }
@@ -2246,6 +2325,13 @@ struct GpuOp {
     int use_lds;
     int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
                        3 batched-probe, 4 radix-regroup */
+    /* closed-pane index (see CpiBuildArgs) */
+    uint64_t *cpi_entries;      /* [R][C][1+2na] */
+    uint32_t *cpi_off;          /* [R][NR+1] */
+    uint32_t *cpi_cur;          /* [NR] scratch */
+    uint32_t mf_range;          /* fused-merge home-range size (fixed) */
+    uint32_t cpi_nr;            /* C / mf_range */
+    std::vector<char> cpi_ready;
     /* multi-column keys: device dictionary (see k_dict_encode) */
     int mk;                     /* 1 when cfg.n_keys >= 2 */
     int64_t *d_dict_digest;
@@ -2302,6 +2388,7 @@ static thread_local char g_err[512];
 
 static int ring_retire(GpuOp *o, uint32_t slot) {
     size_t na = o->agg.n_aggs;
+    if (slot < o->cpi_ready.size()) o->cpi_ready[slot] = 0;
     int blocks = (int)((o->ring.C + 255) / 256);
     if (blocks > 1024) blocks = 1024;
     if (o->ring.packed) {
@@ -2326,6 +2413,33 @@ static int ring_retire(GpuOp *o, uint32_t slot) {
 }
 
 static int flush_staged(GpuOp *o);
+
+/* compact a freshly-closed pane into its home-range-grouped index */
+static int cpi_build(GpuOp *o, uint32_t slot) {
+    if (!o->cpi_entries) return 0;
+    size_t na = o->agg.n_aggs;
+    HIP_CHECK(o, hipMemsetAsync(o->cpi_off + (size_t)slot * (o->cpi_nr + 1),
+                                0, ((size_t)o->cpi_nr + 1) * 4, o->stream));
+    CpiBuildArgs A = {};
+    A.keys = o->ring.keys + (size_t)slot * o->ring.C;
+    A.state = o->ring.state + (size_t)slot * o->ring.C * na * 2;
+    A.C = o->ring.C;
+    A.na = (int32_t)na;
+    A.range = o->mf_range;
+    A.off = o->cpi_off + (size_t)slot * (o->cpi_nr + 1);
+    A.cur = o->cpi_cur;
+    A.entries = o->cpi_entries + (size_t)slot * o->ring.C * (1 + 2 * na);
+    int blocks = (int)((o->ring.C + 255) / 256);
+    if (blocks > 1024) blocks = 1024;
+    hipLaunchKernelGGL(k_cpi_count, dim3(blocks), dim3(256), 0, o->stream, A);
+    hipLaunchKernelGGL(k_cpi_scan, dim3(1), dim3(1), 0, o->stream,
+                       A.off, A.cur, o->cpi_nr);
+    hipLaunchKernelGGL(k_cpi_scatter, dim3(blocks), dim3(256), 0, o->stream,
+                       A);
+    HIP_CHECK(o, hipGetLastError());
+    o->cpi_ready[slot] = 1;
+    return 0;
+}
 
 API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (!cfg || cfg->n_aggs < 1 || cfg->n_aggs > AMD_MAX_AGGS ||
@@ -2452,6 +2566,25 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
+    /* fused-merge home range fixed at create (CPI grouping depends on it) */
+    o->mf_range = MF_RANGE;
+    if (const char *ev = getenv("ARROYO_AMD_MF_RANGE"))
+        o->mf_range = (uint32_t)atoi(ev);
+    if (o->mf_range < 2) o->mf_range = 2;
+    if (o->mf_range > o->ring.C) o->mf_range = o->ring.C;
+    while (o->mf_range & (o->mf_range - 1)) o->mf_range &= o->mf_range - 1;
+    o->cpi_nr = o->ring.C / o->mf_range;
+    if (!o->ring.packed && na <= MF_MAX_AGGS && !o->cfg.is_tumbling) {
+        int use_cpi = 1;
+        if (const char *ev = getenv("ARROYO_AMD_CPI")) use_cpi = atoi(ev);
+        if (use_cpi) {
+            ALLOC(o->cpi_entries,
+                  (size_t)o->ring.R * o->ring.C * (1 + 2 * na) * 8);
+            ALLOC(o->cpi_off, (size_t)o->ring.R * (o->cpi_nr + 1) * 4);
+            ALLOC(o->cpi_cur, (size_t)o->cpi_nr * 4);
+        }
+    }
+    o->cpi_ready.assign(o->ring.R, 0);
     o->mk = o->cfg.n_keys >= 2;
     if (o->mk) {
         /* dictionary 4x the pane table; ids must stay stable, so no
@@ -2614,7 +2747,10 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     }
     if (o->upd_kind == 4) {
         if (n_rows > o->rdx2_cap) {
-            hipFree(o->d_dict_digest);
+            hipFree(o->cpi_entries);
+    hipFree(o->cpi_off);
+    hipFree(o->cpi_cur);
+    hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
     hipFree(o->d_keyid_in);
@@ -3035,13 +3171,25 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             int mfs = 1024;
             if (const char *ev = getenv("ARROYO_AMD_MF_SLOTS"))
                 mfs = atoi(ev);
-            uint32_t range = MF_RANGE;
-            if (const char *ev2 = getenv("ARROYO_AMD_MF_RANGE"))
-                range = (uint32_t)atoi(ev2);
-            if (range < 2) range = 2;
-            if (range > o->ring.C) range = o->ring.C;
-            while (range & (range - 1)) range &= range - 1;  /* pow2 floor */
+            uint32_t range = o->mf_range;
             M.range = range;
+            M.use_cpi = 0;
+            if (o->cpi_entries && src.size() <= 16) {
+                int all = 1;
+                for (size_t i = 0; i < src.size(); i++)
+                    if (!o->cpi_ready[src[i]]) all = 0;
+                if (all) {
+                    M.use_cpi = 1;
+                    size_t na2 = o->agg.n_aggs;
+                    for (size_t i = 0; i < src.size(); i++) {
+                        M.cpi_entries[i] =
+                            o->cpi_entries +
+                            (size_t)src[i] * o->ring.C * (1 + 2 * na2);
+                        M.cpi_off[i] =
+                            o->cpi_off + (size_t)src[i] * (o->cpi_nr + 1);
+                    }
+                }
+            }
             size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) * na * 16;
             if (o->ring.packed) {
                 if (mfs >= 2048)
@@ -3195,6 +3343,7 @@ static int advance(GpuOp *o) {
     auto op_it = o->open.find(b);
     if (op_it != o->open.end()) {
         o->closed[b] = op_it->second;   /* pane stays in the ring */
+        if (cpi_build(o, op_it->second)) return 1;
         o->open.erase(op_it);
         o->table_bins.insert(b);
     }
@@ -3436,10 +3585,12 @@ API int arroyo_amd_restore(void *h, const int64_t *const *cols,
         HIP_CHECK(o, hipGetLastError());
         HIP_CHECK(o, hipStreamSynchronize(o->stream));
         o->table_bins.insert(bin);
-        if (has_wm && bin < wmb)
+        if (has_wm && bin < wmb) {
             o->closed[bin] = p;
-        else
+            if (cpi_build(o, p)) return 1;
+        } else {
             o->open[bin] = p;
+        }
     }
     /* state machine init (sliding:580-593) */
     if (!o->cfg.is_tumbling) {
@@ -3502,6 +3653,9 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
     hipFree(o->rdx_tmp);
+    hipFree(o->cpi_entries);
+    hipFree(o->cpi_off);
+    hipFree(o->cpi_cur);
     hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
