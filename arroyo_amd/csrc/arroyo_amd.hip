@@ -1919,13 +1919,26 @@ k_cpi_count(CpiBuildArgs A) {
     }
 }
 
-__global__ void k_cpi_scan(uint32_t *off, uint32_t *cur, uint32_t nr) {
-    /* single thread: NR+1 is ~2K entries, control-rate */
-    uint32_t acc = 0;
-    for (uint32_t r = 0; r <= nr; r++) {
-        acc += off[r];
-        off[r] = acc;
-        if (r < nr) cur[r] = acc;
+#define CPI_MAX_NR 4096
+__global__ void __launch_bounds__(256)
+k_cpi_scan(uint32_t *off, uint32_t *cur, uint32_t nr) {
+    /* one block; staged through LDS (a first cut scanned 2K words with
+     * one thread's serialized global RMWs — ~160 us per pane close) */
+    __shared__ uint32_t buf[CPI_MAX_NR + 1];
+    for (uint32_t i = threadIdx.x; i <= nr; i += blockDim.x)
+        buf[i] = off[i];
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t acc = 0;
+        for (uint32_t r = 0; r <= nr; r++) {
+            acc += buf[r];
+            buf[r] = acc;
+        }
+    }
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i <= nr; i += blockDim.x) {
+        off[i] = buf[i];
+        if (i < nr) cur[i] = buf[i];
     }
 }
 
@@ -2574,7 +2587,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (o->mf_range > o->ring.C) o->mf_range = o->ring.C;
     while (o->mf_range & (o->mf_range - 1)) o->mf_range &= o->mf_range - 1;
     o->cpi_nr = o->ring.C / o->mf_range;
-    if (!o->ring.packed && na <= MF_MAX_AGGS && !o->cfg.is_tumbling) {
+    if (!o->ring.packed && na <= MF_MAX_AGGS && !o->cfg.is_tumbling &&
+        o->cpi_nr <= CPI_MAX_NR) {
         int use_cpi = 1;
         if (const char *ev = getenv("ARROYO_AMD_CPI")) use_cpi = atoi(ev);
         if (use_cpi) {
