@@ -1887,16 +1887,18 @@ struct MergeFusedArgs {
      * sparse table + displacement overscan (the fire path was 52% of
      * GPU time on the sparse scan).  use_cpi covers n_src <= 16. */
     int32_t use_cpi;
-    const uint64_t *cpi_entries[16];   /* [n_entries][1 + 2*na] */
-    const uint32_t *cpi_off[16];       /* [C/range + 1] */
+    const uint64_t *cpi_entries[16];   /* [nr][range][1 + 2*na] */
+    const uint32_t *cpi_cnt[16];       /* [nr] entries per range */
 };
 
-/* CPI build: group a closed pane's occupied slots by HOME range
- * (hash(key) & mask) / range — the ownership the fused merge partitions
- * by — with a count/scan/scatter pass.  Runs once per pane close
- * (control-rate). */
+/* Closed-pane index build: one workgroup per home range.  Linear probing
+ * bounds a key's stored slot to [home, home + MAX_PROBES], so range r's
+ * keys all sit in slots [r*range, (r+1)*range + MAX_PROBES): one
+ * overscan sweep per PANE (instead of one per fire) collects them into a
+ * dense per-range segment.  Per-range capacity equals `range` (pane load
+ * factor is bounded well below 1 by construction; exceeding it is a
+ * loud error). */
 #define CPI_MAX_NR 4096
-#define CPI_HB 128              /* hist/scatter grid blocks */
 
 struct CpiBuildArgs {
     const int64_t *keys;      /* pane planes in the ring */
@@ -1905,60 +1907,36 @@ struct CpiBuildArgs {
     int32_t na;
     uint32_t range;
     uint32_t nr;              /* C / range */
-    uint32_t *hist;           /* [nr * CPI_HB + 1] per-(range, block) */
-    uint32_t *off;            /* [nr + 1] compacted range starts (output) */
-    uint64_t *entries;        /* [C][1 + 2*na] */
+    uint32_t *cnt;            /* [nr] entries per range (output) */
+    uint64_t *entries;        /* [nr][range][1 + 2*na] */
+    int *err;
 };
 
-/* per-(range, block) histogram via an LDS table — same-address global
- * atomics on 2K shared counters measured 34 us/close */
 __global__ void __launch_bounds__(256)
-k_cpi_count(CpiBuildArgs A) {
-    extern __shared__ uint32_t cnt[];    /* [nr] */
-    for (uint32_t i = threadIdx.x; i < A.nr; i += blockDim.x) cnt[i] = 0;
+k_cpi_build(CpiBuildArgs A) {
+    __shared__ uint32_t lcur;
+    if (threadIdx.x == 0) lcur = 0;
     __syncthreads();
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    uint32_t mask = A.C - 1;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < (int64_t)A.C; i += stride) {
-        int64_t key = A.keys[i];
+    const uint32_t mask = A.C - 1;
+    const uint32_t a = blockIdx.x * A.range;
+    const uint32_t span = A.range + MAX_PROBES;
+    const int ew = 1 + 2 * A.na;
+    uint64_t *dst = A.entries + (size_t)blockIdx.x * A.range * ew;
+    for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
+        uint32_t idx = (a + t) & mask;
+        int64_t key = A.keys[idx];
         if (key == EMPTY_KEY) continue;
-        uint32_t home = (uint32_t)hash64((uint64_t)key) & mask;
-        atomicAdd(&cnt[home / A.range], 1u);
-    }
-    __syncthreads();
-    for (uint32_t i = threadIdx.x; i < A.nr; i += blockDim.x)
-        A.hist[(size_t)i * CPI_HB + blockIdx.x] = cnt[i];
-}
-
-/* after the exclusive scan of hist, gather the per-range starts */
-__global__ void __launch_bounds__(256)
-k_cpi_offsets(const uint32_t *hist, uint32_t *off, uint32_t nr,
-              uint32_t total_idx) {
-    for (uint32_t i = threadIdx.x; i <= nr; i += blockDim.x)
-        off[i] = i < nr ? hist[(size_t)i * CPI_HB] : hist[total_idx];
-}
-
-__global__ void __launch_bounds__(256)
-k_cpi_scatter(CpiBuildArgs A) {
-    extern __shared__ uint32_t cur[];    /* [nr] */
-    for (uint32_t i = threadIdx.x; i < A.nr; i += blockDim.x)
-        cur[i] = A.hist[(size_t)i * CPI_HB + blockIdx.x];
-    __syncthreads();
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    uint32_t mask = A.C - 1;
-    int ew = 1 + 2 * A.na;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < (int64_t)A.C; i += stride) {
-        int64_t key = A.keys[i];
-        if (key == EMPTY_KEY) continue;
-        uint32_t home = (uint32_t)hash64((uint64_t)key) & mask;
-        uint32_t pos = atomicAdd(&cur[home / A.range], 1u);
-        uint64_t *e = A.entries + (size_t)pos * ew;
+        uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
+        if (rel >= A.range) continue;
+        uint32_t pos = atomicAdd(&lcur, 1u);
+        if (pos >= A.range) { *A.err = ERR_MF_OVERFLOW; continue; }
+        uint64_t *e = dst + (size_t)pos * ew;
         e[0] = (uint64_t)key;
         for (int w = 0; w < 2 * A.na; w++)
-            e[1 + w] = A.state[(size_t)i * A.na * 2 + w];
+            e[1 + w] = A.state[(size_t)idx * A.na * 2 + w];
     }
+    __syncthreads();
+    if (threadIdx.x == 0) A.cnt[blockIdx.x] = lcur < A.range ? lcur : A.range;
 }
 
 template <int SLOTS, bool PACKED = false>
@@ -2021,11 +1999,11 @@ k_merge_fused(MergeFusedArgs M) {
         if (!PACKED && M.use_cpi) {
             /* dense home-range-grouped entries: only occupied slots read,
              * ownership established at build time */
-            const uint32_t *off = M.cpi_off[p];
-            const uint64_t *ent = M.cpi_entries[p];
             const int ew = 1 + 2 * na;
-            uint32_t lo = off[blockIdx.x], hi = off[blockIdx.x + 1];
-            for (uint32_t t = lo + threadIdx.x; t < hi; t += blockDim.x) {
+            const uint32_t n_e = M.cpi_cnt[p][blockIdx.x];
+            const uint64_t *ent = M.cpi_entries[p] +
+                                  (size_t)blockIdx.x * M.range * ew;
+            for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
                 const uint64_t *e = ent + (size_t)t * ew;
                 fold((int64_t)e[0], e + 1, 0, true);
             }
@@ -2340,11 +2318,8 @@ struct GpuOp {
     int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
                        3 batched-probe, 4 radix-regroup */
     /* closed-pane index (see CpiBuildArgs) */
-    uint64_t *cpi_entries;      /* [R][C][1+2na] */
-    uint32_t *cpi_off;          /* [R][NR+1] */
-    uint32_t *cpi_hist;         /* [NR*CPI_HB + 1] scratch */
-    void *cpi_tmp;
-    size_t cpi_tmp_bytes;
+    uint64_t *cpi_entries;      /* [R][NR][range][1+2na] */
+    uint32_t *cpi_cnt;          /* [R][NR] */
     uint32_t mf_range;          /* fused-merge home-range size (fixed) */
     uint32_t cpi_nr;            /* C / mf_range */
     std::vector<char> cpi_ready;
@@ -2441,21 +2416,10 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.na = (int32_t)na;
     A.range = o->mf_range;
     A.nr = o->cpi_nr;
-    A.hist = o->cpi_hist;
-    A.off = o->cpi_off + (size_t)slot * (o->cpi_nr + 1);
+    A.cnt = o->cpi_cnt + (size_t)slot * o->cpi_nr;
     A.entries = o->cpi_entries + (size_t)slot * o->ring.C * (1 + 2 * na);
-    size_t shmem = (size_t)o->cpi_nr * 4;
-    uint32_t total_idx = o->cpi_nr * CPI_HB;
-    HIP_CHECK(o, hipMemsetAsync(o->cpi_hist + total_idx, 0, 4, o->stream));
-    hipLaunchKernelGGL(k_cpi_count, dim3(CPI_HB), dim3(256), shmem,
-                       o->stream, A);
-    size_t tmp = o->cpi_tmp_bytes;
-    hipcub::DeviceScan::ExclusiveSum(o->cpi_tmp, tmp, o->cpi_hist,
-                                     o->cpi_hist, (int)(total_idx + 1),
-                                     o->stream);
-    hipLaunchKernelGGL(k_cpi_offsets, dim3(1), dim3(256), 0, o->stream,
-                       o->cpi_hist, A.off, o->cpi_nr, total_idx);
-    hipLaunchKernelGGL(k_cpi_scatter, dim3(CPI_HB), dim3(256), shmem,
+    A.err = o->ring.err;
+    hipLaunchKernelGGL(k_cpi_build, dim3(o->cpi_nr), dim3(256), 0,
                        o->stream, A);
     HIP_CHECK(o, hipGetLastError());
     o->cpi_ready[slot] = 1;
@@ -2602,13 +2566,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         if (use_cpi) {
             ALLOC(o->cpi_entries,
                   (size_t)o->ring.R * o->ring.C * (1 + 2 * na) * 8);
-            ALLOC(o->cpi_off, (size_t)o->ring.R * (o->cpi_nr + 1) * 4);
-            ALLOC(o->cpi_hist, ((size_t)o->cpi_nr * CPI_HB + 1) * 4);
-            o->cpi_tmp_bytes = 0;
-            hipcub::DeviceScan::ExclusiveSum(
-                nullptr, o->cpi_tmp_bytes, o->cpi_hist, o->cpi_hist,
-                (int)(o->cpi_nr * CPI_HB + 1));
-            ALLOC(o->cpi_tmp, o->cpi_tmp_bytes ? o->cpi_tmp_bytes : 1);
+            ALLOC(o->cpi_cnt, (size_t)o->ring.R * o->cpi_nr * 4);
         }
     }
     o->cpi_ready.assign(o->ring.R, 0);
@@ -2775,9 +2733,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
     if (o->upd_kind == 4) {
         if (n_rows > o->rdx2_cap) {
             hipFree(o->cpi_entries);
-    hipFree(o->cpi_off);
-    hipFree(o->cpi_hist);
-    hipFree(o->cpi_tmp);
+    hipFree(o->cpi_cnt);
     hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
@@ -3213,8 +3169,8 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                         M.cpi_entries[i] =
                             o->cpi_entries +
                             (size_t)src[i] * o->ring.C * (1 + 2 * na2);
-                        M.cpi_off[i] =
-                            o->cpi_off + (size_t)src[i] * (o->cpi_nr + 1);
+                        M.cpi_cnt[i] =
+                            o->cpi_cnt + (size_t)src[i] * o->cpi_nr;
                     }
                 }
             }
@@ -3682,9 +3638,7 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->rdx_hist);
     hipFree(o->rdx_tmp);
     hipFree(o->cpi_entries);
-    hipFree(o->cpi_off);
-    hipFree(o->cpi_hist);
-    hipFree(o->cpi_tmp);
+    hipFree(o->cpi_cnt);
     hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
