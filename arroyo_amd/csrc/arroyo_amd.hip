@@ -1639,8 +1639,14 @@ k_dict_encode(DictEncArgs A) {
                     if (old == EMPTY_KEY) {
                         for (int k = 0; k < A.nk; k++)
                             A.dkeys[i * A.nk + k] = key[k];
-                        __threadfence();
-                        atomicExch(&A.ready[i], 1u);
+                        /* agent-scope release: the tuple words must be
+                         * visible before the flag (a relaxed flag let
+                         * readers see stale L1 tuple lines, falsely
+                         * mismatch, and claim a SECOND id for the same
+                         * key — observed as split aggregates) */
+                        __hip_atomic_store(&A.ready[i], 1u,
+                                           __ATOMIC_RELEASE,
+                                           __HIP_MEMORY_SCOPE_AGENT);
                         A.ids[r] = (int64_t)i;
                         done = true;
                         break;
@@ -1648,7 +1654,8 @@ k_dict_encode(DictEncArgs A) {
                     cur = old;
                 }
                 if (cur == dg) {
-                    if (!atomicAdd(&A.ready[i], 0u))
+                    if (!__hip_atomic_load(&A.ready[i], __ATOMIC_ACQUIRE,
+                                           __HIP_MEMORY_SCOPE_AGENT))
                         break;   /* writer not published yet: retry row */
                     bool eq = true;
                     for (int k = 0; k < A.nk && eq; k++)
