@@ -650,6 +650,11 @@ class ExpJoinOp:
 
 
 COUNT_DISTINCT = 5
+STDDEV = 6
+STDDEV_POP = 7
+VAR = 8
+VAR_POP = 9
+BIT_XOR = 10
 
 
 class AmdUpdatingConfig(ctypes.Structure):

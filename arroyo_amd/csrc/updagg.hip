@@ -152,6 +152,21 @@ struct UpdateArgs {
     int *err;
 };
 
+/* CAS-fold a double add (no f64 atomicAdd ordering requirements here;
+ * contention is per (key, agg)) */
+__device__ inline void fold_f64(uint64_t *w, double dv) {
+    unsigned long long old = *w, assumed;
+    do {
+        assumed = old;
+        double cur;
+        memcpy(&cur, &assumed, 8);
+        cur += dv;
+        unsigned long long nv;
+        memcpy(&nv, &cur, 8);
+        old = atomicCAS((unsigned long long *)w, assumed, nv);
+    } while (old != assumed);
+}
+
 __global__ void __launch_bounds__(256)
 k_updagg_update(UpdateArgs A) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -208,6 +223,18 @@ k_updagg_update(UpdateArgs A) {
             case AMD_AGG_COUNT_DISTINCT:
                 chain_add(A.store, slot, a, v, d, A.err);
                 break;
+            case AMD_AGG_STDDEV:
+            case AMD_AGG_STDDEV_POP:
+            case AMD_AGG_VAR:
+            case AMD_AGG_VAR_POP:
+                fold_f64(&st[2 * a], (double)d * (double)v);
+                fold_f64(&st[2 * a + 1],
+                         (double)d * (double)v * (double)v);
+                break;
+            case AMD_AGG_BIT_XOR:
+                atomicXor((unsigned long long *)&st[2 * a],
+                          (unsigned long long)v);
+                break;
             }
         }
     }
@@ -235,6 +262,30 @@ __device__ inline void ueval_slot(const UStore &S, const AggSpec &agg,
             memcpy(&out[a], &v, 8);
             break;
         }
+        case AMD_AGG_STDDEV:
+        case AMD_AGG_STDDEV_POP:
+        case AMD_AGG_VAR:
+        case AMD_AGG_VAR_POP: {
+            double n = (double)S.rows[slot];
+            double sx, sxx;
+            uint64_t w0 = st[2 * a], w1 = st[2 * a + 1];
+            memcpy(&sx, &w0, 8);
+            memcpy(&sxx, &w1, 8);
+            double mean = sx / n;
+            double m2 = sxx - n * mean * mean;
+            if (m2 < 0.0) m2 = 0.0;
+            int samp = agg.op[a] == AMD_AGG_STDDEV ||
+                       agg.op[a] == AMD_AGG_VAR;
+            double var = samp ? (n > 1.0 ? m2 / (n - 1.0) : NAN) : m2 / n;
+            if (agg.op[a] == AMD_AGG_STDDEV ||
+                agg.op[a] == AMD_AGG_STDDEV_POP)
+                var = sqrt(var);
+            memcpy(&out[a], &var, 8);
+            break;
+        }
+        case AMD_AGG_BIT_XOR:
+            out[a] = (int64_t)st[2 * a];
+            break;
         case AMD_AGG_COUNT_DISTINCT: {
             int64_t cnt = 0;
             int32_t headv = *(volatile int32_t *)
@@ -738,7 +789,9 @@ API int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out) {
         out->cols = (void **)calloc(o->out_cols, sizeof(void *));
         out->is_f64 = (int32_t *)calloc(o->out_cols, sizeof(int32_t));
         for (int a = 0; a < o->cfg.n_aggs; a++)
-            if (o->cfg.agg_ops[a] == AMD_AGG_AVG)
+            if (o->cfg.agg_ops[a] == AMD_AGG_AVG ||
+                (o->cfg.agg_ops[a] >= AMD_AGG_STDDEV &&
+                 o->cfg.agg_ops[a] <= AMD_AGG_VAR_POP))
                 out->is_f64[o->cfg.n_keys + a] = 1;
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(n ? n : 1) * 8);

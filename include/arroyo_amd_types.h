@@ -34,8 +34,18 @@ enum AmdAggOp {
     AMD_AGG_MIN   = 2, /* partial: min          final: min            */
     AMD_AGG_MAX   = 3, /* partial: max          final: max            */
     AMD_AGG_AVG   = 4, /* partial: (count,sum f64)  final: sum/divide */
-    AMD_AGG_COUNT_DISTINCT = 5 /* updating aggregate only: exact count of
-                                  distinct values under append+retract */
+    AMD_AGG_COUNT_DISTINCT = 5, /* updating aggregate only: exact count of
+                                   distinct values under append+retract */
+    /* round 2, updating aggregate only (every_aggregate.sql coverage):
+     * retractable co-moment states.  n is the key's live row count (the
+     * aggregated column is non-null in this ABI), state = (sum(x) f64,
+     * sum(x^2) f64); sample variants emit NaN when n < 2 (the reference
+     * emits SQL NULL there). */
+    AMD_AGG_STDDEV = 6,
+    AMD_AGG_STDDEV_POP = 7,
+    AMD_AGG_VAR = 8,
+    AMD_AGG_VAR_POP = 9,
+    AMD_AGG_BIT_XOR = 10  /* xor is its own inverse: retract == append */
 };
 
 #define AMD_MAX_AGGS 8

@@ -1804,6 +1804,20 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
             case AMD_AGG_COUNT_DISTINCT:
                 ums_add(&e->ms[a], v, d);
                 break;
+            case AMD_AGG_STDDEV:
+            case AMD_AGG_STDDEV_POP:
+            case AMD_AGG_VAR:
+            case AMD_AGG_VAR_POP:
+                /* retractable co-moments (sum x, sum x^2); n = live rows */
+                e->st[2 * a] = d_to_bits(bits_to_d(e->st[2 * a]) +
+                                         (double)d * (double)v);
+                e->st[2 * a + 1] =
+                    d_to_bits(bits_to_d(e->st[2 * a + 1]) +
+                              (double)d * (double)v * (double)v);
+                break;
+            case AMD_AGG_BIT_XOR:
+                e->st[2 * a] ^= v;   /* self-inverse: retract == append */
+                break;
             }
         }
     }
@@ -1831,6 +1845,29 @@ static void ueval(const UOp *o, const UEntry *e, int64_t *out) {
             out[a] = n;
             break;
         }
+        case AMD_AGG_STDDEV:
+        case AMD_AGG_STDDEV_POP:
+        case AMD_AGG_VAR:
+        case AMD_AGG_VAR_POP: {
+            double n = (double)e->rows;
+            double sx = bits_to_d(e->st[2 * a]);
+            double sxx = bits_to_d(e->st[2 * a + 1]);
+            double mean = sx / n;
+            double m2 = sxx - n * mean * mean;
+            if (m2 < 0.0) m2 = 0.0;   /* fp noise */
+            int samp = c->agg_ops[a] == AMD_AGG_STDDEV ||
+                       c->agg_ops[a] == AMD_AGG_VAR;
+            double var = samp ? (e->rows > 1 ? m2 / (n - 1.0) : NAN)
+                              : m2 / n;
+            if (c->agg_ops[a] == AMD_AGG_STDDEV ||
+                c->agg_ops[a] == AMD_AGG_STDDEV_POP)
+                var = sqrt(var);
+            out[a] = d_to_bits(var);
+            break;
+        }
+        case AMD_AGG_BIT_XOR:
+            out[a] = e->st[2 * a];
+            break;
         }
     }
 }
@@ -1883,7 +1920,9 @@ ORACLE_API int oracle_updagg_flush(void *h, AmdOutBatch *out) {
         out->cols = calloc((size_t)o->out_cols, sizeof(void *));
         out->is_f64 = calloc((size_t)o->out_cols, sizeof(int32_t));
         for (int a = 0; a < c->n_aggs; a++)
-            if (c->agg_ops[a] == AMD_AGG_AVG)
+            if (c->agg_ops[a] == AMD_AGG_AVG ||
+                (c->agg_ops[a] >= AMD_AGG_STDDEV &&
+                 c->agg_ops[a] <= AMD_AGG_VAR_POP))
                 out->is_f64[c->n_keys + a] = 1;
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
