@@ -655,6 +655,20 @@ STDDEV_POP = 7
 VAR = 8
 VAR_POP = 9
 BIT_XOR = 10
+COVAR_POP = 11
+COVAR_SAMP = 12
+CORR = 13
+REGR_SLOPE = 14
+REGR_INTERCEPT = 15
+REGR_R2 = 16
+REGR_AVGX = 17
+REGR_AVGY = 18
+REGR_COUNT = 19
+REGR_SXX = 20
+REGR_SYY = 21
+REGR_SXY = 22
+BIT_AND = 23
+BIT_OR = 24
 
 
 class AmdUpdatingConfig(ctypes.Structure):
@@ -662,6 +676,7 @@ class AmdUpdatingConfig(ctypes.Structure):
         ("n_keys", ctypes.c_int32),
         ("n_value_cols", ctypes.c_int32),
         ("n_aggs", ctypes.c_int32),
+        ("agg_col2", ctypes.c_int32 * 8),
         ("agg_ops", ctypes.c_int32 * 8),
         ("agg_col", ctypes.c_int32 * 8),
         ("log2_capacity", ctypes.c_uint32),
@@ -679,9 +694,11 @@ def make_updagg_config(aggs, n_keys=1, n_value_cols=0, log2_capacity=16,
     cfg.n_keys = n_keys
     cfg.n_value_cols = n_value_cols
     cfg.n_aggs = len(aggs)
-    for i, (op, col) in enumerate(aggs):
+    for i, spec in enumerate(aggs):
+        op, col = spec[0], spec[1]
         cfg.agg_ops[i] = op
         cfg.agg_col[i] = col
+        cfg.agg_col2[i] = spec[2] if len(spec) > 2 else -1
     cfg.log2_capacity = log2_capacity
     cfg.log2_nodes = log2_nodes
     cfg.log2_out_cap = log2_out_cap

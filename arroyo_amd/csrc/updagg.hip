@@ -70,10 +70,15 @@ __device__ inline uint64_t hash64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
+#define UAGG_MAX_SW 40   /* total state words cap per key */
+
 struct AggSpec {
     int32_t n_aggs;
     int32_t op[AMD_MAX_AGGS];
     int32_t col[AMD_MAX_AGGS];
+    int32_t col2[AMD_MAX_AGGS];  /* co-moment family second argument */
+    int32_t soff[AMD_MAX_AGGS];  /* agg a's state words at st[soff[a]..] */
+    int32_t SW;                  /* total state words per key */
 };
 
 /* distinct-value multiset node (append-only pool) */
@@ -88,7 +93,7 @@ struct UStore {
     int64_t *keys;        /* [C+1]; slot C = spec (key == EMPTY_KEY) */
     uint32_t *epoch;      /* [C+1] last-touched flush epoch */
     long long *rows;      /* [C+1] live row count (presence) */
-    uint64_t *st;         /* [(C+1) * n_aggs * 2] scalar states */
+    uint64_t *st;         /* [(C+1) * SW] scalar states (variable width) */
     int64_t *last;        /* [(C+1) * n_aggs] last emitted values */
     uint32_t *emitted;    /* [C+1] */
     int32_t *head;        /* [(C+1) * n_aggs] chains (COUNT_DISTINCT) */
@@ -180,44 +185,33 @@ k_updagg_update(UpdateArgs A) {
         A.store.epoch[slot] = A.cur_epoch;
         atomicAdd((unsigned long long *)&A.store.rows[slot],
                   (unsigned long long)d);
-        uint64_t *st = A.store.st + (size_t)slot * A.agg.n_aggs * 2;
+        uint64_t *stp = A.store.st + (size_t)slot * A.agg.SW;
         for (int a = 0; a < A.agg.n_aggs; a++) {
             int64_t v = A.agg.col[a] >= 0
                             ? A.cols[A.n_keys + A.agg.col[a]][r] : 0;
+            uint64_t *st = stp + A.agg.soff[a];
             switch (A.agg.op[a]) {
             case AMD_AGG_COUNT:
-                atomicAdd((unsigned long long *)&st[2 * a],
+                atomicAdd((unsigned long long *)&st[0],
                           (unsigned long long)d);
                 break;
             case AMD_AGG_SUM:
-                atomicAdd((unsigned long long *)&st[2 * a],
+                atomicAdd((unsigned long long *)&st[0],
                           (unsigned long long)(d * v));
                 break;
-            case AMD_AGG_AVG: {
-                atomicAdd((unsigned long long *)&st[2 * a],
+            case AMD_AGG_AVG:
+                atomicAdd((unsigned long long *)&st[0],
                           (unsigned long long)d);
-                double dv = (double)d * (double)v;
-                unsigned long long old = st[2 * a + 1], assumed;
-                do {
-                    assumed = old;
-                    double cur;
-                    memcpy(&cur, &assumed, 8);
-                    cur += dv;
-                    unsigned long long nv;
-                    memcpy(&nv, &cur, 8);
-                    old = atomicCAS((unsigned long long *)&st[2 * a + 1],
-                                    assumed, nv);
-                } while (old != assumed);
+                fold_f64(&st[1], (double)d * (double)v);
                 break;
-            }
             case AMD_AGG_MIN:
                 if (d < 0) { *A.err = UERR_RETRACT; break; }
-                atomicMax((unsigned long long *)&st[2 * a],
+                atomicMax((unsigned long long *)&st[0],
                           (unsigned long long)enc_min(v));
                 break;
             case AMD_AGG_MAX:
                 if (d < 0) { *A.err = UERR_RETRACT; break; }
-                atomicMax((unsigned long long *)&st[2 * a],
+                atomicMax((unsigned long long *)&st[0],
                           (unsigned long long)enc_max(v));
                 break;
             case AMD_AGG_COUNT_DISTINCT:
@@ -227,14 +221,40 @@ k_updagg_update(UpdateArgs A) {
             case AMD_AGG_STDDEV_POP:
             case AMD_AGG_VAR:
             case AMD_AGG_VAR_POP:
-                fold_f64(&st[2 * a], (double)d * (double)v);
-                fold_f64(&st[2 * a + 1],
-                         (double)d * (double)v * (double)v);
+                fold_f64(&st[0], (double)d * (double)v);
+                fold_f64(&st[1], (double)d * (double)v * (double)v);
                 break;
             case AMD_AGG_BIT_XOR:
-                atomicXor((unsigned long long *)&st[2 * a],
+                atomicXor((unsigned long long *)&st[0],
                           (unsigned long long)v);
                 break;
+            case AMD_AGG_COVAR_POP: case AMD_AGG_COVAR_SAMP:
+            case AMD_AGG_CORR: case AMD_AGG_REGR_SLOPE:
+            case AMD_AGG_REGR_INTERCEPT: case AMD_AGG_REGR_R2:
+            case AMD_AGG_REGR_AVGX: case AMD_AGG_REGR_AVGY:
+            case AMD_AGG_REGR_COUNT: case AMD_AGG_REGR_SXX:
+            case AMD_AGG_REGR_SYY: case AMD_AGG_REGR_SXY: {
+                double y = (double)v;
+                double x = (double)A.cols[A.n_keys + A.agg.col2[a]][r];
+                fold_f64(&st[0], d * y);
+                fold_f64(&st[1], d * x);
+                fold_f64(&st[2], d * x * y);
+                fold_f64(&st[3], d * y * y);
+                fold_f64(&st[4], d * x * x);
+                break;
+            }
+            case AMD_AGG_BIT_AND:
+            case AMD_AGG_BIT_OR: {
+                unsigned int *cnt = (unsigned int *)st;
+                uint64_t uv = (uint64_t)v;
+                unsigned int du = (unsigned int)(int)d;
+                while (uv) {
+                    int b = (int)(__ffsll((long long)uv) - 1);
+                    atomicAdd(&cnt[b], du);  /* u32 wraps; net exact */
+                    uv &= uv - 1;
+                }
+                break;
+            }
             }
         }
     }
@@ -245,20 +265,21 @@ k_updagg_update(UpdateArgs A) {
  * dedup (duplicate nodes from racing first-inserts are rare and benign). */
 __device__ inline void ueval_slot(const UStore &S, const AggSpec &agg,
                                   int64_t slot, int64_t *out) {
-    const uint64_t *st = S.st + (size_t)slot * agg.n_aggs * 2;
+    const uint64_t *stp = S.st + (size_t)slot * agg.SW;
     for (int a = 0; a < agg.n_aggs; a++) {
+        const uint64_t *st = stp + agg.soff[a];
         switch (agg.op[a]) {
         case AMD_AGG_COUNT:
         case AMD_AGG_SUM:
-            out[a] = (int64_t)st[2 * a];
+            out[a] = (int64_t)st[0];
             break;
-        case AMD_AGG_MIN: out[a] = dec_min(st[2 * a]); break;
-        case AMD_AGG_MAX: out[a] = dec_max(st[2 * a]); break;
+        case AMD_AGG_MIN: out[a] = dec_min(st[0]); break;
+        case AMD_AGG_MAX: out[a] = dec_max(st[0]); break;
         case AMD_AGG_AVG: {
             double sum;
-            uint64_t w1 = st[2 * a + 1];
+            uint64_t w1 = st[1];
             memcpy(&sum, &w1, 8);
-            double v = sum / (double)(int64_t)st[2 * a];
+            double v = sum / (double)(int64_t)st[0];
             memcpy(&out[a], &v, 8);
             break;
         }
@@ -268,7 +289,7 @@ __device__ inline void ueval_slot(const UStore &S, const AggSpec &agg,
         case AMD_AGG_VAR_POP: {
             double n = (double)S.rows[slot];
             double sx, sxx;
-            uint64_t w0 = st[2 * a], w1 = st[2 * a + 1];
+            uint64_t w0 = st[0], w1 = st[1];
             memcpy(&sx, &w0, 8);
             memcpy(&sxx, &w1, 8);
             double mean = sx / n;
@@ -284,8 +305,74 @@ __device__ inline void ueval_slot(const UStore &S, const AggSpec &agg,
             break;
         }
         case AMD_AGG_BIT_XOR:
-            out[a] = (int64_t)st[2 * a];
+            out[a] = (int64_t)st[0];
             break;
+        case AMD_AGG_COVAR_POP: case AMD_AGG_COVAR_SAMP:
+        case AMD_AGG_CORR: case AMD_AGG_REGR_SLOPE:
+        case AMD_AGG_REGR_INTERCEPT: case AMD_AGG_REGR_R2:
+        case AMD_AGG_REGR_AVGX: case AMD_AGG_REGR_AVGY:
+        case AMD_AGG_REGR_COUNT: case AMD_AGG_REGR_SXX:
+        case AMD_AGG_REGR_SYY: case AMD_AGG_REGR_SXY: {
+            double n = (double)S.rows[slot];
+            double w[5];
+            for (int k = 0; k < 5; k++) {
+                uint64_t b = st[k];
+                memcpy(&w[k], &b, 8);
+            }
+            double my = w[0] / n, mx = w[1] / n;
+            double Sxy = w[2] - n * mx * my;
+            double Syy = w[3] - n * my * my;
+            double Sxx = w[4] - n * mx * mx;
+            if (Sxx < 0.0) Sxx = 0.0;
+            if (Syy < 0.0) Syy = 0.0;
+            double v;
+            switch (agg.op[a]) {
+            case AMD_AGG_COVAR_POP:  v = Sxy / n; break;
+            case AMD_AGG_COVAR_SAMP:
+                v = n > 1.0 ? Sxy / (n - 1.0) : NAN;
+                break;
+            case AMD_AGG_CORR:
+                v = (n > 1.0 && Sxx > 0.0 && Syy > 0.0)
+                        ? Sxy / sqrt(Sxx * Syy) : NAN;
+                break;
+            case AMD_AGG_REGR_SLOPE:
+                v = Sxx > 0.0 ? Sxy / Sxx : NAN;
+                break;
+            case AMD_AGG_REGR_INTERCEPT:
+                v = Sxx > 0.0 ? my - (Sxy / Sxx) * mx : NAN;
+                break;
+            case AMD_AGG_REGR_R2:
+                v = Sxx <= 0.0 ? NAN
+                    : (Syy <= 0.0 ? 1.0 : (Sxy * Sxy) / (Sxx * Syy));
+                break;
+            case AMD_AGG_REGR_AVGX: v = mx; break;
+            case AMD_AGG_REGR_AVGY: v = my; break;
+            case AMD_AGG_REGR_SXX: v = Sxx; break;
+            case AMD_AGG_REGR_SYY: v = Syy; break;
+            case AMD_AGG_REGR_SXY: v = Sxy; break;
+            default: v = 0.0; break;
+            }
+            if (agg.op[a] == AMD_AGG_REGR_COUNT)
+                out[a] = (int64_t)S.rows[slot];
+            else
+                memcpy(&out[a], &v, 8);
+            break;
+        }
+        case AMD_AGG_BIT_AND:
+        case AMD_AGG_BIT_OR: {
+            const unsigned int *cnt = (const unsigned int *)st;
+            long long rows = S.rows[slot];
+            uint64_t r2 = 0;
+            for (int b = 0; b < 64; b++) {
+                unsigned int nb = cnt[b];
+                int set = agg.op[a] == AMD_AGG_BIT_AND
+                              ? (long long)nb == rows && rows > 0
+                              : nb > 0;
+                if (set) r2 |= 1ULL << b;
+            }
+            out[a] = (int64_t)r2;
+            break;
+        }
         case AMD_AGG_COUNT_DISTINCT: {
             int64_t cnt = 0;
             int32_t headv = *(volatile int32_t *)
@@ -450,8 +537,8 @@ k_updagg_expire(ExpireArgs E) {
         }
         S.emitted[slot] = 0;
         S.rows[slot] = 0;
-        for (int w = 0; w < 2 * na; w++)
-            S.st[(size_t)slot * na * 2 + w] = 0;
+        for (int w = 0; w < E.agg.SW; w++)
+            S.st[(size_t)slot * E.agg.SW + w] = 0;
         for (int a = 0; a < na; a++) S.head[(size_t)slot * 8 + a] = -1;
     }
 }
@@ -463,7 +550,7 @@ struct UDrainArgs {
     UStore store;
     AggSpec agg;
     int32_t n_keys, which;
-    int64_t *out[3 * AMD_MAX_AGGS + 4];
+    int64_t *out[UAGG_MAX_SW + AMD_MAX_AGGS + 4];
     unsigned long long *n_out;
     int64_t out_cap;
     int *err;
@@ -486,9 +573,9 @@ k_updagg_drain(UDrainArgs D) {
             int col = 0;
             if (D.n_keys) D.out[col++][r] = key;
             D.out[col++][r] = (int64_t)S.rows[slot];
-            for (int w = 0; w < 2 * na; w++)
+            for (int w = 0; w < D.agg.SW; w++)
                 D.out[col++][r] =
-                    (int64_t)S.st[(size_t)slot * na * 2 + w];
+                    (int64_t)S.st[(size_t)slot * D.agg.SW + w];
             D.out[col++][r] = (int64_t)S.emitted[slot];
             for (int a = 0; a < na; a++)
                 D.out[col++][r] = S.last[(size_t)slot * na + a];
@@ -514,7 +601,7 @@ k_updagg_drain(UDrainArgs D) {
 /* restore which=0: one thread per row writes the slot's scalar state (each
  * key appears once in drained data); which=1: chain-insert value rows */
 struct URestoreArgs {
-    const int64_t *cols[24];
+    const int64_t *cols[UAGG_MAX_SW + AMD_MAX_AGGS + 4];
     int32_t n_cols;
     int64_t n_rows;
     UStore store;
@@ -535,8 +622,8 @@ k_updagg_restore(URestoreArgs R) {
         if (R.which == 0) {
             int col = R.n_keys;
             R.store.rows[slot] = (long long)R.cols[col++][r];
-            for (int w = 0; w < 2 * na; w++)
-                R.store.st[(size_t)slot * na * 2 + w] =
+            for (int w = 0; w < R.agg.SW; w++)
+                R.store.st[(size_t)slot * R.agg.SW + w] =
                     (uint64_t)R.cols[col++][r];
             R.store.emitted[slot] = (uint32_t)R.cols[col++][r];
             for (int a = 0; a < na; a++)
@@ -591,16 +678,34 @@ API void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg) {
     GpuUpdAgg *o = new GpuUpdAgg();
     o->cfg = *cfg;
     o->agg.n_aggs = cfg->n_aggs;
+    o->agg.SW = 0;
     for (int i = 0; i < cfg->n_aggs; i++) {
         o->agg.op[i] = cfg->agg_ops[i];
         o->agg.col[i] = cfg->agg_col[i];
+        o->agg.col2[i] = cfg->agg_col2[i];
+        o->agg.soff[i] = o->agg.SW;
+        int w = 2;
+        if (cfg->agg_ops[i] >= AMD_AGG_COVAR_POP &&
+            cfg->agg_ops[i] <= AMD_AGG_REGR_SXY)
+            w = 5;
+        else if (cfg->agg_ops[i] == AMD_AGG_BIT_AND ||
+                 cfg->agg_ops[i] == AMD_AGG_BIT_OR)
+            w = 32;
+        o->agg.SW += w;
+    }
+    if (o->agg.SW > UAGG_MAX_SW) {
+        snprintf(g_ua_err, sizeof g_ua_err,
+                 "aggregate state too wide (%d words > %d)", o->agg.SW,
+                 UAGG_MAX_SW);
+        delete o;
+        return nullptr;
     }
     o->store.C = 1u << (cfg->log2_capacity ? cfg->log2_capacity : 16);
     o->store.pool_cap = 1ll << (cfg->log2_nodes ? cfg->log2_nodes : 20);
     o->out_cap = 1ll << (cfg->log2_out_cap ? cfg->log2_out_cap : 20);
     o->n_in_cols = cfg->n_keys + cfg->n_value_cols + 1;
     o->out_cols = cfg->n_keys + cfg->n_aggs + 1;
-    o->drain0_cols = cfg->n_keys + 1 + 2 * cfg->n_aggs + 1 + cfg->n_aggs;
+    o->drain0_cols = cfg->n_keys + 1 + o->agg.SW + 1 + cfg->n_aggs;
     o->drain1_cols = cfg->n_keys + 3;
     o->cur_epoch = 1;
     if (hipSetDevice(cfg->device) != hipSuccess) {
@@ -625,7 +730,7 @@ API void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg) {
     UALLOC(o->store.keys, C1 * 8);
     UALLOC(o->store.epoch, C1 * 4);
     UALLOC(o->store.rows, C1 * 8);
-    UALLOC(o->store.st, C1 * na * 16);
+    UALLOC(o->store.st, C1 * (size_t)o->agg.SW * 8);
     UALLOC(o->store.last, C1 * na * 8);
     UALLOC(o->store.emitted, C1 * 4);
     UALLOC(o->store.head, C1 * 8 * 4);
@@ -646,16 +751,10 @@ API void *arroyo_amd_updagg_create(const AmdUpdatingConfig *cfg) {
     hipMemset(o->store.head, 0xFF, C1 * 8 * 4);
     hipMemset(o->store.pool_cur, 0, 8);
     hipMemset(o->d_err, 0, 4);
-    /* MIN/MAX identities are not the zero pattern: init st explicitly */
-    {
-        std::vector<uint64_t> init(na * 2, 0);
-        for (int a = 0; a < (int)na; a++) {
-            if (cfg->agg_ops[a] == AMD_AGG_MIN) init[2 * a] = 0;
-            if (cfg->agg_ops[a] == AMD_AGG_MAX) init[2 * a] = 0;
-        }
-        /* enc_min/enc_max map the identities to 0, so plain zero works */
-        hipMemset(o->store.st, 0, C1 * na * 16);
-    }
+    /* every state's identity is the zero pattern (enc_min/enc_max map the
+     * MIN/MAX identities to 0; f64 sums start at +0.0 = zero bits; bit
+     * counters at 0), so a plain memset initializes the plane */
+    hipMemset(o->store.st, 0, C1 * (size_t)o->agg.SW * 8);
     hipStreamCreate(&o->stream);
     o->stg_cap = 1 << 20;
     for (int c = 0; c < o->n_in_cols; c++) {

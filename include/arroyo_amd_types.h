@@ -45,7 +45,28 @@ enum AmdAggOp {
     AMD_AGG_STDDEV_POP = 7,
     AMD_AGG_VAR = 8,
     AMD_AGG_VAR_POP = 9,
-    AMD_AGG_BIT_XOR = 10  /* xor is its own inverse: retract == append */
+    AMD_AGG_BIT_XOR = 10, /* xor is its own inverse: retract == append */
+    /* two-argument co-moment family (updating aggregate only): state =
+     * (sum y, sum x, sum xy, sum y^2, sum x^2) f64; the FIRST argument
+     * (SQL's Y) is agg_col, the second (X) agg_col2.  Sample variants and
+     * zero-variance cases emit NaN where the reference emits SQL NULL. */
+    AMD_AGG_COVAR_POP = 11,
+    AMD_AGG_COVAR_SAMP = 12,
+    AMD_AGG_CORR = 13,
+    AMD_AGG_REGR_SLOPE = 14,
+    AMD_AGG_REGR_INTERCEPT = 15,
+    AMD_AGG_REGR_R2 = 16,
+    AMD_AGG_REGR_AVGX = 17,
+    AMD_AGG_REGR_AVGY = 18,
+    AMD_AGG_REGR_COUNT = 19,   /* i64 output */
+    AMD_AGG_REGR_SXX = 20,
+    AMD_AGG_REGR_SYY = 21,
+    AMD_AGG_REGR_SXY = 22,
+    /* retractable bit aggregates: 64 per-bit u32 set-counts (32 state
+     * words); AND = bit set in every live row, OR = in any.  bool_and /
+     * bool_or are these over a 0/1 column. */
+    AMD_AGG_BIT_AND = 23,
+    AMD_AGG_BIT_OR = 24
 };
 
 #define AMD_MAX_AGGS 8
@@ -199,6 +220,8 @@ typedef struct {
     int32_t  n_keys;            /* 0 or 1 */
     int32_t  n_value_cols;
     int32_t  n_aggs;
+    int32_t  agg_col2[AMD_MAX_AGGS]; /* second argument (X) of the
+                                        co-moment family; -1 otherwise */
     int32_t  agg_ops[AMD_MAX_AGGS];
     int32_t  agg_col[AMD_MAX_AGGS];
     uint32_t log2_capacity;     /* key slots (GPU) */

@@ -1660,9 +1660,11 @@ typedef struct {
     int64_t *cnt;             /* net refcounts */
 } UMultiset;
 
+#define UAGG_MAX_SW 256   /* total state words cap (BIT ops take 32) */
+
 typedef struct {
     int64_t rows;             /* live row count (presence) */
-    int64_t st[AMD_MAX_AGGS * 2];
+    int64_t st[UAGG_MAX_SW];
     UMultiset ms[AMD_MAX_AGGS];   /* per COUNT_DISTINCT agg */
     int emitted;
     int64_t last[AMD_MAX_AGGS];   /* last emitted values (AVG: f64 bits) */
@@ -1672,6 +1674,10 @@ typedef struct {
 
 typedef struct {
     AmdUpdatingConfig cfg;
+    /* variable-width states: agg a's words at st[soff[a]..+usw[a]) */
+    int soff[AMD_MAX_AGGS];
+    int usw[AMD_MAX_AGGS];
+    int sw_total;
     int64_t map_cap, map_n;
     int64_t *map_keys;
     uint8_t *map_used;
@@ -1682,6 +1688,12 @@ typedef struct {
     int64_t epoch;
     char err[256];
 } UOp;
+
+static int uagg_width(int op) {
+    if (op >= AMD_AGG_COVAR_POP && op <= AMD_AGG_REGR_SXY) return 5;
+    if (op == AMD_AGG_BIT_AND || op == AMD_AGG_BIT_OR) return 32;
+    return 2;
+}
 
 static void ugrow(UOp *o);
 
@@ -1698,9 +1710,8 @@ static int64_t ukey_slot(UOp *o, int64_t key) {
         memset(e, 0, sizeof *e);
         for (int a = 0; a < o->cfg.n_aggs; a++) {
             switch (o->cfg.agg_ops[a]) {
-            case AMD_AGG_MIN: e->st[2 * a] = INT64_MAX; break;
-            case AMD_AGG_MAX: e->st[2 * a] = INT64_MIN; break;
-            case AMD_AGG_AVG: e->st[2 * a + 1] = d_to_bits(0.0); break;
+            case AMD_AGG_MIN: e->st[o->soff[a]] = INT64_MAX; break;
+            case AMD_AGG_MAX: e->st[o->soff[a]] = INT64_MIN; break;
             }
         }
     }
@@ -1729,6 +1740,12 @@ ORACLE_API void *oracle_updagg_create(const AmdUpdatingConfig *cfg) {
         return NULL;
     UOp *o = calloc(1, sizeof(UOp));
     o->cfg = *cfg;
+    for (int a = 0; a < cfg->n_aggs; a++) {
+        o->soff[a] = o->sw_total;
+        o->usw[a] = uagg_width(cfg->agg_ops[a]);
+        o->sw_total += o->usw[a];
+    }
+    if (o->sw_total > UAGG_MAX_SW) { free(o); return NULL; }
     o->map_cap = 64;
     o->map_keys = malloc((size_t)o->map_cap * 8);
     o->map_used = calloc((size_t)o->map_cap, 1);
@@ -1777,13 +1794,13 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
         for (int a = 0; a < c->n_aggs; a++) {
             int64_t v = c->agg_col[a] >= 0
                             ? cols[c->n_keys + c->agg_col[a]][r] : 0;
+            int64_t *st = &e->st[o->soff[a]];
             switch (c->agg_ops[a]) {
-            case AMD_AGG_COUNT: e->st[2 * a] += d; break;
-            case AMD_AGG_SUM:   e->st[2 * a] += d * v; break;
+            case AMD_AGG_COUNT: st[0] += d; break;
+            case AMD_AGG_SUM:   st[0] += d * v; break;
             case AMD_AGG_AVG:
-                e->st[2 * a] += d;
-                e->st[2 * a + 1] =
-                    d_to_bits(bits_to_d(e->st[2 * a + 1]) + (double)d * v);
+                st[0] += d;
+                st[1] = d_to_bits(bits_to_d(st[1]) + (double)d * v);
                 break;
             case AMD_AGG_MIN:
                 if (d < 0) {
@@ -1791,7 +1808,7 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
                              "MIN does not support retraction");
                     return 1;
                 }
-                if (v < e->st[2 * a]) e->st[2 * a] = v;
+                if (v < st[0]) st[0] = v;
                 break;
             case AMD_AGG_MAX:
                 if (d < 0) {
@@ -1799,7 +1816,7 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
                              "MAX does not support retraction");
                     return 1;
                 }
-                if (v > e->st[2 * a]) e->st[2 * a] = v;
+                if (v > st[0]) st[0] = v;
                 break;
             case AMD_AGG_COUNT_DISTINCT:
                 ums_add(&e->ms[a], v, d);
@@ -1809,15 +1826,41 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
             case AMD_AGG_VAR:
             case AMD_AGG_VAR_POP:
                 /* retractable co-moments (sum x, sum x^2); n = live rows */
-                e->st[2 * a] = d_to_bits(bits_to_d(e->st[2 * a]) +
-                                         (double)d * (double)v);
-                e->st[2 * a + 1] =
-                    d_to_bits(bits_to_d(e->st[2 * a + 1]) +
-                              (double)d * (double)v * (double)v);
+                st[0] = d_to_bits(bits_to_d(st[0]) + (double)d * (double)v);
+                st[1] = d_to_bits(bits_to_d(st[1]) +
+                                  (double)d * (double)v * (double)v);
                 break;
             case AMD_AGG_BIT_XOR:
-                e->st[2 * a] ^= v;   /* self-inverse: retract == append */
+                st[0] ^= v;   /* self-inverse: retract == append */
                 break;
+            case AMD_AGG_COVAR_POP: case AMD_AGG_COVAR_SAMP:
+            case AMD_AGG_CORR: case AMD_AGG_REGR_SLOPE:
+            case AMD_AGG_REGR_INTERCEPT: case AMD_AGG_REGR_R2:
+            case AMD_AGG_REGR_AVGX: case AMD_AGG_REGR_AVGY:
+            case AMD_AGG_REGR_COUNT: case AMD_AGG_REGR_SXX:
+            case AMD_AGG_REGR_SYY: case AMD_AGG_REGR_SXY: {
+                /* (sum y, sum x, sum xy, sum y^2, sum x^2); y = agg_col
+                 * (SQL's first arg), x = agg_col2 */
+                double y = (double)v;
+                double x = (double)cols[c->n_keys + c->agg_col2[a]][r];
+                st[0] = d_to_bits(bits_to_d(st[0]) + d * y);
+                st[1] = d_to_bits(bits_to_d(st[1]) + d * x);
+                st[2] = d_to_bits(bits_to_d(st[2]) + d * x * y);
+                st[3] = d_to_bits(bits_to_d(st[3]) + d * y * y);
+                st[4] = d_to_bits(bits_to_d(st[4]) + d * x * x);
+                break;
+            }
+            case AMD_AGG_BIT_AND:
+            case AMD_AGG_BIT_OR: {
+                uint32_t *cnt = (uint32_t *)st;
+                uint64_t uv = (uint64_t)v;
+                while (uv) {
+                    int b = __builtin_ctzll(uv);
+                    cnt[b] += (uint32_t)d;   /* u32 wraps; net is exact */
+                    uv &= uv - 1;
+                }
+                break;
+            }
             }
         }
     }
@@ -1827,16 +1870,17 @@ ORACLE_API int oracle_updagg_process_batch(void *h,
 static void ueval(const UOp *o, const UEntry *e, int64_t *out) {
     const AmdUpdatingConfig *c = &o->cfg;
     for (int a = 0; a < c->n_aggs; a++) {
+        const int64_t *st = &e->st[o->soff[a]];
         switch (c->agg_ops[a]) {
         case AMD_AGG_COUNT:
         case AMD_AGG_SUM:
         case AMD_AGG_MIN:
         case AMD_AGG_MAX:
-            out[a] = e->st[2 * a];
+        case AMD_AGG_BIT_XOR:
+            out[a] = st[0];
             break;
         case AMD_AGG_AVG:
-            out[a] = d_to_bits(bits_to_d(e->st[2 * a + 1]) /
-                               (double)e->st[2 * a]);
+            out[a] = d_to_bits(bits_to_d(st[1]) / (double)st[0]);
             break;
         case AMD_AGG_COUNT_DISTINCT: {
             int64_t n = 0;
@@ -1850,8 +1894,8 @@ static void ueval(const UOp *o, const UEntry *e, int64_t *out) {
         case AMD_AGG_VAR:
         case AMD_AGG_VAR_POP: {
             double n = (double)e->rows;
-            double sx = bits_to_d(e->st[2 * a]);
-            double sxx = bits_to_d(e->st[2 * a + 1]);
+            double sx = bits_to_d(st[0]);
+            double sxx = bits_to_d(st[1]);
             double mean = sx / n;
             double m2 = sxx - n * mean * mean;
             if (m2 < 0.0) m2 = 0.0;   /* fp noise */
@@ -1865,9 +1909,69 @@ static void ueval(const UOp *o, const UEntry *e, int64_t *out) {
             out[a] = d_to_bits(var);
             break;
         }
-        case AMD_AGG_BIT_XOR:
-            out[a] = e->st[2 * a];
+        case AMD_AGG_COVAR_POP: case AMD_AGG_COVAR_SAMP:
+        case AMD_AGG_CORR: case AMD_AGG_REGR_SLOPE:
+        case AMD_AGG_REGR_INTERCEPT: case AMD_AGG_REGR_R2:
+        case AMD_AGG_REGR_AVGX: case AMD_AGG_REGR_AVGY:
+        case AMD_AGG_REGR_COUNT: case AMD_AGG_REGR_SXX:
+        case AMD_AGG_REGR_SYY: case AMD_AGG_REGR_SXY: {
+            double n = (double)e->rows;
+            double sy = bits_to_d(st[0]), sx = bits_to_d(st[1]);
+            double sxy = bits_to_d(st[2]), syy = bits_to_d(st[3]);
+            double sxx2 = bits_to_d(st[4]);
+            double my = sy / n, mx = sx / n;
+            double Sxy = sxy - n * mx * my;
+            double Sxx = sxx2 - n * mx * mx;
+            double Syy = syy - n * my * my;
+            if (Sxx < 0.0) Sxx = 0.0;
+            if (Syy < 0.0) Syy = 0.0;
+            double v;
+            switch (c->agg_ops[a]) {
+            case AMD_AGG_COVAR_POP:  v = Sxy / n; break;
+            case AMD_AGG_COVAR_SAMP:
+                v = e->rows > 1 ? Sxy / (n - 1.0) : NAN;
+                break;
+            case AMD_AGG_CORR:
+                v = (e->rows > 1 && Sxx > 0.0 && Syy > 0.0)
+                        ? Sxy / sqrt(Sxx * Syy) : NAN;
+                break;
+            case AMD_AGG_REGR_SLOPE:
+                v = Sxx > 0.0 ? Sxy / Sxx : NAN;
+                break;
+            case AMD_AGG_REGR_INTERCEPT:
+                v = Sxx > 0.0 ? my - (Sxy / Sxx) * mx : NAN;
+                break;
+            case AMD_AGG_REGR_R2:
+                v = Sxx <= 0.0 ? NAN
+                    : (Syy <= 0.0 ? 1.0 : (Sxy * Sxy) / (Sxx * Syy));
+                break;
+            case AMD_AGG_REGR_AVGX: v = mx; break;
+            case AMD_AGG_REGR_AVGY: v = my; break;
+            case AMD_AGG_REGR_SXX: v = Sxx; break;
+            case AMD_AGG_REGR_SYY: v = Syy; break;
+            case AMD_AGG_REGR_SXY: v = Sxy; break;
+            default: v = 0.0; break;   /* unreachable */
+            }
+            if (c->agg_ops[a] == AMD_AGG_REGR_COUNT)
+                out[a] = e->rows;              /* i64 output */
+            else
+                out[a] = d_to_bits(v);
             break;
+        }
+        case AMD_AGG_BIT_AND:
+        case AMD_AGG_BIT_OR: {
+            const uint32_t *cnt = (const uint32_t *)st;
+            uint64_t r2 = 0;
+            for (int b = 0; b < 64; b++) {
+                uint32_t nb = cnt[b];
+                int set = c->agg_ops[a] == AMD_AGG_BIT_AND
+                              ? (int64_t)nb == e->rows && e->rows > 0
+                              : nb > 0;
+                if (set) r2 |= 1ULL << b;
+            }
+            out[a] = (int64_t)r2;
+            break;
+        }
         }
     }
 }
@@ -1922,7 +2026,10 @@ ORACLE_API int oracle_updagg_flush(void *h, AmdOutBatch *out) {
         for (int a = 0; a < c->n_aggs; a++)
             if (c->agg_ops[a] == AMD_AGG_AVG ||
                 (c->agg_ops[a] >= AMD_AGG_STDDEV &&
-                 c->agg_ops[a] <= AMD_AGG_VAR_POP))
+                 c->agg_ops[a] <= AMD_AGG_VAR_POP) ||
+                (c->agg_ops[a] >= AMD_AGG_COVAR_POP &&
+                 c->agg_ops[a] <= AMD_AGG_REGR_SXY &&
+                 c->agg_ops[a] != AMD_AGG_REGR_COUNT))
                 out->is_f64[c->n_keys + a] = 1;
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(o->out_rows ? o->out_rows : 1) * 8);
@@ -1989,7 +2096,7 @@ ORACLE_API int oracle_updagg_checkpoint_drain(void *h, int32_t which,
     int ncols;
     int64_t total = 0;
     if (which == 0) {
-        ncols = c->n_keys + 1 + 2 * c->n_aggs + 1 + c->n_aggs;
+        ncols = c->n_keys + 1 + o->sw_total + 1 + c->n_aggs;
         for (int64_t s = 0; s < o->map_cap; s++)
             if (o->map_used[s]) total++;
     } else {
@@ -2015,7 +2122,7 @@ ORACLE_API int oracle_updagg_checkpoint_drain(void *h, int32_t which,
             int col = 0;
             if (c->n_keys) ((int64_t *)out->cols[col++])[r] = o->map_keys[s];
             ((int64_t *)out->cols[col++])[r] = e->rows;
-            for (int w = 0; w < 2 * c->n_aggs; w++)
+            for (int w = 0; w < o->sw_total; w++)
                 ((int64_t *)out->cols[col++])[r] = e->st[w];
             ((int64_t *)out->cols[col++])[r] = e->emitted;
             for (int a = 0; a < c->n_aggs; a++)
@@ -2044,7 +2151,7 @@ ORACLE_API int oracle_updagg_restore(void *h, int32_t which,
     UOp *o = h;
     const AmdUpdatingConfig *c = &o->cfg;
     if (which == 0) {
-        int want = c->n_keys + 1 + 2 * c->n_aggs + 1 + c->n_aggs;
+        int want = c->n_keys + 1 + o->sw_total + 1 + c->n_aggs;
         if (n_cols != want) {
             snprintf(o->err, sizeof o->err, "restore(0) expects %d cols",
                      want);
@@ -2055,7 +2162,7 @@ ORACLE_API int oracle_updagg_restore(void *h, int32_t which,
             UEntry *e = &o->ent[s];
             int col = c->n_keys;
             e->rows = cols[col++][r];
-            for (int w = 0; w < 2 * c->n_aggs; w++)
+            for (int w = 0; w < o->sw_total; w++)
                 e->st[w] = cols[col++][r];
             e->emitted = (int)cols[col++][r];
             for (int a = 0; a < c->n_aggs; a++)
