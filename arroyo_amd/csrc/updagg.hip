@@ -70,7 +70,7 @@ __device__ inline uint64_t hash64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
-#define UAGG_MAX_SW 40   /* total state words cap per key */
+#define UAGG_MAX_SW 128  /* total state words cap per key */
 
 struct AggSpec {
     int32_t n_aggs;
