@@ -752,7 +752,9 @@ class UpdAggOp:
         self.cfg = cfg
         self._h = self._fn["create"](ctypes.byref(cfg))
         if not self._h:
-            raise RuntimeError(f"{p}create failed")
+            raise RuntimeError(
+                f"{p}create failed: "
+                f"{self._fn['last_error'](None).decode()}")
 
     def _check(self, rc):
         if rc != 0:
@@ -897,7 +899,9 @@ class WindowFnOp:
         self.cfg = cfg
         self._h = self._fn["create"](ctypes.byref(cfg))
         if not self._h:
-            raise RuntimeError(f"{p}create failed")
+            raise RuntimeError(
+                f"{p}create failed: "
+                f"{self._fn['last_error'](None).decode()}")
 
     def _check(self, rc):
         if rc != 0:
@@ -1041,7 +1045,9 @@ class MapOp:
         self.cfg = cfg
         self._h = self._fn["create"](ctypes.byref(cfg))
         if not self._h:
-            raise RuntimeError(f"{p}create failed")
+            raise RuntimeError(
+                f"{p}create failed: "
+                f"{self._fn['last_error'](None).decode()}")
 
     def process_batch_device(self, dptrs, n_rows):
         """Device-resident map/filter (GPU-only extension): input device

@@ -647,7 +647,7 @@ struct GpuUpdAgg {
     AggSpec agg;
     UStore store;
     uint32_t cur_epoch;
-    int64_t *d_out[3 * AMD_MAX_AGGS + 4];
+    int64_t *d_out[UAGG_MAX_SW + AMD_MAX_AGGS + 4];
     unsigned long long *d_n_out;
     int *d_err;
     int64_t *stg_h[12], *stg_d[12];
