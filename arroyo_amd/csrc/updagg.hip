@@ -890,7 +890,10 @@ API int arroyo_amd_updagg_flush(void *h, AmdOutBatch *out) {
         for (int a = 0; a < o->cfg.n_aggs; a++)
             if (o->cfg.agg_ops[a] == AMD_AGG_AVG ||
                 (o->cfg.agg_ops[a] >= AMD_AGG_STDDEV &&
-                 o->cfg.agg_ops[a] <= AMD_AGG_VAR_POP))
+                 o->cfg.agg_ops[a] <= AMD_AGG_VAR_POP) ||
+                (o->cfg.agg_ops[a] >= AMD_AGG_COVAR_POP &&
+                 o->cfg.agg_ops[a] <= AMD_AGG_REGR_SXY &&
+                 o->cfg.agg_ops[a] != AMD_AGG_REGR_COUNT))
                 out->is_f64[o->cfg.n_keys + a] = 1;
         for (int i = 0; i < o->out_cols; i++) {
             out->cols[i] = malloc((size_t)(n ? n : 1) * 8);
