@@ -1660,7 +1660,8 @@ typedef struct {
     int64_t *cnt;             /* net refcounts */
 } UMultiset;
 
-#define UAGG_MAX_SW 256   /* total state words cap (BIT ops take 32) */
+#define UAGG_MAX_SW 128   /* total state words cap (BIT ops take 32);
+                             matches the GPU path */
 
 typedef struct {
     int64_t rows;             /* live row count (presence) */

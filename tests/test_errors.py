@@ -111,3 +111,27 @@ def test_updagg_wrong_cols():
         op.process_batch([np.zeros(3, dtype=np.int64),
                           np.zeros(3, dtype=np.int64)])
     op.close()
+
+
+def test_updagg_state_width_cap_is_loud():
+    """> UAGG_MAX_SW state words must fail at create with a clear message
+    (oracle mirrors the cap)."""
+    import oracle
+    from arroyo_amd import cabi
+
+    # 8 bit aggregates = 256 words > both caps
+    aggs = [(cabi.BIT_AND, 0)] * 8
+    cfg = cabi.make_updagg_config(aggs, n_keys=1, n_value_cols=1)
+    with pytest.raises(RuntimeError):
+        oracle.make_updagg_op(cfg)
+
+
+def test_window_rejects_more_than_four_keys():
+    import oracle
+    from arroyo_amd import cabi
+    from arroyo_amd.pipeline import NS
+
+    with pytest.raises(RuntimeError):
+        oracle.make_op(cabi.make_config(
+            width_ns=4 * NS, slide_ns=2 * NS, n_keys=5, n_value_cols=0,
+            aggs=[(cabi.COUNT, -1)]))
