@@ -83,6 +83,15 @@ def bind(lib, prefix):
     fn["handle_watermark"].restype = ctypes.c_int
     fn["handle_watermark"].argtypes = [ctypes.c_void_p, ctypes.c_uint64,
                                        ctypes.POINTER(AmdOutBatch)]
+    try:
+        # batched variant (HIP library only; the oracle loops in Python)
+        fn["handle_watermarks"] = g("handle_watermarks")
+        fn["handle_watermarks"].restype = ctypes.c_int
+        fn["handle_watermarks"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint64), ctypes.c_int32,
+            ctypes.POINTER(AmdOutBatch)]
+    except AttributeError:
+        pass
     fn["checkpoint_drain"] = g("checkpoint_drain")
     fn["checkpoint_drain"].restype = ctypes.c_int
     fn["checkpoint_drain"].argtypes = [ctypes.c_void_p,
@@ -289,6 +298,29 @@ class WindowOp:
     def handle_watermark(self, wm):
         out = AmdOutBatch()
         rc = self._fn["handle_watermark"](self._h, wm, ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def handle_watermarks(self, wms):
+        """Batched watermarks with no rows between them: one device-status
+        round trip serves the whole group.  Semantically identical to
+        handle_watermark in order; returns all emissions concatenated.
+        Falls back to a Python loop where the library lacks the entry
+        point (the oracle)."""
+        if "handle_watermarks" not in self._fn:
+            outs = [self.handle_watermark(w) for w in wms]
+            outs = [o for o in outs if o and len(o[0])]
+            if not outs:
+                return []
+            return [np.concatenate([o[c] for o in outs])
+                    for c in range(len(outs[0]))]
+        out = AmdOutBatch()
+        arr = (ctypes.c_uint64 * len(wms))(*wms)
+        rc = self._fn["handle_watermarks"](self._h, arr, len(wms),
+                                           ctypes.byref(out))
         if rc != 0:
             raise RuntimeError(self._fn["last_error"](self._h).decode())
         cols = _out_to_numpy(out)

@@ -144,17 +144,23 @@ def run_gpu(args):
 
     wm_fires = [0]   # watermarks emitted (window fires happen inside these)
 
-    def emit_watermark(step):
+    def wm_value(step):
         """watermark after `step` batches: reference cadence ~1/s of event
         time (watermark_generator.rs), value = max_ts - 1s lateness."""
         b = (step - 1) % BASE_BATCHES
         cycle = (step - 1) // BASE_BATCHES
         max_ts = int(ts[(b + 1) * BATCH_ROWS - 1]) + cycle * span
-        rc = wm_lib.arroyo_amd_handle_watermark(
-            op._h, ctypes.c_uint64(max_ts - NS), None)
+        return max_ts - NS
+
+    def emit_watermarks(steps):
+        """one batched call per fused period: no rows arrive between these
+        watermarks, so one device-status read serves the group."""
+        vals = (ctypes.c_uint64 * len(steps))(*[wm_value(s) for s in steps])
+        rc = wm_lib.arroyo_amd_handle_watermarks(
+            op._h, vals, ctypes.c_int32(len(steps)), None)
         if rc != 0:
             raise RuntimeError(op._fn["last_error"](op._h).decode())
-        wm_fires[0] += 1
+        wm_fires[0] += len(steps)
 
     # Watermark-period fusion: the operator's late-row filter guarantees a
     # row arriving after watermark W can never land in a pane that W's
@@ -186,11 +192,10 @@ def run_gpu(args):
                 take -= sub
             # emit every watermark boundary crossed so far, in order
             first_unemitted = (run_span.last_wm // wm_every + 1) * wm_every
-            w = first_unemitted
-            while w <= s:
-                emit_watermark(w)
-                run_span.last_wm = w
-                w += wm_every
+            group = list(range(first_unemitted, s + 1, wm_every))
+            if group:
+                emit_watermarks(group)
+                run_span.last_wm = group[-1]
 
     run_span.last_wm = 0
 

@@ -90,6 +90,15 @@ int arroyo_amd_sync(void *h);
 int arroyo_amd_handle_watermark(void *h, uint64_t watermark_nanos,
                                 AmdOutBatch *out);
 
+/* Batched form of handle_watermark for a group of watermarks with NO
+ * process_batch between them (the periodic WatermarkGenerator emits such
+ * runs whenever the source idles; the bench's fused periods do too).
+ * Equivalent to calling handle_watermark(wms[i]) in order, but the
+ * device status (pane tags, min non-late bin) is read ONCE for the whole
+ * group — each per-watermark read costs a stream sync (~15 us idle). */
+int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms, int32_t n,
+                                 AmdOutBatch *out);
+
 /* Drain open panes' partial states for a checkpoint barrier.  Columns:
  * [key?, partial state words (AVG takes 2)..., bin _timestamp]. */
 int arroyo_amd_checkpoint_drain(void *h, AmdOutBatch *out);

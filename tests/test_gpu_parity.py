@@ -300,3 +300,36 @@ def test_width_equals_slide():
                          n_value_cols=0, aggs=[(cabi.COUNT, -1)],
                          log2_capacity=14, ring_panes=16)
     assert_parity(got, want)
+
+
+def test_batched_watermarks_equal_sequential():
+    """arroyo_amd_handle_watermarks (one device-status read per group of
+    row-free watermarks, the WatermarkGenerator's idle-source cadence)
+    must be emission-identical to per-watermark handle_watermark calls."""
+    cols = nexmark.bids(400_000, events_per_sec=100_000, seed=23)
+    kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=1,
+              aggs=[(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.MAX, 0)],
+              log2_capacity=15, ring_panes=32)
+    a, b = gpu_op(**kw), gpu_op(**kw)
+    batches = batches_from_columns(cols, 50_000)
+    seq, bat = [], []
+    pend = []
+    for i, cb in enumerate(batches):
+        a.process_batch(cb)
+        b.process_batch(cb)
+        wm = int(cb[-1][-1]) - NS
+        pend.append(wm)
+        if len(pend) == 3 or i == len(batches) - 1:
+            for w in pend:
+                out = a.handle_watermark(w)
+                if out and len(out[0]):
+                    seq.append(out)
+            out = b.handle_watermarks(pend)
+            if out and len(out[0]):
+                bat.append(out)
+            pend = []
+    a.close()
+    b.close()
+    got = concat_outputs(bat)
+    want = concat_outputs(seq)
+    assert_parity(got, want)
