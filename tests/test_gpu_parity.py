@@ -306,7 +306,8 @@ def test_batched_watermarks_equal_sequential():
     """arroyo_amd_handle_watermarks (one device-status read per group of
     row-free watermarks, the WatermarkGenerator's idle-source cadence)
     must be emission-identical to per-watermark handle_watermark calls."""
-    cols = nexmark.bids(400_000, events_per_sec=100_000, seed=23)
+    cols = nexmark.bids(400_000, events_per_sec=100_000, seed=23,
+                        with_price=True)
     kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=1,
               aggs=[(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.MAX, 0)],
               log2_capacity=15, ring_panes=32)
