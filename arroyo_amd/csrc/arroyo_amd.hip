@@ -2304,6 +2304,23 @@ struct GpuOp {
     std::vector<std::vector<int64_t>> host_out;
 
     hipStream_t stream;
+    /* fire path (pane close -> cpi build -> merge -> retire) runs on its
+     * own stream, overlapping the NEXT periods' update kernels: both are
+     * latency-bound (update 87.5% memory-parked, merge ~470 GB/s) so
+     * co-resident waves raise aggregate memory-level parallelism.  Safe
+     * because the late-row filter guarantees no update launched after
+     * watermark W writes a pane W's fires read.  fstream == stream when
+     * ARROYO_AMD_FIRE_STREAM=0 (serial debug mode). */
+    hipStream_t fstream;
+    int own_fstream;
+    hipEvent_t ev_gate;       /* updates-done gate the fire stream waits on */
+    hipEvent_t ev_tail[4];    /* per-group fire-tail ring: bounds the fire
+                                 stream's lag to 4 groups << R/2 periods, so
+                                 a ring slot's retire always completes
+                                 before the slot's bin comes around again */
+    uint64_t fire_group;
+    std::vector<uint64_t> retired_bin;  /* [R] bin whose async retire may
+                                           not yet show in d_status */
     Staged stg;
 
     /* one contiguous device status block [err, min_bin, tags[R]] mirrored
@@ -2384,14 +2401,42 @@ static void harvest_events(GpuOp *o) {
 
 static thread_local char g_err[512];
 
-static int ring_retire(GpuOp *o, uint32_t slot) {
+/* order the fire stream after every update launched so far (one event) */
+static int gate_fire(GpuOp *o) {
+    if (o->fstream == o->stream) return 0;
+    HIP_CHECK(o, hipEventRecord(o->ev_gate, o->stream));
+    HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_gate, 0));
+    return 0;
+}
+
+/* record this watermark group's fire tail (retires included) */
+static int fire_tail(GpuOp *o) {
+    if (o->fstream == o->stream) return 0;
+    HIP_CHECK(o, hipEventRecord(o->ev_tail[o->fire_group & 3], o->fstream));
+    o->fire_group++;
+    return 0;
+}
+
+/* bound the fire stream's lag: before this group's (and later groups')
+ * update launches run, the group from 4 ago must have fully retired --
+ * slot reuse is bin-deterministic (p = q & (R-1)) so R/2 >> 4 periods
+ * separate a retire from the slot's next claim */
+static int bound_fire_lag(GpuOp *o) {
+    if (o->fstream == o->stream || o->fire_group < 4) return 0;
+    HIP_CHECK(o, hipStreamWaitEvent(o->stream,
+                                    o->ev_tail[o->fire_group & 3], 0));
+    return 0;
+}
+
+static int ring_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
     size_t na = o->agg.n_aggs;
     if (slot < o->cpi_ready.size()) o->cpi_ready[slot] = 0;
+    if (slot < o->retired_bin.size()) o->retired_bin[slot] = bin;
     int blocks = (int)((o->ring.C + 255) / 256);
     if (blocks > 1024) blocks = 1024;
     if (o->ring.packed) {
         hipLaunchKernelGGL(k_retire_packed, dim3(blocks), dim3(256), 0,
-                           o->stream,
+                           o->fstream,
                            o->ring.slots + (size_t)slot * o->ring.C * 2,
                            (int64_t)o->ring.C, o->ring.tag + slot,
                            o->ring.spec_used + slot,
@@ -2399,7 +2444,7 @@ static int ring_retire(GpuOp *o, uint32_t slot) {
                            (int)(na * 2));
     } else {
         hipLaunchKernelGGL(k_retire_all, dim3(blocks), dim3(256), 0,
-                           o->stream,
+                           o->fstream,
                            o->ring.keys + (size_t)slot * o->ring.C,
                            o->ring.state + (size_t)slot * o->ring.C * na * 2,
                            (int64_t)o->ring.C, (int)(na * 2),
@@ -2427,7 +2472,7 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.entries = o->cpi_entries + (size_t)slot * o->ring.C * (1 + 2 * na);
     A.err = o->ring.err;
     hipLaunchKernelGGL(k_cpi_build, dim3(o->cpi_nr), dim3(256), 0,
-                       o->stream, A);
+                       o->fstream, A);
     HIP_CHECK(o, hipGetLastError());
     o->cpi_ready[slot] = 1;
     return 0;
@@ -2608,6 +2653,19 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_zero_blob, 0, o->m_zero_bytes);
     hipMemset(o->d_emitted, 0, 8);
     hipStreamCreate(&o->stream);
+    o->own_fstream = 1;
+    if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
+        o->own_fstream = atoi(ev) != 0;
+    if (o->own_fstream) {
+        hipStreamCreate(&o->fstream);
+        hipEventCreateWithFlags(&o->ev_gate, hipEventDisableTiming);
+        for (int i = 0; i < 4; i++)
+            hipEventCreateWithFlags(&o->ev_tail[i], hipEventDisableTiming);
+    } else {
+        o->fstream = o->stream;
+    }
+    o->fire_group = 0;
+    o->retired_bin.assign(o->ring.R, EMPTY_TAG);
     /* pinned staging: 1M rows; enough columns for input batches and for
      * restore's raw-state batches */
     o->stg.cap = 1 << 20;
@@ -3112,6 +3170,10 @@ static int sync_open_panes(GpuOp *o) {
     o->open.clear();
     for (uint32_t s = 0; s < o->ring.R; s++) {
         if (tags[s] == EMPTY_TAG) continue;
+        /* a retire launched on the fire stream may not be visible in this
+         * snapshot yet; its bin can never be re-claimed (late-dropped) */
+        if (s < o->retired_bin.size() && tags[s] == o->retired_bin[s])
+            continue;
         bool is_closed = false;
         auto it = o->closed.find(tags[s]);
         if (it != o->closed.end() && it->second == s) is_closed = true;
@@ -3129,7 +3191,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
      * atomics (see k_merge_fused); legacy merge+compact otherwise */
     if (!raw_states && na <= MF_MAX_AGGS && o->ring.C >= MF_RANGE &&
         src.size() <= 64) {
-        HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+        HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->fstream));
         if (!src.empty()) {
             MergeFusedArgs M = {};
             M.ring = o->ring;
@@ -3186,19 +3248,19 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 if (mfs >= 2048)
                     hipLaunchKernelGGL((k_merge_fused<2048, true>),
                                        dim3(o->ring.C / range), dim3(256),
-                                       shmem, o->stream, M);
+                                       shmem, o->fstream, M);
                 else
                     hipLaunchKernelGGL((k_merge_fused<1024, true>),
                                        dim3(o->ring.C / range), dim3(256),
-                                       shmem, o->stream, M);
+                                       shmem, o->fstream, M);
             } else if (mfs >= 2048)
                 hipLaunchKernelGGL(k_merge_fused<2048>,
                                    dim3(o->ring.C / range), dim3(256),
-                                   shmem, o->stream, M);
+                                   shmem, o->fstream, M);
             else
                 hipLaunchKernelGGL(k_merge_fused<1024>,
                                    dim3(o->ring.C / range), dim3(256),
-                                   shmem, o->stream, M);
+                                   shmem, o->fstream, M);
             HIP_CHECK(o, hipGetLastError());
             if (o->mk) {
                 DictDecArgs DD = {};
@@ -3208,15 +3270,15 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 DD.nk = o->cfg.n_keys;
                 for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = o->d_out[k];
                 hipLaunchKernelGGL(k_dict_decode, dim3(1024), dim3(256), 0,
-                                   o->stream, DD);
+                                   o->fstream, DD);
                 HIP_CHECK(o, hipGetLastError());
             }
         }
         if (!o->cfg.emit_to_host) return 0;
         unsigned long long n = 0;
         HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8,
-                                    hipMemcpyDeviceToHost, o->stream));
-        HIP_CHECK(o, hipStreamSynchronize(o->stream));
+                                    hipMemcpyDeviceToHost, o->fstream));
+        HIP_CHECK(o, hipStreamSynchronize(o->fstream));
         if (n == 0) return 0;
         if ((size_t)o->out_cols > o->host_out.size())
             o->host_out.resize(o->out_cols);
@@ -3225,9 +3287,9 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             o->host_out[i].resize(old + n);
             HIP_CHECK(o, hipMemcpyAsync(o->host_out[i].data() + old,
                                         o->d_out[i], n * 8,
-                                        hipMemcpyDeviceToHost, o->stream));
+                                        hipMemcpyDeviceToHost, o->fstream));
         }
-        HIP_CHECK(o, hipStreamSynchronize(o->stream));
+        HIP_CHECK(o, hipStreamSynchronize(o->fstream));
         return 0;
     }
     HIP_CHECK(o, hipMemsetAsync(o->m_keys, 0xFF, (size_t)o->CM * 8, o->stream));
@@ -3292,13 +3354,16 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
         hipLaunchKernelGGL(k_accum, dim3(1), dim3(1), 0, o->stream,
                            o->d_emitted, o->d_n_out);
         HIP_CHECK(o, hipGetLastError());
-        return 0;
+        /* legacy merges run on the main stream but share d_out/d_n_out
+         * and panes with fire-stream work: re-gate so later fire-stream
+         * launches order after them */
+        return gate_fire(o);
     }
     unsigned long long n = 0;
     HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                                 o->stream));
     HIP_CHECK(o, hipStreamSynchronize(o->stream));
-    if (n == 0) return 0;
+    if (n == 0) return gate_fire(o);
     int ncols = raw_states ? 0 : o->out_cols;
     if (raw_states) {
         ncols = o->cfg.n_keys + 1;
@@ -3316,7 +3381,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
         }
         HIP_CHECK(o, hipStreamSynchronize(o->stream));
     }
-    return 0;
+    return gate_fire(o);
 }
 
 /* advance(): sliding_aggregating_window.rs:115-210, host replica */
@@ -3355,7 +3420,7 @@ static int advance(GpuOp *o) {
     if (fire_window(o, src, E - o->width, E, 0, 0)) return 1;
 
     for (uint64_t bb : dead) {
-        if (ring_retire(o, o->closed[bb])) return 1;
+        if (ring_retire(o, o->closed[bb], bb)) return 1;
         o->closed.erase(bb);
     }
 
@@ -3435,7 +3500,7 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
             uint32_t slot = o->open.begin()->second;
             std::vector<uint32_t> src = {slot};
             if (fire_window(o, src, b, b + o->width, 0, 0)) return 1;
-            if (ring_retire(o, slot)) return 1;
+            if (ring_retire(o, slot, b)) return 1;
             o->open.erase(o->open.begin());
         }
     } else {
@@ -3455,7 +3520,7 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
             if (kv.first + o->slide <= wb && kv.first < base)
                 unreachable.push_back(kv.first);
         for (uint64_t bb : unreachable) {
-            if (ring_retire(o, o->open[bb])) return 1;
+            if (ring_retire(o, o->open[bb], bb)) return 1;
             o->open.erase(bb);
         }
     }
@@ -3465,6 +3530,7 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
 /* fold the device status (error, min non-late bin, pane tags) into the
  * host state machine: ONE copy + sync */
 static int wm_fold_status(GpuOp *o) {
+    if (bound_fire_lag(o)) return 1;
     if (flush_staged(o)) return 1;
     if (check_device_error(o)) return 1;
     harvest_events(o);
@@ -3482,13 +3548,14 @@ static int wm_fold_status(GpuOp *o) {
             o->earliest = minb;
         }
     }
-    return 0;
+    return gate_fire(o);
 }
 
 API int arroyo_amd_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
     GpuOp *o = (GpuOp *)h;
     if (wm_fold_status(o)) return 1;
     if (wm_advance(o, wm)) return 1;
+    if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
 }
@@ -3504,6 +3571,7 @@ API int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms,
     if (wm_fold_status(o)) return 1;
     for (int32_t i = 0; i < n; i++)
         if (wm_advance(o, wms[i])) return 1;
+    if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
 }
@@ -3513,6 +3581,7 @@ API int arroyo_amd_checkpoint_drain(void *h, AmdOutBatch *out) {
      * drained into the state table; raw partial-state layout with the bin
      * as trailing timestamp column. */
     GpuOp *o = (GpuOp *)h;
+    HIP_CHECK(o, hipStreamSynchronize(o->fstream));
     if (flush_staged(o)) return 1;
     if (check_device_error(o)) return 1;
     if (sync_open_panes(o)) return 1;
@@ -3531,6 +3600,8 @@ API int arroyo_amd_restore(void *h, const int64_t *const *cols,
     /* on_start (sliding:556-595): state rows with bin < watermark bin go to
      * the tiered holder (closed panes); others re-open their pane. */
     GpuOp *o = (GpuOp *)h;
+    HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+    o->retired_bin.assign(o->ring.R, EMPTY_TAG);
     int swords_total = 0;
     int32_t swords[AMD_MAX_AGGS];
     for (int a = 0; a < o->agg.n_aggs; a++) {
@@ -3649,7 +3720,13 @@ API void arroyo_amd_destroy(void *h) {
     GpuOp *o = (GpuOp *)h;
     if (!o) return;
     hipStreamSynchronize(o->stream);
+    if (o->fstream != o->stream) hipStreamSynchronize(o->fstream);
     harvest_events(o);
+    if (o->fstream != o->stream) {
+        hipEventDestroy(o->ev_gate);
+        for (int i = 0; i < 4; i++) hipEventDestroy(o->ev_tail[i]);
+        hipStreamDestroy(o->fstream);
+    }
     for (auto &pr : o->ev_pool) {
         hipEventDestroy(pr.first);
         hipEventDestroy(pr.second);
@@ -3694,6 +3771,8 @@ API void arroyo_amd_destroy(void *h) {
 API int arroyo_amd_sync(void *h) {
     GpuOp *o = (GpuOp *)h;
     HIP_CHECK(o, hipStreamSynchronize(o->stream));
+    if (o->fstream != o->stream)
+        HIP_CHECK(o, hipStreamSynchronize(o->fstream));
     return 0;
 }
 
@@ -3713,9 +3792,9 @@ API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
     /* fold in the device-side fire accumulator (no per-fire round trips) */
     unsigned long long dev_emitted = 0;
     HIP_CHECK(o, hipMemcpyAsync(&dev_emitted, o->d_emitted, 8,
-                                hipMemcpyDeviceToHost, o->stream));
-    HIP_CHECK(o, hipStreamSynchronize(o->stream));
-    HIP_CHECK(o, hipMemsetAsync(o->d_emitted, 0, 8, o->stream));
+                                hipMemcpyDeviceToHost, o->fstream));
+    HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+    HIP_CHECK(o, hipMemsetAsync(o->d_emitted, 0, 8, o->fstream));
     *emitted_device_rows = o->emitted_device_rows + (int64_t)dev_emitted;
     o->update_kernel_ms = 0;
     o->sampled_launches = 0;
