@@ -52,13 +52,13 @@ EVENTS_PER_SEC = 1_000_000       # event-time rate of the synthetic stream
 # Infinity Cache (PMC FETCH_SIZE showed a 4-batch ring was fully L3-resident)
 BASE_BATCHES = 512
 WIDTH_S, SLIDE_S = 10, 2
-LOG2_CAPACITY = int(os.environ.get("BENCH_LOG2_CAP", "18"))
+LOG2_CAPACITY = int(os.environ.get("BENCH_LOG2_CAP", "19"))
 # ~130K distinct auctions per 2s pane -> default 2^18 slots (~50% load):
 # the round-2 sweep measured the smaller pane table worth +9% whole-job
 # (less scan/merge/retire/CPI traffic) with loud-error headroom intact
 # merge knobs: the 512-slot home range + 2048-slot LDS dedup table won the
 # round-2 sweep (these only seed defaults; explicit env wins)
-os.environ.setdefault("ARROYO_AMD_MF_RANGE", "512")
+os.environ.setdefault("ARROYO_AMD_MF_RANGE", "1024")
 os.environ.setdefault("ARROYO_AMD_MF_SLOTS", "2048")
 RING_PANES = 16
 HBM_PEAK_GBPS = 8000.0           # spec peak (MI355X_MICROARCH.md)
