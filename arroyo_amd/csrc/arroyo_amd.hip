@@ -3560,7 +3560,6 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
 /* fold the device status (error, min non-late bin, pane tags) into the
  * host state machine: ONE copy + sync */
 static int wm_fold_status(GpuOp *o) {
-    if (bound_fire_lag(o)) return 1;
     if (flush_staged(o)) return 1;
     if (check_device_error(o)) return 1;
     harvest_events(o);
@@ -3578,6 +3577,11 @@ static int wm_fold_status(GpuOp *o) {
             o->earliest = minb;
         }
     }
+    /* the lag wait goes AFTER the blocking status sync above: it only
+     * has to precede the NEXT groups' update launches -- enqueued before
+     * the copy it would stall the host on the whole previous fire group
+     * (measured -25% whole-job) */
+    if (bound_fire_lag(o)) return 1;
     return gate_fire(o);
 }
 
