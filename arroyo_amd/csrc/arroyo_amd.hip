@@ -1384,36 +1384,29 @@ k_update_batch_scalar(UpdateArgs A) {
     fold_min_bin(local_min, A.ring.min_bin);
 }
 
-/* grid-wide completion protocol shared by the retire kernels: the pane's
- * tag must become EMPTY only after EVERY block's plane clears are
- * agent-visible, because with the fire stream overlapping updates a
+/* The pane's tag must become EMPTY only after EVERY block's plane clears
+ * are agent-visible: with the fire stream overlapping updates, a
  * concurrent update kernel may claim the slot the instant the tag reads
- * empty (a tag cleared early would hand it a half-cleared table --
- * silent corruption; cleared last, the worst interleaving is a failed
- * CAS and a loud ERR_RING_CONFLICT). */
-__device__ inline void retire_epilogue(unsigned int *ctr, uint64_t *tag,
-                                       uint32_t *spec_used,
-                                       uint64_t *spec_state, int na2) {
-    __syncthreads();
-    if (threadIdx.x != 0) return;
-    __threadfence();                       /* release this block's clears */
-    unsigned int done = atomicAdd(ctr, 1u);
-    if (done == gridDim.x - 1) {
-        *ctr = 0;                          /* rearm for the next retire */
-        *spec_used = 0;
-        for (int w = 0; w < na2; w++) spec_state[w] = 0;
-        __hip_atomic_store((unsigned long long *)tag,
-                           (unsigned long long)EMPTY_TAG, __ATOMIC_RELEASE,
-                           __HIP_MEMORY_SCOPE_AGENT);
-    }
+ * empty, and a tag cleared early would hand it a half-cleared table
+ * (silent corruption).  The plane-clear kernels therefore do NOT touch
+ * the tag; k_retire_tag follows them on the same stream, and the kernel
+ * boundary is the grid-wide completion + release point.  (A grid-wide
+ * done counter was measured instead: 1024 same-address returning atomics
+ * serialize at ~0.3 us each -- it turned a ~5 us retire into a fire-
+ * stream-dominating one and cost 25% whole-job.) */
+__global__ void k_retire_tag(uint64_t *tag, uint32_t *spec_used,
+                             uint64_t *spec_state, int na2) {
+    *spec_used = 0;
+    for (int w = 0; w < na2; w++) spec_state[w] = 0;
+    __hip_atomic_store((unsigned long long *)tag,
+                       (unsigned long long)EMPTY_TAG, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_AGENT);
 }
 
 /* packed pane clear: {EMPTY_KEY, 0} per slot (a single memset cannot set
  * the two words differently, and count must start at 0) */
 __global__ void __launch_bounds__(256)
-k_retire_packed(uint64_t *slots, int64_t n_slots, uint64_t *tag,
-                uint32_t *spec_used, uint64_t *spec_state, int na2,
-                unsigned int *ctr) {
+k_retire_packed(uint64_t *slots, int64_t n_slots) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < n_slots; i += stride) {
@@ -1422,16 +1415,13 @@ k_retire_packed(uint64_t *slots, int64_t n_slots, uint64_t *tag,
         v.y = 0;
         ((ulonglong2 *)slots)[i] = v;
     }
-    if (tag) retire_epilogue(ctr, tag, spec_used, spec_state, na2);
 }
 
 /* one-launch pane retire for the split/unpacked layout: keys to 0xFF,
  * states to 0, scalar metadata — replaces two fill launches plus a
  * metadata kernel (~5 us each on a busy stream) per retire */
 __global__ void __launch_bounds__(256)
-k_retire_all(int64_t *keys, uint64_t *state, int64_t C, int na2,
-             uint64_t *tag, uint32_t *spec_used, uint64_t *spec_state,
-             unsigned int *ctr) {
+k_retire_all(int64_t *keys, uint64_t *state, int64_t C, int na2) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t kv2 = C >> 1;          /* C is a power of two >= 2 */
     ulonglong2 ff, zz;
@@ -1444,7 +1434,6 @@ k_retire_all(int64_t *keys, uint64_t *state, int64_t C, int na2,
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < sv2; i += stride)
         ((ulonglong2 *)state)[i] = zz;
-    retire_epilogue(ctr, tag, spec_used, spec_state, na2);
 }
 
 /* ------------------------------------------------------------------ */
@@ -2316,8 +2305,6 @@ struct GpuOp {
     int n_out_alloc;
     unsigned long long *d_n_out;
     unsigned long long *d_emitted;  /* running device-side emitted rows */
-    unsigned int *d_retire_ctr;     /* retire grid-done counter (see
-                                       retire_epilogue) */
     int64_t out_rows_cap;
 
     /* host-accumulated emission (emit_to_host) */
@@ -2464,20 +2451,19 @@ static int ring_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
         hipLaunchKernelGGL(k_retire_packed, dim3(blocks), dim3(256), 0,
                            o->fstream,
                            o->ring.slots + (size_t)slot * o->ring.C * 2,
-                           (int64_t)o->ring.C, o->ring.tag + slot,
-                           o->ring.spec_used + slot,
-                           o->ring.spec_state + (size_t)slot * na * 2,
-                           (int)(na * 2), o->d_retire_ctr);
+                           (int64_t)o->ring.C);
     } else {
         hipLaunchKernelGGL(k_retire_all, dim3(blocks), dim3(256), 0,
                            o->fstream,
                            o->ring.keys + (size_t)slot * o->ring.C,
                            o->ring.state + (size_t)slot * o->ring.C * na * 2,
-                           (int64_t)o->ring.C, (int)(na * 2),
-                           o->ring.tag + slot, o->ring.spec_used + slot,
-                           o->ring.spec_state + (size_t)slot * na * 2,
-                           o->d_retire_ctr);
+                           (int64_t)o->ring.C, (int)(na * 2));
     }
+    /* tag LAST, after the plane clears, via stream order */
+    hipLaunchKernelGGL(k_retire_tag, dim3(1), dim3(1), 0, o->fstream,
+                       o->ring.tag + slot, o->ring.spec_used + slot,
+                       o->ring.spec_state + (size_t)slot * na * 2,
+                       (int)(na * 2));
     HIP_CHECK(o, hipGetLastError());
     return 0;
 }
@@ -2630,7 +2616,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
-    ALLOC(o->d_retire_ctr, 4);
     /* fused-merge home range fixed at create (CPI grouping depends on it) */
     o->mf_range = MF_RANGE;
     if (const char *ev = getenv("ARROYO_AMD_MF_RANGE"))
@@ -2666,9 +2651,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (o->ring.packed) {
         int64_t n_slots = (int64_t)o->ring.R * o->ring.C;
         hipLaunchKernelGGL(k_retire_packed, dim3(1024), dim3(256), 0, 0,
-                           o->ring.slots, n_slots, (uint64_t *)nullptr,
-                           (uint32_t *)nullptr, (uint64_t *)nullptr, 0,
-                           (unsigned int *)nullptr);
+                           o->ring.slots, n_slots);
     } else {
         hipMemset(o->ring.keys, 0xFF, (size_t)o->ring.R * o->ring.C * 8);
         hipMemset(o->ring.state, 0, (size_t)o->ring.R * o->ring.C * na * 16);
@@ -2681,7 +2664,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_keys, 0xFF, (size_t)o->CM * 8);
     hipMemset(o->m_zero_blob, 0, o->m_zero_bytes);
     hipMemset(o->d_emitted, 0, 8);
-    hipMemset(o->d_retire_ctr, 0, 4);
     hipStreamCreate(&o->stream);
     o->own_fstream = 1;
     if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
@@ -3776,7 +3758,6 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->m_zero_blob);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
     hipFree(o->d_emitted);
-    hipFree(o->d_retire_ctr);
     hipFree(o->rdx_key);
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
