@@ -2326,6 +2326,8 @@ struct GpuOp {
                                  a ring slot's retire always completes
                                  before the slot's bin comes around again */
     uint64_t fire_group;
+    int fire_lag;             /* groups of slack before the lag wait (<=3;
+                                 reuse distance R/bins-per-group >> this) */
     std::vector<uint64_t> retired_bin;  /* [R] bin whose async retire may
                                            not yet show in d_status */
     Staged stg;
@@ -2435,9 +2437,11 @@ static int fire_tail(GpuOp *o) {
  * measurably raced at ring_panes=16 with large pane memsets: loud
  * ERR_RING_CONFLICT, never corruption). */
 static int bound_fire_lag(GpuOp *o) {
-    if (o->fstream == o->stream || o->fire_group < 1) return 0;
+    if (o->fstream == o->stream ||
+        o->fire_group < (uint64_t)o->fire_lag) return 0;
     HIP_CHECK(o, hipStreamWaitEvent(
-                     o->stream, o->ev_tail[(o->fire_group - 1) & 3], 0));
+                     o->stream,
+                     o->ev_tail[(o->fire_group - o->fire_lag) & 3], 0));
     return 0;
 }
 
@@ -2677,6 +2681,16 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         o->fstream = o->stream;
     }
     o->fire_group = 0;
+    /* 2 groups of slack won the sweep (lag 1 throttles updates behind a
+     * slow fire group, -6%; the reuse distance R/bins-per-group leaves
+     * >= 2 periods of margin at the bench shapes, and a violated margin
+     * is a loud ring-conflict, not corruption) */
+    o->fire_lag = 2;
+    if (const char *ev = getenv("ARROYO_AMD_FIRE_LAG")) {
+        o->fire_lag = atoi(ev);
+        if (o->fire_lag < 1) o->fire_lag = 1;
+        if (o->fire_lag > 3) o->fire_lag = 3;
+    }
     o->retired_bin.assign(o->ring.R, EMPTY_TAG);
     /* pinned staging: 1M rows; enough columns for input batches and for
      * restore's raw-state batches */
