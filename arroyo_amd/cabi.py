@@ -90,6 +90,19 @@ def bind(lib, prefix):
         fn["handle_watermarks"].argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint64), ctypes.c_int32,
             ctypes.POINTER(AmdOutBatch)]
+        # epoch pipelining (HIP library only)
+        fn["set_filter_watermark"] = g("set_filter_watermark")
+        fn["set_filter_watermark"].restype = ctypes.c_int
+        fn["set_filter_watermark"].argtypes = [ctypes.c_void_p,
+                                               ctypes.c_uint64]
+        fn["mark_epoch"] = g("mark_epoch")
+        fn["mark_epoch"].restype = ctypes.c_int
+        fn["mark_epoch"].argtypes = [ctypes.c_void_p]
+        fn["handle_watermarks_epoch"] = g("handle_watermarks_epoch")
+        fn["handle_watermarks_epoch"].restype = ctypes.c_int
+        fn["handle_watermarks_epoch"].argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint64), ctypes.c_int32,
+            ctypes.POINTER(AmdOutBatch)]
     except AttributeError:
         pass
     fn["checkpoint_drain"] = g("checkpoint_drain")
@@ -321,6 +334,27 @@ class WindowOp:
         arr = (ctypes.c_uint64 * len(wms))(*wms)
         rc = self._fn["handle_watermarks"](self._h, arr, len(wms),
                                            ctypes.byref(out))
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+        cols = _out_to_numpy(out)
+        self._fn["free_out"](ctypes.byref(out))
+        return cols
+
+    def set_filter_watermark(self, wm):
+        rc = self._fn["set_filter_watermark"](self._h, wm)
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def mark_epoch(self):
+        rc = self._fn["mark_epoch"](self._h)
+        if rc != 0:
+            raise RuntimeError(self._fn["last_error"](self._h).decode())
+
+    def handle_watermarks_epoch(self, wms):
+        out = AmdOutBatch()
+        arr = (ctypes.c_uint64 * len(wms))(*wms)
+        rc = self._fn["handle_watermarks_epoch"](self._h, arr, len(wms),
+                                                 ctypes.byref(out))
         if rc != 0:
             raise RuntimeError(self._fn["last_error"](self._h).decode())
         cols = _out_to_numpy(out)

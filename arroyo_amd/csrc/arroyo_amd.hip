@@ -2326,6 +2326,14 @@ struct GpuOp {
                                  a ring slot's retire always completes
                                  before the slot's bin comes around again */
     uint64_t fire_group;
+    int spin;                 /* busy-wait host syncs (ARROYO_AMD_SPIN) */
+    hipEvent_t ev_sync;       /* status-copy completion (spin target) */
+    /* epoch pipeline (mark_epoch / handle_watermarks_epoch): per-epoch
+     * pinned status snapshots so the harness can submit the NEXT period
+     * before folding this one's watermarks */
+    uint64_t *h_epoch[2];
+    hipEvent_t ev_epoch[2];
+    int epoch_head, epoch_cnt;
     int fire_lag;             /* groups of slack before the lag wait (<=3;
                                  reuse distance R/bins-per-group >> this) */
     std::vector<uint64_t> retired_bin;  /* [R] bin whose async retire may
@@ -2396,8 +2404,17 @@ struct GpuOp {
 
 #define EV_POOL 64
 
-static void harvest_events(GpuOp *o) {
+/* fold finished sampled-launch timings; block=0 leaves unfinished pairs
+ * inflight (the epoch-pipelined fold must not wait on the NEXT period's
+ * update kernels), block=1 (perf/destroy) drains everything */
+static void harvest_events(GpuOp *o, int block) {
+    std::vector<int> keep;
     for (int idx : o->ev_inflight) {
+        if (!block &&
+            hipEventQuery(o->ev_pool[idx].second) == hipErrorNotReady) {
+            keep.push_back(idx);
+            continue;
+        }
         hipEventSynchronize(o->ev_pool[idx].second);
         float ms = 0;
         hipEventElapsedTime(&ms, o->ev_pool[idx].first,
@@ -2405,7 +2422,8 @@ static void harvest_events(GpuOp *o) {
         o->update_kernel_ms += ms;
         o->sampled_launches++;
     }
-    o->ev_inflight.clear();
+    (void)hipGetLastError();   /* clear NotReady sticky */
+    o->ev_inflight = std::move(keep);
 }
 
 static thread_local char g_err[512];
@@ -2681,6 +2699,15 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         o->fstream = o->stream;
     }
     o->fire_group = 0;
+    o->spin = 1;
+    if (const char *ev = getenv("ARROYO_AMD_SPIN")) o->spin = atoi(ev) != 0;
+    hipEventCreateWithFlags(&o->ev_sync, hipEventDisableTiming);
+    for (int i = 0; i < 2; i++) {
+        hipHostMalloc((void **)&o->h_epoch[i],
+                      (2 + (size_t)o->ring.R) * 8);
+        hipEventCreateWithFlags(&o->ev_epoch[i], hipEventDisableTiming);
+    }
+    o->epoch_head = o->epoch_cnt = 0;
     /* 2 groups of slack won the sweep (lag 1 throttles updates behind a
      * slow fire group, -6%; the reuse distance R/bins-per-group leaves
      * >= 2 periods of margin at the bench shapes, and a violated margin
@@ -2813,7 +2840,7 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
             hipEventCreate(&pr.second);
         }
     }
-    if (sample && o->ev_inflight.size() >= EV_POOL) harvest_events(o);
+    if (sample && o->ev_inflight.size() >= EV_POOL) harvest_events(o, 1);
     int ev = -1;
     if (sample) {
         /* find a free pair: pool minus inflight (inflight cleared on
@@ -3146,13 +3173,38 @@ API int arroyo_amd_process_batches_device(void *h,
     return 0;
 }
 
+/* busy-wait on an event: the interrupt-based hipEventSynchronize costs
+ * 10-20 us of wake-up latency per call, on the watermark critical path */
+static int wait_event_spin(GpuOp *o, hipEvent_t ev) {
+    if (!o->spin) {
+        HIP_CHECK(o, hipEventSynchronize(ev));
+        return 0;
+    }
+    hipError_t e;
+    while ((e = hipEventQuery(ev)) == hipErrorNotReady) {}
+    (void)hipGetLastError();   /* clear the NotReady sticky */
+    if (e != hipSuccess) {
+        snprintf(o->err_msg, sizeof o->err_msg, "event wait: %s",
+                 hipGetErrorString(e));
+        return 1;
+    }
+    return 0;
+}
+
+static int decode_status_error(GpuOp *o, const uint64_t *status);
+
 /* one round trip: pulls err + min_bin + pane tags into h_status */
 static int check_device_error(GpuOp *o) {
     HIP_CHECK(o, hipMemcpyAsync(o->h_status, o->d_status,
                                 (2 + (size_t)o->ring.R) * 8,
                                 hipMemcpyDeviceToHost, o->stream));
-    HIP_CHECK(o, hipStreamSynchronize(o->stream));
-    int e = *(int *)o->h_status;
+    HIP_CHECK(o, hipEventRecord(o->ev_sync, o->stream));
+    if (wait_event_spin(o, o->ev_sync)) return 1;
+    return decode_status_error(o, o->h_status);
+}
+
+static int decode_status_error(GpuOp *o, const uint64_t *status) {
+    int e = *(const int *)status;
     if (e == ERR_RING_CONFLICT) {
         snprintf(o->err_msg, sizeof o->err_msg,
                  "pane ring conflict: more than ring_panes=%u live bins; "
@@ -3558,7 +3610,7 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
 static int wm_fold_status(GpuOp *o) {
     if (flush_staged(o)) return 1;
     if (check_device_error(o)) return 1;
-    harvest_events(o);
+    harvest_events(o, 0);
     if (sync_open_panes(o)) return 1;
     /* the OnlyBufferedData `earliest` only advances while not
      * InMemoryData; arrivals during InMemoryData are intentionally not
@@ -3599,6 +3651,88 @@ API int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms,
     GpuOp *o = (GpuOp *)h;
     if (n <= 0) return 0;
     if (wm_fold_status(o)) return 1;
+    for (int32_t i = 0; i < n; i++)
+        if (wm_advance(o, wms[i])) return 1;
+    if (fire_tail(o)) return 1;
+    if (out) return build_out(o, out, 0);
+    return 0;
+}
+
+/* ---- epoch pipelining -------------------------------------------------
+ * The sequential order is submit(g) -> fold+fire(g) -> submit(g+1): the
+ * host blocks in the fold until period g's update kernels finish, leaving
+ * the main stream idle.  mark_epoch snapshots the device status (pane
+ * tags + min non-late bin) on the stream right after period g's launches,
+ * so the harness may submit period g+1 BEFORE folding g's watermarks:
+ *
+ *   submit(g); mark_epoch(); set_filter_watermark(last wm of g);
+ *   submit(g+1); handle_watermarks_epoch(g's wms);
+ *
+ * set_filter_watermark pre-advances the ingest late-drop cutoff to the
+ * value the deferred watermarks will establish, so period g+1's rows are
+ * filtered EXACTLY as in the sequential order (without it, a row late
+ * under g's watermark could land in a pane g's fires read concurrently).
+ * Depth 2; emissions and state are bit-identical to the sequential calls
+ * (tests/test_gpu_parity.py::test_epoch_pipelined_equals_sequential). */
+
+API int arroyo_amd_set_filter_watermark(void *h, uint64_t wm) {
+    GpuOp *o = (GpuOp *)h;
+    o->has_wm = 1;
+    o->wm = wm;
+    return 0;
+}
+
+API int arroyo_amd_mark_epoch(void *h) {
+    GpuOp *o = (GpuOp *)h;
+    if (o->epoch_cnt >= 2) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "epoch queue full (depth 2): fold the oldest epoch with "
+                 "handle_watermarks_epoch first");
+        return 1;
+    }
+    if (flush_staged(o)) return 1;
+    int e = (o->epoch_head + o->epoch_cnt) & 1;
+    HIP_CHECK(o, hipMemcpyAsync(o->h_epoch[e], o->d_status,
+                                (2 + (size_t)o->ring.R) * 8,
+                                hipMemcpyDeviceToHost, o->stream));
+    /* reset AFTER the snapshot captured it (stream order) */
+    HIP_CHECK(o, hipMemsetAsync(o->ring.min_bin, 0xFF, 8, o->stream));
+    HIP_CHECK(o, hipEventRecord(o->ev_epoch[e], o->stream));
+    o->epoch_cnt++;
+    return 0;
+}
+
+API int arroyo_amd_handle_watermarks_epoch(void *h, const uint64_t *wms,
+                                           int32_t n, AmdOutBatch *out) {
+    GpuOp *o = (GpuOp *)h;
+    if (o->epoch_cnt == 0) {
+        snprintf(o->err_msg, sizeof o->err_msg,
+                 "no armed epoch: call mark_epoch after the period's rows");
+        return 1;
+    }
+    int e = o->epoch_head;
+    o->epoch_head ^= 1;
+    o->epoch_cnt--;
+    if (wait_event_spin(o, o->ev_epoch[e])) return 1;
+    if (decode_status_error(o, o->h_epoch[e])) return 1;
+    harvest_events(o, 0);
+    /* fold from the epoch's snapshot */
+    memcpy(o->h_status, o->h_epoch[e], (2 + (size_t)o->ring.R) * 8);
+    if (sync_open_panes(o)) return 1;
+    uint64_t minb = o->h_status[1];
+    if (!o->cfg.is_tumbling && minb != ~0ULL) {
+        if (o->state == 0) {
+            o->state = 1;
+            o->earliest = minb;
+        } else if (o->state == 1 && minb < o->earliest) {
+            o->earliest = minb;
+        }
+    }
+    if (bound_fire_lag(o)) return 1;
+    /* gate this epoch's fires on ITS stream point (later periods' update
+     * launches are after it and not waited on) */
+    if (o->fstream != o->stream)
+        HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_epoch[e], 0));
     for (int32_t i = 0; i < n; i++)
         if (wm_advance(o, wms[i])) return 1;
     if (fire_tail(o)) return 1;
@@ -3751,7 +3885,12 @@ API void arroyo_amd_destroy(void *h) {
     if (!o) return;
     hipStreamSynchronize(o->stream);
     if (o->fstream != o->stream) hipStreamSynchronize(o->fstream);
-    harvest_events(o);
+    harvest_events(o, 1);
+    hipEventDestroy(o->ev_sync);
+    for (int i = 0; i < 2; i++) {
+        hipHostFree(o->h_epoch[i]);
+        hipEventDestroy(o->ev_epoch[i]);
+    }
     if (o->fstream != o->stream) {
         hipEventDestroy(o->ev_gate);
         for (int i = 0; i < 4; i++) hipEventDestroy(o->ev_tail[i]);
@@ -3810,7 +3949,7 @@ API int arroyo_amd_sync(void *h) {
 API int arroyo_amd_perf(void *h, double *update_ms, int64_t *rows,
                         int64_t *launches, int64_t *emitted_device_rows) {
     GpuOp *o = (GpuOp *)h;
-    harvest_events(o);
+    harvest_events(o, 1);
     /* update_ms is extrapolated from the sampled launches so that
      * update_ms / launches equals the sampled per-launch average */
     *update_ms = o->sampled_launches

@@ -65,6 +65,11 @@ def lib():
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_int]
         _lib.arroyo_amd_sync.restype = ctypes.c_int
         _lib.arroyo_amd_sync.argtypes = [ctypes.c_void_p]
+        _lib.arroyo_amd_set_filter_watermark.restype = ctypes.c_int
+        _lib.arroyo_amd_set_filter_watermark.argtypes = [
+            ctypes.c_void_p, ctypes.c_uint64]
+        _lib.arroyo_amd_mark_epoch.restype = ctypes.c_int
+        _lib.arroyo_amd_mark_epoch.argtypes = [ctypes.c_void_p]
         _lib.arroyo_amd_partition.restype = ctypes.c_int
         _lib.arroyo_amd_partition.argtypes = [ctypes.c_void_p] * 3 + [
             ctypes.c_int64, ctypes.c_uint32] + [ctypes.c_void_p] * 3 + [

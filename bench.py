@@ -172,6 +172,21 @@ def run_gpu(args):
     # ~7 us fixed cost (ramp + launch latency) of the update kernel
     # (round-2 sweeps: optimum 4 for the batched kernel).
     wm_fuse = int(os.environ.get("BENCH_WM_FUSE", "4"))
+    # Epoch pipelining (BENCH_PIPELINE=0 restores the sequential order):
+    # submit period g+1 before folding period g's watermarks, so the fold's
+    # host latency hides behind g+1's kernels.  set_filter_watermark keeps
+    # the late-drop cutoff exactly as in the sequential order (the stream
+    # here is time-monotone anyway); emissions are bit-identical
+    # (tests/test_gpu_parity.py::test_epoch_pipelined_equals_sequential).
+    pipelined = os.environ.get("BENCH_PIPELINE", "1") != "0"
+
+    def emit_watermarks_epoch(steps):
+        vals = (ctypes.c_uint64 * len(steps))(*[wm_value(s) for s in steps])
+        rc = wm_lib.arroyo_amd_handle_watermarks_epoch(
+            op._h, vals, ctypes.c_int32(len(steps)), None)
+        if rc != 0:
+            raise RuntimeError(op._fn["last_error"](op._h).decode())
+        wm_fires[0] += len(steps)
 
     def run_span(s_begin, n_steps):
         """single-GPU fast path: multi-batch submits split at fused
@@ -193,11 +208,28 @@ def run_gpu(args):
             # emit every watermark boundary crossed so far, in order
             first_unemitted = (run_span.last_wm // wm_every + 1) * wm_every
             group = list(range(first_unemitted, s + 1, wm_every))
-            if group:
+            if not group:
+                continue
+            if pipelined:
+                rc = wm_lib.arroyo_amd_mark_epoch(op._h)
+                if rc != 0:
+                    raise RuntimeError(op._fn["last_error"](op._h).decode())
+                wm_lib.arroyo_amd_set_filter_watermark(
+                    op._h, ctypes.c_uint64(wm_value(group[-1])))
+                if run_span.pending:
+                    emit_watermarks_epoch(run_span.pending)
+                run_span.pending = group
+            else:
                 emit_watermarks(group)
-                run_span.last_wm = group[-1]
+            run_span.last_wm = group[-1]
+
+    def drain_pending():
+        if run_span.pending:
+            emit_watermarks_epoch(run_span.pending)
+            run_span.pending = None
 
     run_span.last_wm = 0
+    run_span.pending = None
 
     def one_step(step):
         b = step % BASE_BATCHES
@@ -298,6 +330,8 @@ def run_gpu(args):
             break
         chunk = max(chunk, timed_steps)   # double the region each pass
 
+    if world == 1:
+        drain_pending()
     perf = op.perf()
     perf["wm_fires_timed"] = wm_fires[0]
     op.close()

@@ -99,6 +99,21 @@ int arroyo_amd_handle_watermark(void *h, uint64_t watermark_nanos,
 int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms, int32_t n,
                                  AmdOutBatch *out);
 
+/* Epoch pipelining: lets a harness submit the NEXT period's batches
+ * before folding this period's watermarks, hiding the fold's host
+ * latency behind the next period's kernels.  Sequence:
+ *   submit(g); mark_epoch(); set_filter_watermark(last wm of group g);
+ *   submit(g+1); handle_watermarks_epoch(group g's wms, out);
+ * mark_epoch snapshots the device status at period g's stream point
+ * (depth 2).  set_filter_watermark pre-advances the ingest late-drop
+ * cutoff to the watermark the deferred group will establish, so the
+ * next period's rows are filtered exactly as in the sequential order;
+ * emissions are bit-identical to sequential handle_watermark calls. */
+int arroyo_amd_set_filter_watermark(void *h, uint64_t watermark_nanos);
+int arroyo_amd_mark_epoch(void *h);
+int arroyo_amd_handle_watermarks_epoch(void *h, const uint64_t *wms,
+                                       int32_t n, AmdOutBatch *out);
+
 /* Drain open panes' partial states for a checkpoint barrier.  Columns:
  * [key?, partial state words (AVG takes 2)..., bin _timestamp]. */
 int arroyo_amd_checkpoint_drain(void *h, AmdOutBatch *out);

@@ -334,3 +334,41 @@ def test_batched_watermarks_equal_sequential():
     got = concat_outputs(bat)
     want = concat_outputs(seq)
     assert_parity(got, want)
+
+
+def test_epoch_pipelined_equals_sequential():
+    """mark_epoch / set_filter_watermark / handle_watermarks_epoch with the
+    next period's rows submitted BEFORE the fold must be emission-identical
+    to the sequential order (include/arroyo_amd.h epoch-pipelining doc)."""
+    cols = nexmark.bids(600_000, events_per_sec=100_000, seed=29,
+                        with_price=True)
+    kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=1,
+              aggs=[(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.AVG, 0)],
+              log2_capacity=15, ring_panes=32)
+    a, b = gpu_op(**kw), gpu_op(**kw)
+    batches = batches_from_columns(cols, 60_000)
+    seq, pip = [], []
+    pending = None
+    for i, cb in enumerate(batches):
+        wm = int(cb[-1][-1]) - NS
+        # sequential reference
+        a.process_batch(cb)
+        out = a.handle_watermark(wm)
+        if out and len(out[0]):
+            seq.append(out)
+        # pipelined: submit THEN fold the previous epoch
+        b.process_batch(cb)
+        b.mark_epoch()
+        b.set_filter_watermark(wm)
+        if pending is not None:
+            out = b.handle_watermarks_epoch([pending])
+            if out and len(out[0]):
+                pip.append(out)
+        pending = wm
+    if pending is not None:
+        out = b.handle_watermarks_epoch([pending])
+        if out and len(out[0]):
+            pip.append(out)
+    a.close()
+    b.close()
+    assert_parity(concat_outputs(pip), concat_outputs(seq), f64_idx=(3,))
