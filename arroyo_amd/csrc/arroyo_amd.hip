@@ -1394,13 +1394,25 @@ k_update_batch_scalar(UpdateArgs A) {
  * done counter was measured instead: 1024 same-address returning atomics
  * serialize at ~0.3 us each -- it turned a ~5 us retire into a fire-
  * stream-dominating one and cost 25% whole-job.) */
-__global__ void k_retire_tag(uint64_t *tag, uint32_t *spec_used,
-                             uint64_t *spec_state, int na2) {
-    *spec_used = 0;
-    for (int w = 0; w < na2; w++) spec_state[w] = 0;
-    __hip_atomic_store((unsigned long long *)tag,
-                       (unsigned long long)EMPTY_TAG, __ATOMIC_RELEASE,
-                       __HIP_MEMORY_SCOPE_AGENT);
+struct RetireTagsArgs {
+    uint64_t *tag;            /* [R] */
+    uint32_t *spec_used;      /* [R] */
+    uint64_t *spec_state;     /* [R][na2] */
+    int na2;
+    int n;
+    uint32_t slot[8];
+};
+
+__global__ void k_retire_tags(RetireTagsArgs A) {
+    for (int i = 0; i < A.n; i++) {
+        uint32_t p = A.slot[i];
+        A.spec_used[p] = 0;
+        for (int w = 0; w < A.na2; w++)
+            A.spec_state[(size_t)p * A.na2 + w] = 0;
+        __hip_atomic_store((unsigned long long *)(A.tag + p),
+                           (unsigned long long)EMPTY_TAG, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    }
 }
 
 /* packed pane clear: {EMPTY_KEY, 0} per slot (a single memset cannot set
@@ -1901,7 +1913,9 @@ struct MergeFusedArgs {
      * sparse table + displacement overscan (the fire path was 52% of
      * GPU time on the sparse scan).  use_cpi covers n_src <= 16. */
     int32_t use_cpi;
-    const uint64_t *cpi_entries[16];   /* [nr][range][1 + 2*na] */
+    int32_t cpi_ew;   /* entry words: 1+1 when only word0 of a single agg
+                         is live (COUNT/SUM/MIN/MAX), else 1+2*na */
+    const uint64_t *cpi_entries[16];   /* [nr][range][cpi_ew] */
     const uint32_t *cpi_cnt[16];       /* [nr] entries per range */
 };
 
@@ -1918,6 +1932,7 @@ struct CpiBuildArgs {
     const int64_t *keys;      /* pane planes in the ring */
     const uint64_t *state;
     uint32_t C;
+    int32_t ew;               /* entry words (see MergeFusedArgs.cpi_ew) */
     int32_t na;
     uint32_t range;
     uint32_t nr;              /* C / range */
@@ -1934,7 +1949,7 @@ k_cpi_build(CpiBuildArgs A) {
     const uint32_t mask = A.C - 1;
     const uint32_t a = blockIdx.x * A.range;
     const uint32_t span = A.range + MAX_PROBES;
-    const int ew = 1 + 2 * A.na;
+    const int ew = A.ew;
     uint64_t *dst = A.entries + (size_t)blockIdx.x * A.range * ew;
     for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
         uint32_t idx = (a + t) & mask;
@@ -1945,9 +1960,16 @@ k_cpi_build(CpiBuildArgs A) {
         uint32_t pos = atomicAdd(&lcur, 1u);
         if (pos >= A.range) { *A.err = ERR_MF_OVERFLOW; continue; }
         uint64_t *e = dst + (size_t)pos * ew;
-        e[0] = (uint64_t)key;
-        for (int w = 0; w < 2 * A.na; w++)
-            e[1 + w] = A.state[(size_t)idx * A.na * 2 + w];
+        if (ew == 2) {
+            ulonglong2 v;
+            v.x = (uint64_t)key;
+            v.y = A.state[(size_t)idx * A.na * 2];
+            *(ulonglong2 *)e = v;
+        } else {
+            e[0] = (uint64_t)key;
+            for (int w = 0; w < 2 * A.na; w++)
+                e[1 + w] = A.state[(size_t)idx * A.na * 2 + w];
+        }
     }
     __syncthreads();
     if (threadIdx.x == 0) A.cnt[blockIdx.x] = lcur < A.range ? lcur : A.range;
@@ -2013,13 +2035,23 @@ k_merge_fused(MergeFusedArgs M) {
         if (!PACKED && M.use_cpi) {
             /* dense home-range-grouped entries: only occupied slots read,
              * ownership established at build time */
-            const int ew = 1 + 2 * na;
+            const int ew = M.cpi_ew;
             const uint32_t n_e = M.cpi_cnt[p][blockIdx.x];
             const uint64_t *ent = M.cpi_entries[p] +
                                   (size_t)blockIdx.x * M.range * ew;
-            for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
-                const uint64_t *e = ent + (size_t)t * ew;
-                fold((int64_t)e[0], e + 1, 0, true);
+            if (ew == 2) {
+                /* 16 B entries: one vector load per entry; atomic_merge
+                 * reads only src[0] for the word0-only ops */
+                for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
+                    ulonglong2 e = ((const ulonglong2 *)ent)[t];
+                    uint64_t w0 = (uint64_t)e.y;
+                    fold((int64_t)e.x, &w0, 0, true);
+                }
+            } else {
+                for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
+                    const uint64_t *e = ent + (size_t)t * ew;
+                    fold((int64_t)e[0], e + 1, 0, true);
+                }
             }
         } else if (PACKED) {
             for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
@@ -2305,6 +2337,11 @@ struct GpuOp {
     int n_out_alloc;
     unsigned long long *d_n_out;
     unsigned long long *d_emitted;  /* running device-side emitted rows */
+    unsigned long long *d_fire_cur; /* 64 rotating fused-fire row cursors:
+                                       an 8-byte memset launch per fire
+                                       (~4 us on the fire stream) becomes
+                                       one 512 B memset per 64 fires */
+    uint64_t fire_seq;
     int64_t out_rows_cap;
 
     /* host-accumulated emission (emit_to_host) */
@@ -2326,6 +2363,7 @@ struct GpuOp {
                                  a ring slot's retire always completes
                                  before the slot's bin comes around again */
     uint64_t fire_group;
+    std::vector<uint32_t> retire_pend;  /* slots awaiting the tag clear */
     int spin;                 /* busy-wait host syncs (ARROYO_AMD_SPIN) */
     hipEvent_t ev_sync;       /* status-copy completion (spin target) */
     /* epoch pipeline (mark_epoch / handle_watermarks_epoch): per-epoch
@@ -2359,8 +2397,9 @@ struct GpuOp {
     int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
                        3 batched-probe, 4 radix-regroup */
     /* closed-pane index (see CpiBuildArgs) */
-    uint64_t *cpi_entries;      /* [R][NR][range][1+2na] */
+    uint64_t *cpi_entries;      /* [R][NR][range][cpi_ew] */
     uint32_t *cpi_cnt;          /* [R][NR] */
+    int cpi_ew;
     uint32_t mf_range;          /* fused-merge home-range size (fixed) */
     uint32_t cpi_nr;            /* C / mf_range */
     std::vector<char> cpi_ready;
@@ -2463,7 +2502,11 @@ static int bound_fire_lag(GpuOp *o) {
     return 0;
 }
 
-static int ring_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
+static int ring_retire_flush(GpuOp *o);
+
+/* plane clears only; the tag is cleared LAST by ring_retire_flush (one
+ * kernel for the whole batch of retires), via stream order */
+static int ring_retire_planes(GpuOp *o, uint32_t slot, uint64_t bin) {
     size_t na = o->agg.n_aggs;
     if (slot < o->cpi_ready.size()) o->cpi_ready[slot] = 0;
     if (slot < o->retired_bin.size()) o->retired_bin[slot] = bin;
@@ -2481,13 +2524,30 @@ static int ring_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
                            o->ring.state + (size_t)slot * o->ring.C * na * 2,
                            (int64_t)o->ring.C, (int)(na * 2));
     }
-    /* tag LAST, after the plane clears, via stream order */
-    hipLaunchKernelGGL(k_retire_tag, dim3(1), dim3(1), 0, o->fstream,
-                       o->ring.tag + slot, o->ring.spec_used + slot,
-                       o->ring.spec_state + (size_t)slot * na * 2,
-                       (int)(na * 2));
+    HIP_CHECK(o, hipGetLastError());
+    o->retire_pend.push_back(slot);
+    if ((int)o->retire_pend.size() >= 8) return ring_retire_flush(o);
+    return 0;
+}
+
+static int ring_retire_flush(GpuOp *o) {
+    if (o->retire_pend.empty()) return 0;
+    RetireTagsArgs A = {};
+    A.tag = o->ring.tag;
+    A.spec_used = o->ring.spec_used;
+    A.spec_state = o->ring.spec_state;
+    A.na2 = (int)(o->agg.n_aggs * 2);
+    A.n = (int)o->retire_pend.size();
+    for (int i = 0; i < A.n; i++) A.slot[i] = o->retire_pend[i];
+    o->retire_pend.clear();
+    hipLaunchKernelGGL(k_retire_tags, dim3(1), dim3(1), 0, o->fstream, A);
     HIP_CHECK(o, hipGetLastError());
     return 0;
+}
+
+static int ring_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
+    if (ring_retire_planes(o, slot, bin)) return 1;
+    return ring_retire_flush(o);
 }
 
 static int flush_staged(GpuOp *o);
@@ -2500,11 +2560,12 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.keys = o->ring.keys + (size_t)slot * o->ring.C;
     A.state = o->ring.state + (size_t)slot * o->ring.C * na * 2;
     A.C = o->ring.C;
+    A.ew = o->cpi_ew;
     A.na = (int32_t)na;
     A.range = o->mf_range;
     A.nr = o->cpi_nr;
     A.cnt = o->cpi_cnt + (size_t)slot * o->cpi_nr;
-    A.entries = o->cpi_entries + (size_t)slot * o->ring.C * (1 + 2 * na);
+    A.entries = o->cpi_entries + (size_t)slot * o->ring.C * o->cpi_ew;
     A.err = o->ring.err;
     hipLaunchKernelGGL(k_cpi_build, dim3(o->cpi_nr), dim3(256), 0,
                        o->fstream, A);
@@ -2638,6 +2699,7 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
+    ALLOC(o->d_fire_cur, 64 * 8);
     /* fused-merge home range fixed at create (CPI grouping depends on it) */
     o->mf_range = MF_RANGE;
     if (const char *ev = getenv("ARROYO_AMD_MF_RANGE"))
@@ -2651,8 +2713,10 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
         int use_cpi = 1;
         if (const char *ev = getenv("ARROYO_AMD_CPI")) use_cpi = atoi(ev);
         if (use_cpi) {
+            o->cpi_ew = (na == 1 && o->agg.op[0] != AMD_AGG_AVG)
+                            ? 2 : (int)(1 + 2 * na);
             ALLOC(o->cpi_entries,
-                  (size_t)o->ring.R * o->ring.C * (1 + 2 * na) * 8);
+                  (size_t)o->ring.R * o->ring.C * o->cpi_ew * 8);
             ALLOC(o->cpi_cnt, (size_t)o->ring.R * o->cpi_nr * 4);
         }
     }
@@ -2686,6 +2750,8 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_keys, 0xFF, (size_t)o->CM * 8);
     hipMemset(o->m_zero_blob, 0, o->m_zero_bytes);
     hipMemset(o->d_emitted, 0, 8);
+    hipMemset(o->d_fire_cur, 0, 64 * 8);
+    o->fire_seq = 0;
     hipStreamCreate(&o->stream);
     o->own_fstream = 1;
     if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
@@ -3269,7 +3335,11 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
      * atomics (see k_merge_fused); legacy merge+compact otherwise */
     if (!raw_states && na <= MF_MAX_AGGS && o->ring.C >= MF_RANGE &&
         src.size() <= 64) {
-        HIP_CHECK(o, hipMemsetAsync(o->d_n_out, 0, 8, o->fstream));
+        if ((o->fire_seq & 63) == 0)
+            HIP_CHECK(o, hipMemsetAsync(o->d_fire_cur, 0, 64 * 8,
+                                        o->fstream));
+        unsigned long long *cur = o->d_fire_cur + (o->fire_seq & 63);
+        o->fire_seq++;
         if (!src.empty()) {
             MergeFusedArgs M = {};
             M.ring = o->ring;
@@ -3290,7 +3360,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 for (int i = 0; i < o->n_out_alloc && i < 16; i++)
                     M.out[i] = o->d_out[i];
             }
-            M.n_out = o->d_n_out;
+            M.n_out = cur;
             /* emitted-row accounting folded into the kernel for the
              * device-resident path (saves the separate k_accum launch) */
             M.accum = o->cfg.emit_to_host ? nullptr : o->d_emitted;
@@ -3311,11 +3381,11 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                     if (!o->cpi_ready[src[i]]) all = 0;
                 if (all) {
                     M.use_cpi = 1;
-                    size_t na2 = o->agg.n_aggs;
+                    M.cpi_ew = o->cpi_ew;
                     for (size_t i = 0; i < src.size(); i++) {
                         M.cpi_entries[i] =
                             o->cpi_entries +
-                            (size_t)src[i] * o->ring.C * (1 + 2 * na2);
+                            (size_t)src[i] * o->ring.C * o->cpi_ew;
                         M.cpi_cnt[i] =
                             o->cpi_cnt + (size_t)src[i] * o->cpi_nr;
                     }
@@ -3343,7 +3413,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             if (o->mk) {
                 DictDecArgs DD = {};
                 DD.ids = o->d_keyid_out;
-                DD.n = o->d_n_out;
+                DD.n = cur;
                 DD.dkeys = o->d_dict_keys;
                 DD.nk = o->cfg.n_keys;
                 for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = o->d_out[k];
@@ -3354,7 +3424,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
         }
         if (!o->cfg.emit_to_host) return 0;
         unsigned long long n = 0;
-        HIP_CHECK(o, hipMemcpyAsync(&n, o->d_n_out, 8,
+        HIP_CHECK(o, hipMemcpyAsync(&n, cur, 8,
                                     hipMemcpyDeviceToHost, o->fstream));
         HIP_CHECK(o, hipStreamSynchronize(o->fstream));
         if (n == 0) return 0;
@@ -3498,9 +3568,10 @@ static int advance(GpuOp *o) {
     if (fire_window(o, src, E - o->width, E, 0, 0)) return 1;
 
     for (uint64_t bb : dead) {
-        if (ring_retire(o, o->closed[bb], bb)) return 1;
+        if (ring_retire_planes(o, o->closed[bb], bb)) return 1;
         o->closed.erase(bb);
     }
+    if (ring_retire_flush(o)) return 1;
 
     if (o->closed.empty()) {
         if (!o->table_bins.empty()) {
@@ -3578,9 +3649,10 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
             uint32_t slot = o->open.begin()->second;
             std::vector<uint32_t> src = {slot};
             if (fire_window(o, src, b, b + o->width, 0, 0)) return 1;
-            if (ring_retire(o, slot, b)) return 1;
+            if (ring_retire_planes(o, slot, b)) return 1;
             o->open.erase(o->open.begin());
         }
+        if (ring_retire_flush(o)) return 1;
     } else {
         while (o->state != 0) {
             uint64_t base = (o->state == 1) ? o->earliest : o->next_start;
@@ -3598,9 +3670,10 @@ static int wm_advance(GpuOp *o, uint64_t wm) {
             if (kv.first + o->slide <= wb && kv.first < base)
                 unreachable.push_back(kv.first);
         for (uint64_t bb : unreachable) {
-            if (ring_retire(o, o->open[bb], bb)) return 1;
+            if (ring_retire_planes(o, o->open[bb], bb)) return 1;
             o->open.erase(bb);
         }
+        if (ring_retire_flush(o)) return 1;
     }
     return 0;
 }
@@ -3911,6 +3984,7 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->m_zero_blob);
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
     hipFree(o->d_emitted);
+    hipFree(o->d_fire_cur);
     hipFree(o->rdx_key);
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
