@@ -2341,7 +2341,10 @@ struct GpuOp {
                                        an 8-byte memset launch per fire
                                        (~4 us on the fire stream) becomes
                                        one 512 B memset per 64 fires */
-    uint64_t fire_seq;
+    unsigned long long *d_fire_cur2;  /* fstream2's own ring (a shared
+                                         ring's wrap memset would race the
+                                         other stream's running merge) */
+    uint64_t fire_seq, fire_seq2;
     int64_t out_rows_cap;
 
     /* host-accumulated emission (emit_to_host) */
@@ -2356,6 +2359,15 @@ struct GpuOp {
      * watermark W writes a pane W's fires read.  fstream == stream when
      * ARROYO_AMD_FIRE_STREAM=0 (serial debug mode). */
     hipStream_t fstream;
+    /* second fire stream: a period's two window merges are independent
+     * read-only scans of overlapping closed panes, both latency-bound, so
+     * they run concurrently (alternating streams); retires wait on both */
+    hipStream_t fstream2;
+    hipEvent_t ev_f1, ev_f2, ev_cpi;
+    int f2_dirty;             /* fstream2 has unwaited merge work */
+    uint64_t fire_alt;
+    int64_t *d_out2[16];      /* fstream2's output columns (row ranges of
+                                 concurrent merges must not collide) */
     int own_fstream;
     hipEvent_t ev_gate;       /* updates-done gate the fire stream waits on */
     hipEvent_t ev_tail[4];    /* per-group fire-tail ring: bounds the fire
@@ -2472,12 +2484,17 @@ static int gate_fire(GpuOp *o) {
     if (o->fstream == o->stream) return 0;
     HIP_CHECK(o, hipEventRecord(o->ev_gate, o->stream));
     HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_gate, 0));
+    HIP_CHECK(o, hipStreamWaitEvent(o->fstream2, o->ev_gate, 0));
     return 0;
 }
 
 /* record this watermark group's fire tail (retires included) */
 static int fire_tail(GpuOp *o) {
     if (o->fstream == o->stream) return 0;
+    /* the tail must cover BOTH fire streams (odd merges + cpi builds) */
+    HIP_CHECK(o, hipEventRecord(o->ev_f2, o->fstream2));
+    HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_f2, 0));
+    o->f2_dirty = 0;
     HIP_CHECK(o, hipEventRecord(o->ev_tail[o->fire_group & 3], o->fstream));
     o->fire_group++;
     return 0;
@@ -2508,6 +2525,10 @@ static int ring_retire_flush(GpuOp *o);
  * kernel for the whole batch of retires), via stream order */
 static int ring_retire_planes(GpuOp *o, uint32_t slot, uint64_t bin) {
     size_t na = o->agg.n_aggs;
+    if (o->f2_dirty) {   /* pane clears must follow fstream2's merges */
+        HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_f2, 0));
+        o->f2_dirty = 0;
+    }
     if (slot < o->cpi_ready.size()) o->cpi_ready[slot] = 0;
     if (slot < o->retired_bin.size()) o->retired_bin[slot] = bin;
     int blocks = (int)((o->ring.C + 255) / 256);
@@ -2568,8 +2589,10 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.entries = o->cpi_entries + (size_t)slot * o->ring.C * o->cpi_ew;
     A.err = o->ring.err;
     hipLaunchKernelGGL(k_cpi_build, dim3(o->cpi_nr), dim3(256), 0,
-                       o->fstream, A);
+                       o->fstream2, A);
     HIP_CHECK(o, hipGetLastError());
+    if (o->fstream2 != o->fstream)
+        HIP_CHECK(o, hipEventRecord(o->ev_cpi, o->fstream2));
     o->cpi_ready[slot] = 1;
     return 0;
 }
@@ -2698,8 +2721,12 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     o->out_rows_cap = (int64_t)o->CM + 1;
     for (int i = 0; i < o->n_out_alloc; i++)
         ALLOC(o->d_out[i], (size_t)o->out_rows_cap * 8);
+    if (o->own_fstream)
+        for (int i = 0; i < o->n_out_alloc; i++)
+            ALLOC(o->d_out2[i], (size_t)o->out_rows_cap * 8);
     ALLOC(o->d_emitted, 8);
     ALLOC(o->d_fire_cur, 64 * 8);
+    ALLOC(o->d_fire_cur2, 64 * 8);
     /* fused-merge home range fixed at create (CPI grouping depends on it) */
     o->mf_range = MF_RANGE;
     if (const char *ev = getenv("ARROYO_AMD_MF_RANGE"))
@@ -2751,19 +2778,27 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->m_zero_blob, 0, o->m_zero_bytes);
     hipMemset(o->d_emitted, 0, 8);
     hipMemset(o->d_fire_cur, 0, 64 * 8);
-    o->fire_seq = 0;
+    hipMemset(o->d_fire_cur2, 0, 64 * 8);
+    o->fire_seq = o->fire_seq2 = 0;
     hipStreamCreate(&o->stream);
     o->own_fstream = 1;
     if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
         o->own_fstream = atoi(ev) != 0;
     if (o->own_fstream) {
         hipStreamCreate(&o->fstream);
+        hipStreamCreate(&o->fstream2);
         hipEventCreateWithFlags(&o->ev_gate, hipEventDisableTiming);
+        hipEventCreateWithFlags(&o->ev_f1, hipEventDisableTiming);
+        hipEventCreateWithFlags(&o->ev_f2, hipEventDisableTiming);
+        hipEventCreateWithFlags(&o->ev_cpi, hipEventDisableTiming);
         for (int i = 0; i < 4; i++)
             hipEventCreateWithFlags(&o->ev_tail[i], hipEventDisableTiming);
     } else {
         o->fstream = o->stream;
+        o->fstream2 = o->stream;
     }
+    o->f2_dirty = 0;
+    o->fire_alt = 0;
     o->fire_group = 0;
     o->spin = 1;
     if (const char *ev = getenv("ARROYO_AMD_SPIN")) o->spin = atoi(ev) != 0;
@@ -3335,11 +3370,32 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
      * atomics (see k_merge_fused); legacy merge+compact otherwise */
     if (!raw_states && na <= MF_MAX_AGGS && o->ring.C >= MF_RANGE &&
         src.size() <= 64) {
-        if ((o->fire_seq & 63) == 0)
-            HIP_CHECK(o, hipMemsetAsync(o->d_fire_cur, 0, 64 * 8,
-                                        o->fstream));
-        unsigned long long *cur = o->d_fire_cur + (o->fire_seq & 63);
-        o->fire_seq++;
+        /* alternate the period's independent window merges across the two
+         * fire streams (read-only over the panes; retires wait on both).
+         * mk is excluded: the dictionary-id scratch column is shared. */
+        hipStream_t fs = o->fstream;
+        int64_t *const *outv = o->d_out;
+        int on_f2 = o->own_fstream && !o->mk && (o->fire_alt & 1);
+        o->fire_alt++;
+        unsigned long long *cur;
+        if (on_f2) {
+            /* in-stream order already puts this merge after the group's
+             * cpi builds (same stream) and the gate */
+            fs = o->fstream2;
+            outv = o->d_out2;
+            if ((o->fire_seq2 & 63) == 0)
+                HIP_CHECK(o, hipMemsetAsync(o->d_fire_cur2, 0, 64 * 8, fs));
+            cur = o->d_fire_cur2 + (o->fire_seq2 & 63);
+            o->fire_seq2++;
+        } else {
+            /* cpi builds run on fstream2: order this merge after them */
+            if (o->own_fstream)
+                HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_cpi, 0));
+            if ((o->fire_seq & 63) == 0)
+                HIP_CHECK(o, hipMemsetAsync(o->d_fire_cur, 0, 64 * 8, fs));
+            cur = o->d_fire_cur + (o->fire_seq & 63);
+            o->fire_seq++;
+        }
         if (!src.empty()) {
             MergeFusedArgs M = {};
             M.ring = o->ring;
@@ -3355,10 +3411,10 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 M.out[0] = o->d_keyid_out;
                 for (int i = 1; i < o->n_out_alloc - o->cfg.n_keys + 1 &&
                                 i < 16; i++)
-                    M.out[i] = o->d_out[o->cfg.n_keys - 1 + i];
+                    M.out[i] = outv[o->cfg.n_keys - 1 + i];
             } else {
                 for (int i = 0; i < o->n_out_alloc && i < 16; i++)
-                    M.out[i] = o->d_out[i];
+                    M.out[i] = outv[i];
             }
             M.n_out = cur;
             /* emitted-row accounting folded into the kernel for the
@@ -3396,19 +3452,19 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 if (mfs >= 2048)
                     hipLaunchKernelGGL((k_merge_fused<2048, true>),
                                        dim3(o->ring.C / range), dim3(256),
-                                       shmem, o->fstream, M);
+                                       shmem, fs, M);
                 else
                     hipLaunchKernelGGL((k_merge_fused<1024, true>),
                                        dim3(o->ring.C / range), dim3(256),
-                                       shmem, o->fstream, M);
+                                       shmem, fs, M);
             } else if (mfs >= 2048)
                 hipLaunchKernelGGL(k_merge_fused<2048>,
                                    dim3(o->ring.C / range), dim3(256),
-                                   shmem, o->fstream, M);
+                                   shmem, fs, M);
             else
                 hipLaunchKernelGGL(k_merge_fused<1024>,
                                    dim3(o->ring.C / range), dim3(256),
-                                   shmem, o->fstream, M);
+                                   shmem, fs, M);
             HIP_CHECK(o, hipGetLastError());
             if (o->mk) {
                 DictDecArgs DD = {};
@@ -3416,17 +3472,23 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 DD.n = cur;
                 DD.dkeys = o->d_dict_keys;
                 DD.nk = o->cfg.n_keys;
-                for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = o->d_out[k];
+                for (int k = 0; k < o->cfg.n_keys; k++) DD.out[k] = outv[k];
                 hipLaunchKernelGGL(k_dict_decode, dim3(1024), dim3(256), 0,
-                                   o->fstream, DD);
+                                   fs, DD);
                 HIP_CHECK(o, hipGetLastError());
             }
         }
-        if (!o->cfg.emit_to_host) return 0;
+        if (!o->cfg.emit_to_host) {
+            if (on_f2) {
+                HIP_CHECK(o, hipEventRecord(o->ev_f2, o->fstream2));
+                o->f2_dirty = 1;
+            }
+            return 0;
+        }
         unsigned long long n = 0;
         HIP_CHECK(o, hipMemcpyAsync(&n, cur, 8,
-                                    hipMemcpyDeviceToHost, o->fstream));
-        HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+                                    hipMemcpyDeviceToHost, fs));
+        HIP_CHECK(o, hipStreamSynchronize(fs));
         if (n == 0) return 0;
         if ((size_t)o->out_cols > o->host_out.size())
             o->host_out.resize(o->out_cols);
@@ -3434,10 +3496,10 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
             size_t old = o->host_out[i].size();
             o->host_out[i].resize(old + n);
             HIP_CHECK(o, hipMemcpyAsync(o->host_out[i].data() + old,
-                                        o->d_out[i], n * 8,
-                                        hipMemcpyDeviceToHost, o->fstream));
+                                        outv[i], n * 8,
+                                        hipMemcpyDeviceToHost, fs));
         }
-        HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+        HIP_CHECK(o, hipStreamSynchronize(fs));
         return 0;
     }
     HIP_CHECK(o, hipMemsetAsync(o->m_keys, 0xFF, (size_t)o->CM * 8, o->stream));
@@ -3804,8 +3866,10 @@ API int arroyo_amd_handle_watermarks_epoch(void *h, const uint64_t *wms,
     if (bound_fire_lag(o)) return 1;
     /* gate this epoch's fires on ITS stream point (later periods' update
      * launches are after it and not waited on) */
-    if (o->fstream != o->stream)
+    if (o->fstream != o->stream) {
         HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_epoch[e], 0));
+        HIP_CHECK(o, hipStreamWaitEvent(o->fstream2, o->ev_epoch[e], 0));
+    }
     for (int32_t i = 0; i < n; i++)
         if (wm_advance(o, wms[i])) return 1;
     if (fire_tail(o)) return 1;
@@ -3819,6 +3883,8 @@ API int arroyo_amd_checkpoint_drain(void *h, AmdOutBatch *out) {
      * as trailing timestamp column. */
     GpuOp *o = (GpuOp *)h;
     HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+    if (o->fstream != o->stream)
+        HIP_CHECK(o, hipStreamSynchronize(o->fstream2));
     if (flush_staged(o)) return 1;
     if (check_device_error(o)) return 1;
     if (sync_open_panes(o)) return 1;
@@ -3965,10 +4031,16 @@ API void arroyo_amd_destroy(void *h) {
         hipEventDestroy(o->ev_epoch[i]);
     }
     if (o->fstream != o->stream) {
+        hipStreamSynchronize(o->fstream2);
         hipEventDestroy(o->ev_gate);
+        hipEventDestroy(o->ev_f1);
+        hipEventDestroy(o->ev_f2);
+        hipEventDestroy(o->ev_cpi);
         for (int i = 0; i < 4; i++) hipEventDestroy(o->ev_tail[i]);
         hipStreamDestroy(o->fstream);
+        hipStreamDestroy(o->fstream2);
     }
+    for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out2[i]);
     for (auto &pr : o->ev_pool) {
         hipEventDestroy(pr.first);
         hipEventDestroy(pr.second);
@@ -3985,6 +4057,7 @@ API void arroyo_amd_destroy(void *h) {
     for (int i = 0; i < o->n_out_alloc; i++) hipFree(o->d_out[i]);
     hipFree(o->d_emitted);
     hipFree(o->d_fire_cur);
+    hipFree(o->d_fire_cur2);
     hipFree(o->rdx_key);
     hipFree(o->rdx_ts);
     hipFree(o->rdx_hist);
@@ -4014,8 +4087,10 @@ API void arroyo_amd_destroy(void *h) {
 API int arroyo_amd_sync(void *h) {
     GpuOp *o = (GpuOp *)h;
     HIP_CHECK(o, hipStreamSynchronize(o->stream));
-    if (o->fstream != o->stream)
+    if (o->fstream != o->stream) {
         HIP_CHECK(o, hipStreamSynchronize(o->fstream));
+        HIP_CHECK(o, hipStreamSynchronize(o->fstream2));
+    }
     return 0;
 }
 
