@@ -2618,6 +2618,10 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     }
     GpuOp *o = new GpuOp();
     o->cfg = *cfg;
+    /* read before the allocation block: d_out2 existence depends on it */
+    o->own_fstream = 1;
+    if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
+        o->own_fstream = atoi(ev) != 0;
     /* hop(x, x) is a tumble window (arroyo-planner/src/lib.rs:649-651) */
     if (!o->cfg.is_tumbling && o->cfg.slide_nanos == o->cfg.width_nanos)
         o->cfg.is_tumbling = 1;
@@ -2781,9 +2785,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->d_fire_cur2, 0, 64 * 8);
     o->fire_seq = o->fire_seq2 = 0;
     hipStreamCreate(&o->stream);
-    o->own_fstream = 1;
-    if (const char *ev = getenv("ARROYO_AMD_FIRE_STREAM"))
-        o->own_fstream = atoi(ev) != 0;
     if (o->own_fstream) {
         hipStreamCreate(&o->fstream);
         hipStreamCreate(&o->fstream2);
