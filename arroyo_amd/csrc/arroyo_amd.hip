@@ -1951,14 +1951,12 @@ k_cpi_build(CpiBuildArgs A) {
     const uint32_t span = A.range + MAX_PROBES;
     const int ew = A.ew;
     uint64_t *dst = A.entries + (size_t)blockIdx.x * A.range * ew;
-    for (uint32_t t = threadIdx.x; t < span; t += blockDim.x) {
-        uint32_t idx = (a + t) & mask;
-        int64_t key = A.keys[idx];
-        if (key == EMPTY_KEY) continue;
+    auto emit = [&](int64_t key, uint32_t idx) {
+        if (key == EMPTY_KEY) return;
         uint32_t rel = ((uint32_t)hash64((uint64_t)key) - a) & mask;
-        if (rel >= A.range) continue;
+        if (rel >= A.range) return;
         uint32_t pos = atomicAdd(&lcur, 1u);
-        if (pos >= A.range) { *A.err = ERR_MF_OVERFLOW; continue; }
+        if (pos >= A.range) { *A.err = ERR_MF_OVERFLOW; return; }
         uint64_t *e = dst + (size_t)pos * ew;
         if (ew == 2) {
             ulonglong2 v;
@@ -1970,6 +1968,14 @@ k_cpi_build(CpiBuildArgs A) {
             for (int w = 0; w < 2 * A.na; w++)
                 e[1 + w] = A.state[(size_t)idx * A.na * 2 + w];
         }
+    };
+    /* 2 keys per 16 B load: a and span are even and an even idx never
+     * straddles the wrap (same argument as the merge's pair scan) */
+    for (uint32_t t = 2 * threadIdx.x; t < span; t += 2 * blockDim.x) {
+        uint32_t idx = (a + t) & mask;
+        ulonglong2 kv = *(const ulonglong2 *)&A.keys[idx];
+        emit((int64_t)kv.x, idx);
+        emit((int64_t)kv.y, idx + 1);
     }
     __syncthreads();
     if (threadIdx.x == 0) A.cnt[blockIdx.x] = lcur < A.range ? lcur : A.range;
