@@ -1985,14 +1985,20 @@ template <int SLOTS, bool PACKED = false>
 __global__ void __launch_bounds__(256)
 k_merge_fused(MergeFusedArgs M) {
     __shared__ int64_t lkey[SLOTS];
-    extern __shared__ uint64_t lst[];    /* [SLOTS][n_aggs][2] */
+    extern __shared__ uint64_t lst[];    /* [SLOTS][n_aggs][2], or
+                                            [SLOTS] single words when the
+                                            lone aggregate uses only word0
+                                            (compact: halves the LDS and
+                                            lifts the WGs/CU cap) */
     __shared__ unsigned long long blk_base;
     __shared__ unsigned int blk_cnt;
     const int na = M.agg.n_aggs;
+    const int compact = (!PACKED && M.use_cpi && M.cpi_ew == 2);
+    const int sw = compact ? 1 : na * 2;
     const uint32_t C = M.ring.C, mask = C - 1;
     for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
         lkey[i] = EMPTY_KEY;
-        for (int w = 0; w < na * 2; w++) lst[(size_t)i * na * 2 + w] = 0;
+        for (int w = 0; w < sw; w++) lst[(size_t)i * sw + w] = 0;
     }
     if (threadIdx.x == 0) blk_cnt = 0;
     __syncthreads();
@@ -2030,11 +2036,30 @@ k_merge_fused(MergeFusedArgs M) {
                 }
             }
             if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; return; }
-            uint64_t *d = lst + (size_t)slot * na * 2;
+            uint64_t *d = lst + (size_t)slot * sw;
             if (PACKED) {   /* single COUNT state, read with the key */
                 atomicAdd((unsigned long long *)&d[0],
                           (unsigned long long)pw0);
                 return;
+            }
+            if (compact) {   /* single live word: fold without the pair
+                                stride atomic_merge assumes */
+                switch (M.agg.op[0]) {
+                case AMD_AGG_SUM:
+                    if (M.agg.isf[0]) {
+                        atomicAdd((double *)d, *(const double *)stv);
+                        return;
+                    }
+                    /* fallthrough: integer add */
+                case AMD_AGG_COUNT:
+                    atomicAdd((unsigned long long *)d,
+                              (unsigned long long)stv[0]);
+                    return;
+                default:   /* MIN/MAX: order-preserving u64 encodings */
+                    atomicMax((unsigned long long *)d,
+                              (unsigned long long)stv[0]);
+                    return;
+                }
             }
             atomic_merge(d, stv, M.agg);
         };
@@ -2100,11 +2125,11 @@ k_merge_fused(MergeFusedArgs M) {
         int64_t key = lkey[i];
         if (key == EMPTY_KEY) continue;
         int64_t r = (int64_t)(blk_base + atomicAdd(&blk_cnt, 1u));
-        const uint64_t *d = lst + (size_t)i * na * 2;
+        const uint64_t *d = lst + (size_t)i * sw;
         int col = 0;
         if (M.n_keys) M.out[col++][r] = key;
         for (int ag = 0; ag < na; ag++) {
-            uint64_t w0 = d[2 * ag];
+            uint64_t w0 = d[compact ? 0 : 2 * ag];
             switch (M.agg.op[ag]) {
             case AMD_AGG_COUNT:
             case AMD_AGG_SUM: M.out[col++][r] = (int64_t)w0; break;
@@ -3454,7 +3479,8 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                     }
                 }
             }
-            size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) * na * 16;
+            size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) *
+                           ((M.use_cpi && o->cpi_ew == 2) ? 8 : na * 16);
             if (o->ring.packed) {
                 if (mfs >= 2048)
                     hipLaunchKernelGGL((k_merge_fused<2048, true>),
