@@ -1915,8 +1915,12 @@ struct MergeFusedArgs {
     int32_t use_cpi;
     int32_t cpi_ew;   /* entry words: 1+1 when only word0 of a single agg
                          is live (COUNT/SUM/MIN/MAX), else 1+2*na */
-    const uint64_t *cpi_entries[16];   /* [nr][range][cpi_ew] */
-    const uint32_t *cpi_cnt[16];       /* [nr] entries per range */
+    uint32_t cpi_range;  /* index granularity: the build runs one WG per
+                            cpi_range home slots (smaller than the merge
+                            range so the build grid can hide latency);
+                            a merge WG reads range/cpi_range segments */
+    const uint64_t *cpi_entries[16];   /* [nr][cpi_range][cpi_ew] */
+    const uint32_t *cpi_cnt[16];       /* [nr] entries per sub-range */
 };
 
 /* Closed-pane index build: one workgroup per home range.  Linear probing
@@ -1926,7 +1930,7 @@ struct MergeFusedArgs {
  * dense per-range segment.  Per-range capacity equals `range` (pane load
  * factor is bounded well below 1 by construction; exceeding it is a
  * loud error). */
-#define CPI_MAX_NR 4096
+#define CPI_MAX_NR 8192
 
 struct CpiBuildArgs {
     const int64_t *keys;      /* pane planes in the ring */
@@ -2067,21 +2071,27 @@ k_merge_fused(MergeFusedArgs M) {
             /* dense home-range-grouped entries: only occupied slots read,
              * ownership established at build time */
             const int ew = M.cpi_ew;
-            const uint32_t n_e = M.cpi_cnt[p][blockIdx.x];
-            const uint64_t *ent = M.cpi_entries[p] +
-                                  (size_t)blockIdx.x * M.range * ew;
-            if (ew == 2) {
-                /* 16 B entries: one vector load per entry; atomic_merge
-                 * reads only src[0] for the word0-only ops */
-                for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
-                    ulonglong2 e = ((const ulonglong2 *)ent)[t];
-                    uint64_t w0 = (uint64_t)e.y;
-                    fold((int64_t)e.x, &w0, 0, true);
-                }
-            } else {
-                for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
-                    const uint64_t *e = ent + (size_t)t * ew;
-                    fold((int64_t)e[0], e + 1, 0, true);
+            const uint32_t nsub = M.range / M.cpi_range;
+            for (uint32_t j = 0; j < nsub; j++) {
+                uint32_t seg = blockIdx.x * nsub + j;
+                const uint32_t n_e = M.cpi_cnt[p][seg];
+                const uint64_t *ent = M.cpi_entries[p] +
+                                      (size_t)seg * M.cpi_range * ew;
+                if (ew == 2) {
+                    /* 16 B entries: one vector load per entry;
+                     * atomic_merge reads only src[0] for word0-only ops */
+                    for (uint32_t t = threadIdx.x; t < n_e;
+                         t += blockDim.x) {
+                        ulonglong2 e = ((const ulonglong2 *)ent)[t];
+                        uint64_t w0 = (uint64_t)e.y;
+                        fold((int64_t)e.x, &w0, 0, true);
+                    }
+                } else {
+                    for (uint32_t t = threadIdx.x; t < n_e;
+                         t += blockDim.x) {
+                        const uint64_t *e = ent + (size_t)t * ew;
+                        fold((int64_t)e[0], e + 1, 0, true);
+                    }
                 }
             }
         } else if (PACKED) {
@@ -2440,9 +2450,10 @@ struct GpuOp {
     int upd_kind;   /* 0 lds, 1 packed AoS, 2 split wave-combine,
                        3 batched-probe, 4 radix-regroup */
     /* closed-pane index (see CpiBuildArgs) */
-    uint64_t *cpi_entries;      /* [R][NR][range][cpi_ew] */
+    uint64_t *cpi_entries;      /* [R][NR][cpi_range][cpi_ew] */
     uint32_t *cpi_cnt;          /* [R][NR] */
     int cpi_ew;
+    uint32_t cpi_range;         /* build granularity (<= mf_range) */
     uint32_t mf_range;          /* fused-merge home-range size (fixed) */
     uint32_t cpi_nr;            /* C / mf_range */
     std::vector<char> cpi_ready;
@@ -2614,7 +2625,7 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.C = o->ring.C;
     A.ew = o->cpi_ew;
     A.na = (int32_t)na;
-    A.range = o->mf_range;
+    A.range = o->cpi_range;
     A.nr = o->cpi_nr;
     A.cnt = o->cpi_cnt + (size_t)slot * o->cpi_nr;
     A.entries = o->cpi_entries + (size_t)slot * o->ring.C * o->cpi_ew;
@@ -2769,7 +2780,18 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (o->mf_range < 2) o->mf_range = 2;
     if (o->mf_range > o->ring.C) o->mf_range = o->ring.C;
     while (o->mf_range & (o->mf_range - 1)) o->mf_range &= o->mf_range - 1;
-    o->cpi_nr = o->ring.C / o->mf_range;
+    /* build granularity: small enough sub-ranges that the build grid can
+     * hide memory latency (one WG per 2 CUs at 512 WGs was 2 waves/SIMD
+     * and 30 us; 256-slot sub-ranges quadruple the grid) */
+    o->cpi_range = o->mf_range > 256 ? 256 : o->mf_range;
+    if (const char *ev = getenv("ARROYO_AMD_CPI_RANGE")) {
+        o->cpi_range = (uint32_t)atoi(ev);
+        if (o->cpi_range < 2) o->cpi_range = 2;
+        if (o->cpi_range > o->mf_range) o->cpi_range = o->mf_range;
+        while (o->cpi_range & (o->cpi_range - 1))
+            o->cpi_range &= o->cpi_range - 1;
+    }
+    o->cpi_nr = o->ring.C / o->cpi_range;
     if (!o->ring.packed && na <= MF_MAX_AGGS && !o->cfg.is_tumbling &&
         o->cpi_nr <= CPI_MAX_NR) {
         int use_cpi = 1;
@@ -3470,6 +3492,7 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                 if (all) {
                     M.use_cpi = 1;
                     M.cpi_ew = o->cpi_ew;
+                    M.cpi_range = o->cpi_range;
                     for (size_t i = 0; i < src.size(); i++) {
                         M.cpi_entries[i] =
                             o->cpi_entries +
