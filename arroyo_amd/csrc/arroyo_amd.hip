@@ -2781,9 +2781,10 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     if (o->mf_range > o->ring.C) o->mf_range = o->ring.C;
     while (o->mf_range & (o->mf_range - 1)) o->mf_range &= o->mf_range - 1;
     /* build granularity: small enough sub-ranges that the build grid can
-     * hide memory latency (one WG per 2 CUs at 512 WGs was 2 waves/SIMD
-     * and 30 us; 256-slot sub-ranges quadruple the grid) */
-    o->cpi_range = o->mf_range > 256 ? 256 : o->mf_range;
+     * hide memory latency (512 WGs was 2 waves/SIMD and 30 us), but not
+     * so small that the merge's per-segment scan idles most of its
+     * threads (the sweep: 512 > 256 > 128; whole-job 23.8/22.3/18.1 G) */
+    o->cpi_range = o->mf_range > 512 ? 512 : o->mf_range;
     if (const char *ev = getenv("ARROYO_AMD_CPI_RANGE")) {
         o->cpi_range = (uint32_t)atoi(ev);
         if (o->cpi_range < 2) o->cpi_range = 2;
