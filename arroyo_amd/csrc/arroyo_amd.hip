@@ -1394,31 +1394,6 @@ k_update_batch_scalar(UpdateArgs A) {
  * done counter was measured instead: 1024 same-address returning atomics
  * serialize at ~0.3 us each -- it turned a ~5 us retire into a fire-
  * stream-dominating one and cost 25% whole-job.) */
-/* sparse pane retire: the closed-pane index lists exactly the occupied
- * slots (no inserts can follow a close), so clearing those is equivalent
- * to the full-plane memset at a fraction of the traffic (~25% occupancy
- * at the bench shape: 3.6 MB scattered vs 12 MB streamed). */
-struct RetireSparseArgs {
-    int64_t *keys;
-    uint64_t *state;
-    int na2;
-    const uint32_t *cnt;      /* [nseg] */
-    const uint32_t *idx;      /* [nseg][range] */
-    uint32_t range;
-};
-
-__global__ void __launch_bounds__(256)
-k_retire_sparse(RetireSparseArgs A) {
-    const uint32_t n = A.cnt[blockIdx.x];
-    const uint32_t *seg = A.idx + (size_t)blockIdx.x * A.range;
-    for (uint32_t t = threadIdx.x; t < n; t += blockDim.x) {
-        uint32_t i = seg[t];
-        A.keys[i] = EMPTY_KEY;
-        for (int w = 0; w < A.na2; w++)
-            A.state[(size_t)i * A.na2 + w] = 0;
-    }
-}
-
 struct RetireTagsArgs {
     uint64_t *tag;            /* [R] */
     uint32_t *spec_used;      /* [R] */
@@ -1966,9 +1941,7 @@ struct CpiBuildArgs {
     uint32_t range;
     uint32_t nr;              /* C / range */
     uint32_t *cnt;            /* [nr] entries per range (output) */
-    uint64_t *entries;        /* [nr][range][ew] */
-    uint32_t *idx;            /* [nr][range] slot index of each entry (for
-                                 the sparse retire) */
+    uint64_t *entries;        /* [nr][range][1 + 2*na] */
     int *err;
 };
 
@@ -1988,7 +1961,6 @@ k_cpi_build(CpiBuildArgs A) {
         if (rel >= A.range) return;
         uint32_t pos = atomicAdd(&lcur, 1u);
         if (pos >= A.range) { *A.err = ERR_MF_OVERFLOW; return; }
-        A.idx[(size_t)blockIdx.x * A.range + pos] = idx;
         uint64_t *e = dst + (size_t)pos * ew;
         if (ew == 2) {
             ulonglong2 v;
@@ -2480,7 +2452,6 @@ struct GpuOp {
     /* closed-pane index (see CpiBuildArgs) */
     uint64_t *cpi_entries;      /* [R][NR][cpi_range][cpi_ew] */
     uint32_t *cpi_cnt;          /* [R][NR] */
-    uint32_t *cpi_idx;          /* [R][C] entry slot indices */
     int cpi_ew;
     uint32_t cpi_range;         /* build granularity (<= mf_range) */
     uint32_t mf_range;          /* fused-merge home-range size (fixed) */
@@ -2600,24 +2571,8 @@ static int ring_retire_planes(GpuOp *o, uint32_t slot, uint64_t bin) {
         HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_f2, 0));
         o->f2_dirty = 0;
     }
-    int have_cpi = slot < o->cpi_ready.size() && o->cpi_ready[slot];
     if (slot < o->cpi_ready.size()) o->cpi_ready[slot] = 0;
     if (slot < o->retired_bin.size()) o->retired_bin[slot] = bin;
-    if (have_cpi && !o->ring.packed) {
-        RetireSparseArgs A = {};
-        A.keys = o->ring.keys + (size_t)slot * o->ring.C;
-        A.state = o->ring.state + (size_t)slot * o->ring.C * na * 2;
-        A.na2 = (int)(na * 2);
-        A.cnt = o->cpi_cnt + (size_t)slot * o->cpi_nr;
-        A.idx = o->cpi_idx + (size_t)slot * o->ring.C;
-        A.range = o->cpi_range;
-        hipLaunchKernelGGL(k_retire_sparse, dim3(o->cpi_nr), dim3(256), 0,
-                           o->fstream, A);
-        HIP_CHECK(o, hipGetLastError());
-        o->retire_pend.push_back(slot);
-        if ((int)o->retire_pend.size() >= 8) return ring_retire_flush(o);
-        return 0;
-    }
     int blocks = (int)((o->ring.C + 255) / 256);
     if (blocks > 1024) blocks = 1024;
     if (o->ring.packed) {
@@ -2673,7 +2628,6 @@ static int cpi_build(GpuOp *o, uint32_t slot) {
     A.range = o->cpi_range;
     A.nr = o->cpi_nr;
     A.cnt = o->cpi_cnt + (size_t)slot * o->cpi_nr;
-    A.idx = o->cpi_idx + (size_t)slot * o->ring.C;
     A.entries = o->cpi_entries + (size_t)slot * o->ring.C * o->cpi_ew;
     A.err = o->ring.err;
     hipLaunchKernelGGL(k_cpi_build, dim3(o->cpi_nr), dim3(256), 0,
@@ -2849,7 +2803,6 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
             ALLOC(o->cpi_entries,
                   (size_t)o->ring.R * o->ring.C * o->cpi_ew * 8);
             ALLOC(o->cpi_cnt, (size_t)o->ring.R * o->cpi_nr * 4);
-            ALLOC(o->cpi_idx, (size_t)o->ring.R * o->ring.C * 4);
         }
     }
     o->cpi_ready.assign(o->ring.R, 0);
@@ -3055,7 +3008,6 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
         if (n_rows > o->rdx2_cap) {
             hipFree(o->cpi_entries);
     hipFree(o->cpi_cnt);
-    hipFree(o->cpi_idx);
     hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
@@ -4169,7 +4121,6 @@ API void arroyo_amd_destroy(void *h) {
     hipFree(o->rdx_tmp);
     hipFree(o->cpi_entries);
     hipFree(o->cpi_cnt);
-    hipFree(o->cpi_idx);
     hipFree(o->d_dict_digest);
     hipFree(o->d_dict_keys);
     hipFree(o->d_dict_ready);
