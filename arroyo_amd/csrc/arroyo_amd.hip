@@ -2078,10 +2078,27 @@ k_merge_fused(MergeFusedArgs M) {
                 const uint64_t *ent = M.cpi_entries[p] +
                                       (size_t)seg * M.cpi_range * ew;
                 if (ew == 2) {
-                    /* 16 B entries: one vector load per entry;
-                     * atomic_merge reads only src[0] for word0-only ops */
-                    for (uint32_t t = threadIdx.x; t < n_e;
-                         t += blockDim.x) {
+                    /* 16 B entries; 4 loads issued together per thread
+                     * (memory-level parallelism -- the single-load loop
+                     * serializes a ~900-cycle miss against LDS work,
+                     * exactly the update kernel's Q-batching lesson) */
+                    const uint32_t stride = blockDim.x;
+                    uint32_t t = threadIdx.x;
+                    for (; t + 3 * stride < n_e; t += 4 * stride) {
+                        ulonglong2 e0 = ((const ulonglong2 *)ent)[t];
+                        ulonglong2 e1 =
+                            ((const ulonglong2 *)ent)[t + stride];
+                        ulonglong2 e2 =
+                            ((const ulonglong2 *)ent)[t + 2 * stride];
+                        ulonglong2 e3 =
+                            ((const ulonglong2 *)ent)[t + 3 * stride];
+                        uint64_t w;
+                        w = e0.y; fold((int64_t)e0.x, &w, 0, true);
+                        w = e1.y; fold((int64_t)e1.x, &w, 0, true);
+                        w = e2.y; fold((int64_t)e2.x, &w, 0, true);
+                        w = e3.y; fold((int64_t)e3.x, &w, 0, true);
+                    }
+                    for (; t < n_e; t += stride) {
                         ulonglong2 e = ((const ulonglong2 *)ent)[t];
                         uint64_t w0 = (uint64_t)e.y;
                         fold((int64_t)e.x, &w0, 0, true);
