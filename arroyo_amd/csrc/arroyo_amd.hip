@@ -2256,6 +2256,7 @@ k_merge_fused(MergeFusedArgs M) {
 
 template __global__ void k_merge_fused<1024>(MergeFusedArgs);
 template __global__ void k_merge_fused<2048>(MergeFusedArgs);
+template __global__ void k_merge_fused<4096>(MergeFusedArgs);
 template __global__ void k_merge_fused<1024, true>(MergeFusedArgs);
 template __global__ void k_merge_fused<2048, true>(MergeFusedArgs);
 
@@ -3520,7 +3521,10 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                     }
                 }
             }
-            size_t shmem = (size_t)(mfs >= 2048 ? 2048 : 1024) *
+            int slots = mfs >= 4096 ? 4096 : (mfs >= 2048 ? 2048 : 1024);
+            if (slots == 4096 && !(M.use_cpi && o->cpi_ew == 2))
+                slots = 2048;   /* 4096 two-word states blow the LDS */
+            size_t shmem = (size_t)slots *
                            ((M.use_cpi && o->cpi_ew == 2) ? 8 : na * 16);
             if (o->ring.packed) {
                 if (mfs >= 2048)
@@ -3531,7 +3535,11 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
                     hipLaunchKernelGGL((k_merge_fused<1024, true>),
                                        dim3(o->ring.C / range), dim3(256),
                                        shmem, fs, M);
-            } else if (mfs >= 2048)
+            } else if (slots == 4096)
+                hipLaunchKernelGGL(k_merge_fused<4096>,
+                                   dim3(o->ring.C / range), dim3(256),
+                                   shmem, fs, M);
+            else if (slots == 2048)
                 hipLaunchKernelGGL(k_merge_fused<2048>,
                                    dim3(o->ring.C / range), dim3(256),
                                    shmem, fs, M);
