@@ -2856,10 +2856,16 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipMemset(o->d_fire_cur, 0, 64 * 8);
     hipMemset(o->d_fire_cur2, 0, 64 * 8);
     o->fire_seq = o->fire_seq2 = 0;
-    hipStreamCreate(&o->stream);
+    /* the update stream paces the period: give it scheduler priority over
+     * the fire streams so its blocks win the CU race (fires have slack) */
+    int pr_lo = 0, pr_hi = 0;
+    hipDeviceGetStreamPriorityRange(&pr_lo, &pr_hi);
+    hipStreamCreateWithPriority(&o->stream, hipStreamNonBlocking, pr_hi);
     if (o->own_fstream) {
-        hipStreamCreate(&o->fstream);
-        hipStreamCreate(&o->fstream2);
+        hipStreamCreateWithPriority(&o->fstream, hipStreamNonBlocking,
+                                    pr_lo);
+        hipStreamCreateWithPriority(&o->fstream2, hipStreamNonBlocking,
+                                    pr_lo);
         hipEventCreateWithFlags(&o->ev_gate, hipEventDisableTiming);
         hipEventCreateWithFlags(&o->ev_f1, hipEventDisableTiming);
         hipEventCreateWithFlags(&o->ev_f2, hipEventDisableTiming);
