@@ -46,3 +46,6 @@ for tag in ("fetch", "write"):
         print(" ", k, {cn: f"{v/max(n[(k,cn)],1):.3e}x{n[(k,cn)]}" for cn, v in c.items()})
 EOF
 echo DONE
+# genuine sustained soak (MIN_TIMED_S stretches the timed region)
+BENCH_MIN_TIMED_S=30 timeout 300 python /root/repo/bench.py --skip-cpu-baseline > /root/repo/gpurun_out/r02z_soak30.json 2>/dev/null
+tail -c 400 /root/repo/gpurun_out/r02z_soak30.json
