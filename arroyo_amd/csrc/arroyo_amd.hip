@@ -1973,9 +1973,27 @@ k_cpi_build(CpiBuildArgs A) {
                 e[1 + w] = A.state[(size_t)idx * A.na * 2 + w];
         }
     };
-    /* 2 keys per 16 B load: a and span are even and an even idx never
-     * straddles the wrap (same argument as the merge's pair scan) */
-    for (uint32_t t = 2 * threadIdx.x; t < span; t += 2 * blockDim.x) {
+    /* 2 keys per 16 B load (a and span are even and an even idx never
+     * straddles the wrap, same argument as the merge's pair scan); four
+     * pair loads issued together so the scan is not one long dependent
+     * miss chain */
+    const uint32_t step = 2 * blockDim.x;
+    uint32_t t = 2 * threadIdx.x;
+    for (; t + 3 * step < span; t += 4 * step) {
+        uint32_t i0 = (a + t) & mask;
+        uint32_t i1 = (a + t + step) & mask;
+        uint32_t i2 = (a + t + 2 * step) & mask;
+        uint32_t i3 = (a + t + 3 * step) & mask;
+        ulonglong2 k0 = *(const ulonglong2 *)&A.keys[i0];
+        ulonglong2 k1 = *(const ulonglong2 *)&A.keys[i1];
+        ulonglong2 k2 = *(const ulonglong2 *)&A.keys[i2];
+        ulonglong2 k3 = *(const ulonglong2 *)&A.keys[i3];
+        emit((int64_t)k0.x, i0); emit((int64_t)k0.y, i0 + 1);
+        emit((int64_t)k1.x, i1); emit((int64_t)k1.y, i1 + 1);
+        emit((int64_t)k2.x, i2); emit((int64_t)k2.y, i2 + 1);
+        emit((int64_t)k3.x, i3); emit((int64_t)k3.y, i3 + 1);
+    }
+    for (; t < span; t += step) {
         uint32_t idx = (a + t) & mask;
         ulonglong2 kv = *(const ulonglong2 *)&A.keys[idx];
         emit((int64_t)kv.x, idx);
