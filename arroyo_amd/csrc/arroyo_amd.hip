@@ -47,6 +47,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <algorithm>
 #include <map>
 #include <set>
 #include <string>
@@ -2275,6 +2276,149 @@ k_merge_fused(MergeFusedArgs M) {
 template __global__ void k_merge_fused<1024>(MergeFusedArgs);
 template __global__ void k_merge_fused<2048>(MergeFusedArgs);
 template __global__ void k_merge_fused<4096>(MergeFusedArgs);
+
+/* Dual-window fused merge (COUNT-shaped panes only): two CONSECUTIVE
+ * sliding windows fired in the same watermark group share all but one
+ * source pane, so one kernel scans the union (6 pane segments instead of
+ * 10) and maintains both windows' counts per key in LDS.  A key emits a
+ * row per window where it has a nonzero count (COUNT >= 1 iff the key
+ * occurs in any of that window's panes, so presence needs no extra
+ * bits).  Emissions match two k_merge_fused launches up to row order
+ * within the call (the batched-watermark API returns the group's fires
+ * concatenated; order within the batch is not part of the contract). */
+struct MergeDualArgs {
+    DeviceRing ring;
+    int32_t n_src;
+    uint32_t src[16];
+    uint64_t maskA, maskB;    /* src-index bitmasks of each window */
+    uint64_t wsA, weA, wsB, weB;
+    int32_t n_keys;
+    uint32_t range;
+    uint32_t cpi_range;
+    int64_t *out[8];
+    unsigned long long *n_out;
+    unsigned long long *accum;
+    const uint64_t *cpi_entries[16];
+    const uint32_t *cpi_cnt[16];
+};
+
+template <int SLOTS>
+__global__ void __launch_bounds__(256)
+k_merge_dual(MergeDualArgs M) {
+    __shared__ int64_t lkey[SLOTS];
+    extern __shared__ uint64_t lst[];        /* [2][SLOTS] counts */
+    __shared__ unsigned long long blk_base;
+    __shared__ unsigned int blk_cnt;
+    uint64_t *cA = lst, *cB = lst + SLOTS;
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
+        lkey[i] = EMPTY_KEY;
+        cA[i] = 0;
+        cB[i] = 0;
+    }
+    if (threadIdx.x == 0) blk_cnt = 0;
+    __syncthreads();
+    const uint32_t nsub = M.range / M.cpi_range;
+    for (int p = 0; p < M.n_src; p++) {
+        const int inA = (int)((M.maskA >> p) & 1);
+        const int inB = (int)((M.maskB >> p) & 1);
+        auto fold = [&](int64_t key, uint64_t w0) {
+            if (key == EMPTY_KEY) return;
+            uint32_t h = (uint32_t)hash64((uint64_t)key * 0x9e3779b1u) &
+                         (SLOTS - 1);
+            int64_t slot = -1;
+            for (int pr = 0; pr < SLOTS && slot < 0; pr++) {
+                uint32_t sidx = (h + pr) & (SLOTS - 1);
+                int64_t k = lkey[sidx];
+                if (k == key) {
+                    slot = sidx;
+                } else if (k == EMPTY_KEY) {
+                    int64_t old = (int64_t)atomicCAS(
+                        (unsigned long long *)&lkey[sidx],
+                        (unsigned long long)EMPTY_KEY,
+                        (unsigned long long)key);
+                    if (old == EMPTY_KEY || old == key) slot = sidx;
+                }
+            }
+            if (slot < 0) { *M.ring.err = ERR_MF_OVERFLOW; return; }
+            if (inA)
+                atomicAdd((unsigned long long *)&cA[slot],
+                          (unsigned long long)w0);
+            if (inB)
+                atomicAdd((unsigned long long *)&cB[slot],
+                          (unsigned long long)w0);
+        };
+        for (uint32_t j = 0; j < nsub; j++) {
+            uint32_t seg = blockIdx.x * nsub + j;
+            const uint32_t n_e = M.cpi_cnt[p][seg];
+            const uint64_t *ent = M.cpi_entries[p] +
+                                  (size_t)seg * M.cpi_range * 2;
+            for (uint32_t t = threadIdx.x; t < n_e; t += blockDim.x) {
+                ulonglong2 e = ((const ulonglong2 *)ent)[t];
+                fold((int64_t)e.x, (uint64_t)e.y);
+            }
+        }
+    }
+    __syncthreads();
+    unsigned int mine = 0;
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x)
+        if (lkey[i] != EMPTY_KEY)
+            mine += (cA[i] > 0) + (cB[i] > 0);
+    for (int off = 32; off; off >>= 1)
+        mine += (unsigned)__shfl_down((int)mine, off, 64);
+    if ((threadIdx.x & 63) == 0 && mine) atomicAdd(&blk_cnt, mine);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        blk_base = blk_cnt ? atomicAdd(M.n_out,
+                                       (unsigned long long)blk_cnt)
+                           : 0;
+        if (M.accum && blk_cnt)
+            atomicAdd(M.accum, (unsigned long long)blk_cnt);
+        blk_cnt = 0;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
+        int64_t key = lkey[i];
+        if (key == EMPTY_KEY) continue;
+        for (int w = 0; w < 2; w++) {
+            uint64_t c = w ? cB[i] : cA[i];
+            if (!c) continue;
+            int64_t r = (int64_t)(blk_base + atomicAdd(&blk_cnt, 1u));
+            int col = 0;
+            if (M.n_keys) M.out[col++][r] = key;
+            M.out[col++][r] = (int64_t)c;
+            M.out[col++][r] = (int64_t)(w ? M.wsB : M.wsA);
+            M.out[col++][r] = (int64_t)(w ? M.weB : M.weA);
+            M.out[col][r] = (int64_t)((w ? M.weB : M.weA) - 1);
+        }
+    }
+    /* special (key == sentinel) entries */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        for (int w = 0; w < 2; w++) {
+            uint64_t mask = w ? M.maskB : M.maskA;
+            uint64_t sum = 0;
+            int any = 0;
+            for (int p = 0; p < M.n_src; p++) {
+                if (!((mask >> p) & 1)) continue;
+                uint32_t pp = M.src[p];
+                if (!M.ring.spec_used[pp]) continue;
+                any = 1;
+                /* dual is COUNT-shaped only: na == 1, stride 2 words */
+                sum += M.ring.spec_state[(size_t)pp * 2];
+            }
+            if (!any) continue;
+            int64_t r = (int64_t)atomicAdd(M.n_out, 1ULL);
+            if (M.accum) atomicAdd(M.accum, 1ULL);
+            int col = 0;
+            if (M.n_keys) M.out[col++][r] = EMPTY_KEY;
+            M.out[col++][r] = (int64_t)sum;
+            M.out[col++][r] = (int64_t)(w ? M.wsB : M.wsA);
+            M.out[col++][r] = (int64_t)(w ? M.weB : M.weA);
+            M.out[col][r] = (int64_t)((w ? M.weB : M.weA) - 1);
+        }
+    }
+}
+
+template __global__ void k_merge_dual<2048>(MergeDualArgs);
 template __global__ void k_merge_fused<1024, true>(MergeFusedArgs);
 template __global__ void k_merge_fused<2048, true>(MergeFusedArgs);
 
@@ -2453,6 +2597,12 @@ struct GpuOp {
                                  before the slot's bin comes around again */
     uint64_t fire_group;
     std::vector<uint32_t> retire_pend;  /* slots awaiting the tag clear */
+    /* deferred fires within one batched watermark call, so consecutive
+     * windows can be paired into k_merge_dual (see the kernel's doc) */
+    struct PendFire { std::vector<uint32_t> src; uint64_t ws, we; };
+    std::vector<PendFire> pend_fires;
+    std::vector<std::pair<uint32_t, uint64_t>> pend_retires;
+    int defer_fires;
     int spin;                 /* busy-wait host syncs (ARROYO_AMD_SPIN) */
     hipEvent_t ev_sync;       /* status-copy completion (spin target) */
     /* epoch pipeline (mark_epoch / handle_watermarks_epoch): per-epoch
@@ -3700,6 +3850,127 @@ static int fire_window(GpuOp *o, const std::vector<uint32_t> &src,
     return gate_fire(o);
 }
 
+/* pair launch: one k_merge_dual over the union of two consecutive
+ * windows' panes (COUNT shape; see the kernel) */
+static int fire_dual(GpuOp *o, const GpuOp::PendFire &A,
+                     const GpuOp::PendFire &B) {
+    hipStream_t fs = o->fstream;
+    if (o->own_fstream)
+        HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_cpi, 0));
+    if ((o->fire_seq & 63) == 0)
+        HIP_CHECK(o, hipMemsetAsync(o->d_fire_cur, 0, 64 * 8, fs));
+    unsigned long long *cur = o->d_fire_cur + (o->fire_seq & 63);
+    o->fire_seq++;
+    o->fire_alt++;
+    MergeDualArgs M = {};
+    M.ring = o->ring;
+    std::vector<uint32_t> uni = A.src;
+    for (uint32_t sv : B.src)
+        if (std::find(uni.begin(), uni.end(), sv) == uni.end())
+            uni.push_back(sv);
+    M.n_src = (int)uni.size();
+    for (size_t i = 0; i < uni.size(); i++) {
+        M.src[i] = uni[i];
+        if (std::find(A.src.begin(), A.src.end(), uni[i]) != A.src.end())
+            M.maskA |= 1ULL << i;
+        if (std::find(B.src.begin(), B.src.end(), uni[i]) != B.src.end())
+            M.maskB |= 1ULL << i;
+        M.cpi_entries[i] = o->cpi_entries +
+                           (size_t)uni[i] * o->ring.C * o->cpi_ew;
+        M.cpi_cnt[i] = o->cpi_cnt + (size_t)uni[i] * o->cpi_nr;
+    }
+    M.wsA = A.ws;
+    M.weA = A.we;
+    M.wsB = B.ws;
+    M.weB = B.we;
+    M.n_keys = o->cfg.n_keys ? 1 : 0;
+    M.range = o->mf_range;
+    M.cpi_range = o->cpi_range;
+    for (int i = 0; i < o->n_out_alloc && i < 8; i++) M.out[i] = o->d_out[i];
+    M.n_out = cur;
+    M.accum = o->cfg.emit_to_host ? nullptr : o->d_emitted;
+    size_t shmem = (size_t)2048 * 2 * 8;
+    hipLaunchKernelGGL(k_merge_dual<2048>, dim3(o->ring.C / o->mf_range),
+                       dim3(256), shmem, fs, M);
+    HIP_CHECK(o, hipGetLastError());
+    if (!o->cfg.emit_to_host) return 0;
+    unsigned long long n = 0;
+    HIP_CHECK(o, hipMemcpyAsync(&n, cur, 8, hipMemcpyDeviceToHost, fs));
+    HIP_CHECK(o, hipStreamSynchronize(fs));
+    if (n == 0) return 0;
+    if ((size_t)o->out_cols > o->host_out.size())
+        o->host_out.resize(o->out_cols);
+    for (int i = 0; i < o->out_cols; i++) {
+        size_t old = o->host_out[i].size();
+        o->host_out[i].resize(old + n);
+        HIP_CHECK(o, hipMemcpyAsync(o->host_out[i].data() + old,
+                                    o->d_out[i], n * 8,
+                                    hipMemcpyDeviceToHost, fs));
+    }
+    HIP_CHECK(o, hipStreamSynchronize(fs));
+    return 0;
+}
+
+/* can this op use k_merge_dual at all? */
+static int dual_ok(GpuOp *o) {
+    return o->cpi_entries && o->cpi_ew == 2 && !o->ring.packed && !o->mk &&
+           o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
+           !o->agg.isf[0];
+}
+
+/* fire (or queue, when a batched call may pair it) one window */
+static int queue_fire(GpuOp *o, const std::vector<uint32_t> &src,
+                      uint64_t ws, uint64_t we) {
+    if (!o->defer_fires)
+        return fire_window(o, src, ws, we, 0, 0);
+    o->pend_fires.push_back({src, ws, we});
+    return 0;
+}
+
+static int queue_retire(GpuOp *o, uint32_t slot, uint64_t bin) {
+    if (!o->defer_fires) return ring_retire(o, slot, bin);
+    o->pend_retires.push_back({slot, bin});
+    return 0;
+}
+
+/* launch queued fires (pairing consecutive windows whose panes all have
+ * a closed-pane index) then the queued retires */
+static int flush_pend(GpuOp *o) {
+    size_t i = 0;
+    while (i < o->pend_fires.size()) {
+        GpuOp::PendFire &a = o->pend_fires[i];
+        int paired = 0;
+        if (i + 1 < o->pend_fires.size()) {
+            GpuOp::PendFire &b = o->pend_fires[i + 1];
+            int ready = a.src.size() + b.src.size() <= 16 &&
+                        !a.src.empty() && !b.src.empty() &&
+                        b.ws == a.ws + o->slide && b.we == a.we + o->slide;
+            if (ready)
+                for (uint32_t sv : a.src)
+                    if (!(sv < o->cpi_ready.size() && o->cpi_ready[sv]))
+                        ready = 0;
+            if (ready)
+                for (uint32_t sv : b.src)
+                    if (!(sv < o->cpi_ready.size() && o->cpi_ready[sv]))
+                        ready = 0;
+            if (ready) {
+                if (fire_dual(o, a, b)) return 1;
+                i += 2;
+                paired = 1;
+            }
+        }
+        if (!paired) {
+            if (fire_window(o, a.src, a.ws, a.we, 0, 0)) return 1;
+            i += 1;
+        }
+    }
+    o->pend_fires.clear();
+    for (auto &pr : o->pend_retires)
+        if (ring_retire_planes(o, pr.first, pr.second)) return 1;
+    o->pend_retires.clear();
+    return ring_retire_flush(o);
+}
+
 /* advance(): sliding_aggregating_window.rs:115-210, host replica */
 static int advance(GpuOp *o) {
     uint64_t b = (o->state == 1) ? o->earliest : o->next_start;
@@ -3733,13 +4004,13 @@ static int advance(GpuOp *o) {
     for (auto &kv : o->closed)
         if (kv.first < del) dead.push_back(kv.first);
 
-    if (fire_window(o, src, E - o->width, E, 0, 0)) return 1;
+    if (queue_fire(o, src, E - o->width, E)) return 1;
 
     for (uint64_t bb : dead) {
-        if (ring_retire_planes(o, o->closed[bb], bb)) return 1;
+        if (queue_retire(o, o->closed[bb], bb)) return 1;
         o->closed.erase(bb);
     }
-    if (ring_retire_flush(o)) return 1;
+    if (!o->defer_fires && ring_retire_flush(o)) return 1;
 
     if (o->closed.empty()) {
         if (!o->table_bins.empty()) {
@@ -3877,7 +4148,11 @@ static int wm_fold_status(GpuOp *o) {
 API int arroyo_amd_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
     GpuOp *o = (GpuOp *)h;
     if (wm_fold_status(o)) return 1;
-    if (wm_advance(o, wm)) return 1;
+    o->defer_fires = dual_ok(o);
+    int rc = wm_advance(o, wm);
+    if (!rc) rc = flush_pend(o);
+    o->defer_fires = 0;
+    if (rc) return 1;
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
@@ -3892,8 +4167,12 @@ API int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms,
     GpuOp *o = (GpuOp *)h;
     if (n <= 0) return 0;
     if (wm_fold_status(o)) return 1;
-    for (int32_t i = 0; i < n; i++)
-        if (wm_advance(o, wms[i])) return 1;
+    o->defer_fires = dual_ok(o);
+    int rc = 0;
+    for (int32_t i = 0; i < n && !rc; i++) rc = wm_advance(o, wms[i]);
+    if (!rc) rc = flush_pend(o);
+    o->defer_fires = 0;
+    if (rc) return 1;
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
@@ -3976,8 +4255,12 @@ API int arroyo_amd_handle_watermarks_epoch(void *h, const uint64_t *wms,
         HIP_CHECK(o, hipStreamWaitEvent(o->fstream, o->ev_epoch[e], 0));
         HIP_CHECK(o, hipStreamWaitEvent(o->fstream2, o->ev_epoch[e], 0));
     }
-    for (int32_t i = 0; i < n; i++)
-        if (wm_advance(o, wms[i])) return 1;
+    o->defer_fires = dual_ok(o);
+    int rc = 0;
+    for (int32_t i = 0; i < n && !rc; i++) rc = wm_advance(o, wms[i]);
+    if (!rc) rc = flush_pend(o);
+    o->defer_fires = 0;
+    if (rc) return 1;
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
