@@ -3913,8 +3913,13 @@ static int fire_dual(GpuOp *o, const GpuOp::PendFire &A,
 
 /* can this op use k_merge_dual at all? */
 static int dual_ok(GpuOp *o) {
-    return o->cpi_entries && o->cpi_ew == 2 && !o->ring.packed && !o->mk &&
-           o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
+    static int on = -1;
+    if (on < 0) {
+        const char *ev = getenv("ARROYO_AMD_DUAL");
+        on = ev ? atoi(ev) != 0 : 1;
+    }
+    return on && o->cpi_entries && o->cpi_ew == 2 && !o->ring.packed &&
+           !o->mk && o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
            !o->agg.isf[0];
 }
 
