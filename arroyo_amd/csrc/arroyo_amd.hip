@@ -3030,10 +3030,24 @@ API void *arroyo_amd_create(const AmdWindowConfig *cfg) {
     hipDeviceGetStreamPriorityRange(&pr_lo, &pr_hi);
     hipStreamCreateWithPriority(&o->stream, hipStreamNonBlocking, pr_hi);
     if (o->own_fstream) {
-        hipStreamCreateWithPriority(&o->fstream, hipStreamNonBlocking,
-                                    pr_lo);
-        hipStreamCreateWithPriority(&o->fstream2, hipStreamNonBlocking,
-                                    pr_lo);
+        /* ARROYO_AMD_FIRE_CUS=N confines both fire streams to the last N
+         * CUs so update wavefronts own the rest (hard partition via CU
+         * mask); 0/unset = share the whole chip under priorities */
+        int fire_cus = 0;
+        if (const char *ev = getenv("ARROYO_AMD_FIRE_CUS"))
+            fire_cus = atoi(ev);
+        if (fire_cus > 0 && fire_cus < 256) {
+            uint32_t mask[8] = {};
+            for (int b = 256 - fire_cus; b < 256; b++)
+                mask[b / 32] |= 1u << (b % 32);
+            hipExtStreamCreateWithCUMask(&o->fstream, 8, mask);
+            hipExtStreamCreateWithCUMask(&o->fstream2, 8, mask);
+        } else {
+            hipStreamCreateWithPriority(&o->fstream, hipStreamNonBlocking,
+                                        pr_lo);
+            hipStreamCreateWithPriority(&o->fstream2, hipStreamNonBlocking,
+                                        pr_lo);
+        }
         hipEventCreateWithFlags(&o->ev_gate, hipEventDisableTiming);
         hipEventCreateWithFlags(&o->ev_f1, hipEventDisableTiming);
         hipEventCreateWithFlags(&o->ev_f2, hipEventDisableTiming);
