@@ -55,3 +55,48 @@ def test_create_without_gpu_fails_loudly():
     assert not h, "create must fail without a HIP device (no CPU fallback)"
     msg = fn["last_error"](None).decode()
     assert "hip" in msg.lower() or "device" in msg.lower()
+
+
+def test_oracle_handle_watermarks_fallback():
+    """WindowOp.handle_watermarks falls back to a sequential Python loop
+    on libraries without the batched export (the oracle): emissions must
+    equal per-watermark calls, concatenated."""
+    import numpy as np
+
+    import oracle
+    from arroyo_amd import cabi as c
+
+    def stream(op, batched):
+        rng = np.random.default_rng(11)
+        outs = []
+        for i in range(6):
+            ts = (1_600_000_000 * 10**9 +
+                  np.arange(2000, dtype=np.int64) * 10**6 +
+                  i * 2 * 10**9)
+            key = rng.integers(0, 50, size=2000).astype(np.int64)
+            op.process_batch([key, ts])
+            wms = [int(ts[-1]) - 10**9]
+            if batched:
+                out = op.handle_watermarks(wms)
+                if out and len(out[0]):
+                    outs.append(out)
+            else:
+                for w in wms:
+                    out = op.handle_watermark(w)
+                    if out and len(out[0]):
+                        outs.append(out)
+        return outs
+
+    cfg = lambda: c.make_config(width_ns=4 * 10**9, slide_ns=2 * 10**9,
+                                n_keys=1, n_value_cols=0,
+                                aggs=[(c.COUNT, -1)], log2_capacity=12)
+    a = oracle.make_op(cfg())
+    b = oracle.make_op(cfg())
+    sa = stream(a, False)
+    sb = stream(b, True)
+    a.close()
+    b.close()
+    assert len(sa) == len(sb)
+    for x, y in zip(sa, sb):
+        for cx, cy in zip(x, y):
+            assert np.array_equal(cx, cy)
