@@ -129,11 +129,21 @@ def run_gpu(args):
     views = [(d_key[i * BATCH_ROWS:(i + 1) * BATCH_ROWS],
               d_ts[i * BATCH_ROWS:(i + 1) * BATCH_ROWS])
              for i in range(BASE_BATCHES)]
-    # shuffle scratch (N>1): partitioned send buffers + receive buffers
+    # shuffle scratch (N>1): partitioned send buffers + receive buffers.
+    # The exchange runs once per fused watermark period (not per 64K
+    # batch): one partition kernel + one size gather + one all_to_all
+    # over the period's contiguous ring slice -- collective count drops
+    # ~60x and the update launches stay period-fused like N=1.  Sized for
+    # a full period; (world+1)x headroom because the nexmark hot key
+    # sends ~half of every rank's rows to one owner.
     if world > 1:
-        sk = torch.empty(BATCH_ROWS, dtype=torch.int64, device=dev)
-        st = torch.empty(BATCH_ROWS, dtype=torch.int64, device=dev)
-        rbuf_cap = (world + 1) * BATCH_ROWS  # hot keys can skew the exchange
+        wm_every_est = int(NS // (span // BASE_BATCHES)) + 1
+        period_rows = (wm_every_est *
+                       int(os.environ.get("BENCH_WM_FUSE", "4")) *
+                       BATCH_ROWS)
+        sk = torch.empty(period_rows, dtype=torch.int64, device=dev)
+        st = torch.empty(period_rows, dtype=torch.int64, device=dev)
+        rbuf_cap = (world + 1) * period_rows
         rk = torch.empty(rbuf_cap, dtype=torch.int64, device=dev)
         rt = torch.empty(rbuf_cap, dtype=torch.int64, device=dev)
 
@@ -231,6 +241,58 @@ def run_gpu(args):
     run_span.last_wm = 0
     run_span.pending = None
 
+    def run_span_dist(s_begin, n_steps):
+        """N>1 fast path: the period's contiguous ring slice is
+        partitioned and exchanged in ONE collective round, then ingested
+        as one fused multi-batch launch; watermarks emit batched."""
+        from arroyo_amd.shuffle import recv_splits_of
+        s, end = s_begin, s_begin + n_steps
+        while s < end:
+            period = wm_every * wm_fuse
+            take = min(period - (s % period), end - s)
+            while take:
+                b = s % BASE_BATCHES
+                sub = min(take, BASE_BATCHES - b)
+                nrows = sub * BATCH_ROWS
+                ts_off = (s // BASE_BATCHES) * span
+                # fence the op's consumers of the previous exchange
+                wm_lib.arroyo_amd_sync(op._h)
+                counts = gpu.partition_device(
+                    d_key.data_ptr() + b * BATCH_ROWS * 8, 0,
+                    d_ts.data_ptr() + b * BATCH_ROWS * 8, nrows, world,
+                    sk.data_ptr(), 0, st.data_ptr())
+                send_splits = [int(c) for c in counts]
+                tsizes = torch.tensor(send_splits, dtype=torch.int64,
+                                      device=dev)
+                gathered = torch.empty(world * world, dtype=torch.int64,
+                                       device=dev)
+                dist.all_gather_into_tensor(gathered, tsizes)
+                recv_splits = recv_splits_of(gathered.cpu().tolist(),
+                                             world, rank)
+                n_recv = sum(recv_splits)
+                if n_recv > rbuf_cap:
+                    raise RuntimeError(
+                        f"rank {rank}: skewed exchange overflows the "
+                        f"receive buffer ({n_recv} rows > cap "
+                        f"{rbuf_cap}); raise the rbuf_cap headroom")
+                dist.all_to_all_single(rk[:n_recv], sk[:nrows],
+                                       output_split_sizes=recv_splits,
+                                       input_split_sizes=send_splits)
+                dist.all_to_all_single(rt[:n_recv], st[:nrows],
+                                       output_split_sizes=recv_splits,
+                                       input_split_sizes=send_splits)
+                torch.cuda.current_stream().synchronize()
+                if n_recv:
+                    op.process_batch_device([rk.data_ptr(), rt.data_ptr()],
+                                            n_recv, ts_off)
+                s += sub
+                take -= sub
+            first_unemitted = (run_span.last_wm // wm_every + 1) * wm_every
+            group = list(range(first_unemitted, s + 1, wm_every))
+            if group:
+                emit_watermarks(group)
+                run_span.last_wm = group[-1]
+
     def one_step(step):
         b = step % BASE_BATCHES
         cycle = step // BASE_BATCHES
@@ -283,11 +345,14 @@ def run_gpu(args):
     clock = WatermarkClock()
     torch.cuda.synchronize()
     tw0 = time.perf_counter()
+    step_exchange = os.environ.get("BENCH_STEP_EXCHANGE") == "1"
     if world == 1:
         run_span(0, args.warmup)
-    else:
+    elif step_exchange:
         for s in range(args.warmup):
             one_step(s)
+    else:
+        run_span_dist(0, args.warmup)
     torch.cuda.synchronize()
     tw1 = time.perf_counter()
     op.perf()  # reset kernel-time counters after warmup
@@ -310,9 +375,11 @@ def run_gpu(args):
         t0 = time.perf_counter()
         if world == 1:
             run_span(s_cursor, chunk)
-        else:
+        elif step_exchange:
             for s in range(s_cursor, s_cursor + chunk):
                 one_step(s)
+        else:
+            run_span_dist(s_cursor, chunk)
         torch.cuda.synchronize()
         if dist:
             dist.barrier()

@@ -360,3 +360,43 @@ def test_exchange_split_bookkeeping_simulation():
             assert rs == [int(sends[src][rank]) for src in range(world)]
             total_recv += sum(rs)
         assert total_recv == int(sends.sum())
+
+
+def test_period_fused_exchange_equals_per_batch():
+    """The N>1 bench exchanges once per fused watermark period (one
+    partition + one all_to_all over the period's contiguous slice) instead
+    of per 64K batch.  Simulation: for each rank, the multiset of rows it
+    receives over a period must be identical either way, and within-period
+    arrival order cannot change window results (the same late-filter
+    argument as watermark fusion, tests/test_property_large.py)."""
+    import numpy as np
+
+    def splitmix64(x):
+        x = (x + 0x9E3779B97F4A7C15) & (2**64 - 1)
+        x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & (2**64 - 1)
+        x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & (2**64 - 1)
+        return x ^ (x >> 31)
+
+    rng = np.random.default_rng(17)
+    world, batches, rows = 3, 6, 500
+    streams = [rng.integers(0, 10_000, size=batches * rows).astype(np.int64)
+               for _ in range(world)]
+    rng_size = 2**64 // world
+
+    def owner(k):
+        return int(splitmix64(int(np.uint64(k))) // rng_size)
+
+    # per-batch exchange: each destination's received keys, batch by batch
+    per_batch = [[] for _ in range(world)]
+    for b in range(batches):
+        for r in range(world):
+            seg = streams[r][b * rows:(b + 1) * rows]
+            for k in seg:
+                per_batch[owner(k)].append(int(k))
+    # period-fused: one partition of each rank's whole period slice
+    fused = [[] for _ in range(world)]
+    for r in range(world):
+        for k in streams[r]:
+            fused[owner(k)].append(int(k))
+    for d in range(world):
+        assert sorted(per_batch[d]) == sorted(fused[d])
