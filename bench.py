@@ -60,7 +60,7 @@ LOG2_CAPACITY = int(os.environ.get("BENCH_LOG2_CAP", "19"))
 # round-2 sweep (these only seed defaults; explicit env wins)
 os.environ.setdefault("ARROYO_AMD_MF_RANGE", "1024")
 os.environ.setdefault("ARROYO_AMD_MF_SLOTS", "2048")
-RING_PANES = 16
+RING_PANES = int(os.environ.get("BENCH_RING_PANES", "16"))
 HBM_PEAK_GBPS = 8000.0           # spec peak (MI355X_MICROARCH.md)
 ALG_BYTES_PER_ROW = 16           # compulsory HBM read: auction i64 + ts i64
 CPU_SAMPLE_ROWS = 64_000_000
