@@ -4175,7 +4175,12 @@ API int arroyo_amd_handle_watermark(void *h, uint64_t wm, AmdOutBatch *out) {
     int rc = wm_advance(o, wm);
     if (!rc) rc = flush_pend(o);
     o->defer_fires = 0;
-    if (rc) return 1;
+    if (rc) {
+        o->pend_fires.clear();   /* a failed group must not leak deferred
+                                    work into the next call */
+        o->pend_retires.clear();
+        return 1;
+    }
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
@@ -4195,7 +4200,11 @@ API int arroyo_amd_handle_watermarks(void *h, const uint64_t *wms,
     for (int32_t i = 0; i < n && !rc; i++) rc = wm_advance(o, wms[i]);
     if (!rc) rc = flush_pend(o);
     o->defer_fires = 0;
-    if (rc) return 1;
+    if (rc) {
+        o->pend_fires.clear();
+        o->pend_retires.clear();
+        return 1;
+    }
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
@@ -4283,7 +4292,11 @@ API int arroyo_amd_handle_watermarks_epoch(void *h, const uint64_t *wms,
     for (int32_t i = 0; i < n && !rc; i++) rc = wm_advance(o, wms[i]);
     if (!rc) rc = flush_pend(o);
     o->defer_fires = 0;
-    if (rc) return 1;
+    if (rc) {
+        o->pend_fires.clear();
+        o->pend_retires.clear();
+        return 1;
+    }
     if (fire_tail(o)) return 1;
     if (out) return build_out(o, out, 0);
     return 0;
