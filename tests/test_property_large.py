@@ -153,4 +153,46 @@ def test_q5_fused_watermark_submission_identical():
                      for i in range(len(cols[0]))]
         return sorted(rows)
 
-    assert run(1) == run(2) == run(4)
+    def run_epoch(fuse):
+        """the bench's actual submission shape: fused periods, epoch
+        marked after the period's rows, next period submitted BEFORE the
+        previous epoch's watermarks fold"""
+        op = gpu.make_op(cabi.make_config(
+            width_ns=WIDTH, slide_ns=SLIDE, n_keys=1, n_value_cols=0,
+            aggs=[(cabi.COUNT, -1)], log2_capacity=19, ring_panes=16,
+            emit_to_host=True))
+        outs = []
+        pending = []
+        deferred = None
+        wm_last = 0
+        for b in range(n // batch):
+            op.process_batches_device(
+                [d_key.data_ptr() + b * batch * 8,
+                 d_ts.data_ptr() + b * batch * 8], batch, 1,
+                contiguous=True)
+            mx = int(ts[(b + 1) * batch - 1])
+            if mx - wm_last > NS:
+                wm_last = mx
+                pending.append(mx - NS)
+            if len(pending) >= fuse:
+                op.mark_epoch()
+                op.set_filter_watermark(pending[-1])
+                if deferred is not None:
+                    outs.append(op.handle_watermarks_epoch(deferred))
+                deferred = pending
+                pending = []
+        if deferred is not None:
+            outs.append(op.handle_watermarks_epoch(deferred))
+        for wm in pending:
+            outs.append(op.handle_watermark(wm))
+        outs.append(op.handle_watermark(U64MAX))
+        op.close()
+        rows = []
+        for cols in outs:
+            if cols is None or len(cols) == 0 or len(cols[0]) == 0:
+                continue
+            rows += [tuple(int(c[i]) for c in cols)
+                     for i in range(len(cols[0]))]
+        return sorted(rows)
+
+    assert run(1) == run(2) == run(4) == run_epoch(4)
