@@ -1167,6 +1167,31 @@ API int arroyo_amd_session_process_batch_device(void *h,
     return sess_submit(o, dcols, n_rows, ts_offset);
 }
 
+/* contiguous multi-batch submission (bench replay rings): one C call
+ * enqueues `reps` per-batch update+merge rounds -- the per-batch gap
+ * logic is unchanged (each batch's table merges into the store before
+ * the next batch's update), only the host boundary cost amortizes */
+API int arroyo_amd_session_process_batches_device(void *h,
+                                                  const int64_t *const *dcols,
+                                                  int32_t n_cols,
+                                                  int64_t n_rows,
+                                                  int32_t reps,
+                                                  uint64_t ts_offset) {
+    GpuSession *o = (GpuSession *)h;
+    if (n_cols != o->n_in_cols) {
+        snprintf(o->err_msg, sizeof o->err_msg, "expected %d cols, got %d",
+                 o->n_in_cols, n_cols);
+        return 1;
+    }
+    const int64_t *cols[8];
+    for (int32_t k = 0; k < reps; k++) {
+        for (int c = 0; c < n_cols; c++)
+            cols[c] = dcols[c] + (int64_t)k * n_rows;
+        if (sess_submit(o, cols, n_rows, ts_offset)) return 1;
+    }
+    return 0;
+}
+
 API int arroyo_amd_session_process_batch(void *h, const int64_t *const *cols,
                                          int32_t n_cols, int64_t n_rows) {
     GpuSession *o = (GpuSession *)h;
