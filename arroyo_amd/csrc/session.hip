@@ -546,6 +546,13 @@ struct Store {
      * lines per key (key probe + record) instead of the 5 the split
      * planes cost (profiles/r01_session_note.md). */
     uint32_t *ns;      /* [C+1] live sessions per key (dense) */
+    /* live-slot index: the watermark fire visits only slots with ns > 0
+     * instead of scanning all C+1 (the dense scan wasted 7/8 of its
+     * threads at the bench occupancy).  merge_partial appends a slot on
+     * its 0 -> n transition; the fire compacts survivors into the flip
+     * buffer (no duplicates: n reaches 0 only inside a fire). */
+    uint32_t *live;               /* append target for merges */
+    unsigned long long *live_n;
     int64_t *recs;     /* [(C+1)][rec_w] */
     uint32_t C, MS;
     uint32_t sess_w;   /* words per session = 2 + 2*n_aggs */
@@ -636,6 +643,10 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     wr[1] = ce;
     for (int k = 0; k < sw; k++) ((uint64_t *)(wr + 2))[k] = acc[k];
     S.ns[slot] = w + 1;
+    if (n == 0 && S.live) {
+        unsigned long long p = atomicAdd(S.live_n, 1ULL);
+        S.live[p] = (uint32_t)slot;
+    }
 }
 
 __global__ void __launch_bounds__(256)
@@ -723,6 +734,76 @@ k_sess_fire(FireArgs F) {
             }
         }
         if (w != n) S.ns[slot] = w;
+    }
+}
+
+struct FireIdxArgs {
+    FireArgs f;
+    const uint32_t *old_idx;
+    const unsigned long long *old_n;
+    uint32_t *new_idx;            /* survivors compact here */
+    unsigned long long *new_n;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_fire_idx(FireIdxArgs X) {
+    const FireArgs &F = X.f;
+    const Store &S = F.store;
+    int64_t total = (int64_t)*X.old_n;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         t < total; t += stride) {
+        int64_t slot = (int64_t)X.old_idx[t];
+        uint32_t n = S.ns[slot];
+        if (n == 0) continue;     /* emptied by an earlier fire pass */
+        int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+        uint32_t w = 0;
+        for (uint32_t i = 0; i < n; i++) {
+            int64_t *si = rec + (size_t)i * S.sess_w;
+            const uint64_t *sist = (const uint64_t *)(si + 2);
+            uint64_t close = (uint64_t)(si[1] + (int64_t)F.gap);
+            if (close < F.wm) {
+                int64_t r = (int64_t)atomicAdd(F.n_out, 1ULL);
+                if (r >= F.out_cap) { *F.err = SERR_OUT_CAP; continue; }
+                int col = 0;
+                if (F.n_keys) F.out[col++][r] = key;
+                for (int a = 0; a < F.agg.n_aggs; a++, col++) {
+                    uint64_t w0 = sist[2 * a];
+                    uint64_t w1 = sist[2 * a + 1];
+                    int64_t v;
+                    switch (F.agg.op[a]) {
+                    case AMD_AGG_COUNT:
+                    case AMD_AGG_SUM:
+                    case AMD_AGG_COUNT_DISTINCT: v = (int64_t)w0; break;
+                    case AMD_AGG_MIN: v = dec_min(w0); break;
+                    case AMD_AGG_MAX: v = dec_max(w0); break;
+                    default: {  /* AVG */
+                        double sum;
+                        memcpy(&sum, &w1, 8);
+                        double avg = sum / (double)w0;
+                        memcpy(&v, &avg, 8);
+                        break;
+                    }
+                    }
+                    F.out[col][r] = v;
+                }
+                F.out[col++][r] = si[0];
+                F.out[col++][r] = (int64_t)close;
+                F.out[col][r] = (int64_t)close - 1;
+            } else {
+                if (w != i) {
+                    int64_t *sd = rec + (size_t)w * S.sess_w;
+                    for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
+                }
+                w++;
+            }
+        }
+        if (w != n) S.ns[slot] = w;
+        if (w) {
+            unsigned long long p = atomicAdd(X.new_n, 1ULL);
+            X.new_idx[p] = (uint32_t)slot;
+        }
     }
 }
 
@@ -920,6 +1001,11 @@ struct GpuSession {
     int64_t stg_cap;
     int n_in_cols, out_cols, drain_cols;
     int64_t out_cap;
+    /* live-slot index flip buffers (see Store.live) */
+    uint32_t *live_buf[2];
+    unsigned long long *live_nbuf[2];
+    int live_cur;
+    int use_live;
     int has_wm; uint64_t wm;
     hipStream_t stream;
     char err_msg[512];
@@ -993,6 +1079,10 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     o->store.rec_w = (uint32_t)((MS * (2 + sw) + 7) & ~7u);
     SALLOC(o->store.keys, C1 * 8);
     SALLOC(o->store.ns, C1 * 4);
+    SALLOC(o->live_buf[0], C1 * 4);
+    SALLOC(o->live_buf[1], C1 * 4);
+    SALLOC(o->live_nbuf[0], 8);
+    SALLOC(o->live_nbuf[1], 8);
     SALLOC(o->store.recs, C1 * o->store.rec_w * 8);
     SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
     SALLOC(o->bst, ((size_t)o->B + 1) * (2 + sw) * 8);
@@ -1019,6 +1109,16 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
 #undef SALLOC
     hipMemset(o->store.keys, 0xFF, C1 * 8);
     hipMemset(o->store.ns, 0, C1 * 4);
+    hipMemset(o->live_nbuf[0], 0, 8);
+    hipMemset(o->live_nbuf[1], 0, 8);
+    o->live_cur = 0;
+    o->use_live = 1;
+    if (const char *ev = getenv("ARROYO_AMD_SESS_LIVE_IDX"))
+        o->use_live = atoi(ev) != 0;
+    if (o->use_live) {
+        o->store.live = o->live_buf[0];
+        o->store.live_n = o->live_nbuf[0];
+    }
     hipMemset(o->store.recs, 0, C1 * o->store.rec_w * 8);
     hipMemset(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8);
     hipMemset(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8);
@@ -1269,10 +1369,29 @@ API int arroyo_amd_session_handle_watermark(void *h, uint64_t wm,
     F.out_cap = o->out_cap;
     F.n_keys = o->cfg.n_keys;
     F.err = o->d_err;
-    hipLaunchKernelGGL(k_sess_fire,
-                       dim3(grid_for((int64_t)o->store.C + 1)), dim3(256), 0,
-                       o->stream, F);
-    SHIP(o, hipGetLastError());
+    if (o->use_live) {
+        /* fire over the live index, compacting survivors into the flip
+         * buffer; subsequent merges append there */
+        int nxt = 1 - o->live_cur;
+        SHIP(o, hipMemsetAsync(o->live_nbuf[nxt], 0, 8, o->stream));
+        FireIdxArgs X = {};
+        X.f = F;
+        X.old_idx = o->live_buf[o->live_cur];
+        X.old_n = o->live_nbuf[o->live_cur];
+        X.new_idx = o->live_buf[nxt];
+        X.new_n = o->live_nbuf[nxt];
+        hipLaunchKernelGGL(k_sess_fire_idx, dim3(2048), dim3(256), 0,
+                           o->stream, X);
+        SHIP(o, hipGetLastError());
+        o->live_cur = nxt;
+        o->store.live = o->live_buf[nxt];
+        o->store.live_n = o->live_nbuf[nxt];
+    } else {
+        hipLaunchKernelGGL(k_sess_fire,
+                           dim3(grid_for((int64_t)o->store.C + 1)),
+                           dim3(256), 0, o->stream, F);
+        SHIP(o, hipGetLastError());
+    }
     unsigned long long n = 0;
     SHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
                            o->stream));
@@ -1536,6 +1655,10 @@ API void arroyo_amd_session_destroy(void *h) {
     hipStreamSynchronize(o->stream);
     hipFree(o->store.keys);
     hipFree(o->store.ns);
+    hipFree(o->live_buf[0]);
+    hipFree(o->live_buf[1]);
+    hipFree(o->live_nbuf[0]);
+    hipFree(o->live_nbuf[1]);
     hipFree(o->store.recs);
     hipFree(o->bkeys);
     hipFree(o->bst);
