@@ -24,8 +24,7 @@
  *     interval form of the reference's strictly-within-gap row rule.  One
  *     thread per distinct key means no locks and no CAS loops on session
  *     data; only the key-slot claim is a CAS.
- *   - Watermark firing (k_sess_fire_idx): grid-stride over the live
- *     index, streaming the record arena; each lane scans its slot's
+ *   - Watermark firing (k_sess_fire): one thread per key slot scans its
  *     sessions and emits every session with data_end + gap < watermark
  *     (KeyComputingHolder::watermark_update :559-608) through a global
  *     output cursor; output columns [key?, finals..., window_start,
