@@ -553,12 +553,7 @@ struct Store {
      * buffer (no duplicates: n reaches 0 only inside a fire). */
     uint32_t *live;               /* append target for merges */
     unsigned long long *live_n;
-    uint32_t *pos;     /* [C+1] record arena position (valid when ns>0) */
-    int64_t *recs;     /* record ARENA [(C+1)][rec_w]: records sit at
-                          their live-index position, so the fire reads
-                          and compacts them as contiguous streams instead
-                          of ~2M random 64 B lines (the flip buffer swaps
-                          with the index each fire) */
+    int64_t *recs;     /* [(C+1)][rec_w] */
     uint32_t C, MS;
     uint32_t sess_w;   /* words per session = 2 + 2*n_aggs */
     uint32_t rec_w;    /* words per record  = MS*sess_w padded to 64B */
@@ -590,16 +585,8 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     int64_t slot = key_slot(S.keys, S.C, key, err, SERR_TABLE_FULL);
     if (slot < 0) return;
     int sw = 2 * M.agg.n_aggs;
+    int64_t *rec = S.recs + (size_t)slot * S.rec_w;
     uint32_t n = S.ns[slot];
-    uint32_t posv;
-    if (n == 0) {   /* first live session: claim index + arena position */
-        posv = (uint32_t)atomicAdd(S.live_n, 1ULL);
-        S.live[posv] = (uint32_t)slot;
-        S.pos[slot] = posv;
-    } else {
-        posv = S.pos[slot];
-    }
-    int64_t *rec = S.recs + (size_t)posv * S.rec_w;
     /* absorb every stored session whose gap-closure touches the partial's */
     int64_t cs = pmin, ce = pmax;
     uint64_t acc[AMD_MAX_AGGS * 2];
@@ -656,6 +643,10 @@ __device__ inline void merge_partial(const MergeArgs &M, int64_t key,
     wr[1] = ce;
     for (int k = 0; k < sw; k++) ((uint64_t *)(wr + 2))[k] = acc[k];
     S.ns[slot] = w + 1;
+    if (n == 0 && S.live) {
+        unsigned long long p = atomicAdd(S.live_n, 1ULL);
+        S.live[p] = (uint32_t)slot;
+    }
 }
 
 __global__ void __launch_bounds__(256)
@@ -689,34 +680,21 @@ struct FireArgs {
     int *err;
 };
 
-
-struct FireIdxArgs {
-    FireArgs f;                   /* f.store.recs/live/live_n = NEW side */
-    const uint32_t *old_idx;
-    const unsigned long long *old_n;
-    const int64_t *old_recs;
-    uint32_t *new_idx;            /* survivors compact here */
-    unsigned long long *new_n;
-    int64_t *new_recs;
-};
-
 __global__ void __launch_bounds__(256)
-k_sess_fire_idx(FireIdxArgs X) {
-    const FireArgs &F = X.f;
-    const Store &S = F.store;
-    int64_t total = (int64_t)*X.old_n;
+k_sess_fire(FireArgs F) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         t < total; t += stride) {
-        int64_t slot = (int64_t)X.old_idx[t];
+    const Store &S = F.store;
+    int sw = 2 * F.agg.n_aggs;
+    for (int64_t slot = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         slot <= (int64_t)S.C; slot += stride) {
         uint32_t n = S.ns[slot];
-        if (n == 0) continue;     /* emptied by an earlier fire pass */
+        if (n == 0) continue;
+        if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
-        const int64_t *rec = X.old_recs + (size_t)t * S.rec_w;
-        int64_t *dst = nullptr;   /* new arena record, claimed lazily */
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         uint32_t w = 0;
         for (uint32_t i = 0; i < n; i++) {
-            const int64_t *si = rec + (size_t)i * S.sess_w;
+            int64_t *si = rec + (size_t)i * S.sess_w;
             const uint64_t *sist = (const uint64_t *)(si + 2);
             uint64_t close = (uint64_t)(si[1] + (int64_t)F.gap);
             if (close < F.wm) {
@@ -748,18 +726,84 @@ k_sess_fire_idx(FireIdxArgs X) {
                 F.out[col++][r] = (int64_t)close;
                 F.out[col][r] = (int64_t)close - 1;
             } else {
-                if (!dst) {
-                    unsigned long long p = atomicAdd(X.new_n, 1ULL);
-                    X.new_idx[p] = (uint32_t)slot;
-                    S.pos[slot] = (uint32_t)p;
-                    dst = X.new_recs + (size_t)p * S.rec_w;
+                if (w != i) {
+                    int64_t *sd = rec + (size_t)w * S.sess_w;
+                    for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
                 }
-                int64_t *sd = dst + (size_t)w * S.sess_w;
-                for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
                 w++;
             }
         }
         if (w != n) S.ns[slot] = w;
+    }
+}
+
+struct FireIdxArgs {
+    FireArgs f;
+    const uint32_t *old_idx;
+    const unsigned long long *old_n;
+    uint32_t *new_idx;            /* survivors compact here */
+    unsigned long long *new_n;
+};
+
+__global__ void __launch_bounds__(256)
+k_sess_fire_idx(FireIdxArgs X) {
+    const FireArgs &F = X.f;
+    const Store &S = F.store;
+    int64_t total = (int64_t)*X.old_n;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         t < total; t += stride) {
+        int64_t slot = (int64_t)X.old_idx[t];
+        uint32_t n = S.ns[slot];
+        if (n == 0) continue;     /* emptied by an earlier fire pass */
+        int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
+        int64_t *rec = S.recs + (size_t)slot * S.rec_w;
+        uint32_t w = 0;
+        for (uint32_t i = 0; i < n; i++) {
+            int64_t *si = rec + (size_t)i * S.sess_w;
+            const uint64_t *sist = (const uint64_t *)(si + 2);
+            uint64_t close = (uint64_t)(si[1] + (int64_t)F.gap);
+            if (close < F.wm) {
+                int64_t r = (int64_t)atomicAdd(F.n_out, 1ULL);
+                if (r >= F.out_cap) { *F.err = SERR_OUT_CAP; continue; }
+                int col = 0;
+                if (F.n_keys) F.out[col++][r] = key;
+                for (int a = 0; a < F.agg.n_aggs; a++, col++) {
+                    uint64_t w0 = sist[2 * a];
+                    uint64_t w1 = sist[2 * a + 1];
+                    int64_t v;
+                    switch (F.agg.op[a]) {
+                    case AMD_AGG_COUNT:
+                    case AMD_AGG_SUM:
+                    case AMD_AGG_COUNT_DISTINCT: v = (int64_t)w0; break;
+                    case AMD_AGG_MIN: v = dec_min(w0); break;
+                    case AMD_AGG_MAX: v = dec_max(w0); break;
+                    default: {  /* AVG */
+                        double sum;
+                        memcpy(&sum, &w1, 8);
+                        double avg = sum / (double)w0;
+                        memcpy(&v, &avg, 8);
+                        break;
+                    }
+                    }
+                    F.out[col][r] = v;
+                }
+                F.out[col++][r] = si[0];
+                F.out[col++][r] = (int64_t)close;
+                F.out[col][r] = (int64_t)close - 1;
+            } else {
+                if (w != i) {
+                    int64_t *sd = rec + (size_t)w * S.sess_w;
+                    for (uint32_t k = 0; k < S.sess_w; k++) sd[k] = si[k];
+                }
+                w++;
+            }
+        }
+        if (w != n) S.ns[slot] = w;
+        if (w) {
+            unsigned long long p = atomicAdd(X.new_n, 1ULL);
+            X.new_idx[p] = (uint32_t)slot;
+        }
     }
 }
 
@@ -784,7 +828,7 @@ k_sess_drain(DrainArgs D) {
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
-        const int64_t *rec = S.recs + (size_t)S.pos[slot] * S.rec_w;
+        const int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         int64_t base = (int64_t)atomicAdd(D.n_out, (unsigned long long)n);
         if (base + n > D.out_cap) { *D.err = SERR_OUT_CAP; continue; }
         for (uint32_t i = 0; i < n; i++) {
@@ -865,7 +909,7 @@ k_sess_drain_values(CdDrainArgs D) {
         if (n == 0) continue;
         if (slot < (int64_t)S.C && S.keys[slot] == EMPTY_KEY) continue;
         int64_t key = slot == (int64_t)S.C ? EMPTY_KEY : S.keys[slot];
-        const int64_t *rec = S.recs + (size_t)S.pos[slot] * S.rec_w;
+        const int64_t *rec = S.recs + (size_t)slot * S.rec_w;
         for (uint32_t i = 0; i < n; i++) {
             const int64_t *si = rec + (size_t)i * S.sess_w;
             uint64_t reg1 = (uint64_t)si[2 + 2 * D.cd_agg + 1];
@@ -911,9 +955,8 @@ k_sess_restore_values(CdRestoreArgs R) {
             int64_t slot = key_slot(S.keys, S.C, key, R.err,
                                     SERR_TABLE_FULL);
             if (slot < 0) continue;
+            int64_t *rec = S.recs + (size_t)slot * S.rec_w;
             uint32_t n = S.ns[slot];
-            if (n == 0) continue;
-            int64_t *rec = S.recs + (size_t)S.pos[slot] * S.rec_w;
             for (uint32_t si = 0; si < n; si++) {
                 int64_t *sp = rec + (size_t)si * S.sess_w;
                 if (sp[0] != R.start[i])
@@ -958,11 +1001,11 @@ struct GpuSession {
     int64_t stg_cap;
     int n_in_cols, out_cols, drain_cols;
     int64_t out_cap;
-    /* live-slot index + record arena flip buffers (see Store.live) */
+    /* live-slot index flip buffers (see Store.live) */
     uint32_t *live_buf[2];
     unsigned long long *live_nbuf[2];
-    int64_t *rec_buf[2];
     int live_cur;
+    int use_live;
     int has_wm; uint64_t wm;
     hipStream_t stream;
     char err_msg[512];
@@ -1040,10 +1083,7 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     SALLOC(o->live_buf[1], C1 * 4);
     SALLOC(o->live_nbuf[0], 8);
     SALLOC(o->live_nbuf[1], 8);
-    SALLOC(o->store.pos, C1 * 4);
-    SALLOC(o->rec_buf[0], C1 * o->store.rec_w * 8);
-    SALLOC(o->rec_buf[1], C1 * o->store.rec_w * 8);
-    o->store.recs = o->rec_buf[0];
+    SALLOC(o->store.recs, C1 * o->store.rec_w * 8);
     SALLOC(o->bkeys, ((size_t)o->B + 1) * 8);
     SALLOC(o->bst, ((size_t)o->B + 1) * (2 + sw) * 8);
     int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
@@ -1072,9 +1112,14 @@ API void *arroyo_amd_session_create(const AmdSessionConfig *cfg) {
     hipMemset(o->live_nbuf[0], 0, 8);
     hipMemset(o->live_nbuf[1], 0, 8);
     o->live_cur = 0;
-    o->store.live = o->live_buf[0];
-    o->store.live_n = o->live_nbuf[0];
-    hipMemset(o->rec_buf[0], 0, C1 * o->store.rec_w * 8);
+    o->use_live = 1;
+    if (const char *ev = getenv("ARROYO_AMD_SESS_LIVE_IDX"))
+        o->use_live = atoi(ev) != 0;
+    if (o->use_live) {
+        o->store.live = o->live_buf[0];
+        o->store.live_n = o->live_nbuf[0];
+    }
+    hipMemset(o->store.recs, 0, C1 * o->store.rec_w * 8);
     hipMemset(o->bkeys, 0xFF, ((size_t)o->B + 1) * 8);
     hipMemset(o->bst, 0, ((size_t)o->B + 1) * (2 + sw) * 8);
     hipMemset(o->d_err, 0, 4);
@@ -1324,27 +1369,28 @@ API int arroyo_amd_session_handle_watermark(void *h, uint64_t wm,
     F.out_cap = o->out_cap;
     F.n_keys = o->cfg.n_keys;
     F.err = o->d_err;
-    {
-        /* fire over the live index, streaming the old record arena and
-         * compacting survivors (index + records) into the flip side;
-         * subsequent merges append there */
+    if (o->use_live) {
+        /* fire over the live index, compacting survivors into the flip
+         * buffer; subsequent merges append there */
         int nxt = 1 - o->live_cur;
         SHIP(o, hipMemsetAsync(o->live_nbuf[nxt], 0, 8, o->stream));
         FireIdxArgs X = {};
         X.f = F;
         X.old_idx = o->live_buf[o->live_cur];
         X.old_n = o->live_nbuf[o->live_cur];
-        X.old_recs = o->rec_buf[o->live_cur];
         X.new_idx = o->live_buf[nxt];
         X.new_n = o->live_nbuf[nxt];
-        X.new_recs = o->rec_buf[nxt];
         hipLaunchKernelGGL(k_sess_fire_idx, dim3(2048), dim3(256), 0,
                            o->stream, X);
         SHIP(o, hipGetLastError());
         o->live_cur = nxt;
         o->store.live = o->live_buf[nxt];
         o->store.live_n = o->live_nbuf[nxt];
-        o->store.recs = o->rec_buf[nxt];
+    } else {
+        hipLaunchKernelGGL(k_sess_fire,
+                           dim3(grid_for((int64_t)o->store.C + 1)),
+                           dim3(256), 0, o->stream, F);
+        SHIP(o, hipGetLastError());
     }
     unsigned long long n = 0;
     SHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
@@ -1609,13 +1655,11 @@ API void arroyo_amd_session_destroy(void *h) {
     hipStreamSynchronize(o->stream);
     hipFree(o->store.keys);
     hipFree(o->store.ns);
-    hipFree(o->store.pos);
     hipFree(o->live_buf[0]);
     hipFree(o->live_buf[1]);
     hipFree(o->live_nbuf[0]);
     hipFree(o->live_nbuf[1]);
-    hipFree(o->rec_buf[0]);
-    hipFree(o->rec_buf[1]);
+    hipFree(o->store.recs);
     hipFree(o->bkeys);
     hipFree(o->bst);
     int max_out = o->drain_cols > o->out_cols ? o->drain_cols : o->out_cols;
