@@ -1217,6 +1217,7 @@ k_update_batch_n(UpdateArgs A) {
 }
 
 template __global__ void k_update_batch_n<8>(UpdateArgs);
+template __global__ void k_update_batch_n<12>(UpdateArgs);
 template __global__ void k_update_batch_n<16>(UpdateArgs);
 
 /* ------------------------------------------------------------------ */
@@ -3288,6 +3289,9 @@ static int launch_update(GpuOp *o, const int64_t *const *dcols, int64_t n_rows,
             if (!pvec) bq = 1;
             if (bq >= 16)
                 hipLaunchKernelGGL(k_update_batch_n<16>, dim3(pblocks),
+                                   dim3(256), 0, o->stream, A);
+            else if (bq >= 12)
+                hipLaunchKernelGGL(k_update_batch_n<12>, dim3(pblocks),
                                    dim3(256), 0, o->stream, A);
             else if (bq >= 8)
                 hipLaunchKernelGGL(k_update_batch_n<8>, dim3(pblocks),
