@@ -135,3 +135,34 @@ def test_window_rejects_more_than_four_keys():
         oracle.make_op(cabi.make_config(
             width_ns=4 * NS, slide_ns=2 * NS, n_keys=5, n_value_cols=0,
             aggs=[(cabi.COUNT, -1)]))
+
+
+@pytest.mark.gpu
+def test_create_destroy_leak_free():
+    """50 create/destroy cycles of the window operator (which now owns
+    three streams, seven events, epoch buffers and flip cursors) must not
+    leak device memory beyond allocator noise."""
+    import torch
+
+    from arroyo_amd import cabi, gpu
+
+    def mk():
+        op = gpu.make_op(cabi.make_config(
+            width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=0,
+            aggs=[(cabi.COUNT, -1)], log2_capacity=14, ring_panes=16))
+        op.process_batch([
+            np.arange(1000, dtype=np.int64),
+            (1_600_000_000 * NS +
+             np.arange(1000, dtype=np.int64) * 10**6)])
+        op.handle_watermark(1_600_000_000 * NS + 30 * NS)
+        op.close()
+
+    mk()  # warm allocator/pools
+    torch.cuda.synchronize()
+    free0, _ = torch.cuda.mem_get_info()
+    for _ in range(50):
+        mk()
+    torch.cuda.synchronize()
+    free1, _ = torch.cuda.mem_get_info()
+    leaked = free0 - free1
+    assert leaked < 64 * 1024 * 1024, f"leaked ~{leaked/1e6:.1f} MB over 50 cycles"
