@@ -856,6 +856,22 @@ API int arroyo_amd_expjoin_collect(void *h, AmdOutBatch *out) {
     return 0;
 }
 
+/* device-resident consumption: report and reset the accumulated match
+ * count without copying the match columns to the host (the next pipeline
+ * stage consumes d_out in place -- the same accounting convention as the
+ * window operator's emitted-rows accumulator) */
+API int arroyo_amd_expjoin_match_count(void *h, int64_t *n_matches) {
+    GpuExpJoin *o = (GpuExpJoin *)h;
+    unsigned long long n = 0;
+    EJHIP(o, hipMemcpyAsync(&n, o->d_n_out, 8, hipMemcpyDeviceToHost,
+                            o->stream));
+    EJHIP(o, hipStreamSynchronize(o->stream));
+    if (ej_check_err(o)) return 1;
+    *n_matches = (int64_t)n;
+    EJHIP(o, hipMemsetAsync(o->d_n_out, 0, 8, o->stream));
+    return 0;
+}
+
 API int arroyo_amd_expjoin_handle_watermark(void *h, uint64_t wm) {
     GpuExpJoin *o = (GpuExpJoin *)h;
     o->has_wm = 1;

@@ -243,6 +243,11 @@ int arroyo_amd_expjoin_process_batch_device(void *h, int32_t side,
                                             int32_t n_cols, int64_t n_rows);
 int arroyo_amd_expjoin_collect(void *h, AmdOutBatch *out);
 int arroyo_amd_expjoin_handle_watermark(void *h, uint64_t watermark_nanos);
+
+/* Device-resident consumption: report and reset the accumulated match
+ * count without host copies (the next stage consumes the device match
+ * columns in place). */
+int arroyo_amd_expjoin_match_count(void *h, int64_t *n_matches);
 int arroyo_amd_expjoin_expire(void *h);
 int arroyo_amd_expjoin_checkpoint_drain(void *h, int32_t side,
                                         AmdOutBatch *out);
