@@ -166,3 +166,27 @@ def test_create_destroy_leak_free():
     free1, _ = torch.cuda.mem_get_info()
     leaked = free0 - free1
     assert leaked < 64 * 1024 * 1024, f"leaked ~{leaked/1e6:.1f} MB over 50 cycles"
+
+
+@pytest.mark.gpu
+def test_epoch_api_misuse_is_loud():
+    """Epoch-pipeline misuse must fail loudly: folding with no armed epoch,
+    and arming more than depth 2."""
+    from arroyo_amd import cabi, gpu
+
+    op = gpu.make_op(cabi.make_config(
+        width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=0,
+        aggs=[(cabi.COUNT, -1)], log2_capacity=12, ring_panes=16))
+    with pytest.raises(RuntimeError, match="no armed epoch"):
+        op.handle_watermarks_epoch([1_600_000_000 * NS])
+    op.mark_epoch()
+    op.mark_epoch()
+    with pytest.raises(RuntimeError, match="epoch queue full"):
+        op.mark_epoch()
+    # folding drains the queue in order and recovers the operator
+    out = op.handle_watermarks_epoch([1_600_000_000 * NS])
+    assert out == [] or len(out[0]) == 0
+    op.mark_epoch()   # queue has room again
+    op.handle_watermarks_epoch([1_600_000_001 * NS])
+    op.handle_watermarks_epoch([1_600_000_002 * NS])
+    op.close()
