@@ -3934,12 +3934,10 @@ static int dual_ok(GpuOp *o) {
     /* default OFF: at the bench balance the union table's higher LDS
      * load factor and doubled shared-pane atomics cost ~1.2% more than
      * the 10->6 pane-read saving buys (same-box A/B x2); the trade may
-     * flip for wider windows (width/slide > 5) where the sharing grows */
-    static int on = -1;
-    if (on < 0) {
-        const char *ev = getenv("ARROYO_AMD_DUAL");
-        on = ev ? atoi(ev) != 0 : 0;
-    }
+     * flip for wider windows (width/slide > 5) where the sharing grows.
+     * Read per call (not latched) so tests can exercise the path. */
+    const char *ev = getenv("ARROYO_AMD_DUAL");
+    int on = ev ? atoi(ev) != 0 : 0;
     return on && o->cpi_entries && o->cpi_ew == 2 && !o->ring.packed &&
            !o->mk && o->agg.n_aggs == 1 && o->agg.op[0] == AMD_AGG_COUNT &&
            !o->agg.isf[0];

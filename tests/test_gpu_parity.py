@@ -409,3 +409,36 @@ def test_small_ring_fused_groups_loud_or_correct():
     g.close()
     o.close()
     assert_parity(concat_outputs(got), concat_outputs(want))
+
+
+def test_dual_window_merge_matches_sequential(monkeypatch):
+    """The gated dual-window merge (ARROYO_AMD_DUAL=1: one kernel for two
+    consecutive fires in a batched watermark call) must be
+    emission-identical to the default single-window path."""
+    monkeypatch.setenv("ARROYO_AMD_DUAL", "1")
+    cols = nexmark.bids(500_000, events_per_sec=100_000, seed=37)
+    kw = dict(width_ns=10 * NS, slide_ns=2 * NS, n_keys=1, n_value_cols=0,
+              aggs=[(cabi.COUNT, -1)], log2_capacity=15, ring_panes=32)
+    a = gpu_op(**kw)
+    monkeypatch.setenv("ARROYO_AMD_DUAL", "0")
+    b = gpu_op(**kw)
+    batches = batches_from_columns(cols, 50_000)
+    outs = {0: [], 1: []}
+    pend = []
+    for i, cb in enumerate(batches):
+        a.process_batch(cb)
+        b.process_batch(cb)
+        pend.append(int(cb[-1][-1]) - NS)
+        if len(pend) == 4 or i == len(batches) - 1:
+            monkeypatch.setenv("ARROYO_AMD_DUAL", "1")
+            out = a.handle_watermarks(pend)
+            if out and len(out[0]):
+                outs[0].append(out)
+            monkeypatch.setenv("ARROYO_AMD_DUAL", "0")
+            out = b.handle_watermarks(pend)
+            if out and len(out[0]):
+                outs[1].append(out)
+            pend = []
+    a.close()
+    b.close()
+    assert_parity(concat_outputs(outs[0]), concat_outputs(outs[1]))
