@@ -121,3 +121,103 @@ def test_gpu_multikey_checkpoint_roundtrip():
     g = {tuple(int(col[r]) for col in got) for r in range(len(got[0]))}
     w = {tuple(int(col[r]) for col in want) for r in range(len(want[0]))}
     assert g == w
+
+
+@pytest.mark.gpu
+def test_gpu_multikey_f64_values_match_oracle():
+    """Composite keys AND f64 value columns together: the device key
+    dictionary and the f64 state encodings interact nowhere, which this
+    pins (SUM/MIN/AVG over f64 bit-pattern values, 2 key columns)."""
+    from arroyo_amd import gpu
+
+    rng = np.random.default_rng(23)
+    n = 120_000
+    keys = [rng.integers(0, 7, size=n).astype(np.int64) - 3,
+            rng.integers(0, 5, size=n).astype(np.int64)]
+    fval = rng.normal(0.0, 100.0, size=n)
+    val = fval.view(np.int64).copy()
+    ts = T0 + (np.arange(n, dtype=np.int64) * NS) // 25_000
+    kw = dict(width_ns=4 * NS, slide_ns=2 * NS, n_keys=2, n_value_cols=1,
+              aggs=[(cabi.COUNT, -1), (cabi.SUM, 0), (cabi.MIN, 0),
+                    (cabi.AVG, 0)],
+              val_is_f64=(0,), log2_capacity=13, ring_panes=16)
+    g = gpu.make_op(cabi.make_config(**kw))
+    got = run_op(g, keys, val, ts)
+    o = oracle.make_op(cabi.make_config(**kw))
+    want = run_op(o, keys, val, ts)
+    assert got is not None and want is not None
+    assert len(got[0]) == len(want[0])
+    # [k0, k1, count, sum(f64), min(f64), avg(f64), ws, we, ts]
+    ints = [0, 1, 2, 6, 7, 8]
+    gi = np.lexsort(tuple(got[i] for i in reversed(ints)))
+    wi = np.lexsort(tuple(want[i] for i in reversed(ints)))
+    for i in ints:
+        assert np.array_equal(got[i][gi], want[i][wi]), i
+    for i in (3, 4, 5):
+        np.testing.assert_allclose(got[i][gi], want[i][wi], rtol=1e-9)
+
+
+@pytest.mark.gpu
+def test_checkpoint_under_armed_epoch():
+    """checkpoint_drain while an epoch is armed must drain the correct
+    open-pane state; the armed epoch folds afterwards with unchanged
+    emissions (the snapshot predates the checkpoint, which mutates no
+    pane)."""
+    from arroyo_amd import gpu
+
+    keys, val, ts = gen(n=60_000, nk=2)
+    kw = cfg(2)
+    g = gpu.make_op(kw)
+    batches = batches_from_columns([*keys, val, ts], 8192)
+    half = len(batches) // 2
+    outs = []
+    for b in batches[:half]:
+        g.process_batch(b)
+    wm = int(batches[half - 1][-1][-1]) - NS
+    g.mark_epoch()
+    g.set_filter_watermark(wm)
+    drained = g.checkpoint_drain()          # with the epoch still armed
+    assert drained is not None
+    out = g.handle_watermarks_epoch([wm])   # fold after the drain
+    if out and len(out[0]):
+        outs.append(out)
+    for b in batches[half:]:
+        g.process_batch(b)
+        w2 = int(b[-1][-1]) - NS
+        if w2 > wm:
+            out = g.handle_watermark(w2)
+            wm = w2
+            if out and len(out[0]):
+                outs.append(out)
+    got = concat_outputs(outs)
+    g.close()
+
+    o = oracle.make_op(cfg(2))
+    from arroyo_amd.pipeline import WatermarkGen
+    outs_o = []
+    wm = None
+    wmo = int(batches[half - 1][-1][-1]) - NS
+    for i, b in enumerate(batches):
+        o.process_batch(b)
+        if i == half - 1:
+            out = o.handle_watermark(wmo)
+            wm = wmo
+            if out and len(out[0]):
+                outs_o.append(out)
+        elif i >= half:
+            w2 = int(b[-1][-1]) - NS
+            if w2 > wm:
+                out = o.handle_watermark(w2)
+                wm = w2
+                if out and len(out[0]):
+                    outs_o.append(out)
+    want = concat_outputs(outs_o)
+    o.close()
+    assert (got is None) == (want is None)
+    if got is None:
+        return
+    assert len(got[0]) == len(want[0])
+    gi = np.lexsort(tuple(got[::-1]))
+    wi = np.lexsort(tuple(want[::-1]))
+    for gc, wc in zip(got, want):
+        assert np.array_equal(gc[gi], wc[wi])
