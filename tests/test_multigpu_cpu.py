@@ -27,7 +27,8 @@ from arroyo_amd.shuffle import shuffle_columns
 WORLD = 2
 
 
-def _rank_main(rank, world, port, result_q, op_kind="oracle"):
+def _rank_main(rank, world, port, result_q, op_kind="oracle",
+               fused=False):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -51,23 +52,52 @@ def _rank_main(rank, world, port, result_q, op_kind="oracle"):
     wg = WatermarkGen(NS)
     n = len(ts)
     bsz = 8192
-    for lo in range(0, n, bsz):
-        bk, bt = key[lo:lo + bsz], ts[lo:lo + bsz]
-        # this rank's slice of the upstream stream (round-robin split,
-        # mirroring parallel source subtasks)
-        mine = np.arange(len(bk)) % world == rank
-        bk, bt = bk[mine], bt[mine]
-        # keyed shuffle: all-to-all by key-hash range owner
-        mk, mt = shuffle_columns([bk, bt], world)
-        if len(mk):
-            op.process_batch([mk, mt])
-        # watermark: min across upstream partitions (WatermarkHolder,
-        # context.rs:63-86) -- here all ranks see the same source clock
-        wm = wg.on_batch(ts[lo:lo + bsz])
-        if wm is not None:
-            out = op.handle_watermark(wm)
-            if out and len(out[0]):
-                outs.append(out)
+    if fused:
+        # the bench's period-fused cadence: K batches' local slices are
+        # concatenated, exchanged in ONE shuffle round and submitted as
+        # one batch, then the period's watermarks emit batched.  Output
+        # equality with the per-batch cadence is the same late-filter
+        # argument as watermark fusion (the stream is time-monotone).
+        FUSE = 4
+        acc_k, acc_t, pend_wm = [], [], []
+        for i, lo in enumerate(range(0, n, bsz)):
+            bk, bt = key[lo:lo + bsz], ts[lo:lo + bsz]
+            mine = np.arange(len(bk)) % world == rank
+            acc_k.append(bk[mine])
+            acc_t.append(bt[mine])
+            wm = wg.on_batch(ts[lo:lo + bsz])
+            if wm is not None:
+                pend_wm.append(wm)
+            last = lo + bsz >= n
+            if len(acc_k) == FUSE or last:
+                mk, mt = shuffle_columns([np.concatenate(acc_k),
+                                          np.concatenate(acc_t)], world)
+                if len(mk):
+                    op.process_batch([mk, mt])
+                acc_k, acc_t = [], []
+                if pend_wm:
+                    out = op.handle_watermarks(pend_wm)
+                    pend_wm = []
+                    if out and len(out[0]):
+                        outs.append(out)
+    else:
+        for lo in range(0, n, bsz):
+            bk, bt = key[lo:lo + bsz], ts[lo:lo + bsz]
+            # this rank's slice of the upstream stream (round-robin split,
+            # mirroring parallel source subtasks)
+            mine = np.arange(len(bk)) % world == rank
+            bk, bt = bk[mine], bt[mine]
+            # keyed shuffle: all-to-all by key-hash range owner
+            mk, mt = shuffle_columns([bk, bt], world)
+            if len(mk):
+                op.process_batch([mk, mt])
+            # watermark: min across upstream partitions (WatermarkHolder,
+            # context.rs:63-86) -- here all ranks see the same source clock
+            wm = wg.on_batch(ts[lo:lo + bsz])
+            if wm is not None:
+                out = op.handle_watermark(wm)
+                if out and len(out[0]):
+                    outs.append(out)
     out = op.handle_watermark(U64MAX)
     if out and len(out[0]):
         outs.append(out)
@@ -81,10 +111,11 @@ def _rank_main(rank, world, port, result_q, op_kind="oracle"):
     dist.destroy_process_group()
 
 
-def _run_world(world, port, op_kind="oracle"):
+def _run_world(world, port, op_kind="oracle", fused=False):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main, args=(r, world, port, q, op_kind))
+    procs = [ctx.Process(target=_rank_main,
+                         args=(r, world, port, q, op_kind, fused))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -142,3 +173,15 @@ def test_two_rank_shuffle_hip_ops_match_single():
 @pytest.mark.gpu
 def test_four_rank_shuffle_hip_ops_match_single():
     _run_world(4, 29389, op_kind="gpu")
+
+
+def test_two_rank_period_fused_exchange_matches_single():
+    """bench.py's N>1 cadence (one exchange + one submission + batched
+    watermarks per fused period) through the composed dataflow: merged
+    emissions still equal the single-instance run."""
+    _run_world(2, 29389, fused=True)
+
+
+@pytest.mark.gpu
+def test_two_rank_period_fused_hip_ops_match_single():
+    _run_world(2, 29393, op_kind="gpu", fused=True)
